@@ -1,0 +1,442 @@
+// ORACLE — test infrastructure only (see util.h header note).
+// CPU restatement of the CompactionIterator merge pipeline:
+//   UnfilteredPartitionIterators.merge (UnfilteredPartitionIterators.java:123-218)
+//   UnfilteredRowIterators.merge      (UnfilteredRowIterators.java:400-599)
+//   Row.Merger / ColumnDataReducer    (Row.java:694-860)
+//   Cells.reconcile                   (Cells.java:68-119)
+//   RangeTombstoneMarker.Merger       (RangeTombstoneMarker.java:72-198)
+//   PurgeFunction                     (PurgeFunction.java:26-145)
+#include "compact.h"
+#include <algorithm>
+#include <functional>
+
+namespace oracle {
+
+// ---------------------------------------------------------------------------
+// comparator (ClusteringComparator.java:140-158 + ClusteringPrefix.Kind)
+// ---------------------------------------------------------------------------
+static int kind_comparison(BoundKind k) {
+    static const int tbl[8] = {0, 0, 0, 1, 2, 3, 3, 3};  // Kind(comparison, …) ctor args
+    return tbl[k];
+}
+static int kind_compared_to_clustering(BoundKind k) {
+    static const int tbl[8] = {-1, -1, -1, -1, 0, 1, 1, 1};
+    return tbl[k];
+}
+static int compare_component(const Header& h, size_t i, const ClusterVal& a, const ClusterVal& b) {
+    if (a.state == ClusterVal::NUL) return b.state == ClusterVal::NUL ? 0 : -1;
+    if (b.state == ClusterVal::NUL) return 1;
+    const bytes ea, eb;
+    const bytes& va = a.state == ClusterVal::EMPTY ? ea : a.v;
+    const bytes& vb = b.state == ClusterVal::EMPTY ? eb : b.v;
+    return compare_typed(h.clustering_types[i], va, vb);
+}
+int compare_clustering_prefix(const Header& h, BoundKind ka, const Clustering& a,
+                              BoundKind kb, const Clustering& b) {
+    size_t mn = std::min(a.size(), b.size());
+    for (size_t i = 0; i < mn; i++) {
+        int c = compare_component(h, i, a[i], b[i]);
+        if (c) return c;
+    }
+    if (a.size() == b.size()) {
+        int c1 = kind_comparison(ka), c2 = kind_comparison(kb);
+        return c1 < c2 ? -1 : c1 > c2 ? 1 : 0;
+    }
+    return a.size() < b.size() ? kind_compared_to_clustering(ka) : -kind_compared_to_clustering(kb);
+}
+static int compare_unfiltered(const Header& h, const Unfiltered& a, const Unfiltered& b) {
+    BoundKind ka = a.kind == Unfiltered::ROW ? CLUSTERING_K : a.marker.kind;
+    BoundKind kb = b.kind == Unfiltered::ROW ? CLUSTERING_K : b.marker.kind;
+    return compare_clustering_prefix(h, ka, a.clustering(), kb, b.clustering());
+}
+
+// ---------------------------------------------------------------------------
+// Cells.reconcile (Cells.java:68-119)
+// ---------------------------------------------------------------------------
+static const Cell& reconcile_cells(const Cell& left, const Cell& right) {
+    if (left.ts != right.ts) return left.ts > right.ts ? left : right;
+    bool l_dt = left.ldt != LDT_NONE_U32, r_dt = right.ldt != LDT_NONE_U32;
+    if (l_dt | r_dt) {
+        if (l_dt != r_dt) return l_dt ? left : right;
+        bool l_tomb = !left.expiring(), r_tomb = !right.expiring();
+        if (l_tomb != r_tomb) return l_tomb ? left : right;
+        if (left.ldt != right.ldt)
+            return ldt_to_long(left.ldt) > ldt_to_long(right.ldt) ? left : right;
+    }
+    // compareValues: unsigned lexicographic on value bytes (ValueAccessor.compare)
+    size_t n = std::min(left.value.size(), right.value.size());
+    int c = memcmp(left.value.data(), right.value.data(), n);
+    if (c == 0) c = left.value.size() == right.value.size() ? 0 : (left.value.size() < right.value.size() ? -1 : 1);
+    return c >= 0 ? left : right;
+}
+
+// ---------------------------------------------------------------------------
+// Row.Merger.merge (Row.java:730-781) — simple columns only
+// ---------------------------------------------------------------------------
+static bool row_merge(const std::vector<const Row*>& versions, const DeletionTime& active_deletion,
+                      size_t ncols, Row& out) {
+    // returns false if merged row is null (fully shadowed/empty)
+    int present = 0;
+    const Row* last = nullptr;
+    for (auto* r : versions) if (r) { present++; last = r; }
+    if (present == 1 && active_deletion.live()) { out = *last; return true; }
+
+    LivenessInfo info;          // LivenessInfo.EMPTY
+    DeletionTime row_del;       // Deletion.LIVE
+    const Clustering* clust = nullptr;
+    for (auto* r : versions) {
+        if (!r) continue;
+        clust = &r->clustering;
+        if (r->live.supersedes(info)) info = r->live;
+        if (r->del.supersedes(row_del)) row_del = r->del;
+    }
+    // (rowDeletion.isShadowedBy(rowInfo) only applies to deprecated shadowable
+    //  deletions — always false for oa-era data, Row.java:430-433)
+    DeletionTime active = active_deletion;
+    if (row_del.supersedes(active)) active = row_del;
+    else row_del = DT_LIVE;
+    if (active.deletes(info.ts)) info = LivenessInfo{};
+
+    out = Row{};
+    out.clustering = clust ? *clust : Clustering{};
+    out.live = info;
+    out.del = row_del;
+    out.cells.assign(ncols, std::nullopt);
+    bool any_cell = false;
+    for (size_t ci = 0; ci < ncols; ci++) {
+        const Cell* merged = nullptr;
+        for (auto* r : versions) {  // version order == source order (reduce call order)
+            if (!r || ci >= r->cells.size() || !r->cells[ci]) continue;
+            const Cell& cell = *r->cells[ci];
+            if (active.deletes(cell.ts)) continue;  // ColumnDataReducer.getReduced (Row.java:~840)
+            merged = merged == nullptr ? &cell : &reconcile_cells(*merged, cell);
+        }
+        if (merged) { out.cells[ci] = *merged; any_cell = true; }
+    }
+    if (info.empty() && row_del.live() && !any_cell) return false;  // null row
+    return true;
+}
+
+// ---------------------------------------------------------------------------
+// RangeTombstoneMarker.Merger (RangeTombstoneMarker.java:72-198)
+// ---------------------------------------------------------------------------
+struct MarkerMerger {
+    DeletionTime partition_deletion;
+    std::vector<std::optional<DeletionTime>> open_markers;  // per source
+    int biggest = -1;
+
+    explicit MarkerMerger(size_t k, const DeletionTime& pd)
+        : partition_deletion(pd), open_markers(k) {}
+
+    DeletionTime current_open_in_merged() const {
+        if (biggest < 0) return DT_LIVE;
+        const DeletionTime& d = *open_markers[biggest];
+        return !d.supersedes(partition_deletion) ? DT_LIVE : d;
+    }
+    DeletionTime active_deletion() const {
+        DeletionTime om = current_open_in_merged();
+        return om.live() ? partition_deletion : om;
+    }
+    // versions: per-source marker at this position (or null). Returns merged
+    // marker or nullopt (no marker emitted).
+    std::optional<Marker> merge(const std::vector<const Marker*>& versions) {
+        DeletionTime prev = current_open_in_merged();
+        const Marker* bound_src = nullptr;
+        for (size_t i = 0; i < versions.size(); i++) {
+            const Marker* m = versions[i];
+            if (!m) continue;
+            bound_src = m;  // `bound` = last add()ed marker's clustering
+            if (m->open(false)) open_markers[i] = m->open_dt();
+            else open_markers[i] = std::nullopt;
+        }
+        biggest = -1;
+        for (size_t i = 0; i < open_markers.size(); i++)
+            if (open_markers[i] && (biggest < 0 || open_markers[i]->supersedes(*open_markers[biggest])))
+                biggest = (int)i;
+        DeletionTime next = current_open_in_merged();
+        if (prev == next) return std::nullopt;
+        bool before_clustering = kind_compared_to_clustering(bound_src->kind) < 0;
+        Marker m;
+        m.values = bound_src->values;
+        if (prev.live()) {
+            m.kind = before_clustering ? INCL_START : EXCL_START;
+            m.end_dt = next;
+        } else if (next.live()) {
+            m.kind = before_clustering ? EXCL_END : INCL_END;
+            m.end_dt = prev;
+        } else {
+            m.kind = before_clustering ? EXCL_END_INCL_START : INCL_END_EXCL_START;
+            m.end_dt = prev;
+            m.start_dt = next;
+        }
+        return m;
+    }
+};
+
+// ---------------------------------------------------------------------------
+// partition-version merge (UnfilteredRowIterators.merge)
+// ---------------------------------------------------------------------------
+Partition merge_partition_versions(const std::vector<const Partition*>& versions, const Header& h) {
+    if (versions.size() == 1) return *versions[0];  // merge(List size 1) returns as-is
+    Partition out;
+    out.key = versions[0]->key;
+    out.token = versions[0]->token;
+    // collectPartitionLevelDeletion (UnfilteredRowIterators.java:465-482)
+    DeletionTime del;
+    for (auto* p : versions)
+        if (!del.supersedes(p->del)) del = p->del;
+    out.del = del;
+
+    size_t k = versions.size();
+    std::vector<size_t> pos(k, 0);
+    MarkerMerger marker_merger(k, del);
+    size_t ncols = h.regular_cols.size();
+    while (true) {
+        // find min position among streams
+        int min_src = -1;
+        for (size_t i = 0; i < k; i++) {
+            if (pos[i] >= versions[i]->items.size()) continue;
+            if (min_src < 0 ||
+                compare_unfiltered(h, versions[i]->items[pos[i]], versions[min_src]->items[pos[min_src]]) < 0)
+                min_src = (int)i;
+        }
+        if (min_src < 0) break;
+        const Unfiltered& first = versions[min_src]->items[pos[min_src]];
+        // gather all sources equal to min — reduce() call order is source order
+        std::vector<const Row*> row_versions(k, nullptr);
+        std::vector<const Marker*> marker_versions(k, nullptr);
+        bool is_row = first.kind == Unfiltered::ROW;
+        for (size_t i = 0; i < k; i++) {
+            if (pos[i] >= versions[i]->items.size()) continue;
+            const Unfiltered& u = versions[i]->items[pos[i]];
+            if (compare_unfiltered(h, u, first) != 0) continue;
+            // same position: rows group with rows, markers with markers (kind
+            // comparison separates CLUSTERING from bounds)
+            if (u.kind == Unfiltered::ROW) row_versions[i] = &u.row;
+            else marker_versions[i] = &u.marker;
+            pos[i]++;
+        }
+        if (is_row) {
+            Row merged;
+            if (row_merge({row_versions.begin(), row_versions.end()}, marker_merger.active_deletion(), ncols, merged)) {
+                Unfiltered u;
+                u.kind = Unfiltered::ROW;
+                u.row = std::move(merged);
+                out.items.push_back(std::move(u));
+            }
+        } else {
+            auto m = marker_merger.merge({marker_versions.begin(), marker_versions.end()});
+            if (m) {
+                Unfiltered u;
+                u.kind = Unfiltered::MARKER;
+                u.marker = *m;
+                out.items.push_back(std::move(u));
+            }
+        }
+    }
+    return out;
+}
+
+// ---------------------------------------------------------------------------
+// purge (PurgeFunction.java:26-145, BTreeRow.purge:457-470, AbstractCell.purge:78-99)
+// ---------------------------------------------------------------------------
+struct Purger {
+    int64_t now, gc_before;
+    bool never_purge;
+    const std::vector<PurgeRange>* overlaps;
+    int64_t token;
+
+    // purgeEvaluator(key).test(ts) (CompactionController.java:247-286)
+    bool evaluator(int64_t ts) const {
+        int64_t min_ts = INT64_MAX;
+        bool has = false;
+        for (const auto& r : *overlaps)
+            if (token >= r.tok_lo && token <= r.tok_hi) { has = true; min_ts = std::min(min_ts, r.min_ts); }
+        return !has || ts < min_ts;
+    }
+    bool should_purge(int64_t ts, int64_t ldt) const {  // the DeletionPurger lambda (PurgeFunction.java:40-44)
+        if (never_purge) return false;
+        return ldt < gc_before && evaluator(ts);
+    }
+    bool should_purge(const DeletionTime& dt) const {
+        return !dt.live() && should_purge(dt.mfda, ldt_to_long(dt.ldt));
+    }
+    bool should_purge_liveness(const LivenessInfo& l) const {  // DeletionPurger.shouldPurge(liveness, now)
+        bool is_live = !l.empty() && !l.expired() && (!l.expiring() || now < l.let);
+        return !is_live && should_purge(l.ts, l.let);
+    }
+};
+
+static std::optional<Cell> purge_cell(const Cell& c, const Purger& pg) {
+    // AbstractCell.purge (AbstractCell.java:78-99)
+    if (!c.is_live(pg.now)) {
+        if (pg.should_purge(c.ts, ldt_to_long(c.ldt))) return std::nullopt;
+        if (c.expiring()) {
+            // convert expired cell to tombstone: ldt = localDeletionTime - ttl, then purge again
+            Cell t;
+            t.ts = c.ts;
+            t.ldt = ldt_to_u32(ldt_to_long(c.ldt) - c.ttl);
+            t.ttl = NO_TTL;
+            // value dropped (BufferCell.tombstone has empty value)
+            return purge_cell(t, pg);
+        }
+    }
+    return c;
+}
+
+static bool purge_row(Row& r, const Purger& pg, bool enforce_strict_liveness) {
+    // BTreeRow.purge (BTreeRow.java:457-470); unconditional application is
+    // output-equivalent to the hasDeletion(nowInSec) fast path.
+    if (pg.should_purge_liveness(r.live)) r.live = LivenessInfo{};
+    if (pg.should_purge(r.del)) r.del = DeletionTime{};
+    if (enforce_strict_liveness && r.del.live() && r.live.empty()) return false;
+    bool any = false;
+    for (auto& oc : r.cells) {
+        if (!oc) continue;
+        oc = purge_cell(*oc, pg);
+        if (oc) any = true;
+    }
+    return !(r.live.empty() && r.del.live() && !any);  // empty row -> null
+}
+
+bool purge_partition(Partition& p, int64_t now_sec, int64_t gc_before, bool never_purge,
+                     const std::vector<PurgeRange>& overlaps, bool enforce_strict_liveness) {
+    Purger pg{now_sec, gc_before, never_purge, &overlaps, p.token};
+    if (pg.should_purge(p.del)) p.del = DT_LIVE;
+    std::vector<Unfiltered> kept;
+    for (auto& u : p.items) {
+        if (u.kind == Unfiltered::ROW) {
+            if (purge_row(u.row, pg, enforce_strict_liveness)) kept.push_back(std::move(u));
+        } else {
+            // PurgeFunction.applyToMarker (PurgeFunction.java:113-144), reversed=false
+            Marker& m = u.marker;
+            if (m.boundary()) {
+                bool purge_close = pg.should_purge(m.close_dt());
+                bool purge_open = pg.should_purge(m.open_dt());
+                if (purge_close && purge_open) continue;
+                if (purge_close) {
+                    // boundary -> corresponding open marker
+                    Marker o;
+                    o.kind = m.kind == EXCL_END_INCL_START ? INCL_START : EXCL_START;
+                    o.values = m.values;
+                    o.end_dt = m.start_dt;
+                    u.marker = o;
+                } else if (purge_open) {
+                    Marker c;
+                    c.kind = m.kind == EXCL_END_INCL_START ? EXCL_END : INCL_END;
+                    c.values = m.values;
+                    c.end_dt = m.end_dt;
+                    u.marker = c;
+                }
+                kept.push_back(std::move(u));
+            } else {
+                if (!pg.should_purge(m.end_dt)) kept.push_back(std::move(u));
+            }
+        }
+    }
+    p.items = std::move(kept);
+    return !(p.del.live() && p.items.empty());
+}
+
+// ---------------------------------------------------------------------------
+// SerializationHeader.make (SerializationHeader.java:77-106)
+// ---------------------------------------------------------------------------
+static Header make_output_header(const std::vector<SSTable>& inputs) {
+    std::vector<const SSTable*> ordered;
+    for (auto& t : inputs) ordered.push_back(&t);
+    std::stable_sort(ordered.begin(), ordered.end(),
+                     [](const SSTable* a, const SSTable* b) { return a->generation > b->generation; });
+    // EncodingStats.Collector over StatsMetadata mins, descending generation
+    bool ts_set = false, ldt_set = false, ttl_set = false;
+    int64_t min_ts = INT64_MAX, min_ldt = INT64_MAX;
+    int32_t min_ttl = INT32_MAX;
+    Header h;
+    bool first = true;
+    for (const SSTable* t : ordered) {
+        ts_set = true; min_ts = std::min(min_ts, t->stats.min_timestamp);
+        ldt_set = true; min_ldt = std::min(min_ldt, t->stats.min_ldt);
+        ttl_set = true; min_ttl = std::min(min_ttl, t->stats.min_ttl);
+        if (first) {
+            h.key_type = t->header.key_type;
+            h.clustering_types = t->header.clustering_types;
+            first = false;
+        }
+        // columns.addAll — union preserving sorted-by-name Columns order
+        for (auto cols_sel : {0, 1}) {
+            auto& dst = cols_sel ? h.regular_cols : h.static_cols;
+            auto& src = cols_sel ? t->header.regular_cols : t->header.static_cols;
+            for (auto& c : src) {
+                bool found = false;
+                for (auto& d : dst) if (d.first == c.first) { found = true; break; }
+                if (!found) dst.push_back(c);
+            }
+        }
+    }
+    std::sort(h.regular_cols.begin(), h.regular_cols.end(),
+              [](auto& a, auto& b) { return a.first < b.first; });
+    std::sort(h.static_cols.begin(), h.static_cols.end(),
+              [](auto& a, auto& b) { return a.first < b.first; });
+    // EncodingStats ctor epoch mapping (EncodingStats.java:78-89)
+    h.stats.min_ts = (!ts_set || min_ts == NO_TIMESTAMP) ? TIMESTAMP_EPOCH : min_ts;
+    h.stats.min_ldt = (!ldt_set || min_ldt == NO_EXPIRATION_TIME) ? DELETION_TIME_EPOCH : min_ldt;
+    h.stats.min_ttl = !ttl_set ? 0 : min_ttl;
+    return h;
+}
+
+// ---------------------------------------------------------------------------
+// top-level compaction (CompactionTask.runMayThrow hot loop semantics)
+// ---------------------------------------------------------------------------
+CompactionResult compact(const CompactionJob& job) {
+    CompactionResult res;
+    res.out.header = make_output_header(job.inputs);
+    res.out.comp = job.inputs.empty() ? CompressionParams{} : job.inputs[0].comp;
+    res.merged_partition_counts.assign(job.inputs.size(), 0);
+
+    size_t k = job.inputs.size();
+    std::vector<size_t> pos(k, 0);
+    while (true) {
+        int min_src = -1;
+        for (size_t i = 0; i < k; i++) {
+            // shard filter: skip partitions outside our token range
+            while (pos[i] < job.inputs[i].parts.size() && job.has_shard) {
+                int64_t tok = job.inputs[i].parts[pos[i]].token;
+                if (tok < job.shard_lo || tok > job.shard_hi) pos[i]++;
+                else break;
+            }
+            if (pos[i] >= job.inputs[i].parts.size()) continue;
+            const Partition& p = job.inputs[i].parts[pos[i]];
+            if (min_src < 0) { min_src = (int)i; continue; }
+            const Partition& m = job.inputs[min_src].parts[pos[min_src]];
+            if (compare_decorated_key(p.token, p.key.data(), p.key.size(),
+                                      m.token, m.key.data(), m.key.size()) < 0)
+                min_src = (int)i;
+        }
+        if (min_src < 0) break;
+        const Partition& first = job.inputs[min_src].parts[pos[min_src]];
+        std::vector<const Partition*> versions;
+        for (size_t i = 0; i < k; i++) {
+            if (pos[i] >= job.inputs[i].parts.size()) continue;
+            const Partition& p = job.inputs[i].parts[pos[i]];
+            if (compare_decorated_key(p.token, p.key.data(), p.key.size(),
+                                      first.token, first.key.data(), first.key.size()) == 0) {
+                versions.push_back(&p);
+                pos[i]++;
+            }
+        }
+        res.partitions_in += versions.size();
+        res.merged_partition_counts[versions.size() - 1]++;
+        for (auto* v : versions)
+            for (auto& u : v->items) if (u.kind == Unfiltered::ROW) res.rows_in++;
+
+        Partition merged = merge_partition_versions(versions, res.out.header);
+        if (purge_partition(merged, job.now_sec, job.gc_before, job.never_purge,
+                            job.overlaps, job.enforce_strict_liveness)) {
+            res.partitions_out++;
+            for (auto& u : merged.items) if (u.kind == Unfiltered::ROW) res.rows_out++;
+            res.out.parts.push_back(std::move(merged));
+        }
+    }
+    return res;
+}
+
+}  // namespace oracle

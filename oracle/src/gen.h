@@ -1,0 +1,67 @@
+// ORACLE + shared generator CONTRACT.
+// Deterministic synthetic sstable content derived from (seed, sstable index).
+// The GPU product implements the SAME derivation (cassandra_amd/csrc/gen.hip);
+// both sides must produce identical logical rows so parity tests can compare
+// oracle-written and GPU-written sstables byte for byte.
+#pragma once
+#include "sstable.h"
+
+namespace oracle {
+
+struct GenSpec {
+    uint64_t seed = 42;
+    uint32_t n_sstables = 4;
+    uint64_t rows_per_sstable = 1000;
+    uint64_t key_universe = 0;     // 0 => 10x total rows (low overlap)
+    uint32_t value_len = 1024;
+    uint32_t value_repeat_pct = 55;  // P(8-byte word repeats previous) -> LZ4 ratio knob
+    uint32_t tombstone_pct = 0;      // percent of rows that are row-deletions
+    uint32_t partition_del_pct = 0;  // percent of partitions with partition deletion
+    int64_t base_ts = 1700000000000000LL;   // µs
+    int64_t base_ldt = 1700000000LL;        // seconds
+    uint64_t first_generation = 1;
+};
+
+// logical row for one (sstable, slot)
+struct GenRow {
+    uint64_t key_id;
+    int64_t ts;
+    bool is_tombstone;
+    uint32_t ldt;     // for tombstones (seconds)
+};
+
+// Derivations (all splitmix64-based; GPU mirrors these exactly):
+inline uint64_t gen_key_id(const GenSpec& g, uint32_t sst, uint64_t j) {
+    return splitmix64(g.seed ^ (0x5EEDULL + sst) * 0x9E3779B97f4A7C15ULL ^ (j * 0xA24BAED4963EE407ULL)) % g.key_universe;
+}
+inline int64_t gen_ts(const GenSpec& g, uint32_t sst, uint64_t key_id) {
+    return g.base_ts + (int64_t)(splitmix64(g.seed ^ key_id * 31 ^ ((uint64_t)sst << 48)) % 1000000000ULL);
+}
+inline bool gen_is_tombstone(const GenSpec& g, uint32_t sst, uint64_t key_id) {
+    if (g.tombstone_pct == 0) return false;
+    return splitmix64(g.seed ^ 0xDEADULL ^ key_id ^ ((uint64_t)sst << 32)) % 100 < g.tombstone_pct;
+}
+inline bool gen_has_partition_del(const GenSpec& g, uint32_t sst, uint64_t key_id) {
+    if (g.partition_del_pct == 0) return false;
+    return splitmix64(g.seed ^ 0xFEEDULL ^ key_id ^ ((uint64_t)sst << 32)) % 100 < g.partition_del_pct;
+}
+// value: value_len bytes in 8-byte words; word w repeats word w-1 with prob repeat_pct
+inline void gen_value(const GenSpec& g, uint32_t sst, uint64_t key_id, bytes& out) {
+    out.resize(g.value_len);
+    uint64_t state = g.seed ^ key_id * 0x100000001B3ULL ^ ((uint64_t)sst << 40);
+    uint64_t prev = splitmix64(state);
+    size_t nw = (g.value_len + 7) / 8;
+    for (size_t w = 0; w < nw; w++) {
+        uint64_t r = splitmix64(state + 1 + w);
+        uint64_t word = (r % 100 < g.value_repeat_pct && w > 0) ? prev : splitmix64(r);
+        prev = word;
+        size_t off = w * 8;
+        for (size_t b = 0; b < 8 && off + b < g.value_len; b++)
+            out[off + b] = (uint8_t)(word >> (8 * b));
+    }
+}
+
+// build one full synthetic sstable (sorted, deduped keys)
+SSTable generate_sstable(const GenSpec& g, uint32_t sst_index);
+
+}  // namespace oracle

@@ -1,0 +1,192 @@
+// ORACLE CLI — test infrastructure only (see util.h header note).
+// Subcommands: roundtrip, dump, gen, compact, selftest, bench-compact.
+#include "sstable.h"
+#include "compact.h"
+#include "gen.h"
+#include <chrono>
+#include <cstring>
+#include <map>
+#include <string>
+
+using namespace oracle;
+
+static std::map<std::string, std::string> parse_kv(int argc, char** argv, int start,
+                                                   std::vector<std::string>& positional) {
+    std::map<std::string, std::string> kv;
+    for (int i = start; i < argc; i++) {
+        std::string a = argv[i];
+        auto eq = a.find('=');
+        if (eq == std::string::npos) positional.push_back(a);
+        else kv[a.substr(0, eq)] = a.substr(eq + 1);
+    }
+    return kv;
+}
+
+static int cmd_dump(const std::string& base) {
+    SSTable t = read_sstable(base, true);
+    printf("sstable %s\n  partitions=%zu uncompressed=%zu\n", base.c_str(), t.parts.size(),
+           t.raw_data_uncompressed.size());
+    printf("  header: key=%s clustering=%zu regulars=%zu stats(minTs=%lld minLdt=%lld minTTL=%d)\n",
+           cql_type_name(t.header.key_type), t.header.clustering_types.size(),
+           t.header.regular_cols.size(), (long long)t.header.stats.min_ts,
+           (long long)t.header.stats.min_ldt, t.header.stats.min_ttl);
+    size_t show = std::min<size_t>(5, t.parts.size());
+    for (size_t i = 0; i < show; i++) {
+        auto& p = t.parts[i];
+        printf("  part[%zu] token=%lld keylen=%zu del=%s items=%zu\n", i, (long long)p.token,
+               p.key.size(), p.del.live() ? "live" : "set", p.items.size());
+    }
+    return 0;
+}
+
+static int cmd_roundtrip(const std::string& base) {
+    SSTable t = read_sstable(base, true);
+    // re-serialize Data stream with the header read from the file. Fixtures were
+    // written with the reference test config's column_index_size: 4KiB
+    // (test/conf/cassandra.yaml:24).
+    bytes data, index;
+    for (auto& p : t.parts) serialize_partition(p, t.header, data, index, 4096);
+    int rc = 0;
+    auto check = [&](const char* what, const bytes& got, const bytes& want) {
+        if (got == want) { printf("OK  %-18s %zu bytes\n", what, want.size()); return; }
+        rc = 1;
+        size_t n = std::min(got.size(), want.size()), i = 0;
+        while (i < n && got[i] == want[i]) i++;
+        printf("FAIL %-18s got=%zu want=%zu first_diff=%zu (got=%02x want=%02x)\n", what,
+               got.size(), want.size(), i, i < got.size() ? got[i] : 0, i < want.size() ? want[i] : 0);
+    };
+    check("Data(uncompressed)", data, t.raw_data_uncompressed);
+    ChunkedOut co = chunk_compress(data, t.comp);
+    check("Data.db", co.file, read_file(base + "-Data.db"));
+    check("CompressionInfo.db", make_compression_info(t.comp, data.size(), co.offsets),
+          read_file(base + "-CompressionInfo.db"));
+    check("Index.db", index, read_file(base + "-Index.db"));
+    if (getenv("ORACLE_DUMP_INDEX")) write_file(std::string(getenv("ORACLE_DUMP_INDEX")), index);
+    {
+        std::string d = std::to_string(crc32(co.file.data(), co.file.size()));
+        check("Digest.crc32", bytes(d.begin(), d.end()), read_file(base + "-Digest.crc32"));
+    }
+    {
+        // Filter.db: recompute bloom from keys
+        SSTable tmp;
+        tmp.header = t.header;
+        tmp.comp = t.comp;
+        tmp.parts = t.parts;
+        WriterOut w = write_sstable(tmp);
+        check("Filter.db", w.filter, read_file(base + "-Filter.db"));
+    }
+    return rc;
+}
+
+static GenSpec spec_from_kv(std::map<std::string, std::string>& kv) {
+    GenSpec g;
+    auto geti = [&](const char* k, auto def) -> int64_t {
+        return kv.count(k) ? (int64_t)strtoll(kv[k].c_str(), nullptr, 10) : (int64_t)def;
+    };
+    g.seed = geti("seed", g.seed);
+    g.n_sstables = (uint32_t)geti("n", g.n_sstables);
+    g.rows_per_sstable = geti("rows", g.rows_per_sstable);
+    g.key_universe = geti("universe", g.key_universe);
+    g.value_len = (uint32_t)geti("vlen", g.value_len);
+    g.value_repeat_pct = (uint32_t)geti("vrep", g.value_repeat_pct);
+    g.tombstone_pct = (uint32_t)geti("tomb", g.tombstone_pct);
+    g.partition_del_pct = (uint32_t)geti("pdel", g.partition_del_pct);
+    g.first_generation = geti("gen0", g.first_generation);
+    return g;
+}
+
+static int cmd_gen(const std::string& outdir, std::map<std::string, std::string>& kv) {
+    GenSpec g = spec_from_kv(kv);
+    uint64_t total_unc = 0;
+    for (uint32_t s = 0; s < g.n_sstables; s++) {
+        SSTable t = generate_sstable(g, s);
+        WriterOut w = write_sstable(t);
+        std::string base = outdir + "/oa-" + std::to_string(t.generation) + "-big";
+        write_components(w, base);
+        total_unc += w.uncompressed_data_len;
+        printf("wrote %s: parts=%llu uncompressed=%llu compressed=%zu\n", base.c_str(),
+               (unsigned long long)w.partition_count, (unsigned long long)w.uncompressed_data_len,
+               w.data_db.size());
+    }
+    printf("total_uncompressed=%llu\n", (unsigned long long)total_unc);
+    return 0;
+}
+
+static int cmd_compact(const std::string& outbase, std::vector<std::string>& inputs,
+                       std::map<std::string, std::string>& kv) {
+    using clk = std::chrono::steady_clock;
+    CompactionJob job;
+    auto t0 = clk::now();
+    uint64_t input_unc = 0;
+    for (auto& in : inputs) {
+        job.inputs.push_back(read_sstable(in, true));
+        input_unc += job.inputs.back().raw_data_uncompressed.size();
+        job.inputs.back().raw_data_uncompressed.clear();
+        job.inputs.back().raw_statistics.clear();
+    }
+    job.now_sec = kv.count("now") ? strtoll(kv["now"].c_str(), nullptr, 10) : 1800000000LL;
+    job.gc_before = kv.count("gcbefore") ? strtoll(kv["gcbefore"].c_str(), nullptr, 10) : INT64_MIN;
+    job.never_purge = kv.count("nevergc") && kv["nevergc"] == "1";
+    if (kv.count("shard")) {
+        auto s = kv["shard"];
+        auto c = s.find(':');
+        job.has_shard = true;
+        job.shard_lo = strtoll(s.substr(0, c).c_str(), nullptr, 10);
+        job.shard_hi = strtoll(s.substr(c + 1).c_str(), nullptr, 10);
+    }
+    auto t1 = clk::now();
+    CompactionResult res = compact(job);
+    auto t2 = clk::now();
+    WriterOut w = write_sstable(res.out);
+    write_components(w, outbase);
+    auto t3 = clk::now();
+    double read_s = std::chrono::duration<double>(t1 - t0).count();
+    double merge_s = std::chrono::duration<double>(t2 - t1).count();
+    double write_s = std::chrono::duration<double>(t3 - t2).count();
+    printf("{\"input_uncompressed_bytes\": %llu, \"partitions_in\": %llu, \"partitions_out\": %llu, "
+           "\"rows_in\": %llu, \"rows_out\": %llu, \"read_s\": %.3f, \"merge_s\": %.3f, "
+           "\"write_s\": %.3f, \"total_s\": %.3f, \"mb_per_s\": %.2f}\n",
+           (unsigned long long)input_unc, (unsigned long long)res.partitions_in,
+           (unsigned long long)res.partitions_out, (unsigned long long)res.rows_in,
+           (unsigned long long)res.rows_out, read_s, merge_s, write_s, read_s + merge_s + write_s,
+           input_unc / 1e6 / (read_s + merge_s + write_s));
+    return 0;
+}
+
+static int cmd_selftest() {
+    // vint round trips incl. boundaries
+    for (uint64_t v : std::initializer_list<uint64_t>{0, 1, 127, 128, 16383, 16384,
+                       (uint64_t)INT64_MAX, 0xFFFFFFFFFFFFFFFFULL, 0x8000000000000000ULL}) {
+        bytes b;
+        put_unsigned_vint(b, v);
+        Reader r(b);
+        if (read_unsigned_vint(r) != v || r.pos != b.size()) { printf("vint FAIL %llx\n", (unsigned long long)v); return 1; }
+    }
+    // murmur3: empty key handling + known stability (self-consistency)
+    if (murmur3_token(nullptr, 0) != INT64_MIN) { printf("token(empty) FAIL\n"); return 1; }
+    printf("selftest OK\n");
+    return 0;
+}
+
+int main(int argc, char** argv) {
+    if (argc < 2) { fprintf(stderr, "usage: oracle_tool <dump|roundtrip|gen|compact|selftest> ...\n"); return 2; }
+    std::string cmd = argv[1];
+    try {
+        std::vector<std::string> pos;
+        auto kv = parse_kv(argc, argv, 2, pos);
+        if (cmd == "selftest") return cmd_selftest();
+        if (cmd == "dump") return cmd_dump(pos.at(0));
+        if (cmd == "roundtrip") return cmd_roundtrip(pos.at(0));
+        if (cmd == "gen") return cmd_gen(pos.at(0), kv);
+        if (cmd == "compact") {
+            std::string outbase = pos.at(0);
+            std::vector<std::string> ins(pos.begin() + 1, pos.end());
+            return cmd_compact(outbase, ins, kv);
+        }
+    } catch (const std::exception& e) {
+        fprintf(stderr, "error: %s\n", e.what());
+        return 1;
+    }
+    fprintf(stderr, "unknown command %s\n", cmd.c_str());
+    return 2;
+}
