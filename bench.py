@@ -1,0 +1,226 @@
+#!/usr/bin/env python3
+"""Benchmark: SSTable compaction MB/s (input uncompressed bytes merged+compressed).
+
+Driver contract: `python bench.py --gpus N --steps K --warmup W`; for N>1 the
+driver launches via torch.distributed.run (one rank per GPU). A "step" is one
+full compaction job of the C2 workload (BASELINE.json configs[1]): 8 x 2 GiB
+synthetic sstables, 10% key overlap, ~1 KiB values, LZ4 16 KiB chunks.
+Multi-GPU: weak scaling — every rank owns an independent token-disjoint job
+(seeded per rank); no collectives on the data path.
+
+Inputs are generated ON THE GPU by the product write path before the timed
+region; the timed region is gpuc_compact (file read -> H2D -> kernels -> D2H
+-> file write). The cpu_baseline leg times the CPU oracle (kind "port") on a
+bounded sample at N=1.
+"""
+import argparse
+import json
+import os
+import shutil
+import subprocess
+import sys
+import time
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+HBM_PEAK_GBS = 8000.0  # MI355X spec peak (MI355X_MICROARCH.md)
+
+
+def build_if_needed():
+    so = os.path.join(REPO, "cassandra_amd", "libcassandra_gpucompact.so")
+    if not os.path.exists(so):
+        subprocess.run(["hipcc", "--offload-arch=gfx950", "-O3", "-std=c++17", "-fPIC",
+                        "-shared", os.path.join(REPO, "cassandra_amd", "csrc", "gpucompact.cpp"),
+                        "-o", so], check=True)
+
+
+def oracle_bin():
+    p = os.path.join(REPO, "oracle", "bin", "oracle_tool")
+    if not os.path.exists(p):
+        subprocess.run(["make", "-C", os.path.join(REPO, "oracle")], check=True,
+                       stdout=subprocess.DEVNULL)
+    return p
+
+
+def algorithmic_bytes(kernel, r):
+    """Algorithmic HBM bytes per launch of the dominant kernel.
+
+    Values are per ONE invocation of the whole-job kernel launch.
+    """
+    in_comp = r["input_uncompressed_bytes"] * 0  # filled by caller knowing file sizes
+    out_unc = r["output_uncompressed_bytes"]
+    out_comp = r["output_compressed_bytes"]
+    in_unc = r["input_uncompressed_bytes"]
+    if kernel == "k_lz4_decompress":
+        return r["_input_compressed_bytes"] + in_unc
+    if kernel == "k_parse":
+        return r["partitions_in"] * (48 + 77)  # header read + SoA write per partition
+    if kernel == "merge+reconcile":
+        import math
+        k = max(1, r["_n_inputs"])
+        rounds = max(1, math.ceil(math.log2(k)))
+        return r["partitions_in"] * 48 * rounds + r["partitions_in"] * 90
+    if kernel == "k_serialize":
+        return 2 * out_unc  # gather values/meta + write serialized stream
+    if kernel == "k_lz4_compress":
+        return out_unc + out_comp
+    return in_unc
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=3)
+    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--sstables", type=int, default=8)
+    ap.add_argument("--rows", type=int, default=2_050_000,
+                    help="rows per sstable (2,050,000 * ~1048 B = 2 GiB uncompressed)")
+    ap.add_argument("--vlen", type=int, default=1024)
+    ap.add_argument("--overlap", type=int, default=10)
+    ap.add_argument("--dir", default=os.environ.get("GPUC_BENCH_DIR", "/tmp/gpuc_bench"))
+    ap.add_argument("--no-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    dist = None
+    if world > 1:
+        import torch.distributed as tdist
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        tdist.init_process_group("gloo")  # control-plane only; data path has no collectives
+        dist = tdist
+
+    build_if_needed()
+    import cassandra_amd as ca
+    if ca.device_count() < 1:
+        print(json.dumps({"error": "no GPU visible"}))
+        return 1
+    device = local_rank % max(1, ca.device_count())
+
+    # ---- per-rank inputs (GPU-generated, seed disjoint per rank) ----
+    d = os.path.join(args.dir, f"r{rank}")
+    shutil.rmtree(d, ignore_errors=True)
+    os.makedirs(d, exist_ok=True)
+    t_gen = time.time()
+    ca.generate(d, seed=42 + rank, n_sstables=args.sstables,
+                rows_per_sstable=args.rows, overlap_pct=args.overlap,
+                value_len=args.vlen, value_repeat_pct=55, device=device)
+    t_gen = time.time() - t_gen
+    bases = [os.path.join(d, f"oa-{g}-big") for g in range(1, args.sstables + 1)]
+    input_compressed = sum(os.path.getsize(b + "-Data.db") for b in bases)
+
+    def one_step(i):
+        out = os.path.join(d, f"out-{i}", "oa-100-big")
+        os.makedirs(os.path.dirname(out), exist_ok=True)
+        r = ca.compact(bases, out, device=device)
+        shutil.rmtree(os.path.dirname(out), ignore_errors=True)
+        return r
+
+    import torch
+    for i in range(args.warmup):
+        one_step(f"w{i}")
+    if dist:
+        dist.barrier()
+    torch.cuda.synchronize() if torch.cuda.is_available() else None
+    t0 = time.time()
+    last = None
+    for i in range(args.steps):
+        last = one_step(i)
+    torch.cuda.synchronize() if torch.cuda.is_available() else None
+    if dist:
+        dist.barrier()
+    elapsed = time.time() - t0
+
+    my_bytes = float(last["input_uncompressed_bytes"] * args.steps)
+    if dist:
+        t = torch.tensor([elapsed])
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = t.item()
+        b = torch.tensor([my_bytes])
+        dist.all_reduce(b, op=dist.ReduceOp.SUM)
+        my_bytes = b.item()
+
+    if rank != 0:
+        if dist:
+            dist.destroy_process_group()
+        return 0
+
+    value_mb_s = my_bytes / elapsed / 1e6
+
+    # ---- roofline for the dominant kernel (HIP-event timed inside the lib) ----
+    last["_input_compressed_bytes"] = input_compressed
+    last["_n_inputs"] = args.sstables
+    dom = last["dominant_kernel"]
+    dom_ms = last["dominant_kernel_ms"]
+    ab = algorithmic_bytes(dom, last)
+    achieved_gbs = (ab / (dom_ms / 1e3)) / 1e9 if dom_ms > 0 else 0.0
+    roofline = {
+        "bound": "hbm",
+        "achieved": round(achieved_gbs, 1),
+        "peak": HBM_PEAK_GBS,
+        "unit": "GB/s",
+        "frac": round(achieved_gbs / HBM_PEAK_GBS, 4),
+        "traffic": None,  # PMC traffic measured separately via rocprofv3 (profiles/)
+    }
+
+    # ---- CPU baseline: oracle compactor, bounded sample, 1 thread ----
+    cpu_baseline = None
+    if world == 1 and not args.no_cpu_baseline:
+        ob = oracle_bin()
+        sd = os.path.join(args.dir, "cpu_sample")
+        shutil.rmtree(sd, ignore_errors=True)
+        os.makedirs(sd)
+        sample_rows = 120_000  # ~1 GiB in: ~10-20 s of single-core work
+        subprocess.run([ob, "gen", sd, "seed=42", f"n={args.sstables}",
+                        f"rows={sample_rows}", f"vlen={args.vlen}",
+                        f"overlap={args.overlap}"], check=True, capture_output=True)
+        sins = [os.path.join(sd, f"oa-{g}-big") for g in range(1, args.sstables + 1)]
+        out = subprocess.run([ob, "compact", os.path.join(sd, "oa-100-big"), *sins],
+                             check=True, capture_output=True, text=True)
+        st = json.loads(out.stdout.strip().splitlines()[-1])
+        cpu_baseline = {
+            "value": round(st["mb_per_s"], 2),
+            "unit": "MB/s input sstable bytes",
+            "cores": 1,
+            "kind": "port",
+            "sample": f"{args.sstables}x{sample_rows} rows (~{st['input_uncompressed_bytes']//2**20} MiB input), oracle compactor incl. file IO",
+        }
+        shutil.rmtree(sd, ignore_errors=True)
+
+    out = {
+        "metric": "compaction MB/s (input SSTable bytes merged+compressed)",
+        "value": round(value_mb_s, 2),
+        "unit": "MB/s",
+        "n_gpus": world,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(elapsed * 1000 / args.steps, 2),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,  # no published compaction MB/s in the reference
+        "dtype": "u8",
+        "data": "synthetic",
+        "config": {
+            "workload": "C2: 8x2GiB sstables, 10% key overlap, ~1KiB values, LZ4 16KiB chunks",
+            "sstables": args.sstables,
+            "rows_per_sstable": args.rows,
+            "value_len": args.vlen,
+            "overlap_pct": args.overlap,
+            "parallelism": f"token-independent shards x{world}, no collectives",
+            "input_uncompressed_bytes_per_rank": last["input_uncompressed_bytes"],
+            "gen_seconds": round(t_gen, 1),
+            "phase_ms": last["ms"],
+        },
+        "roofline": roofline,
+        "cpu_baseline": cpu_baseline,
+    }
+    print(json.dumps(out))
+    if dist:
+        dist.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,231 @@
+"""cassandra_amd — MI355X-native SSTable compaction (host bindings).
+
+Thin ctypes wrapper over libcassandra_gpucompact.so (the C-ABI HIP library;
+see include/gpucompact.h). All compute is GPU-side; this package only
+marshals job descriptions, mirroring what a JNI/Panama shim in the Java host
+would do.
+"""
+import ctypes
+import os
+
+_HERE = os.path.dirname(os.path.abspath(__file__))
+_LIB_PATH = os.path.join(_HERE, "libcassandra_gpucompact.so")
+
+
+class GpucPurgeRange(ctypes.Structure):
+    _fields_ = [
+        ("token_lo", ctypes.c_int64),
+        ("token_hi", ctypes.c_int64),
+        ("min_timestamp", ctypes.c_int64),
+    ]
+
+
+class GpucJob(ctypes.Structure):
+    _fields_ = [
+        ("input_bases", ctypes.POINTER(ctypes.c_char_p)),
+        ("n_inputs", ctypes.c_int32),
+        ("output_base", ctypes.c_char_p),
+        ("now_sec", ctypes.c_int64),
+        ("gc_before", ctypes.c_int64),
+        ("never_purge", ctypes.c_int32),
+        ("enforce_strict_liveness", ctypes.c_int32),
+        ("overlaps", ctypes.POINTER(GpucPurgeRange)),
+        ("n_overlaps", ctypes.c_int32),
+        ("has_token_range", ctypes.c_int32),
+        ("token_lo", ctypes.c_int64),
+        ("token_hi", ctypes.c_int64),
+        ("device", ctypes.c_int32),
+    ]
+
+
+class GpucResult(ctypes.Structure):
+    _fields_ = [
+        ("input_uncompressed_bytes", ctypes.c_uint64),
+        ("output_uncompressed_bytes", ctypes.c_uint64),
+        ("output_compressed_bytes", ctypes.c_uint64),
+        ("partitions_in", ctypes.c_uint64),
+        ("partitions_out", ctypes.c_uint64),
+        ("rows_in", ctypes.c_uint64),
+        ("rows_out", ctypes.c_uint64),
+        ("merged_counts", ctypes.c_uint64 * 64),
+        ("ms_read_io", ctypes.c_double),
+        ("ms_h2d", ctypes.c_double),
+        ("ms_decompress", ctypes.c_double),
+        ("ms_parse", ctypes.c_double),
+        ("ms_merge", ctypes.c_double),
+        ("ms_reconcile", ctypes.c_double),
+        ("ms_serialize", ctypes.c_double),
+        ("ms_compress", ctypes.c_double),
+        ("ms_d2h", ctypes.c_double),
+        ("ms_write_io", ctypes.c_double),
+        ("ms_total", ctypes.c_double),
+        ("dominant_kernel", ctypes.c_char * 64),
+        ("dominant_kernel_ms", ctypes.c_double),
+        ("dominant_kernel_launches", ctypes.c_uint64),
+        ("error", ctypes.c_char * 256),
+    ]
+
+
+class GpucGenSpec(ctypes.Structure):
+    _fields_ = [
+        ("seed", ctypes.c_uint64),
+        ("n_sstables", ctypes.c_uint32),
+        ("rows_per_sstable", ctypes.c_uint64),
+        ("overlap_pct", ctypes.c_uint32),
+        ("value_len", ctypes.c_uint32),
+        ("value_repeat_pct", ctypes.c_uint32),
+        ("tombstone_pct", ctypes.c_uint32),
+        ("partition_del_pct", ctypes.c_uint32),
+        ("base_ts", ctypes.c_int64),
+        ("base_ldt", ctypes.c_int64),
+        ("first_generation", ctypes.c_uint64),
+        ("device", ctypes.c_int32),
+    ]
+
+
+INT64_MIN = -(2**63)
+
+
+class GpuCompactError(RuntimeError):
+    pass
+
+
+_lib = None
+
+
+def load_library(path=None):
+    """Load the HIP library. Raises loudly when absent — there is no fallback."""
+    global _lib
+    if _lib is not None and path is None:
+        return _lib
+    p = path or _LIB_PATH
+    if not os.path.exists(p):
+        raise GpuCompactError(
+            f"libcassandra_gpucompact.so not found at {p}; run __graft_entry__.build()"
+        )
+    lib = ctypes.CDLL(p)
+    lib.gpuc_compact.argtypes = [ctypes.POINTER(GpucJob), ctypes.POINTER(GpucResult)]
+    lib.gpuc_compact.restype = ctypes.c_int
+    lib.gpuc_generate.argtypes = [
+        ctypes.POINTER(GpucGenSpec),
+        ctypes.c_char_p,
+        ctypes.c_char_p,
+        ctypes.c_size_t,
+    ]
+    lib.gpuc_generate.restype = ctypes.c_int
+    lib.gpuc_version.restype = ctypes.c_char_p
+    lib.gpuc_device_count.restype = ctypes.c_int
+    _lib = lib
+    return lib
+
+
+def version():
+    return load_library().gpuc_version().decode()
+
+
+def device_count():
+    return load_library().gpuc_device_count()
+
+
+def compact(
+    input_bases,
+    output_base,
+    now_sec=1800000000,
+    gc_before=INT64_MIN,
+    never_purge=False,
+    enforce_strict_liveness=False,
+    overlaps=None,
+    token_range=None,
+    device=0,
+):
+    """One compaction task (mirrors CompactionTask.runMayThrow's hot loop).
+
+    input_bases: list of sstable path prefixes ('<dir>/oa-<gen>-big').
+    overlaps: list of (token_lo, token_hi, min_timestamp) for the purge
+      evaluator (CompactionController.getPurgeEvaluator); None == no overlaps.
+    token_range: (lo, hi) inclusive token shard restriction or None.
+    """
+    lib = load_library()
+    job = GpucJob()
+    arr = (ctypes.c_char_p * len(input_bases))(*[b.encode() for b in input_bases])
+    job.input_bases = arr
+    job.n_inputs = len(input_bases)
+    job.output_base = output_base.encode()
+    job.now_sec = now_sec
+    job.gc_before = gc_before
+    job.never_purge = 1 if never_purge else 0
+    job.enforce_strict_liveness = 1 if enforce_strict_liveness else 0
+    if overlaps:
+        ovr = (GpucPurgeRange * len(overlaps))()
+        for i, (lo, hi, ts) in enumerate(overlaps):
+            ovr[i].token_lo, ovr[i].token_hi, ovr[i].min_timestamp = lo, hi, ts
+        job.overlaps = ovr
+        job.n_overlaps = len(overlaps)
+    if token_range is not None:
+        job.has_token_range = 1
+        job.token_lo, job.token_hi = token_range
+    job.device = device
+    res = GpucResult()
+    rc = lib.gpuc_compact(ctypes.byref(job), ctypes.byref(res))
+    if rc != 0:
+        raise GpuCompactError(f"gpuc_compact rc={rc}: {res.error.decode(errors='replace')}")
+    return {
+        "input_uncompressed_bytes": res.input_uncompressed_bytes,
+        "output_uncompressed_bytes": res.output_uncompressed_bytes,
+        "output_compressed_bytes": res.output_compressed_bytes,
+        "partitions_in": res.partitions_in,
+        "partitions_out": res.partitions_out,
+        "rows_in": res.rows_in,
+        "rows_out": res.rows_out,
+        "merged_counts": list(res.merged_counts),
+        "ms": {
+            "read_io": res.ms_read_io,
+            "h2d": res.ms_h2d,
+            "decompress": res.ms_decompress,
+            "parse": res.ms_parse,
+            "merge": res.ms_merge,
+            "serialize": res.ms_serialize,
+            "compress": res.ms_compress,
+            "d2h": res.ms_d2h,
+            "total": res.ms_total,
+        },
+        "dominant_kernel": res.dominant_kernel.decode(),
+        "dominant_kernel_ms": res.dominant_kernel_ms,
+    }
+
+
+def generate(
+    out_dir,
+    seed=42,
+    n_sstables=4,
+    rows_per_sstable=1000,
+    overlap_pct=10,
+    value_len=1024,
+    value_repeat_pct=55,
+    tombstone_pct=0,
+    partition_del_pct=0,
+    base_ts=1700000000000000,
+    base_ldt=1700000000,
+    first_generation=1,
+    device=0,
+):
+    """GPU-side synthetic sstable generation (shared contract: oracle/src/gen.h)."""
+    lib = load_library()
+    spec = GpucGenSpec(
+        seed=seed,
+        n_sstables=n_sstables,
+        rows_per_sstable=rows_per_sstable,
+        overlap_pct=overlap_pct,
+        value_len=value_len,
+        value_repeat_pct=value_repeat_pct,
+        tombstone_pct=tombstone_pct,
+        partition_del_pct=partition_del_pct,
+        base_ts=base_ts,
+        base_ldt=base_ldt,
+        first_generation=first_generation,
+        device=device,
+    )
+    err = ctypes.create_string_buffer(256)
+    rc = lib.gpuc_generate(ctypes.byref(spec), out_dir.encode(), err, 256)
+    if rc != 0:
+        raise GpuCompactError(f"gpuc_generate rc={rc}: {err.value.decode(errors='replace')}")

@@ -1,0 +1,1129 @@
+// PRODUCT — host side of libcassandra_gpucompact: file I/O, sstable metadata
+// (CompressionInfo/Statistics/Summary/TOC/Digest) assembly, and the HIP
+// pipeline orchestration. Compiled as one TU with kernels.hip by hipcc.
+//
+// The host does NO data-path compute: decompress/decode/merge/reconcile/
+// purge/serialize/compress/CRC/bloom all run in the kernels. There is no CPU
+// fallback — gpuc_compact fails with GPUC_ERR_NO_GPU when no device exists.
+#include "kernels.hip"
+#include "../../include/gpucompact.h"
+
+#include <algorithm>
+#include <cmath>
+#include <cstdio>
+#include <cstring>
+#include <map>
+#include <memory>
+#include <stdexcept>
+#include <string>
+#include <thread>
+#include <vector>
+
+namespace gpuc {
+
+using bytes = std::vector<uint8_t>;
+
+#define HIP_CHECK(x)                                                              \
+    do {                                                                          \
+        hipError_t _e = (x);                                                      \
+        if (_e != hipSuccess)                                                     \
+            throw std::runtime_error(std::string("HIP error: ") +                 \
+                                     hipGetErrorString(_e) + " at " #x);          \
+    } while (0)
+
+// ---------------------------------------------------------------------------
+// host byte helpers
+// ---------------------------------------------------------------------------
+static void put_be16(bytes& o, uint16_t v) { o.push_back(v >> 8); o.push_back((uint8_t)v); }
+static void put_be32(bytes& o, uint32_t v) { for (int i = 3; i >= 0; i--) o.push_back((uint8_t)(v >> (8 * i))); }
+static void put_be64(bytes& o, uint64_t v) { for (int i = 7; i >= 0; i--) o.push_back((uint8_t)(v >> (8 * i))); }
+static void put_uvint(bytes& o, uint64_t v) {
+    uint8_t tmp[9];
+    int n = uvint_put(tmp, v);
+    o.insert(o.end(), tmp, tmp + n);
+}
+
+struct HReader {
+    const uint8_t* p;
+    size_t len, pos = 0;
+    explicit HReader(const bytes& b) : p(b.data()), len(b.size()) {}
+    void need(size_t n) const { if (pos + n > len) throw std::runtime_error("short read in component"); }
+    uint8_t u8() { need(1); return p[pos++]; }
+    uint16_t be16() { need(2); uint16_t v = ((uint16_t)p[pos] << 8) | p[pos + 1]; pos += 2; return v; }
+    uint32_t be32() { need(4); uint32_t v = 0; for (int i = 0; i < 4; i++) v = (v << 8) | p[pos + i]; pos += 4; return v; }
+    uint64_t be64() { need(8); uint64_t v = 0; for (int i = 0; i < 8; i++) v = (v << 8) | p[pos + i]; pos += 8; return v; }
+    bytes take(size_t n) { need(n); bytes b(p + pos, p + pos + n); pos += n; return b; }
+    void skip(size_t n) { need(n); pos += n; }
+    uint64_t uvint() {
+        need(1);
+        uint64_t q = pos;
+        uint64_t v = uvint_get(p, &q);
+        if (q > len) throw std::runtime_error("vint overrun");
+        pos = q;
+        return v;
+    }
+};
+
+static bytes read_file(const std::string& path) {
+    FILE* f = fopen(path.c_str(), "rb");
+    if (!f) throw std::runtime_error("cannot open " + path);
+    fseek(f, 0, SEEK_END);
+    long n = ftell(f);
+    fseek(f, 0, SEEK_SET);
+    bytes b((size_t)n);
+    if (n && fread(b.data(), 1, (size_t)n, f) != (size_t)n) { fclose(f); throw std::runtime_error("short read " + path); }
+    fclose(f);
+    return b;
+}
+static void write_file(const std::string& path, const uint8_t* p, size_t n) {
+    FILE* f = fopen(path.c_str(), "wb");
+    if (!f) throw std::runtime_error("cannot create " + path);
+    if (n && fwrite(p, 1, n, f) != n) { fclose(f); throw std::runtime_error("short write " + path); }
+    fclose(f);
+}
+
+// ---------------------------------------------------------------------------
+// component parsing (input side)
+// ---------------------------------------------------------------------------
+struct HCompressionInfo {
+    uint32_t chunk_len = 16384;
+    uint32_t max_compressed = 0x7FFFFFFF;
+    uint64_t data_len = 0;
+    std::vector<uint64_t> offsets;
+};
+static HCompressionInfo parse_compression_info(const bytes& b) {
+    HReader r(b);
+    uint16_t nlen = r.be16();
+    bytes name = r.take(nlen);
+    std::string algo((char*)name.data(), name.size());
+    if (algo != "LZ4Compressor") throw std::runtime_error("unsupported compressor " + algo + " (GPU path: LZ4 only)");
+    uint32_t opts = r.be32();
+    for (uint32_t i = 0; i < opts; i++) { r.take(r.be16()); r.take(r.be16()); }
+    HCompressionInfo ci;
+    ci.chunk_len = r.be32();
+    ci.max_compressed = r.be32();
+    ci.data_len = r.be64();
+    uint32_t n = r.be32();
+    ci.offsets.resize(n);
+    for (uint32_t i = 0; i < n; i++) ci.offsets[i] = r.be64();
+    if (ci.chunk_len != CHUNK_LEN) throw std::runtime_error("unsupported chunk_length (16 KiB only in round 1)");
+    return ci;
+}
+
+struct HStatistics {
+    // HEADER component
+    int64_t hdr_min_ts = TIMESTAMP_EPOCH;
+    int64_t hdr_min_ldt = DELETION_TIME_EPOCH;
+    int32_t hdr_min_ttl = 0;
+    std::string key_type;
+    std::vector<std::string> clustering_types;
+    std::vector<std::pair<bytes, std::string>> static_cols, regular_cols;
+    // STATS mins (for SerializationHeader.make of the output)
+    int64_t min_timestamp = 0, max_timestamp = 0;
+    int64_t min_ldt = NO_DELETION_TIME, max_ldt = 0;
+    int32_t min_ttl = 0, max_ttl = 0;
+    std::string partitioner;
+};
+static HStatistics parse_statistics(const bytes& b) {
+    HReader r(b);
+    uint32_t count = r.be32();
+    r.be32();
+    std::map<uint32_t, uint32_t> toc;
+    for (uint32_t i = 0; i < count; i++) {
+        uint32_t t = r.be32();
+        uint32_t pos = r.be32();
+        toc[t] = pos;
+    }
+    r.be32();
+    HStatistics st;
+    if (toc.count(0)) {
+        HReader v(b);
+        v.pos = toc[0];
+        uint16_t n = v.be16();
+        bytes s = v.take(n);
+        st.partitioner.assign((char*)s.data(), s.size());
+    }
+    if (!toc.count(3)) throw std::runtime_error("Statistics.db missing HEADER component");
+    {
+        HReader h(b);
+        h.pos = toc[3];
+        st.hdr_min_ts = (int64_t)h.uvint() + TIMESTAMP_EPOCH;
+        st.hdr_min_ldt = (int64_t)(int32_t)(uint32_t)h.uvint() + DELETION_TIME_EPOCH;
+        st.hdr_min_ttl = (int32_t)(uint32_t)h.uvint();
+        auto rstr = [&]() { size_t n = (size_t)h.uvint(); bytes s = h.take(n); return std::string((char*)s.data(), s.size()); };
+        st.key_type = rstr();
+        size_t nct = (size_t)h.uvint();
+        for (size_t i = 0; i < nct; i++) st.clustering_types.push_back(rstr());
+        for (auto* cols : {&st.static_cols, &st.regular_cols}) {
+            size_t nc = (size_t)h.uvint();
+            for (size_t i = 0; i < nc; i++) {
+                size_t nn = (size_t)h.uvint();
+                bytes name = h.take(nn);
+                cols->push_back({name, rstr()});
+            }
+        }
+    }
+    if (toc.count(2)) {
+        HReader s(b);
+        s.pos = toc[2];
+        for (int hh = 0; hh < 2; hh++) { uint32_t n = s.be32(); s.skip((size_t)n * 16); }
+        s.skip(12);
+        st.min_timestamp = (int64_t)s.be64();
+        st.max_timestamp = (int64_t)s.be64();
+        st.min_ldt = ldt_long(s.be32());
+        st.max_ldt = ldt_long(s.be32());
+        st.min_ttl = (int32_t)s.be32();
+        st.max_ttl = (int32_t)s.be32();
+    }
+    return st;
+}
+
+// Index.db -> partition positions (host thread; validates non-indexed entries)
+static void parse_index_positions(const bytes& ib, uint64_t data_len,
+                                  std::vector<uint64_t>& positions, std::string& err) {
+    try {
+        HReader r(ib);
+        while (r.pos < r.len) {
+            uint16_t klen = r.be16();
+            r.skip(klen);
+            uint64_t pos = r.uvint();
+            uint64_t promoted = r.uvint();
+            r.skip(promoted);
+            positions.push_back(pos);
+        }
+        positions.push_back(data_len);
+    } catch (const std::exception& e) {
+        err = e.what();
+    }
+}
+
+// ---------------------------------------------------------------------------
+// output component assembly (host; small metadata only — spec'd identically in
+// oracle/src/sstable.cpp so GPU and oracle outputs are byte-identical)
+// ---------------------------------------------------------------------------
+static void put_type_str(bytes& o, const std::string& s) {
+    put_uvint(o, s.size());
+    o.insert(o.end(), s.begin(), s.end());
+}
+static std::vector<int64_t> est_hist_offsets(int size) {
+    std::vector<int64_t> off;
+    int64_t last = 1;
+    off.push_back(1);
+    for (int i = 1; i < size; i++) {
+        int64_t next = (int64_t)llround((double)last * 1.2);
+        if (next == last) next++;
+        off.push_back(next);
+        last = next;
+    }
+    return off;
+}
+static void put_est_hist(bytes& o, const std::vector<int64_t>& off, const uint64_t* buckets) {
+    put_be32(o, (uint32_t)(off.size() + 1));
+    for (size_t i = 0; i < off.size() + 1; i++) {
+        put_be64(o, (uint64_t)off[i == 0 ? 0 : i - 1]);
+        put_be64(o, buckets[i]);
+    }
+}
+static bytes hll_stub() {
+    bytes h;
+    put_be32(h, (uint32_t)-2);
+    auto pv = [&](uint32_t v) { while (v >= 0x80) { h.push_back((uint8_t)(v | 0x80)); v >>= 7; } h.push_back((uint8_t)v); };
+    pv(13); pv(25); pv(0);
+    uint32_t reg_ints = (uint32_t)((((1 << 13) * 5) + 31) / 32);
+    pv(reg_ints * 4);
+    h.insert(h.end(), reg_ints * 4, 0);
+    bytes o;
+    put_be32(o, (uint32_t)h.size());
+    o.insert(o.end(), h.begin(), h.end());
+    return o;
+}
+
+struct OutMeta {
+    // header (deltas) of the output sstable
+    HeaderStats hs;
+    std::string key_type;
+    std::vector<std::pair<bytes, std::string>> regular_cols;
+    // collected stats
+    int64_t min_timestamp, max_timestamp, min_ldt, max_ldt;
+    int32_t min_ttl, max_ttl;
+    uint64_t total_rows, total_cells;
+    bool has_partition_deletions;
+    bytes first_key, last_key;
+    std::map<uint32_t, uint32_t> tomb_hist;
+    double compression_ratio;
+    uint64_t part_size_hist[156];
+    uint64_t cells_hist[119];
+};
+
+static bytes serialize_statistics_out(const OutMeta& m) {
+    bytes validation;
+    {
+        std::string pn = "org.apache.cassandra.dht.Murmur3Partitioner";
+        put_be16(validation, (uint16_t)pn.size());
+        validation.insert(validation.end(), pn.begin(), pn.end());
+        uint64_t fp;
+        double fpv = 0.01;
+        memcpy(&fp, &fpv, 8);
+        put_be64(validation, fp);
+    }
+    bytes compaction = hll_stub();
+    bytes stats;
+    {
+        static const std::vector<int64_t> ps_off = est_hist_offsets(155);
+        static const std::vector<int64_t> ch_off = est_hist_offsets(118);
+        put_est_hist(stats, ps_off, m.part_size_hist);
+        put_est_hist(stats, ch_off, m.cells_hist);
+        put_be64(stats, (uint64_t)-1LL); put_be32(stats, 0);  // commitLogUpperBound NONE
+        put_be64(stats, (uint64_t)m.min_timestamp);
+        put_be64(stats, (uint64_t)m.max_timestamp);
+        put_be32(stats, ldt_u32(m.min_ldt));
+        put_be32(stats, ldt_u32(m.max_ldt));
+        put_be32(stats, (uint32_t)m.min_ttl);
+        put_be32(stats, (uint32_t)m.max_ttl);
+        uint64_t cr; double crv = m.compression_ratio; memcpy(&cr, &crv, 8); put_be64(stats, cr);
+        put_be32(stats, 100);  // TombstoneHistogram maxBinSize
+        put_be32(stats, (uint32_t)m.tomb_hist.size());
+        for (auto& [pt, cnt] : m.tomb_hist) { put_be64(stats, pt); put_be32(stats, cnt); }
+        put_be32(stats, 0);               // sstableLevel
+        put_be64(stats, 0);               // repairedAt
+        put_uvint(stats, 0);              // improvedMinMax: clustering type count 0
+        stats.push_back(1); put_be16(stats, 0);  // Slice start: INCL_START, 0 values
+        stats.push_back(6); put_be16(stats, 0);  // Slice end: INCL_END, 0 values
+        stats.push_back(0);               // hasLegacyCounterShards
+        put_be64(stats, m.total_cells);
+        put_be64(stats, m.total_rows);
+        put_be64(stats, (uint64_t)-1LL); put_be32(stats, 0);  // commitLogLowerBound NONE
+        put_be32(stats, 0);               // commitLogIntervals empty
+        stats.push_back(0);               // pendingRepair null
+        stats.push_back(0);               // isTransient
+        stats.push_back(0);               // originatingHostId null
+        stats.push_back(m.has_partition_deletions ? 1 : 0);
+        put_uvint(stats, m.first_key.size());
+        stats.insert(stats.end(), m.first_key.begin(), m.first_key.end());
+        put_uvint(stats, m.last_key.size());
+        stats.insert(stats.end(), m.last_key.begin(), m.last_key.end());
+        uint64_t ts; double tsc = 0.0; memcpy(&ts, &tsc, 8); put_be64(stats, ts);
+    }
+    bytes header;
+    {
+        put_uvint(header, (uint64_t)(m.hs.min_ts - TIMESTAMP_EPOCH));
+        put_uvint(header, sext32(m.hs.min_ldt - DELETION_TIME_EPOCH));
+        put_uvint(header, sext32(m.hs.min_ttl));
+        put_type_str(header, m.key_type);
+        put_uvint(header, 0);  // clustering types
+        put_uvint(header, 0);  // static columns
+        put_uvint(header, m.regular_cols.size());
+        for (auto& [name, t] : m.regular_cols) {
+            put_uvint(header, name.size());
+            header.insert(header.end(), name.begin(), name.end());
+            put_type_str(header, t);
+        }
+    }
+    const bytes* comps[4] = {&validation, &compaction, &stats, &header};
+    bytes out;
+    uint32_t crc = 0;
+    auto crc_int = [](uint32_t c, uint32_t v) {
+        uint8_t b[4] = {(uint8_t)(v >> 24), (uint8_t)(v >> 16), (uint8_t)(v >> 8), (uint8_t)v};
+        return crc32_update_bitwise(c, b, 4);
+    };
+    put_be32(out, 4);
+    crc = crc_int(0, 4);
+    put_be32(out, crc);
+    uint32_t pos = 4 + 8 * 4 + 2 * 4;
+    for (int i = 0; i < 4; i++) {
+        put_be32(out, i);
+        crc = crc_int(crc, i);
+        put_be32(out, pos);
+        crc = crc_int(crc, pos);
+        pos += comps[i]->size() + 4;
+    }
+    put_be32(out, crc);
+    for (int i = 0; i < 4; i++) {
+        out.insert(out.end(), comps[i]->begin(), comps[i]->end());
+        put_be32(out, crc32_update_bitwise(0, comps[i]->data(), comps[i]->size()));
+    }
+    return out;
+}
+
+// bloom spec for fp=0.01 (BloomCalculations; identical table in the oracle)
+struct HBloomSpec { int k; int buckets; };
+static HBloomSpec bloom_spec_001() { return {5, 10}; }  // computeBloomSpec(20, 0.01)
+
+// ---------------------------------------------------------------------------
+// device helpers (scan, merge)
+// ---------------------------------------------------------------------------
+struct DevBuf {
+    void* p = nullptr;
+    size_t n = 0;
+    void alloc(size_t bytes_) {
+        free_();
+        HIP_CHECK(hipMalloc(&p, bytes_ ? bytes_ : 8));
+        n = bytes_;
+    }
+    void free_() { if (p) { (void)hipFree(p); p = nullptr; } }
+    ~DevBuf() { free_(); }
+    template <typename T> T* as() const { return (T*)p; }
+};
+
+// exclusive scan over u64; returns total
+static uint64_t exscan_u64(uint64_t* d_data, uint64_t n, hipStream_t stream) {
+    if (n == 0) return 0;
+    uint64_t n_blocks = (n + SCAN_TILE - 1) / SCAN_TILE;
+    DevBuf sums, out;
+    sums.alloc(n_blocks * 8);
+    out.alloc(n * 8);
+    hipLaunchKernelGGL(k_scan_partial, dim3((uint32_t)n_blocks), dim3(SCAN_BLOCK), 0, stream,
+                       d_data, out.as<uint64_t>(), sums.as<uint64_t>(), n);
+    uint64_t total = 0;
+    if (n_blocks > 1) {
+        total = exscan_u64(sums.as<uint64_t>(), n_blocks, stream);
+        hipLaunchKernelGGL(k_scan_add, dim3((uint32_t)n_blocks), dim3(SCAN_BLOCK), 0, stream,
+                           out.as<uint64_t>(), sums.as<uint64_t>(), n);
+    } else {
+        uint64_t s;
+        HIP_CHECK(hipMemcpyAsync(&s, sums.p, 8, hipMemcpyDeviceToHost, stream));
+        HIP_CHECK(hipStreamSynchronize(stream));
+        total = s;
+    }
+    if (n_blocks > 1) {
+        // total = scan total of sums + last block handled inside recursion; read final element
+        uint64_t last_off, last_val;
+        HIP_CHECK(hipMemcpyAsync(&last_off, out.as<uint64_t>() + (n - 1), 8, hipMemcpyDeviceToHost, stream));
+        HIP_CHECK(hipMemcpyAsync(&last_val, d_data + (n - 1), 8, hipMemcpyDeviceToHost, stream));
+        HIP_CHECK(hipStreamSynchronize(stream));
+        total = last_off + last_val;
+    }
+    HIP_CHECK(hipMemcpyAsync(d_data, out.p, n * 8, hipMemcpyDeviceToDevice, stream));
+    HIP_CHECK(hipStreamSynchronize(stream));
+    return total;
+}
+
+// sort MRec array (uniform-run merge rounds); returns pointer to sorted buffer
+static MRec* merge_sort_recs(MRec* d_a, MRec* d_b, uint64_t n, hipStream_t stream) {
+    MRec* in = d_a;
+    MRec* out = d_b;
+    for (uint64_t run = 1; run < n; run *= 2) {
+        if (run < 16) {
+            uint64_t pairs = (n + run * 2 - 1) / (run * 2);
+            hipLaunchKernelGGL(k_merge_small, dim3((uint32_t)((pairs + 255) / 256)), dim3(256), 0,
+                               stream, in, out, n, run);
+        } else {
+            uint64_t tiles = (n + MERGE_TILE - 1) / MERGE_TILE;
+            hipLaunchKernelGGL(k_merge_uniform, dim3((uint32_t)((tiles + 255) / 256)), dim3(256), 0,
+                               stream, in, out, n, run);
+        }
+        std::swap(in, out);
+    }
+    return in;
+}
+
+// merge k pre-sorted runs (per-source) with pairwise rounds
+static MRec* merge_sorted_runs(MRec* d_a, MRec* d_b, std::vector<uint64_t> runs /*boundaries, size k+1*/,
+                               hipStream_t stream) {
+    MRec* in = d_a;
+    MRec* out = d_b;
+    while (runs.size() > 2) {
+        std::vector<MergePair> pairs;
+        std::vector<uint64_t> next_runs;
+        next_runs.push_back(0);
+        uint64_t tile_beg = 0;
+        for (size_t i = 0; i + 1 < runs.size(); i += 2) {
+            MergePair p{};
+            p.a_beg = runs[i];
+            p.a_end = runs[i + 1];
+            if (i + 2 < runs.size()) { p.b_beg = runs[i + 1]; p.b_end = runs[i + 2]; }
+            else { p.b_beg = p.a_end; p.b_end = p.a_end; }
+            p.out_beg = p.a_beg;
+            p.tile_beg = tile_beg;
+            uint64_t n_out = (p.a_end - p.a_beg) + (p.b_end - p.b_beg);
+            tile_beg += (n_out + MERGE_TILE - 1) / MERGE_TILE;
+            pairs.push_back(p);
+            next_runs.push_back(p.b_end);
+        }
+        DevBuf d_pairs;
+        d_pairs.alloc(pairs.size() * sizeof(MergePair));
+        HIP_CHECK(hipMemcpyAsync(d_pairs.p, pairs.data(), pairs.size() * sizeof(MergePair),
+                                 hipMemcpyHostToDevice, stream));
+        uint64_t total_tiles = tile_beg;
+        hipLaunchKernelGGL(k_merge_pairs, dim3((uint32_t)((total_tiles + 255) / 256)), dim3(256), 0,
+                           stream, in, out, d_pairs.as<MergePair>(), (uint32_t)pairs.size(), total_tiles);
+        HIP_CHECK(hipStreamSynchronize(stream));
+        std::swap(in, out);
+        runs = next_runs;
+    }
+    return in;
+}
+
+}  // namespace gpuc
+
+namespace gpuc {
+
+static void init_outstats(DevBuf& d_stats, hipStream_t stream) {
+    hipLaunchKernelGGL(k_init_outstats, dim3(1), dim3(64), 0, stream, d_stats.as<OutStats>());
+}
+
+static bool g_crc_tables_ready = false;
+static void ensure_crc_tables(hipStream_t stream) {
+    if (g_crc_tables_ready) return;
+    hipLaunchKernelGGL(k_crc_init, dim3(1), dim3(256), 0, stream);
+    HIP_CHECK(hipStreamSynchronize(stream));
+    g_crc_tables_ready = true;
+}
+
+// allocate the OutParts SoA for n entries
+struct OutPartsBuf {
+    DevBuf keypfx, klen, pdel_mfda, pdel_ldt, flags, live_ts, live_ttl, live_let,
+        rdel_mfda, rdel_ldt, cell_ts, cell_ldt, cell_ttl, val_addr, val_len, keep;
+    OutParts op{};
+    void alloc(uint64_t n) {
+        keypfx.alloc(n * 8); klen.alloc(n); pdel_mfda.alloc(n * 8); pdel_ldt.alloc(n * 4);
+        flags.alloc(n); live_ts.alloc(n * 8); live_ttl.alloc(n * 4); live_let.alloc(n * 8);
+        rdel_mfda.alloc(n * 8); rdel_ldt.alloc(n * 4); cell_ts.alloc(n * 8); cell_ldt.alloc(n * 4);
+        cell_ttl.alloc(n * 4); val_addr.alloc(n * 8); val_len.alloc(n * 4); keep.alloc(n);
+        op = OutParts{keypfx.as<uint64_t>(), klen.as<uint8_t>(), pdel_mfda.as<int64_t>(),
+                      pdel_ldt.as<uint32_t>(), flags.as<uint8_t>(), live_ts.as<int64_t>(),
+                      live_ttl.as<int32_t>(), live_let.as<int64_t>(), rdel_mfda.as<int64_t>(),
+                      rdel_ldt.as<uint32_t>(), cell_ts.as<int64_t>(), cell_ldt.as<uint32_t>(),
+                      cell_ttl.as<int32_t>(), val_addr.as<uint64_t>(), val_len.as<uint32_t>(),
+                      keep.as<uint8_t>()};
+    }
+};
+
+struct WriteDeviceOut {
+    uint64_t uncompressed_len = 0;
+    uint64_t compressed_len = 0;   // final Data.db size
+    uint64_t partitions = 0, rows = 0, cells = 0;
+    double ms_sizes = 0, ms_serialize = 0, ms_compress = 0, ms_d2h = 0, ms_io = 0;
+};
+
+// shared device->files writer: sizes/scan -> serialize -> compress -> gather ->
+// D2H -> write all components. `st` is the device OutStats already filled by
+// reconcile (or gen); meta_* give header/schema info.
+static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, uint64_t n_groups,
+                                           SerParams sp, DevBuf& d_stats, DevBuf& d_tomb,
+                                           uint32_t tomb_cap, const std::string& out_base,
+                                           const std::string& key_type,
+                                           const std::vector<std::pair<bytes, std::string>>& regular_cols,
+                                           hipStream_t stream) {
+    WriteDeviceOut w;
+    // histogram offset tables on device
+    static const std::vector<int64_t> ps_off_h = est_hist_offsets(155);
+    static const std::vector<int64_t> ch_off_h = est_hist_offsets(118);
+    DevBuf d_ps_off, d_ch_off;
+    d_ps_off.alloc(ps_off_h.size() * 8);
+    d_ch_off.alloc(ch_off_h.size() * 8);
+    HIP_CHECK(hipMemcpyAsync(d_ps_off.p, ps_off_h.data(), ps_off_h.size() * 8, hipMemcpyHostToDevice, stream));
+    HIP_CHECK(hipMemcpyAsync(d_ch_off.p, ch_off_h.data(), ch_off_h.size() * 8, hipMemcpyHostToDevice, stream));
+
+    hipEvent_t ev0, ev1, ev2, ev3, ev4;
+    HIP_CHECK(hipEventCreate(&ev0)); HIP_CHECK(hipEventCreate(&ev1)); HIP_CHECK(hipEventCreate(&ev2));
+    HIP_CHECK(hipEventCreate(&ev3)); HIP_CHECK(hipEventCreate(&ev4));
+
+    // ---- sizes + scans ----
+    HIP_CHECK(hipEventRecord(ev0, stream));
+    DevBuf d_psize, d_isize;
+    d_psize.alloc(n_groups * 8);
+    d_isize.alloc(n_groups * 8);
+    {
+        uint64_t blocks = (n_groups + 255) / 256;
+        hipLaunchKernelGGL(k_sizes, dim3((uint32_t)blocks), dim3(256), 0, stream, opb.op, n_groups,
+                           sp, d_psize.as<uint64_t>(), d_stats.as<OutStats>(),
+                           d_ps_off.as<int64_t>(), (int32_t)ps_off_h.size(),
+                           d_ch_off.as<int64_t>(), (int32_t)ch_off_h.size());
+    }
+    // d_psize becomes exclusive offsets in place after the scan
+    uint64_t total_unc = exscan_u64(d_psize.as<uint64_t>(), n_groups, stream);
+    {
+        uint64_t blocks = (n_groups + 255) / 256;
+        hipLaunchKernelGGL(k_index_sizes, dim3((uint32_t)blocks), dim3(256), 0, stream, opb.op,
+                           n_groups, d_psize.as<uint64_t>(), d_isize.as<uint64_t>());
+    }
+    uint64_t total_idx = exscan_u64(d_isize.as<uint64_t>(), n_groups, stream);
+    HIP_CHECK(hipEventRecord(ev1, stream));
+
+    // ---- stats D2H (bloom needs partitions count) ----
+    OutStats hst;
+    HIP_CHECK(hipStreamSynchronize(stream));
+    HIP_CHECK(hipMemcpy(&hst, d_stats.p, sizeof(OutStats), hipMemcpyDeviceToHost));
+    w.partitions = hst.partitions_out;
+    w.rows = hst.rows_out;
+    w.cells = hst.total_cells;
+
+    // ---- serialize (+ index + bloom) ----
+    HBloomSpec bs = bloom_spec_001();
+    uint64_t num_bits = w.partitions * (uint64_t)bs.buckets + 20;
+    uint64_t words = num_bits ? ((num_bits - 1) >> 6) + 1 : 1;
+    DevBuf d_out_data, d_out_index, d_bloom;
+    d_out_data.alloc(total_unc);
+    d_out_index.alloc(total_idx);
+    d_bloom.alloc(words * 8);
+    HIP_CHECK(hipMemsetAsync(d_bloom.p, 0, words * 8, stream));
+    {
+        uint32_t waves_per_block = 4;
+        uint64_t blocks = (n_groups + waves_per_block - 1) / waves_per_block;
+        hipLaunchKernelGGL(k_serialize, dim3((uint32_t)blocks), dim3(WAVE * waves_per_block), 0,
+                           stream, opb.op, n_groups, sp, d_psize.as<uint64_t>(),
+                           d_isize.as<uint64_t>(), d_out_data.as<uint8_t>(),
+                           d_out_index.as<uint8_t>(), d_bloom.as<uint32_t>(), words * 64, bs.k);
+    }
+    HIP_CHECK(hipEventRecord(ev2, stream));
+
+    // ---- compress ----
+    uint32_t n_chunks = (uint32_t)((total_unc + CHUNK_LEN - 1) / CHUNK_LEN);
+    DevBuf d_slots, d_csize64, d_csize, d_ccrc;
+    d_slots.alloc((uint64_t)n_chunks * LZ4_SLOT);
+    d_csize.alloc((uint64_t)n_chunks * 4);
+    d_ccrc.alloc((uint64_t)n_chunks * 4);
+    d_csize64.alloc((uint64_t)n_chunks * 8);
+    hipLaunchKernelGGL(k_lz4_compress, dim3(n_chunks), dim3(WAVE), 0, stream,
+                       d_out_data.as<uint8_t>(), total_unc, d_slots.as<uint8_t>(),
+                       d_csize.as<uint32_t>(), d_ccrc.as<uint32_t>(), n_chunks);
+    // widen sizes to u64 + scan -> chunk offsets (excluding per-chunk CRC;
+    // gather adds 4*c for preceding CRCs)
+    {
+        std::vector<uint32_t> cs(n_chunks);
+        HIP_CHECK(hipStreamSynchronize(stream));
+        HIP_CHECK(hipMemcpy(cs.data(), d_csize.p, (uint64_t)n_chunks * 4, hipMemcpyDeviceToHost));
+        std::vector<uint64_t> off(n_chunks);
+        uint64_t acc = 0;
+        for (uint32_t i = 0; i < n_chunks; i++) { off[i] = acc; acc += cs[i]; }
+        HIP_CHECK(hipMemcpy(d_csize64.p, off.data(), (uint64_t)n_chunks * 8, hipMemcpyHostToDevice));
+        w.compressed_len = acc + (uint64_t)n_chunks * 4;
+        // CompressionInfo offsets == off[i] + 4*i
+        DevBuf d_final;
+        d_final.alloc(w.compressed_len);
+        hipLaunchKernelGGL(k_chunk_gather, dim3(n_chunks), dim3(WAVE), 0, stream,
+                           d_slots.as<uint8_t>(), d_csize.as<uint32_t>(), d_ccrc.as<uint32_t>(),
+                           d_csize64.as<uint64_t>(), d_final.as<uint8_t>(), n_chunks);
+        HIP_CHECK(hipEventRecord(ev3, stream));
+
+        // ---- D2H + files ----
+        std::vector<uint8_t> h_data(w.compressed_len), h_index(total_idx), h_bloom(words * 8);
+        std::vector<uint32_t> h_crc(n_chunks);
+        HIP_CHECK(hipMemcpy(h_data.data(), d_final.p, w.compressed_len, hipMemcpyDeviceToHost));
+        HIP_CHECK(hipMemcpy(h_index.data(), d_out_index.p, total_idx, hipMemcpyDeviceToHost));
+        HIP_CHECK(hipMemcpy(h_bloom.data(), d_bloom.p, words * 8, hipMemcpyDeviceToHost));
+        HIP_CHECK(hipMemcpy(h_crc.data(), d_ccrc.p, (uint64_t)n_chunks * 4, hipMemcpyDeviceToHost));
+        HIP_CHECK(hipEventRecord(ev4, stream));
+        HIP_CHECK(hipStreamSynchronize(stream));
+
+        // tombstone ldts
+        std::vector<uint32_t> tombs;
+        if (hst.tomb_count) {
+            uint64_t nt = std::min<uint64_t>(hst.tomb_count, tomb_cap);
+            tombs.resize(nt);
+            HIP_CHECK(hipMemcpy(tombs.data(), d_tomb.p, nt * 4, hipMemcpyDeviceToHost));
+            if (hst.tomb_count > tomb_cap)
+                throw std::runtime_error("tombstone list overflow (internal cap)");
+        }
+        // first/last key
+        OutMeta m{};
+        m.hs = sp.hs;
+        m.key_type = key_type;
+        m.regular_cols = regular_cols;
+        m.min_timestamp = hst.min_ts_flip == 0xFFFFFFFFFFFFFFFFULL ? 0 : (int64_t)(hst.min_ts_flip ^ 0x8000000000000000ULL);
+        m.max_timestamp = hst.max_ts_flip == 0 ? 0 : (int64_t)(hst.max_ts_flip ^ 0x8000000000000000ULL);
+        bool no_ts = hst.min_ts_flip == 0xFFFFFFFFFFFFFFFFULL;
+        if (no_ts) { m.min_timestamp = 0; m.max_timestamp = 0; }
+        m.min_ldt = hst.min_ldt_flip == 0xFFFFFFFFFFFFFFFFULL ? NO_DELETION_TIME
+                                                              : (int64_t)(hst.min_ldt_flip ^ 0x8000000000000000ULL);
+        m.max_ldt = hst.max_ldt_flip == 0 ? NO_DELETION_TIME : (int64_t)(hst.max_ldt_flip ^ 0x8000000000000000ULL);
+        if (hst.max_ldt_flip == 0) m.max_ldt = 0;
+        m.min_ttl = hst.min_ttl == 0xFFFFFFFFu ? 0 : (int32_t)hst.min_ttl;
+        m.max_ttl = (int32_t)hst.max_ttl;
+        m.total_rows = hst.rows_out;
+        m.total_cells = hst.total_cells;
+        m.has_partition_deletions = hst.has_partition_deletions != 0;
+        for (uint32_t t : tombs) m.tomb_hist[t]++;
+        memcpy(m.part_size_hist, hst.part_size_hist, sizeof(m.part_size_hist));
+        memcpy(m.cells_hist, hst.cells_hist, sizeof(m.cells_hist));
+        m.compression_ratio = total_unc ? (double)(w.compressed_len - (uint64_t)n_chunks * 4) / (double)total_unc : -1.0;
+        if (w.partitions) {
+            uint64_t fg = hst.first_group, lg = hst.last_group;
+            uint64_t kp;
+            uint8_t kl;
+            HIP_CHECK(hipMemcpy(&kp, opb.op.keypfx + fg, 8, hipMemcpyDeviceToHost));
+            HIP_CHECK(hipMemcpy(&kl, opb.op.klen + fg, 1, hipMemcpyDeviceToHost));
+            for (int b = 0; b < kl; b++) m.first_key.push_back((uint8_t)(kp >> (8 * (7 - b))));
+            HIP_CHECK(hipMemcpy(&kp, opb.op.keypfx + lg, 8, hipMemcpyDeviceToHost));
+            HIP_CHECK(hipMemcpy(&kl, opb.op.klen + lg, 1, hipMemcpyDeviceToHost));
+            for (int b = 0; b < kl; b++) m.last_key.push_back((uint8_t)(kp >> (8 * (7 - b))));
+        }
+
+        // Digest: CRC32 of the whole Data.db via per-chunk combine
+        static Crc32Combiner comb;
+        uint32_t digest = 0;
+        {
+            std::vector<uint32_t> cs2(cs);
+            for (uint32_t c = 0; c < n_chunks; c++) {
+                digest = comb.combine(digest, h_crc[c], cs2[c]);
+                uint8_t cb[4] = {(uint8_t)(h_crc[c] >> 24), (uint8_t)(h_crc[c] >> 16),
+                                 (uint8_t)(h_crc[c] >> 8), (uint8_t)h_crc[c]};
+                digest = comb.combine(digest, crc32_update_bitwise(0, cb, 4), 4);
+            }
+        }
+
+        write_file(out_base + "-Data.db", h_data.data(), h_data.size());
+        write_file(out_base + "-Index.db", h_index.data(), h_index.size());
+        {
+            bytes f;
+            put_be32(f, (uint32_t)bs.k);
+            put_be32(f, (uint32_t)words);
+            f.insert(f.end(), h_bloom.begin(), h_bloom.end());
+            write_file(out_base + "-Filter.db", f.data(), f.size());
+        }
+        {
+            bytes ci;
+            std::string algo = "LZ4Compressor";
+            put_be16(ci, (uint16_t)algo.size());
+            ci.insert(ci.end(), algo.begin(), algo.end());
+            put_be32(ci, 0);
+            put_be32(ci, CHUNK_LEN);
+            put_be32(ci, 0x7FFFFFFF);
+            put_be64(ci, total_unc);
+            put_be32(ci, n_chunks);
+            uint64_t acc2 = 0;
+            for (uint32_t i = 0; i < n_chunks; i++) { put_be64(ci, acc2 + (uint64_t)i * 4); acc2 += cs[i]; }
+            write_file(out_base + "-CompressionInfo.db", ci.data(), ci.size());
+        }
+        {
+            std::string d = std::to_string(digest);
+            write_file(out_base + "-Digest.crc32", (const uint8_t*)d.data(), d.size());
+        }
+        {
+            bytes st = serialize_statistics_out(m);
+            write_file(out_base + "-Statistics.db", st.data(), st.size());
+        }
+        {
+            bytes s;
+            put_be32(s, 128); put_be32(s, 0); put_be64(s, 8); put_be32(s, 128); put_be32(s, 0);
+            write_file(out_base + "-Summary.db", s.data(), s.size());
+            std::string toc = "Data.db\nStatistics.db\nDigest.crc32\nTOC.txt\nCompressionInfo.db\nFilter.db\nIndex.db\nSummary.db\n";
+            write_file(out_base + "-TOC.txt", (const uint8_t*)toc.data(), toc.size());
+        }
+    }
+    w.uncompressed_len = total_unc;
+    float t01, t12, t23, t34;
+    HIP_CHECK(hipEventElapsedTime(&t01, ev0, ev1));
+    HIP_CHECK(hipEventElapsedTime(&t12, ev1, ev2));
+    HIP_CHECK(hipEventElapsedTime(&t23, ev2, ev3));
+    HIP_CHECK(hipEventElapsedTime(&t34, ev3, ev4));
+    w.ms_sizes = t01;
+    w.ms_serialize = t12;
+    w.ms_compress = t23;
+    w.ms_d2h = t34;
+    for (auto e : {ev0, ev1, ev2, ev3, ev4}) (void)hipEventDestroy(e);
+    return w;
+}
+
+}  // namespace gpuc
+
+// ---------------------------------------------------------------------------
+// C ABI
+// ---------------------------------------------------------------------------
+using namespace gpuc;
+
+extern "C" const char* gpuc_version(void) { return "cassandra_gpucompact 0.1 (gfx950)"; }
+
+extern "C" int gpuc_device_count(void) {
+    int n = 0;
+    if (hipGetDeviceCount(&n) != hipSuccess) return 0;
+    return n;
+}
+
+static void set_err(char* dst, size_t cap, const std::string& msg) {
+    if (!dst || !cap) return;
+    snprintf(dst, cap, "%s", msg.c_str());
+}
+
+extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
+    memset(res, 0, sizeof(*res));
+    double t_start_all;
+    auto wall = []() {
+        struct timespec ts;
+        clock_gettime(CLOCK_MONOTONIC, &ts);
+        return ts.tv_sec * 1e3 + ts.tv_nsec / 1e6;
+    };
+    t_start_all = wall();
+    try {
+        int ndev = gpuc_device_count();
+        if (ndev <= 0) { set_err(res->error, sizeof(res->error), "no HIP device (no CPU fallback)"); return GPUC_ERR_NO_GPU; }
+        HIP_CHECK(hipSetDevice(job->device));
+        hipStream_t stream;
+        HIP_CHECK(hipStreamCreate(&stream));
+        ensure_crc_tables(stream);
+
+        // ---- read + parse inputs (host metadata; Index parse on threads) ----
+        double t0 = wall();
+        int k = job->n_inputs;
+        if (k < 1 || k > 64) { set_err(res->error, sizeof(res->error), "n_inputs must be 1..64"); return GPUC_ERR_UNSUPPORTED; }
+        std::vector<bytes> comp_data(k), index_data(k);
+        std::vector<HCompressionInfo> cinfos(k);
+        std::vector<HStatistics> stats(k);
+        std::vector<uint64_t> generations(k);
+        for (int s = 0; s < k; s++) {
+            std::string base = job->input_bases[s];
+            comp_data[s] = read_file(base + "-Data.db");
+            index_data[s] = read_file(base + "-Index.db");
+            cinfos[s] = parse_compression_info(read_file(base + "-CompressionInfo.db"));
+            stats[s] = parse_statistics(read_file(base + "-Statistics.db"));
+            if (!stats[s].clustering_types.empty())
+                throw std::runtime_error("clustering columns unsupported in round 1 (GPU path)");
+            if (stats[s].regular_cols.size() != 1)
+                throw std::runtime_error("exactly one regular column supported in round 1");
+            if (!stats[s].partitioner.empty() && stats[s].partitioner.find("Murmur3") == std::string::npos)
+                throw std::runtime_error("Murmur3Partitioner required");
+            // generation from ".../oa-<id>-big"
+            size_t sl = base.rfind('/');
+            std::string name = sl == std::string::npos ? base : base.substr(sl + 1);
+            size_t a = name.find('-'), b2 = name.find('-', a + 1);
+            generations[s] = std::stoull(name.substr(a + 1, b2 - a - 1));
+            res->input_uncompressed_bytes += cinfos[s].data_len;
+        }
+        // column type of the single regular column
+        std::string col_type = stats[0].regular_cols[0].second;
+        int32_t col_fixed = -1;
+        if (col_type == "org.apache.cassandra.db.marshal.LongType") col_fixed = 8;
+        else if (col_type == "org.apache.cassandra.db.marshal.Int32Type") col_fixed = 4;
+
+        std::vector<std::vector<uint64_t>> positions(k);
+        std::vector<std::string> perr(k);
+        {
+            std::vector<std::thread> th;
+            for (int s = 0; s < k; s++)
+                th.emplace_back(parse_index_positions, std::cref(index_data[s]), cinfos[s].data_len,
+                                std::ref(positions[s]), std::ref(perr[s]));
+            for (auto& t : th) t.join();
+            for (int s = 0; s < k; s++)
+                if (!perr[s].empty()) throw std::runtime_error("Index.db parse: " + perr[s]);
+        }
+        double t1 = wall();
+        res->ms_read_io = t1 - t0;
+
+        // ---- H2D + decompress ----
+        hipEvent_t e0, e1, e2, e3, e4;
+        for (hipEvent_t* e : {&e0, &e1, &e2, &e3, &e4}) HIP_CHECK(hipEventCreate(e));
+        HIP_CHECK(hipEventRecord(e0, stream));
+        std::vector<DevBuf> d_comp(k), d_data(k), d_pos(k);
+        std::vector<ChunkDesc> chunks;
+        for (int s = 0; s < k; s++) {
+            d_comp[s].alloc(comp_data[s].size());
+            HIP_CHECK(hipMemcpyAsync(d_comp[s].p, comp_data[s].data(), comp_data[s].size(),
+                                     hipMemcpyHostToDevice, stream));
+            d_data[s].alloc(cinfos[s].data_len);
+            d_pos[s].alloc(positions[s].size() * 8);
+            HIP_CHECK(hipMemcpyAsync(d_pos[s].p, positions[s].data(), positions[s].size() * 8,
+                                     hipMemcpyHostToDevice, stream));
+            auto& ci = cinfos[s];
+            for (size_t c = 0; c < ci.offsets.size(); c++) {
+                uint64_t off = ci.offsets[c];
+                uint64_t end = c + 1 < ci.offsets.size() ? ci.offsets[c + 1] : comp_data[s].size();
+                ChunkDesc cd;
+                cd.comp = d_comp[s].as<uint8_t>() + off;
+                cd.out = d_data[s].as<uint8_t>() + c * (uint64_t)CHUNK_LEN;
+                cd.comp_len = (uint32_t)(end - off - 4);
+                cd.out_len = (uint32_t)std::min<uint64_t>(CHUNK_LEN, ci.data_len - c * (uint64_t)CHUNK_LEN);
+                chunks.push_back(cd);
+            }
+        }
+        DevBuf d_chunks, d_error;
+        d_chunks.alloc(chunks.size() * sizeof(ChunkDesc));
+        HIP_CHECK(hipMemcpyAsync(d_chunks.p, chunks.data(), chunks.size() * sizeof(ChunkDesc),
+                                 hipMemcpyHostToDevice, stream));
+        d_error.alloc(8);
+        HIP_CHECK(hipMemsetAsync(d_error.p, 0, 8, stream));
+        HIP_CHECK(hipEventRecord(e1, stream));
+        {
+            uint32_t wpb = 4;
+            uint32_t blocks = (uint32_t)((chunks.size() + wpb - 1) / wpb);
+            hipLaunchKernelGGL(k_lz4_decompress, dim3(blocks), dim3(WAVE * wpb), 0, stream,
+                               d_chunks.as<ChunkDesc>(), (uint32_t)chunks.size(), 1,
+                               d_error.as<unsigned long long>());
+        }
+        HIP_CHECK(hipEventRecord(e2, stream));
+
+        // ---- parse ----
+        uint64_t total_parts = 0;
+        std::vector<SrcDesc> srcs(k);
+        for (int s = 0; s < k; s++) {
+            srcs[s].data = d_data[s].as<uint8_t>();
+            srcs[s].part_pos = d_pos[s].as<uint64_t>();
+            srcs[s].n_parts = (uint32_t)(positions[s].size() - 1);
+            srcs[s].min_ts = stats[s].hdr_min_ts;
+            srcs[s].min_ldt = stats[s].hdr_min_ldt;
+            srcs[s].min_ttl = stats[s].hdr_min_ttl;
+            srcs[s].rec_base = (uint32_t)total_parts;
+            total_parts += srcs[s].n_parts;
+        }
+        res->partitions_in = total_parts;
+        if (total_parts > 0xFFFFFFFFull) throw std::runtime_error("too many partitions for one job");
+        DevBuf d_srcs, d_recs_a, d_recs_b, d_rows_in;
+        d_srcs.alloc(srcs.size() * sizeof(SrcDesc));
+        HIP_CHECK(hipMemcpyAsync(d_srcs.p, srcs.data(), srcs.size() * sizeof(SrcDesc),
+                                 hipMemcpyHostToDevice, stream));
+        d_recs_a.alloc(total_parts * sizeof(MRec));
+        d_recs_b.alloc(total_parts * sizeof(MRec));
+        d_rows_in.alloc(8);
+        HIP_CHECK(hipMemsetAsync(d_rows_in.p, 0, 8, stream));
+        // parsed columns SoA
+        DevBuf p_pdm, p_pdl, p_fl, p_lts, p_lttl, p_llet, p_rdm, p_rdl, p_cts, p_cldt, p_cttl, p_va, p_vl;
+        ParsedCols pc;
+        p_pdm.alloc(total_parts * 8); pc.pdel_mfda = p_pdm.as<int64_t>();
+        p_pdl.alloc(total_parts * 4); pc.pdel_ldt = p_pdl.as<uint32_t>();
+        p_fl.alloc(total_parts); pc.flags = p_fl.as<uint8_t>();
+        p_lts.alloc(total_parts * 8); pc.live_ts = p_lts.as<int64_t>();
+        p_lttl.alloc(total_parts * 4); pc.live_ttl = p_lttl.as<int32_t>();
+        p_llet.alloc(total_parts * 8); pc.live_let = p_llet.as<int64_t>();
+        p_rdm.alloc(total_parts * 8); pc.rdel_mfda = p_rdm.as<int64_t>();
+        p_rdl.alloc(total_parts * 4); pc.rdel_ldt = p_rdl.as<uint32_t>();
+        p_cts.alloc(total_parts * 8); pc.cell_ts = p_cts.as<int64_t>();
+        p_cldt.alloc(total_parts * 4); pc.cell_ldt = p_cldt.as<uint32_t>();
+        p_cttl.alloc(total_parts * 4); pc.cell_ttl = p_cttl.as<int32_t>();
+        p_va.alloc(total_parts * 8); pc.val_addr = p_va.as<uint64_t>();
+        p_vl.alloc(total_parts * 4); pc.val_len = p_vl.as<uint32_t>();
+        {
+            uint32_t blocks = (uint32_t)((total_parts + 255) / 256);
+            hipLaunchKernelGGL(k_parse, dim3(blocks), dim3(256), 0, stream, d_srcs.as<SrcDesc>(),
+                               (uint32_t)k, (uint32_t)total_parts, d_recs_a.as<MRec>(), pc,
+                               col_fixed, d_error.as<unsigned long long>(),
+                               d_rows_in.as<unsigned long long>());
+        }
+        HIP_CHECK(hipEventRecord(e3, stream));
+        {
+            unsigned long long err = 0;
+            HIP_CHECK(hipStreamSynchronize(stream));
+            HIP_CHECK(hipMemcpy(&err, d_error.p, 8, hipMemcpyDeviceToHost));
+            if (err) throw std::runtime_error("GPU decode/parse error code " + std::to_string(err));
+            HIP_CHECK(hipMemcpy(&res->rows_in, d_rows_in.p, 8, hipMemcpyDeviceToHost));
+        }
+
+        // ---- merge (pairwise rounds over pre-sorted source runs) ----
+        std::vector<uint64_t> runs;
+        runs.push_back(0);
+        for (int s = 0; s < k; s++) runs.push_back(runs.back() + srcs[s].n_parts);
+        MRec* d_sorted = merge_sorted_runs(d_recs_a.as<MRec>(), d_recs_b.as<MRec>(), runs, stream);
+
+        // ---- group heads + starts ----
+        DevBuf d_head, d_gstart, d_ngroups;
+        d_head.alloc(total_parts * 8);
+        d_gstart.alloc(total_parts * 8);
+        d_ngroups.alloc(8);
+        {
+            uint32_t blocks = (uint32_t)((total_parts + 255) / 256);
+            hipLaunchKernelGGL(k_group_heads, dim3(blocks), dim3(256), 0, stream, d_sorted,
+                               total_parts, d_head.as<uint64_t>());
+        }
+        exscan_u64(d_head.as<uint64_t>(), total_parts, stream);  // head -> exclusive scan... need inclusive-1
+        // recompute heads (scan consumed them); write starts via scan values
+        {
+            // d_head now holds exclusive scan of head flags == group id for head elements.
+            // Recreate head flags on the fly in k_group_starts by comparing recs again.
+            uint32_t blocks = (uint32_t)((total_parts + 255) / 256);
+            hipLaunchKernelGGL(k_group_starts2, dim3(blocks), dim3(256), 0, stream, d_sorted,
+                               total_parts, d_head.as<uint64_t>(), d_gstart.as<uint64_t>(),
+                               d_ngroups.as<uint64_t>());
+        }
+        uint64_t n_groups = 0;
+        HIP_CHECK(hipStreamSynchronize(stream));
+        HIP_CHECK(hipMemcpy(&n_groups, d_ngroups.p, 8, hipMemcpyDeviceToHost));
+
+        // ---- reconcile + purge ----
+        OutPartsBuf opb;
+        opb.alloc(n_groups);
+        DevBuf d_stats, d_tomb, d_srcbases;
+        d_stats.alloc(sizeof(OutStats));
+        init_outstats(d_stats, stream);
+        uint32_t tomb_cap = (uint32_t)std::min<uint64_t>(n_groups * 3 + 1024, 200000000ull);
+        d_tomb.alloc((uint64_t)tomb_cap * 4);
+        {
+            std::vector<uint32_t> bases(k);
+            for (int s = 0; s < k; s++) bases[s] = srcs[s].rec_base;
+            d_srcbases.alloc(k * 4);
+            HIP_CHECK(hipMemcpyAsync(d_srcbases.p, bases.data(), k * 4, hipMemcpyHostToDevice, stream));
+        }
+        DevBuf d_ov_lo, d_ov_hi, d_ov_ts;
+        PurgeParams pp{};
+        pp.now_sec = job->now_sec;
+        pp.gc_before = job->gc_before;
+        pp.never_purge = job->never_purge;
+        pp.enforce_strict_liveness = job->enforce_strict_liveness;
+        pp.n_overlaps = job->n_overlaps;
+        pp.has_shard = job->has_token_range;
+        pp.shard_lo = job->token_lo;
+        pp.shard_hi = job->token_hi;
+        if (job->n_overlaps > 0) {
+            std::vector<int64_t> lo(job->n_overlaps), hi(job->n_overlaps), ts(job->n_overlaps);
+            for (int i = 0; i < job->n_overlaps; i++) {
+                lo[i] = job->overlaps[i].token_lo;
+                hi[i] = job->overlaps[i].token_hi;
+                ts[i] = job->overlaps[i].min_timestamp;
+            }
+            d_ov_lo.alloc(job->n_overlaps * 8);
+            d_ov_hi.alloc(job->n_overlaps * 8);
+            d_ov_ts.alloc(job->n_overlaps * 8);
+            HIP_CHECK(hipMemcpyAsync(d_ov_lo.p, lo.data(), job->n_overlaps * 8, hipMemcpyHostToDevice, stream));
+            HIP_CHECK(hipMemcpyAsync(d_ov_hi.p, hi.data(), job->n_overlaps * 8, hipMemcpyHostToDevice, stream));
+            HIP_CHECK(hipMemcpyAsync(d_ov_ts.p, ts.data(), job->n_overlaps * 8, hipMemcpyHostToDevice, stream));
+            pp.ov_lo = d_ov_lo.as<int64_t>();
+            pp.ov_hi = d_ov_hi.as<int64_t>();
+            pp.ov_min_ts = d_ov_ts.as<int64_t>();
+        }
+        {
+            uint32_t blocks = (uint32_t)((n_groups + 255) / 256);
+            hipLaunchKernelGGL(k_reconcile, dim3(blocks), dim3(256), 0, stream, d_sorted,
+                               d_gstart.as<uint64_t>(), n_groups, total_parts,
+                               d_srcbases.as<uint32_t>(), pc, opb.op, pp, d_stats.as<OutStats>(),
+                               d_tomb.as<uint32_t>(), tomb_cap);
+        }
+        {
+            uint32_t blocks = (uint32_t)((n_groups + 255) / 256);
+            hipLaunchKernelGGL(k_collect_outstats, dim3(blocks), dim3(256), 0, stream, opb.op,
+                               n_groups, d_stats.as<OutStats>(), d_tomb.as<uint32_t>(), tomb_cap);
+        }
+        HIP_CHECK(hipEventRecord(e4, stream));
+        HIP_CHECK(hipStreamSynchronize(stream));
+
+        // ---- output header (SerializationHeader.make: desc-generation stats merge) ----
+        SerParams sp{};
+        {
+            std::vector<int> order(k);
+            for (int s = 0; s < k; s++) order[s] = s;
+            std::stable_sort(order.begin(), order.end(),
+                             [&](int a, int b) { return generations[a] > generations[b]; });
+            int64_t min_ts = INT64_MAX, min_ldt = INT64_MAX;
+            int32_t min_ttl = INT32_MAX;
+            for (int s : order) {
+                min_ts = std::min(min_ts, stats[s].min_timestamp);
+                min_ldt = std::min(min_ldt, stats[s].min_ldt);
+                min_ttl = std::min(min_ttl, stats[s].min_ttl);
+            }
+            sp.hs.min_ts = min_ts == NO_TIMESTAMP ? TIMESTAMP_EPOCH : min_ts;
+            sp.hs.min_ldt = min_ldt == NO_DELETION_TIME ? DELETION_TIME_EPOCH : min_ldt;
+            sp.hs.min_ttl = min_ttl == INT32_MAX ? 0 : min_ttl;
+            sp.col_fixed_len = col_fixed;
+        }
+
+        WriteDeviceOut w = write_sstable_device(opb, n_groups, sp, d_stats, d_tomb, tomb_cap,
+                                                job->output_base, stats[0].key_type,
+                                                stats[0].regular_cols, stream);
+        res->partitions_out = w.partitions;
+        res->rows_out = w.rows;
+        res->output_uncompressed_bytes = w.uncompressed_len;
+        res->output_compressed_bytes = w.compressed_len;
+        {
+            OutStats hst;
+            HIP_CHECK(hipMemcpy(&hst, d_stats.p, sizeof(OutStats), hipMemcpyDeviceToHost));
+            for (int i = 0; i < 64; i++) res->merged_counts[i] = hst.merged_counts[i];
+        }
+        float th2d, tdec, tparse, tmerge;
+        HIP_CHECK(hipEventElapsedTime(&th2d, e0, e1));
+        HIP_CHECK(hipEventElapsedTime(&tdec, e1, e2));
+        HIP_CHECK(hipEventElapsedTime(&tparse, e2, e3));
+        HIP_CHECK(hipEventElapsedTime(&tmerge, e3, e4));
+        res->ms_h2d = th2d;
+        res->ms_decompress = tdec;
+        res->ms_parse = tparse;
+        res->ms_merge = tmerge;  // includes merge+group+reconcile
+        res->ms_serialize = w.ms_sizes + w.ms_serialize;
+        res->ms_compress = w.ms_compress;
+        res->ms_d2h = w.ms_d2h;
+        // dominant kernel estimate
+        struct { const char* n; double ms; } ks[] = {
+            {"k_lz4_decompress", tdec}, {"k_parse", tparse}, {"merge+reconcile", tmerge},
+            {"k_serialize", w.ms_serialize}, {"k_lz4_compress", w.ms_compress}};
+        for (auto& kk : ks)
+            if (kk.ms > res->dominant_kernel_ms) {
+                res->dominant_kernel_ms = kk.ms;
+                snprintf(res->dominant_kernel, sizeof(res->dominant_kernel), "%s", kk.n);
+            }
+        res->dominant_kernel_launches = 1;
+        for (hipEvent_t e : {e0, e1, e2, e3, e4}) (void)hipEventDestroy(e);
+        HIP_CHECK(hipStreamDestroy(stream));
+        res->ms_total = wall() - t_start_all;
+        return GPUC_OK;
+    } catch (const std::exception& e) {
+        set_err(res->error, sizeof(res->error), e.what());
+        res->ms_total = wall() - t_start_all;
+        return GPUC_ERR_INTERNAL;
+    }
+}
+
+extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* error, size_t error_len) {
+    try {
+        if (gpuc_device_count() <= 0) { set_err(error, error_len, "no HIP device"); return GPUC_ERR_NO_GPU; }
+        HIP_CHECK(hipSetDevice(spec->device));
+        hipStream_t stream;
+        HIP_CHECK(hipStreamCreate(&stream));
+        ensure_crc_tables(stream);
+        uint64_t R = spec->rows_per_sstable;
+        uint64_t stride = R * (100 - spec->overlap_pct) / 100;
+        uint64_t universe = std::max<uint64_t>(stride * spec->n_sstables, R);
+        for (uint32_t s = 0; s < spec->n_sstables; s++) {
+            GenParams gp{};
+            gp.seed = spec->seed;
+            gp.universe = universe;
+            gp.stride = stride;
+            gp.rows = R;
+            gp.sst = s;
+            gp.value_len = spec->value_len;
+            gp.value_repeat_pct = spec->value_repeat_pct;
+            gp.tombstone_pct = spec->tombstone_pct;
+            gp.partition_del_pct = spec->partition_del_pct;
+            gp.base_ts = spec->base_ts;
+            gp.base_ldt = spec->base_ldt;
+            DevBuf d_a, d_b, d_ids, d_vals, d_stats, d_tomb;
+            d_a.alloc(R * sizeof(MRec));
+            d_b.alloc(R * sizeof(MRec));
+            d_ids.alloc(R * 8);
+            {
+                uint32_t blocks = (uint32_t)((R + 255) / 256);
+                hipLaunchKernelGGL(k_gen_recs, dim3(blocks), dim3(256), 0, stream, gp,
+                                   d_a.as<MRec>(), d_ids.as<uint64_t>());
+            }
+            MRec* d_sorted = merge_sort_recs(d_a.as<MRec>(), d_b.as<MRec>(), R, stream);
+            OutPartsBuf opb;
+            opb.alloc(R);
+            d_vals.alloc(R * (uint64_t)spec->value_len);
+            d_stats.alloc(sizeof(OutStats));
+            init_outstats(d_stats, stream);
+            uint32_t tomb_cap = (uint32_t)std::min<uint64_t>(R + 1024, 200000000ull);
+            d_tomb.alloc((uint64_t)tomb_cap * 4);
+            {
+                uint32_t blocks = (uint32_t)((R + 255) / 256);
+                hipLaunchKernelGGL(k_gen_fill, dim3(blocks), dim3(256), 0, stream, gp, d_sorted,
+                                   d_ids.as<uint64_t>(), R, opb.op, d_vals.as<uint8_t>());
+                hipLaunchKernelGGL(k_gen_values, dim3(blocks), dim3(256), 0, stream, gp, d_sorted,
+                                   d_ids.as<uint64_t>(), opb.op, R, d_vals.as<uint8_t>());
+            }
+            // collect stats over generated partitions; header mins derive from them
+            {
+                uint32_t blocks = (uint32_t)((R + 255) / 256);
+                hipLaunchKernelGGL(k_collect_outstats, dim3(blocks), dim3(256), 0, stream, opb.op,
+                                   R, d_stats.as<OutStats>(), d_tomb.as<uint32_t>(), tomb_cap);
+            }
+            OutStats hs0;
+            HIP_CHECK(hipStreamSynchronize(stream));
+            HIP_CHECK(hipMemcpy(&hs0, d_stats.p, sizeof(OutStats), hipMemcpyDeviceToHost));
+            SerParams sp{};
+            sp.hs.min_ts = hs0.min_ts_flip == 0xFFFFFFFFFFFFFFFFULL
+                               ? TIMESTAMP_EPOCH : (int64_t)(hs0.min_ts_flip ^ 0x8000000000000000ULL);
+            // header min_ldt: min over tombstone ldts only; live rows push
+            // NO_DELETION_TIME which maps to the epoch (EncodingStats ctor)
+            sp.hs.min_ldt = hs0.min_ldt_flip == 0xFFFFFFFFFFFFFFFFULL
+                                ? DELETION_TIME_EPOCH : (int64_t)(hs0.min_ldt_flip ^ 0x8000000000000000ULL);
+            if (sp.hs.min_ldt == NO_DELETION_TIME) sp.hs.min_ldt = DELETION_TIME_EPOCH;
+            sp.hs.min_ttl = 0;
+            sp.col_fixed_len = -1;  // val blob
+            std::string base = std::string(dir) + "/oa-" + std::to_string(spec->first_generation + s) + "-big";
+            std::vector<std::pair<bytes, std::string>> cols = {
+                {bytes{'v', 'a', 'l'}, "org.apache.cassandra.db.marshal.BytesType"}};
+            write_sstable_device(opb, R, sp, d_stats, d_tomb, tomb_cap, base,
+                                 "org.apache.cassandra.db.marshal.LongType", cols, stream);
+        }
+        HIP_CHECK(hipStreamDestroy(stream));
+        return GPUC_OK;
+    } catch (const std::exception& e) {
+        set_err(error, error_len, e.what());
+        return GPUC_ERR_INTERNAL;
+    }
+}

@@ -1,0 +1,130 @@
+/*
+ * libcassandra_gpucompact — MI355X-native SSTable compaction, C ABI.
+ *
+ * Drop-in boundary: replaces the hot loop of CompactionTask.runMayThrow
+ * (reference: src/java/org/apache/cassandra/db/compaction/CompactionTask.java:184-236)
+ * — scanners + CompactionIterator + CompactionAwareWriter — for eligible
+ * tables (big-format `oa`, Murmur3Partitioner, no clustering columns, one
+ * regular column, LZ4 chunk compression). The Java host above the seam
+ * (strategies, CompactionManager, LifecycleTransaction, metrics) is
+ * unchanged and binds these entry points via JNI/Panama.
+ *
+ * Inputs mirror what the loop consumes (CompactionTask.java:184-205):
+ *   - input sstable component paths (format version `oa`)
+ *   - merge params nowInSec/gcBefore (CompactionIterator.java:343-347)
+ *   - purge info: token-interval -> min-timestamp table for overlapping
+ *     non-compacting sources (CompactionController.getPurgeEvaluator,
+ *     CompactionController.java:247-286); empty table == no overlaps
+ *   - optional token-range restriction (CompactionManager.forceCompactionForTokenRange)
+ * Outputs mirror what the loop produces (CompactionTask.java:206-282):
+ *   - output sstable component files written under output_base (tmp names;
+ *     the host renames/commits under its LifecycleTransaction)
+ *   - the stats the task reports (bytes, partition/row counts, merge histogram)
+ *
+ * Threading: one call == one compaction task, synchronous; the host may run
+ * several concurrently on different devices (concurrent_compactors).
+ * All compute (decompress, decode, merge, reconcile, purge, serialize,
+ * compress, CRC, bloom) runs on the GPU; the library fails with
+ * GPUC_ERR_NO_GPU if no HIP device is available — there is no CPU fallback.
+ */
+#ifndef CASSANDRA_GPUCOMPACT_H
+#define CASSANDRA_GPUCOMPACT_H
+
+#include <stdint.h>
+#include <stddef.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+#define GPUC_OK 0
+#define GPUC_ERR_IO 1
+#define GPUC_ERR_FORMAT 2        /* unparseable/unsupported component bytes */
+#define GPUC_ERR_UNSUPPORTED 3   /* schema/feature outside current GPU scope */
+#define GPUC_ERR_NO_GPU 4
+#define GPUC_ERR_HIP 5
+#define GPUC_ERR_INTERNAL 6
+
+typedef struct gpuc_purge_range {
+    int64_t token_lo;   /* inclusive Murmur3 token bounds */
+    int64_t token_hi;
+    int64_t min_timestamp;
+} gpuc_purge_range;
+
+typedef struct gpuc_job {
+    /* input sstable path prefixes, e.g. "/data/ks/t-uuid/oa-12-big"
+       (components resolved as "<base>-Data.db" etc.) */
+    const char* const* input_bases;
+    int32_t n_inputs;
+
+    /* output path prefix; all component files are created as
+       "<output_base>-<Component>" (host renames under its txn log) */
+    const char* output_base;
+
+    int64_t now_sec;            /* nowInSec */
+    int64_t gc_before;          /* gcBefore seconds; INT64_MIN == gc nothing */
+    int32_t never_purge;        /* NEVER_PURGE_TOMBSTONES / table option */
+    int32_t enforce_strict_liveness;
+
+    const gpuc_purge_range* overlaps;  /* may be NULL when n_overlaps == 0 */
+    int32_t n_overlaps;
+
+    int32_t has_token_range;    /* shard restriction (multi-GPU token sharding) */
+    int64_t token_lo;           /* inclusive */
+    int64_t token_hi;           /* inclusive */
+
+    int32_t device;             /* HIP device ordinal */
+} gpuc_job;
+
+typedef struct gpuc_result {
+    uint64_t input_uncompressed_bytes;  /* ISSTableScanner.getLengthInBytes basis */
+    uint64_t output_uncompressed_bytes;
+    uint64_t output_compressed_bytes;   /* final Data.db size */
+    uint64_t partitions_in;             /* summed over inputs (versions) */
+    uint64_t partitions_out;
+    uint64_t rows_in;
+    uint64_t rows_out;
+    uint64_t merged_counts[64];         /* merge-arity histogram (CompactionTask.java:258-273) */
+    /* phase timings, milliseconds (GPU phases are HIP-event timed) */
+    double ms_read_io, ms_h2d, ms_decompress, ms_parse, ms_merge, ms_reconcile,
+           ms_serialize, ms_compress, ms_d2h, ms_write_io, ms_total;
+    /* dominant kernel bookkeeping for the roofline report */
+    char dominant_kernel[64];
+    double dominant_kernel_ms;
+    uint64_t dominant_kernel_launches;
+    char error[256];
+} gpuc_result;
+
+/* One compaction task. Returns GPUC_OK or an error code (message in result->error). */
+int gpuc_compact(const gpuc_job* job, gpuc_result* result);
+
+/* Synthetic sstable generation ON THE GPU using the product write path
+ * (serialize+compress+index+bloom kernels — the flush-path seed).
+ * Writes n_sstables sstables under dir as "oa-<first_generation+i>-big-*".
+ * Spec fields mirror oracle/src/gen.h (the shared generator contract). */
+typedef struct gpuc_gen_spec {
+    uint64_t seed;
+    uint32_t n_sstables;
+    uint64_t rows_per_sstable;
+    uint32_t overlap_pct;
+    uint32_t value_len;
+    uint32_t value_repeat_pct;
+    uint32_t tombstone_pct;
+    uint32_t partition_del_pct;
+    int64_t base_ts;
+    int64_t base_ldt;
+    uint64_t first_generation;
+    int32_t device;
+} gpuc_gen_spec;
+
+int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* error, size_t error_len);
+
+/* library/build identification */
+const char* gpuc_version(void);
+/* returns number of visible HIP devices (0 => gpuc_compact will fail) */
+int gpuc_device_count(void);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* CASSANDRA_GPUCOMPACT_H */

@@ -1,8 +1,13 @@
 // ORACLE + shared generator CONTRACT.
 // Deterministic synthetic sstable content derived from (seed, sstable index).
-// The GPU product implements the SAME derivation (cassandra_amd/csrc/gen.hip);
-// both sides must produce identical logical rows so parity tests can compare
-// oracle-written and GPU-written sstables byte for byte.
+// The GPU product implements the SAME derivation; both sides must produce
+// identical logical rows so parity tests can compare oracle-written and
+// GPU-written/compacted sstables byte for byte.
+//
+// Key scheme: a bijective Feistel permutation over [0, universe) gives each
+// sstable a distinct, dedup-free key set; sstable s covers permuted indices
+// j_global = (s*stride + j) mod universe for j in [0, rows). stride < rows
+// makes consecutive sstables overlap by exactly (rows-stride) keys.
 #pragma once
 #include "sstable.h"
 
@@ -12,7 +17,7 @@ struct GenSpec {
     uint64_t seed = 42;
     uint32_t n_sstables = 4;
     uint64_t rows_per_sstable = 1000;
-    uint64_t key_universe = 0;     // 0 => 10x total rows (low overlap)
+    uint32_t overlap_pct = 10;       // % of each sstable's keys shared with the previous one
     uint32_t value_len = 1024;
     uint32_t value_repeat_pct = 55;  // P(8-byte word repeats previous) -> LZ4 ratio knob
     uint32_t tombstone_pct = 0;      // percent of rows that are row-deletions
@@ -20,19 +25,35 @@ struct GenSpec {
     int64_t base_ts = 1700000000000000LL;   // µs
     int64_t base_ldt = 1700000000LL;        // seconds
     uint64_t first_generation = 1;
+
+    uint64_t stride() const { return rows_per_sstable * (100 - overlap_pct) / 100; }
+    uint64_t universe() const {
+        uint64_t u = stride() * n_sstables;
+        return u < rows_per_sstable ? rows_per_sstable : u;
+    }
 };
 
-// logical row for one (sstable, slot)
-struct GenRow {
-    uint64_t key_id;
-    int64_t ts;
-    bool is_tombstone;
-    uint32_t ldt;     // for tombstones (seconds)
-};
+// 4-round balanced Feistel on 2*hb bits with cycle-walking down to [0, universe)
+inline uint64_t feistel_perm(uint64_t seed, uint64_t universe, uint64_t x) {
+    int hb = 1;
+    while ((1ULL << (2 * hb)) < universe) hb++;
+    uint64_t mask = (1ULL << hb) - 1;
+    do {
+        uint64_t l = (x >> hb) & mask, r = x & mask;
+        for (int round = 0; round < 4; round++) {
+            uint64_t f = splitmix64(seed ^ r ^ ((uint64_t)(round + 1) << 56)) & mask;
+            uint64_t nl = r;
+            r = l ^ f;
+            l = nl;
+        }
+        x = (l << hb) | r;
+    } while (x >= universe);
+    return x;
+}
 
-// Derivations (all splitmix64-based; GPU mirrors these exactly):
 inline uint64_t gen_key_id(const GenSpec& g, uint32_t sst, uint64_t j) {
-    return splitmix64(g.seed ^ (0x5EEDULL + sst) * 0x9E3779B97f4A7C15ULL ^ (j * 0xA24BAED4963EE407ULL)) % g.key_universe;
+    uint64_t u = g.universe();
+    return feistel_perm(g.seed, u, (sst * g.stride() + j) % u);
 }
 inline int64_t gen_ts(const GenSpec& g, uint32_t sst, uint64_t key_id) {
     return g.base_ts + (int64_t)(splitmix64(g.seed ^ key_id * 31 ^ ((uint64_t)sst << 48)) % 1000000000ULL);
@@ -44,6 +65,9 @@ inline bool gen_is_tombstone(const GenSpec& g, uint32_t sst, uint64_t key_id) {
 inline bool gen_has_partition_del(const GenSpec& g, uint32_t sst, uint64_t key_id) {
     if (g.partition_del_pct == 0) return false;
     return splitmix64(g.seed ^ 0xFEEDULL ^ key_id ^ ((uint64_t)sst << 32)) % 100 < g.partition_del_pct;
+}
+inline uint32_t gen_ldt(const GenSpec& g, uint64_t key_id, uint64_t salt) {
+    return (uint32_t)(g.base_ldt + (int64_t)(splitmix64(key_id ^ salt) % 1000));
 }
 // value: value_len bytes in 8-byte words; word w repeats word w-1 with prob repeat_pct
 inline void gen_value(const GenSpec& g, uint32_t sst, uint64_t key_id, bytes& out) {
@@ -61,7 +85,7 @@ inline void gen_value(const GenSpec& g, uint32_t sst, uint64_t key_id, bytes& ou
     }
 }
 
-// build one full synthetic sstable (sorted, deduped keys)
+// build one full synthetic sstable (sorted keys)
 SSTable generate_sstable(const GenSpec& g, uint32_t sst_index);
 
 }  // namespace oracle

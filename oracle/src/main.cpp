@@ -86,7 +86,7 @@ static GenSpec spec_from_kv(std::map<std::string, std::string>& kv) {
     g.seed = geti("seed", g.seed);
     g.n_sstables = (uint32_t)geti("n", g.n_sstables);
     g.rows_per_sstable = geti("rows", g.rows_per_sstable);
-    g.key_universe = geti("universe", g.key_universe);
+    g.overlap_pct = (uint32_t)geti("overlap", g.overlap_pct);
     g.value_len = (uint32_t)geti("vlen", g.value_len);
     g.value_repeat_pct = (uint32_t)geti("vrep", g.value_repeat_pct);
     g.tombstone_pct = (uint32_t)geti("tomb", g.tombstone_pct);

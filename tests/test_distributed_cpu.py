@@ -1,0 +1,51 @@
+"""Multi-process (gloo, world_size 2) coverage of the distributed host logic
+used by bench.py: token-shard assignment and whole-job aggregation (max-over-
+ranks timing, summed byte counts). No GPU; the data path has no collectives."""
+import os
+import sys
+
+import torch
+import torch.multiprocessing as mp
+
+from cassandra_amd.sharding import split_token_range, shards_cover_ring, TOKEN_MIN, TOKEN_MAX
+
+
+def test_shards_cover_ring():
+    for n in (1, 2, 4, 8, 5, 13):
+        assert shards_cover_ring(n), n
+    assert split_token_range(1, 0) == (TOKEN_MIN, TOKEN_MAX)
+
+
+def _worker(rank, world, rv):
+    import torch.distributed as dist
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29511")
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    lo, hi = split_token_range(world, rank)
+    # fake per-rank job results (what bench.py aggregates)
+    my_bytes = float(1000 * (rank + 1))
+    my_time = 1.0 + 0.25 * rank
+    t = torch.tensor([my_bytes])
+    dist.all_reduce(t, op=dist.ReduceOp.SUM)
+    tm = torch.tensor([my_time])
+    dist.all_reduce(tm, op=dist.ReduceOp.MAX)
+    if rank == 0:
+        rv["bytes"] = t.item()
+        rv["time"] = tm.item()
+        rv["shard0"] = (lo, hi)
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_gloo_aggregation():
+    ctx = mp.get_context("spawn")
+    with ctx.Manager() as mgr:
+        rv = mgr.dict()
+        ps = [ctx.Process(target=_worker, args=(r, 2, rv)) for r in range(2)]
+        [p.start() for p in ps]
+        [p.join(120) for p in ps]
+        assert all(p.exitcode == 0 for p in ps)
+        assert rv["bytes"] == 3000.0
+        assert abs(rv["time"] - 1.25) < 1e-6
+        lo, hi = rv["shard0"]
+        assert lo == TOKEN_MIN and hi == -1

@@ -1,0 +1,145 @@
+"""GPU parity tests: the product's GPU compaction output must be
+BYTE-IDENTICAL to the CPU oracle on the same inputs, across every component
+file. Inputs are synthetic (shared generator contract) — /root/reference is
+never touched at run time (it does not exist on the GPU box).
+"""
+import filecmp
+import os
+import subprocess
+
+import pytest
+
+from conftest import ORACLE
+
+pytestmark = pytest.mark.gpu
+
+COMPONENTS = ["Data.db", "Index.db", "CompressionInfo.db", "Filter.db",
+              "Digest.crc32", "Statistics.db", "Summary.db", "TOC.txt"]
+
+
+@pytest.fixture(scope="module")
+def ca(product_lib):
+    import cassandra_amd as mod
+    if mod.device_count() < 1:
+        pytest.skip("no GPU")
+    return mod
+
+
+def _oracle_gen(d, **kw):
+    args = [f"{k}={v}" for k, v in kw.items()]
+    subprocess.run([ORACLE, "gen", d, *args], check=True, capture_output=True)
+
+
+def _oracle_compact(outbase, inputs, **kw):
+    args = [f"{k}={v}" for k, v in kw.items()]
+    subprocess.run([ORACLE, "compact", outbase, *inputs, *args], check=True,
+                   capture_output=True)
+
+
+def _assert_dirs_equal(base_a, base_b, components=COMPONENTS):
+    for c in components:
+        fa, fb = f"{base_a}-{c}", f"{base_b}-{c}"
+        assert os.path.exists(fa), f"missing {fa}"
+        assert os.path.exists(fb), f"missing {fb}"
+        if not filecmp.cmp(fa, fb, shallow=False):
+            a = open(fa, "rb").read()
+            b = open(fb, "rb").read()
+            i = next((j for j in range(min(len(a), len(b))) if a[j] != b[j]), min(len(a), len(b)))
+            raise AssertionError(
+                f"{c}: len {len(a)} vs {len(b)}, first diff at {i} "
+                f"({a[i:i+8].hex() if i < len(a) else '-'} vs {b[i:i+8].hex() if i < len(b) else '-'})")
+
+
+def test_generator_parity(ca, oracle_bin, tmp_path):
+    """GPU write path (serialize+compress+index+bloom kernels) == oracle writer."""
+    dg, do = str(tmp_path / "gpu"), str(tmp_path / "cpu")
+    os.makedirs(dg), os.makedirs(do)
+    kw = dict(seed=13, n=3, rows=2000, vlen=700, overlap=25, tomb=10, pdel=2)
+    ca.generate(dg, seed=13, n_sstables=3, rows_per_sstable=2000, value_len=700,
+                overlap_pct=25, tombstone_pct=10, partition_del_pct=2)
+    _oracle_gen(do, **kw)
+    for g in (1, 2, 3):
+        _assert_dirs_equal(f"{dg}/oa-{g}-big", f"{do}/oa-{g}-big")
+
+
+@pytest.mark.parametrize("case", [
+    dict(name="plain", gen=dict(seed=42, n=4, rows=3000, vlen=512, overlap=10), job={}),
+    dict(name="tombstones_nogc", gen=dict(seed=17, n=4, rows=2500, vlen=256, overlap=30,
+                                          tomb=20, pdel=5), job={}),
+    dict(name="tombstones_gc", gen=dict(seed=17, n=4, rows=2500, vlen=256, overlap=30,
+                                        tomb=20, pdel=5),
+         job=dict(gc_before=2000000000)),
+    dict(name="never_purge", gen=dict(seed=17, n=4, rows=2500, vlen=256, overlap=30, tomb=20),
+         job=dict(gc_before=2000000000, never_purge=True)),
+    dict(name="single_input", gen=dict(seed=8, n=1, rows=4000, vlen=300, overlap=0, tomb=10),
+         job={}),
+    dict(name="all_overlap", gen=dict(seed=29, n=4, rows=1500, vlen=128, overlap=100), job={}),
+    dict(name="incompressible", gen=dict(seed=31, n=2, rows=2000, vlen=900, overlap=10,
+                                         vrep=0), job={}),
+    dict(name="highly_compressible", gen=dict(seed=33, n=2, rows=2000, vlen=900, overlap=10,
+                                              vrep=97), job={}),
+])
+def test_compaction_parity(ca, oracle_bin, tmp_path, case):
+    d = str(tmp_path)
+    gen = case["gen"]
+    _oracle_gen(d, **gen)
+    ins = [f"{d}/oa-{g}-big" for g in range(1, gen["n"] + 1)]
+    job = dict(case["job"])
+    okw = {}
+    if "gc_before" in job:
+        okw["gcbefore"] = job["gc_before"]
+    if job.pop("never_purge", False):
+        okw["nevergc"] = 1
+        job["never_purge"] = True
+    _oracle_compact(f"{d}/oa-90-big", ins, **okw)
+    ca.compact(ins, f"{d}/oa-91-big", **job)
+    _assert_dirs_equal(f"{d}/oa-90-big", f"{d}/oa-91-big")
+
+
+def test_purge_overlap_table_parity(ca, oracle_bin, tmp_path):
+    """gcBefore purge gated by the token-interval min-timestamp table."""
+    d = str(tmp_path)
+    _oracle_gen(d, seed=55, n=3, rows=2000, vlen=128, overlap=20, tomb=30)
+    ins = [f"{d}/oa-{g}-big" for g in (1, 2, 3)]
+    # overlap table: left half of the token space has an old overlapping source
+    # (min_timestamp below every generated ts -> nothing there may purge)
+    lo, hi, ts = -(2**63), 0, 1
+    # oracle CLI has no overlap flag yet -> emulate with two sharded runs:
+    # shard A (overlapped: purge denied == nevergc), shard B (purge allowed)
+    _oracle_compact(f"{d}/oa-80-big", ins, shard=f"{lo}:{hi}",
+                    gcbefore=2000000000, nevergc=1)
+    _oracle_compact(f"{d}/oa-81-big", ins, shard=f"{hi+1}:{2**63-1}",
+                    gcbefore=2000000000)
+    ca.compact(ins, f"{d}/oa-90-big", gc_before=2000000000,
+               overlaps=[(lo, hi, ts)], token_range=(lo, hi))
+    ca.compact(ins, f"{d}/oa-91-big", gc_before=2000000000,
+               overlaps=[(lo, hi, ts)], token_range=(hi + 1, 2**63 - 1))
+    _assert_dirs_equal(f"{d}/oa-80-big", f"{d}/oa-90-big", ["Data.db", "Index.db"])
+    _assert_dirs_equal(f"{d}/oa-81-big", f"{d}/oa-91-big", ["Data.db", "Index.db"])
+
+
+def test_token_shards_compose(ca, oracle_bin, tmp_path):
+    """Disjoint token shards partition the full output exactly (§8(e))."""
+    d = str(tmp_path)
+    _oracle_gen(d, seed=61, n=2, rows=3000, vlen=200, overlap=10)
+    ins = [f"{d}/oa-{g}-big" for g in (1, 2)]
+    full = ca.compact(ins, f"{d}/oa-90-big")
+    bounds = [-(2**63), -(2**62), 0, 2**62, 2**63 - 1]
+    total = 0
+    for i in range(4):
+        lo = bounds[i] if i == 0 else bounds[i] + 1
+        r = ca.compact(ins, f"{d}/oa-9{i+1}-big", token_range=(lo, bounds[i + 1]))
+        total += r["partitions_out"]
+    assert total == full["partitions_out"]
+
+
+def test_reads_reference_written_inputs(ca, oracle_bin, tmp_path):
+    """GPU path consumes oracle-written sstables (which byte-round-trip the
+    reference's own fixtures) — covers the full read path against the same
+    wire format the reference writes."""
+    d = str(tmp_path)
+    _oracle_gen(d, seed=77, n=2, rows=1000, vlen=100, overlap=0)
+    ins = [f"{d}/oa-{g}-big" for g in (1, 2)]
+    r = ca.compact(ins, f"{d}/oa-90-big")
+    assert r["partitions_in"] == 2000
+    assert r["input_uncompressed_bytes"] > 0
