@@ -710,10 +710,11 @@ WriterOut write_sstable(const SSTable& t) {
     if (!t.parts.empty()) { st.first_key = t.parts.front().key; st.last_key = t.parts.back().key; }
 
     ChunkedOut co = chunk_compress(data_raw, t.comp);
-    st.compression_ratio = (double)co.file.size() / (double)std::max<size_t>(1, data_raw.size());
-    // NOTE: reference compressionRatio = compressed/uncompressed from MetadataCollector
-    // (MetadataCollector.addCompressionRatio) — chunk bytes without CRCs. Kept approximate
-    // for round 1 (Statistics.db STATS is not in the judged bit-exact set).
+    // compressionRatio = compressedSize/uncompressedSize where compressedSize
+    // excludes the per-chunk CRC words (CompressedSequentialWriter.java:158,
+    // chunkOffset += compressedLength + 4) — matches the product's writer.
+    st.compression_ratio = data_raw.empty() ? -1.0
+        : (double)(co.file.size() - 4 * co.offsets.size()) / (double)data_raw.size();
     w.data_db = std::move(co.file);
     w.compression_info = make_compression_info(t.comp, data_raw.size(), co.offsets);
     w.filter = serialize_bloom(bloom);
