@@ -651,7 +651,9 @@ __global__ void k_group_starts2(const MRec* recs, uint64_t n, const uint64_t* he
     if (i >= n) return;
     bool head = (i == 0) || !mrec_eq(recs[i], recs[i - 1]);
     if (head) group_start[head_scan[i]] = i;
-    if (i == n - 1) *n_groups = head_scan[i] + 1;
+    // total groups = exclusive-scan of flags at the last element PLUS its own
+    // flag (the last record need not start a group)
+    if (i == n - 1) *n_groups = head_scan[i] + (head ? 1 : 0);
 }
 
 __global__ void k_init_outstats(OutStats* st) {
