@@ -463,10 +463,26 @@ static void init_outstats(DevBuf& d_stats, hipStream_t stream) {
 }
 
 static bool g_crc_tables_ready = false;
+static void* g_crc256 = nullptr;  // 256-entry CRC table, device-resident
 static void ensure_crc_tables(hipStream_t stream) {
     if (g_crc_tables_ready) return;
     hipLaunchKernelGGL(k_crc_init, dim3(1), dim3(256), 0, stream);
+    // plain 256-entry table for the wave LZ4 kernels
+    uint32_t tab[256];
+    crc32_make_table(tab);
+    HIP_CHECK(hipMalloc(&g_crc256, sizeof(tab)));
+    HIP_CHECK(hipMemcpyAsync(g_crc256, tab, sizeof(tab), hipMemcpyHostToDevice, stream));
+    // verify the LDS same-address write-order rule the wave compressor relies on
+    unsigned int* d_probe;
+    HIP_CHECK(hipMalloc(&d_probe, 8));
+    hipLaunchKernelGGL(k_probe_lds_order, dim3(1), dim3(WAVE), 0, stream, d_probe);
+    unsigned int probe[2];
     HIP_CHECK(hipStreamSynchronize(stream));
+    HIP_CHECK(hipMemcpy(probe, d_probe, 8, hipMemcpyDeviceToHost));
+    (void)hipFree(d_probe);
+    if (probe[0] != 63 || probe[1] != 0)
+        throw std::runtime_error("LDS conflicting-write order is not highest-lane-wins on this device; "
+                                 "wave LZ4 compressor would be incorrect (refusing to run)");
     g_crc_tables_ready = true;
 }
 
@@ -575,9 +591,10 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, uint64_t n_gr
     d_csize.alloc((uint64_t)n_chunks * 4);
     d_ccrc.alloc((uint64_t)n_chunks * 4);
     d_csize64.alloc((uint64_t)n_chunks * 8);
-    hipLaunchKernelGGL(k_lz4_compress, dim3(n_chunks), dim3(WAVE), 0, stream,
+    hipLaunchKernelGGL(k_lz4_compress_wave, dim3(n_chunks), dim3(WAVE), 0, stream,
                        d_out_data.as<uint8_t>(), total_unc, d_slots.as<uint8_t>(),
-                       d_csize.as<uint32_t>(), d_ccrc.as<uint32_t>(), n_chunks);
+                       d_csize.as<uint32_t>(), d_ccrc.as<uint32_t>(), n_chunks,
+                       (const uint32_t*)g_crc256);
     // widen sizes to u64 + scan -> chunk offsets (excluding per-chunk CRC;
     // gather adds 4*c for preceding CRCs)
     {
@@ -833,13 +850,9 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
         d_error.alloc(8);
         HIP_CHECK(hipMemsetAsync(d_error.p, 0, 8, stream));
         HIP_CHECK(hipEventRecord(e1, stream));
-        {
-            uint32_t wpb = 4;
-            uint32_t blocks = (uint32_t)((chunks.size() + wpb - 1) / wpb);
-            hipLaunchKernelGGL(k_lz4_decompress, dim3(blocks), dim3(WAVE * wpb), 0, stream,
-                               d_chunks.as<ChunkDesc>(), (uint32_t)chunks.size(), 1,
-                               d_error.as<unsigned long long>());
-        }
+        hipLaunchKernelGGL(k_lz4_decompress_wave, dim3((uint32_t)chunks.size()), dim3(WAVE), 0,
+                           stream, d_chunks.as<ChunkDesc>(), (uint32_t)chunks.size(), 1,
+                           d_error.as<unsigned long long>(), (const uint32_t*)g_crc256);
         HIP_CHECK(hipEventRecord(e2, stream));
 
         // ---- parse ----

@@ -609,38 +609,85 @@ __global__ void k_reconcile(const MRec* recs, const uint64_t* group_start, uint6
 // ---------------------------------------------------------------------------
 __global__ void k_collect_outstats(OutParts op, uint64_t n, OutStats* st,
                                    uint32_t* tomb_ldts, uint32_t tomb_cap) {
+    __shared__ unsigned long long sh_parts, sh_rows, sh_cells, sh_mints, sh_maxts,
+        sh_minldt, sh_maxldt, sh_first, sh_last, sh_haspdel;
+    __shared__ unsigned int sh_minttl, sh_maxttl;
+    if (threadIdx.x == 0) {
+        sh_parts = sh_rows = sh_cells = 0;
+        sh_mints = sh_minldt = 0xFFFFFFFFFFFFFFFFULL;
+        sh_maxts = sh_maxldt = 0;
+        sh_first = 0xFFFFFFFFFFFFFFFFULL;
+        sh_last = 0;
+        sh_haspdel = 0;
+        sh_minttl = 0xFFFFFFFFu;
+        sh_maxttl = 0;
+    }
+    __syncthreads();
     uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (g >= n || !op.keep[g]) return;
-    uint8_t of = op.flags[g];
-    atomicAdd(&st->partitions_out, 1ull);
-    if (of & PF_HAS_ROW) atomicAdd(&st->rows_out, 1ull);
-    atomicMin(&st->first_group, (unsigned long long)g);
-    atomicMax(&st->last_group, (unsigned long long)g);
-    bool pdel_live = op.pdel_mfda[g] == INT64_MIN && op.pdel_ldt[g] == LDT_NONE_U32;
-    if (!pdel_live) {
-        atomicExch(&st->has_partition_deletions, 1ull);
-        stat_ts(st, op.pdel_mfda[g]);
-        stat_ldt(st, ldt_long(op.pdel_ldt[g]));
-        tomb_push(st, tomb_ldts, tomb_cap, op.pdel_ldt[g]);
+    if (g < n && op.keep[g]) {
+        uint8_t of = op.flags[g];
+        atomicAdd(&sh_parts, 1ull);
+        if (of & PF_HAS_ROW) atomicAdd(&sh_rows, 1ull);
+        atomicMin(&sh_first, (unsigned long long)g);
+        atomicMax(&sh_last, (unsigned long long)g);
+        auto lts = [&](int64_t ts) {
+            if (ts == NO_TIMESTAMP) return;
+            unsigned long long v = (unsigned long long)ts ^ 0x8000000000000000ULL;
+            atomicMin(&sh_mints, v);
+            atomicMax(&sh_maxts, v);
+        };
+        auto lldt = [&](int64_t l) {
+            unsigned long long v = (unsigned long long)l ^ 0x8000000000000000ULL;
+            atomicMin(&sh_minldt, v);
+            atomicMax(&sh_maxldt, v);
+        };
+        bool pdel_live = op.pdel_mfda[g] == INT64_MIN && op.pdel_ldt[g] == LDT_NONE_U32;
+        if (!pdel_live) {
+            atomicExch(&sh_haspdel, 1ull);
+            lts(op.pdel_mfda[g]);
+            lldt(ldt_long(op.pdel_ldt[g]));
+            tomb_push(st, tomb_ldts, tomb_cap, op.pdel_ldt[g]);
+        }
+        if (of & PF_LIVE_TS) {
+            lts(op.live_ts[g]);
+            if (op.live_ttl[g] != 0) {
+                lldt(op.live_let[g]);
+                atomicMin(&sh_minttl, (unsigned)op.live_ttl[g]);
+                atomicMax(&sh_maxttl, (unsigned)op.live_ttl[g]);
+            } else lldt(NO_DELETION_TIME);
+        }
+        if (of & PF_ROW_DEL) {
+            lts(op.rdel_mfda[g]);
+            lldt(ldt_long(op.rdel_ldt[g]));
+            tomb_push(st, tomb_ldts, tomb_cap, op.rdel_ldt[g]);
+        }
+        if (of & PF_HAS_CELL) {
+            atomicAdd(&sh_cells, 1ull);
+            lts(op.cell_ts[g]);
+            uint32_t cldt = op.cell_ldt[g];
+            int32_t cttl = op.cell_ttl[g];
+            if (cldt != LDT_NONE_U32 && cttl == 0) { lldt(ldt_long(cldt)); tomb_push(st, tomb_ldts, tomb_cap, cldt); }
+            else if (cttl != 0) {
+                lldt(ldt_long(cldt));
+                atomicMin(&sh_minttl, (unsigned)cttl);
+                atomicMax(&sh_maxttl, (unsigned)cttl);
+            } else lldt(NO_DELETION_TIME);
+        }
     }
-    if (of & PF_LIVE_TS) {
-        stat_ts(st, op.live_ts[g]);
-        if (op.live_ttl[g] != 0) { stat_ldt(st, op.live_let[g]); stat_ttl(st, op.live_ttl[g]); }
-        else stat_ldt(st, NO_DELETION_TIME);
-    }
-    if (of & PF_ROW_DEL) {
-        stat_ts(st, op.rdel_mfda[g]);
-        stat_ldt(st, ldt_long(op.rdel_ldt[g]));
-        tomb_push(st, tomb_ldts, tomb_cap, op.rdel_ldt[g]);
-    }
-    if (of & PF_HAS_CELL) {
-        atomicAdd(&st->total_cells, 1ull);
-        stat_ts(st, op.cell_ts[g]);
-        uint32_t cldt = op.cell_ldt[g];
-        int32_t cttl = op.cell_ttl[g];
-        if (cldt != LDT_NONE_U32 && cttl == 0) { stat_ldt(st, ldt_long(cldt)); tomb_push(st, tomb_ldts, tomb_cap, cldt); }
-        else if (cttl != 0) { stat_ldt(st, ldt_long(cldt)); stat_ttl(st, cttl); }
-        else stat_ldt(st, NO_DELETION_TIME);
+    __syncthreads();
+    if (threadIdx.x == 0 && sh_parts) {
+        atomicAdd(&st->partitions_out, sh_parts);
+        atomicAdd(&st->rows_out, sh_rows);
+        atomicAdd(&st->total_cells, sh_cells);
+        atomicMin(&st->min_ts_flip, sh_mints);
+        atomicMax(&st->max_ts_flip, sh_maxts);
+        atomicMin(&st->min_ldt_flip, sh_minldt);
+        atomicMax(&st->max_ldt_flip, sh_maxldt);
+        atomicMin(&st->min_ttl, sh_minttl);
+        atomicMax(&st->max_ttl, sh_maxttl);
+        atomicMin(&st->first_group, sh_first);
+        atomicMax(&st->last_group, sh_last);
+        if (sh_haspdel) atomicExch(&st->has_partition_deletions, 1ull);
     }
 }
 
@@ -749,23 +796,30 @@ __device__ inline uint64_t d_partition_size(const OutParts& op, uint64_t g, cons
 __global__ void k_sizes(OutParts op, uint64_t n, SerParams sp, uint64_t* psize,
                         OutStats* st, const int64_t* ps_hist_off, int32_t ps_hist_n,
                         const int64_t* ch_hist_off, int32_t ch_hist_n) {
+    __shared__ unsigned int sh_ps[156], sh_ch[119];
+    for (int i = threadIdx.x; i < 156; i += blockDim.x) sh_ps[i] = 0;
+    for (int i = threadIdx.x; i < 119; i += blockDim.x) sh_ch[i] = 0;
+    __syncthreads();
     uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (g >= n) return;
-    uint64_t s = d_partition_size(op, g, sp);
-    psize[g] = s;
-    if (!s) return;
-    // EstimatedHistogram.add: first offset >= value (Java Arrays.binarySearch insertion)
-    {
-        int lo = 0, hi = ps_hist_n;  // offsets count; bucket index in [0, n] (n = overflow)
-        while (lo < hi) { int mid = (lo + hi) >> 1; if ((uint64_t)ps_hist_off[mid] < s) lo = mid + 1; else hi = mid; }
-        atomicAdd(&st->part_size_hist[lo], 1ull);
+    if (g < n) {
+        uint64_t s = d_partition_size(op, g, sp);
+        psize[g] = s;
+        if (s) {
+            // EstimatedHistogram.add: first offset >= value
+            int lo = 0, hi = ps_hist_n;
+            while (lo < hi) { int mid = (lo + hi) >> 1; if ((uint64_t)ps_hist_off[mid] < s) lo = mid + 1; else hi = mid; }
+            atomicAdd(&sh_ps[lo], 1u);
+            uint64_t cells = (op.flags[g] & PF_HAS_CELL) ? 1 : 0;
+            lo = 0; hi = ch_hist_n;
+            while (lo < hi) { int mid = (lo + hi) >> 1; if ((uint64_t)ch_hist_off[mid] < cells) lo = mid + 1; else hi = mid; }
+            atomicAdd(&sh_ch[lo], 1u);
+        }
     }
-    {
-        uint64_t cells = (op.flags[g] & PF_HAS_CELL) ? 1 : 0;
-        int lo = 0, hi = ch_hist_n;
-        while (lo < hi) { int mid = (lo + hi) >> 1; if ((uint64_t)ch_hist_off[mid] < cells) lo = mid + 1; else hi = mid; }
-        atomicAdd(&st->cells_hist[lo], 1ull);
-    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < 156; i += blockDim.x)
+        if (sh_ps[i]) atomicAdd(&st->part_size_hist[i], (unsigned long long)sh_ps[i]);
+    for (int i = threadIdx.x; i < 119; i += blockDim.x)
+        if (sh_ch[i]) atomicAdd(&st->cells_hist[i], (unsigned long long)sh_ch[i]);
 }
 
 __global__ void k_index_sizes(OutParts op, uint64_t n, const uint64_t* data_off,
@@ -1057,3 +1111,5 @@ __global__ void k_merge_uniform(const MRec* in, MRec* out, uint64_t n, uint64_t 
 }
 
 }  // namespace gpuc
+
+#include "lz4_wave.h"

@@ -1,0 +1,360 @@
+// PRODUCT — wave-cooperative LZ4 kernels for CDNA4 (gfx950).
+//
+// Bit-exact reimplementation of LZ4_compress_default (liblz4 1.9.3, byU16,
+// acceleration 1 — the scalar spec lives in lz4_model.h and is fuzz-pinned to
+// the system library) where the sequential hash-probe chain is executed as
+// 64-position speculative windows across the wavefront:
+//
+//   * the probe sequence q_{k+1} = q_k + ((64+k)>>6) advances with a CONSTANT
+//     stride inside each 64-probe window (stride w+1 in window w), so a wave
+//     evaluates one window per step: parallel read32+hash, candidate lookup,
+//     match test, then commits table updates only up to the first match/abort
+//     — byte-identical to the scalar loop.
+//   * in-window hash duplicates are resolved with two marker write/read-back
+//     rounds on the LDS hash table. This relies on the CDNA LDS conflict rule
+//     that simultaneous writes to one address retire in lane order (highest
+//     lane wins); k_probe_lds_order verifies it at init and the host refuses
+//     to run if it does not hold.
+//   * match extension, catch-up, literal copies and 255-run emission are
+//     wave-parallel (ballots + strided copies).
+//
+// Decompression runs the token chain redundantly on all 64 lanes (uniform
+// control flow; operands come from LDS broadcasts) with wave-parallel copies;
+// overlapped match copies use the periodic form dst[i] = src[i % offset].
+#pragma once
+#include <hip/hip_runtime.h>
+#include "codec.h"
+#include "lz4_model.h"
+
+namespace gpuc {
+
+#ifndef WAVE
+#define WAVE 64
+#endif
+
+__device__ inline uint64_t wave_ballot(bool p) { return __ballot(p); }
+
+// probe: do conflicting LDS writes retire highest-lane-last?
+__global__ void k_probe_lds_order(unsigned int* out) {
+    __shared__ unsigned int slot[1];
+    __shared__ unsigned int slot2[1];
+    int lane = threadIdx.x & (WAVE - 1);
+    slot[0] = 0xFFFF;
+    __syncthreads();
+    slot[0] = (unsigned)lane;          // all 64 lanes, same address
+    slot2[0] = (unsigned)(63 - lane);  // reversed
+    __syncthreads();
+    if (threadIdx.x == 0) { out[0] = slot[0]; out[1] = slot2[0]; }
+}
+
+// ---------------------------------------------------------------------------
+// wave-cooperative compress: one wave per 16 KiB chunk
+// s_chunk: staged input; s_table: u16[8192] zeroed; dst: global output (after
+// the caller-written 4-byte LE header). Returns compressed size via lane 0.
+// ---------------------------------------------------------------------------
+__device__ inline uint32_t lds_read32(const uint8_t* s, uint32_t p) {
+    uint32_t v;
+    memcpy(&v, s + p, 4);
+    return v;
+}
+
+__device__ inline int lz4_wave_compress(const uint8_t* __restrict__ s_chunk, int srcSize,
+                                        uint8_t* __restrict__ dst, uint16_t* __restrict__ s_table,
+                                        int lane) {
+    const int mflimitPlusOne = srcSize - LZ4M_MFLIMIT + 1;
+    const int matchlimit = srcSize - LZ4M_LASTLITERALS;
+    int ip = 0, anchor = 0;
+    uint32_t op = 0;
+
+    auto emit_last_literals = [&]() {
+        int lastRun = srcSize - anchor;
+        if (lastRun >= (int)LZ4M_RUN_MASK) {
+            int acc = lastRun - LZ4M_RUN_MASK;
+            if (lane == 0) dst[op] = (uint8_t)(LZ4M_RUN_MASK << LZ4M_ML_BITS);
+            op++;
+            int n255 = acc / 255;
+            for (int i = lane; i < n255; i += WAVE) dst[op + i] = 255;
+            op += n255;
+            if (lane == 0) dst[op] = (uint8_t)(acc % 255);
+            op++;
+        } else {
+            if (lane == 0) dst[op] = (uint8_t)(lastRun << LZ4M_ML_BITS);
+            op++;
+        }
+        for (int i = lane; i < lastRun; i += WAVE) dst[op + i] = s_chunk[anchor + i];
+        op += lastRun;
+    };
+
+    if (srcSize < LZ4M_MFLIMIT + 1) {
+        emit_last_literals();
+        return (int)op;
+    }
+
+    // first byte insert (all lanes uniform)
+    s_table[lz4m_hash(lds_read32(s_chunk, 0))] = 0;
+    ip = 1;
+
+    bool done = false;
+    while (!done) {
+        // ================= match finder: 64-probe windows =================
+        int match = -1;           // matched candidate position
+        {
+            int wstart = ip;      // first probe of window
+            int stride = 1;
+            bool found = false, aborted = false;
+            while (true) {
+                int p_l = wstart + lane * stride;
+                // abort condition for probe k: q_k + stride > mflimitPlusOne
+                bool valid = (p_l + stride) <= mflimitPlusOne;
+                uint32_t v_l = 0, h_l = 0;
+                if (p_l + 4 <= srcSize) v_l = lds_read32(s_chunk, (uint32_t)p_l);
+                h_l = lz4m_hash(v_l);
+                uint16_t t_l = s_table[h_l];              // pre-window candidate
+                // marker rounds: max-lane and min-lane of each hash group
+                s_table[h_l] = (uint16_t)lane;
+                int maxlane = (int)s_table[h_l];
+                s_table[h_l] = (uint16_t)(63 - lane);
+                int minlane = 63 - (int)s_table[h_l];
+                bool dup_earlier = minlane < lane;
+                // provisional match using the pre-window candidate
+                bool m_l = valid && !dup_earlier && lds_read32(s_chunk, t_l) == v_l;
+                uint64_t abort_mask = wave_ballot(!valid);
+                int first_abort = abort_mask ? (int)__ffsll((long long)abort_mask) - 1 : WAVE;
+                uint64_t match_mask = wave_ballot(m_l);
+                int first_event = match_mask ? (int)__ffsll((long long)match_mask) - 1 : WAVE;
+                // resolve ambiguous (dup_earlier) lanes below the provisional event
+                uint64_t amb = wave_ballot(dup_earlier && valid);
+                while (amb) {
+                    int d = (int)__ffsll((long long)amb) - 1;
+                    amb &= amb - 1;
+                    if (d >= first_event || d >= first_abort) continue;
+                    uint32_t h_d = (uint32_t)__shfl(h_l, d);
+                    uint32_t v_d = (uint32_t)__shfl((int)v_l, d);
+                    uint64_t eq = wave_ballot(h_l == h_d) & ((1ULL << d) - 1);
+                    bool md;
+                    if (eq) {
+                        int pred = 63 - (int)__clzll((long long)eq);
+                        uint32_t v_p = (uint32_t)__shfl((int)v_l, pred);
+                        md = v_p == v_d;
+                    } else {
+                        uint16_t t_d = (uint16_t)__shfl((int)t_l, d);
+                        md = lds_read32(s_chunk, t_d) == v_d;
+                    }
+                    if (md && d < first_event) first_event = d;
+                }
+                first_event = min(first_event, WAVE);
+                bool have_match = first_event < first_abort && first_event < WAVE;
+                int commit_hi = have_match ? first_event : (first_abort < WAVE ? first_abort - 1 : WAVE - 1);
+                // restore pre-window entries everywhere, then commit probed range
+                s_table[h_l] = t_l;
+                if (lane <= commit_hi && (have_match || lane < first_abort || first_abort == WAVE))
+                    s_table[h_l] = (uint16_t)p_l;
+                if (have_match) {
+                    ip = wstart + first_event * stride;
+                    // candidate of the matching lane
+                    uint64_t eq = wave_ballot(h_l == (uint32_t)__shfl(h_l, first_event)) &
+                                  ((1ULL << first_event) - 1);
+                    int m;
+                    if (eq) m = wstart + (63 - (int)__clzll((long long)eq)) * stride;
+                    else m = (int)(uint16_t)__shfl((int)t_l, first_event);
+                    match = m;
+                    found = true;
+                } else if (first_abort < WAVE) {
+                    aborted = true;
+                }
+                if (found || aborted) break;
+                wstart = wstart + WAVE * stride;  // next window
+                stride += 1;
+            }
+            if (aborted) { emit_last_literals(); return (int)op; }
+        }
+
+        // ================= catch up =================
+        while (true) {
+            int e = lane + 1;
+            bool ok = (ip - e + 1 > anchor) && (match - e + 1 > 0) &&
+                      s_chunk[ip - e] == s_chunk[match - e];
+            uint64_t bad = wave_ballot(!ok);
+            int run = bad ? (int)__ffsll((long long)bad) - 1 : WAVE;
+            ip -= run;
+            match -= run;
+            if (run < WAVE) break;
+        }
+
+        // ================= literals =================
+        {
+            int lit = ip - anchor;
+            uint32_t token_pos = op++;
+            uint8_t token;
+            if (lit >= (int)LZ4M_RUN_MASK) {
+                token = (uint8_t)(LZ4M_RUN_MASK << LZ4M_ML_BITS);
+                int len = lit - LZ4M_RUN_MASK;
+                int n255 = len / 255;
+                for (int i = lane; i < n255; i += WAVE) dst[op + i] = 255;
+                op += n255;
+                if (lane == 0) dst[op] = (uint8_t)(len % 255);
+                op++;
+            } else {
+                token = (uint8_t)(lit << LZ4M_ML_BITS);
+            }
+            for (int i = lane; i < lit; i += WAVE) dst[op + i] = s_chunk[anchor + i];
+            op += lit;
+
+            // ================= offset + match length =================
+            while (true) {
+                int offv = ip - match;
+                if (lane == 0) { dst[op] = (uint8_t)offv; dst[op + 1] = (uint8_t)(offv >> 8); }
+                op += 2;
+                // match length from ip+4 / match+4 up to matchlimit
+                int mc = 0;
+                {
+                    int offb = 0;
+                    while (true) {
+                        int pi = ip + 4 + offb + lane;
+                        bool eqb = pi < matchlimit && s_chunk[pi] == s_chunk[match + 4 + offb + lane];
+                        uint64_t ne = wave_ballot(!eqb);
+                        int run = ne ? (int)__ffsll((long long)ne) - 1 : WAVE;
+                        mc = offb + run;
+                        if (run < WAVE) break;
+                        offb += WAVE;
+                    }
+                }
+                ip += mc + LZ4M_MINMATCH;
+                if (mc >= (int)LZ4M_ML_MASK) {
+                    token += LZ4M_ML_MASK;
+                    int rem = mc - LZ4M_ML_MASK;
+                    int n255 = rem / 255;
+                    for (int i = lane; i < n255; i += WAVE) dst[op + i] = 255;
+                    op += n255;
+                    if (lane == 0) dst[op] = (uint8_t)(rem % 255);
+                    op++;
+                } else {
+                    token += (uint8_t)mc;
+                }
+                if (lane == 0) dst[token_pos] = token;
+
+                anchor = ip;
+                if (ip >= mflimitPlusOne) { done = true; break; }
+                // fill table at ip-2
+                s_table[lz4m_hash(lds_read32(s_chunk, (uint32_t)(ip - 2)))] = (uint16_t)(ip - 2);
+                // immediate test at ip
+                uint32_t v = lds_read32(s_chunk, (uint32_t)ip);
+                uint32_t h = lz4m_hash(v);
+                int m2 = (int)s_table[h];
+                s_table[h] = (uint16_t)ip;
+                if (lds_read32(s_chunk, (uint32_t)m2) == v) {
+                    match = m2;
+                    token_pos = op++;
+                    token = 0;
+                    continue;  // another match with empty literal run
+                }
+                ip += 1;
+                break;
+            }
+            if (done) break;
+        }
+    }
+    emit_last_literals();
+    return (int)op;
+}
+
+__global__ void __launch_bounds__(WAVE) k_lz4_compress_wave(const uint8_t* data, uint64_t data_len,
+                                                            uint8_t* slots, uint32_t* csize,
+                                                            uint32_t* ccrc, uint32_t n_chunks,
+                                                            const uint32_t* crc_table) {
+    __shared__ uint8_t s_chunk[CHUNK_LEN];
+    __shared__ uint16_t s_table[LZ4M_HASHTABLESIZE_U16];
+    uint32_t c = blockIdx.x;
+    if (c >= n_chunks) return;
+    int lane = threadIdx.x;
+    uint64_t off = (uint64_t)c * CHUNK_LEN;
+    uint32_t len = (uint32_t)min((uint64_t)CHUNK_LEN, data_len - off);
+    for (uint32_t i = lane * 4; i + 4 <= len; i += WAVE * 4)
+        *(uint32_t*)&s_chunk[i] = *(const uint32_t*)((const uint8_t*)data + off + i);
+    for (uint32_t i = (len & ~3u) + lane; i < len; i += WAVE) s_chunk[i] = data[off + i];
+    for (int i = lane; i < LZ4M_HASHTABLESIZE_U16; i += WAVE) s_table[i] = 0;
+    __syncthreads();
+    uint8_t* dst = slots + (uint64_t)c * LZ4_SLOT;
+    if (lane == 0) {
+        dst[0] = (uint8_t)len; dst[1] = (uint8_t)(len >> 8);
+        dst[2] = (uint8_t)(len >> 16); dst[3] = (uint8_t)(len >> 24);
+    }
+    int csz = lz4_wave_compress(s_chunk, (int)len, dst + 4, s_table, lane);
+    uint32_t total = (uint32_t)csz + 4;
+    if (lane == 0) csize[c] = total;
+    // wave-parallel CRC over the compressed bytes: per-lane slices + combine
+    // (round 1: lane0 table CRC — slices TBD)
+    if (lane == 0) {
+        uint32_t crc = 0xFFFFFFFFu;
+        for (uint32_t i = 0; i < total; i++) crc = crc_table[(crc ^ dst[i]) & 0xFF] ^ (crc >> 8);
+        ccrc[c] = ~crc;
+    }
+}
+
+// ---------------------------------------------------------------------------
+// wave-cooperative decompress: comp chunk staged in LDS; the token chain runs
+// redundantly on all lanes (uniform); copies are wave-parallel.
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(WAVE) k_lz4_decompress_wave(const ChunkDesc* chunks, uint32_t n,
+                                                              int verify_crc,
+                                                              unsigned long long* error,
+                                                              const uint32_t* crc_table) {
+    __shared__ uint8_t s_comp[LZ4_SLOT];
+    uint32_t c = blockIdx.x;
+    if (c >= n) return;
+    int lane = threadIdx.x;
+    ChunkDesc ch = chunks[c];
+    if (ch.comp_len > LZ4_SLOT) { if (lane == 0) atomicExch(error, 9ull); return; }
+    for (uint32_t i = lane * 4; i + 4 <= ch.comp_len; i += WAVE * 4) {
+        uint32_t v;
+        memcpy(&v, ch.comp + i, 4);
+        *(uint32_t*)&s_comp[i] = v;
+    }
+    for (uint32_t i = (ch.comp_len & ~3u) + lane; i < ch.comp_len; i += WAVE) s_comp[i] = ch.comp[i];
+    __syncthreads();
+    if (verify_crc && lane == 0) {
+        uint32_t crc = 0xFFFFFFFFu;
+        for (uint32_t i = 0; i < ch.comp_len; i++) crc = crc_table[(crc ^ s_comp[i]) & 0xFF] ^ (crc >> 8);
+        crc = ~crc;
+        uint32_t stored = ((uint32_t)ch.comp[ch.comp_len] << 24) | ((uint32_t)ch.comp[ch.comp_len + 1] << 16) |
+                          ((uint32_t)ch.comp[ch.comp_len + 2] << 8) | ch.comp[ch.comp_len + 3];
+        if (crc != stored) atomicExch(error, 1ull);
+    }
+    uint32_t hdr = s_comp[0] | (s_comp[1] << 8) | (s_comp[2] << 16) | ((uint32_t)s_comp[3] << 24);
+    if (hdr != ch.out_len) { if (lane == 0) atomicExch(error, 2ull); return; }
+    // uniform decode on all lanes
+    uint32_t ip = 4, iend = ch.comp_len;
+    uint32_t opos = 0, olen = ch.out_len;
+    uint8_t* out = ch.out;
+    while (ip < iend) {
+        uint32_t token = s_comp[ip++];
+        uint32_t lit = token >> 4;
+        if (lit == 15) {
+            uint32_t s;
+            do { s = s_comp[ip++]; lit += s; } while (s == 255);
+        }
+        if (opos + lit > olen || ip + lit > iend) { if (lane == 0) atomicExch(error, 3ull); return; }
+        for (uint32_t i = lane; i < lit; i += WAVE) out[opos + i] = s_comp[ip + i];
+        opos += lit; ip += lit;
+        if (ip >= iend) break;
+        uint32_t off = s_comp[ip] | (s_comp[ip + 1] << 8);
+        ip += 2;
+        uint32_t ml = (token & 15) + 4;
+        if (ml == 19) {
+            uint32_t s;
+            do { s = s_comp[ip++]; ml += s; } while (s == 255);
+        }
+        if (off == 0 || opos < off || opos + ml > olen) { if (lane == 0) atomicExch(error, 4ull); return; }
+        const uint8_t* src = out + opos - off;
+        if (off >= WAVE) {
+            for (uint32_t i = lane; i < ml; i += WAVE) out[opos + i] = src[i];
+        } else {
+            // overlapped copy == periodic repetition of the last `off` bytes
+            for (uint32_t i = lane; i < ml; i += WAVE) out[opos + i] = src[i % off];
+        }
+        opos += ml;
+    }
+    if (opos != olen && lane == 0) atomicExch(error, 5ull);
+}
+
+}  // namespace gpuc

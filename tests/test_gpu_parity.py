@@ -78,6 +78,8 @@ def test_generator_parity(ca, oracle_bin, tmp_path):
                                          vrep=0), job={}),
     dict(name="highly_compressible", gen=dict(seed=33, n=2, rows=2000, vlen=900, overlap=10,
                                               vrep=97), job={}),
+    dict(name="constant_values", gen=dict(seed=35, n=2, rows=1500, vlen=1000, overlap=0,
+                                          vrep=100), job={}),
 ])
 def test_compaction_parity(ca, oracle_bin, tmp_path, case):
     d = str(tmp_path)
