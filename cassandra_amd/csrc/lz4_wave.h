@@ -5,16 +5,17 @@
 // the system library) where the sequential hash-probe chain is executed as
 // 64-position speculative windows across the wavefront:
 //
-//   * the probe sequence q_{k+1} = q_k + ((64+k)>>6) advances with a CONSTANT
-//     stride inside each 64-probe window (stride w+1 in window w), so a wave
-//     evaluates one window per step: parallel read32+hash, candidate lookup,
-//     match test, then commits table updates only up to the first match/abort
+//   * the probe sequence advances by adv_0=1, adv_m=(63+m)>>6 (the scalar
+//     loop's skip acceleration); a wave evaluates 64 probes per step at the
+//     exact closed-form positions (lz4_adv_sum), tests candidates in
+//     parallel, then commits table updates only up to the first match/abort
 //     — byte-identical to the scalar loop.
-//   * in-window hash duplicates are resolved with two marker write/read-back
-//     rounds on the LDS hash table. This relies on the CDNA LDS conflict rule
-//     that simultaneous writes to one address retire in lane order (highest
-//     lane wins); k_probe_lds_order verifies it at init and the host refuses
-//     to run if it does not hold.
+//   * in-window hash duplicates are resolved exactly: one marker write/read
+//     round on the LDS hash table names each hash group's max lane (CDNA LDS
+//     conflict rule: simultaneous writes to one address retire in lane order,
+//     highest lane wins — k_probe_lds_order verifies this at init and the
+//     host refuses to run if it does not hold), then a per-group ballot walk
+//     hands every lane its exact in-window predecessor.
 //   * match extension, catch-up, literal copies and 255-run emission are
 //     wave-parallel (ballots + strided copies).
 //
@@ -58,9 +59,32 @@ __device__ inline uint32_t lds_read32(const uint8_t* s, uint32_t p) {
     return v;
 }
 
+// advance of the scalar probe loop after probe m (lz4.c: step initialised to 1,
+// then step = searchMatchNb++ >> 6 with searchMatchNb starting at 64):
+// adv_0 = 1, adv_m = (63+m)>>6 for m >= 1. lz4_adv_sum(a,b) = sum adv_m, m in [a,b).
+__device__ inline int lz4_adv_g(int T) {  // sum_{t=0}^{T} t>>6
+    if (T < 0) return 0;
+    int S = T >> 6;
+    return 64 * (S * (S - 1) / 2) + S * (T - 64 * S + 1);
+}
+__device__ inline int lz4_adv_f(int x) {  // sum_{m=1}^{x} (63+m)>>6
+    return lz4_adv_g(63 + x) - lz4_adv_g(63);
+}
+__device__ inline int lz4_adv_sum(int a, int b) {
+    if (b <= a) return 0;
+    int d = 0;
+    if (a == 0) {
+        d += 1;
+        a = 1;
+        if (b <= a) return d;
+    }
+    return d + lz4_adv_f(b - 1) - lz4_adv_f(a - 1);
+}
+__device__ inline int lz4_adv(int m) { return m == 0 ? 1 : (63 + m) >> 6; }
+
 __device__ inline int lz4_wave_compress(const uint8_t* __restrict__ s_chunk, int srcSize,
                                         uint8_t* __restrict__ dst, uint16_t* __restrict__ s_table,
-                                        int lane) {
+                                        int lane, uint32_t* dbg = nullptr) {
     const int mflimitPlusOne = srcSize - LZ4M_MFLIMIT + 1;
     const int matchlimit = srcSize - LZ4M_LASTLITERALS;
     int ip = 0, anchor = 0;
@@ -99,72 +123,78 @@ __device__ inline int lz4_wave_compress(const uint8_t* __restrict__ s_chunk, int
         // ================= match finder: 64-probe windows =================
         int match = -1;           // matched candidate position
         {
-            int wstart = ip;      // first probe of window
-            int stride = 1;
+            int k0 = 0;           // probe index of this window's lane 0
+            int P0 = ip;          // its position
             bool found = false, aborted = false;
             while (true) {
-                int p_l = wstart + lane * stride;
-                // abort condition for probe k: q_k + stride > mflimitPlusOne
-                bool valid = (p_l + stride) <= mflimitPlusOne;
-                uint32_t v_l = 0, h_l = 0;
-                if (p_l + 4 <= srcSize) v_l = lds_read32(s_chunk, (uint32_t)p_l);
-                h_l = lz4m_hash(v_l);
+                int p_l = P0 + lz4_adv_sum(k0, k0 + lane);
+                // scalar loop aborts probe m when q_m + adv_m > mflimitPlusOne
+                bool valid = (p_l + lz4_adv(k0 + lane)) <= mflimitPlusOne;
+                uint32_t v_l = 0;
+                if (p_l + 4 <= srcSize && p_l >= 0) v_l = lds_read32(s_chunk, (uint32_t)p_l);
+                uint32_t h_l = lz4m_hash(v_l);
                 uint16_t t_l = s_table[h_l];              // pre-window candidate
-                // marker rounds: max-lane and min-lane of each hash group
-                s_table[h_l] = (uint16_t)lane;
-                int maxlane = (int)s_table[h_l];
-                s_table[h_l] = (uint16_t)(63 - lane);
-                int minlane = 63 - (int)s_table[h_l];
-                bool dup_earlier = minlane < lane;
-                // provisional match using the pre-window candidate
-                bool m_l = valid && !dup_earlier && lds_read32(s_chunk, t_l) == v_l;
+                // marker round: conflicting LDS writes retire highest-lane-last,
+                // so the read-back names each hash group's max lane. volatile
+                // stops the compiler from forwarding the lane's own store —
+                // the value read must come from the LDS conflict resolution.
+                {
+                    volatile uint16_t* vt = s_table;
+                    vt[h_l] = (uint16_t)lane;
+                }
+                int maxgroup;
+                {
+                    volatile const uint16_t* vt = s_table;
+                    maxgroup = (int)vt[h_l];
+                }
+                // exact in-window predecessor per lane via group masks
+                int pred = -1;
+                {
+                    uint64_t G = wave_ballot(maxgroup != lane);  // non-max members
+                    while (G) {
+                        int g = (int)__ffsll((long long)G) - 1;
+                        int mg = __shfl(maxgroup, g);
+                        uint64_t members = wave_ballot(maxgroup == mg);
+                        if (maxgroup == mg) {
+                            uint64_t below = members & ((1ULL << lane) - 1);
+                            if (below) pred = 63 - (int)__clzll((long long)below);
+                        }
+                        G &= ~members;
+                    }
+                }
+                // exact match test: in-window predecessor position if any,
+                // else the pre-window table entry. The predecessor's position
+                // is closed-form per lane (no divergent cross-lane read).
+                uint32_t cand_pos = pred >= 0 ? (uint32_t)(P0 + lz4_adv_sum(k0, k0 + pred))
+                                              : (uint32_t)t_l;
+                bool m_l = valid && lds_read32(s_chunk, cand_pos) == v_l;
                 uint64_t abort_mask = wave_ballot(!valid);
                 int first_abort = abort_mask ? (int)__ffsll((long long)abort_mask) - 1 : WAVE;
                 uint64_t match_mask = wave_ballot(m_l);
                 int first_event = match_mask ? (int)__ffsll((long long)match_mask) - 1 : WAVE;
-                // resolve ambiguous (dup_earlier) lanes below the provisional event
-                uint64_t amb = wave_ballot(dup_earlier && valid);
-                while (amb) {
-                    int d = (int)__ffsll((long long)amb) - 1;
-                    amb &= amb - 1;
-                    if (d >= first_event || d >= first_abort) continue;
-                    uint32_t h_d = (uint32_t)__shfl(h_l, d);
-                    uint32_t v_d = (uint32_t)__shfl((int)v_l, d);
-                    uint64_t eq = wave_ballot(h_l == h_d) & ((1ULL << d) - 1);
-                    bool md;
-                    if (eq) {
-                        int pred = 63 - (int)__clzll((long long)eq);
-                        uint32_t v_p = (uint32_t)__shfl((int)v_l, pred);
-                        md = v_p == v_d;
-                    } else {
-                        uint16_t t_d = (uint16_t)__shfl((int)t_l, d);
-                        md = lds_read32(s_chunk, t_d) == v_d;
-                    }
-                    if (md && d < first_event) first_event = d;
-                }
-                first_event = min(first_event, WAVE);
                 bool have_match = first_event < first_abort && first_event < WAVE;
                 int commit_hi = have_match ? first_event : (first_abort < WAVE ? first_abort - 1 : WAVE - 1);
-                // restore pre-window entries everywhere, then commit probed range
-                s_table[h_l] = t_l;
-                if (lane <= commit_hi && (have_match || lane < first_abort || first_abort == WAVE))
-                    s_table[h_l] = (uint16_t)p_l;
+                // restore pre-window entries, then commit the probed range.
+                // These MUST stay two separate wave-wide stores (restore pass
+                // by all lanes, then commit pass by lanes <= commit_hi):
+                // merging them into one select-store changes the cross-lane
+                // winner when a higher non-committing lane shares a slot —
+                // volatile forbids the compiler from combining them.
+                {
+                    volatile uint16_t* vt = s_table;
+                    vt[h_l] = t_l;
+                    if (lane <= commit_hi) vt[h_l] = (uint16_t)p_l;
+                }
                 if (have_match) {
-                    ip = wstart + first_event * stride;
-                    // candidate of the matching lane
-                    uint64_t eq = wave_ballot(h_l == (uint32_t)__shfl(h_l, first_event)) &
-                                  ((1ULL << first_event) - 1);
-                    int m;
-                    if (eq) m = wstart + (63 - (int)__clzll((long long)eq)) * stride;
-                    else m = (int)(uint16_t)__shfl((int)t_l, first_event);
-                    match = m;
+                    ip = P0 + lz4_adv_sum(k0, k0 + first_event);
+                    match = (int)(uint32_t)__shfl((int)cand_pos, first_event);
                     found = true;
                 } else if (first_abort < WAVE) {
                     aborted = true;
                 }
                 if (found || aborted) break;
-                wstart = wstart + WAVE * stride;  // next window
-                stride += 1;
+                P0 += lz4_adv_sum(k0, k0 + WAVE);
+                k0 += WAVE;
             }
             if (aborted) { emit_last_literals(); return (int)op; }
         }
@@ -183,7 +213,7 @@ __device__ inline int lz4_wave_compress(const uint8_t* __restrict__ s_chunk, int
 
         // ================= literals =================
         {
-            int lit = ip - anchor;
+            const int lit = ip - anchor;
             uint32_t token_pos = op++;
             uint8_t token;
             if (lit >= (int)LZ4M_RUN_MASK) {
@@ -203,6 +233,10 @@ __device__ inline int lz4_wave_compress(const uint8_t* __restrict__ s_chunk, int
             // ================= offset + match length =================
             while (true) {
                 int offv = ip - match;
+                if (dbg && lane == 0) {
+                    uint32_t n = ++dbg[0];
+                    if (n < 2000) { dbg[n * 3] = (uint32_t)ip; dbg[n * 3 + 1] = (uint32_t)match; dbg[n * 3 + 2] = (uint32_t)lit; }
+                }
                 if (lane == 0) { dst[op] = (uint8_t)offv; dst[op + 1] = (uint8_t)(offv >> 8); }
                 op += 2;
                 // match length from ip+4 / match+4 up to matchlimit
@@ -261,7 +295,8 @@ __device__ inline int lz4_wave_compress(const uint8_t* __restrict__ s_chunk, int
 __global__ void __launch_bounds__(WAVE) k_lz4_compress_wave(const uint8_t* data, uint64_t data_len,
                                                             uint8_t* slots, uint32_t* csize,
                                                             uint32_t* ccrc, uint32_t n_chunks,
-                                                            const uint32_t* crc_table) {
+                                                            const uint32_t* crc_table,
+                                                            uint32_t* dbg = nullptr) {
     __shared__ uint8_t s_chunk[CHUNK_LEN];
     __shared__ uint16_t s_table[LZ4M_HASHTABLESIZE_U16];
     uint32_t c = blockIdx.x;
@@ -279,7 +314,7 @@ __global__ void __launch_bounds__(WAVE) k_lz4_compress_wave(const uint8_t* data,
         dst[0] = (uint8_t)len; dst[1] = (uint8_t)(len >> 8);
         dst[2] = (uint8_t)(len >> 16); dst[3] = (uint8_t)(len >> 24);
     }
-    int csz = lz4_wave_compress(s_chunk, (int)len, dst + 4, s_table, lane);
+    int csz = lz4_wave_compress(s_chunk, (int)len, dst + 4, s_table, lane, dbg);
     uint32_t total = (uint32_t)csz + 4;
     if (lane == 0) csize[c] = total;
     // wave-parallel CRC over the compressed bytes: per-lane slices + combine
@@ -346,10 +381,12 @@ __global__ void __launch_bounds__(WAVE) k_lz4_decompress_wave(const ChunkDesc* c
         }
         if (off == 0 || opos < off || opos + ml > olen) { if (lane == 0) atomicExch(error, 4ull); return; }
         const uint8_t* src = out + opos - off;
-        if (off >= WAVE) {
+        if (ml <= off) {
             for (uint32_t i = lane; i < ml; i += WAVE) out[opos + i] = src[i];
         } else {
-            // overlapped copy == periodic repetition of the last `off` bytes
+            // overlapped copy (ml > off) == periodic repetition of the last
+            // `off` bytes; the modulo form avoids reading bytes this same
+            // copy has not written yet
             for (uint32_t i = lane; i < ml; i += WAVE) out[opos + i] = src[i % off];
         }
         opos += ml;
