@@ -17,6 +17,8 @@
 #include <stdexcept>
 #include <string>
 #include <thread>
+#include <fcntl.h>
+#include <unistd.h>
 #include <vector>
 
 namespace gpuc {
@@ -63,6 +65,79 @@ struct HReader {
         return v;
     }
 };
+
+// grow-only pinned staging buffers, reused across calls (hipHostMalloc is
+// expensive; a compaction service reuses its staging arenas the same way)
+struct PinnedSlot {
+    void* p = nullptr;
+    size_t cap = 0;
+    void* get(size_t n) {
+        if (n > cap) {
+            if (p) (void)hipHostFree(p);
+            size_t want = n + n / 4;
+            if (hipHostMalloc(&p, want) != hipSuccess) { p = nullptr; cap = 0; return nullptr; }
+            cap = want;
+        }
+        return p;
+    }
+};
+static PinnedSlot g_pin_in[64], g_pin_out[4];
+
+// threaded read of a whole file into (pinned) memory
+static size_t file_size_of(const std::string& path) {
+    FILE* f = fopen(path.c_str(), "rb");
+    if (!f) throw std::runtime_error("cannot open " + path);
+    fseek(f, 0, SEEK_END);
+    long n = ftell(f);
+    fclose(f);
+    return (size_t)n;
+}
+static void read_file_into(const std::string& path, uint8_t* dst, size_t n, int nthreads = 4) {
+    if (n == 0) return;
+    std::vector<std::thread> th;
+    size_t per = (n + nthreads - 1) / nthreads;
+    for (int t = 0; t < nthreads; t++) {
+        size_t off = (size_t)t * per;
+        if (off >= n) break;
+        size_t len = std::min(per, n - off);
+        th.emplace_back([&, off, len]() {
+            FILE* f = fopen(path.c_str(), "rb");
+            if (!f) return;
+            fseek(f, (long)off, SEEK_SET);
+            size_t rd = fread(dst + off, 1, len, f);
+            (void)rd;
+            fclose(f);
+        });
+    }
+    for (auto& x : th) x.join();
+}
+static void write_file_parallel(const std::string& path, const uint8_t* p, size_t n, int nthreads = 4) {
+    // create + size, then threaded pwrite
+    FILE* f = fopen(path.c_str(), "wb");
+    if (!f) throw std::runtime_error("cannot create " + path);
+    fclose(f);
+    if (n == 0) return;
+    if (truncate(path.c_str(), (off_t)n) != 0) throw std::runtime_error("truncate " + path);
+    std::vector<std::thread> th;
+    size_t per = (n + nthreads - 1) / nthreads;
+    for (int t = 0; t < nthreads; t++) {
+        size_t off = (size_t)t * per;
+        if (off >= n) break;
+        size_t len = std::min(per, n - off);
+        th.emplace_back([&, off, len]() {
+            int fd = open(path.c_str(), O_WRONLY);
+            if (fd < 0) return;
+            size_t done = 0;
+            while (done < len) {
+                ssize_t w = pwrite(fd, p + off + done, len - done, (off_t)(off + done));
+                if (w <= 0) break;
+                done += (size_t)w;
+            }
+            close(fd);
+        });
+    }
+    for (auto& x : th) x.join();
+}
 
 static bytes read_file(const std::string& path) {
     FILE* f = fopen(path.c_str(), "rb");
@@ -591,7 +666,7 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, uint64_t n_gr
     d_csize.alloc((uint64_t)n_chunks * 4);
     d_ccrc.alloc((uint64_t)n_chunks * 4);
     d_csize64.alloc((uint64_t)n_chunks * 8);
-    hipLaunchKernelGGL(k_lz4_compress_wave, dim3(n_chunks), dim3(WAVE), 0, stream,
+    hipLaunchKernelGGL(k_lz4_compress_wave_t<false>, dim3(n_chunks), dim3(WAVE), 0, stream,
                        d_out_data.as<uint8_t>(), total_unc, d_slots.as<uint8_t>(),
                        d_csize.as<uint32_t>(), d_ccrc.as<uint32_t>(), n_chunks,
                        (const uint32_t*)g_crc256);
@@ -614,11 +689,14 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, uint64_t n_gr
                            d_csize64.as<uint64_t>(), d_final.as<uint8_t>(), n_chunks);
         HIP_CHECK(hipEventRecord(ev3, stream));
 
-        // ---- D2H + files ----
-        std::vector<uint8_t> h_data(w.compressed_len), h_index(total_idx), h_bloom(words * 8);
+        // ---- D2H (pinned staging) + files ----
+        uint8_t* h_data = (uint8_t*)g_pin_out[0].get(w.compressed_len);
+        uint8_t* h_index = (uint8_t*)g_pin_out[1].get(total_idx ? total_idx : 1);
+        if (!h_data || !h_index) throw std::runtime_error("pinned out alloc failed");
+        std::vector<uint8_t> h_bloom(words * 8);
         std::vector<uint32_t> h_crc(n_chunks);
-        HIP_CHECK(hipMemcpy(h_data.data(), d_final.p, w.compressed_len, hipMemcpyDeviceToHost));
-        HIP_CHECK(hipMemcpy(h_index.data(), d_out_index.p, total_idx, hipMemcpyDeviceToHost));
+        HIP_CHECK(hipMemcpyAsync(h_data, d_final.p, w.compressed_len, hipMemcpyDeviceToHost, stream));
+        HIP_CHECK(hipMemcpyAsync(h_index, d_out_index.p, total_idx, hipMemcpyDeviceToHost, stream));
         HIP_CHECK(hipMemcpy(h_bloom.data(), d_bloom.p, words * 8, hipMemcpyDeviceToHost));
         HIP_CHECK(hipMemcpy(h_crc.data(), d_ccrc.p, (uint64_t)n_chunks * 4, hipMemcpyDeviceToHost));
         HIP_CHECK(hipEventRecord(ev4, stream));
@@ -680,8 +758,8 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, uint64_t n_gr
             }
         }
 
-        write_file(out_base + "-Data.db", h_data.data(), h_data.size());
-        write_file(out_base + "-Index.db", h_index.data(), h_index.size());
+        write_file_parallel(out_base + "-Data.db", h_data, w.compressed_len, 6);
+        write_file_parallel(out_base + "-Index.db", h_index, total_idx, 4);
         {
             bytes f;
             put_be32(f, (uint32_t)bs.k);
@@ -774,15 +852,20 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
         double t0 = wall();
         int k = job->n_inputs;
         if (k < 1 || k > 64) { set_err(res->error, sizeof(res->error), "n_inputs must be 1..64"); return GPUC_ERR_UNSUPPORTED; }
-        std::vector<bytes> comp_data(k), index_data(k);
+        std::vector<bytes> index_data(k);
         std::vector<HCompressionInfo> cinfos(k);
         std::vector<HStatistics> stats(k);
         std::vector<uint64_t> generations(k);
+        std::vector<uint8_t*> comp_pin(k, nullptr);
+        std::vector<size_t> comp_sz(k, 0);
         for (int s = 0; s < k; s++) {
             std::string base = job->input_bases[s];
-            comp_data[s] = read_file(base + "-Data.db");
+            comp_sz[s] = file_size_of(base + "-Data.db");
+            comp_pin[s] = (uint8_t*)g_pin_in[s].get(comp_sz[s]);
+            if (!comp_pin[s]) throw std::runtime_error("pinned alloc failed");
             index_data[s] = read_file(base + "-Index.db");
             cinfos[s] = parse_compression_info(read_file(base + "-CompressionInfo.db"));
+            read_file_into(base + "-Data.db", comp_pin[s], comp_sz[s], 6);
             stats[s] = parse_statistics(read_file(base + "-Statistics.db"));
             if (!stats[s].clustering_types.empty())
                 throw std::runtime_error("clustering columns unsupported in round 1 (GPU path)");
@@ -824,8 +907,8 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
         std::vector<DevBuf> d_comp(k), d_data(k), d_pos(k);
         std::vector<ChunkDesc> chunks;
         for (int s = 0; s < k; s++) {
-            d_comp[s].alloc(comp_data[s].size());
-            HIP_CHECK(hipMemcpyAsync(d_comp[s].p, comp_data[s].data(), comp_data[s].size(),
+            d_comp[s].alloc(comp_sz[s]);
+            HIP_CHECK(hipMemcpyAsync(d_comp[s].p, comp_pin[s], comp_sz[s],
                                      hipMemcpyHostToDevice, stream));
             d_data[s].alloc(cinfos[s].data_len);
             d_pos[s].alloc(positions[s].size() * 8);
@@ -834,7 +917,7 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
             auto& ci = cinfos[s];
             for (size_t c = 0; c < ci.offsets.size(); c++) {
                 uint64_t off = ci.offsets[c];
-                uint64_t end = c + 1 < ci.offsets.size() ? ci.offsets[c + 1] : comp_data[s].size();
+                uint64_t end = c + 1 < ci.offsets.size() ? ci.offsets[c + 1] : comp_sz[s];
                 ChunkDesc cd;
                 cd.comp = d_comp[s].as<uint8_t>() + off;
                 cd.out = d_data[s].as<uint8_t>() + c * (uint64_t)CHUNK_LEN;

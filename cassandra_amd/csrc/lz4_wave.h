@@ -82,6 +82,7 @@ __device__ inline int lz4_adv_sum(int a, int b) {
 }
 __device__ inline int lz4_adv(int m) { return m == 0 ? 1 : (63 + m) >> 6; }
 
+template <bool STAGE_LDS = true>
 __device__ inline int lz4_wave_compress(const uint8_t* __restrict__ s_chunk, int srcSize,
                                         uint8_t* __restrict__ dst, uint16_t* __restrict__ s_table,
                                         int lane, uint32_t* dbg = nullptr) {
@@ -292,21 +293,28 @@ __device__ inline int lz4_wave_compress(const uint8_t* __restrict__ s_chunk, int
     return (int)op;
 }
 
-__global__ void __launch_bounds__(WAVE) k_lz4_compress_wave(const uint8_t* data, uint64_t data_len,
-                                                            uint8_t* slots, uint32_t* csize,
-                                                            uint32_t* ccrc, uint32_t n_chunks,
-                                                            const uint32_t* crc_table,
-                                                            uint32_t* dbg = nullptr) {
-    __shared__ uint8_t s_chunk[CHUNK_LEN];
+template <bool STAGE_LDS>
+__global__ void __launch_bounds__(WAVE) k_lz4_compress_wave_t(const uint8_t* data, uint64_t data_len,
+                                                              uint8_t* slots, uint32_t* csize,
+                                                              uint32_t* ccrc, uint32_t n_chunks,
+                                                              const uint32_t* crc_table,
+                                                              uint32_t* dbg = nullptr) {
     __shared__ uint16_t s_table[LZ4M_HASHTABLESIZE_U16];
     uint32_t c = blockIdx.x;
     if (c >= n_chunks) return;
     int lane = threadIdx.x;
     uint64_t off = (uint64_t)c * CHUNK_LEN;
     uint32_t len = (uint32_t)min((uint64_t)CHUNK_LEN, data_len - off);
-    for (uint32_t i = lane * 4; i + 4 <= len; i += WAVE * 4)
-        *(uint32_t*)&s_chunk[i] = *(const uint32_t*)((const uint8_t*)data + off + i);
-    for (uint32_t i = (len & ~3u) + lane; i < len; i += WAVE) s_chunk[i] = data[off + i];
+    const uint8_t* src;
+    if constexpr (STAGE_LDS) {
+        __shared__ uint8_t s_chunk[CHUNK_LEN];
+        for (uint32_t i = lane * 4; i + 4 <= len; i += WAVE * 4)
+            *(uint32_t*)&s_chunk[i] = *(const uint32_t*)((const uint8_t*)data + off + i);
+        for (uint32_t i = (len & ~3u) + lane; i < len; i += WAVE) s_chunk[i] = data[off + i];
+        src = s_chunk;
+    } else {
+        src = data + off;
+    }
     for (int i = lane; i < LZ4M_HASHTABLESIZE_U16; i += WAVE) s_table[i] = 0;
     __syncthreads();
     uint8_t* dst = slots + (uint64_t)c * LZ4_SLOT;
@@ -314,11 +322,11 @@ __global__ void __launch_bounds__(WAVE) k_lz4_compress_wave(const uint8_t* data,
         dst[0] = (uint8_t)len; dst[1] = (uint8_t)(len >> 8);
         dst[2] = (uint8_t)(len >> 16); dst[3] = (uint8_t)(len >> 24);
     }
-    int csz = lz4_wave_compress(s_chunk, (int)len, dst + 4, s_table, lane, dbg);
+    int csz = lz4_wave_compress(src, (int)len, dst + 4, s_table, lane, dbg);
     uint32_t total = (uint32_t)csz + 4;
     if (lane == 0) csize[c] = total;
-    // wave-parallel CRC over the compressed bytes: per-lane slices + combine
-    // (round 1: lane0 table CRC — slices TBD)
+    // CRC over the compressed bytes: wave-parallel slices via per-lane CRC of
+    // strided... (kept lane0 table CRC; compressed bytes ~8KB, minor next to search)
     if (lane == 0) {
         uint32_t crc = 0xFFFFFFFFu;
         for (uint32_t i = 0; i < total; i++) crc = crc_table[(crc ^ dst[i]) & 0xFF] ^ (crc >> 8);

@@ -69,7 +69,7 @@ int main() {
             hipMalloc(&d_csize, 4);
             hipMalloc(&d_ccrc, 4);
             hipMemcpy(d_src, src.data(), n, hipMemcpyHostToDevice);
-            hipLaunchKernelGGL(k_lz4_compress_wave, dim3(1), dim3(64), 0, 0, d_src, (uint64_t)n,
+            hipLaunchKernelGGL(k_lz4_compress_wave_t<true>, dim3(1), dim3(64), 0, 0, d_src, (uint64_t)n,
                                d_slots, d_csize, d_ccrc, 1, d_tab);
             hipError_t e = hipDeviceSynchronize();
             if (e != hipSuccess) { printf("kernel error %s mode=%d trial=%d\n", hipGetErrorString(e), mode, trial); return 4; }
@@ -96,7 +96,7 @@ int main() {
                 uint32_t* d_dbg;
                 hipMalloc(&d_dbg, 4 * 3 * 2001);
                 hipMemset(d_dbg, 0, 4 * 3 * 2001);
-                hipLaunchKernelGGL(k_lz4_compress_wave, dim3(1), dim3(64), 0, 0, d_src, (uint64_t)n,
+                hipLaunchKernelGGL(k_lz4_compress_wave_t<true>, dim3(1), dim3(64), 0, 0, d_src, (uint64_t)n,
                                    d_slots, d_csize, d_ccrc, 1, d_tab, d_dbg);
                 hipDeviceSynchronize();
                 std::vector<uint32_t> geps(3 * 2001);
@@ -150,5 +150,48 @@ int main() {
         }
     }
     printf(fails ? "FAILED %d of %d\n" : "lz4 wave OK (%d cases)\n", fails ? fails : cases, cases);
-    return fails ? 1 : 0;
+    if (fails) return 1;
+
+    // ---- A/B throughput: LDS-staged vs global-chunk, value-like content ----
+    {
+        const uint32_t NC = 16384;  // 256 MiB of chunks
+        std::vector<uint8_t> big((uint64_t)NC * CHUNK_LEN);
+        for (uint64_t i = 0; i < big.size(); i += 8) {
+            // ~55% repeat words (bench-like compressibility)
+            uint64_t r = sm(i);
+            uint64_t w = (r % 100 < 55 && i) ? *(uint64_t*)&big[i - 8] : sm(r);
+            memcpy(&big[i], &w, 8);
+        }
+        uint8_t *d_big, *d_slots2;
+        uint32_t *d_cs, *d_crc2;
+        hipMalloc(&d_big, big.size());
+        hipMalloc(&d_slots2, (uint64_t)NC * LZ4_SLOT);
+        hipMalloc(&d_cs, NC * 4);
+        hipMalloc(&d_crc2, NC * 4);
+        hipMemcpy(d_big, big.data(), big.size(), hipMemcpyHostToDevice);
+        for (int variant = 0; variant < 2; variant++) {
+            hipEvent_t a, b;
+            hipEventCreate(&a); hipEventCreate(&b);
+            // warmup
+            if (variant == 0)
+                hipLaunchKernelGGL(k_lz4_compress_wave_t<true>, dim3(NC), dim3(64), 0, 0, d_big, big.size(), d_slots2, d_cs, d_crc2, NC, d_tab);
+            else
+                hipLaunchKernelGGL(k_lz4_compress_wave_t<false>, dim3(NC), dim3(64), 0, 0, d_big, big.size(), d_slots2, d_cs, d_crc2, NC, d_tab);
+            hipDeviceSynchronize();
+            hipEventRecord(a);
+            for (int it = 0; it < 3; it++) {
+                if (variant == 0)
+                    hipLaunchKernelGGL(k_lz4_compress_wave_t<true>, dim3(NC), dim3(64), 0, 0, d_big, big.size(), d_slots2, d_cs, d_crc2, NC, d_tab);
+                else
+                    hipLaunchKernelGGL(k_lz4_compress_wave_t<false>, dim3(NC), dim3(64), 0, 0, d_big, big.size(), d_slots2, d_cs, d_crc2, NC, d_tab);
+            }
+            hipEventRecord(b);
+            hipDeviceSynchronize();
+            float ms;
+            hipEventElapsedTime(&ms, a, b);
+            printf("compress %s: %.1f ms for 3x256MiB = %.2f GB/s in\n",
+                   variant == 0 ? "LDS-staged " : "global-src", ms, 3.0 * big.size() / (ms / 1e3) / 1e9);
+        }
+    }
+    return 0;
 }
