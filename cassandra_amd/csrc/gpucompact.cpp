@@ -14,6 +14,7 @@
 #include <cstring>
 #include <map>
 #include <memory>
+#include <mutex>
 #include <stdexcept>
 #include <string>
 #include <thread>
@@ -427,15 +428,65 @@ static HBloomSpec bloom_spec_001() { return {5, 10}; }  // computeBloomSpec(20, 
 // ---------------------------------------------------------------------------
 // device helpers (scan, merge)
 // ---------------------------------------------------------------------------
+// caching device allocator: hipMalloc of multi-GB buffers costs 100+ ms, and
+// every compaction call uses the same transient working set, so freed blocks
+// are pooled by power-of-two size class and reused across calls.
+class DevPool {
+    std::mutex mu_;
+    std::map<size_t, std::vector<void*>> free_;
+
+public:
+    static size_t cls(size_t n) {
+        size_t c = 4096;
+        while (c < n) c <<= 1;
+        return c;
+    }
+    void* get(size_t n) {
+        size_t c = cls(n);
+        {
+            std::lock_guard<std::mutex> g(mu_);
+            auto it = free_.find(c);
+            if (it != free_.end() && !it->second.empty()) {
+                void* p = it->second.back();
+                it->second.pop_back();
+                return p;
+            }
+        }
+        void* p = nullptr;
+        if (hipMalloc(&p, c) != hipSuccess) {
+            trim();
+            HIP_CHECK(hipMalloc(&p, c));
+        }
+        return p;
+    }
+    void put(size_t n, void* p) {
+        std::lock_guard<std::mutex> g(mu_);
+        free_[cls(n)].push_back(p);
+    }
+    void trim() {
+        std::lock_guard<std::mutex> g(mu_);
+        for (auto& [c, v] : free_) {
+            for (void* p : v) (void)hipFree(p);
+            v.clear();
+        }
+    }
+};
+static DevPool g_dev_pool;
+
 struct DevBuf {
     void* p = nullptr;
     size_t n = 0;
     void alloc(size_t bytes_) {
         free_();
-        HIP_CHECK(hipMalloc(&p, bytes_ ? bytes_ : 8));
-        n = bytes_;
+        n = bytes_ ? bytes_ : 8;
+        p = g_dev_pool.get(n);
     }
-    void free_() { if (p) { (void)hipFree(p); p = nullptr; } }
+    void free_() {
+        if (p) {
+            g_dev_pool.put(n, p);
+            p = nullptr;
+        }
+    }
     ~DevBuf() { free_(); }
     template <typename T> T* as() const { return (T*)p; }
 };
