@@ -8,6 +8,7 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
     SSTable t;
     t.generation = g.first_generation + sst;
     t.header.key_type = CqlType::LONG;  // pk bigint
+    if (g.clustering_rows > 0) t.header.clustering_types = {CqlType::LONG};  // ck bigint
     t.header.regular_cols = {{bytes{'v', 'a', 'l'}, CqlType::BYTES}};  // val blob
 
     struct Ent { int64_t token; bytes key; uint64_t id; };
@@ -26,7 +27,81 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
     });
 
     int64_t min_ts = INT64_MAX, min_ldt_l = INT64_MAX;
+    auto be8 = [](int64_t v) {
+        bytes b(8);
+        for (int i = 0; i < 8; i++) b[i] = (uint8_t)((uint64_t)v >> (8 * (7 - i)));
+        return b;
+    };
     for (const Ent& e : ents) {
+        if (g.clustering_rows > 0) {
+            // ---- wide partition: clustering rows + optional range tombstone ----
+            Partition p;
+            p.key = e.key;
+            p.token = e.token;
+            bool has_rt = gen_has_range_tomb(g, sst, e.id);
+            int64_t rlo = 0, rhi = 0, rts = 0;
+            uint32_t rldt = 0;
+            if (has_rt) {
+                gen_range_bounds(g, sst, e.id, &rlo, &rhi);
+                rts = gen_range_ts(g, sst, e.id);
+                rldt = gen_ldt(g, e.id, 0xCC);
+                min_ts = std::min(min_ts, rts);
+                min_ldt_l = std::min<int64_t>(min_ldt_l, rldt);
+            }
+            bool rt_open = false;
+            for (uint32_t j = 0; j < g.clustering_rows; j++) {
+                int64_t ck = gen_ck(g, e.id, j);
+                if (has_rt && !rt_open && ck > rlo) {
+                    Unfiltered u;
+                    u.kind = Unfiltered::MARKER;
+                    u.marker.kind = INCL_START;
+                    u.marker.values = {ClusterVal{ClusterVal::VALUE, be8(rlo)}};
+                    u.marker.end_dt = DeletionTime{rts, rldt};
+                    // marker sits at bound position rlo (before this row)
+                    p.items.push_back(std::move(u));
+                    rt_open = true;
+                }
+                Unfiltered u;
+                u.kind = Unfiltered::ROW;
+                Row& r = u.row;
+                r.clustering = {ClusterVal{ClusterVal::VALUE, be8(ck)}};
+                r.cells.resize(1);
+                int64_t ts = gen_row_ts(g, sst, e.id, j);
+                min_ts = std::min(min_ts, ts);
+                if (gen_row_is_tombstone(g, sst, e.id, j)) {
+                    r.del.mfda = ts;
+                    r.del.ldt = gen_ldt(g, e.id * 1000 + j, 0xEE);
+                    min_ldt_l = std::min<int64_t>(min_ldt_l, r.del.ldt);
+                } else {
+                    r.live.ts = ts;
+                    Cell cell;
+                    cell.ts = ts;
+                    gen_value(g, sst, e.id ^ ((uint64_t)j << 52) ^ j, cell.value);
+                    r.cells[0] = std::move(cell);
+                }
+                p.items.push_back(std::move(u));
+                if (has_rt && rt_open && j + 1 < g.clustering_rows &&
+                    gen_ck(g, e.id, j + 1) > rhi) {
+                    Unfiltered m;
+                    m.kind = Unfiltered::MARKER;
+                    m.marker.kind = INCL_END;
+                    m.marker.values = {ClusterVal{ClusterVal::VALUE, be8(rhi)}};
+                    m.marker.end_dt = DeletionTime{rts, rldt};
+                    p.items.push_back(std::move(m));
+                    rt_open = false;
+                }
+            }
+            if (has_rt && rt_open) {
+                Unfiltered m;
+                m.kind = Unfiltered::MARKER;
+                m.marker.kind = INCL_END;
+                m.marker.values = {ClusterVal{ClusterVal::VALUE, be8(rhi)}};
+                m.marker.end_dt = DeletionTime{rts, rldt};
+                p.items.push_back(std::move(m));
+            }
+            t.parts.push_back(std::move(p));
+            continue;
+        }
         Partition p;
         p.key = e.key;
         p.token = e.token;

@@ -25,6 +25,12 @@ struct GenSpec {
     int64_t base_ts = 1700000000000000LL;   // µs
     int64_t base_ldt = 1700000000LL;        // seconds
     uint64_t first_generation = 1;
+    // wide-partition (C4) shape: clustering_rows > 0 makes each partition hold
+    // clustering_rows rows keyed by a LongType clustering column; tombstone_pct
+    // then applies per ROW, and range_tomb_pct partitions get one range
+    // tombstone [a, b] over the clustering space.
+    uint32_t clustering_rows = 0;
+    uint32_t range_tomb_pct = 0;
 
     uint64_t stride() const { return rows_per_sstable * (100 - overlap_pct) / 100; }
     uint64_t universe() const {
@@ -61,6 +67,34 @@ inline int64_t gen_ts(const GenSpec& g, uint32_t sst, uint64_t key_id) {
 inline bool gen_is_tombstone(const GenSpec& g, uint32_t sst, uint64_t key_id) {
     if (g.tombstone_pct == 0) return false;
     return splitmix64(g.seed ^ 0xDEADULL ^ key_id ^ ((uint64_t)sst << 32)) % 100 < g.tombstone_pct;
+}
+// wide-partition derivations (rows keyed by (key_id, row j))
+inline int64_t gen_ck(const GenSpec& g, uint64_t key_id, uint32_t j) {
+    (void)g; (void)key_id;
+    return (int64_t)j * 16;  // ascending, gaps so range bounds can fall between rows
+}
+inline int64_t gen_row_ts(const GenSpec& g, uint32_t sst, uint64_t key_id, uint32_t j) {
+    return g.base_ts + (int64_t)(splitmix64(g.seed ^ key_id * 31 ^ ((uint64_t)sst << 48) ^ (uint64_t)(j + 1) * 0x9E37ULL) % 1000000000ULL);
+}
+inline bool gen_row_is_tombstone(const GenSpec& g, uint32_t sst, uint64_t key_id, uint32_t j) {
+    if (g.tombstone_pct == 0) return false;
+    return splitmix64(g.seed ^ 0xDEADULL ^ key_id ^ ((uint64_t)sst << 32) ^ (uint64_t)(j + 7) * 131) % 100 < g.tombstone_pct;
+}
+inline bool gen_has_range_tomb(const GenSpec& g, uint32_t sst, uint64_t key_id) {
+    if (g.range_tomb_pct == 0) return false;
+    return splitmix64(g.seed ^ 0xBEEFULL ^ key_id ^ ((uint64_t)sst << 32)) % 100 < g.range_tomb_pct;
+}
+// range [lo, hi] clustering values (inclusive bounds), lo < hi
+inline void gen_range_bounds(const GenSpec& g, uint32_t sst, uint64_t key_id, int64_t* lo, int64_t* hi) {
+    uint64_t r = splitmix64(g.seed ^ 0xB00BULL ^ key_id ^ ((uint64_t)sst << 32));
+    uint32_t a = (uint32_t)(r % g.clustering_rows);
+    uint32_t b = a + 1 + (uint32_t)((r >> 32) % (g.clustering_rows - a));
+    *lo = gen_ck(g, key_id, a) + 8;  // between rows a and a+1
+    *hi = gen_ck(g, key_id, b) + 8;
+    // NOTE: with +8 offsets the bounds never collide with row clusterings
+}
+inline int64_t gen_range_ts(const GenSpec& g, uint32_t sst, uint64_t key_id) {
+    return g.base_ts + (int64_t)(splitmix64(g.seed ^ 0xAB1EULL ^ key_id ^ ((uint64_t)sst << 48)) % 1000000000ULL);
 }
 inline bool gen_has_partition_del(const GenSpec& g, uint32_t sst, uint64_t key_id) {
     if (g.partition_del_pct == 0) return false;

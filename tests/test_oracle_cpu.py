@@ -97,3 +97,24 @@ def test_compact_is_idempotent(oracle_bin, tmp_path):
     a = open(f"{d}/oa-50-big-Data.db", "rb").read()
     b = open(f"{d}/oa-51-big-Data.db", "rb").read()
     assert a == b
+
+
+def test_wide_partitions_roundtrip_and_merge(oracle_bin, tmp_path):
+    """C4-shaped: clustering rows + row/range tombstones (oracle path)."""
+    d = str(tmp_path)
+    _gen(d, n=3, rows=60, crows=40, vlen=200, overlap=20, tomb=15, rtomb=30, seed=3)
+    ins = [f"{d}/oa-{g}-big" for g in (1, 2, 3)]
+    for b in ins:
+        r = oracle_run("roundtrip", b)
+        assert "FAIL" not in r.stdout, r.stdout
+    s = _compact(f"{d}/oa-90-big", ins)
+    assert s["rows_out"] < s["rows_in"]  # range tombstones shadow rows on merge
+    r = oracle_run("roundtrip", f"{d}/oa-90-big")
+    assert "FAIL" not in r.stdout, r.stdout
+    # idempotence
+    s2 = _compact(f"{d}/oa-91-big", [f"{d}/oa-90-big"])
+    assert open(f"{d}/oa-90-big-Data.db", "rb").read() == open(f"{d}/oa-91-big-Data.db", "rb").read()
+    assert s2["rows_out"] == s["rows_out"]
+    # gc purge drops tombstones
+    s3 = _compact(f"{d}/oa-92-big", ins, gcbefore=2000000000)
+    assert s3["rows_out"] < s["rows_out"]
