@@ -76,6 +76,8 @@ class GpucGenSpec(ctypes.Structure):
         ("value_repeat_pct", ctypes.c_uint32),
         ("tombstone_pct", ctypes.c_uint32),
         ("partition_del_pct", ctypes.c_uint32),
+        ("clustering_rows", ctypes.c_uint32),
+        ("range_tomb_pct", ctypes.c_uint32),
         ("base_ts", ctypes.c_int64),
         ("base_ldt", ctypes.c_int64),
         ("first_generation", ctypes.c_uint64),
@@ -204,6 +206,8 @@ def generate(
     value_repeat_pct=55,
     tombstone_pct=0,
     partition_del_pct=0,
+    clustering_rows=0,
+    range_tomb_pct=0,
     base_ts=1700000000000000,
     base_ldt=1700000000,
     first_generation=1,
@@ -220,6 +224,8 @@ def generate(
         value_repeat_pct=value_repeat_pct,
         tombstone_pct=tombstone_pct,
         partition_del_pct=partition_del_pct,
+        clustering_rows=clustering_rows,
+        range_tomb_pct=range_tomb_pct,
         base_ts=base_ts,
         base_ldt=base_ldt,
         first_generation=first_generation,

@@ -27,18 +27,23 @@ __host__ __device__ inline bool mrec_eq(const MRec& a, const MRec& b) {
     return a.tok == b.tok && a.pfx == b.pfx && a.klen == b.klen;
 }
 
-// parsed per input-partition fields (SoA, concatenated across sources; index
-// space == MRec (src, idx) resolved through per-source base offsets)
-struct ParsedCols {
-    int64_t* pdel_mfda;
-    uint32_t* pdel_ldt;
-    uint8_t* flags;       // bit0 has_row, bit1 row_live_ts, bit2 row_del,
-                          // bit3 has_cell, bit4 cell_has_value, bit5 cell_expiring
+// unfiltered-level SoA (rows AND range-tombstone markers), used for parsed
+// input rows and for reconciled output rows. One clustering column of fixed
+// width (LongType/Int32Type) or none; `ck` is the sortable encoding
+// (big-endian fixed value as signed -> value ^ SIGN so unsigned compare ==
+// the ClusteringComparator order). `rkind` is the ClusteringPrefix.Kind
+// ordinal (CLUSTERING_K = 4 for rows; bound/boundary kinds for markers).
+struct UnfCols {
+    uint64_t* ck;
+    uint8_t* rkind;
+    uint8_t* flags;      // PF_* for rows
     int64_t* live_ts;
     int32_t* live_ttl;
     int64_t* live_let;
-    int64_t* rdel_mfda;
+    int64_t* rdel_mfda;  // row deletion, or marker end/close deletion
     uint32_t* rdel_ldt;
+    int64_t* start_mfda;  // marker open deletion (boundary markers)
+    uint32_t* start_ldt;
     int64_t* cell_ts;
     uint32_t* cell_ldt;
     int32_t* cell_ttl;
@@ -49,6 +54,30 @@ enum : uint8_t {
     PF_HAS_ROW = 1, PF_LIVE_TS = 2, PF_ROW_DEL = 4, PF_HAS_CELL = 8,
     PF_CELL_VALUE = 16, PF_CELL_EXPIRING = 32
 };
+// ClusteringPrefix.Kind ordinals (ClusteringPrefix.java:65-85)
+enum : uint8_t {
+    BK_EXCL_END = 0, BK_INCL_START = 1, BK_EXCL_END_INCL_START = 2, BK_STATIC = 3,
+    BK_CLUSTERING = 4, BK_INCL_END_EXCL_START = 5, BK_INCL_END = 6, BK_EXCL_START = 7
+};
+__host__ __device__ inline int bk_comparison(uint8_t k) {
+    // Kind(comparison, …) ctor args (ClusteringPrefix.java:70-79)
+    const int tbl[8] = {0, 0, 0, 1, 2, 3, 3, 3};
+    return tbl[k];
+}
+__host__ __device__ inline bool bk_is_boundary(uint8_t k) {
+    return k == BK_EXCL_END_INCL_START || k == BK_INCL_END_EXCL_START;
+}
+__host__ __device__ inline bool bk_is_open(uint8_t k) {
+    return k == BK_INCL_START || k == BK_EXCL_START || bk_is_boundary(k);
+}
+
+// parsed per input-partition fields (partition level)
+struct ParsedCols {
+    int64_t* pdel_mfda;
+    uint32_t* pdel_ldt;
+    uint32_t* row_count;  // unfiltereds in this partition
+    uint64_t* row_base;   // start into the per-source-concatenated UnfCols
+};
 
 // reconciled output partitions (also produced directly by the generator)
 struct OutParts {
@@ -56,17 +85,8 @@ struct OutParts {
     uint8_t* klen;
     int64_t* pdel_mfda;
     uint32_t* pdel_ldt;
-    uint8_t* flags;      // PF_* as above; PF_HAS_ROW==0 -> deletion-only partition
-    int64_t* live_ts;
-    int32_t* live_ttl;
-    int64_t* live_let;
-    int64_t* rdel_mfda;
-    uint32_t* rdel_ldt;
-    int64_t* cell_ts;
-    uint32_t* cell_ldt;
-    int32_t* cell_ttl;
-    uint64_t* val_addr;
-    uint32_t* val_len;
+    uint64_t* row_base;  // into the output UnfCols arena
+    uint32_t* row_count; // merged+purged unfiltereds
     uint8_t* keep;       // 0 = dropped (purged empty / out of shard)
 };
 

@@ -6,6 +6,7 @@
 // purge/serialize/compress/CRC/bloom all run in the kernels. There is no CPU
 // fallback — gpuc_compact fails with GPUC_ERR_NO_GPU when no device exists.
 #include "kernels.hip"
+#include "kernels_rows.hip"
 #include "../../include/gpucompact.h"
 
 #include <algorithm>
@@ -318,6 +319,7 @@ struct OutMeta {
     // header (deltas) of the output sstable
     HeaderStats hs;
     std::string key_type;
+    std::string ck_type;  // empty = no clustering column
     std::vector<std::pair<bytes, std::string>> regular_cols;
     // collected stats
     int64_t min_timestamp, max_timestamp, min_ldt, max_ldt;
@@ -362,7 +364,10 @@ static bytes serialize_statistics_out(const OutMeta& m) {
         for (auto& [pt, cnt] : m.tomb_hist) { put_be64(stats, pt); put_be32(stats, cnt); }
         put_be32(stats, 0);               // sstableLevel
         put_be64(stats, 0);               // repairedAt
-        put_uvint(stats, 0);              // improvedMinMax: clustering type count 0
+        // improvedMinMax: clustering type list + covered Slice (spec'd with the
+        // oracle: BOTTOM..TOP bounds, no values)
+        put_uvint(stats, m.ck_type.empty() ? 0 : 1);
+        if (!m.ck_type.empty()) put_type_str(stats, m.ck_type);
         stats.push_back(1); put_be16(stats, 0);  // Slice start: INCL_START, 0 values
         stats.push_back(6); put_be16(stats, 0);  // Slice end: INCL_END, 0 values
         stats.push_back(0);               // hasLegacyCounterShards
@@ -386,7 +391,8 @@ static bytes serialize_statistics_out(const OutMeta& m) {
         put_uvint(header, sext32(m.hs.min_ldt - DELETION_TIME_EPOCH));
         put_uvint(header, sext32(m.hs.min_ttl));
         put_type_str(header, m.key_type);
-        put_uvint(header, 0);  // clustering types
+        put_uvint(header, m.ck_type.empty() ? 0 : 1);
+        if (!m.ck_type.empty()) put_type_str(header, m.ck_type);
         put_uvint(header, 0);  // static columns
         put_uvint(header, m.regular_cols.size());
         for (auto& [name, t] : m.regular_cols) {
@@ -614,20 +620,34 @@ static void ensure_crc_tables(hipStream_t stream) {
 
 // allocate the OutParts SoA for n entries
 struct OutPartsBuf {
-    DevBuf keypfx, klen, pdel_mfda, pdel_ldt, flags, live_ts, live_ttl, live_let,
-        rdel_mfda, rdel_ldt, cell_ts, cell_ldt, cell_ttl, val_addr, val_len, keep;
+    DevBuf keypfx, klen, pdel_mfda, pdel_ldt, row_base, row_count, keep;
     OutParts op{};
     void alloc(uint64_t n) {
         keypfx.alloc(n * 8); klen.alloc(n); pdel_mfda.alloc(n * 8); pdel_ldt.alloc(n * 4);
-        flags.alloc(n); live_ts.alloc(n * 8); live_ttl.alloc(n * 4); live_let.alloc(n * 8);
-        rdel_mfda.alloc(n * 8); rdel_ldt.alloc(n * 4); cell_ts.alloc(n * 8); cell_ldt.alloc(n * 4);
-        cell_ttl.alloc(n * 4); val_addr.alloc(n * 8); val_len.alloc(n * 4); keep.alloc(n);
+        row_base.alloc(n * 8); row_count.alloc(n * 4); keep.alloc(n);
         op = OutParts{keypfx.as<uint64_t>(), klen.as<uint8_t>(), pdel_mfda.as<int64_t>(),
-                      pdel_ldt.as<uint32_t>(), flags.as<uint8_t>(), live_ts.as<int64_t>(),
-                      live_ttl.as<int32_t>(), live_let.as<int64_t>(), rdel_mfda.as<int64_t>(),
-                      rdel_ldt.as<uint32_t>(), cell_ts.as<int64_t>(), cell_ldt.as<uint32_t>(),
-                      cell_ttl.as<int32_t>(), val_addr.as<uint64_t>(), val_len.as<uint32_t>(),
+                      pdel_ldt.as<uint32_t>(), row_base.as<uint64_t>(), row_count.as<uint32_t>(),
                       keep.as<uint8_t>()};
+    }
+};
+
+// allocate an UnfCols arena for n unfiltereds
+struct UnfColsBuf {
+    DevBuf ck, rkind, flags, live_ts, live_ttl, live_let, rdel_mfda, rdel_ldt,
+        start_mfda, start_ldt, cell_ts, cell_ldt, cell_ttl, val_addr, val_len;
+    UnfCols uc{};
+    void alloc(uint64_t n) {
+        if (!n) n = 1;
+        ck.alloc(n * 8); rkind.alloc(n); flags.alloc(n); live_ts.alloc(n * 8);
+        live_ttl.alloc(n * 4); live_let.alloc(n * 8); rdel_mfda.alloc(n * 8);
+        rdel_ldt.alloc(n * 4); start_mfda.alloc(n * 8); start_ldt.alloc(n * 4);
+        cell_ts.alloc(n * 8); cell_ldt.alloc(n * 4); cell_ttl.alloc(n * 4);
+        val_addr.alloc(n * 8); val_len.alloc(n * 4);
+        uc = UnfCols{ck.as<uint64_t>(), rkind.as<uint8_t>(), flags.as<uint8_t>(),
+                     live_ts.as<int64_t>(), live_ttl.as<int32_t>(), live_let.as<int64_t>(),
+                     rdel_mfda.as<int64_t>(), rdel_ldt.as<uint32_t>(), start_mfda.as<int64_t>(),
+                     start_ldt.as<uint32_t>(), cell_ts.as<int64_t>(), cell_ldt.as<uint32_t>(),
+                     cell_ttl.as<int32_t>(), val_addr.as<uint64_t>(), val_len.as<uint32_t>()};
     }
 };
 
@@ -641,14 +661,15 @@ struct WriteDeviceOut {
 // shared device->files writer: sizes/scan -> serialize -> compress -> gather ->
 // D2H -> write all components. `st` is the device OutStats already filled by
 // reconcile (or gen); meta_* give header/schema info.
-static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, uint64_t n_groups,
-                                           SerParams sp, DevBuf& d_stats, DevBuf& d_tomb,
-                                           uint32_t tomb_cap, const std::string& out_base,
+static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfColsBuf& rows,
+                                           uint64_t n_groups, SerParams2 sp, DevBuf& d_stats,
+                                           DevBuf& d_tomb, uint32_t tomb_cap,
+                                           const std::string& out_base,
                                            const std::string& key_type,
+                                           const std::string& ck_type,
                                            const std::vector<std::pair<bytes, std::string>>& regular_cols,
                                            hipStream_t stream) {
     WriteDeviceOut w;
-    // histogram offset tables on device
     static const std::vector<int64_t> ps_off_h = est_hist_offsets(155);
     static const std::vector<int64_t> ch_off_h = est_hist_offsets(118);
     DevBuf d_ps_off, d_ch_off;
@@ -661,29 +682,32 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, uint64_t n_gr
     HIP_CHECK(hipEventCreate(&ev0)); HIP_CHECK(hipEventCreate(&ev1)); HIP_CHECK(hipEventCreate(&ev2));
     HIP_CHECK(hipEventCreate(&ev3)); HIP_CHECK(hipEventCreate(&ev4));
 
+    // ---- collect stats (needed before bloom sizing) ----
+    {
+        uint64_t blocks = (n_groups + 255) / 256;
+        hipLaunchKernelGGL(k_collect_rows, dim3((uint32_t)blocks), dim3(256), 0, stream, opb.op,
+                           rows.uc, n_groups, d_stats.as<OutStats>(), d_tomb.as<uint32_t>(), tomb_cap);
+    }
+
     // ---- sizes + scans ----
     HIP_CHECK(hipEventRecord(ev0, stream));
-    DevBuf d_psize, d_isize;
+    DevBuf d_psize, d_isize, d_nblocks, d_infsz;
     d_psize.alloc(n_groups * 8);
     d_isize.alloc(n_groups * 8);
+    d_nblocks.alloc(n_groups * 4);
+    d_infsz.alloc(n_groups * 8);
     {
         uint64_t blocks = (n_groups + 255) / 256;
-        hipLaunchKernelGGL(k_sizes, dim3((uint32_t)blocks), dim3(256), 0, stream, opb.op, n_groups,
-                           sp, d_psize.as<uint64_t>(), d_stats.as<OutStats>(),
-                           d_ps_off.as<int64_t>(), (int32_t)ps_off_h.size(),
+        hipLaunchKernelGGL(k_sizes_rows, dim3((uint32_t)blocks), dim3(256), 0, stream, opb.op,
+                           rows.uc, n_groups, sp, d_psize.as<uint64_t>(), d_isize.as<uint64_t>(),
+                           d_nblocks.as<uint32_t>(), d_infsz.as<uint64_t>(),
+                           d_stats.as<OutStats>(), d_ps_off.as<int64_t>(), (int32_t)ps_off_h.size(),
                            d_ch_off.as<int64_t>(), (int32_t)ch_off_h.size());
     }
-    // d_psize becomes exclusive offsets in place after the scan
     uint64_t total_unc = exscan_u64(d_psize.as<uint64_t>(), n_groups, stream);
-    {
-        uint64_t blocks = (n_groups + 255) / 256;
-        hipLaunchKernelGGL(k_index_sizes, dim3((uint32_t)blocks), dim3(256), 0, stream, opb.op,
-                           n_groups, d_psize.as<uint64_t>(), d_isize.as<uint64_t>());
-    }
     uint64_t total_idx = exscan_u64(d_isize.as<uint64_t>(), n_groups, stream);
     HIP_CHECK(hipEventRecord(ev1, stream));
 
-    // ---- stats D2H (bloom needs partitions count) ----
     OutStats hst;
     HIP_CHECK(hipStreamSynchronize(stream));
     HIP_CHECK(hipMemcpy(&hst, d_stats.p, sizeof(OutStats), hipMemcpyDeviceToHost));
@@ -703,10 +727,11 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, uint64_t n_gr
     {
         uint32_t waves_per_block = 4;
         uint64_t blocks = (n_groups + waves_per_block - 1) / waves_per_block;
-        hipLaunchKernelGGL(k_serialize, dim3((uint32_t)blocks), dim3(WAVE * waves_per_block), 0,
-                           stream, opb.op, n_groups, sp, d_psize.as<uint64_t>(),
-                           d_isize.as<uint64_t>(), d_out_data.as<uint8_t>(),
-                           d_out_index.as<uint8_t>(), d_bloom.as<uint32_t>(), words * 64, bs.k);
+        hipLaunchKernelGGL(k_serialize_rows, dim3((uint32_t)blocks), dim3(WAVE * waves_per_block),
+                           0, stream, opb.op, rows.uc, n_groups, sp, d_psize.as<uint64_t>(),
+                           d_isize.as<uint64_t>(), d_nblocks.as<uint32_t>(), d_infsz.as<uint64_t>(),
+                           d_out_data.as<uint8_t>(), d_out_index.as<uint8_t>(),
+                           d_bloom.as<uint32_t>(), words * 64, bs.k);
     }
     HIP_CHECK(hipEventRecord(ev2, stream));
 
@@ -721,8 +746,6 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, uint64_t n_gr
                        d_out_data.as<uint8_t>(), total_unc, d_slots.as<uint8_t>(),
                        d_csize.as<uint32_t>(), d_ccrc.as<uint32_t>(), n_chunks,
                        (const uint32_t*)g_crc256);
-    // widen sizes to u64 + scan -> chunk offsets (excluding per-chunk CRC;
-    // gather adds 4*c for preceding CRCs)
     {
         std::vector<uint32_t> cs(n_chunks);
         HIP_CHECK(hipStreamSynchronize(stream));
@@ -732,7 +755,6 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, uint64_t n_gr
         for (uint32_t i = 0; i < n_chunks; i++) { off[i] = acc; acc += cs[i]; }
         HIP_CHECK(hipMemcpy(d_csize64.p, off.data(), (uint64_t)n_chunks * 8, hipMemcpyHostToDevice));
         w.compressed_len = acc + (uint64_t)n_chunks * 4;
-        // CompressionInfo offsets == off[i] + 4*i
         DevBuf d_final;
         d_final.alloc(w.compressed_len);
         hipLaunchKernelGGL(k_chunk_gather, dim3(n_chunks), dim3(WAVE), 0, stream,
@@ -753,7 +775,6 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, uint64_t n_gr
         HIP_CHECK(hipEventRecord(ev4, stream));
         HIP_CHECK(hipStreamSynchronize(stream));
 
-        // tombstone ldts
         std::vector<uint32_t> tombs;
         if (hst.tomb_count) {
             uint64_t nt = std::min<uint64_t>(hst.tomb_count, tomb_cap);
@@ -762,25 +783,23 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, uint64_t n_gr
             if (hst.tomb_count > tomb_cap)
                 throw std::runtime_error("tombstone list overflow (internal cap)");
         }
-        // first/last key
         OutMeta m{};
         m.hs = sp.hs;
         m.key_type = key_type;
+        m.ck_type = ck_type;
         m.regular_cols = regular_cols;
-        m.min_timestamp = hst.min_ts_flip == 0xFFFFFFFFFFFFFFFFULL ? 0 : (int64_t)(hst.min_ts_flip ^ 0x8000000000000000ULL);
-        m.max_timestamp = hst.max_ts_flip == 0 ? 0 : (int64_t)(hst.max_ts_flip ^ 0x8000000000000000ULL);
         bool no_ts = hst.min_ts_flip == 0xFFFFFFFFFFFFFFFFULL;
-        if (no_ts) { m.min_timestamp = 0; m.max_timestamp = 0; }
+        m.min_timestamp = no_ts ? 0 : (int64_t)(hst.min_ts_flip ^ 0x8000000000000000ULL);
+        m.max_timestamp = no_ts ? 0 : (int64_t)(hst.max_ts_flip ^ 0x8000000000000000ULL);
         m.min_ldt = hst.min_ldt_flip == 0xFFFFFFFFFFFFFFFFULL ? NO_DELETION_TIME
                                                               : (int64_t)(hst.min_ldt_flip ^ 0x8000000000000000ULL);
-        m.max_ldt = hst.max_ldt_flip == 0 ? NO_DELETION_TIME : (int64_t)(hst.max_ldt_flip ^ 0x8000000000000000ULL);
-        if (hst.max_ldt_flip == 0) m.max_ldt = 0;
+        m.max_ldt = hst.max_ldt_flip == 0 ? 0 : (int64_t)(hst.max_ldt_flip ^ 0x8000000000000000ULL);
         m.min_ttl = hst.min_ttl == 0xFFFFFFFFu ? 0 : (int32_t)hst.min_ttl;
         m.max_ttl = (int32_t)hst.max_ttl;
         m.total_rows = hst.rows_out;
         m.total_cells = hst.total_cells;
         m.has_partition_deletions = hst.has_partition_deletions != 0;
-        for (uint32_t t : tombs) m.tomb_hist[t]++;
+        for (uint32_t tv : tombs) m.tomb_hist[tv]++;
         memcpy(m.part_size_hist, hst.part_size_hist, sizeof(m.part_size_hist));
         memcpy(m.cells_hist, hst.cells_hist, sizeof(m.cells_hist));
         m.compression_ratio = total_unc ? (double)(w.compressed_len - (uint64_t)n_chunks * 4) / (double)total_unc : -1.0;
@@ -795,18 +814,13 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, uint64_t n_gr
             HIP_CHECK(hipMemcpy(&kl, opb.op.klen + lg, 1, hipMemcpyDeviceToHost));
             for (int b = 0; b < kl; b++) m.last_key.push_back((uint8_t)(kp >> (8 * (7 - b))));
         }
-
-        // Digest: CRC32 of the whole Data.db via per-chunk combine
         static Crc32Combiner comb;
         uint32_t digest = 0;
-        {
-            std::vector<uint32_t> cs2(cs);
-            for (uint32_t c = 0; c < n_chunks; c++) {
-                digest = comb.combine(digest, h_crc[c], cs2[c]);
-                uint8_t cb[4] = {(uint8_t)(h_crc[c] >> 24), (uint8_t)(h_crc[c] >> 16),
-                                 (uint8_t)(h_crc[c] >> 8), (uint8_t)h_crc[c]};
-                digest = comb.combine(digest, crc32_update_bitwise(0, cb, 4), 4);
-            }
+        for (uint32_t c = 0; c < n_chunks; c++) {
+            digest = comb.combine(digest, h_crc[c], cs[c]);
+            uint8_t cb[4] = {(uint8_t)(h_crc[c] >> 24), (uint8_t)(h_crc[c] >> 16),
+                             (uint8_t)(h_crc[c] >> 8), (uint8_t)h_crc[c]};
+            digest = comb.combine(digest, crc32_update_bitwise(0, cb, 4), 4);
         }
 
         write_file_parallel(out_base + "-Data.db", h_data, w.compressed_len, 6);
@@ -918,8 +932,8 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
             cinfos[s] = parse_compression_info(read_file(base + "-CompressionInfo.db"));
             read_file_into(base + "-Data.db", comp_pin[s], comp_sz[s], 6);
             stats[s] = parse_statistics(read_file(base + "-Statistics.db"));
-            if (!stats[s].clustering_types.empty())
-                throw std::runtime_error("clustering columns unsupported in round 1 (GPU path)");
+            if (stats[s].clustering_types.size() > 1)
+                throw std::runtime_error("at most one clustering column supported");
             if (stats[s].regular_cols.size() != 1)
                 throw std::runtime_error("exactly one regular column supported in round 1");
             if (!stats[s].partitioner.empty() && stats[s].partitioner.find("Murmur3") == std::string::npos)
@@ -936,6 +950,15 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
         int32_t col_fixed = -1;
         if (col_type == "org.apache.cassandra.db.marshal.LongType") col_fixed = 8;
         else if (col_type == "org.apache.cassandra.db.marshal.Int32Type") col_fixed = 4;
+        // clustering column (0 or 1, fixed width)
+        int32_t ck_width = 0;
+        std::string ck_type_str;
+        if (!stats[0].clustering_types.empty()) {
+            ck_type_str = stats[0].clustering_types[0];
+            if (ck_type_str == "org.apache.cassandra.db.marshal.LongType") ck_width = 8;
+            else if (ck_type_str == "org.apache.cassandra.db.marshal.Int32Type") ck_width = 4;
+            else throw std::runtime_error("unsupported clustering type " + ck_type_str);
+        }
 
         std::vector<std::vector<uint64_t>> positions(k);
         std::vector<std::string> perr(k);
@@ -989,9 +1012,9 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
                            d_error.as<unsigned long long>(), (const uint32_t*)g_crc256);
         HIP_CHECK(hipEventRecord(e2, stream));
 
-        // ---- parse ----
+        // ---- parse (pass A: count + partition meta; pass B: row decode) ----
         uint64_t total_parts = 0;
-        std::vector<SrcDesc> srcs(k);
+        std::vector<SrcDesc2> srcs(k);
         for (int s = 0; s < k; s++) {
             srcs[s].data = d_data[s].as<uint8_t>();
             srcs[s].part_pos = d_pos[s].as<uint64_t>();
@@ -1004,35 +1027,40 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
         }
         res->partitions_in = total_parts;
         if (total_parts > 0xFFFFFFFFull) throw std::runtime_error("too many partitions for one job");
+        SchemaParams sch{};
+        sch.ck_width = ck_width;
+        sch.col_fixed_len = col_fixed;
+        sch.column_index_size = 64 * 1024;
         DevBuf d_srcs, d_recs_a, d_recs_b, d_rows_in;
-        d_srcs.alloc(srcs.size() * sizeof(SrcDesc));
-        HIP_CHECK(hipMemcpyAsync(d_srcs.p, srcs.data(), srcs.size() * sizeof(SrcDesc),
+        d_srcs.alloc(srcs.size() * sizeof(SrcDesc2));
+        HIP_CHECK(hipMemcpyAsync(d_srcs.p, srcs.data(), srcs.size() * sizeof(SrcDesc2),
                                  hipMemcpyHostToDevice, stream));
         d_recs_a.alloc(total_parts * sizeof(MRec));
         d_recs_b.alloc(total_parts * sizeof(MRec));
         d_rows_in.alloc(8);
         HIP_CHECK(hipMemsetAsync(d_rows_in.p, 0, 8, stream));
-        // parsed columns SoA
-        DevBuf p_pdm, p_pdl, p_fl, p_lts, p_lttl, p_llet, p_rdm, p_rdl, p_cts, p_cldt, p_cttl, p_va, p_vl;
-        ParsedCols pc;
+        DevBuf p_pdm, p_pdl, p_rcnt, p_rbase;
+        ParsedCols pc{};
         p_pdm.alloc(total_parts * 8); pc.pdel_mfda = p_pdm.as<int64_t>();
         p_pdl.alloc(total_parts * 4); pc.pdel_ldt = p_pdl.as<uint32_t>();
-        p_fl.alloc(total_parts); pc.flags = p_fl.as<uint8_t>();
-        p_lts.alloc(total_parts * 8); pc.live_ts = p_lts.as<int64_t>();
-        p_lttl.alloc(total_parts * 4); pc.live_ttl = p_lttl.as<int32_t>();
-        p_llet.alloc(total_parts * 8); pc.live_let = p_llet.as<int64_t>();
-        p_rdm.alloc(total_parts * 8); pc.rdel_mfda = p_rdm.as<int64_t>();
-        p_rdl.alloc(total_parts * 4); pc.rdel_ldt = p_rdl.as<uint32_t>();
-        p_cts.alloc(total_parts * 8); pc.cell_ts = p_cts.as<int64_t>();
-        p_cldt.alloc(total_parts * 4); pc.cell_ldt = p_cldt.as<uint32_t>();
-        p_cttl.alloc(total_parts * 4); pc.cell_ttl = p_cttl.as<int32_t>();
-        p_va.alloc(total_parts * 8); pc.val_addr = p_va.as<uint64_t>();
-        p_vl.alloc(total_parts * 4); pc.val_len = p_vl.as<uint32_t>();
+        p_rcnt.alloc(total_parts * 4); pc.row_count = p_rcnt.as<uint32_t>();
+        p_rbase.alloc(total_parts * 8); pc.row_base = p_rbase.as<uint64_t>();
         {
             uint32_t blocks = (uint32_t)((total_parts + 255) / 256);
-            hipLaunchKernelGGL(k_parse, dim3(blocks), dim3(256), 0, stream, d_srcs.as<SrcDesc>(),
-                               (uint32_t)k, (uint32_t)total_parts, d_recs_a.as<MRec>(), pc,
-                               col_fixed, d_error.as<unsigned long long>(),
+            hipLaunchKernelGGL(k_parse_count, dim3(blocks), dim3(256), 0, stream,
+                               d_srcs.as<SrcDesc2>(), (uint32_t)k, (uint32_t)total_parts,
+                               d_recs_a.as<MRec>(), pc, sch, d_error.as<unsigned long long>());
+            hipLaunchKernelGGL(k_widen_u32, dim3(blocks), dim3(256), 0, stream,
+                               pc.row_count, pc.row_base, total_parts);
+        }
+        uint64_t total_in_rows = exscan_u64(pc.row_base, total_parts, stream);
+        UnfColsBuf in_rows;
+        in_rows.alloc(total_in_rows);
+        {
+            uint32_t blocks = (uint32_t)((total_parts + 255) / 256);
+            hipLaunchKernelGGL(k_parse_rows, dim3(blocks), dim3(256), 0, stream,
+                               d_srcs.as<SrcDesc2>(), (uint32_t)k, (uint32_t)total_parts, pc,
+                               in_rows.uc, sch, d_error.as<unsigned long long>(),
                                d_rows_in.as<unsigned long long>());
         }
         HIP_CHECK(hipEventRecord(e3, stream));
@@ -1060,11 +1088,8 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
             hipLaunchKernelGGL(k_group_heads, dim3(blocks), dim3(256), 0, stream, d_sorted,
                                total_parts, d_head.as<uint64_t>());
         }
-        exscan_u64(d_head.as<uint64_t>(), total_parts, stream);  // head -> exclusive scan... need inclusive-1
-        // recompute heads (scan consumed them); write starts via scan values
+        exscan_u64(d_head.as<uint64_t>(), total_parts, stream);
         {
-            // d_head now holds exclusive scan of head flags == group id for head elements.
-            // Recreate head flags on the fly in k_group_starts by comparing recs again.
             uint32_t blocks = (uint32_t)((total_parts + 255) / 256);
             hipLaunchKernelGGL(k_group_starts2, dim3(blocks), dim3(256), 0, stream, d_sorted,
                                total_parts, d_head.as<uint64_t>(), d_gstart.as<uint64_t>(),
@@ -1075,21 +1100,32 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
         HIP_CHECK(hipMemcpy(&n_groups, d_ngroups.p, 8, hipMemcpyDeviceToHost));
 
         // ---- reconcile + purge ----
-        OutPartsBuf opb;
-        opb.alloc(n_groups);
-        DevBuf d_stats, d_tomb, d_srcbases;
-        d_stats.alloc(sizeof(OutStats));
-        init_outstats(d_stats, stream);
-        uint32_t tomb_cap = (uint32_t)std::min<uint64_t>(n_groups * 3 + 1024, 200000000ull);
-        d_tomb.alloc((uint64_t)tomb_cap * 4);
+        DevBuf d_srcbases, d_group_rows;
         {
             std::vector<uint32_t> bases(k);
             for (int s = 0; s < k; s++) bases[s] = srcs[s].rec_base;
             d_srcbases.alloc(k * 4);
             HIP_CHECK(hipMemcpyAsync(d_srcbases.p, bases.data(), k * 4, hipMemcpyHostToDevice, stream));
         }
+        d_group_rows.alloc(n_groups * 8);
+        {
+            uint32_t blocks = (uint32_t)((n_groups + 255) / 256);
+            hipLaunchKernelGGL(k_group_row_sums, dim3(blocks), dim3(256), 0, stream, d_sorted,
+                               d_gstart.as<uint64_t>(), n_groups, total_parts,
+                               d_srcbases.as<uint32_t>(), pc, d_group_rows.as<uint64_t>());
+        }
+        uint64_t total_out_rows = exscan_u64(d_group_rows.as<uint64_t>(), n_groups, stream);
+        OutPartsBuf opb;
+        opb.alloc(n_groups);
+        UnfColsBuf out_rows;
+        out_rows.alloc(total_out_rows);
+        DevBuf d_stats, d_tomb;
+        d_stats.alloc(sizeof(OutStats));
+        init_outstats(d_stats, stream);
+        uint32_t tomb_cap = (uint32_t)std::min<uint64_t>(total_out_rows * 2 + n_groups + 1024, 400000000ull);
+        d_tomb.alloc((uint64_t)tomb_cap * 4);
         DevBuf d_ov_lo, d_ov_hi, d_ov_ts;
-        PurgeParams pp{};
+        PurgeParams2 pp{};
         pp.now_sec = job->now_sec;
         pp.gc_before = job->gc_before;
         pp.never_purge = job->never_purge;
@@ -1117,21 +1153,23 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
         }
         {
             uint32_t blocks = (uint32_t)((n_groups + 255) / 256);
-            hipLaunchKernelGGL(k_reconcile, dim3(blocks), dim3(256), 0, stream, d_sorted,
+            hipLaunchKernelGGL(k_reconcile_rows, dim3(blocks), dim3(256), 0, stream, d_sorted,
                                d_gstart.as<uint64_t>(), n_groups, total_parts,
-                               d_srcbases.as<uint32_t>(), pc, opb.op, pp, d_stats.as<OutStats>(),
-                               d_tomb.as<uint32_t>(), tomb_cap);
-        }
-        {
-            uint32_t blocks = (uint32_t)((n_groups + 255) / 256);
-            hipLaunchKernelGGL(k_collect_outstats, dim3(blocks), dim3(256), 0, stream, opb.op,
-                               n_groups, d_stats.as<OutStats>(), d_tomb.as<uint32_t>(), tomb_cap);
+                               d_srcbases.as<uint32_t>(), pc, in_rows.uc, opb.op, out_rows.uc,
+                               d_group_rows.as<uint64_t>(), sch, pp, d_stats.as<OutStats>(),
+                               d_error.as<unsigned long long>());
         }
         HIP_CHECK(hipEventRecord(e4, stream));
         HIP_CHECK(hipStreamSynchronize(stream));
+        {
+            unsigned long long err = 0;
+            HIP_CHECK(hipMemcpy(&err, d_error.p, 8, hipMemcpyDeviceToHost));
+            if (err) throw std::runtime_error("GPU reconcile error code " + std::to_string(err));
+        }
 
         // ---- output header (SerializationHeader.make: desc-generation stats merge) ----
-        SerParams sp{};
+        SerParams2 sp2{};
+        sp2.sch = sch;
         {
             std::vector<int> order(k);
             for (int s = 0; s < k; s++) order[s] = s;
@@ -1144,15 +1182,13 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
                 min_ldt = std::min(min_ldt, stats[s].min_ldt);
                 min_ttl = std::min(min_ttl, stats[s].min_ttl);
             }
-            sp.hs.min_ts = min_ts == NO_TIMESTAMP ? TIMESTAMP_EPOCH : min_ts;
-            sp.hs.min_ldt = min_ldt == NO_DELETION_TIME ? DELETION_TIME_EPOCH : min_ldt;
-            sp.hs.min_ttl = min_ttl == INT32_MAX ? 0 : min_ttl;
-            sp.col_fixed_len = col_fixed;
+            sp2.hs.min_ts = min_ts == NO_TIMESTAMP ? TIMESTAMP_EPOCH : min_ts;
+            sp2.hs.min_ldt = min_ldt == NO_DELETION_TIME ? DELETION_TIME_EPOCH : min_ldt;
+            sp2.hs.min_ttl = min_ttl == INT32_MAX ? 0 : min_ttl;
         }
-
-        WriteDeviceOut w = write_sstable_device(opb, n_groups, sp, d_stats, d_tomb, tomb_cap,
-                                                job->output_base, stats[0].key_type,
-                                                stats[0].regular_cols, stream);
+        WriteDeviceOut w = write_sstable_device(opb, out_rows, n_groups, sp2, d_stats, d_tomb,
+                                                tomb_cap, job->output_base, stats[0].key_type,
+                                                ck_type_str, stats[0].regular_cols, stream);
         res->partitions_out = w.partitions;
         res->rows_out = w.rows;
         res->output_uncompressed_bytes = w.uncompressed_len;
@@ -1206,7 +1242,7 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
         uint64_t stride = R * (100 - spec->overlap_pct) / 100;
         uint64_t universe = std::max<uint64_t>(stride * spec->n_sstables, R);
         for (uint32_t s = 0; s < spec->n_sstables; s++) {
-            GenParams gp{};
+            GenParams2 gp{};
             gp.seed = spec->seed;
             gp.universe = universe;
             gp.stride = stride;
@@ -1216,56 +1252,64 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
             gp.value_repeat_pct = spec->value_repeat_pct;
             gp.tombstone_pct = spec->tombstone_pct;
             gp.partition_del_pct = spec->partition_del_pct;
+            gp.clustering_rows = spec->clustering_rows;
+            gp.range_tomb_pct = spec->range_tomb_pct;
             gp.base_ts = spec->base_ts;
             gp.base_ldt = spec->base_ldt;
-            DevBuf d_a, d_b, d_ids, d_vals, d_stats, d_tomb;
+            DevBuf d_a, d_b, d_ids, d_vals, d_stats, d_tomb, d_prows;
             d_a.alloc(R * sizeof(MRec));
             d_b.alloc(R * sizeof(MRec));
             d_ids.alloc(R * 8);
-            {
-                uint32_t blocks = (uint32_t)((R + 255) / 256);
-                hipLaunchKernelGGL(k_gen_recs, dim3(blocks), dim3(256), 0, stream, gp,
-                                   d_a.as<MRec>(), d_ids.as<uint64_t>());
-            }
+            uint32_t blocks = (uint32_t)((R + 255) / 256);
+            hipLaunchKernelGGL(k_gen_recs2, dim3(blocks), dim3(256), 0, stream, gp,
+                               d_a.as<MRec>(), d_ids.as<uint64_t>());
             MRec* d_sorted = merge_sort_recs(d_a.as<MRec>(), d_b.as<MRec>(), R, stream);
             OutPartsBuf opb;
             opb.alloc(R);
-            d_vals.alloc(R * (uint64_t)spec->value_len);
+            d_prows.alloc(R * 8);
+            hipLaunchKernelGGL(k_gen_count, dim3(blocks), dim3(256), 0, stream, gp, d_sorted,
+                               d_ids.as<uint64_t>(), R, opb.op, d_prows.as<uint64_t>());
+            uint64_t total_rows = exscan_u64(d_prows.as<uint64_t>(), R, stream);
+            UnfColsBuf rows;
+            rows.alloc(total_rows);
+            d_vals.alloc(total_rows * (uint64_t)spec->value_len);
+            hipLaunchKernelGGL(k_gen_fill2, dim3(blocks), dim3(256), 0, stream, gp, d_sorted,
+                               d_ids.as<uint64_t>(), R, opb.op, rows.uc, d_prows.as<uint64_t>(),
+                               d_vals.as<uint8_t>());
+            hipLaunchKernelGGL(k_gen_values2, dim3(blocks), dim3(256), 0, stream, gp, d_sorted,
+                               d_ids.as<uint64_t>(), opb.op, rows.uc, R, d_vals.as<uint8_t>(),
+                               d_prows.as<uint64_t>());
             d_stats.alloc(sizeof(OutStats));
             init_outstats(d_stats, stream);
-            uint32_t tomb_cap = (uint32_t)std::min<uint64_t>(R + 1024, 200000000ull);
+            uint32_t tomb_cap = (uint32_t)std::min<uint64_t>(total_rows * 2 + R + 1024, 400000000ull);
             d_tomb.alloc((uint64_t)tomb_cap * 4);
             {
-                uint32_t blocks = (uint32_t)((R + 255) / 256);
-                hipLaunchKernelGGL(k_gen_fill, dim3(blocks), dim3(256), 0, stream, gp, d_sorted,
-                                   d_ids.as<uint64_t>(), R, opb.op, d_vals.as<uint8_t>());
-                hipLaunchKernelGGL(k_gen_values, dim3(blocks), dim3(256), 0, stream, gp, d_sorted,
-                                   d_ids.as<uint64_t>(), opb.op, R, d_vals.as<uint8_t>());
-            }
-            // collect stats over generated partitions; header mins derive from them
-            {
-                uint32_t blocks = (uint32_t)((R + 255) / 256);
-                hipLaunchKernelGGL(k_collect_outstats, dim3(blocks), dim3(256), 0, stream, opb.op,
-                                   R, d_stats.as<OutStats>(), d_tomb.as<uint32_t>(), tomb_cap);
+                hipLaunchKernelGGL(k_collect_rows, dim3(blocks), dim3(256), 0, stream, opb.op,
+                                   rows.uc, R, d_stats.as<OutStats>(), d_tomb.as<uint32_t>(), tomb_cap);
             }
             OutStats hs0;
             HIP_CHECK(hipStreamSynchronize(stream));
             HIP_CHECK(hipMemcpy(&hs0, d_stats.p, sizeof(OutStats), hipMemcpyDeviceToHost));
-            SerParams sp{};
+            SerParams2 sp{};
             sp.hs.min_ts = hs0.min_ts_flip == 0xFFFFFFFFFFFFFFFFULL
                                ? TIMESTAMP_EPOCH : (int64_t)(hs0.min_ts_flip ^ 0x8000000000000000ULL);
-            // header min_ldt: min over tombstone ldts only; live rows push
-            // NO_DELETION_TIME which maps to the epoch (EncodingStats ctor)
             sp.hs.min_ldt = hs0.min_ldt_flip == 0xFFFFFFFFFFFFFFFFULL
                                 ? DELETION_TIME_EPOCH : (int64_t)(hs0.min_ldt_flip ^ 0x8000000000000000ULL);
             if (sp.hs.min_ldt == NO_DELETION_TIME) sp.hs.min_ldt = DELETION_TIME_EPOCH;
             sp.hs.min_ttl = 0;
-            sp.col_fixed_len = -1;  // val blob
+            sp.sch.ck_width = spec->clustering_rows ? 8 : 0;
+            sp.sch.col_fixed_len = -1;  // val blob
+            sp.sch.column_index_size = 64 * 1024;
+            // reset and recollect so the writer sees fresh stats (collect ran
+            // once above only to derive the header mins)
+            init_outstats(d_stats, stream);
             std::string base = std::string(dir) + "/oa-" + std::to_string(spec->first_generation + s) + "-big";
             std::vector<std::pair<bytes, std::string>> cols = {
                 {bytes{'v', 'a', 'l'}, "org.apache.cassandra.db.marshal.BytesType"}};
-            write_sstable_device(opb, R, sp, d_stats, d_tomb, tomb_cap, base,
-                                 "org.apache.cassandra.db.marshal.LongType", cols, stream);
+            write_sstable_device(opb, rows, R, sp, d_stats, d_tomb, tomb_cap, base,
+                                 "org.apache.cassandra.db.marshal.LongType",
+                                 spec->clustering_rows ? "org.apache.cassandra.db.marshal.LongType" : "",
+                                 cols, stream);
         }
         HIP_CHECK(hipStreamDestroy(stream));
         return GPUC_OK;
@@ -1274,3 +1318,4 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
         return GPUC_ERR_INTERNAL;
     }
 }
+

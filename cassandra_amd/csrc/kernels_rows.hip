@@ -1,0 +1,1418 @@
+// PRODUCT — general (wide-partition) kernels: rows + range-tombstone markers
+// under one fixed-width clustering column (or none). Replaces the round-1
+// single-row kernels; the simple schema is the row_count<=1 special case of
+// the same code path.
+//
+// Reference semantics restated here:
+//   partition frame / row wire format: SortedTablePartitionWriter.java:97-166,
+//     UnfilteredSerializer.java:36-305, Cell.java:240-306
+//   merge: UnfilteredRowIterators.merge (:400-599), Row.Merger (Row.java:694-791),
+//     Cells.reconcile (Cells.java:68-119), RangeTombstoneMarker.Merger
+//     (RangeTombstoneMarker.java:72-198)
+//   purge: PurgeFunction.java:26-145, CompactionController.java:247-286
+//   promoted index: BigFormatPartitionWriter.java:128-245, IndexInfo.java:90-118,
+//     RowIndexEntry.java:460-483,625-647
+#pragma once
+#include <hip/hip_runtime.h>
+#include "codec.h"
+#include "gpu_structs.h"
+
+namespace gpuc {
+
+__device__ inline void tomb_push(OutStats* st, uint32_t* ldts, uint32_t cap, uint32_t ldt) {
+    unsigned long long i = atomicAdd(&st->tomb_count, 1ull);
+    if (i < cap) ldts[i] = ldt;
+}
+
+// schema/runtime constants shared by the general kernels
+struct SchemaParams {
+    int32_t ck_width;        // 0 = no clustering column; 4/8 fixed width
+    int32_t col_fixed_len;   // regular column: -1 variable else fixed width
+    uint32_t column_index_size;  // promoted-index granularity (64 KiB default)
+};
+
+// sortable ck encoding: big-endian fixed-width signed value -> flip sign bit
+__device__ inline uint64_t ck_sortable(const uint8_t* p, int width) {
+    uint64_t v = 0;
+    for (int b = 0; b < width; b++) v = (v << 8) | p[b];
+    // values compare as signed integers of `width` bytes: flip the sign bit
+    return v ^ (1ULL << (8 * width - 1));
+}
+__device__ inline void ck_bytes(uint64_t ck, int width, uint8_t* out) {
+    uint64_t v = ck ^ (1ULL << (8 * width - 1));
+    for (int b = 0; b < width; b++) out[b] = (uint8_t)(v >> (8 * (width - 1 - b)));
+}
+// position compare: (ck, Kind.comparison); returns <0, 0, >0
+__device__ inline int pos_cmp(uint64_t cka, uint8_t ka, uint64_t ckb, uint8_t kb, int ck_width) {
+    if (ck_width > 0) {
+        if (cka != ckb) return cka < ckb ? -1 : 1;
+    }
+    int c1 = bk_comparison(ka), c2 = bk_comparison(kb);
+    return c1 - c2;
+}
+
+// ---------------------------------------------------------------------------
+// parse pass A: one thread per partition — key/token/partition deletion,
+// count unfiltereds (skip-walk via the size vints)
+// ---------------------------------------------------------------------------
+struct SrcDesc2 {
+    const uint8_t* data;
+    const uint64_t* part_pos;  // n_parts+1
+    uint32_t n_parts;
+    int64_t min_ts, min_ldt;
+    int32_t min_ttl;
+    uint32_t rec_base;         // partition index base in the concatenated arrays
+};
+
+__global__ void k_parse_count(const SrcDesc2* srcs, uint32_t n_srcs, uint32_t total,
+                              MRec* recs, ParsedCols pc, SchemaParams sp,
+                              unsigned long long* error) {
+    uint32_t gi = blockIdx.x * blockDim.x + threadIdx.x;
+    if (gi >= total) return;
+    uint32_t s = 0;
+    while (s + 1 < n_srcs && gi >= srcs[s + 1].rec_base) s++;
+    const SrcDesc2& sd = srcs[s];
+    uint32_t li = gi - sd.rec_base;
+    const uint8_t* base = sd.data;
+    uint64_t pos = sd.part_pos[li];
+    uint64_t end = sd.part_pos[li + 1];
+
+    uint32_t klen = ((uint32_t)base[pos] << 8) | base[pos + 1];
+    pos += 2;
+    if (klen == 0 || klen > 8) { atomicExch(error, 10ull); return; }
+    uint64_t pfx = 0;
+    uint8_t keyb[8];
+    for (uint32_t b = 0; b < klen; b++) { keyb[b] = base[pos + b]; pfx |= (uint64_t)keyb[b] << (8 * (7 - b)); }
+    pos += klen;
+    int64_t token = murmur3_token(keyb, klen);
+    MRec r{(uint64_t)token ^ 0x8000000000000000ULL, pfx, li, (uint16_t)s, (uint8_t)klen, 0};
+    recs[gi] = r;
+
+    int64_t pdm = INT64_MIN;
+    uint32_t pdl = LDT_NONE_U32;
+    {
+        uint8_t f = base[pos];
+        if (f & 0x80) { pos++; if (f != 0x80) { atomicExch(error, 11ull); return; } }
+        else {
+            uint64_t v = 0;
+            for (int i = 0; i < 8; i++) v = (v << 8) | base[pos + i];
+            pdm = (int64_t)v;
+            pdl = ((uint32_t)base[pos + 8] << 24) | ((uint32_t)base[pos + 9] << 16) |
+                  ((uint32_t)base[pos + 10] << 8) | base[pos + 11];
+            pos += 12;
+        }
+    }
+    pc.pdel_mfda[gi] = pdm;
+    pc.pdel_ldt[gi] = pdl;
+
+    // skip-walk the unfiltereds
+    uint32_t count = 0;
+    while (true) {
+        uint8_t flags = base[pos++];
+        if (flags & 0x01) break;  // END_OF_PARTITION
+        if (flags & 0xC0) { atomicExch(error, 12ull); return; }  // complex/extension
+        if (flags & 0x02) {
+            // marker: kind, u16 size, values
+            uint8_t kind = base[pos++];
+            uint32_t nv = ((uint32_t)base[pos] << 8) | base[pos + 1];
+            pos += 2;
+            if (kind == BK_STATIC || nv > 1 || (nv == 1 && sp.ck_width == 0)) { atomicExch(error, 15ull); return; }
+            if (nv == 0 && sp.ck_width != 0) { atomicExch(error, 16ull); return; }  // 0-value bound unsupported
+            if (nv) { uint64_t hdr = uvint_get(base, &pos); if (hdr) { atomicExch(error, 17ull); return; } pos += sp.ck_width; }
+        } else if (sp.ck_width) {
+            uint64_t hdr = uvint_get(base, &pos);
+            if (hdr) { atomicExch(error, 17ull); return; }  // null/empty clustering unsupported
+            pos += sp.ck_width;
+        }
+        uint64_t size = uvint_get(base, &pos);
+        uint64_t prev = uvint_get(base, &pos);
+        pos += size - uvint_size(prev);  // body
+        count++;
+        if (pos > end) { atomicExch(error, 14ull); return; }
+    }
+    if (pos != end) { atomicExch(error, 14ull); return; }
+    pc.row_count[gi] = count;
+}
+
+// ---------------------------------------------------------------------------
+// parse pass B: full decode into the UnfCols arena (row_base prescanned)
+// ---------------------------------------------------------------------------
+__global__ void k_parse_rows(const SrcDesc2* srcs, uint32_t n_srcs, uint32_t total,
+                             ParsedCols pc, UnfCols rc, SchemaParams sp,
+                             unsigned long long* error, unsigned long long* rows_in) {
+    uint32_t gi = blockIdx.x * blockDim.x + threadIdx.x;
+    if (gi >= total) return;
+    uint32_t s = 0;
+    while (s + 1 < n_srcs && gi >= srcs[s + 1].rec_base) s++;
+    const SrcDesc2& sd = srcs[s];
+    uint32_t li = gi - sd.rec_base;
+    const uint8_t* base = sd.data;
+    uint64_t pos = sd.part_pos[li];
+    // skip key + partition deletion
+    uint32_t klen = ((uint32_t)base[pos] << 8) | base[pos + 1];
+    pos += 2 + klen;
+    pos += (base[pos] & 0x80) ? 1 : 12;
+
+    uint64_t out = pc.row_base[gi];
+    uint32_t emitted = 0;
+    while (true) {
+        uint8_t flags = base[pos++];
+        if (flags & 0x01) break;
+        uint64_t o = out + emitted;
+        if (flags & 0x02) {
+            // ---- marker ----
+            uint8_t kind = base[pos++];
+            uint32_t nv = ((uint32_t)base[pos] << 8) | base[pos + 1];
+            pos += 2;
+            uint64_t ck = 0;
+            if (nv) { uvint_get(base, &pos); ck = ck_sortable(base + pos, sp.ck_width); pos += sp.ck_width; }
+            uvint_get(base, &pos);  // size
+            uvint_get(base, &pos);  // prev
+            rc.rkind[o] = kind;
+            rc.ck[o] = ck;
+            rc.flags[o] = 0;
+            // deltas: boundary = end then start; bound = single deletion
+            int64_t em = (int64_t)uvint_get(base, &pos) + sd.min_ts;
+            uint32_t el = ldt_u32((int64_t)(int32_t)(uint32_t)uvint_get(base, &pos) + sd.min_ldt);
+            rc.rdel_mfda[o] = em;
+            rc.rdel_ldt[o] = el;
+            if (bk_is_boundary(kind)) {
+                int64_t sm = (int64_t)uvint_get(base, &pos) + sd.min_ts;
+                uint32_t sl = ldt_u32((int64_t)(int32_t)(uint32_t)uvint_get(base, &pos) + sd.min_ldt);
+                rc.start_mfda[o] = sm;
+                rc.start_ldt[o] = sl;
+            } else {
+                rc.start_mfda[o] = INT64_MIN;
+                rc.start_ldt[o] = LDT_NONE_U32;
+            }
+            rc.live_ts[o] = NO_TIMESTAMP;
+            rc.live_ttl[o] = 0;
+            rc.live_let[o] = NO_DELETION_TIME;
+            rc.cell_ts[o] = NO_TIMESTAMP;
+            rc.cell_ldt[o] = LDT_NONE_U32;
+            rc.cell_ttl[o] = 0;
+            rc.val_addr[o] = 0;
+            rc.val_len[o] = 0;
+        } else {
+            // ---- row ----
+            uint64_t ck = 0;
+            if (sp.ck_width) { uvint_get(base, &pos); ck = ck_sortable(base + pos, sp.ck_width); pos += sp.ck_width; }
+            uvint_get(base, &pos);  // size
+            uvint_get(base, &pos);  // prev
+            uint8_t pf = PF_HAS_ROW;
+            int64_t lts = NO_TIMESTAMP, llet = NO_DELETION_TIME, rdm = INT64_MIN;
+            int32_t lttl = 0;
+            uint32_t rdl = LDT_NONE_U32;
+            int64_t cts = NO_TIMESTAMP;
+            uint32_t cldt = LDT_NONE_U32;
+            int32_t cttl = 0;
+            uint64_t vaddr = 0;
+            uint32_t vlen = 0;
+            if (flags & 0x04) { pf |= PF_LIVE_TS; lts = (int64_t)uvint_get(base, &pos) + sd.min_ts; }
+            if (flags & 0x08) {
+                lttl = (int32_t)(uint32_t)uvint_get(base, &pos) + sd.min_ttl;
+                llet = (int64_t)(int32_t)(uint32_t)uvint_get(base, &pos) + sd.min_ldt;
+            }
+            if (flags & 0x10) {
+                pf |= PF_ROW_DEL;
+                rdm = (int64_t)uvint_get(base, &pos) + sd.min_ts;
+                rdl = ldt_u32((int64_t)(int32_t)(uint32_t)uvint_get(base, &pos) + sd.min_ldt);
+            }
+            bool has_cell;
+            if (flags & 0x20) has_cell = true;
+            else has_cell = !(uvint_get(base, &pos) & 1);
+            if (has_cell) {
+                pf |= PF_HAS_CELL;
+                uint8_t cf = base[pos++];
+                cts = (cf & 8) ? lts : (int64_t)uvint_get(base, &pos) + sd.min_ts;
+                bool dead = cf & 1, exp = cf & 2;
+                int64_t ldtl;
+                if (cf & 16) ldtl = llet;
+                else if (dead || exp) ldtl = (int64_t)(int32_t)(uint32_t)uvint_get(base, &pos) + sd.min_ldt;
+                else ldtl = NO_DELETION_TIME;
+                cttl = (cf & 16) ? lttl : (exp ? (int32_t)(uint32_t)uvint_get(base, &pos) + sd.min_ttl : 0);
+                cldt = ldt_u32(ldtl);
+                if (exp) pf |= PF_CELL_EXPIRING;
+                if (!(cf & 4)) {
+                    pf |= PF_CELL_VALUE;
+                    vlen = sp.col_fixed_len >= 0 ? (uint32_t)sp.col_fixed_len : (uint32_t)uvint_get(base, &pos);
+                    vaddr = (uint64_t)(base + pos);
+                    pos += vlen;
+                }
+            }
+            rc.rkind[o] = BK_CLUSTERING;
+            rc.ck[o] = ck;
+            rc.flags[o] = pf;
+            rc.live_ts[o] = lts;
+            rc.live_ttl[o] = lttl;
+            rc.live_let[o] = llet;
+            rc.rdel_mfda[o] = rdm;
+            rc.rdel_ldt[o] = rdl;
+            rc.start_mfda[o] = INT64_MIN;
+            rc.start_ldt[o] = LDT_NONE_U32;
+            rc.cell_ts[o] = cts;
+            rc.cell_ldt[o] = cldt;
+            rc.cell_ttl[o] = cttl;
+            rc.val_addr[o] = vaddr;
+            rc.val_len[o] = vlen;
+            atomicAdd(rows_in, 1ull);
+        }
+        emitted++;
+    }
+    (void)error;
+}
+
+__global__ void k_widen_u32(const uint32_t* in, uint64_t* out, uint64_t n) {
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < n) out[i] = in[i];
+}
+
+// ---------------------------------------------------------------------------
+// group input row totals (for the output arena layout)
+// ---------------------------------------------------------------------------
+__global__ void k_group_row_sums(const MRec* recs, const uint64_t* group_start,
+                                 uint64_t n_groups, uint64_t n_recs,
+                                 const uint32_t* src_bases, ParsedCols pc,
+                                 uint64_t* group_rows) {
+    uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (g >= n_groups) return;
+    uint64_t beg = group_start[g];
+    uint64_t endi = g + 1 < n_groups ? group_start[g + 1] : n_recs;
+    uint64_t sum = 0;
+    for (uint64_t m = beg; m < endi; m++)
+        sum += pc.row_count[src_bases[recs[m].src] + recs[m].idx];
+    group_rows[g] = sum;
+}
+
+// ---------------------------------------------------------------------------
+// reconcile + purge, general: one thread per group merges the ≤k versions'
+// row/marker streams and purges, writing merged unfiltereds to the out arena.
+// ---------------------------------------------------------------------------
+struct PurgeParams2 {
+    int64_t now_sec, gc_before;
+    int32_t never_purge;
+    int32_t enforce_strict_liveness;
+    const int64_t* ov_lo;
+    const int64_t* ov_hi;
+    const int64_t* ov_min_ts;
+    int32_t n_overlaps;
+    int32_t has_shard;
+    int64_t shard_lo, shard_hi;
+};
+
+__device__ inline bool purge_eval2(const PurgeParams2& pp, int64_t token, int64_t ts) {
+    int64_t min_ts = INT64_MAX;
+    bool has = false;
+    for (int i = 0; i < pp.n_overlaps; i++)
+        if (token >= pp.ov_lo[i] && token <= pp.ov_hi[i]) { has = true; min_ts = min(min_ts, pp.ov_min_ts[i]); }
+    return !has || ts < min_ts;
+}
+__device__ inline bool should_purge2(const PurgeParams2& pp, int64_t token, int64_t ts, int64_t ldt) {
+    if (pp.never_purge) return false;
+    return ldt < pp.gc_before && purge_eval2(pp, token, ts);
+}
+__device__ inline bool dt_sup(int64_t am, uint32_t al, int64_t bm, uint32_t bl) {
+    return am > bm || (am == bm && ldt_long(al) > ldt_long(bl));
+}
+
+#define GPUC_MAX_ARITY 64
+
+__global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
+                                 uint64_t n_groups, uint64_t n_recs,
+                                 const uint32_t* src_bases, ParsedCols pc, UnfCols in,
+                                 OutParts op, UnfCols out, const uint64_t* out_base,
+                                 SchemaParams sp, PurgeParams2 pp, OutStats* st,
+                                 unsigned long long* error) {
+    uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (g >= n_groups) return;
+    uint64_t beg = group_start[g];
+    uint64_t endi = g + 1 < n_groups ? group_start[g + 1] : n_recs;
+    uint32_t k = (uint32_t)(endi - beg);
+    atomicAdd(&st->merged_counts[k > 64 ? 63 : k - 1], 1ull);
+    if (k > GPUC_MAX_ARITY) { atomicExch(error, 20ull); return; }
+
+    const MRec r0 = recs[beg];
+    int64_t token = (int64_t)(r0.tok ^ 0x8000000000000000ULL);
+    op.keypfx[g] = r0.pfx;
+    op.klen[g] = r0.klen;
+    op.row_base[g] = out_base[g];
+    op.row_count[g] = 0;
+    if (pp.has_shard && (token < pp.shard_lo || token > pp.shard_hi)) { op.keep[g] = 0; return; }
+
+    // member streams
+    uint64_t mb[GPUC_MAX_ARITY];
+    uint32_t mcnt[GPUC_MAX_ARITY], mpos[GPUC_MAX_ARITY];
+    int64_t pdm = INT64_MIN;
+    uint32_t pdl = LDT_NONE_U32;
+    for (uint32_t m = 0; m < k; m++) {
+        uint32_t i = src_bases[recs[beg + m].src] + recs[beg + m].idx;
+        mb[m] = pc.row_base[i];
+        mcnt[m] = pc.row_count[i];
+        mpos[m] = 0;
+        if (!dt_sup(pdm, pdl, pc.pdel_mfda[i], pc.pdel_ldt[i])) { pdm = pc.pdel_mfda[i]; pdl = pc.pdel_ldt[i]; }
+    }
+    bool pdel_live0 = pdm == INT64_MIN && pdl == LDT_NONE_U32;
+    if (!pdel_live0 && should_purge2(pp, token, pdm, ldt_long(pdl))) { pdm = INT64_MIN; pdl = LDT_NONE_U32; }
+    // NOTE: purge of the partition deletion happens AFTER merge in the
+    // reference pipeline, but merge uses the UNPURGED value as activeDeletion.
+    int64_t adm = INT64_MIN;   // merged partition deletion for shadow filtering
+    uint32_t adl = LDT_NONE_U32;
+    {
+        // recompute un-purged merged value (pdm/pdl may have been purged above)
+        int64_t m2 = INT64_MIN;
+        uint32_t l2 = LDT_NONE_U32;
+        for (uint32_t m = 0; m < k; m++) {
+            uint32_t i = src_bases[recs[beg + m].src] + recs[beg + m].idx;
+            if (!dt_sup(m2, l2, pc.pdel_mfda[i], pc.pdel_ldt[i])) { m2 = pc.pdel_mfda[i]; l2 = pc.pdel_ldt[i]; }
+        }
+        adm = m2;
+        adl = l2;
+    }
+
+    // marker-merger state (RangeTombstoneMarker.Merger)
+    int64_t om_m[GPUC_MAX_ARITY];
+    uint32_t om_l[GPUC_MAX_ARITY];
+    uint64_t om_set = 0;  // bit per member
+
+    auto merged_open = [&](int64_t* mm, uint32_t* ml) {
+        // currentOpenDeletionTimeInMerged: biggest open not superseding pdel -> LIVE
+        int64_t bm = INT64_MIN;
+        uint32_t bl = LDT_NONE_U32;
+        bool any = false;
+        for (uint32_t m = 0; m < k; m++)
+            if (om_set & (1ULL << m))
+                if (!any || dt_sup(om_m[m], om_l[m], bm, bl)) { bm = om_m[m]; bl = om_l[m]; any = true; }
+        if (!any || !dt_sup(bm, bl, adm, adl)) { *mm = INT64_MIN; *ml = LDT_NONE_U32; }
+        else { *mm = bm; *ml = bl; }
+    };
+
+    uint64_t obase = out_base[g];
+    uint32_t ocount = 0;
+    auto emit = [&](uint8_t kind, uint64_t ck, uint8_t flags, int64_t lts, int32_t lttl,
+                    int64_t llet, int64_t rdm, uint32_t rdl, int64_t smf, uint32_t sld,
+                    int64_t cts, uint32_t cldt, int32_t cttl, uint64_t va, uint32_t vl) {
+        uint64_t o = obase + ocount++;
+        out.rkind[o] = kind;
+        out.ck[o] = ck;
+        out.flags[o] = flags;
+        out.live_ts[o] = lts;
+        out.live_ttl[o] = lttl;
+        out.live_let[o] = llet;
+        out.rdel_mfda[o] = rdm;
+        out.rdel_ldt[o] = rdl;
+        out.start_mfda[o] = smf;
+        out.start_ldt[o] = sld;
+        out.cell_ts[o] = cts;
+        out.cell_ldt[o] = cldt;
+        out.cell_ttl[o] = cttl;
+        out.val_addr[o] = va;
+        out.val_len[o] = vl;
+    };
+
+    // purge helpers
+    auto purge_dt = [&](int64_t m, uint32_t l) { return m != INT64_MIN && should_purge2(pp, token, m, ldt_long(l)); };
+
+    while (true) {
+        // find min position
+        int first = -1;
+        for (uint32_t m = 0; m < k; m++) {
+            if (mpos[m] >= mcnt[m]) continue;
+            uint64_t o = mb[m] + mpos[m];
+            if (first < 0) { first = (int)m; continue; }
+            uint64_t f = mb[first] + mpos[first];
+            if (pos_cmp(in.ck[o], in.rkind[o], in.ck[f], in.rkind[f], sp.ck_width) < 0) first = (int)m;
+        }
+        if (first < 0) break;
+        uint64_t fo = mb[first] + mpos[first];
+        uint64_t fck = in.ck[fo];
+        uint8_t fkind = in.rkind[fo];
+        bool is_row = bk_comparison(fkind) == 2;
+        // gather members at this position
+        uint64_t members = 0;
+        uint32_t nmem = 0;
+        int lastm = -1;
+        for (uint32_t m = 0; m < k; m++) {
+            if (mpos[m] >= mcnt[m]) continue;
+            uint64_t o = mb[m] + mpos[m];
+            if (pos_cmp(in.ck[o], in.rkind[o], fck, fkind, sp.ck_width) == 0) {
+                members |= 1ULL << m;
+                nmem++;
+                lastm = (int)m;
+            }
+        }
+
+        if (is_row) {
+            // active deletion for rows = open marker in merged stream, else pdel
+            int64_t am, al_;
+            uint32_t al;
+            {
+                int64_t mm;
+                uint32_t ml;
+                merged_open(&mm, &ml);
+                if (mm == INT64_MIN && ml == LDT_NONE_U32) { am = adm; al = adl; }
+                else { am = mm; al = ml; }
+                al_ = 0;
+                (void)al_;
+            }
+            bool active_live = am == INT64_MIN && al == LDT_NONE_U32;
+            // merge row versions (Row.Merger); k==1 group passthrough applies
+            // at the PARTITION level (UnfilteredRowIterators.merge of 1 iterator)
+            uint8_t of = 0;
+            int64_t lts = NO_TIMESTAMP, llet = NO_DELETION_TIME, rdm = INT64_MIN;
+            int32_t lttl = 0;
+            uint32_t rdl = LDT_NONE_U32;
+            int64_t cts = NO_TIMESTAMP;
+            uint32_t cldt = LDT_NONE_U32;
+            int32_t cttl = 0;
+            uint64_t va = 0;
+            uint32_t vl = 0;
+            bool have_cell = false, cell_val = false, cell_exp = false;
+            if (k == 1 || (nmem == 1 && active_live)) {
+                uint64_t o = mb[lastm] + mpos[lastm];
+                of = in.flags[o];
+                lts = in.live_ts[o]; lttl = in.live_ttl[o]; llet = in.live_let[o];
+                rdm = in.rdel_mfda[o]; rdl = in.rdel_ldt[o];
+                cts = in.cell_ts[o]; cldt = in.cell_ldt[o]; cttl = in.cell_ttl[o];
+                va = in.val_addr[o]; vl = in.val_len[o];
+            } else {
+                bool has_live = false;
+                for (uint32_t m = 0; m < k; m++) {
+                    if (!(members & (1ULL << m))) continue;
+                    uint64_t o = mb[m] + mpos[m];
+                    uint8_t f = in.flags[o];
+                    if (f & PF_LIVE_TS) {
+                        int64_t t2 = in.live_ts[o];
+                        int32_t ttl2 = in.live_ttl[o];
+                        int64_t let2 = in.live_let[o];
+                        bool sup;
+                        if (!has_live) sup = true;
+                        else if (t2 != lts) sup = t2 > lts;
+                        else {
+                            bool e1 = lttl == INT32_MAX, e2 = ttl2 == INT32_MAX;
+                            if (e1 != e2) sup = e2;
+                            else if ((lttl != 0) == (ttl2 != 0)) sup = let2 > llet;
+                            else sup = ttl2 != 0;
+                        }
+                        if (sup) { lts = t2; lttl = ttl2; llet = let2; has_live = true; }
+                    }
+                    if (f & PF_ROW_DEL)
+                        if (dt_sup(in.rdel_mfda[o], in.rdel_ldt[o], rdm, rdl)) { rdm = in.rdel_mfda[o]; rdl = in.rdel_ldt[o]; }
+                }
+                if (has_live) of |= PF_LIVE_TS;
+                int64_t am2 = am;
+                uint32_t al2 = al;
+                bool row_del_kept = false;
+                if (dt_sup(rdm, rdl, am2, al2)) { am2 = rdm; al2 = rdl; row_del_kept = true; }
+                if (!row_del_kept) { rdm = INT64_MIN; rdl = LDT_NONE_U32; }
+                else of |= PF_ROW_DEL;
+                if (has_live && lts <= am2) { of &= ~PF_LIVE_TS; lts = NO_TIMESTAMP; lttl = 0; llet = NO_DELETION_TIME; }
+                for (uint32_t m = 0; m < k; m++) {
+                    if (!(members & (1ULL << m))) continue;
+                    uint64_t o = mb[m] + mpos[m];
+                    uint8_t f = in.flags[o];
+                    if (!(f & PF_HAS_CELL)) continue;
+                    int64_t ts2 = in.cell_ts[o];
+                    if (ts2 <= am2) continue;
+                    if (!have_cell) {
+                        have_cell = true;
+                        cts = ts2; cldt = in.cell_ldt[o]; cttl = in.cell_ttl[o];
+                        va = in.val_addr[o]; vl = in.val_len[o];
+                        cell_val = f & PF_CELL_VALUE;
+                        cell_exp = f & PF_CELL_EXPIRING;
+                        continue;
+                    }
+                    bool take_right = false;
+                    uint32_t rl = in.cell_ldt[o];
+                    bool l_dt = cldt != LDT_NONE_U32, r_dt = rl != LDT_NONE_U32;
+                    if (cts != ts2) take_right = ts2 > cts;
+                    else if (l_dt || r_dt) {
+                        if (l_dt != r_dt) take_right = r_dt;
+                        else {
+                            bool l_tomb = !cell_exp, r_tomb = !(f & PF_CELL_EXPIRING);
+                            if (l_tomb != r_tomb) take_right = r_tomb;
+                            else if (cldt != rl) take_right = ldt_long(rl) > ldt_long(cldt);
+                            else {
+                                // compareValues (unsigned lexicographic)
+                                const uint8_t* lp = (const uint8_t*)va;
+                                const uint8_t* rp = (const uint8_t*)in.val_addr[o];
+                                uint32_t nn = min(vl, in.val_len[o]);
+                                int c = 0;
+                                for (uint32_t x = 0; x < nn; x++)
+                                    if (lp[x] != rp[x]) { c = lp[x] < rp[x] ? -1 : 1; break; }
+                                if (c == 0) c = vl == in.val_len[o] ? 0 : (vl < in.val_len[o] ? -1 : 1);
+                                take_right = c < 0;
+                            }
+                        }
+                    } else {
+                        const uint8_t* lp = (const uint8_t*)va;
+                        const uint8_t* rp = (const uint8_t*)in.val_addr[o];
+                        uint32_t nn = min(vl, in.val_len[o]);
+                        int c = 0;
+                        for (uint32_t x = 0; x < nn; x++)
+                            if (lp[x] != rp[x]) { c = lp[x] < rp[x] ? -1 : 1; break; }
+                        if (c == 0) c = vl == in.val_len[o] ? 0 : (vl < in.val_len[o] ? -1 : 1);
+                        take_right = c < 0;
+                    }
+                    if (take_right) {
+                        cts = ts2; cldt = rl; cttl = in.cell_ttl[o];
+                        va = in.val_addr[o]; vl = in.val_len[o];
+                        cell_val = f & PF_CELL_VALUE;
+                        cell_exp = f & PF_CELL_EXPIRING;
+                    }
+                }
+                if (have_cell) of |= PF_HAS_CELL;
+                if (cell_val) of |= PF_CELL_VALUE;
+                if (cell_exp) of |= PF_CELL_EXPIRING;
+                if (of & (PF_LIVE_TS | PF_ROW_DEL | PF_HAS_CELL)) of |= PF_HAS_ROW;
+                else of = 0;
+            }
+            // ---- purge the merged row (BTreeRow.purge + AbstractCell.purge) ----
+            if (of & PF_HAS_ROW) {
+                if (of & PF_LIVE_TS) {
+                    bool is_live = lttl == INT32_MAX ? false : (lttl != 0 ? pp.now_sec < llet : true);
+                    if (!is_live && should_purge2(pp, token, lts, llet)) { of &= ~PF_LIVE_TS; lts = NO_TIMESTAMP; lttl = 0; llet = NO_DELETION_TIME; }
+                }
+                if ((of & PF_ROW_DEL) && purge_dt(rdm, rdl)) { of &= ~PF_ROW_DEL; rdm = INT64_MIN; rdl = LDT_NONE_U32; }
+                if (of & PF_HAS_CELL) {
+                    bool live_cell = cldt == LDT_NONE_U32 || (cttl != 0 && pp.now_sec < ldt_long(cldt));
+                    if (!live_cell) {
+                        if (should_purge2(pp, token, cts, ldt_long(cldt))) of &= ~(PF_HAS_CELL | PF_CELL_VALUE | PF_CELL_EXPIRING);
+                        else if (cttl != 0) {
+                            int64_t nldt = ldt_long(cldt) - cttl;
+                            if (should_purge2(pp, token, cts, nldt)) of &= ~(PF_HAS_CELL | PF_CELL_VALUE | PF_CELL_EXPIRING);
+                            else { cldt = ldt_u32(nldt); cttl = 0; of &= ~(PF_CELL_VALUE | PF_CELL_EXPIRING); vl = 0; }
+                        }
+                    }
+                }
+                if (!(of & (PF_LIVE_TS | PF_ROW_DEL | PF_HAS_CELL))) of = 0;
+                else if (pp.enforce_strict_liveness && !(of & PF_LIVE_TS) && !(of & PF_ROW_DEL)) of = 0;
+            }
+            if (of & PF_HAS_ROW)
+                emit(BK_CLUSTERING, fck, of, lts, lttl, llet, rdm, rdl, INT64_MIN, LDT_NONE_U32,
+                     cts, cldt, cttl, va, (of & PF_CELL_VALUE) ? vl : 0);
+        } else if (k == 1) {
+            // single-version partition: UnfilteredRowIterators.merge of one
+            // iterator returns it unchanged — markers pass through as-is and
+            // only the Purger applies (PurgeFunction.applyToMarker).
+            uint64_t o = mb[0] + mpos[0];
+            uint8_t kd = in.rkind[o];
+            int64_t e_m = in.rdel_mfda[o], s_m = in.start_mfda[o];
+            uint32_t e_l = in.rdel_ldt[o], s_l = in.start_ldt[o];
+            if (bk_is_boundary(kd)) {
+                bool purge_close = purge_dt(e_m, e_l);
+                bool purge_open = purge_dt(s_m, s_l);
+                if (purge_close && purge_open) {
+                } else if (purge_close) {
+                    emit(kd == BK_EXCL_END_INCL_START ? BK_INCL_START : BK_EXCL_START, fck, 0,
+                         NO_TIMESTAMP, 0, NO_DELETION_TIME, s_m, s_l, INT64_MIN, LDT_NONE_U32,
+                         NO_TIMESTAMP, LDT_NONE_U32, 0, 0, 0);
+                } else if (purge_open) {
+                    emit(kd == BK_EXCL_END_INCL_START ? BK_EXCL_END : BK_INCL_END, fck, 0,
+                         NO_TIMESTAMP, 0, NO_DELETION_TIME, e_m, e_l, INT64_MIN, LDT_NONE_U32,
+                         NO_TIMESTAMP, LDT_NONE_U32, 0, 0, 0);
+                } else {
+                    emit(kd, fck, 0, NO_TIMESTAMP, 0, NO_DELETION_TIME, e_m, e_l, s_m, s_l,
+                         NO_TIMESTAMP, LDT_NONE_U32, 0, 0, 0);
+                }
+            } else {
+                if (!purge_dt(e_m, e_l))
+                    emit(kd, fck, 0, NO_TIMESTAMP, 0, NO_DELETION_TIME, e_m, e_l,
+                         INT64_MIN, LDT_NONE_U32, NO_TIMESTAMP, LDT_NONE_U32, 0, 0, 0);
+            }
+        } else {
+            // ---- marker event (RangeTombstoneMarker.Merger.merge) ----
+            int64_t prev_m;
+            uint32_t prev_l;
+            merged_open(&prev_m, &prev_l);
+            for (uint32_t m = 0; m < k; m++) {
+                if (!(members & (1ULL << m))) continue;
+                uint64_t o = mb[m] + mpos[m];
+                uint8_t kd = in.rkind[o];
+                if (bk_is_open(kd)) {
+                    om_m[m] = bk_is_boundary(kd) ? in.start_mfda[o] : in.rdel_mfda[o];
+                    om_l[m] = bk_is_boundary(kd) ? in.start_ldt[o] : in.rdel_ldt[o];
+                    om_set |= 1ULL << m;
+                } else {
+                    om_set &= ~(1ULL << m);
+                }
+            }
+            int64_t next_m;
+            uint32_t next_l;
+            merged_open(&next_m, &next_l);
+            bool prev_live = prev_m == INT64_MIN && prev_l == LDT_NONE_U32;
+            bool next_live = next_m == INT64_MIN && next_l == LDT_NONE_U32;
+            if (!(prev_m == next_m && prev_l == next_l)) {
+                // kind selection (RangeTombstoneMarker.java:124-147), reversed=false
+                int ctc_tbl[8] = {-1, -1, -1, -1, 0, 1, 1, 1};
+                bool before_clustering = ctc_tbl[fkind] < 0;
+                uint8_t okind;
+                int64_t e_m = INT64_MIN, s_m = INT64_MIN;
+                uint32_t e_l = LDT_NONE_U32, s_l = LDT_NONE_U32;
+                if (prev_live) {
+                    okind = before_clustering ? BK_INCL_START : BK_EXCL_START;
+                    e_m = next_m; e_l = next_l;
+                } else if (next_live) {
+                    okind = before_clustering ? BK_EXCL_END : BK_INCL_END;
+                    e_m = prev_m; e_l = prev_l;
+                } else {
+                    okind = before_clustering ? BK_EXCL_END_INCL_START : BK_INCL_END_EXCL_START;
+                    e_m = prev_m; e_l = prev_l;
+                    s_m = next_m; s_l = next_l;
+                }
+                // ---- purge (PurgeFunction.applyToMarker, reversed=false) ----
+                if (bk_is_boundary(okind)) {
+                    bool purge_close = purge_dt(e_m, e_l);
+                    bool purge_open = purge_dt(s_m, s_l);
+                    if (purge_close && purge_open) {
+                    } else if (purge_close) {
+                        emit(okind == BK_EXCL_END_INCL_START ? BK_INCL_START : BK_EXCL_START, fck, 0,
+                             NO_TIMESTAMP, 0, NO_DELETION_TIME, s_m, s_l, INT64_MIN, LDT_NONE_U32,
+                             NO_TIMESTAMP, LDT_NONE_U32, 0, 0, 0);
+                    } else if (purge_open) {
+                        emit(okind == BK_EXCL_END_INCL_START ? BK_EXCL_END : BK_INCL_END, fck, 0,
+                             NO_TIMESTAMP, 0, NO_DELETION_TIME, e_m, e_l, INT64_MIN, LDT_NONE_U32,
+                             NO_TIMESTAMP, LDT_NONE_U32, 0, 0, 0);
+                    } else {
+                        emit(okind, fck, 0, NO_TIMESTAMP, 0, NO_DELETION_TIME, e_m, e_l, s_m, s_l,
+                             NO_TIMESTAMP, LDT_NONE_U32, 0, 0, 0);
+                    }
+                } else {
+                    if (!purge_dt(e_m, e_l))
+                        emit(okind, fck, 0, NO_TIMESTAMP, 0, NO_DELETION_TIME, e_m, e_l,
+                             INT64_MIN, LDT_NONE_U32, NO_TIMESTAMP, LDT_NONE_U32, 0, 0, 0);
+                }
+            }
+        }
+        // advance members
+        for (uint32_t m = 0; m < k; m++)
+            if (members & (1ULL << m)) mpos[m]++;
+    }
+    op.row_count[g] = ocount;
+    bool pdel_live = pdm == INT64_MIN && pdl == LDT_NONE_U32;
+    op.pdel_mfda[g] = pdm;
+    op.pdel_ldt[g] = pdl;
+    op.keep[g] = (!pdel_live || ocount > 0) ? 1 : 0;
+    if (op.keep[g]) {
+        atomicMin(&st->first_group, (unsigned long long)g);
+        atomicMax(&st->last_group, (unsigned long long)g);
+    }
+}
+
+}  // namespace gpuc
+
+namespace gpuc {
+
+// ---------------------------------------------------------------------------
+// serialization (SIZE pass then EMIT pass share one templated walk)
+// ---------------------------------------------------------------------------
+struct SerParams2 {
+    HeaderStats hs;
+    SchemaParams sch;
+};
+
+// row/marker body size + flag bytes (mirrors the oracle serializers)
+__device__ inline uint32_t unf_body_size(const UnfCols& u, uint64_t o, const SerParams2& sp,
+                                         uint8_t* out_flags, uint8_t* out_cflags) {
+    uint8_t kind = u.rkind[o];
+    if (kind != BK_CLUSTERING) {
+        uint32_t body = uvint_size((uint64_t)(u.rdel_mfda[o] - sp.hs.min_ts)) +
+                        uvint_size(sext32(ldt_long(u.rdel_ldt[o]) - sp.hs.min_ldt));
+        if (bk_is_boundary(kind))
+            body += uvint_size((uint64_t)(u.start_mfda[o] - sp.hs.min_ts)) +
+                    uvint_size(sext32(ldt_long(u.start_ldt[o]) - sp.hs.min_ldt));
+        if (out_flags) *out_flags = 0x02;
+        if (out_cflags) *out_cflags = 0;
+        return body;
+    }
+    uint8_t f = u.flags[o];
+    uint32_t body = 0;
+    uint8_t rflags = 0;
+    bool live = f & PF_LIVE_TS;
+    bool exp_live = live && u.live_ttl[o] != NO_TTL;
+    if (live) { rflags |= 0x04; body += uvint_size((uint64_t)(u.live_ts[o] - sp.hs.min_ts)); }
+    if (exp_live) {
+        rflags |= 0x08;
+        body += uvint_size(sext32(u.live_ttl[o] - sp.hs.min_ttl));
+        body += uvint_size(sext32(u.live_let[o] - sp.hs.min_ldt));
+    }
+    if (f & PF_ROW_DEL) {
+        rflags |= 0x10;
+        body += uvint_size((uint64_t)(u.rdel_mfda[o] - sp.hs.min_ts));
+        body += uvint_size(sext32(ldt_long(u.rdel_ldt[o]) - sp.hs.min_ldt));
+    }
+    if (f & PF_HAS_CELL) rflags |= 0x20;
+    else body += uvint_size(1);
+    uint8_t cflags = 0;
+    if (f & PF_HAS_CELL) {
+        int64_t cts = u.cell_ts[o];
+        uint32_t cldt = u.cell_ldt[o];
+        int32_t cttl = u.cell_ttl[o];
+        uint32_t vlen = u.val_len[o];
+        bool has_value = vlen > 0 && (f & PF_CELL_VALUE);
+        bool deleted = cldt != LDT_NONE_U32 && cttl == NO_TTL;
+        bool expiring = cttl != NO_TTL;
+        bool use_row_ts = live && cts == u.live_ts[o];
+        bool use_row_ttl = expiring && exp_live && cttl == u.live_ttl[o] && ldt_long(cldt) == u.live_let[o];
+        if (!has_value) cflags |= 4;
+        if (deleted) cflags |= 1;
+        else if (expiring) cflags |= 2;
+        if (use_row_ts) cflags |= 8;
+        if (use_row_ttl) cflags |= 16;
+        body += 1;
+        if (!use_row_ts) body += uvint_size((uint64_t)(cts - sp.hs.min_ts));
+        if ((deleted || expiring) && !use_row_ttl) body += uvint_size(sext32(ldt_long(cldt) - sp.hs.min_ldt));
+        if (expiring && !use_row_ttl) body += uvint_size(sext32(cttl - sp.hs.min_ttl));
+        if (has_value) body += (sp.sch.col_fixed_len >= 0 ? 0 : uvint_size(vlen)) + vlen;
+    }
+    if (out_flags) *out_flags = rflags;
+    if (out_cflags) *out_cflags = cflags;
+    return body;
+}
+
+// serialized size of a ClusteringPrefix (full serializer: kind byte [+ u16
+// size for bounds] + values-without-size) — IndexInfo first/last names
+__device__ inline uint32_t prefix_full_size(uint8_t kind, const SchemaParams& sch) {
+    uint32_t s = 1;                       // kind byte
+    if (kind != BK_CLUSTERING) s += 2;    // u16 value count
+    if (sch.ck_width) s += 1 + sch.ck_width;  // header vint + fixed value
+    return s;
+}
+__device__ inline uint32_t dt_ser_size(int64_t m, uint32_t l) {
+    return (m == INT64_MIN && l == LDT_NONE_U32) ? 1 : 12;
+}
+
+// one partition: Data bytes (+ IndexInfo bookkeeping). EMIT=false computes
+// sizes only (thread, lane==0); EMIT=true writes bytes (wave; lane 0 meta,
+// all lanes value copies). Returns psize; outputs isize/nblocks/infos_size.
+template <bool EMIT>
+__device__ uint64_t part_walk(const OutParts& op, const UnfCols& out, uint64_t g,
+                              const SerParams2& sp, uint64_t data_off, uint64_t idx_off,
+                              uint64_t infos_size_hint, uint32_t nblocks_hint,
+                              uint8_t* out_data, uint8_t* out_index, int lane,
+                              uint64_t* isize_out, uint32_t* nblocks_out,
+                              uint64_t* infos_size_out) {
+    if (!op.keep[g]) {
+        if (isize_out) *isize_out = 0;
+        if (nblocks_out) *nblocks_out = 0;
+        if (infos_size_out) *infos_size_out = 0;
+        return 0;
+    }
+    uint32_t klen = op.klen[g];
+    uint64_t kp = op.keypfx[g];
+    bool pdel_live = op.pdel_mfda[g] == INT64_MIN && op.pdel_ldt[g] == LDT_NONE_U32;
+    uint64_t header_len = 2 + klen + (pdel_live ? 1 : 12);
+    uint64_t rb = op.row_base[g];
+    uint32_t nrows = op.row_count[g];
+    uint8_t keyb[8];
+    for (uint32_t b = 0; b < klen; b++) keyb[b] = (uint8_t)(kp >> (8 * (7 - b)));
+
+    uint64_t pos = 0;
+    auto emit8 = [&](uint8_t v) {
+        if (EMIT && lane == 0) out_data[data_off + pos] = v;
+        pos++;
+    };
+    auto emit_uv = [&](uint64_t v) {
+        if (EMIT && lane == 0) {
+            uint8_t tmp[9];
+            int n = uvint_put(tmp, v);
+            for (int i = 0; i < n; i++) out_data[data_off + pos + i] = tmp[i];
+            pos += n;
+        } else pos += uvint_size(v);
+    };
+    auto emit_dt = [&](int64_t m, uint32_t l) {
+        if (m == INT64_MIN && l == LDT_NONE_U32) { emit8(0x80); return; }
+        if (EMIT && lane == 0) {
+            for (int i = 0; i < 8; i++) out_data[data_off + pos + i] = (uint8_t)((uint64_t)m >> (8 * (7 - i)));
+            for (int i = 0; i < 4; i++) out_data[data_off + pos + 8 + i] = (uint8_t)(l >> (8 * (3 - i)));
+        }
+        pos += 12;
+    };
+    auto emit_ck = [&](uint64_t ck) {
+        if (sp.sch.ck_width == 0) return;
+        if (EMIT && lane == 0) {
+            out_data[data_off + pos] = 0;  // 32-batch header: one non-null value
+            ck_bytes(ck, sp.sch.ck_width, &out_data[data_off + pos + 1]);
+        }
+        pos += 1 + sp.sch.ck_width;
+    };
+
+    // ---- index-entry layout (EMIT uses the SIZE pass results) ----
+    uint64_t fields_size = uvint_size(header_len) + dt_ser_size(op.pdel_mfda[g], op.pdel_ldt[g]) +
+                           uvint_size(nblocks_hint);
+    uint64_t S_hint = fields_size + infos_size_hint + (uint64_t)nblocks_hint * 4;
+    uint64_t infos_begin = 0, offsets_begin = 0;
+    if (EMIT && nblocks_hint > 1) {
+        infos_begin = idx_off + 2 + klen + uvint_size(data_off) + uvint_size(S_hint) + fields_size;
+        offsets_begin = infos_begin + infos_size_hint;
+    }
+
+    // ---- partition header ----
+    emit8((uint8_t)(klen >> 8));
+    emit8((uint8_t)klen);
+    for (uint32_t b = 0; b < klen; b++) emit8(keyb[b]);
+    emit_dt(op.pdel_mfda[g], op.pdel_ldt[g]);
+
+    // ---- walk unfiltereds ----
+    uint64_t prev_start = 0;
+    bool block_open = false;
+    uint64_t block_start = 0, block_first_ck = 0;
+    uint8_t block_first_kind = 0;
+    uint64_t last_ck = 0;
+    uint8_t last_kind = 0;
+    int64_t open_m = INT64_MIN;
+    uint32_t open_l = LDT_NONE_U32;
+    uint32_t nblocks = 0;
+    uint64_t infos_size = 0;
+    uint64_t info_cursor = infos_begin;
+
+    auto emit_prefix_idx = [&](uint8_t kk, uint64_t ck, uint64_t q) -> uint64_t {
+        if (EMIT && lane == 0) out_index[q] = kk;
+        q++;
+        if (kk != BK_CLUSTERING) {
+            if (EMIT && lane == 0) { out_index[q] = 0; out_index[q + 1] = (uint8_t)(sp.sch.ck_width ? 1 : 0); }
+            q += 2;
+        }
+        if (sp.sch.ck_width) {
+            if (EMIT && lane == 0) {
+                out_index[q] = 0;
+                ck_bytes(ck, sp.sch.ck_width, &out_index[q + 1]);
+            }
+            q += 1 + sp.sch.ck_width;
+        }
+        return q;
+    };
+    auto flush_block = [&](uint64_t end_pos) {
+        uint64_t width = end_pos - block_start;
+        bool has_open = !(open_m == INT64_MIN && open_l == LDT_NONE_U32);
+        uint64_t isz = prefix_full_size(block_first_kind, sp.sch) + prefix_full_size(last_kind, sp.sch) +
+                       uvint_size(block_start) + uvint_size(zigzag((int64_t)width - 65536)) + 1 +
+                       (has_open ? dt_ser_size(open_m, open_l) : 0);
+        if (EMIT && nblocks_hint > 1) {
+            if (lane == 0) {
+                uint64_t rel = info_cursor - infos_begin;
+                uint64_t slot = offsets_begin + (uint64_t)nblocks * 4;
+                out_index[slot] = (uint8_t)(rel >> 24);
+                out_index[slot + 1] = (uint8_t)(rel >> 16);
+                out_index[slot + 2] = (uint8_t)(rel >> 8);
+                out_index[slot + 3] = (uint8_t)rel;
+            }
+            uint64_t q = emit_prefix_idx(block_first_kind, block_first_ck, info_cursor);
+            q = emit_prefix_idx(last_kind, last_ck, q);
+            if (lane == 0) {
+                uint8_t tmp[10];
+                int n = uvint_put(tmp, block_start);
+                for (int i = 0; i < n; i++) out_index[q + i] = tmp[i];
+            }
+            q += uvint_size(block_start);
+            if (lane == 0) {
+                uint8_t tmp[10];
+                int n = uvint_put(tmp, zigzag((int64_t)width - 65536));
+                for (int i = 0; i < n; i++) out_index[q + i] = tmp[i];
+            }
+            q += uvint_size(zigzag((int64_t)width - 65536));
+            if (lane == 0) out_index[q] = has_open ? 1 : 0;
+            q += 1;
+            if (has_open) {
+                if (lane == 0) {
+                    for (int i = 0; i < 8; i++) out_index[q + i] = (uint8_t)((uint64_t)open_m >> (8 * (7 - i)));
+                    for (int i = 0; i < 4; i++) out_index[q + 8 + i] = (uint8_t)(open_l >> (8 * (3 - i)));
+                }
+                q += 12;
+            }
+            info_cursor = q;
+        }
+        infos_size += isz;
+        nblocks++;
+        block_open = false;
+    };
+
+    for (uint32_t j = 0; j < nrows; j++) {
+        uint64_t o = rb + j;
+        uint64_t upos = pos;
+        if (!block_open) {
+            block_open = true;
+            block_start = upos;
+            block_first_ck = out.ck[o];
+            block_first_kind = out.rkind[o];
+        }
+        uint8_t kind = out.rkind[o];
+        uint8_t rflags = 0, cflags = 0;
+        uint32_t body = unf_body_size(out, o, sp, &rflags, &cflags);
+        uint64_t prev_sz = j == 0 ? header_len : upos - prev_start;
+        if (kind != BK_CLUSTERING) {
+            emit8(0x02);
+            emit8(kind);
+            emit8(0);
+            emit8(sp.sch.ck_width ? 1 : 0);
+            emit_ck(out.ck[o]);
+            emit_uv(body + uvint_size(prev_sz));
+            emit_uv(prev_sz);
+            emit_uv((uint64_t)(out.rdel_mfda[o] - sp.hs.min_ts));
+            emit_uv(sext32(ldt_long(out.rdel_ldt[o]) - sp.hs.min_ldt));
+            if (bk_is_boundary(kind)) {
+                emit_uv((uint64_t)(out.start_mfda[o] - sp.hs.min_ts));
+                emit_uv(sext32(ldt_long(out.start_ldt[o]) - sp.hs.min_ldt));
+            }
+            open_m = bk_is_open(kind) ? (bk_is_boundary(kind) ? out.start_mfda[o] : out.rdel_mfda[o]) : INT64_MIN;
+            open_l = bk_is_open(kind) ? (bk_is_boundary(kind) ? out.start_ldt[o] : out.rdel_ldt[o]) : LDT_NONE_U32;
+        } else {
+            emit8(rflags);
+            emit_ck(out.ck[o]);
+            emit_uv(body + uvint_size(prev_sz));
+            emit_uv(prev_sz);
+            uint8_t f = out.flags[o];
+            if (rflags & 0x04) emit_uv((uint64_t)(out.live_ts[o] - sp.hs.min_ts));
+            if (rflags & 0x08) {
+                emit_uv(sext32(out.live_ttl[o] - sp.hs.min_ttl));
+                emit_uv(sext32(out.live_let[o] - sp.hs.min_ldt));
+            }
+            if (rflags & 0x10) {
+                emit_uv((uint64_t)(out.rdel_mfda[o] - sp.hs.min_ts));
+                emit_uv(sext32(ldt_long(out.rdel_ldt[o]) - sp.hs.min_ldt));
+            }
+            if (!(rflags & 0x20)) emit_uv(1);
+            if (f & PF_HAS_CELL) {
+                emit8(cflags);
+                if (!(cflags & 8)) emit_uv((uint64_t)(out.cell_ts[o] - sp.hs.min_ts));
+                bool deleted = cflags & 1, expiring = cflags & 2;
+                if ((deleted || expiring) && !(cflags & 16))
+                    emit_uv(sext32(ldt_long(out.cell_ldt[o]) - sp.hs.min_ldt));
+                if (expiring && !(cflags & 16)) emit_uv(sext32(out.cell_ttl[o] - sp.hs.min_ttl));
+                if (!(cflags & 4)) {
+                    uint32_t vlen = out.val_len[o];
+                    if (sp.sch.col_fixed_len < 0) emit_uv(vlen);
+                    if (EMIT) {
+                        const uint8_t* src = (const uint8_t*)out.val_addr[o];
+                        for (uint32_t i = (uint32_t)lane; i < vlen; i += WAVE)
+                            out_data[data_off + pos + i] = src[i];
+                    }
+                    pos += vlen;
+                }
+            }
+        }
+        prev_start = upos;
+        last_ck = out.ck[o];
+        last_kind = kind;
+        if (pos - block_start >= sp.sch.column_index_size) flush_block(pos);
+    }
+    emit8(0x01);  // END_OF_PARTITION
+    // the final block's width includes the end byte (addIndexBlock runs after
+    // writeEndOfPartition in BigFormatPartitionWriter.finish)
+    if (nrows > 0 && block_open) flush_block(pos);
+
+    // ---- index entry ----
+    uint64_t entry = 2 + klen + uvint_size(data_off);
+    if (nblocks > 1) {
+        uint64_t S = fields_size + infos_size + (uint64_t)nblocks * 4;
+        entry += uvint_size(S) + S;
+        if (EMIT) {
+            uint64_t q = idx_off;
+            if (lane == 0) {
+                out_index[q] = (uint8_t)(klen >> 8);
+                out_index[q + 1] = (uint8_t)klen;
+                for (uint32_t b = 0; b < klen; b++) out_index[q + 2 + b] = keyb[b];
+            }
+            q += 2 + klen;
+            uint8_t tmp[10];
+            if (lane == 0) {
+                int n = uvint_put(tmp, data_off);
+                for (int i = 0; i < n; i++) out_index[q + i] = tmp[i];
+            }
+            q += uvint_size(data_off);
+            if (lane == 0) {
+                int n = uvint_put(tmp, S);
+                for (int i = 0; i < n; i++) out_index[q + i] = tmp[i];
+            }
+            q += uvint_size(S);
+            if (lane == 0) {
+                int n = uvint_put(tmp, header_len);
+                for (int i = 0; i < n; i++) out_index[q + i] = tmp[i];
+            }
+            q += uvint_size(header_len);
+            if (pdel_live) {
+                if (lane == 0) out_index[q] = 0x80;
+                q += 1;
+            } else {
+                if (lane == 0) {
+                    for (int i = 0; i < 8; i++) out_index[q + i] = (uint8_t)((uint64_t)op.pdel_mfda[g] >> (8 * (7 - i)));
+                    for (int i = 0; i < 4; i++) out_index[q + 8 + i] = (uint8_t)(op.pdel_ldt[g] >> (8 * (3 - i)));
+                }
+                q += 12;
+            }
+            if (lane == 0) {
+                int n = uvint_put(tmp, nblocks);
+                for (int i = 0; i < n; i++) out_index[q + i] = tmp[i];
+            }
+            q += uvint_size(nblocks);
+            // infos + offsets were written during the walk at infos_begin/offsets_begin
+        }
+    } else {
+        entry += 1;  // vint promoted-size 0 (RowIndexEntry.java:468-473)
+        if (EMIT && lane == 0) {
+            uint64_t q = idx_off;
+            out_index[q] = (uint8_t)(klen >> 8);
+            out_index[q + 1] = (uint8_t)klen;
+            for (uint32_t b = 0; b < klen; b++) out_index[q + 2 + b] = keyb[b];
+            q += 2 + klen;
+            uint8_t tmp[10];
+            int n = uvint_put(tmp, data_off);
+            for (int i = 0; i < n; i++) out_index[q + i] = tmp[i];
+            q += n;
+            out_index[q] = 0;
+        }
+    }
+    if (isize_out) *isize_out = entry;
+    if (nblocks_out) *nblocks_out = nblocks;
+    if (infos_size_out) *infos_size_out = infos_size;
+    return pos;
+}
+
+// SIZE pass: psize/isize (+nblocks, infos_size) + output-stats histograms
+__global__ void k_sizes_rows(OutParts op, UnfCols out, uint64_t n, SerParams2 sp,
+                             uint64_t* psize, uint64_t* isize, uint32_t* nblocks,
+                             uint64_t* infos_size, OutStats* st,
+                             const int64_t* ps_hist_off, int32_t ps_hist_n,
+                             const int64_t* ch_hist_off, int32_t ch_hist_n) {
+    __shared__ unsigned int sh_ps[156], sh_ch[119];
+    for (int i = threadIdx.x; i < 156; i += blockDim.x) sh_ps[i] = 0;
+    for (int i = threadIdx.x; i < 119; i += blockDim.x) sh_ch[i] = 0;
+    __syncthreads();
+    uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (g < n) {
+        uint64_t is, inf;
+        uint32_t nb;
+        uint64_t s = part_walk<false>(op, out, g, sp, 0, 0, 0, 0, nullptr, nullptr, 0, &is, &nb, &inf);
+        psize[g] = s;
+        isize[g] = is;
+        nblocks[g] = nb;
+        infos_size[g] = inf;
+        if (s) {
+            int lo = 0, hi = ps_hist_n;
+            while (lo < hi) { int mid = (lo + hi) >> 1; if ((uint64_t)ps_hist_off[mid] < s) lo = mid + 1; else hi = mid; }
+            atomicAdd(&sh_ps[lo], 1u);
+            uint64_t cells = 0;
+            for (uint32_t j = 0; j < op.row_count[g]; j++)
+                if (out.flags[op.row_base[g] + j] & PF_HAS_CELL) cells++;
+            lo = 0; hi = ch_hist_n;
+            while (lo < hi) { int mid = (lo + hi) >> 1; if ((uint64_t)ch_hist_off[mid] < cells) lo = mid + 1; else hi = mid; }
+            atomicAdd(&sh_ch[lo], 1u);
+        }
+    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < 156; i += blockDim.x)
+        if (sh_ps[i]) atomicAdd(&st->part_size_hist[i], (unsigned long long)sh_ps[i]);
+    for (int i = threadIdx.x; i < 119; i += blockDim.x)
+        if (sh_ch[i]) atomicAdd(&st->cells_hist[i], (unsigned long long)sh_ch[i]);
+}
+
+// EMIT pass: wave per partition (Data + Index + bloom)
+__global__ void k_serialize_rows(OutParts op, UnfCols out, uint64_t n, SerParams2 sp,
+                                 const uint64_t* data_off, const uint64_t* idx_off,
+                                 const uint32_t* nblocks, const uint64_t* infos_size,
+                                 uint8_t* out_data, uint8_t* out_index,
+                                 uint32_t* bloom_bits, uint64_t bloom_bitlen, int32_t bloom_k) {
+    uint64_t g = blockIdx.x * (blockDim.x / WAVE) + (threadIdx.x / WAVE);
+    if (g >= n || !op.keep[g]) return;
+    int lane = threadIdx.x & (WAVE - 1);
+    part_walk<true>(op, out, g, sp, data_off[g], idx_off[g], infos_size[g], nblocks[g],
+                    out_data, out_index, lane, nullptr, nullptr, nullptr);
+    if (lane == 0) {
+        uint32_t klen = op.klen[g];
+        uint8_t keyb[8];
+        for (uint32_t b = 0; b < klen; b++) keyb[b] = (uint8_t)(op.keypfx[g] >> (8 * (7 - b)));
+        uint64_t h[2];
+        murmur3_128(keyb, klen, 0, h);
+        int64_t base = (int64_t)h[1], inc = (int64_t)h[0];
+        for (int i = 0; i < bloom_k; i++) {
+            int64_t m = base % (int64_t)bloom_bitlen;
+            uint64_t idx = (uint64_t)((m ^ (m >> 63)) - (m >> 63));
+            atomicOr(&bloom_bits[idx >> 5], 1u << (idx & 31));
+            base += inc;
+        }
+    }
+}
+
+// output stats over the final OutParts/UnfCols (MetadataCollector semantics)
+__global__ void k_collect_rows(OutParts op, UnfCols out, uint64_t n, OutStats* st,
+                               uint32_t* tomb_ldts, uint32_t tomb_cap) {
+    __shared__ unsigned long long sh_parts, sh_rows, sh_cells, sh_mints, sh_maxts,
+        sh_minldt, sh_maxldt, sh_haspdel;
+    __shared__ unsigned int sh_minttl, sh_maxttl;
+    if (threadIdx.x == 0) {
+        sh_parts = sh_rows = sh_cells = 0;
+        sh_mints = sh_minldt = 0xFFFFFFFFFFFFFFFFULL;
+        sh_maxts = sh_maxldt = 0;
+        sh_haspdel = 0;
+        sh_minttl = 0xFFFFFFFFu;
+        sh_maxttl = 0;
+    }
+    __syncthreads();
+    uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (g < n && op.keep[g]) {
+        auto lts = [&](int64_t ts) {
+            if (ts == NO_TIMESTAMP) return;
+            unsigned long long v = (unsigned long long)ts ^ 0x8000000000000000ULL;
+            atomicMin(&sh_mints, v);
+            atomicMax(&sh_maxts, v);
+        };
+        auto lldt = [&](int64_t l) {
+            unsigned long long v = (unsigned long long)l ^ 0x8000000000000000ULL;
+            atomicMin(&sh_minldt, v);
+            atomicMax(&sh_maxldt, v);
+        };
+        atomicAdd(&sh_parts, 1ull);
+        bool pdel_live = op.pdel_mfda[g] == INT64_MIN && op.pdel_ldt[g] == LDT_NONE_U32;
+        if (!pdel_live) {
+            atomicExch(&sh_haspdel, 1ull);
+            lts(op.pdel_mfda[g]);
+            lldt(ldt_long(op.pdel_ldt[g]));
+            tomb_push(st, tomb_ldts, tomb_cap, op.pdel_ldt[g]);
+        }
+        for (uint32_t j = 0; j < op.row_count[g]; j++) {
+            uint64_t o = op.row_base[g] + j;
+            uint8_t kind = out.rkind[o];
+            if (kind != BK_CLUSTERING) {
+                lts(out.rdel_mfda[o]);
+                lldt(ldt_long(out.rdel_ldt[o]));
+                tomb_push(st, tomb_ldts, tomb_cap, out.rdel_ldt[o]);
+                if (bk_is_boundary(kind)) {
+                    lts(out.start_mfda[o]);
+                    lldt(ldt_long(out.start_ldt[o]));
+                    tomb_push(st, tomb_ldts, tomb_cap, out.start_ldt[o]);
+                }
+                continue;
+            }
+            atomicAdd(&sh_rows, 1ull);
+            uint8_t of = out.flags[o];
+            if (of & PF_LIVE_TS) {
+                lts(out.live_ts[o]);
+                if (out.live_ttl[o] != 0) {
+                    lldt(out.live_let[o]);
+                    atomicMin(&sh_minttl, (unsigned)out.live_ttl[o]);
+                    atomicMax(&sh_maxttl, (unsigned)out.live_ttl[o]);
+                } else lldt(NO_DELETION_TIME);
+            }
+            if (of & PF_ROW_DEL) {
+                lts(out.rdel_mfda[o]);
+                lldt(ldt_long(out.rdel_ldt[o]));
+                tomb_push(st, tomb_ldts, tomb_cap, out.rdel_ldt[o]);
+            }
+            if (of & PF_HAS_CELL) {
+                atomicAdd(&sh_cells, 1ull);
+                lts(out.cell_ts[o]);
+                uint32_t cldt = out.cell_ldt[o];
+                int32_t cttl = out.cell_ttl[o];
+                if (cldt != LDT_NONE_U32 && cttl == 0) { lldt(ldt_long(cldt)); tomb_push(st, tomb_ldts, tomb_cap, cldt); }
+                else if (cttl != 0) {
+                    lldt(ldt_long(cldt));
+                    atomicMin(&sh_minttl, (unsigned)cttl);
+                    atomicMax(&sh_maxttl, (unsigned)cttl);
+                } else lldt(NO_DELETION_TIME);
+            }
+        }
+        atomicMin(&st->first_group, (unsigned long long)g);
+        atomicMax(&st->last_group, (unsigned long long)g);
+    }
+    __syncthreads();
+    if (threadIdx.x == 0 && sh_parts) {
+        atomicAdd(&st->partitions_out, sh_parts);
+        atomicAdd(&st->rows_out, sh_rows);
+        atomicAdd(&st->total_cells, sh_cells);
+        atomicMin(&st->min_ts_flip, sh_mints);
+        atomicMax(&st->max_ts_flip, sh_maxts);
+        atomicMin(&st->min_ldt_flip, sh_minldt);
+        atomicMax(&st->max_ldt_flip, sh_maxldt);
+        atomicMin(&st->min_ttl, sh_minttl);
+        atomicMax(&st->max_ttl, sh_maxttl);
+        if (sh_haspdel) atomicExch(&st->has_partition_deletions, 1ull);
+    }
+}
+
+// ---------------------------------------------------------------------------
+// generator (general): thread per partition fills rows (+ markers) mirroring
+// oracle/src/gen.h; values generated per live row into the values arena.
+// ---------------------------------------------------------------------------
+struct GenParams2 {
+    uint64_t seed, universe, stride, rows;
+    uint32_t sst;
+    uint32_t value_len, value_repeat_pct, tombstone_pct, partition_del_pct;
+    uint32_t clustering_rows, range_tomb_pct;
+    int64_t base_ts, base_ldt;
+};
+
+__global__ void k_gen_recs2(GenParams2 gp, MRec* recs, uint64_t* ids) {
+    uint64_t j = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (j >= gp.rows) return;
+    uint64_t id = feistel_perm(gp.seed, gp.universe, (gp.sst * gp.stride + j) % gp.universe);
+    ids[j] = id;
+    uint8_t key[8];
+    for (int b = 0; b < 8; b++) key[b] = (uint8_t)(id >> (8 * (7 - b)));
+    int64_t tok = murmur3_token(key, 8);
+    recs[j] = MRec{(uint64_t)tok ^ 0x8000000000000000ULL, id, (uint32_t)j, 0, 8, 0};
+}
+
+__device__ inline uint32_t gen2_ldt(const GenParams2& gp, uint64_t key_id, uint64_t salt) {
+    return (uint32_t)(gp.base_ldt + (int64_t)(splitmix64(key_id ^ salt) % 1000));
+}
+
+// per-partition unfiltered count (upper bound used for layout: exact)
+__global__ void k_gen_count(GenParams2 gp, const MRec* sorted, const uint64_t* ids, uint64_t n,
+                            OutParts op, uint64_t* prow_count) {
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    uint64_t id = ids[sorted[i].idx];
+    uint32_t cnt;
+    if (gp.clustering_rows == 0) {
+        bool pdel = gp.partition_del_pct && (splitmix64(gp.seed ^ 0xFEEDULL ^ id ^ ((uint64_t)gp.sst << 32)) % 100 < gp.partition_del_pct);
+        cnt = pdel ? 0 : 1;
+    } else {
+        cnt = gp.clustering_rows;
+        if (gp.range_tomb_pct && (splitmix64(gp.seed ^ 0xBEEFULL ^ id ^ ((uint64_t)gp.sst << 32)) % 100 < gp.range_tomb_pct))
+            cnt += 2;  // open + close bound
+    }
+    prow_count[i] = cnt;
+    op.keypfx[i] = sorted[i].pfx;
+    op.klen[i] = 8;
+    op.keep[i] = 1;
+}
+
+__global__ void k_gen_fill2(GenParams2 gp, const MRec* sorted, const uint64_t* ids, uint64_t n,
+                            OutParts op, UnfCols out, const uint64_t* row_base,
+                            uint8_t* values) {
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    uint64_t id = ids[sorted[i].idx];
+    uint64_t ob = row_base[i];
+    op.row_base[i] = ob;
+    uint32_t emitted = 0;
+    auto put_marker = [&](uint8_t kind, int64_t ck, int64_t m, uint32_t l) {
+        uint64_t o = ob + emitted++;
+        out.rkind[o] = kind;
+        out.ck[o] = (uint64_t)ck ^ 0x8000000000000000ULL;
+        out.flags[o] = 0;
+        out.live_ts[o] = NO_TIMESTAMP;
+        out.live_ttl[o] = 0;
+        out.live_let[o] = NO_DELETION_TIME;
+        out.rdel_mfda[o] = m;
+        out.rdel_ldt[o] = l;
+        out.start_mfda[o] = INT64_MIN;
+        out.start_ldt[o] = LDT_NONE_U32;
+        out.cell_ts[o] = NO_TIMESTAMP;
+        out.cell_ldt[o] = LDT_NONE_U32;
+        out.cell_ttl[o] = 0;
+        out.val_addr[o] = 0;
+        out.val_len[o] = 0;
+    };
+    auto put_row = [&](int64_t ck, bool has_ck, int64_t ts, bool tomb, uint32_t tomb_ldt,
+                       uint64_t val_slot) {
+        uint64_t o = ob + emitted++;
+        out.rkind[o] = BK_CLUSTERING;
+        out.ck[o] = has_ck ? ((uint64_t)ck ^ 0x8000000000000000ULL) : 0;
+        out.start_mfda[o] = INT64_MIN;
+        out.start_ldt[o] = LDT_NONE_U32;
+        out.live_ttl[o] = 0;
+        out.live_let[o] = NO_DELETION_TIME;
+        out.cell_ttl[o] = 0;
+        if (tomb) {
+            out.flags[o] = PF_HAS_ROW | PF_ROW_DEL;
+            out.live_ts[o] = NO_TIMESTAMP;
+            out.rdel_mfda[o] = ts;
+            out.rdel_ldt[o] = tomb_ldt;
+            out.cell_ts[o] = NO_TIMESTAMP;
+            out.cell_ldt[o] = LDT_NONE_U32;
+            out.val_addr[o] = 0;
+            out.val_len[o] = 0;
+        } else {
+            out.flags[o] = PF_HAS_ROW | PF_LIVE_TS | PF_HAS_CELL | PF_CELL_VALUE;
+            out.live_ts[o] = ts;
+            out.rdel_mfda[o] = INT64_MIN;
+            out.rdel_ldt[o] = LDT_NONE_U32;
+            out.cell_ts[o] = ts;
+            out.cell_ldt[o] = LDT_NONE_U32;
+            out.val_addr[o] = (uint64_t)(values + val_slot * (uint64_t)gp.value_len);
+            out.val_len[o] = gp.value_len;
+        }
+    };
+
+    if (gp.clustering_rows == 0) {
+        int64_t ts = gp.base_ts + (int64_t)(splitmix64(gp.seed ^ id * 31 ^ ((uint64_t)gp.sst << 48)) % 1000000000ULL);
+        bool pdel = gp.partition_del_pct && (splitmix64(gp.seed ^ 0xFEEDULL ^ id ^ ((uint64_t)gp.sst << 32)) % 100 < gp.partition_del_pct);
+        bool tomb = !pdel && gp.tombstone_pct && (splitmix64(gp.seed ^ 0xDEADULL ^ id ^ ((uint64_t)gp.sst << 32)) % 100 < gp.tombstone_pct);
+        if (pdel) {
+            op.pdel_mfda[i] = ts;
+            op.pdel_ldt[i] = gen2_ldt(gp, id, 0xDD);
+        } else {
+            op.pdel_mfda[i] = INT64_MIN;
+            op.pdel_ldt[i] = LDT_NONE_U32;
+            put_row(0, false, ts, tomb, tomb ? gen2_ldt(gp, id, 0xEE) : 0, ob + 0);
+        }
+    } else {
+        op.pdel_mfda[i] = INT64_MIN;
+        op.pdel_ldt[i] = LDT_NONE_U32;
+        bool has_rt = gp.range_tomb_pct && (splitmix64(gp.seed ^ 0xBEEFULL ^ id ^ ((uint64_t)gp.sst << 32)) % 100 < gp.range_tomb_pct);
+        int64_t rlo = 0, rhi = 0, rts = 0;
+        uint32_t rldt = 0;
+        if (has_rt) {
+            uint64_t r = splitmix64(gp.seed ^ 0xB00BULL ^ id ^ ((uint64_t)gp.sst << 32));
+            uint32_t a = (uint32_t)(r % gp.clustering_rows);
+            uint32_t b = a + 1 + (uint32_t)((r >> 32) % (gp.clustering_rows - a));
+            rlo = (int64_t)a * 16 + 8;
+            rhi = (int64_t)b * 16 + 8;
+            rts = gp.base_ts + (int64_t)(splitmix64(gp.seed ^ 0xAB1EULL ^ id ^ ((uint64_t)gp.sst << 48)) % 1000000000ULL);
+            rldt = gen2_ldt(gp, id, 0xCC);
+        }
+        bool rt_open = false;
+        for (uint32_t j = 0; j < gp.clustering_rows; j++) {
+            int64_t ck = (int64_t)j * 16;
+            if (has_rt && !rt_open && ck > rlo) {
+                put_marker(BK_INCL_START, rlo, rts, rldt);
+                rt_open = true;
+            }
+            int64_t ts = gp.base_ts + (int64_t)(splitmix64(gp.seed ^ id * 31 ^ ((uint64_t)gp.sst << 48) ^ (uint64_t)(j + 1) * 0x9E37ULL) % 1000000000ULL);
+            bool tomb = gp.tombstone_pct && (splitmix64(gp.seed ^ 0xDEADULL ^ id ^ ((uint64_t)gp.sst << 32) ^ (uint64_t)(j + 7) * 131) % 100 < gp.tombstone_pct);
+            put_row(ck, true, ts, tomb, tomb ? gen2_ldt(gp, id * 1000 + j, 0xEE) : 0, ob + j);
+            if (has_rt && rt_open && j + 1 < gp.clustering_rows && ((int64_t)(j + 1) * 16) > rhi) {
+                put_marker(BK_INCL_END, rhi, rts, rldt);
+                rt_open = false;
+            }
+        }
+        if (has_rt && rt_open) put_marker(BK_INCL_END, rhi, rts, rldt);
+    }
+    op.row_count[i] = emitted;
+}
+
+// values keyed like the oracle generator: simple schema seeds by key id; wide
+// partitions seed by (id ^ (j<<52) ^ j)
+__global__ void k_gen_values2(GenParams2 gp, const MRec* sorted, const uint64_t* ids,
+                              OutParts op, UnfCols out, uint64_t n, uint8_t* values,
+                              const uint64_t* row_base) {
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    uint64_t id = ids[sorted[i].idx];
+    uint64_t rb = op.row_base[i];
+    uint32_t cnt = op.row_count[i];
+    uint32_t rowj = 0;
+    for (uint32_t u = 0; u < cnt; u++) {
+        uint64_t o = rb + u;
+        if (out.rkind[o] != BK_CLUSTERING) continue;
+        uint32_t j = rowj++;
+        if (!(out.flags[o] & PF_CELL_VALUE)) continue;
+        uint64_t seed_id = gp.clustering_rows == 0 ? id : (id ^ ((uint64_t)j << 52) ^ j);
+        uint8_t* outp = (uint8_t*)out.val_addr[o];
+        uint64_t state = gp.seed ^ seed_id * 0x100000001B3ULL ^ ((uint64_t)gp.sst << 40);
+        uint64_t prev = splitmix64(state);
+        uint32_t nw = (gp.value_len + 7) / 8;
+        for (uint32_t w = 0; w < nw; w++) {
+            uint64_t r = splitmix64(state + 1 + w);
+            uint64_t word = (r % 100 < gp.value_repeat_pct && w > 0) ? prev : splitmix64(r);
+            prev = word;
+            uint32_t off = w * 8;
+            for (uint32_t b = 0; b < 8 && off + b < gp.value_len; b++)
+                outp[off + b] = (uint8_t)(word >> (8 * b));
+        }
+    }
+    (void)values;
+    (void)row_base;
+}
+
+}  // namespace gpuc

@@ -558,6 +558,7 @@ struct StatsComponentInput {
     std::map<uint32_t, uint32_t> tombstone_hist;  // ldt(seconds)->count, ≤100 bins
     double token_space_coverage = 0;
     size_t clustering_count = 0;
+    CqlType clustering_type = CqlType::LONG;
     bool has_partition_deletions = false;
 };
 static bytes serialize_stats_component(const StatsComponentInput& s) {
@@ -579,10 +580,10 @@ static bytes serialize_stats_component(const StatsComponentInput& s) {
     for (auto& [pt, cnt] : s.tombstone_hist) { put_be64(out, pt); put_be32(out, cnt); }
     put_be32(out, 0);               // sstableLevel
     put_be64(out, 0);               // repairedAt = UNREPAIRED_SSTABLE
-    // improvedMinMax (oa, no legacy): typeSerializer.serializeList + Slice
+    // improvedMinMax: typeSerializer.serializeList + Slice (spec'd with the GPU
+    // writer: type strings then BOTTOM..TOP bounds without values)
     put_unsigned_vint(out, s.clustering_count);
-    // (round 1: tables with clustering types in STATS only when clustering_count>0 — then
-    //  we'd need the covered slice; simple schema => 0 types, Slice = BOTTOM..TOP)
+    for (size_t i = 0; i < s.clustering_count; i++) put_type_str(out, s.clustering_type);
     out.push_back((uint8_t)INCL_START); put_be16(out, 0);  // start bound, 0 values
     out.push_back((uint8_t)INCL_END);   put_be16(out, 0);  // end bound, 0 values
     out.push_back(0);                                       // hasLegacyCounterShards=false
@@ -707,6 +708,7 @@ WriterOut write_sstable(const SSTable& t) {
     st.mins.total_rows = total_rows;
     st.mins.total_columns_set = total_cells;
     st.clustering_count = t.header.clustering_types.size();
+    if (st.clustering_count) st.clustering_type = t.header.clustering_types[0];
     if (!t.parts.empty()) { st.first_key = t.parts.front().key; st.last_key = t.parts.back().key; }
 
     ChunkedOut co = chunk_compress(data_raw, t.comp);

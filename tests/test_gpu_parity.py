@@ -62,6 +62,18 @@ def test_generator_parity(ca, oracle_bin, tmp_path):
         _assert_dirs_equal(f"{dg}/oa-{g}-big", f"{do}/oa-{g}-big")
 
 
+def test_generator_parity_wide(ca, oracle_bin, tmp_path):
+    """Wide partitions: clustering rows + range tombstones + promoted index."""
+    dg, do = str(tmp_path / "gpu"), str(tmp_path / "cpu")
+    os.makedirs(dg), os.makedirs(do)
+    kw = dict(seed=23, n=2, rows=30, crows=200, vlen=600, overlap=30, tomb=10, rtomb=40)
+    ca.generate(dg, seed=23, n_sstables=2, rows_per_sstable=30, clustering_rows=200,
+                value_len=600, overlap_pct=30, tombstone_pct=10, range_tomb_pct=40)
+    _oracle_gen(do, **kw)
+    for g in (1, 2):
+        _assert_dirs_equal(f"{dg}/oa-{g}-big", f"{do}/oa-{g}-big")
+
+
 @pytest.mark.parametrize("case", [
     dict(name="plain", gen=dict(seed=42, n=4, rows=3000, vlen=512, overlap=10), job={}),
     dict(name="tombstones_nogc", gen=dict(seed=17, n=4, rows=2500, vlen=256, overlap=30,
@@ -80,6 +92,16 @@ def test_generator_parity(ca, oracle_bin, tmp_path):
                                               vrep=97), job={}),
     dict(name="constant_values", gen=dict(seed=35, n=2, rows=1500, vlen=1000, overlap=0,
                                           vrep=100), job={}),
+    dict(name="wide_plain", gen=dict(seed=51, n=3, rows=80, crows=50, vlen=300, overlap=20),
+         job={}),
+    dict(name="wide_tombstones", gen=dict(seed=52, n=3, rows=60, crows=40, vlen=200,
+                                          overlap=30, tomb=15, rtomb=30), job={}),
+    dict(name="wide_gc", gen=dict(seed=52, n=3, rows=60, crows=40, vlen=200, overlap=30,
+                                  tomb=15, rtomb=30), job=dict(gc_before=2000000000)),
+    dict(name="wide_promoted_index", gen=dict(seed=53, n=2, rows=12, crows=300, vlen=800,
+                                              overlap=50, rtomb=40), job={}),
+    dict(name="wide_single_input", gen=dict(seed=54, n=1, rows=40, crows=60, vlen=250,
+                                            tomb=10, rtomb=25, overlap=0), job={}),
 ])
 def test_compaction_parity(ca, oracle_bin, tmp_path, case):
     d = str(tmp_path)
