@@ -1083,7 +1083,7 @@ __global__ void k_sizes_rows(OutParts op, UnfCols out, uint64_t n, SerParams2 sp
         uint32_t nb;
         uint64_t s = part_walk<false>(op, out, g, sp, 0, 0, 0, 0, nullptr, nullptr, 0, &is, &nb, &inf);
         psize[g] = s;
-        isize[g] = is;
+        isize[g] = 0;  // computed by k_index_sizes_rows once data offsets exist
         nblocks[g] = nb;
         infos_size[g] = inf;
         if (s) {
@@ -1103,6 +1103,29 @@ __global__ void k_sizes_rows(OutParts op, UnfCols out, uint64_t n, SerParams2 sp
         if (sh_ps[i]) atomicAdd(&st->part_size_hist[i], (unsigned long long)sh_ps[i]);
     for (int i = threadIdx.x; i < 119; i += blockDim.x)
         if (sh_ch[i]) atomicAdd(&st->cells_hist[i], (unsigned long long)sh_ch[i]);
+}
+
+// index entry sizes — needs the scanned Data offsets (the position vint's
+// width depends on the partition's absolute position)
+__global__ void k_index_sizes_rows(OutParts op, uint64_t n, const uint64_t* data_off,
+                                   const uint32_t* nblocks, const uint64_t* infos_size,
+                                   uint64_t* isize) {
+    uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (g >= n) return;
+    if (!op.keep[g]) { isize[g] = 0; return; }
+    uint32_t klen = op.klen[g];
+    bool pdel_live = op.pdel_mfda[g] == INT64_MIN && op.pdel_ldt[g] == LDT_NONE_U32;
+    uint64_t header_len = 2 + klen + (pdel_live ? 1 : 12);
+    uint64_t e = 2 + klen + uvint_size(data_off[g]);
+    uint32_t nb = nblocks[g];
+    if (nb > 1) {
+        uint64_t S = uvint_size(header_len) + dt_ser_size(op.pdel_mfda[g], op.pdel_ldt[g]) +
+                     uvint_size(nb) + infos_size[g] + (uint64_t)nb * 4;
+        e += uvint_size(S) + S;
+    } else {
+        e += 1;
+    }
+    isize[g] = e;
 }
 
 // EMIT pass: wave per partition (Data + Index + bloom)
