@@ -1290,8 +1290,11 @@ __global__ void k_gen_count(GenParams2 gp, const MRec* sorted, const uint64_t* i
         cnt = pdel ? 0 : 1;
     } else {
         cnt = gp.clustering_rows;
-        if (gp.range_tomb_pct && (splitmix64(gp.seed ^ 0xBEEFULL ^ id ^ ((uint64_t)gp.sst << 32)) % 100 < gp.range_tomb_pct))
-            cnt += 2;  // open + close bound
+        if (gp.range_tomb_pct && (splitmix64(gp.seed ^ 0xBEEFULL ^ id ^ ((uint64_t)gp.sst << 32)) % 100 < gp.range_tomb_pct)) {
+            uint64_t r = splitmix64(gp.seed ^ 0xB00BULL ^ id ^ ((uint64_t)gp.sst << 32));
+            uint32_t a = (uint32_t)(r % gp.clustering_rows);
+            if (a + 1 < gp.clustering_rows) cnt += 2;  // one open + one close bound
+        }
     }
     prow_count[i] = cnt;
     op.keypfx[i] = sorted[i].pfx;
@@ -1384,10 +1387,10 @@ __global__ void k_gen_fill2(GenParams2 gp, const MRec* sorted, const uint64_t* i
             rts = gp.base_ts + (int64_t)(splitmix64(gp.seed ^ 0xAB1EULL ^ id ^ ((uint64_t)gp.sst << 48)) % 1000000000ULL);
             rldt = gen2_ldt(gp, id, 0xCC);
         }
-        bool rt_open = false;
+        bool rt_open = false, rt_done = false;
         for (uint32_t j = 0; j < gp.clustering_rows; j++) {
             int64_t ck = (int64_t)j * 16;
-            if (has_rt && !rt_open && ck > rlo) {
+            if (has_rt && !rt_open && !rt_done && ck > rlo) {
                 put_marker(BK_INCL_START, rlo, rts, rldt);
                 rt_open = true;
             }
@@ -1397,6 +1400,7 @@ __global__ void k_gen_fill2(GenParams2 gp, const MRec* sorted, const uint64_t* i
             if (has_rt && rt_open && j + 1 < gp.clustering_rows && ((int64_t)(j + 1) * 16) > rhi) {
                 put_marker(BK_INCL_END, rhi, rts, rldt);
                 rt_open = false;
+                rt_done = true;
             }
         }
         if (has_rt && rt_open) put_marker(BK_INCL_END, rhi, rts, rldt);

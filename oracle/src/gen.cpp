@@ -48,10 +48,10 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
                 min_ts = std::min(min_ts, rts);
                 min_ldt_l = std::min<int64_t>(min_ldt_l, rldt);
             }
-            bool rt_open = false;
+            bool rt_open = false, rt_done = false;
             for (uint32_t j = 0; j < g.clustering_rows; j++) {
                 int64_t ck = gen_ck(g, e.id, j);
-                if (has_rt && !rt_open && ck > rlo) {
+                if (has_rt && !rt_open && !rt_done && ck > rlo) {
                     Unfiltered u;
                     u.kind = Unfiltered::MARKER;
                     u.marker.kind = INCL_START;
@@ -89,6 +89,7 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
                     m.marker.end_dt = DeletionTime{rts, rldt};
                     p.items.push_back(std::move(m));
                     rt_open = false;
+                    rt_done = true;
                 }
             }
             if (has_rt && rt_open) {
