@@ -78,6 +78,10 @@ def main():
                     help="rows per sstable (2,050,000 * ~1048 B = 2 GiB uncompressed)")
     ap.add_argument("--vlen", type=int, default=1024)
     ap.add_argument("--overlap", type=int, default=10)
+    ap.add_argument("--clustering-rows", type=int, default=0,
+                    help="C4 shape: rows per partition under a bigint clustering column")
+    ap.add_argument("--tombstone-pct", type=int, default=0)
+    ap.add_argument("--range-tomb-pct", type=int, default=0)
     ap.add_argument("--dir", default=os.environ.get("GPUC_BENCH_DIR", "/tmp/gpuc_bench"))
     ap.add_argument("--no-cpu-baseline", action="store_true")
     args = ap.parse_args()
@@ -106,7 +110,10 @@ def main():
     t_gen = time.time()
     ca.generate(d, seed=42 + rank, n_sstables=args.sstables,
                 rows_per_sstable=args.rows, overlap_pct=args.overlap,
-                value_len=args.vlen, value_repeat_pct=55, device=device)
+                value_len=args.vlen, value_repeat_pct=55, device=device,
+                clustering_rows=args.clustering_rows,
+                tombstone_pct=args.tombstone_pct,
+                range_tomb_pct=args.range_tomb_pct)
     t_gen = time.time() - t_gen
     bases = [os.path.join(d, f"oa-{g}-big") for g in range(1, args.sstables + 1)]
     input_compressed = sum(os.path.getsize(b + "-Data.db") for b in bases)
@@ -173,9 +180,14 @@ def main():
         shutil.rmtree(sd, ignore_errors=True)
         os.makedirs(sd)
         sample_rows = 120_000  # ~1 GiB in: ~10-20 s of single-core work
-        subprocess.run([ob, "gen", sd, "seed=42", f"n={args.sstables}",
-                        f"rows={sample_rows}", f"vlen={args.vlen}",
-                        f"overlap={args.overlap}"], check=True, capture_output=True)
+        if args.clustering_rows:
+            sample_rows = max(1, sample_rows // args.clustering_rows)
+        gen_args = [ob, "gen", sd, "seed=42", f"n={args.sstables}",
+                    f"rows={sample_rows}", f"vlen={args.vlen}", f"overlap={args.overlap}"]
+        if args.clustering_rows:
+            gen_args += [f"crows={args.clustering_rows}", f"tomb={args.tombstone_pct}",
+                         f"rtomb={args.range_tomb_pct}"]
+        subprocess.run(gen_args, check=True, capture_output=True)
         sins = [os.path.join(sd, f"oa-{g}-big") for g in range(1, args.sstables + 1)]
         out = subprocess.run([ob, "compact", os.path.join(sd, "oa-100-big"), *sins],
                              check=True, capture_output=True, text=True)
@@ -203,11 +215,18 @@ def main():
         "dtype": "u8",
         "data": "synthetic",
         "config": {
-            "workload": "C2: 8x2GiB sstables, 10% key overlap, ~1KiB values, LZ4 16KiB chunks",
+            "workload": ("C4: 8x2GiB wide-partition sstables "
+                         f"({args.clustering_rows} clustering rows/partition, "
+                         f"{args.tombstone_pct}% row + {args.range_tomb_pct}% range tombstones)"
+                         if args.clustering_rows else
+                         "C2: 8x2GiB sstables, 10% key overlap, ~1KiB values, LZ4 16KiB chunks"),
             "sstables": args.sstables,
             "rows_per_sstable": args.rows,
             "value_len": args.vlen,
             "overlap_pct": args.overlap,
+            "clustering_rows": args.clustering_rows,
+            "tombstone_pct": args.tombstone_pct,
+            "range_tomb_pct": args.range_tomb_pct,
             "parallelism": f"token-independent shards x{world}, no collectives",
             "input_uncompressed_bytes_per_rank": last["input_uncompressed_bytes"],
             "gen_seconds": round(t_gen, 1),
