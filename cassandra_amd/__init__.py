@@ -186,9 +186,11 @@ def compact(
             "decompress": res.ms_decompress,
             "parse": res.ms_parse,
             "merge": res.ms_merge,
+            "reconcile": res.ms_reconcile,
             "serialize": res.ms_serialize,
             "compress": res.ms_compress,
             "d2h": res.ms_d2h,
+            "write_io": res.ms_write_io,
             "total": res.ms_total,
         },
         "dominant_kernel": res.dominant_kernel.decode(),

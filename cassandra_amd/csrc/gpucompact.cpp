@@ -829,6 +829,8 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
             digest = comb.combine(digest, crc32_update_bitwise(0, cb, 4), 4);
         }
 
+        struct timespec ts0, ts1;
+        clock_gettime(CLOCK_MONOTONIC, &ts0);
         write_file_parallel(out_base + "-Data.db", h_data, w.compressed_len, 6);
         write_file_parallel(out_base + "-Index.db", h_index, total_idx, 4);
         {
@@ -867,6 +869,8 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
             std::string toc = "Data.db\nStatistics.db\nDigest.crc32\nTOC.txt\nCompressionInfo.db\nFilter.db\nIndex.db\nSummary.db\n";
             write_file(out_base + "-TOC.txt", (const uint8_t*)toc.data(), toc.size());
         }
+        clock_gettime(CLOCK_MONOTONIC, &ts1);
+        w.ms_io = (ts1.tv_sec - ts0.tv_sec) * 1e3 + (ts1.tv_nsec - ts0.tv_nsec) / 1e6;
     }
     w.uncompressed_len = total_unc;
     float t01, t12, t23, t34;
@@ -929,14 +933,15 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
         std::vector<uint64_t> generations(k);
         std::vector<uint8_t*> comp_pin(k, nullptr);
         std::vector<size_t> comp_sz(k, 0);
+        std::vector<std::string> in_bases(k);
         for (int s = 0; s < k; s++) {
             std::string base = job->input_bases[s];
+            in_bases[s] = base;
             comp_sz[s] = file_size_of(base + "-Data.db");
             comp_pin[s] = (uint8_t*)g_pin_in[s].get(comp_sz[s]);
             if (!comp_pin[s]) throw std::runtime_error("pinned alloc failed");
             index_data[s] = read_file(base + "-Index.db");
             cinfos[s] = parse_compression_info(read_file(base + "-CompressionInfo.db"));
-            read_file_into(base + "-Data.db", comp_pin[s], comp_sz[s], 6);
             stats[s] = parse_statistics(read_file(base + "-Statistics.db"));
             if (stats[s].clustering_types.size() > 1)
                 throw std::runtime_error("at most one clustering column supported");
@@ -966,6 +971,17 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
             else throw std::runtime_error("unsupported clustering type " + ck_type_str);
         }
 
+        // Data.db reads for ALL inputs start in parallel (host threads into
+        // pinned arenas); Index.db parses run alongside; per-sstable H2D is
+        // issued on a copy stream as soon as that file's read completes, and
+        // its decompress launch follows on the compute stream — read / H2D /
+        // decompress form a 3-deep pipeline instead of three serial phases.
+        std::vector<std::thread> data_readers(k);
+        for (int s = 0; s < k; s++)
+            data_readers[s] = std::thread([&, s] {
+                read_file_into(in_bases[s] + "-Data.db", comp_pin[s], comp_sz[s], 3);
+            });
+
         std::vector<std::vector<uint64_t>> positions(k);
         std::vector<std::string> perr(k);
         {
@@ -978,23 +994,36 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
                 if (!perr[s].empty()) throw std::runtime_error("Index.db parse: " + perr[s]);
         }
         double t1 = wall();
-        res->ms_read_io = t1 - t0;
+        res->ms_read_io = t1 - t0;  // metadata + index parse; Data reads overlap H2D below
 
-        // ---- H2D + decompress ----
+        // ---- H2D + decompress (pipelined per sstable) ----
+        hipStream_t copy_stream;
+        HIP_CHECK(hipStreamCreate(&copy_stream));
         hipEvent_t e0, e1, e2, e3, e4;
         for (hipEvent_t* e : {&e0, &e1, &e2, &e3, &e4}) HIP_CHECK(hipEventCreate(e));
         HIP_CHECK(hipEventRecord(e0, stream));
-        std::vector<DevBuf> d_comp(k), d_data(k), d_pos(k);
-        std::vector<ChunkDesc> chunks;
+        std::vector<DevBuf> d_comp(k), d_data(k), d_pos(k), d_chunks_s(k);
+        DevBuf d_error;
+        d_error.alloc(8);
+        HIP_CHECK(hipMemsetAsync(d_error.p, 0, 8, stream));
+        std::vector<hipEvent_t> ev_h2d(k);
+        std::vector<std::vector<ChunkDesc>> chunks_all(k);  // kept alive past async copies
+        size_t n_chunks_total = 0;
+        double ms_read_data = 0;
         for (int s = 0; s < k; s++) {
+            double tr = wall();
+            data_readers[s].join();  // in-order wait; all reads run concurrently
+            ms_read_data = wall() - tr + ms_read_data;
             d_comp[s].alloc(comp_sz[s]);
             HIP_CHECK(hipMemcpyAsync(d_comp[s].p, comp_pin[s], comp_sz[s],
-                                     hipMemcpyHostToDevice, stream));
+                                     hipMemcpyHostToDevice, copy_stream));
             d_data[s].alloc(cinfos[s].data_len);
             d_pos[s].alloc(positions[s].size() * 8);
             HIP_CHECK(hipMemcpyAsync(d_pos[s].p, positions[s].data(), positions[s].size() * 8,
-                                     hipMemcpyHostToDevice, stream));
+                                     hipMemcpyHostToDevice, copy_stream));
             auto& ci = cinfos[s];
+            auto& chunks = chunks_all[s];
+            chunks.reserve(ci.offsets.size());
             for (size_t c = 0; c < ci.offsets.size(); c++) {
                 uint64_t off = ci.offsets[c];
                 uint64_t end = c + 1 < ci.offsets.size() ? ci.offsets[c + 1] : comp_sz[s];
@@ -1005,17 +1034,19 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
                 cd.out_len = (uint32_t)std::min<uint64_t>(CHUNK_LEN, ci.data_len - c * (uint64_t)CHUNK_LEN);
                 chunks.push_back(cd);
             }
+            n_chunks_total += chunks.size();
+            d_chunks_s[s].alloc(chunks.size() * sizeof(ChunkDesc));
+            HIP_CHECK(hipMemcpyAsync(d_chunks_s[s].p, chunks.data(), chunks.size() * sizeof(ChunkDesc),
+                                     hipMemcpyHostToDevice, copy_stream));
+            HIP_CHECK(hipEventCreate(&ev_h2d[s]));
+            HIP_CHECK(hipEventRecord(ev_h2d[s], copy_stream));
+            HIP_CHECK(hipStreamWaitEvent(stream, ev_h2d[s], 0));
+            hipLaunchKernelGGL(k_lz4_decompress_wave, dim3((uint32_t)chunks.size()), dim3(WAVE), 0,
+                               stream, d_chunks_s[s].as<ChunkDesc>(), (uint32_t)chunks.size(), 1,
+                               d_error.as<unsigned long long>(), (const uint32_t*)g_crc256);
         }
-        DevBuf d_chunks, d_error;
-        d_chunks.alloc(chunks.size() * sizeof(ChunkDesc));
-        HIP_CHECK(hipMemcpyAsync(d_chunks.p, chunks.data(), chunks.size() * sizeof(ChunkDesc),
-                                 hipMemcpyHostToDevice, stream));
-        d_error.alloc(8);
-        HIP_CHECK(hipMemsetAsync(d_error.p, 0, 8, stream));
-        HIP_CHECK(hipEventRecord(e1, stream));
-        hipLaunchKernelGGL(k_lz4_decompress_wave, dim3((uint32_t)chunks.size()), dim3(WAVE), 0,
-                           stream, d_chunks.as<ChunkDesc>(), (uint32_t)chunks.size(), 1,
-                           d_error.as<unsigned long long>(), (const uint32_t*)g_crc256);
+        res->ms_read_io += ms_read_data;  // residual read wait not hidden by the pipeline
+        HIP_CHECK(hipEventRecord(e1, stream));  // e0..e1: H2D+decompress pipeline (overlapped)
         HIP_CHECK(hipEventRecord(e2, stream));
 
         // ---- parse (pass A: count + partition meta; pass B: row decode) ----
@@ -1157,6 +1188,10 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
             pp.ov_hi = d_ov_hi.as<int64_t>();
             pp.ov_min_ts = d_ov_ts.as<int64_t>();
         }
+        hipEvent_t er0, er1;
+        HIP_CHECK(hipEventCreate(&er0));
+        HIP_CHECK(hipEventCreate(&er1));
+        HIP_CHECK(hipEventRecord(er0, stream));
         {
             uint32_t blocks = (uint32_t)((n_groups + 255) / 256);
             hipLaunchKernelGGL(k_reconcile_rows, dim3(blocks), dim3(256), 0, stream, d_sorted,
@@ -1165,6 +1200,7 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
                                d_group_rows.as<uint64_t>(), sch, pp, d_stats.as<OutStats>(),
                                d_error.as<unsigned long long>());
         }
+        HIP_CHECK(hipEventRecord(er1, stream));
         HIP_CHECK(hipEventRecord(e4, stream));
         HIP_CHECK(hipStreamSynchronize(stream));
         {
@@ -1204,21 +1240,24 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
             HIP_CHECK(hipMemcpy(&hst, d_stats.p, sizeof(OutStats), hipMemcpyDeviceToHost));
             for (int i = 0; i < 64; i++) res->merged_counts[i] = hst.merged_counts[i];
         }
-        float th2d, tdec, tparse, tmerge;
+        float th2d, tdec, tparse, tmerge, trec;
         HIP_CHECK(hipEventElapsedTime(&th2d, e0, e1));
         HIP_CHECK(hipEventElapsedTime(&tdec, e1, e2));
         HIP_CHECK(hipEventElapsedTime(&tparse, e2, e3));
         HIP_CHECK(hipEventElapsedTime(&tmerge, e3, e4));
-        res->ms_h2d = th2d;
-        res->ms_decompress = tdec;
+        HIP_CHECK(hipEventElapsedTime(&trec, er0, er1));
+        res->ms_h2d = 0;         // overlapped into the ingest pipeline (e0..e1)
+        res->ms_decompress = th2d + tdec;  // read/H2D/decompress pipeline, GPU side
         res->ms_parse = tparse;
-        res->ms_merge = tmerge;  // includes merge+group+reconcile
+        res->ms_reconcile = trec;
+        res->ms_merge = tmerge - trec;  // merge rounds + grouping
         res->ms_serialize = w.ms_sizes + w.ms_serialize;
         res->ms_compress = w.ms_compress;
         res->ms_d2h = w.ms_d2h;
+        res->ms_write_io = w.ms_io;
         // dominant kernel estimate
         struct { const char* n; double ms; } ks[] = {
-            {"k_lz4_decompress", tdec}, {"k_parse", tparse}, {"merge+reconcile", tmerge},
+            {"k_lz4_decompress", (double)th2d + tdec}, {"k_parse", tparse}, {"merge+reconcile", tmerge},
             {"k_serialize", w.ms_serialize}, {"k_lz4_compress", w.ms_compress}};
         for (auto& kk : ks)
             if (kk.ms > res->dominant_kernel_ms) {
@@ -1226,7 +1265,9 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
                 snprintf(res->dominant_kernel, sizeof(res->dominant_kernel), "%s", kk.n);
             }
         res->dominant_kernel_launches = 1;
-        for (hipEvent_t e : {e0, e1, e2, e3, e4}) (void)hipEventDestroy(e);
+        for (hipEvent_t e : {e0, e1, e2, e3, e4, er0, er1}) (void)hipEventDestroy(e);
+        for (int s = 0; s < k; s++) (void)hipEventDestroy(ev_h2d[s]);
+        HIP_CHECK(hipStreamDestroy(copy_stream));
         HIP_CHECK(hipStreamDestroy(stream));
         res->ms_total = wall() - t_start_all;
         return GPUC_OK;
