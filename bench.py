@@ -118,11 +118,20 @@ def main():
     bases = [os.path.join(d, f"oa-{g}-big") for g in range(1, args.sstables + 1)]
     input_compressed = sum(os.path.getsize(b + "-Data.db") for b in bases)
 
-    def one_step(i):
+    import threading
+    cleaners = []
+
+    def one_step(i, prev=[None]):
         out = os.path.join(d, f"out-{i}", "oa-100-big")
         os.makedirs(os.path.dirname(out), exist_ok=True)
         r = ca.compact(bases, out, device=device)
-        shutil.rmtree(os.path.dirname(out), ignore_errors=True)
+        # previous step's output is deleted in the background (bounded disk,
+        # no serial rmtree inside the measured path)
+        if prev[0]:
+            t = threading.Thread(target=shutil.rmtree, args=(prev[0],), kwargs={"ignore_errors": True})
+            t.start()
+            cleaners.append(t)
+        prev[0] = os.path.dirname(out)
         return r
 
     import torch
@@ -139,6 +148,8 @@ def main():
     if dist:
         dist.barrier()
     elapsed = time.time() - t0
+    for t in cleaners:
+        t.join()
 
     my_bytes = float(last["input_uncompressed_bytes"] * args.steps)
     if dist:

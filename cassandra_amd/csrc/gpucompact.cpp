@@ -19,6 +19,7 @@
 #include <stdexcept>
 #include <string>
 #include <thread>
+#include <future>
 #include <fcntl.h>
 #include <unistd.h>
 #include <vector>
@@ -741,45 +742,133 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
     }
     HIP_CHECK(hipEventRecord(ev2, stream));
 
-    // ---- compress ----
+    // ---- compress + gather + D2H + write: slab-pipelined ----
+    // Compress launches for all slabs are enqueued on `stream` back to back;
+    // a second stream drains finished slabs (sizes -> local offsets -> gather
+    // -> D2H into rotating pinned buffers) while later slabs still compress,
+    // and host writer threads pwrite each drained slab into Data.db at its
+    // final offset. Output I/O therefore rides under the compress kernel
+    // instead of following it.
     uint32_t n_chunks = (uint32_t)((total_unc + CHUNK_LEN - 1) / CHUNK_LEN);
-    DevBuf d_slots, d_csize64, d_csize, d_ccrc;
-    d_slots.alloc((uint64_t)n_chunks * LZ4_SLOT);
-    d_csize.alloc((uint64_t)n_chunks * 4);
-    d_ccrc.alloc((uint64_t)n_chunks * 4);
-    d_csize64.alloc((uint64_t)n_chunks * 8);
-    hipLaunchKernelGGL(k_lz4_compress_wave_t<false>, dim3(n_chunks), dim3(WAVE), 0, stream,
-                       d_out_data.as<uint8_t>(), total_unc, d_slots.as<uint8_t>(),
-                       d_csize.as<uint32_t>(), d_ccrc.as<uint32_t>(), n_chunks,
-                       (const uint32_t*)g_crc256);
-    {
-        std::vector<uint32_t> cs(n_chunks);
-        HIP_CHECK(hipStreamSynchronize(stream));
-        HIP_CHECK(hipMemcpy(cs.data(), d_csize.p, (uint64_t)n_chunks * 4, hipMemcpyDeviceToHost));
-        std::vector<uint64_t> off(n_chunks);
-        uint64_t acc = 0;
-        for (uint32_t i = 0; i < n_chunks; i++) { off[i] = acc; acc += cs[i]; }
-        HIP_CHECK(hipMemcpy(d_csize64.p, off.data(), (uint64_t)n_chunks * 8, hipMemcpyHostToDevice));
-        w.compressed_len = acc + (uint64_t)n_chunks * 4;
-        DevBuf d_final;
-        d_final.alloc(w.compressed_len);
-        hipLaunchKernelGGL(k_chunk_gather, dim3(n_chunks), dim3(WAVE), 0, stream,
-                           d_slots.as<uint8_t>(), d_csize.as<uint32_t>(), d_ccrc.as<uint32_t>(),
-                           d_csize64.as<uint64_t>(), d_final.as<uint8_t>(), n_chunks);
-        HIP_CHECK(hipEventRecord(ev3, stream));
+    const uint32_t SLAB = 32768;  // 512 MiB uncompressed per slab
+    uint32_t n_slabs = n_chunks ? (n_chunks + SLAB - 1) / SLAB : 0;
+    DevBuf d_slots, d_csize, d_ccrc;
+    d_slots.alloc((uint64_t)n_chunks * LZ4_SLOT + 16);
+    d_csize.alloc((uint64_t)n_chunks * 4 + 16);
+    d_ccrc.alloc((uint64_t)n_chunks * 4 + 16);
+    hipStream_t cstream;
+    HIP_CHECK(hipStreamCreate(&cstream));
 
-        // ---- D2H (pinned staging) + files ----
-        uint8_t* h_data = (uint8_t*)g_pin_out[0].get(w.compressed_len);
-        uint8_t* h_index = (uint8_t*)g_pin_out[1].get(total_idx ? total_idx : 1);
-        if (!h_data || !h_index) throw std::runtime_error("pinned out alloc failed");
+    // Index.db image is final right after serialize: drain it early on cstream
+    uint8_t* h_index = (uint8_t*)g_pin_out[1].get(total_idx ? total_idx : 1);
+    if (!h_index) throw std::runtime_error("pinned out alloc failed");
+    HIP_CHECK(hipStreamWaitEvent(cstream, ev2, 0));
+    HIP_CHECK(hipMemcpyAsync(h_index, d_out_index.p, total_idx, hipMemcpyDeviceToHost, cstream));
+
+    std::vector<hipEvent_t> ev_c(n_slabs);
+    for (uint32_t i = 0; i < n_slabs; i++) {
+        uint32_t cb = i * SLAB, m = std::min(SLAB, n_chunks - cb);
+        // all kernel arguments shift uniformly per chunk, so a slab launch is
+        // just base-offset pointers with a local chunk count
+        hipLaunchKernelGGL(k_lz4_compress_wave_t<false>, dim3(m), dim3(WAVE), 0, stream,
+                           d_out_data.as<uint8_t>() + (uint64_t)cb * CHUNK_LEN,
+                           total_unc - (uint64_t)cb * CHUNK_LEN,
+                           d_slots.as<uint8_t>() + (uint64_t)cb * LZ4_SLOT,
+                           d_csize.as<uint32_t>() + cb, d_ccrc.as<uint32_t>() + cb, m,
+                           (const uint32_t*)g_crc256);
+        HIP_CHECK(hipEventCreate(&ev_c[i]));
+        HIP_CHECK(hipEventRecord(ev_c[i], stream));
+    }
+    HIP_CHECK(hipEventRecord(ev3, stream));  // ev2..ev3: pure compress GPU time
+
+    std::string data_path = out_base + "-Data.db";
+    {   // create/truncate so writers can pwrite into it
+        FILE* f = fopen(data_path.c_str(), "wb");
+        if (!f) throw std::runtime_error("cannot create " + data_path);
+        fclose(f);
+    }
+    const uint64_t worst_slab = (uint64_t)SLAB * (LZ4_SLOT + 4);
+    const int NSLOTS = 3;
+    uint8_t* h_slab0 = (uint8_t*)g_pin_out[0].get(worst_slab * NSLOTS);
+    if (!h_slab0) throw std::runtime_error("pinned out alloc failed");
+    DevBuf d_gat[2], d_foff[2];
+    if (n_slabs) {
+        d_gat[0].alloc(worst_slab); d_gat[1].alloc(worst_slab);
+        d_foff[0].alloc((uint64_t)SLAB * 8); d_foff[1].alloc((uint64_t)SLAB * 8);
+    }
+    std::vector<uint32_t> cs(n_chunks);
+    std::vector<uint64_t> foff_h[2];
+    foff_h[0].resize(SLAB); foff_h[1].resize(SLAB);
+    std::future<void> wfut[NSLOTS];
+    double t_drain0 = 0, t_drain1 = 0;
+    {
+        struct timespec tsd; clock_gettime(CLOCK_MONOTONIC, &tsd);
+        t_drain0 = tsd.tv_sec * 1e3 + tsd.tv_nsec / 1e6;
+    }
+    uint64_t file_off = 0;
+    for (uint32_t i = 0; i < n_slabs; i++) {
+        uint32_t cb = i * SLAB, m = std::min(SLAB, n_chunks - cb);
+        HIP_CHECK(hipStreamWaitEvent(cstream, ev_c[i], 0));
+        HIP_CHECK(hipMemcpyAsync(cs.data() + cb, d_csize.as<uint32_t>() + cb, (uint64_t)m * 4,
+                                 hipMemcpyDeviceToHost, cstream));
+        HIP_CHECK(hipStreamSynchronize(cstream));
+        auto& fo = foff_h[i & 1];
+        uint64_t acc = 0;
+        for (uint32_t j = 0; j < m; j++) { fo[j] = acc; acc += cs[cb + j]; }
+        uint64_t slab_bytes = acc + (uint64_t)m * 4;
+        HIP_CHECK(hipMemcpyAsync(d_foff[i & 1].p, fo.data(), (uint64_t)m * 8,
+                                 hipMemcpyHostToDevice, cstream));
+        hipLaunchKernelGGL(k_chunk_gather, dim3(m), dim3(WAVE), 0, cstream,
+                           d_slots.as<uint8_t>() + (uint64_t)cb * LZ4_SLOT,
+                           d_csize.as<uint32_t>() + cb, d_ccrc.as<uint32_t>() + cb,
+                           d_foff[i & 1].as<uint64_t>(), d_gat[i & 1].as<uint8_t>(), m);
+        int slot = (int)(i % NSLOTS);
+        if (wfut[slot].valid()) wfut[slot].get();  // pinned buffer free again
+        uint8_t* hbuf = h_slab0 + (uint64_t)slot * worst_slab;
+        HIP_CHECK(hipMemcpyAsync(hbuf, d_gat[i & 1].p, slab_bytes, hipMemcpyDeviceToHost, cstream));
+        HIP_CHECK(hipStreamSynchronize(cstream));
+        uint64_t off0 = file_off;
+        wfut[slot] = std::async(std::launch::async, [=]() {
+            int nth = 4;
+            std::vector<std::thread> th;
+            size_t per = (slab_bytes + nth - 1) / nth;
+            for (int t = 0; t < nth; t++) {
+                size_t o = (size_t)t * per;
+                if (o >= slab_bytes) break;
+                size_t len = std::min<size_t>(per, slab_bytes - o);
+                th.emplace_back([=]() {
+                    int fd = open(data_path.c_str(), O_WRONLY);
+                    if (fd < 0) return;
+                    size_t done = 0;
+                    while (done < len) {
+                        ssize_t ww = pwrite(fd, hbuf + o + done, len - done, (off_t)(off0 + o + done));
+                        if (ww <= 0) break;
+                        done += (size_t)ww;
+                    }
+                    close(fd);
+                });
+            }
+            for (auto& x : th) x.join();
+        });
+        file_off += slab_bytes;
+    }
+    for (int sfin = 0; sfin < NSLOTS; sfin++)
+        if (wfut[sfin].valid()) wfut[sfin].get();
+    {
+        struct timespec tsd; clock_gettime(CLOCK_MONOTONIC, &tsd);
+        t_drain1 = tsd.tv_sec * 1e3 + tsd.tv_nsec / 1e6;
+    }
+    w.compressed_len = file_off;
+    HIP_CHECK(hipEventRecord(ev4, stream));
+
+    {
         std::vector<uint8_t> h_bloom(words * 8);
         std::vector<uint32_t> h_crc(n_chunks);
-        HIP_CHECK(hipMemcpyAsync(h_data, d_final.p, w.compressed_len, hipMemcpyDeviceToHost, stream));
-        HIP_CHECK(hipMemcpyAsync(h_index, d_out_index.p, total_idx, hipMemcpyDeviceToHost, stream));
         HIP_CHECK(hipMemcpy(h_bloom.data(), d_bloom.p, words * 8, hipMemcpyDeviceToHost));
-        HIP_CHECK(hipMemcpy(h_crc.data(), d_ccrc.p, (uint64_t)n_chunks * 4, hipMemcpyDeviceToHost));
-        HIP_CHECK(hipEventRecord(ev4, stream));
+        if (n_chunks)
+            HIP_CHECK(hipMemcpy(h_crc.data(), d_ccrc.p, (uint64_t)n_chunks * 4, hipMemcpyDeviceToHost));
         HIP_CHECK(hipStreamSynchronize(stream));
+        HIP_CHECK(hipStreamSynchronize(cstream));
 
         std::vector<uint32_t> tombs;
         if (hst.tomb_count) {
@@ -824,14 +913,13 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
         uint32_t digest = 0;
         for (uint32_t c = 0; c < n_chunks; c++) {
             digest = comb.combine(digest, h_crc[c], cs[c]);
-            uint8_t cb[4] = {(uint8_t)(h_crc[c] >> 24), (uint8_t)(h_crc[c] >> 16),
-                             (uint8_t)(h_crc[c] >> 8), (uint8_t)h_crc[c]};
-            digest = comb.combine(digest, crc32_update_bitwise(0, cb, 4), 4);
+            uint8_t cb2[4] = {(uint8_t)(h_crc[c] >> 24), (uint8_t)(h_crc[c] >> 16),
+                              (uint8_t)(h_crc[c] >> 8), (uint8_t)h_crc[c]};
+            digest = comb.combine(digest, crc32_update_bitwise(0, cb2, 4), 4);
         }
 
         struct timespec ts0, ts1;
         clock_gettime(CLOCK_MONOTONIC, &ts0);
-        write_file_parallel(out_base + "-Data.db", h_data, w.compressed_len, 6);
         write_file_parallel(out_base + "-Index.db", h_index, total_idx, 4);
         {
             bytes f;
@@ -870,18 +958,20 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
             write_file(out_base + "-TOC.txt", (const uint8_t*)toc.data(), toc.size());
         }
         clock_gettime(CLOCK_MONOTONIC, &ts1);
+        // residual (non-overlapped) component writes + the Data.db drain tail
         w.ms_io = (ts1.tv_sec - ts0.tv_sec) * 1e3 + (ts1.tv_nsec - ts0.tv_nsec) / 1e6;
+        w.ms_d2h = t_drain1 - t_drain0;  // slab drain wall (overlaps compress)
     }
+    for (uint32_t i = 0; i < n_slabs; i++) (void)hipEventDestroy(ev_c[i]);
+    HIP_CHECK(hipStreamDestroy(cstream));
     w.uncompressed_len = total_unc;
-    float t01, t12, t23, t34;
+    float t01, t12, t23;
     HIP_CHECK(hipEventElapsedTime(&t01, ev0, ev1));
     HIP_CHECK(hipEventElapsedTime(&t12, ev1, ev2));
     HIP_CHECK(hipEventElapsedTime(&t23, ev2, ev3));
-    HIP_CHECK(hipEventElapsedTime(&t34, ev3, ev4));
     w.ms_sizes = t01;
     w.ms_serialize = t12;
     w.ms_compress = t23;
-    w.ms_d2h = t34;
     for (auto e : {ev0, ev1, ev2, ev3, ev4}) (void)hipEventDestroy(e);
     return w;
 }
