@@ -86,6 +86,20 @@ struct PinnedSlot {
 };
 static PinnedSlot g_pin_in[64], g_pin_out[4];
 
+// GPUC_TRACE=1: wall-clock checkpoints on stderr (host-gap hunting)
+static bool g_trace = getenv("GPUC_TRACE") != nullptr;
+static double g_trace_t0 = 0;
+static double trace_wall() {
+    struct timespec ts; clock_gettime(CLOCK_MONOTONIC, &ts);
+    return ts.tv_sec * 1e3 + ts.tv_nsec / 1e6;
+}
+static void TR(const char* tag) {
+    if (!g_trace) return;
+    double t = trace_wall();
+    fprintf(stderr, "[gpuc %8.1f] %s\n", t - g_trace_t0, tag);
+}
+
+
 // threaded read of a whole file into (pinned) memory
 static size_t file_size_of(const std::string& path) {
     FILE* f = fopen(path.c_str(), "rb");
@@ -671,6 +685,7 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
                                            const std::vector<std::pair<bytes, std::string>>& regular_cols,
                                            hipStream_t stream) {
     WriteDeviceOut w;
+    TR("wsd: enter");
     static const std::vector<int64_t> ps_off_h = est_hist_offsets(155);
     static const std::vector<int64_t> ch_off_h = est_hist_offsets(118);
     DevBuf d_ps_off, d_ch_off;
@@ -717,6 +732,7 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
 
     OutStats hst;
     HIP_CHECK(hipStreamSynchronize(stream));
+    TR("wsd: sizes synced");
     HIP_CHECK(hipMemcpy(&hst, d_stats.p, sizeof(OutStats), hipMemcpyDeviceToHost));
     w.partitions = hst.partitions_out;
     w.rows = hst.rows_out;
@@ -780,6 +796,7 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
         HIP_CHECK(hipEventRecord(ev_c[i], stream));
     }
     HIP_CHECK(hipEventRecord(ev3, stream));  // ev2..ev3: pure compress GPU time
+    TR("wsd: compress issued");
 
     std::string data_path = out_base + "-Data.db";
     {   // create/truncate so writers can pwrite into it
@@ -859,6 +876,7 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
         t_drain1 = tsd.tv_sec * 1e3 + tsd.tv_nsec / 1e6;
     }
     w.compressed_len = file_off;
+    TR("wsd: drain done");
     HIP_CHECK(hipEventRecord(ev4, stream));
 
     {
@@ -958,6 +976,7 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
             write_file(out_base + "-TOC.txt", (const uint8_t*)toc.data(), toc.size());
         }
         clock_gettime(CLOCK_MONOTONIC, &ts1);
+        TR("wsd: components written");
         // residual (non-overlapped) component writes + the Data.db drain tail
         w.ms_io = (ts1.tv_sec - ts0.tv_sec) * 1e3 + (ts1.tv_nsec - ts0.tv_nsec) / 1e6;
         w.ms_d2h = t_drain1 - t_drain0;  // slab drain wall (overlaps compress)
@@ -1005,6 +1024,7 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
         return ts.tv_sec * 1e3 + ts.tv_nsec / 1e6;
     };
     t_start_all = wall();
+    if (g_trace) g_trace_t0 = trace_wall();
     try {
         int ndev = gpuc_device_count();
         if (ndev <= 0) { set_err(res->error, sizeof(res->error), "no HIP device (no CPU fallback)"); return GPUC_ERR_NO_GPU; }
@@ -1085,6 +1105,7 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
         }
         double t1 = wall();
         res->ms_read_io = t1 - t0;  // metadata + index parse; Data reads overlap H2D below
+        TR("meta+index parsed");
 
         // ---- H2D + decompress (pipelined per sstable) ----
         hipStream_t copy_stream;
@@ -1136,6 +1157,7 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
                                d_error.as<unsigned long long>(), (const uint32_t*)g_crc256);
         }
         res->ms_read_io += ms_read_data;  // residual read wait not hidden by the pipeline
+        TR("ingest issued");
         HIP_CHECK(hipEventRecord(e1, stream));  // e0..e1: H2D+decompress pipeline (overlapped)
         HIP_CHECK(hipEventRecord(e2, stream));
 
@@ -1194,6 +1216,7 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
         {
             unsigned long long err = 0;
             HIP_CHECK(hipStreamSynchronize(stream));
+            TR("parse synced");
             HIP_CHECK(hipMemcpy(&err, d_error.p, 8, hipMemcpyDeviceToHost));
             if (err) throw std::runtime_error("GPU decode/parse error code " + std::to_string(err));
             HIP_CHECK(hipMemcpy(&res->rows_in, d_rows_in.p, 8, hipMemcpyDeviceToHost));
@@ -1204,6 +1227,7 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
         runs.push_back(0);
         for (int s = 0; s < k; s++) runs.push_back(runs.back() + srcs[s].n_parts);
         MRec* d_sorted = merge_sorted_runs(d_recs_a.as<MRec>(), d_recs_b.as<MRec>(), runs, stream);
+        TR("merge issued");
 
         // ---- group heads + starts ----
         DevBuf d_head, d_gstart, d_ngroups;
@@ -1225,6 +1249,7 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
         uint64_t n_groups = 0;
         HIP_CHECK(hipStreamSynchronize(stream));
         HIP_CHECK(hipMemcpy(&n_groups, d_ngroups.p, 8, hipMemcpyDeviceToHost));
+        TR("groups known");
 
         // ---- reconcile + purge ----
         DevBuf d_srcbases, d_group_rows;
@@ -1242,6 +1267,7 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
                                d_srcbases.as<uint32_t>(), pc, d_group_rows.as<uint64_t>());
         }
         uint64_t total_out_rows = exscan_u64(d_group_rows.as<uint64_t>(), n_groups, stream);
+        TR("out rows scanned");
         OutPartsBuf opb;
         opb.alloc(n_groups);
         UnfColsBuf out_rows;
@@ -1292,6 +1318,7 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
         }
         HIP_CHECK(hipEventRecord(er1, stream));
         HIP_CHECK(hipEventRecord(e4, stream));
+        TR("reconcile issued");
         HIP_CHECK(hipStreamSynchronize(stream));
         {
             unsigned long long err = 0;
@@ -1321,6 +1348,7 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
         WriteDeviceOut w = write_sstable_device(opb, out_rows, n_groups, sp2, d_stats, d_tomb,
                                                 tomb_cap, job->output_base, stats[0].key_type,
                                                 ck_type_str, stats[0].regular_cols, stream);
+        TR("writer done");
         res->partitions_out = w.partitions;
         res->rows_out = w.rows;
         res->output_uncompressed_bytes = w.uncompressed_len;

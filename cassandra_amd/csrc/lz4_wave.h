@@ -127,14 +127,27 @@ __device__ inline int lz4_wave_compress(const uint8_t* __restrict__ s_chunk, int
             int k0 = 0;           // probe index of this window's lane 0
             int P0 = ip;          // its position
             bool found = false, aborted = false;
+            // preload this window's probe dwords; inside the loop the NEXT
+            // window's dwords are prefetched while the current one resolves
+            // (pure loads: the probe positions are closed-form, independent
+            // of the window outcome, so semantics are unchanged and the
+            // global-load latency hides under the LDS/ballot chain)
+            int p_l = P0 + lz4_adv_sum(k0, k0 + lane);
+            uint32_t v_l = (p_l >= 0 && p_l + 4 <= srcSize) ? lds_read32(s_chunk, (uint32_t)p_l) : 0;
             while (true) {
-                int p_l = P0 + lz4_adv_sum(k0, k0 + lane);
                 // scalar loop aborts probe m when q_m + adv_m > mflimitPlusOne
                 bool valid = (p_l + lz4_adv(k0 + lane)) <= mflimitPlusOne;
-                uint32_t v_l = 0;
-                if (p_l + 4 <= srcSize && p_l >= 0) v_l = lds_read32(s_chunk, (uint32_t)p_l);
                 uint32_t h_l = lz4m_hash(v_l);
                 uint16_t t_l = s_table[h_l];              // pre-window candidate
+                // speculative candidate dword for the (common) no-duplicate
+                // case: issue the load before the marker/ballot section.
+                // Loads only on `valid` lanes: an invalid lane's (in-window
+                // predecessor) position can lie past the chunk end — the
+                // scalar code never evaluates those candidates at all.
+                uint32_t spec_cand = valid ? lds_read32(s_chunk, (uint32_t)t_l) : 0;
+                // prefetch next window probes
+                int p_n = P0 + lz4_adv_sum(k0, k0 + WAVE + lane);
+                uint32_t v_n = (p_n >= 0 && p_n + 4 <= srcSize) ? lds_read32(s_chunk, (uint32_t)p_n) : 0;
                 // marker round: conflicting LDS writes retire highest-lane-last,
                 // so the read-back names each hash group's max lane. volatile
                 // stops the compiler from forwarding the lane's own store —
@@ -168,7 +181,9 @@ __device__ inline int lz4_wave_compress(const uint8_t* __restrict__ s_chunk, int
                 // is closed-form per lane (no divergent cross-lane read).
                 uint32_t cand_pos = pred >= 0 ? (uint32_t)(P0 + lz4_adv_sum(k0, k0 + pred))
                                               : (uint32_t)t_l;
-                bool m_l = valid && lds_read32(s_chunk, cand_pos) == v_l;
+                uint32_t cand_val = pred >= 0 ? (valid ? lds_read32(s_chunk, cand_pos) : 0)
+                                              : spec_cand;
+                bool m_l = valid && cand_val == v_l;
                 uint64_t abort_mask = wave_ballot(!valid);
                 int first_abort = abort_mask ? (int)__ffsll((long long)abort_mask) - 1 : WAVE;
                 uint64_t match_mask = wave_ballot(m_l);
@@ -196,6 +211,8 @@ __device__ inline int lz4_wave_compress(const uint8_t* __restrict__ s_chunk, int
                 if (found || aborted) break;
                 P0 += lz4_adv_sum(k0, k0 + WAVE);
                 k0 += WAVE;
+                p_l = p_n;
+                v_l = v_n;
             }
             if (aborted) { emit_last_literals(); return (int)op; }
         }
