@@ -882,11 +882,14 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
     {
         std::vector<uint8_t> h_bloom(words * 8);
         std::vector<uint32_t> h_crc(n_chunks);
+        TR("wsd: tail vectors");
         HIP_CHECK(hipMemcpy(h_bloom.data(), d_bloom.p, words * 8, hipMemcpyDeviceToHost));
+        TR("wsd: bloom d2h");
         if (n_chunks)
             HIP_CHECK(hipMemcpy(h_crc.data(), d_ccrc.p, (uint64_t)n_chunks * 4, hipMemcpyDeviceToHost));
         HIP_CHECK(hipStreamSynchronize(stream));
         HIP_CHECK(hipStreamSynchronize(cstream));
+        TR("wsd: tail synced");
 
         std::vector<uint32_t> tombs;
         if (hst.tomb_count) {
@@ -927,15 +930,40 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
             HIP_CHECK(hipMemcpy(&kl, opb.op.klen + lg, 1, hipMemcpyDeviceToHost));
             for (int b = 0; b < kl; b++) m.last_key.push_back((uint8_t)(kp >> (8 * (7 - b))));
         }
+        TR("wsd: meta built");
         static Crc32Combiner comb;
+        // Digest = CRC of the whole Data.db, folded from per-chunk CRCs.
+        // Thread-parallel: each worker folds a contiguous chunk range into
+        // (crc, len); the partials combine left-to-right (CRC concatenation
+        // is associative over (crc, len) pairs).
         uint32_t digest = 0;
-        for (uint32_t c = 0; c < n_chunks; c++) {
-            digest = comb.combine(digest, h_crc[c], cs[c]);
-            uint8_t cb2[4] = {(uint8_t)(h_crc[c] >> 24), (uint8_t)(h_crc[c] >> 16),
-                              (uint8_t)(h_crc[c] >> 8), (uint8_t)h_crc[c]};
-            digest = comb.combine(digest, crc32_update_bitwise(0, cb2, 4), 4);
+        {
+            int nth = n_chunks > 4096 ? 16 : 1;
+            std::vector<uint32_t> pcrc(nth, 0);
+            std::vector<uint64_t> plen(nth, 0);
+            uint32_t per = (n_chunks + nth - 1) / nth;
+            std::vector<std::thread> th;
+            for (int t = 0; t < nth; t++) {
+                th.emplace_back([&, t]() {
+                    uint32_t c0 = t * per, c1 = std::min(n_chunks, (t + 1) * per);
+                    uint32_t d = 0;
+                    uint64_t l = 0;
+                    for (uint32_t c = c0; c < c1; c++) {
+                        d = comb.combine(d, h_crc[c], cs[c]);
+                        uint8_t cb2[4] = {(uint8_t)(h_crc[c] >> 24), (uint8_t)(h_crc[c] >> 16),
+                                          (uint8_t)(h_crc[c] >> 8), (uint8_t)h_crc[c]};
+                        d = comb.combine(d, crc32_update_bitwise(0, cb2, 4), 4);
+                        l += cs[c] + 4;
+                    }
+                    pcrc[t] = d;
+                    plen[t] = l;
+                });
+            }
+            for (auto& x : th) x.join();
+            for (int t = 0; t < nth; t++) digest = comb.combine(digest, pcrc[t], plen[t]);
         }
 
+        TR("wsd: digest done");
         struct timespec ts0, ts1;
         clock_gettime(CLOCK_MONOTONIC, &ts0);
         write_file_parallel(out_base + "-Index.db", h_index, total_idx, 4);
@@ -1044,15 +1072,31 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
         std::vector<uint8_t*> comp_pin(k, nullptr);
         std::vector<size_t> comp_sz(k, 0);
         std::vector<std::string> in_bases(k);
+        // metadata components of all inputs read+parsed on one thread per
+        // sstable (Index.db alone is ~40 MB per input at C2 scale)
+        {
+            std::vector<std::thread> mth;
+            std::vector<std::string> merr(k);
+            for (int s = 0; s < k; s++) {
+                in_bases[s] = job->input_bases[s];
+                mth.emplace_back([&, s]() {
+                    try {
+                        const std::string& base = in_bases[s];
+                        index_data[s] = read_file(base + "-Index.db");
+                        cinfos[s] = parse_compression_info(read_file(base + "-CompressionInfo.db"));
+                        stats[s] = parse_statistics(read_file(base + "-Statistics.db"));
+                    } catch (const std::exception& e) { merr[s] = e.what(); }
+                });
+            }
+            for (auto& t : mth) t.join();
+            for (int s = 0; s < k; s++)
+                if (!merr[s].empty()) throw std::runtime_error(merr[s]);
+        }
         for (int s = 0; s < k; s++) {
-            std::string base = job->input_bases[s];
-            in_bases[s] = base;
+            std::string base = in_bases[s];
             comp_sz[s] = file_size_of(base + "-Data.db");
             comp_pin[s] = (uint8_t*)g_pin_in[s].get(comp_sz[s]);
             if (!comp_pin[s]) throw std::runtime_error("pinned alloc failed");
-            index_data[s] = read_file(base + "-Index.db");
-            cinfos[s] = parse_compression_info(read_file(base + "-CompressionInfo.db"));
-            stats[s] = parse_statistics(read_file(base + "-Statistics.db"));
             if (stats[s].clustering_types.size() > 1)
                 throw std::runtime_error("at most one clustering column supported");
             if (stats[s].regular_cols.size() != 1)
