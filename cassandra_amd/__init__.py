@@ -78,6 +78,7 @@ class GpucGenSpec(ctypes.Structure):
         ("partition_del_pct", ctypes.c_uint32),
         ("clustering_rows", ctypes.c_uint32),
         ("range_tomb_pct", ctypes.c_uint32),
+        ("key_len", ctypes.c_uint32),
         ("base_ts", ctypes.c_int64),
         ("base_ldt", ctypes.c_int64),
         ("first_generation", ctypes.c_uint64),
@@ -210,6 +211,7 @@ def generate(
     partition_del_pct=0,
     clustering_rows=0,
     range_tomb_pct=0,
+    key_len=8,
     base_ts=1700000000000000,
     base_ldt=1700000000,
     first_generation=1,
@@ -228,6 +230,7 @@ def generate(
         partition_del_pct=partition_del_pct,
         clustering_rows=clustering_rows,
         range_tomb_pct=range_tomb_pct,
+        key_len=key_len,
         base_ts=base_ts,
         base_ldt=base_ldt,
         first_generation=first_generation,

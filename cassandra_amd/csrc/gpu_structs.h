@@ -11,20 +11,50 @@ namespace gpuc {
 // zero-padded big-endian in pfx, ties broken by klen).
 struct MRec {
     uint64_t tok;   // token ^ 0x8000000000000000
-    uint64_t pfx;   // key bytes, big-endian, zero-padded
+    uint64_t pfx;   // first min(8,klen) key bytes, big-endian, zero-padded
     uint32_t idx;   // partition ordinal within source
     uint16_t src;
-    uint8_t klen;
-    uint8_t pad;
+    uint16_t klen;  // full key length (Cassandra keys are u16-length)
 };
 
-__host__ __device__ inline bool mrec_less(const MRec& a, const MRec& b) {
+// Key lookup for exact DecoratedKey comparison beyond the 8-byte prefix:
+// base[src] + pos[src][idx] + 2 is the address of partition idx's key bytes
+// in the decompressed input (the short-length field precedes them). The
+// compaction path enables it after parse; the generator path disables it
+// (generated keys have unique 8-byte prefixes, so prefix+klen is exact).
+struct KeyLut {
+    const uint8_t* base[64];
+    const uint64_t* pos[64];
+    int enabled;
+};
+__device__ inline KeyLut g_key_lut;  // zero-init: disabled
+
+__device__ inline const uint8_t* mrec_key_bytes(const MRec& r) {
+    return g_key_lut.base[r.src] + g_key_lut.pos[r.src][r.idx] + 2;
+}
+// DecoratedKey order (DecoratedKey.java:79-92): token, then key bytes
+// compared as unsigned lexicographic with shorter-is-less on prefix equality
+// (ByteBufferUtil.compareUnsigned). The zero-padded 8-byte prefix compare is
+// exact whenever it differs; ties fall back to a byte walk from offset 8.
+__device__ inline bool mrec_less(const MRec& a, const MRec& b) {
     if (a.tok != b.tok) return a.tok < b.tok;
     if (a.pfx != b.pfx) return a.pfx < b.pfx;
+    if (!g_key_lut.enabled || (a.klen <= 8 && b.klen <= 8)) return a.klen < b.klen;
+    const uint8_t* ka = mrec_key_bytes(a);
+    const uint8_t* kb = mrec_key_bytes(b);
+    uint32_t n = a.klen < b.klen ? a.klen : b.klen;
+    for (uint32_t i = 8; i < n; i++)
+        if (ka[i] != kb[i]) return ka[i] < kb[i];
     return a.klen < b.klen;
 }
-__host__ __device__ inline bool mrec_eq(const MRec& a, const MRec& b) {
-    return a.tok == b.tok && a.pfx == b.pfx && a.klen == b.klen;
+__device__ inline bool mrec_eq(const MRec& a, const MRec& b) {
+    if (a.tok != b.tok || a.pfx != b.pfx || a.klen != b.klen) return false;
+    if (!g_key_lut.enabled || a.klen <= 8) return true;
+    const uint8_t* ka = mrec_key_bytes(a);
+    const uint8_t* kb = mrec_key_bytes(b);
+    for (uint32_t i = 8; i < a.klen; i++)
+        if (ka[i] != kb[i]) return false;
+    return true;
 }
 
 // unfiltered-level SoA (rows AND range-tombstone markers), used for parsed
@@ -77,12 +107,14 @@ struct ParsedCols {
     uint32_t* pdel_ldt;
     uint32_t* row_count;  // unfiltereds in this partition
     uint64_t* row_base;   // start into the per-source-concatenated UnfCols
+    uint64_t* key_addr;   // device address of the key bytes in input data
 };
 
 // reconciled output partitions (also produced directly by the generator)
 struct OutParts {
-    uint64_t* keypfx;    // big-endian zero-padded key
-    uint8_t* klen;
+    uint64_t* keypfx;    // big-endian zero-padded key prefix (debug/aux)
+    uint64_t* key_addr;  // device address of the full key bytes
+    uint16_t* klen;
     int64_t* pdel_mfda;
     uint32_t* pdel_ldt;
     uint64_t* row_base;  // into the output UnfCols arena

@@ -635,12 +635,14 @@ static void ensure_crc_tables(hipStream_t stream) {
 
 // allocate the OutParts SoA for n entries
 struct OutPartsBuf {
-    DevBuf keypfx, klen, pdel_mfda, pdel_ldt, row_base, row_count, keep;
+    DevBuf keypfx, key_addr, klen, pdel_mfda, pdel_ldt, row_base, row_count, keep;
     OutParts op{};
     void alloc(uint64_t n) {
-        keypfx.alloc(n * 8); klen.alloc(n); pdel_mfda.alloc(n * 8); pdel_ldt.alloc(n * 4);
+        keypfx.alloc(n * 8); key_addr.alloc(n * 8); klen.alloc(n * 2);
+        pdel_mfda.alloc(n * 8); pdel_ldt.alloc(n * 4);
         row_base.alloc(n * 8); row_count.alloc(n * 4); keep.alloc(n);
-        op = OutParts{keypfx.as<uint64_t>(), klen.as<uint8_t>(), pdel_mfda.as<int64_t>(),
+        op = OutParts{keypfx.as<uint64_t>(), key_addr.as<uint64_t>(), klen.as<uint16_t>(),
+                      pdel_mfda.as<int64_t>(),
                       pdel_ldt.as<uint32_t>(), row_base.as<uint64_t>(), row_count.as<uint32_t>(),
                       keep.as<uint8_t>()};
     }
@@ -921,14 +923,16 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
         m.compression_ratio = total_unc ? (double)(w.compressed_len - (uint64_t)n_chunks * 4) / (double)total_unc : -1.0;
         if (w.partitions) {
             uint64_t fg = hst.first_group, lg = hst.last_group;
-            uint64_t kp;
-            uint8_t kl;
-            HIP_CHECK(hipMemcpy(&kp, opb.op.keypfx + fg, 8, hipMemcpyDeviceToHost));
-            HIP_CHECK(hipMemcpy(&kl, opb.op.klen + fg, 1, hipMemcpyDeviceToHost));
-            for (int b = 0; b < kl; b++) m.first_key.push_back((uint8_t)(kp >> (8 * (7 - b))));
-            HIP_CHECK(hipMemcpy(&kp, opb.op.keypfx + lg, 8, hipMemcpyDeviceToHost));
-            HIP_CHECK(hipMemcpy(&kl, opb.op.klen + lg, 1, hipMemcpyDeviceToHost));
-            for (int b = 0; b < kl; b++) m.last_key.push_back((uint8_t)(kp >> (8 * (7 - b))));
+            for (int which = 0; which < 2; which++) {
+                uint64_t g = which ? lg : fg;
+                uint64_t ka;
+                uint16_t kl;
+                HIP_CHECK(hipMemcpy(&ka, opb.op.key_addr + g, 8, hipMemcpyDeviceToHost));
+                HIP_CHECK(hipMemcpy(&kl, opb.op.klen + g, 2, hipMemcpyDeviceToHost));
+                bytes kb(kl);
+                HIP_CHECK(hipMemcpy(kb.data(), (const void*)ka, kl, hipMemcpyDeviceToHost));
+                (which ? m.last_key : m.first_key) = kb;
+            }
         }
         TR("wsd: meta built");
         static Crc32Combiner comb;
@@ -1238,6 +1242,8 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
         p_pdl.alloc(total_parts * 4); pc.pdel_ldt = p_pdl.as<uint32_t>();
         p_rcnt.alloc(total_parts * 4); pc.row_count = p_rcnt.as<uint32_t>();
         p_rbase.alloc(total_parts * 8); pc.row_base = p_rbase.as<uint64_t>();
+        DevBuf p_kaddr;
+        p_kaddr.alloc(total_parts * 8); pc.key_addr = p_kaddr.as<uint64_t>();
         {
             uint32_t blocks = (uint32_t)((total_parts + 255) / 256);
             hipLaunchKernelGGL(k_parse_count, dim3(blocks), dim3(256), 0, stream,
@@ -1264,6 +1270,18 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
             HIP_CHECK(hipMemcpy(&err, d_error.p, 8, hipMemcpyDeviceToHost));
             if (err) throw std::runtime_error("GPU decode/parse error code " + std::to_string(err));
             HIP_CHECK(hipMemcpy(&res->rows_in, d_rows_in.p, 8, hipMemcpyDeviceToHost));
+        }
+
+        // exact-key comparator fallback: point the merge kernels at the
+        // decompressed inputs (keys > 8 bytes tie-break by byte walk)
+        {
+            KeyLut lut{};
+            for (int s2 = 0; s2 < k; s2++) {
+                lut.base[s2] = d_data[s2].as<uint8_t>();
+                lut.pos[s2] = d_pos[s2].as<uint64_t>();
+            }
+            lut.enabled = 1;
+            HIP_CHECK(hipMemcpyToSymbol(HIP_SYMBOL(g_key_lut), &lut, sizeof(lut)));
         }
 
         // ---- merge (pairwise rounds over pre-sorted source runs) ----
@@ -1465,19 +1483,28 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
             gp.range_tomb_pct = spec->range_tomb_pct;
             gp.base_ts = spec->base_ts;
             gp.base_ldt = spec->base_ldt;
-            DevBuf d_a, d_b, d_ids, d_vals, d_stats, d_tomb, d_prows;
+            gp.key_len = spec->key_len ? spec->key_len : 8;
+            if (gp.key_len < 8 || gp.key_len > 255)
+                throw std::runtime_error("key_len must be 8..255 (generator contract)");
+            {
+                KeyLut lut{};  // generated 8-byte prefixes are unique: prefix compare is exact
+                HIP_CHECK(hipMemcpyToSymbol(HIP_SYMBOL(g_key_lut), &lut, sizeof(lut)));
+            }
+            DevBuf d_a, d_b, d_ids, d_keys, d_vals, d_stats, d_tomb, d_prows;
             d_a.alloc(R * sizeof(MRec));
             d_b.alloc(R * sizeof(MRec));
             d_ids.alloc(R * 8);
+            d_keys.alloc(R * (uint64_t)gp.key_len);
             uint32_t blocks = (uint32_t)((R + 255) / 256);
             hipLaunchKernelGGL(k_gen_recs2, dim3(blocks), dim3(256), 0, stream, gp,
-                               d_a.as<MRec>(), d_ids.as<uint64_t>());
+                               d_a.as<MRec>(), d_ids.as<uint64_t>(), d_keys.as<uint8_t>());
             MRec* d_sorted = merge_sort_recs(d_a.as<MRec>(), d_b.as<MRec>(), R, stream);
             OutPartsBuf opb;
             opb.alloc(R);
             d_prows.alloc(R * 8);
             hipLaunchKernelGGL(k_gen_count, dim3(blocks), dim3(256), 0, stream, gp, d_sorted,
-                               d_ids.as<uint64_t>(), R, opb.op, d_prows.as<uint64_t>());
+                               d_ids.as<uint64_t>(), R, opb.op, d_prows.as<uint64_t>(),
+                               d_keys.as<uint8_t>());
             uint64_t total_rows = exscan_u64(d_prows.as<uint64_t>(), R, stream);
             UnfColsBuf rows;
             rows.alloc(total_rows);
@@ -1516,7 +1543,8 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
             std::vector<std::pair<bytes, std::string>> cols = {
                 {bytes{'v', 'a', 'l'}, "org.apache.cassandra.db.marshal.BytesType"}};
             write_sstable_device(opb, rows, R, sp, d_stats, d_tomb, tomb_cap, base,
-                                 "org.apache.cassandra.db.marshal.LongType",
+                                 gp.key_len > 8 ? "org.apache.cassandra.db.marshal.BytesType"
+                                                : "org.apache.cassandra.db.marshal.LongType",
                                  spec->clustering_rows ? "org.apache.cassandra.db.marshal.LongType" : "",
                                  cols, stream);
         }

@@ -79,13 +79,14 @@ __global__ void k_parse_count(const SrcDesc2* srcs, uint32_t n_srcs, uint32_t to
 
     uint32_t klen = ((uint32_t)base[pos] << 8) | base[pos + 1];
     pos += 2;
-    if (klen == 0 || klen > 8) { atomicExch(error, 10ull); return; }
+    if (klen == 0) { atomicExch(error, 10ull); return; }
     uint64_t pfx = 0;
-    uint8_t keyb[8];
-    for (uint32_t b = 0; b < klen; b++) { keyb[b] = base[pos + b]; pfx |= (uint64_t)keyb[b] << (8 * (7 - b)); }
+    uint32_t pb = klen < 8 ? klen : 8;
+    for (uint32_t b = 0; b < pb; b++) pfx |= (uint64_t)base[pos + b] << (8 * (7 - b));
+    int64_t token = murmur3_token(base + pos, klen);
+    pc.key_addr[gi] = (uint64_t)(base + pos);
     pos += klen;
-    int64_t token = murmur3_token(keyb, klen);
-    MRec r{(uint64_t)token ^ 0x8000000000000000ULL, pfx, li, (uint16_t)s, (uint8_t)klen, 0};
+    MRec r{(uint64_t)token ^ 0x8000000000000000ULL, pfx, li, (uint16_t)s, (uint16_t)klen};
     recs[gi] = r;
 
     int64_t pdm = INT64_MIN;
@@ -334,6 +335,7 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
     const MRec r0 = recs[beg];
     int64_t token = (int64_t)(r0.tok ^ 0x8000000000000000ULL);
     op.keypfx[g] = r0.pfx;
+    op.key_addr[g] = pc.key_addr[src_bases[r0.src] + r0.idx];
     op.klen[g] = r0.klen;
     op.row_base[g] = out_base[g];
     op.row_count[g] = 0;
@@ -798,13 +800,11 @@ __device__ uint64_t part_walk(const OutParts& op, const UnfCols& out, uint64_t g
         return 0;
     }
     uint32_t klen = op.klen[g];
-    uint64_t kp = op.keypfx[g];
+    const uint8_t* keyb = (const uint8_t*)op.key_addr[g];
     bool pdel_live = op.pdel_mfda[g] == INT64_MIN && op.pdel_ldt[g] == LDT_NONE_U32;
     uint64_t header_len = 2 + klen + (pdel_live ? 1 : 12);
     uint64_t rb = op.row_base[g];
     uint32_t nrows = op.row_count[g];
-    uint8_t keyb[8];
-    for (uint32_t b = 0; b < klen; b++) keyb[b] = (uint8_t)(kp >> (8 * (7 - b)));
 
     uint64_t pos = 0;
     auto emit8 = [&](uint8_t v) {
@@ -1141,10 +1141,8 @@ __global__ void k_serialize_rows(OutParts op, UnfCols out, uint64_t n, SerParams
                     out_data, out_index, lane, nullptr, nullptr, nullptr);
     if (lane == 0) {
         uint32_t klen = op.klen[g];
-        uint8_t keyb[8];
-        for (uint32_t b = 0; b < klen; b++) keyb[b] = (uint8_t)(op.keypfx[g] >> (8 * (7 - b)));
         uint64_t h[2];
-        murmur3_128(keyb, klen, 0, h);
+        murmur3_128((const uint8_t*)op.key_addr[g], klen, 0, h);
         int64_t base = (int64_t)h[1], inc = (int64_t)h[0];
         for (int i = 0; i < bloom_k; i++) {
             int64_t m = base % (int64_t)bloom_bitlen;
@@ -1260,18 +1258,26 @@ struct GenParams2 {
     uint32_t sst;
     uint32_t value_len, value_repeat_pct, tombstone_pct, partition_del_pct;
     uint32_t clustering_rows, range_tomb_pct;
+    uint32_t key_len;   // 8 (default) .. 255; bytes 8.. are gen2_key_salt(id, j)
     int64_t base_ts, base_ldt;
 };
 
-__global__ void k_gen_recs2(GenParams2 gp, MRec* recs, uint64_t* ids) {
+// shared generator contract (oracle/src/gen.h): key = 8-byte BE id, then
+// salt bytes for key_len > 8
+__device__ __host__ inline uint8_t gen2_key_salt(uint64_t id, uint32_t j) {
+    return (uint8_t)splitmix64(id ^ (0xC0FFEE5EEDULL + j));
+}
+
+__global__ void k_gen_recs2(GenParams2 gp, MRec* recs, uint64_t* ids, uint8_t* keys) {
     uint64_t j = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (j >= gp.rows) return;
     uint64_t id = feistel_perm(gp.seed, gp.universe, (gp.sst * gp.stride + j) % gp.universe);
     ids[j] = id;
-    uint8_t key[8];
+    uint8_t* key = keys + j * gp.key_len;
     for (int b = 0; b < 8; b++) key[b] = (uint8_t)(id >> (8 * (7 - b)));
-    int64_t tok = murmur3_token(key, 8);
-    recs[j] = MRec{(uint64_t)tok ^ 0x8000000000000000ULL, id, (uint32_t)j, 0, 8, 0};
+    for (uint32_t b = 8; b < gp.key_len; b++) key[b] = gen2_key_salt(id, b);
+    int64_t tok = murmur3_token(key, gp.key_len);
+    recs[j] = MRec{(uint64_t)tok ^ 0x8000000000000000ULL, id, (uint32_t)j, 0, (uint16_t)gp.key_len};
 }
 
 __device__ inline uint32_t gen2_ldt(const GenParams2& gp, uint64_t key_id, uint64_t salt) {
@@ -1280,7 +1286,7 @@ __device__ inline uint32_t gen2_ldt(const GenParams2& gp, uint64_t key_id, uint6
 
 // per-partition unfiltered count (upper bound used for layout: exact)
 __global__ void k_gen_count(GenParams2 gp, const MRec* sorted, const uint64_t* ids, uint64_t n,
-                            OutParts op, uint64_t* prow_count) {
+                            OutParts op, uint64_t* prow_count, const uint8_t* keys) {
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
     uint64_t id = ids[sorted[i].idx];
@@ -1298,7 +1304,8 @@ __global__ void k_gen_count(GenParams2 gp, const MRec* sorted, const uint64_t* i
     }
     prow_count[i] = cnt;
     op.keypfx[i] = sorted[i].pfx;
-    op.klen[i] = 8;
+    op.key_addr[i] = (uint64_t)(keys + (uint64_t)sorted[i].idx * gp.key_len);
+    op.klen[i] = (uint16_t)gp.key_len;
     op.keep[i] = 1;
 }
 

@@ -114,6 +114,7 @@ typedef struct gpuc_gen_spec {
     uint32_t partition_del_pct;
     uint32_t clustering_rows;   /* >0: wide partitions (LongType ck), this many rows each */
     uint32_t range_tomb_pct;    /* % of wide partitions with one range tombstone */
+    uint32_t key_len;           /* partition key bytes, 8..255 (0 == 8); >8 appends salt bytes */
     int64_t base_ts;
     int64_t base_ldt;
     uint64_t first_generation;

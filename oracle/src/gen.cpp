@@ -7,7 +7,7 @@ namespace oracle {
 SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
     SSTable t;
     t.generation = g.first_generation + sst;
-    t.header.key_type = CqlType::LONG;  // pk bigint
+    t.header.key_type = g.key_len > 8 ? CqlType::BYTES : CqlType::LONG;  // pk bigint / blob
     if (g.clustering_rows > 0) t.header.clustering_types = {CqlType::LONG};  // ck bigint
     t.header.regular_cols = {{bytes{'v', 'a', 'l'}, CqlType::BYTES}};  // val blob
 
@@ -16,8 +16,7 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
     ents.reserve(g.rows_per_sstable);
     for (uint64_t j = 0; j < g.rows_per_sstable; j++) {
         uint64_t id = gen_key_id(g, sst, j);
-        bytes key(8);
-        for (int b = 0; b < 8; b++) key[b] = (uint8_t)(id >> (8 * (7 - b)));  // LongType BE
+        bytes key = gen_key_bytes(g, id);
         int64_t tok = murmur3_token(key.data(), key.size());
         ents.push_back({tok, std::move(key), id});
     }

@@ -31,6 +31,10 @@ struct GenSpec {
     // tombstone [a, b] over the clustering space.
     uint32_t clustering_rows = 0;
     uint32_t range_tomb_pct = 0;
+    // partition key width in bytes, 8..255: first 8 = big-endian key id,
+    // bytes 8.. = splitmix64(id ^ (0xC0FFEE5EED + j)) (GPU: gen2_key_salt).
+    // key_len > 8 switches the declared key type LongType -> BytesType.
+    uint32_t key_len = 8;
 
     uint64_t stride() const { return rows_per_sstable * (100 - overlap_pct) / 100; }
     uint64_t universe() const {
@@ -60,6 +64,13 @@ inline uint64_t feistel_perm(uint64_t seed, uint64_t universe, uint64_t x) {
 inline uint64_t gen_key_id(const GenSpec& g, uint32_t sst, uint64_t j) {
     uint64_t u = g.universe();
     return feistel_perm(g.seed, u, (sst * g.stride() + j) % u);
+}
+inline bytes gen_key_bytes(const GenSpec& g, uint64_t id) {
+    bytes key(g.key_len ? g.key_len : 8);
+    for (int b = 0; b < 8; b++) key[b] = (uint8_t)(id >> (8 * (7 - b)));
+    for (size_t b = 8; b < key.size(); b++)
+        key[b] = (uint8_t)splitmix64(id ^ (0xC0FFEE5EEDULL + (uint64_t)b));
+    return key;
 }
 inline int64_t gen_ts(const GenSpec& g, uint32_t sst, uint64_t key_id) {
     return g.base_ts + (int64_t)(splitmix64(g.seed ^ key_id * 31 ^ ((uint64_t)sst << 48)) % 1000000000ULL);

@@ -101,6 +101,7 @@ static GenSpec spec_from_kv(std::map<std::string, std::string>& kv) {
     g.partition_del_pct = (uint32_t)geti("pdel", g.partition_del_pct);
     g.clustering_rows = (uint32_t)geti("crows", g.clustering_rows);
     g.range_tomb_pct = (uint32_t)geti("rtomb", g.range_tomb_pct);
+    g.key_len = (uint32_t)geti("keylen", g.key_len);
     g.first_generation = geti("gen0", g.first_generation);
     return g;
 }

@@ -74,6 +74,18 @@ def test_generator_parity_wide(ca, oracle_bin, tmp_path):
         _assert_dirs_equal(f"{dg}/oa-{g}-big", f"{do}/oa-{g}-big")
 
 
+def test_generator_parity_long_keys(ca, oracle_bin, tmp_path):
+    """Arbitrary-length partition keys (BytesType, 24 B): write-path parity."""
+    dg, do = str(tmp_path / "gpu"), str(tmp_path / "cpu")
+    os.makedirs(dg), os.makedirs(do)
+    kw = dict(seed=61, n=3, rows=1500, vlen=400, overlap=25, tomb=10, keylen=24)
+    ca.generate(dg, seed=61, n_sstables=3, rows_per_sstable=1500, value_len=400,
+                overlap_pct=25, tombstone_pct=10, key_len=24)
+    _oracle_gen(do, **kw)
+    for g in (1, 2, 3):
+        _assert_dirs_equal(f"{dg}/oa-{g}-big", f"{do}/oa-{g}-big")
+
+
 @pytest.mark.parametrize("case", [
     dict(name="plain", gen=dict(seed=42, n=4, rows=3000, vlen=512, overlap=10), job={}),
     dict(name="tombstones_nogc", gen=dict(seed=17, n=4, rows=2500, vlen=256, overlap=30,
@@ -102,6 +114,14 @@ def test_generator_parity_wide(ca, oracle_bin, tmp_path):
                                               overlap=50, rtomb=40), job={}),
     dict(name="wide_single_input", gen=dict(seed=54, n=1, rows=40, crows=60, vlen=250,
                                             tomb=10, rtomb=25, overlap=0), job={}),
+    dict(name="long_keys", gen=dict(seed=62, n=4, rows=2000, vlen=300, overlap=40,
+                                    tomb=10, keylen=24), job={}),
+    dict(name="long_keys_max", gen=dict(seed=63, n=2, rows=800, vlen=200, overlap=50,
+                                        keylen=255), job={}),
+    dict(name="long_keys_wide", gen=dict(seed=64, n=3, rows=40, crows=60, vlen=250,
+                                         overlap=30, tomb=10, rtomb=30, keylen=32), job={}),
+    dict(name="long_keys_gc", gen=dict(seed=65, n=3, rows=1500, vlen=200, overlap=40,
+                                       tomb=25, keylen=24), job=dict(gc_before=2000000000)),
 ])
 def test_compaction_parity(ca, oracle_bin, tmp_path, case):
     d = str(tmp_path)
