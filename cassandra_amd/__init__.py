@@ -79,6 +79,7 @@ class GpucGenSpec(ctypes.Structure):
         ("clustering_rows", ctypes.c_uint32),
         ("range_tomb_pct", ctypes.c_uint32),
         ("key_len", ctypes.c_uint32),
+        ("ck_text", ctypes.c_uint32),
         ("base_ts", ctypes.c_int64),
         ("base_ldt", ctypes.c_int64),
         ("first_generation", ctypes.c_uint64),
@@ -212,6 +213,7 @@ def generate(
     clustering_rows=0,
     range_tomb_pct=0,
     key_len=8,
+    ck_text=False,
     base_ts=1700000000000000,
     base_ldt=1700000000,
     first_generation=1,
@@ -231,6 +233,7 @@ def generate(
         clustering_rows=clustering_rows,
         range_tomb_pct=range_tomb_pct,
         key_len=key_len,
+        ck_text=1 if ck_text else 0,
         base_ts=base_ts,
         base_ldt=base_ldt,
         first_generation=first_generation,

@@ -79,6 +79,12 @@ struct UnfCols {
     int32_t* cell_ttl;
     uint64_t* val_addr;   // absolute device address of value bytes
     uint32_t* val_len;
+    // variable-width clustering (ck_width == -1, e.g. UTF8Type/BytesType):
+    // `ck` then holds the first min(8,len) value bytes big-endian zero-padded
+    // (unsigned lexicographic prefix, no sign flip); full bytes live at
+    // ck_addr/ck_len and break prefix ties exactly.
+    uint64_t* ck_addr;
+    uint32_t* ck_len;
 };
 enum : uint8_t {
     PF_HAS_ROW = 1, PF_LIVE_TS = 2, PF_ROW_DEL = 4, PF_HAS_CELL = 8,

@@ -651,7 +651,7 @@ struct OutPartsBuf {
 // allocate an UnfCols arena for n unfiltereds
 struct UnfColsBuf {
     DevBuf ck, rkind, flags, live_ts, live_ttl, live_let, rdel_mfda, rdel_ldt,
-        start_mfda, start_ldt, cell_ts, cell_ldt, cell_ttl, val_addr, val_len;
+        start_mfda, start_ldt, cell_ts, cell_ldt, cell_ttl, val_addr, val_len, ck_addr, ck_len;
     UnfCols uc{};
     void alloc(uint64_t n) {
         if (!n) n = 1;
@@ -660,11 +660,13 @@ struct UnfColsBuf {
         rdel_ldt.alloc(n * 4); start_mfda.alloc(n * 8); start_ldt.alloc(n * 4);
         cell_ts.alloc(n * 8); cell_ldt.alloc(n * 4); cell_ttl.alloc(n * 4);
         val_addr.alloc(n * 8); val_len.alloc(n * 4);
+        ck_addr.alloc(n * 8); ck_len.alloc(n * 4);
         uc = UnfCols{ck.as<uint64_t>(), rkind.as<uint8_t>(), flags.as<uint8_t>(),
                      live_ts.as<int64_t>(), live_ttl.as<int32_t>(), live_let.as<int64_t>(),
                      rdel_mfda.as<int64_t>(), rdel_ldt.as<uint32_t>(), start_mfda.as<int64_t>(),
                      start_ldt.as<uint32_t>(), cell_ts.as<int64_t>(), cell_ldt.as<uint32_t>(),
-                     cell_ttl.as<int32_t>(), val_addr.as<uint64_t>(), val_len.as<uint32_t>()};
+                     cell_ttl.as<int32_t>(), val_addr.as<uint64_t>(), val_len.as<uint32_t>(),
+                     ck_addr.as<uint64_t>(), ck_len.as<uint32_t>()};
     }
 };
 
@@ -1126,6 +1128,10 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
             ck_type_str = stats[0].clustering_types[0];
             if (ck_type_str == "org.apache.cassandra.db.marshal.LongType") ck_width = 8;
             else if (ck_type_str == "org.apache.cassandra.db.marshal.Int32Type") ck_width = 4;
+            else if (ck_type_str == "org.apache.cassandra.db.marshal.UTF8Type" ||
+                     ck_type_str == "org.apache.cassandra.db.marshal.AsciiType" ||
+                     ck_type_str == "org.apache.cassandra.db.marshal.BytesType")
+                ck_width = -1;  // variable width, unsigned-lex comparator
             else throw std::runtime_error("unsupported clustering type " + ck_type_str);
         }
 
@@ -1484,6 +1490,9 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
             gp.base_ts = spec->base_ts;
             gp.base_ldt = spec->base_ldt;
             gp.key_len = spec->key_len ? spec->key_len : 8;
+            gp.ck_text = spec->ck_text;
+            if (gp.ck_text && (uint64_t)gp.clustering_rows * 16 >= 100000000ull)
+                throw std::runtime_error("ck_text needs clustering_rows*16 < 1e8 (8-digit order)");
             if (gp.key_len < 8 || gp.key_len > 255)
                 throw std::runtime_error("key_len must be 8..255 (generator contract)");
             {
@@ -1509,9 +1518,11 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
             UnfColsBuf rows;
             rows.alloc(total_rows);
             d_vals.alloc(total_rows * (uint64_t)spec->value_len);
+            DevBuf d_ckarena;
+            if (gp.ck_text) d_ckarena.alloc(total_rows * 16 + 16);
             hipLaunchKernelGGL(k_gen_fill2, dim3(blocks), dim3(256), 0, stream, gp, d_sorted,
                                d_ids.as<uint64_t>(), R, opb.op, rows.uc, d_prows.as<uint64_t>(),
-                               d_vals.as<uint8_t>());
+                               d_vals.as<uint8_t>(), gp.ck_text ? d_ckarena.as<uint8_t>() : nullptr);
             hipLaunchKernelGGL(k_gen_values2, dim3(blocks), dim3(256), 0, stream, gp, d_sorted,
                                d_ids.as<uint64_t>(), opb.op, rows.uc, R, d_vals.as<uint8_t>(),
                                d_prows.as<uint64_t>());
@@ -1533,7 +1544,7 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
                                 ? DELETION_TIME_EPOCH : (int64_t)(hs0.min_ldt_flip ^ 0x8000000000000000ULL);
             if (sp.hs.min_ldt == NO_DELETION_TIME) sp.hs.min_ldt = DELETION_TIME_EPOCH;
             sp.hs.min_ttl = 0;
-            sp.sch.ck_width = spec->clustering_rows ? 8 : 0;
+            sp.sch.ck_width = spec->clustering_rows ? (gp.ck_text ? -1 : 8) : 0;
             sp.sch.col_fixed_len = -1;  // val blob
             sp.sch.column_index_size = 64 * 1024;
             // reset and recollect so the writer sees fresh stats (collect ran
@@ -1545,7 +1556,10 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
             write_sstable_device(opb, rows, R, sp, d_stats, d_tomb, tomb_cap, base,
                                  gp.key_len > 8 ? "org.apache.cassandra.db.marshal.BytesType"
                                                 : "org.apache.cassandra.db.marshal.LongType",
-                                 spec->clustering_rows ? "org.apache.cassandra.db.marshal.LongType" : "",
+                                 spec->clustering_rows
+                                     ? (gp.ck_text ? "org.apache.cassandra.db.marshal.UTF8Type"
+                                                   : "org.apache.cassandra.db.marshal.LongType")
+                                     : "",
                                  cols, stream);
         }
         HIP_CHECK(hipStreamDestroy(stream));

@@ -26,7 +26,7 @@ __device__ inline void tomb_push(OutStats* st, uint32_t* ldts, uint32_t cap, uin
 
 // schema/runtime constants shared by the general kernels
 struct SchemaParams {
-    int32_t ck_width;        // 0 = no clustering column; 4/8 fixed width
+    int32_t ck_width;        // 0 = none; 4/8 fixed width; -1 variable (UTF8/Bytes)
     int32_t col_fixed_len;   // regular column: -1 variable else fixed width
     uint32_t column_index_size;  // promoted-index granularity (64 KiB default)
 };
@@ -38,16 +38,38 @@ __device__ inline uint64_t ck_sortable(const uint8_t* p, int width) {
     // values compare as signed integers of `width` bytes: flip the sign bit
     return v ^ (1ULL << (8 * width - 1));
 }
+// unsigned-lex sortable prefix of a variable-width clustering value
+__device__ inline uint64_t ck_prefix_var(const uint8_t* p, uint32_t len) {
+    uint64_t v = 0;
+    uint32_t n = len < 8 ? len : 8;
+    for (uint32_t b = 0; b < n; b++) v |= (uint64_t)p[b] << (8 * (7 - b));
+    return v;
+}
 __device__ inline void ck_bytes(uint64_t ck, int width, uint8_t* out) {
     uint64_t v = ck ^ (1ULL << (8 * width - 1));
     for (int b = 0; b < width; b++) out[b] = (uint8_t)(v >> (8 * (width - 1 - b)));
 }
-// position compare: (ck, Kind.comparison); returns <0, 0, >0
-__device__ inline int pos_cmp(uint64_t cka, uint8_t ka, uint64_t ckb, uint8_t kb, int ck_width) {
-    if (ck_width > 0) {
-        if (cka != ckb) return cka < ckb ? -1 : 1;
+// position compare: (ck, Kind.comparison); returns <0, 0, >0.
+// Fixed-width ck compares via the sortable u64; variable-width compares the
+// 8-byte prefix then byte-walks the full values (unsigned lex, shorter-first
+// on prefix equality — ClusteringComparator over UTF8/Bytes values).
+__device__ inline int pos_cmp(const UnfCols& in, uint64_t a, uint64_t b, int ck_width) {
+    if (ck_width != 0) {
+        uint64_t ca = in.ck[a], cb = in.ck[b];
+        if (ca != cb) return ca < cb ? -1 : 1;
+        if (ck_width < 0) {
+            uint32_t la = in.ck_len[a], lb = in.ck_len[b];
+            if (la > 8 || lb > 8) {
+                const uint8_t* pa = (const uint8_t*)in.ck_addr[a];
+                const uint8_t* pb = (const uint8_t*)in.ck_addr[b];
+                uint32_t n = la < lb ? la : lb;
+                for (uint32_t i = 8; i < n; i++)
+                    if (pa[i] != pb[i]) return pa[i] < pb[i] ? -1 : 1;
+            }
+            if (la != lb) return la < lb ? -1 : 1;
+        }
     }
-    int c1 = bk_comparison(ka), c2 = bk_comparison(kb);
+    int c1 = bk_comparison(in.rkind[a]), c2 = bk_comparison(in.rkind[b]);
     return c1 - c2;
 }
 
@@ -119,11 +141,15 @@ __global__ void k_parse_count(const SrcDesc2* srcs, uint32_t n_srcs, uint32_t to
             pos += 2;
             if (kind == BK_STATIC || nv > 1 || (nv == 1 && sp.ck_width == 0)) { atomicExch(error, 15ull); return; }
             if (nv == 0 && sp.ck_width != 0) { atomicExch(error, 16ull); return; }  // 0-value bound unsupported
-            if (nv) { uint64_t hdr = uvint_get(base, &pos); if (hdr) { atomicExch(error, 17ull); return; } pos += sp.ck_width; }
+            if (nv) {
+                uint64_t hdr = uvint_get(base, &pos);
+                if (hdr) { atomicExch(error, 17ull); return; }
+                pos += sp.ck_width > 0 ? (uint64_t)sp.ck_width : uvint_get(base, &pos);
+            }
         } else if (sp.ck_width) {
             uint64_t hdr = uvint_get(base, &pos);
             if (hdr) { atomicExch(error, 17ull); return; }  // null/empty clustering unsupported
-            pos += sp.ck_width;
+            pos += sp.ck_width > 0 ? (uint64_t)sp.ck_width : uvint_get(base, &pos);
         }
         uint64_t size = uvint_get(base, &pos);
         uint64_t prev = uvint_get(base, &pos);
@@ -165,12 +191,26 @@ __global__ void k_parse_rows(const SrcDesc2* srcs, uint32_t n_srcs, uint32_t tot
             uint8_t kind = base[pos++];
             uint32_t nv = ((uint32_t)base[pos] << 8) | base[pos + 1];
             pos += 2;
-            uint64_t ck = 0;
-            if (nv) { uvint_get(base, &pos); ck = ck_sortable(base + pos, sp.ck_width); pos += sp.ck_width; }
+            uint64_t ck = 0, ckaddr = 0;
+            uint32_t cklen = 0;
+            if (nv) {
+                uvint_get(base, &pos);
+                if (sp.ck_width > 0) {
+                    ck = ck_sortable(base + pos, sp.ck_width);
+                    cklen = (uint32_t)sp.ck_width;
+                } else {
+                    cklen = (uint32_t)uvint_get(base, &pos);
+                    ck = ck_prefix_var(base + pos, cklen);
+                }
+                ckaddr = (uint64_t)(base + pos);
+                pos += cklen;
+            }
             uvint_get(base, &pos);  // size
             uvint_get(base, &pos);  // prev
             rc.rkind[o] = kind;
             rc.ck[o] = ck;
+            rc.ck_addr[o] = ckaddr;
+            rc.ck_len[o] = cklen;
             rc.flags[o] = 0;
             // deltas: boundary = end then start; bound = single deletion
             int64_t em = (int64_t)uvint_get(base, &pos) + sd.min_ts;
@@ -196,8 +236,20 @@ __global__ void k_parse_rows(const SrcDesc2* srcs, uint32_t n_srcs, uint32_t tot
             rc.val_len[o] = 0;
         } else {
             // ---- row ----
-            uint64_t ck = 0;
-            if (sp.ck_width) { uvint_get(base, &pos); ck = ck_sortable(base + pos, sp.ck_width); pos += sp.ck_width; }
+            uint64_t ck = 0, ckaddr = 0;
+            uint32_t cklen = 0;
+            if (sp.ck_width) {
+                uvint_get(base, &pos);
+                if (sp.ck_width > 0) {
+                    ck = ck_sortable(base + pos, sp.ck_width);
+                    cklen = (uint32_t)sp.ck_width;
+                } else {
+                    cklen = (uint32_t)uvint_get(base, &pos);
+                    ck = ck_prefix_var(base + pos, cklen);
+                }
+                ckaddr = (uint64_t)(base + pos);
+                pos += cklen;
+            }
             uvint_get(base, &pos);  // size
             uvint_get(base, &pos);  // prev
             uint8_t pf = PF_HAS_ROW;
@@ -243,6 +295,8 @@ __global__ void k_parse_rows(const SrcDesc2* srcs, uint32_t n_srcs, uint32_t tot
             }
             rc.rkind[o] = BK_CLUSTERING;
             rc.ck[o] = ck;
+            rc.ck_addr[o] = ckaddr;
+            rc.ck_len[o] = cklen;
             rc.flags[o] = pf;
             rc.live_ts[o] = lts;
             rc.live_ttl[o] = lttl;
@@ -390,12 +444,16 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
 
     uint64_t obase = out_base[g];
     uint32_t ocount = 0;
+    uint64_t cur_ck_addr = 0;
+    uint32_t cur_ck_len = 0;
     auto emit = [&](uint8_t kind, uint64_t ck, uint8_t flags, int64_t lts, int32_t lttl,
                     int64_t llet, int64_t rdm, uint32_t rdl, int64_t smf, uint32_t sld,
                     int64_t cts, uint32_t cldt, int32_t cttl, uint64_t va, uint32_t vl) {
         uint64_t o = obase + ocount++;
         out.rkind[o] = kind;
         out.ck[o] = ck;
+        out.ck_addr[o] = cur_ck_addr;
+        out.ck_len[o] = cur_ck_len;
         out.flags[o] = flags;
         out.live_ts[o] = lts;
         out.live_ttl[o] = lttl;
@@ -422,11 +480,13 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
             uint64_t o = mb[m] + mpos[m];
             if (first < 0) { first = (int)m; continue; }
             uint64_t f = mb[first] + mpos[first];
-            if (pos_cmp(in.ck[o], in.rkind[o], in.ck[f], in.rkind[f], sp.ck_width) < 0) first = (int)m;
+            if (pos_cmp(in, o, f, sp.ck_width) < 0) first = (int)m;
         }
         if (first < 0) break;
         uint64_t fo = mb[first] + mpos[first];
         uint64_t fck = in.ck[fo];
+        cur_ck_addr = in.ck_addr[fo];
+        cur_ck_len = in.ck_len[fo];
         uint8_t fkind = in.rkind[fo];
         bool is_row = bk_comparison(fkind) == 2;
         // gather members at this position
@@ -436,7 +496,7 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
         for (uint32_t m = 0; m < k; m++) {
             if (mpos[m] >= mcnt[m]) continue;
             uint64_t o = mb[m] + mpos[m];
-            if (pos_cmp(in.ck[o], in.rkind[o], fck, fkind, sp.ck_width) == 0) {
+            if (pos_cmp(in, o, fo, sp.ck_width) == 0) {
                 members |= 1ULL << m;
                 nmem++;
                 lastm = (int)m;
@@ -773,10 +833,11 @@ __device__ inline uint32_t unf_body_size(const UnfCols& u, uint64_t o, const Ser
 
 // serialized size of a ClusteringPrefix (full serializer: kind byte [+ u16
 // size for bounds] + values-without-size) — IndexInfo first/last names
-__device__ inline uint32_t prefix_full_size(uint8_t kind, const SchemaParams& sch) {
+__device__ inline uint32_t prefix_full_size(uint8_t kind, const SchemaParams& sch, uint32_t vlen) {
     uint32_t s = 1;                       // kind byte
     if (kind != BK_CLUSTERING) s += 2;    // u16 value count
-    if (sch.ck_width) s += 1 + sch.ck_width;  // header vint + fixed value
+    if (sch.ck_width > 0) s += 1 + sch.ck_width;           // header vint + fixed value
+    else if (sch.ck_width < 0) s += 1 + uvint_size(vlen) + vlen;  // header + vint len + bytes
     return s;
 }
 __device__ inline uint32_t dt_ser_size(int64_t m, uint32_t l) {
@@ -827,13 +888,27 @@ __device__ uint64_t part_walk(const OutParts& op, const UnfCols& out, uint64_t g
         }
         pos += 12;
     };
-    auto emit_ck = [&](uint64_t ck) {
+    auto emit_ck = [&](uint64_t o) {
         if (sp.sch.ck_width == 0) return;
-        if (EMIT && lane == 0) {
-            out_data[data_off + pos] = 0;  // 32-batch header: one non-null value
-            ck_bytes(ck, sp.sch.ck_width, &out_data[data_off + pos + 1]);
+        if (sp.sch.ck_width > 0) {
+            if (EMIT && lane == 0) {
+                out_data[data_off + pos] = 0;  // 32-batch header: one non-null value
+                ck_bytes(out.ck[o], sp.sch.ck_width, &out_data[data_off + pos + 1]);
+            }
+            pos += 1 + sp.sch.ck_width;
+        } else {
+            uint32_t vlen = out.ck_len[o];
+            if (EMIT && lane == 0) {
+                out_data[data_off + pos] = 0;
+                uint8_t tmp[9];
+                int n = uvint_put(tmp, vlen);
+                for (int i = 0; i < n; i++) out_data[data_off + pos + 1 + i] = tmp[i];
+                const uint8_t* src = (const uint8_t*)out.ck_addr[o];
+                for (uint32_t i = 0; i < vlen; i++)
+                    out_data[data_off + pos + 1 + n + i] = src[i];
+            }
+            pos += 1 + uvint_size(vlen) + vlen;
         }
-        pos += 1 + sp.sch.ck_width;
     };
 
     // ---- index-entry layout (EMIT uses the SIZE pass results) ----
@@ -855,36 +930,49 @@ __device__ uint64_t part_walk(const OutParts& op, const UnfCols& out, uint64_t g
     // ---- walk unfiltereds ----
     uint64_t prev_start = 0;
     bool block_open = false;
-    uint64_t block_start = 0, block_first_ck = 0;
-    uint8_t block_first_kind = 0;
-    uint64_t last_ck = 0;
-    uint8_t last_kind = 0;
+    uint64_t block_start = 0;
+    uint64_t block_first_o = 0, last_o = 0;  // row indices (kind/ck read back)
     int64_t open_m = INT64_MIN;
     uint32_t open_l = LDT_NONE_U32;
     uint32_t nblocks = 0;
     uint64_t infos_size = 0;
     uint64_t info_cursor = infos_begin;
 
-    auto emit_prefix_idx = [&](uint8_t kk, uint64_t ck, uint64_t q) -> uint64_t {
+    auto emit_prefix_idx = [&](uint64_t o, uint64_t q) -> uint64_t {
+        uint8_t kk = out.rkind[o];
         if (EMIT && lane == 0) out_index[q] = kk;
         q++;
         if (kk != BK_CLUSTERING) {
             if (EMIT && lane == 0) { out_index[q] = 0; out_index[q + 1] = (uint8_t)(sp.sch.ck_width ? 1 : 0); }
             q += 2;
         }
-        if (sp.sch.ck_width) {
+        if (sp.sch.ck_width > 0) {
             if (EMIT && lane == 0) {
                 out_index[q] = 0;
-                ck_bytes(ck, sp.sch.ck_width, &out_index[q + 1]);
+                ck_bytes(out.ck[o], sp.sch.ck_width, &out_index[q + 1]);
             }
             q += 1 + sp.sch.ck_width;
+        } else if (sp.sch.ck_width < 0) {
+            uint32_t vlen = out.ck_len[o];
+            if (EMIT && lane == 0) {
+                out_index[q] = 0;
+                uint8_t tmp[9];
+                int n = uvint_put(tmp, vlen);
+                for (int i = 0; i < n; i++) out_index[q + 1 + i] = tmp[i];
+                const uint8_t* src = (const uint8_t*)out.ck_addr[o];
+                for (uint32_t i = 0; i < vlen; i++) out_index[q + 1 + n + i] = src[i];
+            }
+            q += 1 + uvint_size(vlen) + vlen;
         }
         return q;
     };
     auto flush_block = [&](uint64_t end_pos) {
         uint64_t width = end_pos - block_start;
         bool has_open = !(open_m == INT64_MIN && open_l == LDT_NONE_U32);
-        uint64_t isz = prefix_full_size(block_first_kind, sp.sch) + prefix_full_size(last_kind, sp.sch) +
+        uint64_t isz = prefix_full_size(out.rkind[block_first_o], sp.sch,
+                                        sp.sch.ck_width < 0 ? out.ck_len[block_first_o] : 0) +
+                       prefix_full_size(out.rkind[last_o], sp.sch,
+                                        sp.sch.ck_width < 0 ? out.ck_len[last_o] : 0) +
                        uvint_size(block_start) + uvint_size(zigzag((int64_t)width - 65536)) + 1 +
                        (has_open ? dt_ser_size(open_m, open_l) : 0);
         if (EMIT && nblocks_hint > 1) {
@@ -896,8 +984,8 @@ __device__ uint64_t part_walk(const OutParts& op, const UnfCols& out, uint64_t g
                 out_index[slot + 2] = (uint8_t)(rel >> 8);
                 out_index[slot + 3] = (uint8_t)rel;
             }
-            uint64_t q = emit_prefix_idx(block_first_kind, block_first_ck, info_cursor);
-            q = emit_prefix_idx(last_kind, last_ck, q);
+            uint64_t q = emit_prefix_idx(block_first_o, info_cursor);
+            q = emit_prefix_idx(last_o, q);
             if (lane == 0) {
                 uint8_t tmp[10];
                 int n = uvint_put(tmp, block_start);
@@ -932,8 +1020,7 @@ __device__ uint64_t part_walk(const OutParts& op, const UnfCols& out, uint64_t g
         if (!block_open) {
             block_open = true;
             block_start = upos;
-            block_first_ck = out.ck[o];
-            block_first_kind = out.rkind[o];
+            block_first_o = o;
         }
         uint8_t kind = out.rkind[o];
         uint8_t rflags = 0, cflags = 0;
@@ -944,7 +1031,7 @@ __device__ uint64_t part_walk(const OutParts& op, const UnfCols& out, uint64_t g
             emit8(kind);
             emit8(0);
             emit8(sp.sch.ck_width ? 1 : 0);
-            emit_ck(out.ck[o]);
+            emit_ck(o);
             emit_uv(body + uvint_size(prev_sz));
             emit_uv(prev_sz);
             emit_uv((uint64_t)(out.rdel_mfda[o] - sp.hs.min_ts));
@@ -957,7 +1044,7 @@ __device__ uint64_t part_walk(const OutParts& op, const UnfCols& out, uint64_t g
             open_l = bk_is_open(kind) ? (bk_is_boundary(kind) ? out.start_ldt[o] : out.rdel_ldt[o]) : LDT_NONE_U32;
         } else {
             emit8(rflags);
-            emit_ck(out.ck[o]);
+            emit_ck(o);
             emit_uv(body + uvint_size(prev_sz));
             emit_uv(prev_sz);
             uint8_t f = out.flags[o];
@@ -991,8 +1078,7 @@ __device__ uint64_t part_walk(const OutParts& op, const UnfCols& out, uint64_t g
             }
         }
         prev_start = upos;
-        last_ck = out.ck[o];
-        last_kind = kind;
+        last_o = o;
         if (pos - block_start >= sp.sch.column_index_size) flush_block(pos);
     }
     emit8(0x01);  // END_OF_PARTITION
@@ -1259,11 +1345,25 @@ struct GenParams2 {
     uint32_t value_len, value_repeat_pct, tombstone_pct, partition_del_pct;
     uint32_t clustering_rows, range_tomb_pct;
     uint32_t key_len;   // 8 (default) .. 255; bytes 8.. are gen2_key_salt(id, j)
+    uint32_t ck_text;   // clustering values as UTF8 strings (oracle gen_ck_bytes)
     int64_t base_ts, base_ldt;
 };
 
 // shared generator contract (oracle/src/gen.h): key = 8-byte BE id, then
 // salt bytes for key_len > 8
+// text clustering value (oracle gen_ck_bytes contract): 8 zero-padded
+// decimal digits of the numeric position; rows append (pos/16 %% 3) 'x' bytes
+__device__ inline uint32_t gen2_ck_text(int64_t ckval, bool is_row, uint8_t* out) {
+    uint64_t v = (uint64_t)ckval;
+    for (int i = 7; i >= 0; i--) { out[i] = (uint8_t)('0' + v % 10); v /= 10; }
+    uint32_t len = 8;
+    if (is_row) {
+        uint32_t sfx = (uint32_t)(((uint64_t)ckval / 16) % 3);
+        for (uint32_t i = 0; i < sfx; i++) out[len + i] = 'x';
+        len += sfx;
+    }
+    return len;
+}
 __device__ __host__ inline uint8_t gen2_key_salt(uint64_t id, uint32_t j) {
     return (uint8_t)splitmix64(id ^ (0xC0FFEE5EEDULL + j));
 }
@@ -1311,17 +1411,31 @@ __global__ void k_gen_count(GenParams2 gp, const MRec* sorted, const uint64_t* i
 
 __global__ void k_gen_fill2(GenParams2 gp, const MRec* sorted, const uint64_t* ids, uint64_t n,
                             OutParts op, UnfCols out, const uint64_t* row_base,
-                            uint8_t* values) {
+                            uint8_t* values, uint8_t* ck_arena) {
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
     uint64_t id = ids[sorted[i].idx];
     uint64_t ob = row_base[i];
     op.row_base[i] = ob;
     uint32_t emitted = 0;
+    auto put_ck = [&](uint64_t o, int64_t ck, bool has_ck, bool is_row) {
+        if (!has_ck) { out.ck[o] = 0; out.ck_addr[o] = 0; out.ck_len[o] = 0; return; }
+        if (gp.ck_text) {
+            uint8_t* dst = ck_arena + o * 16;
+            uint32_t len = gen2_ck_text(ck, is_row, dst);
+            out.ck[o] = ck_prefix_var(dst, len);
+            out.ck_addr[o] = (uint64_t)dst;
+            out.ck_len[o] = len;
+        } else {
+            out.ck[o] = (uint64_t)ck ^ 0x8000000000000000ULL;
+            out.ck_addr[o] = 0;
+            out.ck_len[o] = 8;
+        }
+    };
     auto put_marker = [&](uint8_t kind, int64_t ck, int64_t m, uint32_t l) {
         uint64_t o = ob + emitted++;
         out.rkind[o] = kind;
-        out.ck[o] = (uint64_t)ck ^ 0x8000000000000000ULL;
+        put_ck(o, ck, true, false);
         out.flags[o] = 0;
         out.live_ts[o] = NO_TIMESTAMP;
         out.live_ttl[o] = 0;
@@ -1340,7 +1454,7 @@ __global__ void k_gen_fill2(GenParams2 gp, const MRec* sorted, const uint64_t* i
                        uint64_t val_slot) {
         uint64_t o = ob + emitted++;
         out.rkind[o] = BK_CLUSTERING;
-        out.ck[o] = has_ck ? ((uint64_t)ck ^ 0x8000000000000000ULL) : 0;
+        put_ck(o, ck, has_ck, true);
         out.start_mfda[o] = INT64_MIN;
         out.start_ldt[o] = LDT_NONE_U32;
         out.live_ttl[o] = 0;

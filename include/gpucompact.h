@@ -115,6 +115,7 @@ typedef struct gpuc_gen_spec {
     uint32_t clustering_rows;   /* >0: wide partitions (LongType ck), this many rows each */
     uint32_t range_tomb_pct;    /* % of wide partitions with one range tombstone */
     uint32_t key_len;           /* partition key bytes, 8..255 (0 == 8); >8 appends salt bytes */
+    uint32_t ck_text;           /* clustering values as UTF8 strings (variable width) */
     int64_t base_ts;
     int64_t base_ldt;
     uint64_t first_generation;

@@ -8,7 +8,8 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
     SSTable t;
     t.generation = g.first_generation + sst;
     t.header.key_type = g.key_len > 8 ? CqlType::BYTES : CqlType::LONG;  // pk bigint / blob
-    if (g.clustering_rows > 0) t.header.clustering_types = {CqlType::LONG};  // ck bigint
+    if (g.clustering_rows > 0)
+        t.header.clustering_types = {g.ck_text ? CqlType::UTF8 : CqlType::LONG};
     t.header.regular_cols = {{bytes{'v', 'a', 'l'}, CqlType::BYTES}};  // val blob
 
     struct Ent { int64_t token; bytes key; uint64_t id; };
@@ -54,7 +55,7 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
                     Unfiltered u;
                     u.kind = Unfiltered::MARKER;
                     u.marker.kind = INCL_START;
-                    u.marker.values = {ClusterVal{ClusterVal::VALUE, be8(rlo)}};
+                    u.marker.values = {ClusterVal{ClusterVal::VALUE, gen_ck_bytes(g, rlo, false)}};
                     u.marker.end_dt = DeletionTime{rts, rldt};
                     // marker sits at bound position rlo (before this row)
                     p.items.push_back(std::move(u));
@@ -63,7 +64,7 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
                 Unfiltered u;
                 u.kind = Unfiltered::ROW;
                 Row& r = u.row;
-                r.clustering = {ClusterVal{ClusterVal::VALUE, be8(ck)}};
+                r.clustering = {ClusterVal{ClusterVal::VALUE, gen_ck_bytes(g, ck, true)}};
                 r.cells.resize(1);
                 int64_t ts = gen_row_ts(g, sst, e.id, j);
                 min_ts = std::min(min_ts, ts);
@@ -84,7 +85,7 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
                     Unfiltered m;
                     m.kind = Unfiltered::MARKER;
                     m.marker.kind = INCL_END;
-                    m.marker.values = {ClusterVal{ClusterVal::VALUE, be8(rhi)}};
+                    m.marker.values = {ClusterVal{ClusterVal::VALUE, gen_ck_bytes(g, rhi, false)}};
                     m.marker.end_dt = DeletionTime{rts, rldt};
                     p.items.push_back(std::move(m));
                     rt_open = false;
@@ -95,7 +96,7 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
                 Unfiltered m;
                 m.kind = Unfiltered::MARKER;
                 m.marker.kind = INCL_END;
-                m.marker.values = {ClusterVal{ClusterVal::VALUE, be8(rhi)}};
+                m.marker.values = {ClusterVal{ClusterVal::VALUE, gen_ck_bytes(g, rhi, false)}};
                 m.marker.end_dt = DeletionTime{rts, rldt};
                 p.items.push_back(std::move(m));
             }

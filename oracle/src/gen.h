@@ -31,6 +31,11 @@ struct GenSpec {
     // tombstone [a, b] over the clustering space.
     uint32_t clustering_rows = 0;
     uint32_t range_tomb_pct = 0;
+    // ck_text: clustering values become UTF8 strings instead of bigint —
+    // "%08u" decimal of the numeric position (so byte order == numeric
+    // order) plus, for ROW values only, (j %% 3) 'x' suffix bytes to
+    // exercise variable width. GPU mirror: gen2_ck_text.
+    uint32_t ck_text = 0;
     // partition key width in bytes, 8..255: first 8 = big-endian key id,
     // bytes 8.. = splitmix64(id ^ (0xC0FFEE5EED + j)) (GPU: gen2_key_salt).
     // key_len > 8 switches the declared key type LongType -> BytesType.
@@ -80,6 +85,19 @@ inline bool gen_is_tombstone(const GenSpec& g, uint32_t sst, uint64_t key_id) {
     return splitmix64(g.seed ^ 0xDEADULL ^ key_id ^ ((uint64_t)sst << 32)) % 100 < g.tombstone_pct;
 }
 // wide-partition derivations (rows keyed by (key_id, row j))
+inline bytes gen_ck_bytes(const GenSpec& g, int64_t ckval, bool is_row) {
+    if (!g.ck_text) {
+        bytes b(8);
+        for (int i = 0; i < 8; i++) b[i] = (uint8_t)((uint64_t)ckval >> (8 * (7 - i)));
+        return b;
+    }
+    char buf[12];
+    snprintf(buf, sizeof(buf), "%08llu", (unsigned long long)ckval);
+    bytes out(buf, buf + 8);
+    if (is_row)
+        for (int i = 0; i < (int)((ckval / 16) % 3); i++) out.push_back('x');
+    return out;
+}
 inline int64_t gen_ck(const GenSpec& g, uint64_t key_id, uint32_t j) {
     (void)g; (void)key_id;
     return (int64_t)j * 16;  // ascending, gaps so range bounds can fall between rows

@@ -102,6 +102,7 @@ static GenSpec spec_from_kv(std::map<std::string, std::string>& kv) {
     g.clustering_rows = (uint32_t)geti("crows", g.clustering_rows);
     g.range_tomb_pct = (uint32_t)geti("rtomb", g.range_tomb_pct);
     g.key_len = (uint32_t)geti("keylen", g.key_len);
+    g.ck_text = (uint32_t)geti("cktext", g.ck_text);
     g.first_generation = geti("gen0", g.first_generation);
     return g;
 }

@@ -74,6 +74,20 @@ def test_generator_parity_wide(ca, oracle_bin, tmp_path):
         _assert_dirs_equal(f"{dg}/oa-{g}-big", f"{do}/oa-{g}-big")
 
 
+def test_generator_parity_text_ck(ca, oracle_bin, tmp_path):
+    """Variable-width (UTF8) clustering values: write-path parity."""
+    dg, do = str(tmp_path / "gpu"), str(tmp_path / "cpu")
+    os.makedirs(dg), os.makedirs(do)
+    kw = dict(seed=75, n=2, rows=40, crows=120, vlen=400, overlap=30, tomb=10, rtomb=40,
+              cktext=1)
+    ca.generate(dg, seed=75, n_sstables=2, rows_per_sstable=40, clustering_rows=120,
+                value_len=400, overlap_pct=30, tombstone_pct=10, range_tomb_pct=40,
+                ck_text=True)
+    _oracle_gen(do, **kw)
+    for g in (1, 2):
+        _assert_dirs_equal(f"{dg}/oa-{g}-big", f"{do}/oa-{g}-big")
+
+
 def test_generator_parity_long_keys(ca, oracle_bin, tmp_path):
     """Arbitrary-length partition keys (BytesType, 24 B): write-path parity."""
     dg, do = str(tmp_path / "gpu"), str(tmp_path / "cpu")
@@ -122,6 +136,18 @@ def test_generator_parity_long_keys(ca, oracle_bin, tmp_path):
                                          overlap=30, tomb=10, rtomb=30, keylen=32), job={}),
     dict(name="long_keys_gc", gen=dict(seed=65, n=3, rows=1500, vlen=200, overlap=40,
                                        tomb=25, keylen=24), job=dict(gc_before=2000000000)),
+    dict(name="text_ck_plain", gen=dict(seed=71, n=3, rows=60, crows=50, vlen=300,
+                                        overlap=25, cktext=1), job={}),
+    dict(name="text_ck_tombstones", gen=dict(seed=72, n=3, rows=50, crows=40, vlen=200,
+                                             overlap=30, tomb=15, rtomb=35, cktext=1), job={}),
+    dict(name="text_ck_gc", gen=dict(seed=72, n=3, rows=50, crows=40, vlen=200, overlap=30,
+                                     tomb=15, rtomb=35, cktext=1),
+         job=dict(gc_before=2000000000)),
+    dict(name="text_ck_promoted_index", gen=dict(seed=73, n=2, rows=10, crows=400, vlen=700,
+                                                 overlap=50, rtomb=40, cktext=1), job={}),
+    dict(name="text_ck_long_keys", gen=dict(seed=74, n=3, rows=40, crows=50, vlen=250,
+                                            overlap=30, tomb=10, rtomb=25, cktext=1,
+                                            keylen=32), job={}),
 ])
 def test_compaction_parity(ca, oracle_bin, tmp_path, case):
     d = str(tmp_path)

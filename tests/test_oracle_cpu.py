@@ -142,3 +142,16 @@ def test_long_keys_roundtrip_and_merge(oracle_bin, tmp_path):
         a = open(f"{d}/nz/oa-50-big-{c}", "rb").read()
         b = open(f"{d}/nz/oa-51-big-{c}", "rb").read()
         assert a == b, c
+
+
+def test_text_clustering_roundtrip_and_merge(oracle_bin, tmp_path):
+    """UTF8 (variable-width) clustering values: write -> reread -> compact."""
+    d = str(tmp_path)
+    _gen(d, n=3, rows=40, crows=60, vlen=200, overlap=30, tomb=10, rtomb=30,
+         cktext=1, seed=76)
+    ins = [f"{d}/oa-{g}-big" for g in (1, 2, 3)]
+    stats = _compact(f"{d}/oa-50-big", ins)
+    assert stats["partitions_out"] > 0
+    assert stats["rows_out"] > 0
+    r = oracle_run("roundtrip", f"{d}/oa-50-big")
+    assert "FAIL" not in r.stdout, r.stdout
