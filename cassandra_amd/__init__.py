@@ -80,6 +80,8 @@ class GpucGenSpec(ctypes.Structure):
         ("range_tomb_pct", ctypes.c_uint32),
         ("key_len", ctypes.c_uint32),
         ("ck_text", ctypes.c_uint32),
+        ("n_value_cols", ctypes.c_uint32),
+        ("col_missing_pct", ctypes.c_uint32),
         ("base_ts", ctypes.c_int64),
         ("base_ldt", ctypes.c_int64),
         ("first_generation", ctypes.c_uint64),
@@ -214,6 +216,8 @@ def generate(
     range_tomb_pct=0,
     key_len=8,
     ck_text=False,
+    n_value_cols=1,
+    col_missing_pct=0,
     base_ts=1700000000000000,
     base_ldt=1700000000,
     first_generation=1,
@@ -234,6 +238,8 @@ def generate(
         range_tomb_pct=range_tomb_pct,
         key_len=key_len,
         ck_text=1 if ck_text else 0,
+        n_value_cols=n_value_cols,
+        col_missing_pct=col_missing_pct,
         base_ts=base_ts,
         base_ldt=base_ldt,
         first_generation=first_generation,

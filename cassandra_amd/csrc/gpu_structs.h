@@ -74,11 +74,14 @@ struct UnfCols {
     uint32_t* rdel_ldt;
     int64_t* start_mfda;  // marker open deletion (boundary markers)
     uint32_t* start_ldt;
+    // regular-column cells, row-major strided: index = row_o * n_cols + c
+    // (n_cols from SchemaParams; n_cols == 1 degenerates to the old layout)
     int64_t* cell_ts;
     uint32_t* cell_ldt;
     int32_t* cell_ttl;
     uint64_t* val_addr;   // absolute device address of value bytes
     uint32_t* val_len;
+    uint8_t* cell_flags;  // CF_* per cell
     // variable-width clustering (ck_width == -1, e.g. UTF8Type/BytesType):
     // `ck` then holds the first min(8,len) value bytes big-endian zero-padded
     // (unsigned lexicographic prefix, no sign flip); full bytes live at
@@ -86,10 +89,9 @@ struct UnfCols {
     uint64_t* ck_addr;
     uint32_t* ck_len;
 };
-enum : uint8_t {
-    PF_HAS_ROW = 1, PF_LIVE_TS = 2, PF_ROW_DEL = 4, PF_HAS_CELL = 8,
-    PF_CELL_VALUE = 16, PF_CELL_EXPIRING = 32
-};
+enum : uint8_t { PF_HAS_ROW = 1, PF_LIVE_TS = 2, PF_ROW_DEL = 4 };
+// per-cell flags (cell_flags array)
+enum : uint8_t { CELLF_PRESENT = 1, CELLF_HAS_VALUE = 2, CELLF_EXPIRING = 4 };
 // ClusteringPrefix.Kind ordinals (ClusteringPrefix.java:65-85)
 enum : uint8_t {
     BK_EXCL_END = 0, BK_INCL_START = 1, BK_EXCL_END_INCL_START = 2, BK_STATIC = 3,

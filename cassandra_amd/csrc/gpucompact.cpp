@@ -651,21 +651,24 @@ struct OutPartsBuf {
 // allocate an UnfCols arena for n unfiltereds
 struct UnfColsBuf {
     DevBuf ck, rkind, flags, live_ts, live_ttl, live_let, rdel_mfda, rdel_ldt,
-        start_mfda, start_ldt, cell_ts, cell_ldt, cell_ttl, val_addr, val_len, ck_addr, ck_len;
+        start_mfda, start_ldt, cell_ts, cell_ldt, cell_ttl, val_addr, val_len, ck_addr, ck_len,
+        cell_flags;
     UnfCols uc{};
-    void alloc(uint64_t n) {
+    void alloc(uint64_t n, uint32_t n_cols) {
         if (!n) n = 1;
+        uint64_t nc = n * n_cols;
         ck.alloc(n * 8); rkind.alloc(n); flags.alloc(n); live_ts.alloc(n * 8);
         live_ttl.alloc(n * 4); live_let.alloc(n * 8); rdel_mfda.alloc(n * 8);
         rdel_ldt.alloc(n * 4); start_mfda.alloc(n * 8); start_ldt.alloc(n * 4);
-        cell_ts.alloc(n * 8); cell_ldt.alloc(n * 4); cell_ttl.alloc(n * 4);
-        val_addr.alloc(n * 8); val_len.alloc(n * 4);
+        cell_ts.alloc(nc * 8); cell_ldt.alloc(nc * 4); cell_ttl.alloc(nc * 4);
+        val_addr.alloc(nc * 8); val_len.alloc(nc * 4); cell_flags.alloc(nc);
         ck_addr.alloc(n * 8); ck_len.alloc(n * 4);
         uc = UnfCols{ck.as<uint64_t>(), rkind.as<uint8_t>(), flags.as<uint8_t>(),
                      live_ts.as<int64_t>(), live_ttl.as<int32_t>(), live_let.as<int64_t>(),
                      rdel_mfda.as<int64_t>(), rdel_ldt.as<uint32_t>(), start_mfda.as<int64_t>(),
                      start_ldt.as<uint32_t>(), cell_ts.as<int64_t>(), cell_ldt.as<uint32_t>(),
                      cell_ttl.as<int32_t>(), val_addr.as<uint64_t>(), val_len.as<uint32_t>(),
+                     cell_flags.as<uint8_t>(),
                      ck_addr.as<uint64_t>(), ck_len.as<uint32_t>()};
     }
 };
@@ -706,7 +709,8 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
     {
         uint64_t blocks = (n_groups + 255) / 256;
         hipLaunchKernelGGL(k_collect_rows, dim3((uint32_t)blocks), dim3(256), 0, stream, opb.op,
-                           rows.uc, n_groups, d_stats.as<OutStats>(), d_tomb.as<uint32_t>(), tomb_cap);
+                           rows.uc, n_groups, sp.sch.n_cols, d_stats.as<OutStats>(),
+                           d_tomb.as<uint32_t>(), tomb_cap);
     }
 
     // ---- sizes + scans ----
@@ -1105,8 +1109,8 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
             if (!comp_pin[s]) throw std::runtime_error("pinned alloc failed");
             if (stats[s].clustering_types.size() > 1)
                 throw std::runtime_error("at most one clustering column supported");
-            if (stats[s].regular_cols.size() != 1)
-                throw std::runtime_error("exactly one regular column supported in round 1");
+            if (stats[s].regular_cols.empty() || stats[s].regular_cols.size() > 63)
+                throw std::runtime_error("1..63 regular columns supported");
             if (!stats[s].partitioner.empty() && stats[s].partitioner.find("Murmur3") == std::string::npos)
                 throw std::runtime_error("Murmur3Partitioner required");
             // generation from ".../oa-<id>-big"
@@ -1116,11 +1120,18 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
             generations[s] = std::stoull(name.substr(a + 1, b2 - a - 1));
             res->input_uncompressed_bytes += cinfos[s].data_len;
         }
-        // column type of the single regular column
-        std::string col_type = stats[0].regular_cols[0].second;
-        int32_t col_fixed = -1;
-        if (col_type == "org.apache.cassandra.db.marshal.LongType") col_fixed = 8;
-        else if (col_type == "org.apache.cassandra.db.marshal.Int32Type") col_fixed = 4;
+        // per-column fixed widths (−1 = variable) from the header types
+        std::vector<int32_t> col_fixed_h;
+        for (auto& [nm, ct] : stats[0].regular_cols) {
+            (void)nm;
+            if (ct == "org.apache.cassandra.db.marshal.LongType") col_fixed_h.push_back(8);
+            else if (ct == "org.apache.cassandra.db.marshal.Int32Type") col_fixed_h.push_back(4);
+            else if (ct == "org.apache.cassandra.db.marshal.BytesType" ||
+                     ct == "org.apache.cassandra.db.marshal.UTF8Type" ||
+                     ct == "org.apache.cassandra.db.marshal.AsciiType")
+                col_fixed_h.push_back(-1);
+            else throw std::runtime_error("unsupported column type " + ct);
+        }
         // clustering column (0 or 1, fixed width)
         int32_t ck_width = 0;
         std::string ck_type_str;
@@ -1232,7 +1243,12 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
         if (total_parts > 0xFFFFFFFFull) throw std::runtime_error("too many partitions for one job");
         SchemaParams sch{};
         sch.ck_width = ck_width;
-        sch.col_fixed_len = col_fixed;
+        sch.n_cols = (uint32_t)col_fixed_h.size();
+        DevBuf d_col_fixed;
+        d_col_fixed.alloc(col_fixed_h.size() * 4);
+        HIP_CHECK(hipMemcpyAsync(d_col_fixed.p, col_fixed_h.data(), col_fixed_h.size() * 4,
+                                 hipMemcpyHostToDevice, stream));
+        sch.col_fixed = d_col_fixed.as<int32_t>();
         sch.column_index_size = 64 * 1024;
         DevBuf d_srcs, d_recs_a, d_recs_b, d_rows_in;
         d_srcs.alloc(srcs.size() * sizeof(SrcDesc2));
@@ -1260,7 +1276,7 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
         }
         uint64_t total_in_rows = exscan_u64(pc.row_base, total_parts, stream);
         UnfColsBuf in_rows;
-        in_rows.alloc(total_in_rows);
+        in_rows.alloc(total_in_rows, sch.n_cols);
         {
             uint32_t blocks = (uint32_t)((total_parts + 255) / 256);
             hipLaunchKernelGGL(k_parse_rows, dim3(blocks), dim3(256), 0, stream,
@@ -1339,7 +1355,7 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
         OutPartsBuf opb;
         opb.alloc(n_groups);
         UnfColsBuf out_rows;
-        out_rows.alloc(total_out_rows);
+        out_rows.alloc(total_out_rows, sch.n_cols);
         DevBuf d_stats, d_tomb;
         d_stats.alloc(sizeof(OutStats));
         init_outstats(d_stats, stream);
@@ -1491,6 +1507,9 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
             gp.base_ldt = spec->base_ldt;
             gp.key_len = spec->key_len ? spec->key_len : 8;
             gp.ck_text = spec->ck_text;
+            gp.n_value_cols = spec->n_value_cols ? spec->n_value_cols : 1;
+            gp.col_missing_pct = spec->col_missing_pct;
+            if (gp.n_value_cols > 63) throw std::runtime_error("n_value_cols must be 1..63");
             if (gp.ck_text && (uint64_t)gp.clustering_rows * 16 >= 100000000ull)
                 throw std::runtime_error("ck_text needs clustering_rows*16 < 1e8 (8-digit order)");
             if (gp.key_len < 8 || gp.key_len > 255)
@@ -1516,8 +1535,8 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
                                d_keys.as<uint8_t>());
             uint64_t total_rows = exscan_u64(d_prows.as<uint64_t>(), R, stream);
             UnfColsBuf rows;
-            rows.alloc(total_rows);
-            d_vals.alloc(total_rows * (uint64_t)spec->value_len);
+            rows.alloc(total_rows, gp.n_value_cols);
+            d_vals.alloc(total_rows * (uint64_t)gp.n_value_cols * spec->value_len);
             DevBuf d_ckarena;
             if (gp.ck_text) d_ckarena.alloc(total_rows * 16 + 16);
             hipLaunchKernelGGL(k_gen_fill2, dim3(blocks), dim3(256), 0, stream, gp, d_sorted,
@@ -1532,7 +1551,8 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
             d_tomb.alloc((uint64_t)tomb_cap * 4);
             {
                 hipLaunchKernelGGL(k_collect_rows, dim3(blocks), dim3(256), 0, stream, opb.op,
-                                   rows.uc, R, d_stats.as<OutStats>(), d_tomb.as<uint32_t>(), tomb_cap);
+                                   rows.uc, R, gp.n_value_cols, d_stats.as<OutStats>(),
+                                   d_tomb.as<uint32_t>(), tomb_cap);
             }
             OutStats hs0;
             HIP_CHECK(hipStreamSynchronize(stream));
@@ -1545,14 +1565,28 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
             if (sp.hs.min_ldt == NO_DELETION_TIME) sp.hs.min_ldt = DELETION_TIME_EPOCH;
             sp.hs.min_ttl = 0;
             sp.sch.ck_width = spec->clustering_rows ? (gp.ck_text ? -1 : 8) : 0;
-            sp.sch.col_fixed_len = -1;  // val blob
+            sp.sch.n_cols = gp.n_value_cols;
+            std::vector<int32_t> gcf(gp.n_value_cols, -1);  // val blobs
+            DevBuf d_gcf;
+            d_gcf.alloc(gcf.size() * 4);
+            HIP_CHECK(hipMemcpyAsync(d_gcf.p, gcf.data(), gcf.size() * 4,
+                                     hipMemcpyHostToDevice, stream));
+            sp.sch.col_fixed = d_gcf.as<int32_t>();
             sp.sch.column_index_size = 64 * 1024;
             // reset and recollect so the writer sees fresh stats (collect ran
             // once above only to derive the header mins)
             init_outstats(d_stats, stream);
             std::string base = std::string(dir) + "/oa-" + std::to_string(spec->first_generation + s) + "-big";
-            std::vector<std::pair<bytes, std::string>> cols = {
-                {bytes{'v', 'a', 'l'}, "org.apache.cassandra.db.marshal.BytesType"}};
+            std::vector<std::pair<bytes, std::string>> cols;
+            if (gp.n_value_cols == 1) {
+                cols.push_back({bytes{'v', 'a', 'l'}, "org.apache.cassandra.db.marshal.BytesType"});
+            } else {
+                for (uint32_t c = 0; c < gp.n_value_cols; c++) {
+                    std::string nm = "val" + std::to_string(c);
+                    cols.push_back({bytes(nm.begin(), nm.end()),
+                                    "org.apache.cassandra.db.marshal.BytesType"});
+                }
+            }
             write_sstable_device(opb, rows, R, sp, d_stats, d_tomb, tomb_cap, base,
                                  gp.key_len > 8 ? "org.apache.cassandra.db.marshal.BytesType"
                                                 : "org.apache.cassandra.db.marshal.LongType",

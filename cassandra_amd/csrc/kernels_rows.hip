@@ -27,7 +27,8 @@ __device__ inline void tomb_push(OutStats* st, uint32_t* ldts, uint32_t cap, uin
 // schema/runtime constants shared by the general kernels
 struct SchemaParams {
     int32_t ck_width;        // 0 = none; 4/8 fixed width; -1 variable (UTF8/Bytes)
-    int32_t col_fixed_len;   // regular column: -1 variable else fixed width
+    uint32_t n_cols;         // regular columns (1..63; header superset order)
+    const int32_t* col_fixed;  // per column: -1 variable else fixed width
     uint32_t column_index_size;  // promoted-index granularity (64 KiB default)
 };
 
@@ -229,11 +230,7 @@ __global__ void k_parse_rows(const SrcDesc2* srcs, uint32_t n_srcs, uint32_t tot
             rc.live_ts[o] = NO_TIMESTAMP;
             rc.live_ttl[o] = 0;
             rc.live_let[o] = NO_DELETION_TIME;
-            rc.cell_ts[o] = NO_TIMESTAMP;
-            rc.cell_ldt[o] = LDT_NONE_U32;
-            rc.cell_ttl[o] = 0;
-            rc.val_addr[o] = 0;
-            rc.val_len[o] = 0;
+            for (uint32_t c = 0; c < sp.n_cols; c++) rc.cell_flags[o * sp.n_cols + c] = 0;
         } else {
             // ---- row ----
             uint64_t ck = 0, ckaddr = 0;
@@ -256,11 +253,6 @@ __global__ void k_parse_rows(const SrcDesc2* srcs, uint32_t n_srcs, uint32_t tot
             int64_t lts = NO_TIMESTAMP, llet = NO_DELETION_TIME, rdm = INT64_MIN;
             int32_t lttl = 0;
             uint32_t rdl = LDT_NONE_U32;
-            int64_t cts = NO_TIMESTAMP;
-            uint32_t cldt = LDT_NONE_U32;
-            int32_t cttl = 0;
-            uint64_t vaddr = 0;
-            uint32_t vlen = 0;
             if (flags & 0x04) { pf |= PF_LIVE_TS; lts = (int64_t)uvint_get(base, &pos) + sd.min_ts; }
             if (flags & 0x08) {
                 lttl = (int32_t)(uint32_t)uvint_get(base, &pos) + sd.min_ttl;
@@ -271,27 +263,40 @@ __global__ void k_parse_rows(const SrcDesc2* srcs, uint32_t n_srcs, uint32_t tot
                 rdm = (int64_t)uvint_get(base, &pos) + sd.min_ts;
                 rdl = ldt_u32((int64_t)(int32_t)(uint32_t)uvint_get(base, &pos) + sd.min_ldt);
             }
-            bool has_cell;
-            if (flags & 0x20) has_cell = true;
-            else has_cell = !(uvint_get(base, &pos) & 1);
-            if (has_cell) {
-                pf |= PF_HAS_CELL;
+            // columns subset (Columns.serializeSubset: vint bitmap of MISSING)
+            uint64_t missing = 0;
+            if (!(flags & 0x20)) missing = uvint_get(base, &pos);
+            for (uint32_t c = 0; c < sp.n_cols; c++) {
+                uint64_t oc = o * sp.n_cols + c;
+                if (missing & (1ULL << c)) {
+                    rc.cell_flags[oc] = 0;
+                    continue;
+                }
+                uint8_t cfl = CELLF_PRESENT;
                 uint8_t cf = base[pos++];
-                cts = (cf & 8) ? lts : (int64_t)uvint_get(base, &pos) + sd.min_ts;
+                int64_t cts = (cf & 8) ? lts : (int64_t)uvint_get(base, &pos) + sd.min_ts;
                 bool dead = cf & 1, exp = cf & 2;
                 int64_t ldtl;
                 if (cf & 16) ldtl = llet;
                 else if (dead || exp) ldtl = (int64_t)(int32_t)(uint32_t)uvint_get(base, &pos) + sd.min_ldt;
                 else ldtl = NO_DELETION_TIME;
-                cttl = (cf & 16) ? lttl : (exp ? (int32_t)(uint32_t)uvint_get(base, &pos) + sd.min_ttl : 0);
-                cldt = ldt_u32(ldtl);
-                if (exp) pf |= PF_CELL_EXPIRING;
+                int32_t cttl = (cf & 16) ? lttl : (exp ? (int32_t)(uint32_t)uvint_get(base, &pos) + sd.min_ttl : 0);
+                if (exp) cfl |= CELLF_EXPIRING;
+                uint64_t vaddr = 0;
+                uint32_t vlen = 0;
                 if (!(cf & 4)) {
-                    pf |= PF_CELL_VALUE;
-                    vlen = sp.col_fixed_len >= 0 ? (uint32_t)sp.col_fixed_len : (uint32_t)uvint_get(base, &pos);
+                    cfl |= CELLF_HAS_VALUE;
+                    int32_t fw = sp.col_fixed[c];
+                    vlen = fw >= 0 ? (uint32_t)fw : (uint32_t)uvint_get(base, &pos);
                     vaddr = (uint64_t)(base + pos);
                     pos += vlen;
                 }
+                rc.cell_flags[oc] = cfl;
+                rc.cell_ts[oc] = cts;
+                rc.cell_ldt[oc] = ldt_u32(ldtl);
+                rc.cell_ttl[oc] = cttl;
+                rc.val_addr[oc] = vaddr;
+                rc.val_len[oc] = vlen;
             }
             rc.rkind[o] = BK_CLUSTERING;
             rc.ck[o] = ck;
@@ -305,11 +310,6 @@ __global__ void k_parse_rows(const SrcDesc2* srcs, uint32_t n_srcs, uint32_t tot
             rc.rdel_ldt[o] = rdl;
             rc.start_mfda[o] = INT64_MIN;
             rc.start_ldt[o] = LDT_NONE_U32;
-            rc.cell_ts[o] = cts;
-            rc.cell_ldt[o] = cldt;
-            rc.cell_ttl[o] = cttl;
-            rc.val_addr[o] = vaddr;
-            rc.val_len[o] = vlen;
             atomicAdd(rows_in, 1ull);
         }
         emitted++;
@@ -368,6 +368,16 @@ __device__ inline bool should_purge2(const PurgeParams2& pp, int64_t token, int6
 }
 __device__ inline bool dt_sup(int64_t am, uint32_t al, int64_t bm, uint32_t bl) {
     return am > bm || (am == bm && ldt_long(al) > ldt_long(bl));
+}
+
+// Cell.compareValues tie-break: unsigned lexicographic, shorter-first
+__device__ inline int cmp_values(uint64_t la, uint32_t ll, uint64_t ra, uint32_t rl) {
+    const uint8_t* lp = (const uint8_t*)la;
+    const uint8_t* rp = (const uint8_t*)ra;
+    uint32_t nn = ll < rl ? ll : rl;
+    for (uint32_t x = 0; x < nn; x++)
+        if (lp[x] != rp[x]) return lp[x] < rp[x] ? -1 : 1;
+    return ll == rl ? 0 : (ll < rl ? -1 : 1);
 }
 
 #define GPUC_MAX_ARITY 64
@@ -446,10 +456,14 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
     uint32_t ocount = 0;
     uint64_t cur_ck_addr = 0;
     uint32_t cur_ck_len = 0;
+    // cells_written: the row path fills out.cell_* at slot obase+ocount
+    // BEFORE calling emit; markers pass false and get all-absent cells
     auto emit = [&](uint8_t kind, uint64_t ck, uint8_t flags, int64_t lts, int32_t lttl,
                     int64_t llet, int64_t rdm, uint32_t rdl, int64_t smf, uint32_t sld,
-                    int64_t cts, uint32_t cldt, int32_t cttl, uint64_t va, uint32_t vl) {
+                    bool cells_written) {
         uint64_t o = obase + ocount++;
+        if (!cells_written)
+            for (uint32_t c = 0; c < sp.n_cols; c++) out.cell_flags[o * sp.n_cols + c] = 0;
         out.rkind[o] = kind;
         out.ck[o] = ck;
         out.ck_addr[o] = cur_ck_addr;
@@ -462,11 +476,6 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
         out.rdel_ldt[o] = rdl;
         out.start_mfda[o] = smf;
         out.start_ldt[o] = sld;
-        out.cell_ts[o] = cts;
-        out.cell_ldt[o] = cldt;
-        out.cell_ttl[o] = cttl;
-        out.val_addr[o] = va;
-        out.val_len[o] = vl;
     };
 
     // purge helpers
@@ -523,19 +532,26 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
             int64_t lts = NO_TIMESTAMP, llet = NO_DELETION_TIME, rdm = INT64_MIN;
             int32_t lttl = 0;
             uint32_t rdl = LDT_NONE_U32;
-            int64_t cts = NO_TIMESTAMP;
-            uint32_t cldt = LDT_NONE_U32;
-            int32_t cttl = 0;
-            uint64_t va = 0;
-            uint32_t vl = 0;
-            bool have_cell = false, cell_val = false, cell_exp = false;
+            const uint32_t NC = sp.n_cols;
+            uint64_t oslot = obase + ocount;  // cells written here pre-emit
+            bool any_cell = false;
             if (k == 1 || (nmem == 1 && active_live)) {
                 uint64_t o = mb[lastm] + mpos[lastm];
                 of = in.flags[o];
                 lts = in.live_ts[o]; lttl = in.live_ttl[o]; llet = in.live_let[o];
                 rdm = in.rdel_mfda[o]; rdl = in.rdel_ldt[o];
-                cts = in.cell_ts[o]; cldt = in.cell_ldt[o]; cttl = in.cell_ttl[o];
-                va = in.val_addr[o]; vl = in.val_len[o];
+                for (uint32_t c = 0; c < NC; c++) {
+                    uint64_t ic = o * NC + c, ocx = oslot * NC + c;
+                    uint8_t cfl = in.cell_flags[ic];
+                    out.cell_flags[ocx] = cfl;
+                    if (!(cfl & CELLF_PRESENT)) continue;
+                    any_cell = true;
+                    out.cell_ts[ocx] = in.cell_ts[ic];
+                    out.cell_ldt[ocx] = in.cell_ldt[ic];
+                    out.cell_ttl[ocx] = in.cell_ttl[ic];
+                    out.val_addr[ocx] = in.val_addr[ic];
+                    out.val_len[ocx] = in.val_len[ic];
+                }
             } else {
                 bool has_live = false;
                 for (uint32_t m = 0; m < k; m++) {
@@ -568,64 +584,66 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
                 if (!row_del_kept) { rdm = INT64_MIN; rdl = LDT_NONE_U32; }
                 else of |= PF_ROW_DEL;
                 if (has_live && lts <= am2) { of &= ~PF_LIVE_TS; lts = NO_TIMESTAMP; lttl = 0; llet = NO_DELETION_TIME; }
-                for (uint32_t m = 0; m < k; m++) {
-                    if (!(members & (1ULL << m))) continue;
-                    uint64_t o = mb[m] + mpos[m];
-                    uint8_t f = in.flags[o];
-                    if (!(f & PF_HAS_CELL)) continue;
-                    int64_t ts2 = in.cell_ts[o];
-                    if (ts2 <= am2) continue;
-                    if (!have_cell) {
-                        have_cell = true;
-                        cts = ts2; cldt = in.cell_ldt[o]; cttl = in.cell_ttl[o];
-                        va = in.val_addr[o]; vl = in.val_len[o];
-                        cell_val = f & PF_CELL_VALUE;
-                        cell_exp = f & PF_CELL_EXPIRING;
-                        continue;
-                    }
-                    bool take_right = false;
-                    uint32_t rl = in.cell_ldt[o];
-                    bool l_dt = cldt != LDT_NONE_U32, r_dt = rl != LDT_NONE_U32;
-                    if (cts != ts2) take_right = ts2 > cts;
-                    else if (l_dt || r_dt) {
-                        if (l_dt != r_dt) take_right = r_dt;
-                        else {
-                            bool l_tomb = !cell_exp, r_tomb = !(f & PF_CELL_EXPIRING);
-                            if (l_tomb != r_tomb) take_right = r_tomb;
-                            else if (cldt != rl) take_right = ldt_long(rl) > ldt_long(cldt);
-                            else {
-                                // compareValues (unsigned lexicographic)
-                                const uint8_t* lp = (const uint8_t*)va;
-                                const uint8_t* rp = (const uint8_t*)in.val_addr[o];
-                                uint32_t nn = min(vl, in.val_len[o]);
-                                int c = 0;
-                                for (uint32_t x = 0; x < nn; x++)
-                                    if (lp[x] != rp[x]) { c = lp[x] < rp[x] ? -1 : 1; break; }
-                                if (c == 0) c = vl == in.val_len[o] ? 0 : (vl < in.val_len[o] ? -1 : 1);
-                                take_right = c < 0;
-                            }
+                // per-column Cells.reconcile (Cells.java:145-179) + activeDeletion filter
+                for (uint32_t c = 0; c < NC; c++) {
+                    int64_t cts = NO_TIMESTAMP;
+                    uint32_t cldt = LDT_NONE_U32;
+                    int32_t cttl = 0;
+                    uint64_t va = 0;
+                    uint32_t vl = 0;
+                    bool have_cell = false, cell_val = false, cell_exp = false;
+                    for (uint32_t m = 0; m < k; m++) {
+                        if (!(members & (1ULL << m))) continue;
+                        uint64_t o = (mb[m] + mpos[m]) * NC + c;
+                        uint8_t f = in.cell_flags[o];
+                        if (!(f & CELLF_PRESENT)) continue;
+                        int64_t ts2 = in.cell_ts[o];
+                        if (ts2 <= am2) continue;
+                        if (!have_cell) {
+                            have_cell = true;
+                            cts = ts2; cldt = in.cell_ldt[o]; cttl = in.cell_ttl[o];
+                            va = in.val_addr[o]; vl = in.val_len[o];
+                            cell_val = f & CELLF_HAS_VALUE;
+                            cell_exp = f & CELLF_EXPIRING;
+                            continue;
                         }
-                    } else {
-                        const uint8_t* lp = (const uint8_t*)va;
-                        const uint8_t* rp = (const uint8_t*)in.val_addr[o];
-                        uint32_t nn = min(vl, in.val_len[o]);
-                        int c = 0;
-                        for (uint32_t x = 0; x < nn; x++)
-                            if (lp[x] != rp[x]) { c = lp[x] < rp[x] ? -1 : 1; break; }
-                        if (c == 0) c = vl == in.val_len[o] ? 0 : (vl < in.val_len[o] ? -1 : 1);
-                        take_right = c < 0;
+                        bool take_right = false;
+                        uint32_t rl = in.cell_ldt[o];
+                        bool l_dt = cldt != LDT_NONE_U32, r_dt = rl != LDT_NONE_U32;
+                        if (cts != ts2) take_right = ts2 > cts;
+                        else if (l_dt || r_dt) {
+                            if (l_dt != r_dt) take_right = r_dt;
+                            else {
+                                bool l_tomb = !cell_exp, r_tomb = !(f & CELLF_EXPIRING);
+                                if (l_tomb != r_tomb) take_right = r_tomb;
+                                else if (cldt != rl) take_right = ldt_long(rl) > ldt_long(cldt);
+                                else take_right = cmp_values(va, vl, in.val_addr[o], in.val_len[o]) < 0;
+                            }
+                        } else {
+                            take_right = cmp_values(va, vl, in.val_addr[o], in.val_len[o]) < 0;
+                        }
+                        if (take_right) {
+                            cts = ts2; cldt = rl; cttl = in.cell_ttl[o];
+                            va = in.val_addr[o]; vl = in.val_len[o];
+                            cell_val = f & CELLF_HAS_VALUE;
+                            cell_exp = f & CELLF_EXPIRING;
+                        }
                     }
-                    if (take_right) {
-                        cts = ts2; cldt = rl; cttl = in.cell_ttl[o];
-                        va = in.val_addr[o]; vl = in.val_len[o];
-                        cell_val = f & PF_CELL_VALUE;
-                        cell_exp = f & PF_CELL_EXPIRING;
+                    uint64_t ocx = oslot * NC + c;
+                    if (have_cell) {
+                        any_cell = true;
+                        out.cell_flags[ocx] = CELLF_PRESENT | (cell_val ? CELLF_HAS_VALUE : 0) |
+                                              (cell_exp ? CELLF_EXPIRING : 0);
+                        out.cell_ts[ocx] = cts;
+                        out.cell_ldt[ocx] = cldt;
+                        out.cell_ttl[ocx] = cttl;
+                        out.val_addr[ocx] = va;
+                        out.val_len[ocx] = vl;
+                    } else {
+                        out.cell_flags[ocx] = 0;
                     }
                 }
-                if (have_cell) of |= PF_HAS_CELL;
-                if (cell_val) of |= PF_CELL_VALUE;
-                if (cell_exp) of |= PF_CELL_EXPIRING;
-                if (of & (PF_LIVE_TS | PF_ROW_DEL | PF_HAS_CELL)) of |= PF_HAS_ROW;
+                if ((of & (PF_LIVE_TS | PF_ROW_DEL)) || any_cell) of |= PF_HAS_ROW;
                 else of = 0;
             }
             // ---- purge the merged row (BTreeRow.purge + AbstractCell.purge) ----
@@ -635,23 +653,34 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
                     if (!is_live && should_purge2(pp, token, lts, llet)) { of &= ~PF_LIVE_TS; lts = NO_TIMESTAMP; lttl = 0; llet = NO_DELETION_TIME; }
                 }
                 if ((of & PF_ROW_DEL) && purge_dt(rdm, rdl)) { of &= ~PF_ROW_DEL; rdm = INT64_MIN; rdl = LDT_NONE_U32; }
-                if (of & PF_HAS_CELL) {
+                any_cell = false;
+                for (uint32_t c = 0; c < NC; c++) {
+                    uint64_t ocx = oslot * NC + c;
+                    uint8_t cfl = out.cell_flags[ocx];
+                    if (!(cfl & CELLF_PRESENT)) continue;
+                    int64_t cts = out.cell_ts[ocx];
+                    uint32_t cldt = out.cell_ldt[ocx];
+                    int32_t cttl = out.cell_ttl[ocx];
                     bool live_cell = cldt == LDT_NONE_U32 || (cttl != 0 && pp.now_sec < ldt_long(cldt));
                     if (!live_cell) {
-                        if (should_purge2(pp, token, cts, ldt_long(cldt))) of &= ~(PF_HAS_CELL | PF_CELL_VALUE | PF_CELL_EXPIRING);
-                        else if (cttl != 0) {
+                        if (should_purge2(pp, token, cts, ldt_long(cldt))) { out.cell_flags[ocx] = 0; continue; }
+                        if (cttl != 0) {
                             int64_t nldt = ldt_long(cldt) - cttl;
-                            if (should_purge2(pp, token, cts, nldt)) of &= ~(PF_HAS_CELL | PF_CELL_VALUE | PF_CELL_EXPIRING);
-                            else { cldt = ldt_u32(nldt); cttl = 0; of &= ~(PF_CELL_VALUE | PF_CELL_EXPIRING); vl = 0; }
+                            if (should_purge2(pp, token, cts, nldt)) { out.cell_flags[ocx] = 0; continue; }
+                            out.cell_ldt[ocx] = ldt_u32(nldt);
+                            out.cell_ttl[ocx] = 0;
+                            out.cell_flags[ocx] = CELLF_PRESENT;  // expired -> tombstone, value dropped
+                            out.val_len[ocx] = 0;
                         }
                     }
+                    any_cell = true;
                 }
-                if (!(of & (PF_LIVE_TS | PF_ROW_DEL | PF_HAS_CELL))) of = 0;
+                if (!(of & (PF_LIVE_TS | PF_ROW_DEL)) && !any_cell) of = 0;
                 else if (pp.enforce_strict_liveness && !(of & PF_LIVE_TS) && !(of & PF_ROW_DEL)) of = 0;
             }
             if (of & PF_HAS_ROW)
                 emit(BK_CLUSTERING, fck, of, lts, lttl, llet, rdm, rdl, INT64_MIN, LDT_NONE_U32,
-                     cts, cldt, cttl, va, (of & PF_CELL_VALUE) ? vl : 0);
+                     true);
         } else if (k == 1) {
             // single-version partition: UnfilteredRowIterators.merge of one
             // iterator returns it unchanged — markers pass through as-is and
@@ -666,20 +695,17 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
                 if (purge_close && purge_open) {
                 } else if (purge_close) {
                     emit(kd == BK_EXCL_END_INCL_START ? BK_INCL_START : BK_EXCL_START, fck, 0,
-                         NO_TIMESTAMP, 0, NO_DELETION_TIME, s_m, s_l, INT64_MIN, LDT_NONE_U32,
-                         NO_TIMESTAMP, LDT_NONE_U32, 0, 0, 0);
+                         NO_TIMESTAMP, 0, NO_DELETION_TIME, s_m, s_l, INT64_MIN, LDT_NONE_U32, false);
                 } else if (purge_open) {
                     emit(kd == BK_EXCL_END_INCL_START ? BK_EXCL_END : BK_INCL_END, fck, 0,
-                         NO_TIMESTAMP, 0, NO_DELETION_TIME, e_m, e_l, INT64_MIN, LDT_NONE_U32,
-                         NO_TIMESTAMP, LDT_NONE_U32, 0, 0, 0);
+                         NO_TIMESTAMP, 0, NO_DELETION_TIME, e_m, e_l, INT64_MIN, LDT_NONE_U32, false);
                 } else {
-                    emit(kd, fck, 0, NO_TIMESTAMP, 0, NO_DELETION_TIME, e_m, e_l, s_m, s_l,
-                         NO_TIMESTAMP, LDT_NONE_U32, 0, 0, 0);
+                    emit(kd, fck, 0, NO_TIMESTAMP, 0, NO_DELETION_TIME, e_m, e_l, s_m, s_l, false);
                 }
             } else {
                 if (!purge_dt(e_m, e_l))
                     emit(kd, fck, 0, NO_TIMESTAMP, 0, NO_DELETION_TIME, e_m, e_l,
-                         INT64_MIN, LDT_NONE_U32, NO_TIMESTAMP, LDT_NONE_U32, 0, 0, 0);
+                         INT64_MIN, LDT_NONE_U32, false);
             }
         } else {
             // ---- marker event (RangeTombstoneMarker.Merger.merge) ----
@@ -728,20 +754,17 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
                     if (purge_close && purge_open) {
                     } else if (purge_close) {
                         emit(okind == BK_EXCL_END_INCL_START ? BK_INCL_START : BK_EXCL_START, fck, 0,
-                             NO_TIMESTAMP, 0, NO_DELETION_TIME, s_m, s_l, INT64_MIN, LDT_NONE_U32,
-                             NO_TIMESTAMP, LDT_NONE_U32, 0, 0, 0);
+                             NO_TIMESTAMP, 0, NO_DELETION_TIME, s_m, s_l, INT64_MIN, LDT_NONE_U32, false);
                     } else if (purge_open) {
                         emit(okind == BK_EXCL_END_INCL_START ? BK_EXCL_END : BK_INCL_END, fck, 0,
-                             NO_TIMESTAMP, 0, NO_DELETION_TIME, e_m, e_l, INT64_MIN, LDT_NONE_U32,
-                             NO_TIMESTAMP, LDT_NONE_U32, 0, 0, 0);
+                             NO_TIMESTAMP, 0, NO_DELETION_TIME, e_m, e_l, INT64_MIN, LDT_NONE_U32, false);
                     } else {
-                        emit(okind, fck, 0, NO_TIMESTAMP, 0, NO_DELETION_TIME, e_m, e_l, s_m, s_l,
-                             NO_TIMESTAMP, LDT_NONE_U32, 0, 0, 0);
+                        emit(okind, fck, 0, NO_TIMESTAMP, 0, NO_DELETION_TIME, e_m, e_l, s_m, s_l, false);
                     }
                 } else {
                     if (!purge_dt(e_m, e_l))
                         emit(okind, fck, 0, NO_TIMESTAMP, 0, NO_DELETION_TIME, e_m, e_l,
-                             INT64_MIN, LDT_NONE_U32, NO_TIMESTAMP, LDT_NONE_U32, 0, 0, 0);
+                             INT64_MIN, LDT_NONE_U32, false);
                 }
             }
         }
@@ -771,6 +794,35 @@ struct SerParams2 {
     HeaderStats hs;
     SchemaParams sch;
 };
+
+// one cell: serialized body size + flags byte (Cell.Serializer)
+__device__ inline uint32_t cell_body_size(const UnfCols& u, uint64_t oc, const SerParams2& sp,
+                                          int32_t fixed, bool live, bool exp_live, uint64_t o,
+                                          uint8_t* out_cf) {
+    uint8_t f = u.cell_flags[oc];
+    int64_t cts = u.cell_ts[oc];
+    uint32_t cldt = u.cell_ldt[oc];
+    int32_t cttl = u.cell_ttl[oc];
+    uint32_t vlen = u.val_len[oc];
+    bool has_value = vlen > 0 && (f & CELLF_HAS_VALUE);
+    bool deleted = cldt != LDT_NONE_U32 && cttl == NO_TTL;
+    bool expiring = cttl != NO_TTL;
+    bool use_row_ts = live && cts == u.live_ts[o];
+    bool use_row_ttl = expiring && exp_live && cttl == u.live_ttl[o] && ldt_long(cldt) == u.live_let[o];
+    uint8_t cflags = 0;
+    if (!has_value) cflags |= 4;
+    if (deleted) cflags |= 1;
+    else if (expiring) cflags |= 2;
+    if (use_row_ts) cflags |= 8;
+    if (use_row_ttl) cflags |= 16;
+    uint32_t body = 1;
+    if (!use_row_ts) body += uvint_size((uint64_t)(cts - sp.hs.min_ts));
+    if ((deleted || expiring) && !use_row_ttl) body += uvint_size(sext32(ldt_long(cldt) - sp.hs.min_ldt));
+    if (expiring && !use_row_ttl) body += uvint_size(sext32(cttl - sp.hs.min_ttl));
+    if (has_value) body += (fixed >= 0 ? 0 : uvint_size(vlen)) + vlen;
+    *out_cf = cflags;
+    return body;
+}
 
 // row/marker body size + flag bytes (mirrors the oracle serializers)
 __device__ inline uint32_t unf_body_size(const UnfCols& u, uint64_t o, const SerParams2& sp,
@@ -802,32 +854,21 @@ __device__ inline uint32_t unf_body_size(const UnfCols& u, uint64_t o, const Ser
         body += uvint_size((uint64_t)(u.rdel_mfda[o] - sp.hs.min_ts));
         body += uvint_size(sext32(ldt_long(u.rdel_ldt[o]) - sp.hs.min_ldt));
     }
-    if (f & PF_HAS_CELL) rflags |= 0x20;
-    else body += uvint_size(1);
-    uint8_t cflags = 0;
-    if (f & PF_HAS_CELL) {
-        int64_t cts = u.cell_ts[o];
-        uint32_t cldt = u.cell_ldt[o];
-        int32_t cttl = u.cell_ttl[o];
-        uint32_t vlen = u.val_len[o];
-        bool has_value = vlen > 0 && (f & PF_CELL_VALUE);
-        bool deleted = cldt != LDT_NONE_U32 && cttl == NO_TTL;
-        bool expiring = cttl != NO_TTL;
-        bool use_row_ts = live && cts == u.live_ts[o];
-        bool use_row_ttl = expiring && exp_live && cttl == u.live_ttl[o] && ldt_long(cldt) == u.live_let[o];
-        if (!has_value) cflags |= 4;
-        if (deleted) cflags |= 1;
-        else if (expiring) cflags |= 2;
-        if (use_row_ts) cflags |= 8;
-        if (use_row_ttl) cflags |= 16;
-        body += 1;
-        if (!use_row_ts) body += uvint_size((uint64_t)(cts - sp.hs.min_ts));
-        if ((deleted || expiring) && !use_row_ttl) body += uvint_size(sext32(ldt_long(cldt) - sp.hs.min_ldt));
-        if (expiring && !use_row_ttl) body += uvint_size(sext32(cttl - sp.hs.min_ttl));
-        if (has_value) body += (sp.sch.col_fixed_len >= 0 ? 0 : uvint_size(vlen)) + vlen;
+    // columns subset + per-cell sizes (cell_flags_byte shared with EMIT)
+    uint64_t present_mask = 0;
+    uint32_t present = 0;
+    for (uint32_t c = 0; c < sp.sch.n_cols; c++)
+        if (u.cell_flags[o * sp.sch.n_cols + c] & CELLF_PRESENT) { present_mask |= 1ULL << c; present++; }
+    if (present == sp.sch.n_cols) rflags |= 0x20;
+    else body += uvint_size(((1ULL << sp.sch.n_cols) - 1) & ~present_mask);
+    for (uint32_t c = 0; c < sp.sch.n_cols; c++) {
+        uint64_t oc = o * sp.sch.n_cols + c;
+        if (!(u.cell_flags[oc] & CELLF_PRESENT)) continue;
+        uint8_t cflags;
+        body += cell_body_size(u, oc, sp, sp.sch.col_fixed[c], live, exp_live, o, &cflags);
     }
     if (out_flags) *out_flags = rflags;
-    if (out_cflags) *out_cflags = cflags;
+    if (out_cflags) *out_cflags = 0;
     return body;
 }
 
@@ -1057,19 +1098,29 @@ __device__ uint64_t part_walk(const OutParts& op, const UnfCols& out, uint64_t g
                 emit_uv((uint64_t)(out.rdel_mfda[o] - sp.hs.min_ts));
                 emit_uv(sext32(ldt_long(out.rdel_ldt[o]) - sp.hs.min_ldt));
             }
-            if (!(rflags & 0x20)) emit_uv(1);
-            if (f & PF_HAS_CELL) {
-                emit8(cflags);
-                if (!(cflags & 8)) emit_uv((uint64_t)(out.cell_ts[o] - sp.hs.min_ts));
-                bool deleted = cflags & 1, expiring = cflags & 2;
-                if ((deleted || expiring) && !(cflags & 16))
-                    emit_uv(sext32(ldt_long(out.cell_ldt[o]) - sp.hs.min_ldt));
-                if (expiring && !(cflags & 16)) emit_uv(sext32(out.cell_ttl[o] - sp.hs.min_ttl));
-                if (!(cflags & 4)) {
-                    uint32_t vlen = out.val_len[o];
-                    if (sp.sch.col_fixed_len < 0) emit_uv(vlen);
+            if (!(rflags & 0x20)) {
+                uint64_t present_mask = 0;
+                for (uint32_t c = 0; c < sp.sch.n_cols; c++)
+                    if (out.cell_flags[o * sp.sch.n_cols + c] & CELLF_PRESENT) present_mask |= 1ULL << c;
+                emit_uv(((1ULL << sp.sch.n_cols) - 1) & ~present_mask);
+            }
+            bool live = rflags & 0x04, exp_live = rflags & 0x08;
+            for (uint32_t c = 0; c < sp.sch.n_cols; c++) {
+                uint64_t oc = o * sp.sch.n_cols + c;
+                if (!(out.cell_flags[oc] & CELLF_PRESENT)) continue;
+                uint8_t cfb;
+                cell_body_size(out, oc, sp, sp.sch.col_fixed[c], live, exp_live, o, &cfb);
+                emit8(cfb);
+                if (!(cfb & 8)) emit_uv((uint64_t)(out.cell_ts[oc] - sp.hs.min_ts));
+                bool deleted = cfb & 1, expiring = cfb & 2;
+                if ((deleted || expiring) && !(cfb & 16))
+                    emit_uv(sext32(ldt_long(out.cell_ldt[oc]) - sp.hs.min_ldt));
+                if (expiring && !(cfb & 16)) emit_uv(sext32(out.cell_ttl[oc] - sp.hs.min_ttl));
+                if (!(cfb & 4)) {
+                    uint32_t vlen = out.val_len[oc];
+                    if (sp.sch.col_fixed[c] < 0) emit_uv(vlen);
                     if (EMIT) {
-                        const uint8_t* src = (const uint8_t*)out.val_addr[o];
+                        const uint8_t* src = (const uint8_t*)out.val_addr[oc];
                         for (uint32_t i = (uint32_t)lane; i < vlen; i += WAVE)
                             out_data[data_off + pos + i] = src[i];
                     }
@@ -1178,7 +1229,8 @@ __global__ void k_sizes_rows(OutParts op, UnfCols out, uint64_t n, SerParams2 sp
             atomicAdd(&sh_ps[lo], 1u);
             uint64_t cells = 0;
             for (uint32_t j = 0; j < op.row_count[g]; j++)
-                if (out.flags[op.row_base[g] + j] & PF_HAS_CELL) cells++;
+                for (uint32_t c = 0; c < sp.sch.n_cols; c++)
+                    if (out.cell_flags[(op.row_base[g] + j) * sp.sch.n_cols + c] & CELLF_PRESENT) cells++;
             lo = 0; hi = ch_hist_n;
             while (lo < hi) { int mid = (lo + hi) >> 1; if ((uint64_t)ch_hist_off[mid] < cells) lo = mid + 1; else hi = mid; }
             atomicAdd(&sh_ch[lo], 1u);
@@ -1240,7 +1292,7 @@ __global__ void k_serialize_rows(OutParts op, UnfCols out, uint64_t n, SerParams
 }
 
 // output stats over the final OutParts/UnfCols (MetadataCollector semantics)
-__global__ void k_collect_rows(OutParts op, UnfCols out, uint64_t n, OutStats* st,
+__global__ void k_collect_rows(OutParts op, UnfCols out, uint64_t n, uint32_t n_cols, OutStats* st,
                                uint32_t* tomb_ldts, uint32_t tomb_cap) {
     __shared__ unsigned long long sh_parts, sh_rows, sh_cells, sh_mints, sh_maxts,
         sh_minldt, sh_maxldt, sh_haspdel;
@@ -1304,11 +1356,13 @@ __global__ void k_collect_rows(OutParts op, UnfCols out, uint64_t n, OutStats* s
                 lldt(ldt_long(out.rdel_ldt[o]));
                 tomb_push(st, tomb_ldts, tomb_cap, out.rdel_ldt[o]);
             }
-            if (of & PF_HAS_CELL) {
+            for (uint32_t c = 0; c < n_cols; c++) {
+                uint64_t oc = o * n_cols + c;
+                if (!(out.cell_flags[oc] & CELLF_PRESENT)) continue;
                 atomicAdd(&sh_cells, 1ull);
-                lts(out.cell_ts[o]);
-                uint32_t cldt = out.cell_ldt[o];
-                int32_t cttl = out.cell_ttl[o];
+                lts(out.cell_ts[oc]);
+                uint32_t cldt = out.cell_ldt[oc];
+                int32_t cttl = out.cell_ttl[oc];
                 if (cldt != LDT_NONE_U32 && cttl == 0) { lldt(ldt_long(cldt)); tomb_push(st, tomb_ldts, tomb_cap, cldt); }
                 else if (cttl != 0) {
                     lldt(ldt_long(cldt));
@@ -1346,6 +1400,8 @@ struct GenParams2 {
     uint32_t clustering_rows, range_tomb_pct;
     uint32_t key_len;   // 8 (default) .. 255; bytes 8.. are gen2_key_salt(id, j)
     uint32_t ck_text;   // clustering values as UTF8 strings (oracle gen_ck_bytes)
+    uint32_t n_value_cols;    // regular columns val0..valN-1 (1..63)
+    uint32_t col_missing_pct; // P(cell absent) per live row and column
     int64_t base_ts, base_ldt;
 };
 
@@ -1432,8 +1488,10 @@ __global__ void k_gen_fill2(GenParams2 gp, const MRec* sorted, const uint64_t* i
             out.ck_len[o] = 8;
         }
     };
+    const uint32_t NCV = gp.n_value_cols;
     auto put_marker = [&](uint8_t kind, int64_t ck, int64_t m, uint32_t l) {
         uint64_t o = ob + emitted++;
+        for (uint32_t c = 0; c < NCV; c++) out.cell_flags[o * NCV + c] = 0;
         out.rkind[o] = kind;
         put_ck(o, ck, true, false);
         out.flags[o] = 0;
@@ -1451,7 +1509,7 @@ __global__ void k_gen_fill2(GenParams2 gp, const MRec* sorted, const uint64_t* i
         out.val_len[o] = 0;
     };
     auto put_row = [&](int64_t ck, bool has_ck, int64_t ts, bool tomb, uint32_t tomb_ldt,
-                       uint64_t val_slot) {
+                       uint32_t rowj) {
         uint64_t o = ob + emitted++;
         out.rkind[o] = BK_CLUSTERING;
         put_ck(o, ck, has_ck, true);
@@ -1459,25 +1517,31 @@ __global__ void k_gen_fill2(GenParams2 gp, const MRec* sorted, const uint64_t* i
         out.start_ldt[o] = LDT_NONE_U32;
         out.live_ttl[o] = 0;
         out.live_let[o] = NO_DELETION_TIME;
-        out.cell_ttl[o] = 0;
         if (tomb) {
             out.flags[o] = PF_HAS_ROW | PF_ROW_DEL;
             out.live_ts[o] = NO_TIMESTAMP;
             out.rdel_mfda[o] = ts;
             out.rdel_ldt[o] = tomb_ldt;
-            out.cell_ts[o] = NO_TIMESTAMP;
-            out.cell_ldt[o] = LDT_NONE_U32;
-            out.val_addr[o] = 0;
-            out.val_len[o] = 0;
+            for (uint32_t c = 0; c < NCV; c++) out.cell_flags[o * NCV + c] = 0;
         } else {
-            out.flags[o] = PF_HAS_ROW | PF_LIVE_TS | PF_HAS_CELL | PF_CELL_VALUE;
+            out.flags[o] = PF_HAS_ROW | PF_LIVE_TS;
             out.live_ts[o] = ts;
             out.rdel_mfda[o] = INT64_MIN;
             out.rdel_ldt[o] = LDT_NONE_U32;
-            out.cell_ts[o] = ts;
-            out.cell_ldt[o] = LDT_NONE_U32;
-            out.val_addr[o] = (uint64_t)(values + val_slot * (uint64_t)gp.value_len);
-            out.val_len[o] = gp.value_len;
+            for (uint32_t c = 0; c < NCV; c++) {
+                uint64_t oc = o * NCV + c;
+                bool miss = gp.col_missing_pct &&
+                            splitmix64(gp.seed ^ 0xC011C011ULL ^ id ^ ((uint64_t)(rowj + 1) << 40) ^
+                                       ((uint64_t)(c + 1) << 56) ^ ((uint64_t)gp.sst << 32)) % 100 <
+                                gp.col_missing_pct;
+                if (miss) { out.cell_flags[oc] = 0; continue; }
+                out.cell_flags[oc] = CELLF_PRESENT | CELLF_HAS_VALUE;
+                out.cell_ts[oc] = ts;
+                out.cell_ldt[oc] = LDT_NONE_U32;
+                out.cell_ttl[oc] = 0;
+                out.val_addr[oc] = (uint64_t)(values + oc * (uint64_t)gp.value_len);
+                out.val_len[oc] = gp.value_len;
+            }
         }
     };
 
@@ -1491,7 +1555,7 @@ __global__ void k_gen_fill2(GenParams2 gp, const MRec* sorted, const uint64_t* i
         } else {
             op.pdel_mfda[i] = INT64_MIN;
             op.pdel_ldt[i] = LDT_NONE_U32;
-            put_row(0, false, ts, tomb, tomb ? gen2_ldt(gp, id, 0xEE) : 0, ob + 0);
+            put_row(0, false, ts, tomb, tomb ? gen2_ldt(gp, id, 0xEE) : 0, 0);
         }
     } else {
         op.pdel_mfda[i] = INT64_MIN;
@@ -1517,7 +1581,7 @@ __global__ void k_gen_fill2(GenParams2 gp, const MRec* sorted, const uint64_t* i
             }
             int64_t ts = gp.base_ts + (int64_t)(splitmix64(gp.seed ^ id * 31 ^ ((uint64_t)gp.sst << 48) ^ (uint64_t)(j + 1) * 0x9E37ULL) % 1000000000ULL);
             bool tomb = gp.tombstone_pct && (splitmix64(gp.seed ^ 0xDEADULL ^ id ^ ((uint64_t)gp.sst << 32) ^ (uint64_t)(j + 7) * 131) % 100 < gp.tombstone_pct);
-            put_row(ck, true, ts, tomb, tomb ? gen2_ldt(gp, id * 1000 + j, 0xEE) : 0, ob + j);
+            put_row(ck, true, ts, tomb, tomb ? gen2_ldt(gp, id * 1000 + j, 0xEE) : 0, j);
             if (has_rt && rt_open && j + 1 < gp.clustering_rows && ((int64_t)(j + 1) * 16) > rhi) {
                 put_marker(BK_INCL_END, rhi, rts, rldt);
                 rt_open = false;
@@ -1544,9 +1608,12 @@ __global__ void k_gen_values2(GenParams2 gp, const MRec* sorted, const uint64_t*
         uint64_t o = rb + u;
         if (out.rkind[o] != BK_CLUSTERING) continue;
         uint32_t j = rowj++;
-        if (!(out.flags[o] & PF_CELL_VALUE)) continue;
+        for (uint32_t c = 0; c < gp.n_value_cols; c++) {
+        uint64_t oc = o * gp.n_value_cols + c;
+        if (!(out.cell_flags[oc] & CELLF_HAS_VALUE)) continue;
         uint64_t seed_id = gp.clustering_rows == 0 ? id : (id ^ ((uint64_t)j << 52) ^ j);
-        uint8_t* outp = (uint8_t*)out.val_addr[o];
+        seed_id += (uint64_t)c * 0xA5A5A5A5A5A5A5ULL;
+        uint8_t* outp = (uint8_t*)out.val_addr[oc];
         uint64_t state = gp.seed ^ seed_id * 0x100000001B3ULL ^ ((uint64_t)gp.sst << 40);
         uint64_t prev = splitmix64(state);
         uint32_t nw = (gp.value_len + 7) / 8;
@@ -1557,6 +1624,7 @@ __global__ void k_gen_values2(GenParams2 gp, const MRec* sorted, const uint64_t*
             uint32_t off = w * 8;
             for (uint32_t b = 0; b < 8 && off + b < gp.value_len; b++)
                 outp[off + b] = (uint8_t)(word >> (8 * b));
+        }
         }
     }
     (void)values;

@@ -10,7 +10,14 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
     t.header.key_type = g.key_len > 8 ? CqlType::BYTES : CqlType::LONG;  // pk bigint / blob
     if (g.clustering_rows > 0)
         t.header.clustering_types = {g.ck_text ? CqlType::UTF8 : CqlType::LONG};
-    t.header.regular_cols = {{bytes{'v', 'a', 'l'}, CqlType::BYTES}};  // val blob
+    if (g.n_value_cols <= 1) {
+        t.header.regular_cols = {{bytes{'v', 'a', 'l'}, CqlType::BYTES}};  // val blob
+    } else {
+        for (uint32_t cc = 0; cc < g.n_value_cols; cc++) {
+            std::string nm = "val" + std::to_string(cc);
+            t.header.regular_cols.push_back({bytes(nm.begin(), nm.end()), CqlType::BYTES});
+        }
+    }
 
     struct Ent { int64_t token; bytes key; uint64_t id; };
     std::vector<Ent> ents;
@@ -65,7 +72,8 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
                 u.kind = Unfiltered::ROW;
                 Row& r = u.row;
                 r.clustering = {ClusterVal{ClusterVal::VALUE, gen_ck_bytes(g, ck, true)}};
-                r.cells.resize(1);
+                uint32_t ncols = g.n_value_cols ? g.n_value_cols : 1;
+                r.cells.resize(ncols);
                 int64_t ts = gen_row_ts(g, sst, e.id, j);
                 min_ts = std::min(min_ts, ts);
                 if (gen_row_is_tombstone(g, sst, e.id, j)) {
@@ -74,10 +82,14 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
                     min_ldt_l = std::min<int64_t>(min_ldt_l, r.del.ldt);
                 } else {
                     r.live.ts = ts;
-                    Cell cell;
-                    cell.ts = ts;
-                    gen_value(g, sst, e.id ^ ((uint64_t)j << 52) ^ j, cell.value);
-                    r.cells[0] = std::move(cell);
+                    uint64_t seed_id = e.id ^ ((uint64_t)j << 52) ^ j;
+                    for (uint32_t cc = 0; cc < ncols; cc++) {
+                        if (gen_col_missing(g, sst, e.id, j, cc)) continue;
+                        Cell cell;
+                        cell.ts = ts;
+                        gen_value(g, sst, gen_col_value_id(seed_id, cc), cell.value);
+                        r.cells[cc] = std::move(cell);
+                    }
                 }
                 p.items.push_back(std::move(u));
                 if (has_rt && rt_open && j + 1 < g.clustering_rows &&
@@ -118,17 +130,21 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
         Unfiltered u;
         u.kind = Unfiltered::ROW;
         Row& r = u.row;
-        r.cells.resize(1);
+        uint32_t ncols = g.n_value_cols ? g.n_value_cols : 1;
+        r.cells.resize(ncols);
         if (gen_is_tombstone(g, sst, e.id)) {
             r.del.mfda = ts;
             r.del.ldt = gen_ldt(g, e.id, 0xEE);
             min_ldt_l = std::min<int64_t>(min_ldt_l, r.del.ldt);
         } else {
             r.live.ts = ts;
-            Cell c;
-            c.ts = ts;
-            gen_value(g, sst, e.id, c.value);
-            r.cells[0] = std::move(c);
+            for (uint32_t cc = 0; cc < ncols; cc++) {
+                if (gen_col_missing(g, sst, e.id, 0, cc)) continue;
+                Cell c;
+                c.ts = ts;
+                gen_value(g, sst, gen_col_value_id(e.id, cc), c.value);
+                r.cells[cc] = std::move(c);
+            }
         }
         p.items.push_back(std::move(u));
         t.parts.push_back(std::move(p));

@@ -40,6 +40,11 @@ struct GenSpec {
     // bytes 8.. = splitmix64(id ^ (0xC0FFEE5EED + j)) (GPU: gen2_key_salt).
     // key_len > 8 switches the declared key type LongType -> BytesType.
     uint32_t key_len = 8;
+    // multi-column: n_value_cols regular blob columns named val0..valN-1
+    // (1 keeps the single column "val"); col_missing_pct drops cells per
+    // (row, column). GPU mirror: GenParams2.n_value_cols/col_missing_pct.
+    uint32_t n_value_cols = 1;
+    uint32_t col_missing_pct = 0;
 
     uint64_t stride() const { return rows_per_sstable * (100 - overlap_pct) / 100; }
     uint64_t universe() const {
@@ -79,6 +84,15 @@ inline bytes gen_key_bytes(const GenSpec& g, uint64_t id) {
 }
 inline int64_t gen_ts(const GenSpec& g, uint32_t sst, uint64_t key_id) {
     return g.base_ts + (int64_t)(splitmix64(g.seed ^ key_id * 31 ^ ((uint64_t)sst << 48)) % 1000000000ULL);
+}
+inline bool gen_col_missing(const GenSpec& g, uint32_t sst, uint64_t key_id, uint32_t j,
+                            uint32_t c) {
+    if (g.col_missing_pct == 0) return false;
+    return splitmix64(g.seed ^ 0xC011C011ULL ^ key_id ^ ((uint64_t)(j + 1) << 40) ^
+                      ((uint64_t)(c + 1) << 56) ^ ((uint64_t)sst << 32)) % 100 < g.col_missing_pct;
+}
+inline uint64_t gen_col_value_id(uint64_t seed_id, uint32_t c) {
+    return seed_id + (uint64_t)c * 0xA5A5A5A5A5A5A5ULL;
 }
 inline bool gen_is_tombstone(const GenSpec& g, uint32_t sst, uint64_t key_id) {
     if (g.tombstone_pct == 0) return false;

@@ -74,6 +74,18 @@ def test_generator_parity_wide(ca, oracle_bin, tmp_path):
         _assert_dirs_equal(f"{dg}/oa-{g}-big", f"{do}/oa-{g}-big")
 
 
+def test_generator_parity_multi_column(ca, oracle_bin, tmp_path):
+    """N regular columns + per-cell subset bitmaps: write-path parity."""
+    dg, do = str(tmp_path / "gpu"), str(tmp_path / "cpu")
+    os.makedirs(dg), os.makedirs(do)
+    kw = dict(seed=86, n=2, rows=1000, vlen=180, overlap=25, tomb=10, ncols=4, colmiss=25)
+    ca.generate(dg, seed=86, n_sstables=2, rows_per_sstable=1000, value_len=180,
+                overlap_pct=25, tombstone_pct=10, n_value_cols=4, col_missing_pct=25)
+    _oracle_gen(do, **kw)
+    for g in (1, 2):
+        _assert_dirs_equal(f"{dg}/oa-{g}-big", f"{do}/oa-{g}-big")
+
+
 def test_generator_parity_text_ck(ca, oracle_bin, tmp_path):
     """Variable-width (UTF8) clustering values: write-path parity."""
     dg, do = str(tmp_path / "gpu"), str(tmp_path / "cpu")
@@ -148,6 +160,17 @@ def test_generator_parity_long_keys(ca, oracle_bin, tmp_path):
     dict(name="text_ck_long_keys", gen=dict(seed=74, n=3, rows=40, crows=50, vlen=250,
                                             overlap=30, tomb=10, rtomb=25, cktext=1,
                                             keylen=32), job={}),
+    dict(name="multi_column", gen=dict(seed=82, n=4, rows=1500, vlen=200, overlap=30,
+                                       tomb=10, ncols=4, colmiss=25), job={}),
+    dict(name="multi_column_wide", gen=dict(seed=83, n=3, rows=40, crows=50, vlen=150,
+                                            overlap=30, tomb=10, rtomb=30, ncols=3,
+                                            colmiss=20), job={}),
+    dict(name="multi_column_gc", gen=dict(seed=84, n=3, rows=1200, vlen=120, overlap=40,
+                                          tomb=25, ncols=5, colmiss=30),
+         job=dict(gc_before=2000000000)),
+    dict(name="multi_column_everything", gen=dict(seed=85, n=3, rows=30, crows=40, vlen=180,
+                                                  overlap=30, tomb=12, rtomb=25, ncols=4,
+                                                  colmiss=20, cktext=1, keylen=24), job={}),
 ])
 def test_compaction_parity(ca, oracle_bin, tmp_path, case):
     d = str(tmp_path)

@@ -155,3 +155,14 @@ def test_text_clustering_roundtrip_and_merge(oracle_bin, tmp_path):
     assert stats["rows_out"] > 0
     r = oracle_run("roundtrip", f"{d}/oa-50-big")
     assert "FAIL" not in r.stdout, r.stdout
+
+
+def test_multi_column_roundtrip_and_merge(oracle_bin, tmp_path):
+    """N regular columns with per-cell subsets (Columns.serializeSubset)."""
+    d = str(tmp_path)
+    _gen(d, n=3, rows=600, vlen=150, overlap=30, tomb=10, ncols=4, colmiss=25, seed=81)
+    ins = [f"{d}/oa-{g}-big" for g in (1, 2, 3)]
+    stats = _compact(f"{d}/oa-50-big", ins)
+    assert stats["partitions_out"] > 0
+    r = oracle_run("roundtrip", f"{d}/oa-50-big")
+    assert "FAIL" not in r.stdout, r.stdout
