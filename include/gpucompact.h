@@ -4,9 +4,10 @@
  * Drop-in boundary: replaces the hot loop of CompactionTask.runMayThrow
  * (reference: src/java/org/apache/cassandra/db/compaction/CompactionTask.java:184-236)
  * — scanners + CompactionIterator + CompactionAwareWriter — for eligible
- * tables (big-format `oa`, Murmur3Partitioner, at most one fixed-width
- * clustering column, one regular column, LZ4 chunk compression; row, cell,
- * partition and range tombstones all supported). The Java host above the seam
+ * tables (big-format `oa`, Murmur3Partitioner, any partition key, at most one
+ * clustering column of fixed or variable width, 1..63 regular columns with
+ * cell subsets, LZ4 chunk compression; row, cell, partition and range
+ * tombstones all supported). The Java host above the seam
  * (strategies, CompactionManager, LifecycleTransaction, metrics) is
  * unchanged and binds these entry points via JNI/Panama.
  *
