@@ -144,6 +144,17 @@ GPUC_HD inline uint32_t crc32_update_t(uint32_t crc, const uint8_t* buf, size_t 
     for (size_t i = 0; i < len; i++) crc = table[(crc ^ buf[i]) & 0xFF] ^ (crc >> 8);
     return ~crc;
 }
+// slicing-by-8: t[k][b] = table-walk of byte b followed by k zero bytes
+inline void crc32_make_table8(uint32_t* t /* [8*256] */) {
+    for (uint32_t i = 0; i < 256; i++) {
+        uint32_t c = i;
+        for (int j = 0; j < 8; j++) c = (c & 1) ? 0xEDB88320u ^ (c >> 1) : c >> 1;
+        t[i] = c;
+    }
+    for (int k = 1; k < 8; k++)
+        for (uint32_t i = 0; i < 256; i++)
+            t[k * 256 + i] = t[(t[(k - 1) * 256 + i] & 0xFF)] ^ (t[(k - 1) * 256 + i] >> 8);
+}
 inline void crc32_make_table(uint32_t* table) {
     for (uint32_t i = 0; i < 256; i++) {
         uint32_t c = i;

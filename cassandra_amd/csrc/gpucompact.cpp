@@ -610,15 +610,17 @@ static void init_outstats(DevBuf& d_stats, hipStream_t stream) {
 }
 
 static bool g_crc_tables_ready = false;
-static void* g_crc256 = nullptr;  // 256-entry CRC table, device-resident
+static void* g_crc256 = nullptr;  // 8x256 sliced CRC tables (first 256 = base)
 static void ensure_crc_tables(hipStream_t stream) {
     if (g_crc_tables_ready) return;
     hipLaunchKernelGGL(k_crc_init, dim3(1), dim3(256), 0, stream);
     // plain 256-entry table for the wave LZ4 kernels
     uint32_t tab[256];
     crc32_make_table(tab);
-    HIP_CHECK(hipMalloc(&g_crc256, sizeof(tab)));
-    HIP_CHECK(hipMemcpyAsync(g_crc256, tab, sizeof(tab), hipMemcpyHostToDevice, stream));
+    static uint32_t tab8[8 * 256];
+    crc32_make_table8(tab8);
+    HIP_CHECK(hipMalloc(&g_crc256, sizeof(tab8)));
+    HIP_CHECK(hipMemcpyAsync(g_crc256, tab8, sizeof(tab8), hipMemcpyHostToDevice, stream));
     // verify the LDS same-address write-order rule the wave compressor relies on
     unsigned int* d_probe;
     HIP_CHECK(hipMalloc(&d_probe, 8));
@@ -1394,11 +1396,16 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
         HIP_CHECK(hipEventRecord(er0, stream));
         {
             uint32_t blocks = (uint32_t)((n_groups + 255) / 256);
-            hipLaunchKernelGGL(k_reconcile_rows, dim3(blocks), dim3(256), 0, stream, d_sorted,
-                               d_gstart.as<uint64_t>(), n_groups, total_parts,
-                               d_srcbases.as<uint32_t>(), pc, in_rows.uc, opb.op, out_rows.uc,
-                               d_group_rows.as<uint64_t>(), sch, pp, d_stats.as<OutStats>(),
-                               d_error.as<unsigned long long>());
+            auto launch_rec = [&](auto kern) {
+                hipLaunchKernelGGL(kern, dim3(blocks), dim3(256), 0, stream, d_sorted,
+                                   d_gstart.as<uint64_t>(), n_groups, total_parts,
+                                   d_srcbases.as<uint32_t>(), pc, in_rows.uc, opb.op, out_rows.uc,
+                                   d_group_rows.as<uint64_t>(), sch, pp, d_stats.as<OutStats>(),
+                                   d_error.as<unsigned long long>());
+            };
+            if (k <= 8) launch_rec(k_reconcile_rows<8>);
+            else if (k <= 16) launch_rec(k_reconcile_rows<16>);
+            else launch_rec(k_reconcile_rows<64>);
         }
         HIP_CHECK(hipEventRecord(er1, stream));
         HIP_CHECK(hipEventRecord(e4, stream));

@@ -382,6 +382,10 @@ __device__ inline int cmp_values(uint64_t la, uint32_t ll, uint64_t ra, uint32_t
 
 #define GPUC_MAX_ARITY 64
 
+// MA = max merge arity this instantiation supports (member-stream arrays are
+// MA-sized locals; small MA keeps them out of scratch — the host picks the
+// smallest instantiation >= n_inputs)
+template <int MA>
 __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
                                  uint64_t n_groups, uint64_t n_recs,
                                  const uint32_t* src_bases, ParsedCols pc, UnfCols in,
@@ -394,7 +398,7 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
     uint64_t endi = g + 1 < n_groups ? group_start[g + 1] : n_recs;
     uint32_t k = (uint32_t)(endi - beg);
     atomicAdd(&st->merged_counts[k > 64 ? 63 : k - 1], 1ull);
-    if (k > GPUC_MAX_ARITY) { atomicExch(error, 20ull); return; }
+    if (k > (uint32_t)MA) { atomicExch(error, 20ull); return; }
 
     const MRec r0 = recs[beg];
     int64_t token = (int64_t)(r0.tok ^ 0x8000000000000000ULL);
@@ -406,8 +410,8 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
     if (pp.has_shard && (token < pp.shard_lo || token > pp.shard_hi)) { op.keep[g] = 0; return; }
 
     // member streams
-    uint64_t mb[GPUC_MAX_ARITY];
-    uint32_t mcnt[GPUC_MAX_ARITY], mpos[GPUC_MAX_ARITY];
+    uint64_t mb[MA];
+    uint32_t mcnt[MA], mpos[MA];
     int64_t pdm = INT64_MIN;
     uint32_t pdl = LDT_NONE_U32;
     for (uint32_t m = 0; m < k; m++) {

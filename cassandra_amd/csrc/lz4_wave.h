@@ -359,22 +359,30 @@ __global__ void __launch_bounds__(WAVE) k_lz4_decompress_wave(const ChunkDesc* c
                                                               int verify_crc,
                                                               unsigned long long* error,
                                                               const uint32_t* crc_table) {
-    __shared__ uint8_t s_comp[LZ4_SLOT];
+    // compressed bytes are read through L1/L2 (all-lane same-address reads
+    // broadcast; no LDS staging -> higher occupancy, same trade as the
+    // compressor's global-src variant). crc_table is the 8x256 sliced set.
     uint32_t c = blockIdx.x;
     if (c >= n) return;
     int lane = threadIdx.x;
     ChunkDesc ch = chunks[c];
     if (ch.comp_len > LZ4_SLOT) { if (lane == 0) atomicExch(error, 9ull); return; }
-    for (uint32_t i = lane * 4; i + 4 <= ch.comp_len; i += WAVE * 4) {
-        uint32_t v;
-        memcpy(&v, ch.comp + i, 4);
-        *(uint32_t*)&s_comp[i] = v;
-    }
-    for (uint32_t i = (ch.comp_len & ~3u) + lane; i < ch.comp_len; i += WAVE) s_comp[i] = ch.comp[i];
-    __syncthreads();
+    const uint8_t* s_comp = ch.comp;
     if (verify_crc && lane == 0) {
+        // CRC32 of the compressed bytes, slicing-by-8 (zlib technique)
         uint32_t crc = 0xFFFFFFFFu;
-        for (uint32_t i = 0; i < ch.comp_len; i++) crc = crc_table[(crc ^ s_comp[i]) & 0xFF] ^ (crc >> 8);
+        uint32_t i = 0;
+        for (; i + 8 <= ch.comp_len; i += 8) {
+            uint32_t lo, hi;
+            memcpy(&lo, s_comp + i, 4);
+            memcpy(&hi, s_comp + i + 4, 4);
+            lo ^= crc;
+            crc = crc_table[7 * 256 + (lo & 0xFF)] ^ crc_table[6 * 256 + ((lo >> 8) & 0xFF)] ^
+                  crc_table[5 * 256 + ((lo >> 16) & 0xFF)] ^ crc_table[4 * 256 + (lo >> 24)] ^
+                  crc_table[3 * 256 + (hi & 0xFF)] ^ crc_table[2 * 256 + ((hi >> 8) & 0xFF)] ^
+                  crc_table[1 * 256 + ((hi >> 16) & 0xFF)] ^ crc_table[0 * 256 + (hi >> 24)];
+        }
+        for (; i < ch.comp_len; i++) crc = crc_table[(crc ^ s_comp[i]) & 0xFF] ^ (crc >> 8);
         crc = ~crc;
         uint32_t stored = ((uint32_t)ch.comp[ch.comp_len] << 24) | ((uint32_t)ch.comp[ch.comp_len + 1] << 16) |
                           ((uint32_t)ch.comp[ch.comp_len + 2] << 8) | ch.comp[ch.comp_len + 3];

@@ -22,9 +22,11 @@ int main() {
     if (hipGetDeviceCount(&ndev) != hipSuccess || ndev < 1) { printf("no gpu\n"); return 2; }
     uint32_t tab[256];
     crc32_make_table(tab);
+    static uint32_t tab8[8 * 256];
+    crc32_make_table8(tab8);  // decompress kernel uses the sliced set
     uint32_t* d_tab;
-    hipMalloc(&d_tab, sizeof(tab));
-    hipMemcpy(d_tab, tab, sizeof(tab), hipMemcpyHostToDevice);
+    hipMalloc(&d_tab, sizeof(tab8));
+    hipMemcpy(d_tab, tab8, sizeof(tab8), hipMemcpyHostToDevice);
     // probe LDS order
     {
         unsigned int* d_p;
