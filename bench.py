@@ -83,6 +83,9 @@ def main():
     ap.add_argument("--tombstone-pct", type=int, default=0)
     ap.add_argument("--range-tomb-pct", type=int, default=0)
     ap.add_argument("--dir", default=os.environ.get("GPUC_BENCH_DIR", "/tmp/gpuc_bench"))
+    ap.add_argument("--shards", type=int, default=1,
+                    help="n_output_shards per job (UCS-style sharded outputs; "
+                         "2 shards run concurrently inside the library)")
     ap.add_argument("--no-cpu-baseline", action="store_true")
     args = ap.parse_args()
 
@@ -124,7 +127,7 @@ def main():
     def one_step(i, prev=[None]):
         out = os.path.join(d, f"out-{i}", "oa-100-big")
         os.makedirs(os.path.dirname(out), exist_ok=True)
-        r = ca.compact(bases, out, device=device)
+        r = ca.compact(bases, out, device=device, n_output_shards=args.shards)
         # previous step's output is deleted in the background (bounded disk,
         # no serial rmtree inside the measured path)
         if prev[0]:
@@ -238,7 +241,8 @@ def main():
             "clustering_rows": args.clustering_rows,
             "tombstone_pct": args.tombstone_pct,
             "range_tomb_pct": args.range_tomb_pct,
-            "parallelism": f"token-independent shards x{world}, no collectives",
+            "parallelism": f"token-independent shards x{world}, no collectives"
+                           + (f"; {args.shards} sharded outputs/job" if args.shards > 1 else ""),
             "input_uncompressed_bytes_per_rank": last["input_uncompressed_bytes"],
             "gen_seconds": round(t_gen, 1),
             "phase_ms": last["ms"],

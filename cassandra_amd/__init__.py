@@ -35,6 +35,7 @@ class GpucJob(ctypes.Structure):
         ("token_lo", ctypes.c_int64),
         ("token_hi", ctypes.c_int64),
         ("device", ctypes.c_int32),
+        ("n_output_shards", ctypes.c_int32),
     ]
 
 
@@ -143,6 +144,7 @@ def compact(
     overlaps=None,
     token_range=None,
     device=0,
+    n_output_shards=1,
 ):
     """One compaction task (mirrors CompactionTask.runMayThrow's hot loop).
 
@@ -171,6 +173,7 @@ def compact(
         job.has_token_range = 1
         job.token_lo, job.token_hi = token_range
     job.device = device
+    job.n_output_shards = n_output_shards
     res = GpucResult()
     rc = lib.gpuc_compact(ctypes.byref(job), ctypes.byref(res))
     if rc != 0:

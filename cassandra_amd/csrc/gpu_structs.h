@@ -27,31 +27,30 @@ struct KeyLut {
     const uint64_t* pos[64];
     int enabled;
 };
-__device__ inline KeyLut g_key_lut;  // zero-init: disabled
 
-__device__ inline const uint8_t* mrec_key_bytes(const MRec& r) {
-    return g_key_lut.base[r.src] + g_key_lut.pos[r.src][r.idx] + 2;
+__device__ inline const uint8_t* mrec_key_bytes(const KeyLut& lut, const MRec& r) {
+    return lut.base[r.src] + lut.pos[r.src][r.idx] + 2;
 }
 // DecoratedKey order (DecoratedKey.java:79-92): token, then key bytes
 // compared as unsigned lexicographic with shorter-is-less on prefix equality
 // (ByteBufferUtil.compareUnsigned). The zero-padded 8-byte prefix compare is
 // exact whenever it differs; ties fall back to a byte walk from offset 8.
-__device__ inline bool mrec_less(const MRec& a, const MRec& b) {
+__device__ inline bool mrec_less(const KeyLut& lut, const MRec& a, const MRec& b) {
     if (a.tok != b.tok) return a.tok < b.tok;
     if (a.pfx != b.pfx) return a.pfx < b.pfx;
-    if (!g_key_lut.enabled || (a.klen <= 8 && b.klen <= 8)) return a.klen < b.klen;
-    const uint8_t* ka = mrec_key_bytes(a);
-    const uint8_t* kb = mrec_key_bytes(b);
+    if (!lut.enabled || (a.klen <= 8 && b.klen <= 8)) return a.klen < b.klen;
+    const uint8_t* ka = mrec_key_bytes(lut, a);
+    const uint8_t* kb = mrec_key_bytes(lut, b);
     uint32_t n = a.klen < b.klen ? a.klen : b.klen;
     for (uint32_t i = 8; i < n; i++)
         if (ka[i] != kb[i]) return ka[i] < kb[i];
     return a.klen < b.klen;
 }
-__device__ inline bool mrec_eq(const MRec& a, const MRec& b) {
+__device__ inline bool mrec_eq(const KeyLut& lut, const MRec& a, const MRec& b) {
     if (a.tok != b.tok || a.pfx != b.pfx || a.klen != b.klen) return false;
-    if (!g_key_lut.enabled || a.klen <= 8) return true;
-    const uint8_t* ka = mrec_key_bytes(a);
-    const uint8_t* kb = mrec_key_bytes(b);
+    if (!lut.enabled || a.klen <= 8) return true;
+    const uint8_t* ka = mrec_key_bytes(lut, a);
+    const uint8_t* kb = mrec_key_bytes(lut, b);
     for (uint32_t i = 8; i < a.klen; i++)
         if (ka[i] != kb[i]) return false;
     return true;

@@ -20,6 +20,7 @@
 #include <string>
 #include <thread>
 #include <future>
+#include <atomic>
 #include <fcntl.h>
 #include <unistd.h>
 #include <vector>
@@ -84,7 +85,7 @@ struct PinnedSlot {
         return p;
     }
 };
-static PinnedSlot g_pin_in[64], g_pin_out[4];
+static PinnedSlot g_pin_in[2][64], g_pin_out[2][4];
 
 // GPUC_TRACE=1: wall-clock checkpoints on stderr (host-gap hunting)
 static bool g_trace = getenv("GPUC_TRACE") != nullptr;
@@ -128,6 +129,30 @@ static void read_file_into(const std::string& path, uint8_t* dst, size_t n, int 
     }
     for (auto& x : th) x.join();
 }
+static void read_file_range(const std::string& path, uint8_t* dst, uint64_t off, size_t n,
+                            int nthreads = 3) {
+    if (n == 0) return;
+    std::vector<std::thread> th;
+    size_t per = (n + nthreads - 1) / nthreads;
+    for (int t = 0; t < nthreads; t++) {
+        size_t o = (size_t)t * per;
+        if (o >= n) break;
+        size_t len = std::min(per, n - o);
+        th.emplace_back([&, o, len]() {
+            int fd = open(path.c_str(), O_RDONLY);
+            if (fd < 0) return;
+            size_t done = 0;
+            while (done < len) {
+                ssize_t r = pread(fd, dst + o + done, len - done, (off_t)(off + o + done));
+                if (r <= 0) break;
+                done += (size_t)r;
+            }
+            close(fd);
+        });
+    }
+    for (auto& x : th) x.join();
+}
+
 static void write_file_parallel(const std::string& path, const uint8_t* p, size_t n, int nthreads = 4) {
     // create + size, then threaded pwrite
     FILE* f = fopen(path.c_str(), "wb");
@@ -272,10 +297,12 @@ static HStatistics parse_statistics(const bytes& b) {
 
 // Index.db -> partition positions (host thread; validates non-indexed entries)
 static void parse_index_positions(const bytes& ib, uint64_t data_len,
-                                  std::vector<uint64_t>& positions, std::string& err) {
+                                  std::vector<uint64_t>& positions,
+                                  std::vector<uint64_t>& entry_offs, std::string& err) {
     try {
         HReader r(ib);
         while (r.pos < r.len) {
+            entry_offs.push_back(r.pos);
             uint16_t klen = r.be16();
             r.skip(klen);
             uint64_t pos = r.uvint();
@@ -546,18 +573,19 @@ static uint64_t exscan_u64(uint64_t* d_data, uint64_t n, hipStream_t stream) {
 }
 
 // sort MRec array (uniform-run merge rounds); returns pointer to sorted buffer
-static MRec* merge_sort_recs(MRec* d_a, MRec* d_b, uint64_t n, hipStream_t stream) {
+static MRec* merge_sort_recs(MRec* d_a, MRec* d_b, uint64_t n, hipStream_t stream,
+                             const KeyLut& lut) {
     MRec* in = d_a;
     MRec* out = d_b;
     for (uint64_t run = 1; run < n; run *= 2) {
         if (run < 16) {
             uint64_t pairs = (n + run * 2 - 1) / (run * 2);
             hipLaunchKernelGGL(k_merge_small, dim3((uint32_t)((pairs + 255) / 256)), dim3(256), 0,
-                               stream, in, out, n, run);
+                               stream, in, out, n, run, lut);
         } else {
             uint64_t tiles = (n + MERGE_TILE - 1) / MERGE_TILE;
             hipLaunchKernelGGL(k_merge_uniform, dim3((uint32_t)((tiles + 255) / 256)), dim3(256), 0,
-                               stream, in, out, n, run);
+                               stream, in, out, n, run, lut);
         }
         std::swap(in, out);
     }
@@ -566,7 +594,7 @@ static MRec* merge_sort_recs(MRec* d_a, MRec* d_b, uint64_t n, hipStream_t strea
 
 // merge k pre-sorted runs (per-source) with pairwise rounds
 static MRec* merge_sorted_runs(MRec* d_a, MRec* d_b, std::vector<uint64_t> runs /*boundaries, size k+1*/,
-                               hipStream_t stream) {
+                               hipStream_t stream, const KeyLut& lut) {
     MRec* in = d_a;
     MRec* out = d_b;
     while (runs.size() > 2) {
@@ -593,7 +621,7 @@ static MRec* merge_sorted_runs(MRec* d_a, MRec* d_b, std::vector<uint64_t> runs 
                                  hipMemcpyHostToDevice, stream));
         uint64_t total_tiles = tile_beg;
         hipLaunchKernelGGL(k_merge_pairs, dim3((uint32_t)((total_tiles + 255) / 256)), dim3(256), 0,
-                           stream, in, out, d_pairs.as<MergePair>(), (uint32_t)pairs.size(), total_tiles);
+                           stream, in, out, d_pairs.as<MergePair>(), (uint32_t)pairs.size(), total_tiles, lut);
         HIP_CHECK(hipStreamSynchronize(stream));
         std::swap(in, out);
         runs = next_runs;
@@ -611,7 +639,9 @@ static void init_outstats(DevBuf& d_stats, hipStream_t stream) {
 
 static bool g_crc_tables_ready = false;
 static void* g_crc256 = nullptr;  // 8x256 sliced CRC tables (first 256 = base)
+static std::mutex g_crc_init_mu;
 static void ensure_crc_tables(hipStream_t stream) {
+    std::lock_guard<std::mutex> crc_g(g_crc_init_mu);
     if (g_crc_tables_ready) return;
     hipLaunchKernelGGL(k_crc_init, dim3(1), dim3(256), 0, stream);
     // plain 256-entry table for the wave LZ4 kernels
@@ -692,7 +722,7 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
                                            const std::string& key_type,
                                            const std::string& ck_type,
                                            const std::vector<std::pair<bytes, std::string>>& regular_cols,
-                                           hipStream_t stream) {
+                                           hipStream_t stream, int wslot = 0) {
     WriteDeviceOut w;
     TR("wsd: enter");
     static const std::vector<int64_t> ps_off_h = est_hist_offsets(155);
@@ -786,7 +816,7 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
     HIP_CHECK(hipStreamCreate(&cstream));
 
     // Index.db image is final right after serialize: drain it early on cstream
-    uint8_t* h_index = (uint8_t*)g_pin_out[1].get(total_idx ? total_idx : 1);
+    uint8_t* h_index = (uint8_t*)g_pin_out[wslot][1].get(total_idx ? total_idx : 1);
     if (!h_index) throw std::runtime_error("pinned out alloc failed");
     HIP_CHECK(hipStreamWaitEvent(cstream, ev2, 0));
     HIP_CHECK(hipMemcpyAsync(h_index, d_out_index.p, total_idx, hipMemcpyDeviceToHost, cstream));
@@ -816,7 +846,7 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
     }
     const uint64_t worst_slab = (uint64_t)SLAB * (LZ4_SLOT + 4);
     const int NSLOTS = 3;
-    uint8_t* h_slab0 = (uint8_t*)g_pin_out[0].get(worst_slab * NSLOTS);
+    uint8_t* h_slab0 = (uint8_t*)g_pin_out[wslot][0].get(worst_slab * NSLOTS);
     if (!h_slab0) throw std::runtime_error("pinned out alloc failed");
     DevBuf d_gat[2], d_foff[2];
     if (n_slabs) {
@@ -1055,124 +1085,87 @@ static void set_err(char* dst, size_t cap, const std::string& msg) {
     snprintf(dst, cap, "%s", msg.c_str());
 }
 
-extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
-    memset(res, 0, sizeof(*res));
-    double t_start_all;
+struct CompactSetup {
+    int k = 0;
+    std::vector<std::string> in_bases;
+    std::vector<bytes> index_data;
+    std::vector<HCompressionInfo> cinfos;
+    std::vector<HStatistics> stats;
+    std::vector<uint64_t> generations;
+    std::vector<std::vector<uint64_t>> positions;   // n_parts+1 absolute offsets
+    std::vector<std::vector<uint64_t>> entry_offs;  // Index.db entry byte offsets
+    std::vector<size_t> comp_file_sz;
+    std::vector<int32_t> col_fixed_h;
+    int32_t ck_width = 0;
+    std::string ck_type_str;
+};
+
+// Everything from data read through component write for ONE output sstable,
+// restricted to the partition window pr[s] = [blo, bhi) per input (the full
+// range in the unsharded case). wslot selects the pinned-arena set; shards
+// run two at a time so front-phase kernels of one shard overlap the other
+// shard's LDS-bound compress (they use disjoint CU resources).
+static void compact_one(const gpuc_job* job, const CompactSetup& su,
+                        const std::vector<std::pair<uint32_t, uint32_t>>& pr,
+                        const std::string& out_base_str, int wslot,
+                        gpuc_result* res, std::mutex& res_mu) {
     auto wall = []() {
         struct timespec ts;
         clock_gettime(CLOCK_MONOTONIC, &ts);
         return ts.tv_sec * 1e3 + ts.tv_nsec / 1e6;
     };
-    t_start_all = wall();
-    if (g_trace) g_trace_t0 = trace_wall();
-    try {
-        int ndev = gpuc_device_count();
-        if (ndev <= 0) { set_err(res->error, sizeof(res->error), "no HIP device (no CPU fallback)"); return GPUC_ERR_NO_GPU; }
-        HIP_CHECK(hipSetDevice(job->device));
-        hipStream_t stream;
-        HIP_CHECK(hipStreamCreate(&stream));
-        ensure_crc_tables(stream);
-
-        // ---- read + parse inputs (host metadata; Index parse on threads) ----
+    HIP_CHECK(hipSetDevice(job->device));
+    hipStream_t stream;
+    HIP_CHECK(hipStreamCreate(&stream));
+    {
+        int k = su.k;
+        const auto& in_bases = su.in_bases;
+        const auto& cinfos = su.cinfos;
+        const auto& stats = su.stats;
+        const auto& generations = su.generations;
+        const auto& positions = su.positions;
+        const auto& col_fixed_h = su.col_fixed_h;
+        int32_t ck_width = su.ck_width;
+        const std::string& ck_type_str = su.ck_type_str;
         double t0 = wall();
-        int k = job->n_inputs;
-        if (k < 1 || k > 64) { set_err(res->error, sizeof(res->error), "n_inputs must be 1..64"); return GPUC_ERR_UNSUPPORTED; }
-        std::vector<bytes> index_data(k);
-        std::vector<HCompressionInfo> cinfos(k);
-        std::vector<HStatistics> stats(k);
-        std::vector<uint64_t> generations(k);
+        // per-input window: partition range -> chunk range -> compressed range
+        std::vector<uint32_t> win_blo(k), win_n(k);
+        std::vector<uint64_t> win_chunk_lo(k), win_comp_lo(k);
+        std::vector<size_t> comp_sz(k, 0);        // compressed bytes in window
         std::vector<uint8_t*> comp_pin(k, nullptr);
-        std::vector<size_t> comp_sz(k, 0);
-        std::vector<std::string> in_bases(k);
-        // metadata components of all inputs read+parsed on one thread per
-        // sstable (Index.db alone is ~40 MB per input at C2 scale)
-        {
-            std::vector<std::thread> mth;
-            std::vector<std::string> merr(k);
-            for (int s = 0; s < k; s++) {
-                in_bases[s] = job->input_bases[s];
-                mth.emplace_back([&, s]() {
-                    try {
-                        const std::string& base = in_bases[s];
-                        index_data[s] = read_file(base + "-Index.db");
-                        cinfos[s] = parse_compression_info(read_file(base + "-CompressionInfo.db"));
-                        stats[s] = parse_statistics(read_file(base + "-Statistics.db"));
-                    } catch (const std::exception& e) { merr[s] = e.what(); }
-                });
-            }
-            for (auto& t : mth) t.join();
-            for (int s = 0; s < k; s++)
-                if (!merr[s].empty()) throw std::runtime_error(merr[s]);
-        }
+        std::vector<uint64_t> data_lo(k), data_hi(k);  // uncompressed byte window
         for (int s = 0; s < k; s++) {
-            std::string base = in_bases[s];
-            comp_sz[s] = file_size_of(base + "-Data.db");
-            comp_pin[s] = (uint8_t*)g_pin_in[s].get(comp_sz[s]);
+            uint32_t blo = pr[s].first, bhi = pr[s].second;
+            win_blo[s] = blo;
+            win_n[s] = bhi - blo;
+            data_lo[s] = positions[s][blo];
+            data_hi[s] = positions[s][bhi];
+            uint64_t c_lo = data_lo[s] / CHUNK_LEN;
+            uint64_t c_hi = (data_hi[s] + CHUNK_LEN - 1) / CHUNK_LEN;
+            uint64_t n_chunks_file = su.cinfos[s].offsets.size();
+            if (c_hi > n_chunks_file) c_hi = n_chunks_file;
+            if (win_n[s] == 0) { c_lo = c_hi = 0; }
+            win_chunk_lo[s] = c_lo;
+            uint64_t comp_lo = c_lo < n_chunks_file ? cinfos[s].offsets[c_lo] : su.comp_file_sz[s];
+            uint64_t comp_hi = c_hi < n_chunks_file ? cinfos[s].offsets[c_hi] : su.comp_file_sz[s];
+            if (win_n[s] == 0) comp_lo = comp_hi = 0;
+            win_comp_lo[s] = comp_lo;
+            comp_sz[s] = comp_hi - comp_lo;
+            comp_pin[s] = (uint8_t*)g_pin_in[wslot][s].get(comp_sz[s] ? comp_sz[s] : 1);
             if (!comp_pin[s]) throw std::runtime_error("pinned alloc failed");
-            if (stats[s].clustering_types.size() > 1)
-                throw std::runtime_error("at most one clustering column supported");
-            if (stats[s].regular_cols.empty() || stats[s].regular_cols.size() > 63)
-                throw std::runtime_error("1..63 regular columns supported");
-            if (!stats[s].partitioner.empty() && stats[s].partitioner.find("Murmur3") == std::string::npos)
-                throw std::runtime_error("Murmur3Partitioner required");
-            // generation from ".../oa-<id>-big"
-            size_t sl = base.rfind('/');
-            std::string name = sl == std::string::npos ? base : base.substr(sl + 1);
-            size_t a = name.find('-'), b2 = name.find('-', a + 1);
-            generations[s] = std::stoull(name.substr(a + 1, b2 - a - 1));
-            res->input_uncompressed_bytes += cinfos[s].data_len;
         }
-        // per-column fixed widths (−1 = variable) from the header types
-        std::vector<int32_t> col_fixed_h;
-        for (auto& [nm, ct] : stats[0].regular_cols) {
-            (void)nm;
-            if (ct == "org.apache.cassandra.db.marshal.LongType") col_fixed_h.push_back(8);
-            else if (ct == "org.apache.cassandra.db.marshal.Int32Type") col_fixed_h.push_back(4);
-            else if (ct == "org.apache.cassandra.db.marshal.BytesType" ||
-                     ct == "org.apache.cassandra.db.marshal.UTF8Type" ||
-                     ct == "org.apache.cassandra.db.marshal.AsciiType")
-                col_fixed_h.push_back(-1);
-            else throw std::runtime_error("unsupported column type " + ct);
-        }
-        // clustering column (0 or 1, fixed width)
-        int32_t ck_width = 0;
-        std::string ck_type_str;
-        if (!stats[0].clustering_types.empty()) {
-            ck_type_str = stats[0].clustering_types[0];
-            if (ck_type_str == "org.apache.cassandra.db.marshal.LongType") ck_width = 8;
-            else if (ck_type_str == "org.apache.cassandra.db.marshal.Int32Type") ck_width = 4;
-            else if (ck_type_str == "org.apache.cassandra.db.marshal.UTF8Type" ||
-                     ck_type_str == "org.apache.cassandra.db.marshal.AsciiType" ||
-                     ck_type_str == "org.apache.cassandra.db.marshal.BytesType")
-                ck_width = -1;  // variable width, unsigned-lex comparator
-            else throw std::runtime_error("unsupported clustering type " + ck_type_str);
-        }
-
-        // Data.db reads for ALL inputs start in parallel (host threads into
-        // pinned arenas); Index.db parses run alongside; per-sstable H2D is
-        // issued on a copy stream as soon as that file's read completes, and
-        // its decompress launch follows on the compute stream — read / H2D /
-        // decompress form a 3-deep pipeline instead of three serial phases.
         std::vector<std::thread> data_readers(k);
         for (int s = 0; s < k; s++)
             data_readers[s] = std::thread([&, s] {
-                read_file_into(in_bases[s] + "-Data.db", comp_pin[s], comp_sz[s], 3);
+                if (comp_sz[s])
+                    read_file_range(in_bases[s] + "-Data.db", comp_pin[s], win_comp_lo[s],
+                                    comp_sz[s], 3);
             });
-
-        std::vector<std::vector<uint64_t>> positions(k);
-        std::vector<std::string> perr(k);
         {
-            std::vector<std::thread> th;
-            for (int s = 0; s < k; s++)
-                th.emplace_back(parse_index_positions, std::cref(index_data[s]), cinfos[s].data_len,
-                                std::ref(positions[s]), std::ref(perr[s]));
-            for (auto& t : th) t.join();
-            for (int s = 0; s < k; s++)
-                if (!perr[s].empty()) throw std::runtime_error("Index.db parse: " + perr[s]);
+            std::lock_guard<std::mutex> g(res_mu);
+            res->ms_read_io += wall() - t0;
         }
-        double t1 = wall();
-        res->ms_read_io = t1 - t0;  // metadata + index parse; Data reads overlap H2D below
-        TR("meta+index parsed");
+        TR("window resolved");
 
         // ---- H2D + decompress (pipelined per sstable) ----
         hipStream_t copy_stream;
@@ -1188,42 +1181,58 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
         std::vector<std::vector<ChunkDesc>> chunks_all(k);  // kept alive past async copies
         size_t n_chunks_total = 0;
         double ms_read_data = 0;
+        std::vector<const uint8_t*> vbase(k, nullptr);  // virtual decompressed origin
         for (int s = 0; s < k; s++) {
             double tr = wall();
             data_readers[s].join();  // in-order wait; all reads run concurrently
             ms_read_data = wall() - tr + ms_read_data;
-            d_comp[s].alloc(comp_sz[s]);
-            HIP_CHECK(hipMemcpyAsync(d_comp[s].p, comp_pin[s], comp_sz[s],
-                                     hipMemcpyHostToDevice, copy_stream));
-            d_data[s].alloc(cinfos[s].data_len);
-            d_pos[s].alloc(positions[s].size() * 8);
-            HIP_CHECK(hipMemcpyAsync(d_pos[s].p, positions[s].data(), positions[s].size() * 8,
-                                     hipMemcpyHostToDevice, copy_stream));
             auto& ci = cinfos[s];
+            uint64_t n_chunks_file = ci.offsets.size();
+            uint64_t c_lo = win_chunk_lo[s];
+            uint64_t c_hi = win_n[s] ? (data_hi[s] + CHUNK_LEN - 1) / CHUNK_LEN : c_lo;
+            if (c_hi > n_chunks_file) c_hi = n_chunks_file;
+            uint64_t n_wchunks = c_hi - c_lo;
+            d_comp[s].alloc(comp_sz[s] ? comp_sz[s] : 1);
+            if (comp_sz[s])
+                HIP_CHECK(hipMemcpyAsync(d_comp[s].p, comp_pin[s], comp_sz[s],
+                                         hipMemcpyHostToDevice, copy_stream));
+            d_data[s].alloc(n_wchunks * (uint64_t)CHUNK_LEN + 16);
+            // positions stay ABSOLUTE: the decompressed window is addressed
+            // through a virtual origin so partition offsets need no rewrite
+            vbase[s] = d_data[s].as<uint8_t>() - c_lo * (uint64_t)CHUNK_LEN;
+            uint64_t n_wpos = (uint64_t)win_n[s] + 1;
+            d_pos[s].alloc(n_wpos * 8);
+            HIP_CHECK(hipMemcpyAsync(d_pos[s].p, positions[s].data() + win_blo[s], n_wpos * 8,
+                                     hipMemcpyHostToDevice, copy_stream));
             auto& chunks = chunks_all[s];
-            chunks.reserve(ci.offsets.size());
-            for (size_t c = 0; c < ci.offsets.size(); c++) {
+            chunks.reserve(n_wchunks);
+            for (uint64_t c = c_lo; c < c_hi; c++) {
                 uint64_t off = ci.offsets[c];
-                uint64_t end = c + 1 < ci.offsets.size() ? ci.offsets[c + 1] : comp_sz[s];
+                uint64_t end = c + 1 < n_chunks_file ? ci.offsets[c + 1] : su.comp_file_sz[s];
                 ChunkDesc cd;
-                cd.comp = d_comp[s].as<uint8_t>() + off;
-                cd.out = d_data[s].as<uint8_t>() + c * (uint64_t)CHUNK_LEN;
+                cd.comp = d_comp[s].as<uint8_t>() + (off - win_comp_lo[s]);
+                cd.out = d_data[s].as<uint8_t>() + (c - c_lo) * (uint64_t)CHUNK_LEN;
                 cd.comp_len = (uint32_t)(end - off - 4);
                 cd.out_len = (uint32_t)std::min<uint64_t>(CHUNK_LEN, ci.data_len - c * (uint64_t)CHUNK_LEN);
                 chunks.push_back(cd);
             }
             n_chunks_total += chunks.size();
-            d_chunks_s[s].alloc(chunks.size() * sizeof(ChunkDesc));
-            HIP_CHECK(hipMemcpyAsync(d_chunks_s[s].p, chunks.data(), chunks.size() * sizeof(ChunkDesc),
-                                     hipMemcpyHostToDevice, copy_stream));
+            d_chunks_s[s].alloc(chunks.size() * sizeof(ChunkDesc) + 16);
+            if (!chunks.empty())
+                HIP_CHECK(hipMemcpyAsync(d_chunks_s[s].p, chunks.data(), chunks.size() * sizeof(ChunkDesc),
+                                         hipMemcpyHostToDevice, copy_stream));
             HIP_CHECK(hipEventCreate(&ev_h2d[s]));
             HIP_CHECK(hipEventRecord(ev_h2d[s], copy_stream));
             HIP_CHECK(hipStreamWaitEvent(stream, ev_h2d[s], 0));
-            hipLaunchKernelGGL(k_lz4_decompress_wave, dim3((uint32_t)chunks.size()), dim3(WAVE), 0,
-                               stream, d_chunks_s[s].as<ChunkDesc>(), (uint32_t)chunks.size(), 1,
-                               d_error.as<unsigned long long>(), (const uint32_t*)g_crc256);
+            if (!chunks.empty())
+                hipLaunchKernelGGL(k_lz4_decompress_wave, dim3((uint32_t)chunks.size()), dim3(WAVE), 0,
+                                   stream, d_chunks_s[s].as<ChunkDesc>(), (uint32_t)chunks.size(), 1,
+                                   d_error.as<unsigned long long>(), (const uint32_t*)g_crc256);
         }
-        res->ms_read_io += ms_read_data;  // residual read wait not hidden by the pipeline
+        {
+            std::lock_guard<std::mutex> g(res_mu);
+            res->ms_read_io += ms_read_data;  // residual read wait not hidden by the pipeline
+        }
         TR("ingest issued");
         HIP_CHECK(hipEventRecord(e1, stream));  // e0..e1: H2D+decompress pipeline (overlapped)
         HIP_CHECK(hipEventRecord(e2, stream));
@@ -1232,16 +1241,15 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
         uint64_t total_parts = 0;
         std::vector<SrcDesc2> srcs(k);
         for (int s = 0; s < k; s++) {
-            srcs[s].data = d_data[s].as<uint8_t>();
+            srcs[s].data = vbase[s];
             srcs[s].part_pos = d_pos[s].as<uint64_t>();
-            srcs[s].n_parts = (uint32_t)(positions[s].size() - 1);
+            srcs[s].n_parts = win_n[s];
             srcs[s].min_ts = stats[s].hdr_min_ts;
             srcs[s].min_ldt = stats[s].hdr_min_ldt;
             srcs[s].min_ttl = stats[s].hdr_min_ttl;
             srcs[s].rec_base = (uint32_t)total_parts;
             total_parts += srcs[s].n_parts;
         }
-        res->partitions_in = total_parts;
         if (total_parts > 0xFFFFFFFFull) throw std::runtime_error("too many partitions for one job");
         SchemaParams sch{};
         sch.ck_width = ck_width;
@@ -1293,26 +1301,26 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
             TR("parse synced");
             HIP_CHECK(hipMemcpy(&err, d_error.p, 8, hipMemcpyDeviceToHost));
             if (err) throw std::runtime_error("GPU decode/parse error code " + std::to_string(err));
-            HIP_CHECK(hipMemcpy(&res->rows_in, d_rows_in.p, 8, hipMemcpyDeviceToHost));
+            uint64_t rows_in_local = 0;
+            HIP_CHECK(hipMemcpy(&rows_in_local, d_rows_in.p, 8, hipMemcpyDeviceToHost));
+            std::lock_guard<std::mutex> g(res_mu);
+            res->rows_in += rows_in_local;
         }
 
         // exact-key comparator fallback: point the merge kernels at the
         // decompressed inputs (keys > 8 bytes tie-break by byte walk)
-        {
-            KeyLut lut{};
-            for (int s2 = 0; s2 < k; s2++) {
-                lut.base[s2] = d_data[s2].as<uint8_t>();
-                lut.pos[s2] = d_pos[s2].as<uint64_t>();
-            }
-            lut.enabled = 1;
-            HIP_CHECK(hipMemcpyToSymbol(HIP_SYMBOL(g_key_lut), &lut, sizeof(lut)));
+        KeyLut lut{};
+        for (int s2 = 0; s2 < k; s2++) {
+            lut.base[s2] = vbase[s2];
+            lut.pos[s2] = d_pos[s2].as<uint64_t>();
         }
+        lut.enabled = 1;
 
         // ---- merge (pairwise rounds over pre-sorted source runs) ----
         std::vector<uint64_t> runs;
         runs.push_back(0);
         for (int s = 0; s < k; s++) runs.push_back(runs.back() + srcs[s].n_parts);
-        MRec* d_sorted = merge_sorted_runs(d_recs_a.as<MRec>(), d_recs_b.as<MRec>(), runs, stream);
+        MRec* d_sorted = merge_sorted_runs(d_recs_a.as<MRec>(), d_recs_b.as<MRec>(), runs, stream, lut);
         TR("merge issued");
 
         // ---- group heads + starts ----
@@ -1323,14 +1331,14 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
         {
             uint32_t blocks = (uint32_t)((total_parts + 255) / 256);
             hipLaunchKernelGGL(k_group_heads, dim3(blocks), dim3(256), 0, stream, d_sorted,
-                               total_parts, d_head.as<uint64_t>());
+                               total_parts, d_head.as<uint64_t>(), lut);
         }
         exscan_u64(d_head.as<uint64_t>(), total_parts, stream);
         {
             uint32_t blocks = (uint32_t)((total_parts + 255) / 256);
             hipLaunchKernelGGL(k_group_starts2, dim3(blocks), dim3(256), 0, stream, d_sorted,
                                total_parts, d_head.as<uint64_t>(), d_gstart.as<uint64_t>(),
-                               d_ngroups.as<uint64_t>());
+                               d_ngroups.as<uint64_t>(), lut);
         }
         uint64_t n_groups = 0;
         HIP_CHECK(hipStreamSynchronize(stream));
@@ -1437,47 +1445,243 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
             sp2.hs.min_ttl = min_ttl == INT32_MAX ? 0 : min_ttl;
         }
         WriteDeviceOut w = write_sstable_device(opb, out_rows, n_groups, sp2, d_stats, d_tomb,
-                                                tomb_cap, job->output_base, stats[0].key_type,
-                                                ck_type_str, stats[0].regular_cols, stream);
+                                                tomb_cap, out_base_str, stats[0].key_type,
+                                                ck_type_str, stats[0].regular_cols, stream, wslot);
         TR("writer done");
-        res->partitions_out = w.partitions;
-        res->rows_out = w.rows;
-        res->output_uncompressed_bytes = w.uncompressed_len;
-        res->output_compressed_bytes = w.compressed_len;
         {
             OutStats hst;
             HIP_CHECK(hipMemcpy(&hst, d_stats.p, sizeof(OutStats), hipMemcpyDeviceToHost));
-            for (int i = 0; i < 64; i++) res->merged_counts[i] = hst.merged_counts[i];
+            float th2d, tdec, tparse, tmerge, trec;
+            HIP_CHECK(hipEventElapsedTime(&th2d, e0, e1));
+            HIP_CHECK(hipEventElapsedTime(&tdec, e1, e2));
+            HIP_CHECK(hipEventElapsedTime(&tparse, e2, e3));
+            HIP_CHECK(hipEventElapsedTime(&tmerge, e3, e4));
+            HIP_CHECK(hipEventElapsedTime(&trec, er0, er1));
+            std::lock_guard<std::mutex> g(res_mu);
+            res->partitions_out += w.partitions;
+            res->rows_out += w.rows;
+            res->output_uncompressed_bytes += w.uncompressed_len;
+            res->output_compressed_bytes += w.compressed_len;
+            for (int i = 0; i < 64; i++) res->merged_counts[i] += hst.merged_counts[i];
+            res->ms_h2d = 0;         // overlapped into the ingest pipeline (e0..e1)
+            res->ms_decompress += th2d + tdec;  // read/H2D/decompress pipeline, GPU side
+            res->ms_parse += tparse;
+            res->ms_reconcile += trec;
+            res->ms_merge += tmerge - trec;  // merge rounds + grouping
+            res->ms_serialize += w.ms_sizes + w.ms_serialize;
+            res->ms_compress += w.ms_compress;
+            res->ms_d2h += w.ms_d2h;
+            res->ms_write_io += w.ms_io;
+            // dominant kernel estimate
+            struct { const char* n; double ms; } ks[] = {
+                {"k_lz4_decompress", (double)th2d + tdec}, {"k_parse", (double)tparse},
+                {"merge+reconcile", (double)tmerge},
+                {"k_serialize", w.ms_serialize}, {"k_lz4_compress", w.ms_compress}};
+            for (auto& kk : ks)
+                if (kk.ms > res->dominant_kernel_ms) {
+                    res->dominant_kernel_ms = kk.ms;
+                    snprintf(res->dominant_kernel, sizeof(res->dominant_kernel), "%s", kk.n);
+                }
+            res->dominant_kernel_launches = 1;
         }
-        float th2d, tdec, tparse, tmerge, trec;
-        HIP_CHECK(hipEventElapsedTime(&th2d, e0, e1));
-        HIP_CHECK(hipEventElapsedTime(&tdec, e1, e2));
-        HIP_CHECK(hipEventElapsedTime(&tparse, e2, e3));
-        HIP_CHECK(hipEventElapsedTime(&tmerge, e3, e4));
-        HIP_CHECK(hipEventElapsedTime(&trec, er0, er1));
-        res->ms_h2d = 0;         // overlapped into the ingest pipeline (e0..e1)
-        res->ms_decompress = th2d + tdec;  // read/H2D/decompress pipeline, GPU side
-        res->ms_parse = tparse;
-        res->ms_reconcile = trec;
-        res->ms_merge = tmerge - trec;  // merge rounds + grouping
-        res->ms_serialize = w.ms_sizes + w.ms_serialize;
-        res->ms_compress = w.ms_compress;
-        res->ms_d2h = w.ms_d2h;
-        res->ms_write_io = w.ms_io;
-        // dominant kernel estimate
-        struct { const char* n; double ms; } ks[] = {
-            {"k_lz4_decompress", (double)th2d + tdec}, {"k_parse", tparse}, {"merge+reconcile", tmerge},
-            {"k_serialize", w.ms_serialize}, {"k_lz4_compress", w.ms_compress}};
-        for (auto& kk : ks)
-            if (kk.ms > res->dominant_kernel_ms) {
-                res->dominant_kernel_ms = kk.ms;
-                snprintf(res->dominant_kernel, sizeof(res->dominant_kernel), "%s", kk.n);
-            }
-        res->dominant_kernel_launches = 1;
         for (hipEvent_t e : {e0, e1, e2, e3, e4, er0, er1}) (void)hipEventDestroy(e);
         for (int s = 0; s < k; s++) (void)hipEventDestroy(ev_h2d[s]);
         HIP_CHECK(hipStreamDestroy(copy_stream));
-        HIP_CHECK(hipStreamDestroy(stream));
+    }
+    HIP_CHECK(hipStreamDestroy(stream));
+}
+
+
+
+// ---------------------------------------------------------------------------
+// compaction setup (host metadata shared by all shards) + sharded driver
+// ---------------------------------------------------------------------------
+
+static void compact_setup(const gpuc_job* job, CompactSetup& su) {
+    int k = job->n_inputs;
+    su.k = k;
+    su.in_bases.resize(k);
+    su.index_data.resize(k);
+    su.cinfos.resize(k);
+    su.stats.resize(k);
+    su.generations.resize(k);
+    su.positions.resize(k);
+    su.entry_offs.resize(k);
+    su.comp_file_sz.resize(k);
+    {
+        std::vector<std::thread> mth;
+        std::vector<std::string> merr(k);
+        for (int s = 0; s < k; s++) {
+            su.in_bases[s] = job->input_bases[s];
+            mth.emplace_back([&, s]() {
+                try {
+                    const std::string& base = su.in_bases[s];
+                    su.index_data[s] = read_file(base + "-Index.db");
+                    su.cinfos[s] = parse_compression_info(read_file(base + "-CompressionInfo.db"));
+                    su.stats[s] = parse_statistics(read_file(base + "-Statistics.db"));
+                    su.comp_file_sz[s] = file_size_of(base + "-Data.db");
+                } catch (const std::exception& e) { merr[s] = e.what(); }
+            });
+        }
+        for (auto& t : mth) t.join();
+        for (int s = 0; s < k; s++)
+            if (!merr[s].empty()) throw std::runtime_error(merr[s]);
+    }
+    for (int s = 0; s < k; s++) {
+        const std::string& base = su.in_bases[s];
+        auto& st = su.stats[s];
+        if (st.clustering_types.size() > 1)
+            throw std::runtime_error("at most one clustering column supported");
+        if (st.regular_cols.empty() || st.regular_cols.size() > 63)
+            throw std::runtime_error("1..63 regular columns supported");
+        if (!st.partitioner.empty() && st.partitioner.find("Murmur3") == std::string::npos)
+            throw std::runtime_error("Murmur3Partitioner required");
+        size_t sl = base.rfind('/');
+        std::string name = sl == std::string::npos ? base : base.substr(sl + 1);
+        size_t a = name.find('-'), b2 = name.find('-', a + 1);
+        su.generations[s] = std::stoull(name.substr(a + 1, b2 - a - 1));
+    }
+    for (auto& [nm, ct] : su.stats[0].regular_cols) {
+        (void)nm;
+        if (ct == "org.apache.cassandra.db.marshal.LongType") su.col_fixed_h.push_back(8);
+        else if (ct == "org.apache.cassandra.db.marshal.Int32Type") su.col_fixed_h.push_back(4);
+        else if (ct == "org.apache.cassandra.db.marshal.BytesType" ||
+                 ct == "org.apache.cassandra.db.marshal.UTF8Type" ||
+                 ct == "org.apache.cassandra.db.marshal.AsciiType")
+            su.col_fixed_h.push_back(-1);
+        else throw std::runtime_error("unsupported column type " + ct);
+    }
+    if (!su.stats[0].clustering_types.empty()) {
+        su.ck_type_str = su.stats[0].clustering_types[0];
+        if (su.ck_type_str == "org.apache.cassandra.db.marshal.LongType") su.ck_width = 8;
+        else if (su.ck_type_str == "org.apache.cassandra.db.marshal.Int32Type") su.ck_width = 4;
+        else if (su.ck_type_str == "org.apache.cassandra.db.marshal.UTF8Type" ||
+                 su.ck_type_str == "org.apache.cassandra.db.marshal.AsciiType" ||
+                 su.ck_type_str == "org.apache.cassandra.db.marshal.BytesType")
+            su.ck_width = -1;  // variable width, unsigned-lex comparator
+        else throw std::runtime_error("unsupported clustering type " + su.ck_type_str);
+    }
+    {
+        std::vector<std::thread> th;
+        std::vector<std::string> perr(k);
+        for (int s = 0; s < k; s++)
+            th.emplace_back(parse_index_positions, std::cref(su.index_data[s]),
+                            su.cinfos[s].data_len, std::ref(su.positions[s]),
+                            std::ref(su.entry_offs[s]), std::ref(perr[s]));
+        for (auto& t : th) t.join();
+        for (int s = 0; s < k; s++)
+            if (!perr[s].empty()) throw std::runtime_error("Index.db parse: " + perr[s]);
+    }
+}
+
+// Murmur3 token of the key in Index.db entry at byte offset `off`
+static int64_t index_entry_token(const bytes& ib, uint64_t off) {
+    uint32_t klen = ((uint32_t)ib[off] << 8) | ib[off + 1];
+    return murmur3_token(ib.data() + off + 2, klen);
+}
+
+extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
+    memset(res, 0, sizeof(*res));
+    double t_start_all;
+    auto wall = []() {
+        struct timespec ts;
+        clock_gettime(CLOCK_MONOTONIC, &ts);
+        return ts.tv_sec * 1e3 + ts.tv_nsec / 1e6;
+    };
+    t_start_all = wall();
+    if (g_trace) g_trace_t0 = trace_wall();
+    try {
+        int ndev = gpuc_device_count();
+        if (ndev <= 0) { set_err(res->error, sizeof(res->error), "no HIP device (no CPU fallback)"); return GPUC_ERR_NO_GPU; }
+        HIP_CHECK(hipSetDevice(job->device));
+        {
+            hipStream_t s0;
+            HIP_CHECK(hipStreamCreate(&s0));
+            ensure_crc_tables(s0);
+            HIP_CHECK(hipStreamDestroy(s0));
+        }
+        if (job->n_inputs < 1 || job->n_inputs > 64) {
+            set_err(res->error, sizeof(res->error), "n_inputs must be 1..64");
+            return GPUC_ERR_UNSUPPORTED;
+        }
+        double t0 = wall();
+        CompactSetup su;
+        compact_setup(job, su);
+        res->ms_read_io = wall() - t0;
+        TR("meta+index parsed");
+        for (int s = 0; s < su.k; s++)
+            res->input_uncompressed_bytes += su.cinfos[s].data_len;
+        for (int s = 0; s < su.k; s++)
+            res->partitions_in += su.positions[s].size() - 1;
+
+        int S = job->n_output_shards > 1 ? job->n_output_shards : 1;
+        if (S > 64) throw std::runtime_error("n_output_shards must be <= 64");
+        std::mutex res_mu;
+        if (S == 1) {
+            std::vector<std::pair<uint32_t, uint32_t>> pr(su.k);
+            for (int s = 0; s < su.k; s++)
+                pr[s] = {0u, (uint32_t)(su.positions[s].size() - 1)};
+            compact_one(job, su, pr, job->output_base, 0, res, res_mu);
+        } else {
+            if (job->has_token_range)
+                throw std::runtime_error("token_range + n_output_shards unsupported");
+            // output bases: generation g, g+1, ... (a compaction may emit
+            // several sstables — UCS shard model / SplittingCompactionWriter)
+            std::string ob = job->output_base;
+            size_t sl = ob.rfind('/');
+            std::string dir = sl == std::string::npos ? std::string(".") : ob.substr(0, sl);
+            std::string name = sl == std::string::npos ? ob : ob.substr(sl + 1);
+            size_t a = name.find('-'), b2 = name.find('-', a + 1);
+            uint64_t gen0 = std::stoull(name.substr(a + 1, b2 - a - 1));
+            std::string pre = name.substr(0, a + 1), post = name.substr(b2);
+            // equal token-range split (cassandra_amd/sharding.py contract)
+            std::vector<int64_t> lo_tok(S + 1);
+            for (int i = 0; i < S; i++) {
+                uint64_t lo_u = (uint64_t)(((unsigned __int128)i << 64) / (unsigned)S);
+                lo_tok[i] = (int64_t)(lo_u + 0x8000000000000000ULL);
+            }
+            // per input: shard boundary partition = first with token >= lo_tok[i]
+            std::vector<std::vector<uint32_t>> bounds(su.k, std::vector<uint32_t>(S + 1));
+            for (int s = 0; s < su.k; s++) {
+                uint32_t n = (uint32_t)(su.positions[s].size() - 1);
+                bounds[s][0] = 0;
+                bounds[s][S] = n;
+                for (int i = 1; i < S; i++) {
+                    uint32_t lo = 0, hi = n;  // lower_bound(token >= lo_tok[i])
+                    while (lo < hi) {
+                        uint32_t mid = (lo + hi) >> 1;
+                        if (index_entry_token(su.index_data[s], su.entry_offs[s][mid]) < lo_tok[i])
+                            lo = mid + 1;
+                        else hi = mid;
+                    }
+                    bounds[s][i] = lo;
+                }
+            }
+            // two workers: front-phase kernels of one shard overlap the
+            // other shard's compress
+            std::atomic<int> next{0};
+            std::vector<std::string> werr(2);
+            auto workfn = [&](int w) {
+                for (;;) {
+                    int i = next.fetch_add(1);
+                    if (i >= S) break;
+                    std::vector<std::pair<uint32_t, uint32_t>> pr(su.k);
+                    for (int s = 0; s < su.k; s++) pr[s] = {bounds[s][i], bounds[s][i + 1]};
+                    std::string out_i = dir + "/" + pre + std::to_string(gen0 + i) + post;
+                    try {
+                        compact_one(job, su, pr, out_i, w, res, res_mu);
+                    } catch (const std::exception& e) {
+                        werr[w] = e.what();
+                        break;
+                    }
+                }
+            };
+            std::thread w0(workfn, 0), w1(workfn, 1);
+            w0.join();
+            w1.join();
+            for (auto& e : werr)
+                if (!e.empty()) throw std::runtime_error(e);
+        }
         res->ms_total = wall() - t_start_all;
         return GPUC_OK;
     } catch (const std::exception& e) {
@@ -1486,6 +1690,7 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
         return GPUC_ERR_INTERNAL;
     }
 }
+
 
 extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* error, size_t error_len) {
     try {
@@ -1521,10 +1726,7 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
                 throw std::runtime_error("ck_text needs clustering_rows*16 < 1e8 (8-digit order)");
             if (gp.key_len < 8 || gp.key_len > 255)
                 throw std::runtime_error("key_len must be 8..255 (generator contract)");
-            {
-                KeyLut lut{};  // generated 8-byte prefixes are unique: prefix compare is exact
-                HIP_CHECK(hipMemcpyToSymbol(HIP_SYMBOL(g_key_lut), &lut, sizeof(lut)));
-            }
+            KeyLut lut{};  // generated 8-byte prefixes are unique: prefix compare is exact
             DevBuf d_a, d_b, d_ids, d_keys, d_vals, d_stats, d_tomb, d_prows;
             d_a.alloc(R * sizeof(MRec));
             d_b.alloc(R * sizeof(MRec));
@@ -1533,7 +1735,7 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
             uint32_t blocks = (uint32_t)((R + 255) / 256);
             hipLaunchKernelGGL(k_gen_recs2, dim3(blocks), dim3(256), 0, stream, gp,
                                d_a.as<MRec>(), d_ids.as<uint64_t>(), d_keys.as<uint8_t>());
-            MRec* d_sorted = merge_sort_recs(d_a.as<MRec>(), d_b.as<MRec>(), R, stream);
+            MRec* d_sorted = merge_sort_recs(d_a.as<MRec>(), d_b.as<MRec>(), R, stream, lut);
             OutPartsBuf opb;
             opb.alloc(R);
             d_prows.alloc(R * 8);

@@ -75,7 +75,7 @@ struct MergePair {
 #define MERGE_TILE 16
 
 __global__ void k_merge_pairs(const MRec* in, MRec* out, const MergePair* pairs,
-                              uint32_t n_pairs, uint64_t total_tiles) {
+                              uint32_t n_pairs, uint64_t total_tiles, KeyLut lut) {
     uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (t >= total_tiles) return;
     // find pair by binary search over tile_beg
@@ -98,7 +98,7 @@ __global__ void k_merge_pairs(const MRec* in, MRec* out, const MergePair* pairs,
     while (ilo < ihi) {
         uint64_t i = (ilo + ihi + 1) >> 1;
         // A[i-1] goes before B[k-i] iff !(B < A) — ties favor A
-        if (!mrec_less(in[p.b_beg + (k - i)], in[p.a_beg + i - 1])) ilo = i;
+        if (!mrec_less(lut, in[p.b_beg + (k - i)], in[p.a_beg + i - 1])) ilo = i;
         else ihi = i - 1;
     }
     uint64_t i = ilo, j = k - ilo;
@@ -106,7 +106,7 @@ __global__ void k_merge_pairs(const MRec* in, MRec* out, const MergePair* pairs,
         bool take_a;
         if (i >= an) take_a = false;
         else if (j >= bn) take_a = true;
-        else take_a = !mrec_less(in[p.b_beg + j], in[p.a_beg + i]);  // ties -> A
+        else take_a = !mrec_less(lut, in[p.b_beg + j], in[p.a_beg + i]);  // ties -> A
         out[p.out_beg + o] = take_a ? in[p.a_beg + i++] : in[p.b_beg + j++];
     }
 }
@@ -114,10 +114,10 @@ __global__ void k_merge_pairs(const MRec* in, MRec* out, const MergePair* pairs,
 // ---------------------------------------------------------------------------
 // group heads + exclusive scan utilities (u64)
 // ---------------------------------------------------------------------------
-__global__ void k_group_heads(const MRec* recs, uint64_t n, uint64_t* head) {
+__global__ void k_group_heads(const MRec* recs, uint64_t n, uint64_t* head, KeyLut lut) {
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
-    head[i] = (i == 0) || !mrec_eq(recs[i], recs[i - 1]);
+    head[i] = (i == 0) || !mrec_eq(lut, recs[i], recs[i - 1]);
 }
 
 #define SCAN_BLOCK 256
@@ -178,10 +178,10 @@ __global__ void k_group_starts(const uint64_t* head, const uint64_t* head_scan, 
 // ---------------------------------------------------------------------------
 // group starts using the scanned head values (heads recomputed from recs)
 __global__ void k_group_starts2(const MRec* recs, uint64_t n, const uint64_t* head_scan,
-                                uint64_t* group_start, uint64_t* n_groups) {
+                                uint64_t* group_start, uint64_t* n_groups, KeyLut lut) {
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
-    bool head = (i == 0) || !mrec_eq(recs[i], recs[i - 1]);
+    bool head = (i == 0) || !mrec_eq(lut, recs[i], recs[i - 1]);
     if (head) group_start[head_scan[i]] = i;
     // total groups = exclusive-scan of flags at the last element PLUS its own
     // flag (the last record need not start a group)
@@ -225,7 +225,7 @@ __global__ void k_chunk_gather(const uint8_t* slots, const uint32_t* csize, cons
 // synthetic generator kernels (shared contract with oracle/src/gen.h)
 // ---------------------------------------------------------------------------
 // small uniform runs (run_size < 16): one thread merges a whole pair
-__global__ void k_merge_small(const MRec* in, MRec* out, uint64_t n, uint64_t run_size) {
+__global__ void k_merge_small(const MRec* in, MRec* out, uint64_t n, uint64_t run_size, KeyLut lut) {
     uint64_t p = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     uint64_t a_beg = p * run_size * 2;
     if (a_beg >= n) return;
@@ -234,13 +234,13 @@ __global__ void k_merge_small(const MRec* in, MRec* out, uint64_t n, uint64_t ru
     uint64_t b_end = min(a_beg + run_size * 2, n);
     uint64_t i = a_beg, j = b_beg, o = a_beg;
     while (i < a_end && j < b_end)
-        out[o++] = !mrec_less(in[j], in[i]) ? in[i++] : in[j++];
+        out[o++] = !mrec_less(lut, in[j], in[i]) ? in[i++] : in[j++];
     while (i < a_end) out[o++] = in[i++];
     while (j < b_end) out[o++] = in[j++];
 }
 
 // uniform-run merge round for the generator sort (runs of size run_size >= 16)
-__global__ void k_merge_uniform(const MRec* in, MRec* out, uint64_t n, uint64_t run_size) {
+__global__ void k_merge_uniform(const MRec* in, MRec* out, uint64_t n, uint64_t run_size, KeyLut lut) {
     uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     uint64_t k0 = t * MERGE_TILE;
     if (k0 >= n) return;
@@ -258,7 +258,7 @@ __global__ void k_merge_uniform(const MRec* in, MRec* out, uint64_t n, uint64_t 
     uint64_t ihi = k < an ? k : an;
     while (ilo < ihi) {
         uint64_t i = (ilo + ihi + 1) >> 1;
-        if (!mrec_less(in[b_beg + (k - i)], in[a_beg + i - 1])) ilo = i;
+        if (!mrec_less(lut, in[b_beg + (k - i)], in[a_beg + i - 1])) ilo = i;
         else ihi = i - 1;
     }
     uint64_t i = ilo, j = k - ilo;
@@ -266,7 +266,7 @@ __global__ void k_merge_uniform(const MRec* in, MRec* out, uint64_t n, uint64_t 
         bool take_a;
         if (i >= an) take_a = false;
         else if (j >= bn) take_a = true;
-        else take_a = !mrec_less(in[b_beg + j], in[a_beg + i]);
+        else take_a = !mrec_less(lut, in[b_beg + j], in[a_beg + i]);
         out[a_beg + o] = take_a ? in[a_beg + i++] : in[b_beg + j++];
     }
 }

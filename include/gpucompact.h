@@ -76,6 +76,13 @@ typedef struct gpuc_job {
     int64_t token_hi;           /* inclusive */
 
     int32_t device;             /* HIP device ordinal */
+
+    /* 0/1 = one output sstable. N>1: the job is split into N equal Murmur3
+       token ranges; each produces its own complete output sstable
+       (generation, generation+1, ...) — the UCS shard model / a splitting
+       compaction writer. Two ranges are processed concurrently so front-phase
+       kernels overlap the LDS-bound compressor. */
+    int32_t n_output_shards;
 } gpuc_job;
 
 typedef struct gpuc_result {

@@ -189,6 +189,41 @@ def test_compaction_parity(ca, oracle_bin, tmp_path, case):
     _assert_dirs_equal(f"{d}/oa-90-big", f"{d}/oa-91-big")
 
 
+def test_sharded_output_parity(ca, oracle_bin, tmp_path):
+    """n_output_shards=S: one call emits S sstables over equal token ranges,
+    each byte-identical to an oracle compaction restricted to that range
+    (shards run two at a time inside the library)."""
+    import sys
+    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    from cassandra_amd.sharding import split_token_range
+    d = str(tmp_path)
+    S = 3
+    _oracle_gen(d, seed=91, n=4, rows=2500, vlen=300, overlap=30, tomb=10)
+    ins = [f"{d}/oa-{g}-big" for g in range(1, 5)]
+    ca.compact(ins, f"{d}/oa-200-big", n_output_shards=S)
+    for i in range(S):
+        lo, hi = split_token_range(S, i)
+        _oracle_compact(f"{d}/oa-{300+i}-big", ins, shard=f"{lo}:{hi}")
+        _assert_dirs_equal(f"{d}/oa-{300+i}-big", f"{d}/oa-{200+i}-big")
+
+
+def test_sharded_output_parity_wide(ca, oracle_bin, tmp_path):
+    """Sharded outputs with wide partitions + long keys + multi-column."""
+    import sys
+    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    from cassandra_amd.sharding import split_token_range
+    d = str(tmp_path)
+    S = 2
+    _oracle_gen(d, seed=92, n=3, rows=40, crows=50, vlen=200, overlap=30, tomb=10,
+                rtomb=30, ncols=3, keylen=24)
+    ins = [f"{d}/oa-{g}-big" for g in range(1, 4)]
+    ca.compact(ins, f"{d}/oa-200-big", n_output_shards=S)
+    for i in range(S):
+        lo, hi = split_token_range(S, i)
+        _oracle_compact(f"{d}/oa-{300+i}-big", ins, shard=f"{lo}:{hi}")
+        _assert_dirs_equal(f"{d}/oa-{300+i}-big", f"{d}/oa-{200+i}-big")
+
+
 def test_purge_overlap_table_parity(ca, oracle_bin, tmp_path):
     """gcBefore purge gated by the token-interval min-timestamp table."""
     d = str(tmp_path)
