@@ -1098,6 +1098,14 @@ struct CompactSetup {
     std::vector<int32_t> col_fixed_h;
     int32_t ck_width = 0;
     std::string ck_type_str;
+    // unsharded fast path: whole-file Data.db reads started during index
+    // parse (compact_one adopts them when its window covers the full file)
+    mutable std::vector<std::thread> full_readers;
+    std::vector<uint8_t*> full_pin;
+    ~CompactSetup() {
+        for (auto& t : full_readers)
+            if (t.joinable()) t.join();
+    }
 };
 
 // Everything from data read through component write for ONE output sstable,
@@ -1151,16 +1159,23 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
             if (win_n[s] == 0) comp_lo = comp_hi = 0;
             win_comp_lo[s] = comp_lo;
             comp_sz[s] = comp_hi - comp_lo;
-            comp_pin[s] = (uint8_t*)g_pin_in[wslot][s].get(comp_sz[s] ? comp_sz[s] : 1);
-            if (!comp_pin[s]) throw std::runtime_error("pinned alloc failed");
         }
-        std::vector<std::thread> data_readers(k);
-        for (int s = 0; s < k; s++)
-            data_readers[s] = std::thread([&, s] {
-                if (comp_sz[s])
-                    read_file_range(in_bases[s] + "-Data.db", comp_pin[s], win_comp_lo[s],
-                                    comp_sz[s], 3);
-            });
+        bool adopt = !su.full_pin.empty();
+        std::vector<std::thread> data_readers(adopt ? 0 : k);
+        for (int s = 0; s < k; s++) {
+            if (adopt) {
+                // setup already reads the whole file into the slot-0 arenas
+                comp_pin[s] = su.full_pin[s] + win_comp_lo[s];
+            } else {
+                comp_pin[s] = (uint8_t*)g_pin_in[wslot][s].get(comp_sz[s] ? comp_sz[s] : 1);
+                if (!comp_pin[s]) throw std::runtime_error("pinned alloc failed");
+                data_readers[s] = std::thread([&, s] {
+                    if (comp_sz[s])
+                        read_file_range(in_bases[s] + "-Data.db", comp_pin[s], win_comp_lo[s],
+                                        comp_sz[s], 3);
+                });
+            }
+        }
         {
             std::lock_guard<std::mutex> g(res_mu);
             res->ms_read_io += wall() - t0;
@@ -1184,7 +1199,8 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         std::vector<const uint8_t*> vbase(k, nullptr);  // virtual decompressed origin
         for (int s = 0; s < k; s++) {
             double tr = wall();
-            data_readers[s].join();  // in-order wait; all reads run concurrently
+            if (adopt) su.full_readers[s].join();
+            else data_readers[s].join();  // in-order wait; all reads run concurrently
             ms_read_data = wall() - tr + ms_read_data;
             auto& ci = cinfos[s];
             uint64_t n_chunks_file = ci.offsets.size();
@@ -1497,7 +1513,7 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
 // compaction setup (host metadata shared by all shards) + sharded driver
 // ---------------------------------------------------------------------------
 
-static void compact_setup(const gpuc_job* job, CompactSetup& su) {
+static void compact_setup(const gpuc_job* job, CompactSetup& su, bool preread_full) {
     int k = job->n_inputs;
     su.k = k;
     su.in_bases.resize(k);
@@ -1540,6 +1556,18 @@ static void compact_setup(const gpuc_job* job, CompactSetup& su) {
         std::string name = sl == std::string::npos ? base : base.substr(sl + 1);
         size_t a = name.find('-'), b2 = name.find('-', a + 1);
         su.generations[s] = std::stoull(name.substr(a + 1, b2 - a - 1));
+    }
+    if (preread_full) {
+        su.full_pin.resize(k, nullptr);
+        su.full_readers.resize(k);
+        for (int s = 0; s < k; s++) {
+            su.full_pin[s] = (uint8_t*)g_pin_in[0][s].get(su.comp_file_sz[s] ? su.comp_file_sz[s] : 1);
+            if (!su.full_pin[s]) throw std::runtime_error("pinned alloc failed");
+            su.full_readers[s] = std::thread([&su, s] {
+                read_file_range(su.in_bases[s] + "-Data.db", su.full_pin[s], 0,
+                                su.comp_file_sz[s], 3);
+            });
+        }
     }
     for (auto& [nm, ct] : su.stats[0].regular_cols) {
         (void)nm;
@@ -1605,8 +1633,9 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
             return GPUC_ERR_UNSUPPORTED;
         }
         double t0 = wall();
+        int S_pre = job->n_output_shards > 1 ? job->n_output_shards : 1;
         CompactSetup su;
-        compact_setup(job, su);
+        compact_setup(job, su, S_pre == 1);
         res->ms_read_io = wall() - t0;
         TR("meta+index parsed");
         for (int s = 0; s < su.k; s++)
@@ -1614,7 +1643,7 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
         for (int s = 0; s < su.k; s++)
             res->partitions_in += su.positions[s].size() - 1;
 
-        int S = job->n_output_shards > 1 ? job->n_output_shards : 1;
+        int S = S_pre;
         if (S > 64) throw std::runtime_error("n_output_shards must be <= 64");
         std::mutex res_mu;
         if (S == 1) {
