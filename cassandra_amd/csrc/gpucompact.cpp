@@ -1524,18 +1524,34 @@ static void compact_setup(const gpuc_job* job, CompactSetup& su, bool preread_fu
     su.positions.resize(k);
     su.entry_offs.resize(k);
     su.comp_file_sz.resize(k);
+    // Data.db reads start FIRST (they are the long pole); metadata parse and
+    // index decode run while they stream into the pinned arenas
+    if (preread_full) {
+        su.full_pin.resize(k, nullptr);
+        su.full_readers.resize(k);
+        for (int s = 0; s < k; s++) {
+            su.in_bases[s] = job->input_bases[s];
+            su.comp_file_sz[s] = file_size_of(su.in_bases[s] + "-Data.db");
+            su.full_pin[s] = (uint8_t*)g_pin_in[0][s].get(su.comp_file_sz[s] ? su.comp_file_sz[s] : 1);
+            if (!su.full_pin[s]) throw std::runtime_error("pinned alloc failed");
+            su.full_readers[s] = std::thread([&su, s] {
+                read_file_range(su.in_bases[s] + "-Data.db", su.full_pin[s], 0,
+                                su.comp_file_sz[s], 4);
+            });
+        }
+    }
     {
         std::vector<std::thread> mth;
         std::vector<std::string> merr(k);
         for (int s = 0; s < k; s++) {
             su.in_bases[s] = job->input_bases[s];
-            mth.emplace_back([&, s]() {
+            mth.emplace_back([&, s, preread_full]() {
                 try {
                     const std::string& base = su.in_bases[s];
                     su.index_data[s] = read_file(base + "-Index.db");
                     su.cinfos[s] = parse_compression_info(read_file(base + "-CompressionInfo.db"));
                     su.stats[s] = parse_statistics(read_file(base + "-Statistics.db"));
-                    su.comp_file_sz[s] = file_size_of(base + "-Data.db");
+                    if (!preread_full) su.comp_file_sz[s] = file_size_of(base + "-Data.db");
                 } catch (const std::exception& e) { merr[s] = e.what(); }
             });
         }
@@ -1556,18 +1572,6 @@ static void compact_setup(const gpuc_job* job, CompactSetup& su, bool preread_fu
         std::string name = sl == std::string::npos ? base : base.substr(sl + 1);
         size_t a = name.find('-'), b2 = name.find('-', a + 1);
         su.generations[s] = std::stoull(name.substr(a + 1, b2 - a - 1));
-    }
-    if (preread_full) {
-        su.full_pin.resize(k, nullptr);
-        su.full_readers.resize(k);
-        for (int s = 0; s < k; s++) {
-            su.full_pin[s] = (uint8_t*)g_pin_in[0][s].get(su.comp_file_sz[s] ? su.comp_file_sz[s] : 1);
-            if (!su.full_pin[s]) throw std::runtime_error("pinned alloc failed");
-            su.full_readers[s] = std::thread([&su, s] {
-                read_file_range(su.in_bases[s] + "-Data.db", su.full_pin[s], 0,
-                                su.comp_file_sz[s], 3);
-            });
-        }
     }
     for (auto& [nm, ct] : su.stats[0].regular_cols) {
         (void)nm;
