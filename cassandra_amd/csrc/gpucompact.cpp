@@ -298,11 +298,13 @@ static HStatistics parse_statistics(const bytes& b) {
 // Index.db -> partition positions (host thread; validates non-indexed entries)
 static void parse_index_positions(const bytes& ib, uint64_t data_len,
                                   std::vector<uint64_t>& positions,
-                                  std::vector<uint64_t>& entry_offs, std::string& err) {
+                                  std::vector<uint64_t>* entry_offs, std::string& err) {
     try {
         HReader r(ib);
+        positions.reserve(ib.size() / 18 + 2);
+        if (entry_offs) entry_offs->reserve(ib.size() / 18 + 2);
         while (r.pos < r.len) {
-            entry_offs.push_back(r.pos);
+            if (entry_offs) entry_offs->push_back(r.pos);
             uint16_t klen = r.be16();
             r.skip(klen);
             uint64_t pos = r.uvint();
@@ -1599,7 +1601,7 @@ static void compact_setup(const gpuc_job* job, CompactSetup& su, bool preread_fu
         for (int s = 0; s < k; s++)
             th.emplace_back(parse_index_positions, std::cref(su.index_data[s]),
                             su.cinfos[s].data_len, std::ref(su.positions[s]),
-                            std::ref(su.entry_offs[s]), std::ref(perr[s]));
+                            preread_full ? nullptr : &su.entry_offs[s], std::ref(perr[s]));
         for (auto& t : th) t.join();
         for (int s = 0; s < k; s++)
             if (!perr[s].empty()) throw std::runtime_error("Index.db parse: " + perr[s]);
