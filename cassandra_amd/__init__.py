@@ -120,10 +120,22 @@ def load_library(path=None):
         ctypes.c_size_t,
     ]
     lib.gpuc_generate.restype = ctypes.c_int
+    lib.gpuc_verify.argtypes = [ctypes.c_char_p, ctypes.c_int32, ctypes.c_char_p, ctypes.c_size_t]
+    lib.gpuc_verify.restype = ctypes.c_int
     lib.gpuc_version.restype = ctypes.c_char_p
     lib.gpuc_device_count.restype = ctypes.c_int
     _lib = lib
     return lib
+
+
+def verify(input_base, device=0):
+    """sstableverify --extended equivalent: raises GpuCompactError on any
+    corruption (chunk CRC, digest, row format, key order, bloom mismatch)."""
+    lib = load_library()
+    err = ctypes.create_string_buffer(256)
+    rc = lib.gpuc_verify(input_base.encode(), device, err, 256)
+    if rc != 0:
+        raise GpuCompactError(f"gpuc_verify rc={rc}: {err.value.decode(errors='replace')}")
 
 
 def version():

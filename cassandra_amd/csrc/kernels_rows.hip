@@ -1394,6 +1394,34 @@ __global__ void k_collect_rows(OutParts op, UnfCols out, uint64_t n, uint32_t n_
 }
 
 // ---------------------------------------------------------------------------
+// verify epilogue (Verifier.java semantics: key order + bloom recheck)
+// ---------------------------------------------------------------------------
+// adjacent DecoratedKey order: every partition must sort strictly after its
+// predecessor (Verifier.java "out of order" check)
+__global__ void k_verify_order(const MRec* recs, uint64_t n, KeyLut lut,
+                               unsigned long long* error) {
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i == 0 || i >= n) return;
+    if (!mrec_less(lut, recs[i - 1], recs[i])) atomicExch(error, 30ull);
+}
+
+// recompute bloom bits from the parsed keys (FilterComponent recheck)
+__global__ void k_verify_bloom(ParsedCols pc, const MRec* recs, uint64_t n,
+                               uint32_t* bloom_bits, uint64_t bloom_bitlen, int32_t bloom_k) {
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n) return;
+    uint64_t h[2];
+    murmur3_128((const uint8_t*)pc.key_addr[i], recs[i].klen, 0, h);
+    int64_t base = (int64_t)h[1], inc = (int64_t)h[0];
+    for (int j = 0; j < bloom_k; j++) {
+        int64_t m = base % (int64_t)bloom_bitlen;
+        uint64_t idx = (uint64_t)((m ^ (m >> 63)) - (m >> 63));
+        atomicOr(&bloom_bits[idx >> 5], 1u << (idx & 31));
+        base += inc;
+    }
+}
+
+// ---------------------------------------------------------------------------
 // generator (general): thread per partition fills rows (+ markers) mirroring
 // oracle/src/gen.h; values generated per live row into the values arena.
 // ---------------------------------------------------------------------------

@@ -134,6 +134,13 @@ typedef struct gpuc_gen_spec {
 
 int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* error, size_t error_len);
 
+/* Verify one sstable (Verifier.java / sstableverify --extended semantics):
+ * CompressionInfo frame walk, per-chunk CRC32, full row-format walk,
+ * strict DecoratedKey order, Digest.crc32 recomputation, bloom-filter
+ * rebuild compared against Filter.db. GPUC_OK or GPUC_ERR_FORMAT with a
+ * message. */
+int gpuc_verify(const char* input_base, int32_t device, char* error, size_t error_len);
+
 /* library/build identification */
 const char* gpuc_version(void);
 /* returns number of visible HIP devices (0 => gpuc_compact will fail) */

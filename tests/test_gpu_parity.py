@@ -224,6 +224,35 @@ def test_sharded_output_parity_wide(ca, oracle_bin, tmp_path):
         _assert_dirs_equal(f"{d}/oa-{300+i}-big", f"{d}/oa-{200+i}-big")
 
 
+def test_verify_epilogue(ca, oracle_bin, tmp_path):
+    """gpuc_verify passes on valid sstables (oracle- and GPU-written, all
+    schema shapes) and fails on corruption of data, digest or filter."""
+    import shutil
+    d = str(tmp_path)
+    _oracle_gen(d, seed=95, n=1, rows=2000, vlen=300, overlap=0, tomb=10)
+    _oracle_gen(d + "/w", seed=96, n=1, rows=30, crows=60, vlen=200, rtomb=30,
+                cktext=1, keylen=24, ncols=3, overlap=0)
+    ca.generate(d + "/g", seed=97, n_sstables=1, rows_per_sstable=2000, value_len=256)
+    for b in [f"{d}/oa-1-big", f"{d}/w/oa-1-big", f"{d}/g/oa-1-big"]:
+        ca.verify(b)
+    # corrupt one data byte mid-file -> chunk CRC must fail
+    for comp, what in [("Data.db", "crc"), ("Digest.crc32", "digest"), ("Filter.db", "bloom")]:
+        cd = f"{d}/corrupt_{what}"
+        os.makedirs(cd)
+        for c in COMPONENTS:
+            shutil.copy(f"{d}/oa-1-big-{c}", f"{cd}/oa-1-big-{c}")
+        with open(f"{cd}/oa-1-big-{comp}", "r+b") as f:
+            f.seek(os.path.getsize(f"{cd}/oa-1-big-{comp}") // 2)
+            b0 = f.read(1)
+            f.seek(-1, 1)
+            f.write(bytes([b0[0] ^ 0x01]))
+        try:
+            ca.verify(f"{cd}/oa-1-big")
+            raise AssertionError(f"verify accepted corrupted {comp}")
+        except Exception as e:
+            assert "gpuc_verify" in str(e), e
+
+
 def test_purge_overlap_table_parity(ca, oracle_bin, tmp_path):
     """gcBefore purge gated by the token-interval min-timestamp table."""
     d = str(tmp_path)
