@@ -229,6 +229,7 @@ def test_verify_epilogue(ca, oracle_bin, tmp_path):
     schema shapes) and fails on corruption of data, digest or filter."""
     import shutil
     d = str(tmp_path)
+    os.makedirs(d + "/w"), os.makedirs(d + "/g")
     _oracle_gen(d, seed=95, n=1, rows=2000, vlen=300, overlap=0, tomb=10)
     _oracle_gen(d + "/w", seed=96, n=1, rows=30, crows=60, vlen=200, rtomb=30,
                 cktext=1, keylen=24, ncols=3, overlap=0)
