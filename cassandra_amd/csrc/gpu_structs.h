@@ -81,12 +81,15 @@ struct UnfCols {
     uint64_t* val_addr;   // absolute device address of value bytes
     uint32_t* val_len;
     uint8_t* cell_flags;  // CF_* per cell
-    // variable-width clustering (ck_width == -1, e.g. UTF8Type/BytesType):
-    // `ck` then holds the first min(8,len) value bytes big-endian zero-padded
-    // (unsigned lexicographic prefix, no sign flip); full bytes live at
-    // ck_addr/ck_len and break prefix ties exactly.
+    // clustering components, strided by n_ck: index = row_o * n_ck + c.
+    // Fixed-width components store the sign-flipped big-endian value in `ck`;
+    // variable-width components (UTF8/Bytes) store the first min(8,len) bytes
+    // big-endian zero-padded (unsigned lex prefix, no flip) with the full
+    // bytes at ck_addr/ck_len breaking prefix ties. ck_count = components
+    // present (rows: n_ck; bounds may be shorter prefixes).
     uint64_t* ck_addr;
     uint32_t* ck_len;
+    uint8_t* ck_count;
 };
 enum : uint8_t { PF_HAS_ROW = 1, PF_LIVE_TS = 2, PF_ROW_DEL = 4 };
 // per-cell flags (cell_flags array)

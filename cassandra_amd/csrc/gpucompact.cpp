@@ -363,7 +363,7 @@ struct OutMeta {
     // header (deltas) of the output sstable
     HeaderStats hs;
     std::string key_type;
-    std::string ck_type;  // empty = no clustering column
+    std::vector<std::string> ck_types;  // empty = no clustering columns
     std::vector<std::pair<bytes, std::string>> regular_cols;
     // collected stats
     int64_t min_timestamp, max_timestamp, min_ldt, max_ldt;
@@ -410,8 +410,8 @@ static bytes serialize_statistics_out(const OutMeta& m) {
         put_be64(stats, 0);               // repairedAt
         // improvedMinMax: clustering type list + covered Slice (spec'd with the
         // oracle: BOTTOM..TOP bounds, no values)
-        put_uvint(stats, m.ck_type.empty() ? 0 : 1);
-        if (!m.ck_type.empty()) put_type_str(stats, m.ck_type);
+        put_uvint(stats, m.ck_types.size());
+        for (auto& t : m.ck_types) put_type_str(stats, t);
         stats.push_back(1); put_be16(stats, 0);  // Slice start: INCL_START, 0 values
         stats.push_back(6); put_be16(stats, 0);  // Slice end: INCL_END, 0 values
         stats.push_back(0);               // hasLegacyCounterShards
@@ -435,8 +435,8 @@ static bytes serialize_statistics_out(const OutMeta& m) {
         put_uvint(header, sext32(m.hs.min_ldt - DELETION_TIME_EPOCH));
         put_uvint(header, sext32(m.hs.min_ttl));
         put_type_str(header, m.key_type);
-        put_uvint(header, m.ck_type.empty() ? 0 : 1);
-        if (!m.ck_type.empty()) put_type_str(header, m.ck_type);
+        put_uvint(header, m.ck_types.size());
+        for (auto& t : m.ck_types) put_type_str(header, t);
         put_uvint(header, 0);  // static columns
         put_uvint(header, m.regular_cols.size());
         for (auto& [name, t] : m.regular_cols) {
@@ -686,24 +686,25 @@ struct OutPartsBuf {
 struct UnfColsBuf {
     DevBuf ck, rkind, flags, live_ts, live_ttl, live_let, rdel_mfda, rdel_ldt,
         start_mfda, start_ldt, cell_ts, cell_ldt, cell_ttl, val_addr, val_len, ck_addr, ck_len,
-        cell_flags;
+        cell_flags, ck_count;
     UnfCols uc{};
-    void alloc(uint64_t n, uint32_t n_cols) {
+    void alloc(uint64_t n, uint32_t n_cols, uint32_t n_ck) {
         if (!n) n = 1;
         uint64_t nc = n * n_cols;
-        ck.alloc(n * 8); rkind.alloc(n); flags.alloc(n); live_ts.alloc(n * 8);
+        uint64_t nk = n * (n_ck ? n_ck : 1);
+        ck.alloc(nk * 8); rkind.alloc(n); flags.alloc(n); live_ts.alloc(n * 8);
         live_ttl.alloc(n * 4); live_let.alloc(n * 8); rdel_mfda.alloc(n * 8);
         rdel_ldt.alloc(n * 4); start_mfda.alloc(n * 8); start_ldt.alloc(n * 4);
         cell_ts.alloc(nc * 8); cell_ldt.alloc(nc * 4); cell_ttl.alloc(nc * 4);
         val_addr.alloc(nc * 8); val_len.alloc(nc * 4); cell_flags.alloc(nc);
-        ck_addr.alloc(n * 8); ck_len.alloc(n * 4);
+        ck_addr.alloc(nk * 8); ck_len.alloc(nk * 4); ck_count.alloc(n);
         uc = UnfCols{ck.as<uint64_t>(), rkind.as<uint8_t>(), flags.as<uint8_t>(),
                      live_ts.as<int64_t>(), live_ttl.as<int32_t>(), live_let.as<int64_t>(),
                      rdel_mfda.as<int64_t>(), rdel_ldt.as<uint32_t>(), start_mfda.as<int64_t>(),
                      start_ldt.as<uint32_t>(), cell_ts.as<int64_t>(), cell_ldt.as<uint32_t>(),
                      cell_ttl.as<int32_t>(), val_addr.as<uint64_t>(), val_len.as<uint32_t>(),
                      cell_flags.as<uint8_t>(),
-                     ck_addr.as<uint64_t>(), ck_len.as<uint32_t>()};
+                     ck_addr.as<uint64_t>(), ck_len.as<uint32_t>(), ck_count.as<uint8_t>()};
     }
 };
 
@@ -722,7 +723,7 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
                                            DevBuf& d_tomb, uint32_t tomb_cap,
                                            const std::string& out_base,
                                            const std::string& key_type,
-                                           const std::string& ck_type,
+                                           const std::vector<std::string>& ck_types,
                                            const std::vector<std::pair<bytes, std::string>>& regular_cols,
                                            hipStream_t stream, int wslot = 0) {
     WriteDeviceOut w;
@@ -944,7 +945,7 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
         OutMeta m{};
         m.hs = sp.hs;
         m.key_type = key_type;
-        m.ck_type = ck_type;
+        m.ck_types = ck_types;
         m.regular_cols = regular_cols;
         bool no_ts = hst.min_ts_flip == 0xFFFFFFFFFFFFFFFFULL;
         m.min_timestamp = no_ts ? 0 : (int64_t)(hst.min_ts_flip ^ 0x8000000000000000ULL);
@@ -1087,6 +1088,17 @@ static void set_err(char* dst, size_t cap, const std::string& msg) {
     snprintf(dst, cap, "%s", msg.c_str());
 }
 
+// marshal type string -> clustering component width (-1 variable)
+static int32_t ck_type_width(const std::string& t) {
+    if (t == "org.apache.cassandra.db.marshal.LongType") return 8;
+    if (t == "org.apache.cassandra.db.marshal.Int32Type") return 4;
+    if (t == "org.apache.cassandra.db.marshal.UTF8Type" ||
+        t == "org.apache.cassandra.db.marshal.AsciiType" ||
+        t == "org.apache.cassandra.db.marshal.BytesType")
+        return -1;
+    throw std::runtime_error("unsupported clustering type " + t);
+}
+
 struct CompactSetup {
     int k = 0;
     std::vector<std::string> in_bases;
@@ -1098,8 +1110,7 @@ struct CompactSetup {
     std::vector<std::vector<uint64_t>> entry_offs;  // Index.db entry byte offsets
     std::vector<size_t> comp_file_sz;
     std::vector<int32_t> col_fixed_h;
-    int32_t ck_width = 0;
-    std::string ck_type_str;
+    std::vector<int32_t> ck_widths;      // per clustering column
     // unsharded fast path: whole-file Data.db reads started during index
     // parse (compact_one adopts them when its window covers the full file)
     mutable std::vector<std::thread> full_readers;
@@ -1135,8 +1146,7 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         const auto& generations = su.generations;
         const auto& positions = su.positions;
         const auto& col_fixed_h = su.col_fixed_h;
-        int32_t ck_width = su.ck_width;
-        const std::string& ck_type_str = su.ck_type_str;
+
         double t0 = wall();
         // per-input window: partition range -> chunk range -> compressed range
         std::vector<uint32_t> win_blo(k), win_n(k);
@@ -1270,7 +1280,13 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         }
         if (total_parts > 0xFFFFFFFFull) throw std::runtime_error("too many partitions for one job");
         SchemaParams sch{};
-        sch.ck_width = ck_width;
+        sch.n_ck = (uint32_t)su.ck_widths.size();
+        DevBuf d_ck_w;
+        d_ck_w.alloc(su.ck_widths.size() * 4 + 8);
+        if (sch.n_ck)
+            HIP_CHECK(hipMemcpyAsync(d_ck_w.p, su.ck_widths.data(), su.ck_widths.size() * 4,
+                                     hipMemcpyHostToDevice, stream));
+        sch.ck_w = d_ck_w.as<int32_t>();
         sch.n_cols = (uint32_t)col_fixed_h.size();
         DevBuf d_col_fixed;
         d_col_fixed.alloc(col_fixed_h.size() * 4);
@@ -1304,7 +1320,7 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         }
         uint64_t total_in_rows = exscan_u64(pc.row_base, total_parts, stream);
         UnfColsBuf in_rows;
-        in_rows.alloc(total_in_rows, sch.n_cols);
+        in_rows.alloc(total_in_rows, sch.n_cols, sch.n_ck);
         {
             uint32_t blocks = (uint32_t)((total_parts + 255) / 256);
             hipLaunchKernelGGL(k_parse_rows, dim3(blocks), dim3(256), 0, stream,
@@ -1383,7 +1399,7 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         OutPartsBuf opb;
         opb.alloc(n_groups);
         UnfColsBuf out_rows;
-        out_rows.alloc(total_out_rows, sch.n_cols);
+        out_rows.alloc(total_out_rows, sch.n_cols, sch.n_ck);
         DevBuf d_stats, d_tomb;
         d_stats.alloc(sizeof(OutStats));
         init_outstats(d_stats, stream);
@@ -1464,7 +1480,8 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         }
         WriteDeviceOut w = write_sstable_device(opb, out_rows, n_groups, sp2, d_stats, d_tomb,
                                                 tomb_cap, out_base_str, stats[0].key_type,
-                                                ck_type_str, stats[0].regular_cols, stream, wslot);
+                                                stats[0].clustering_types, stats[0].regular_cols,
+                                                stream, wslot);
         TR("writer done");
         {
             OutStats hst;
@@ -1564,8 +1581,6 @@ static void compact_setup(const gpuc_job* job, CompactSetup& su, bool preread_fu
     for (int s = 0; s < k; s++) {
         const std::string& base = su.in_bases[s];
         auto& st = su.stats[s];
-        if (st.clustering_types.size() > 1)
-            throw std::runtime_error("at most one clustering column supported");
         if (st.regular_cols.empty() || st.regular_cols.size() > 63)
             throw std::runtime_error("1..63 regular columns supported");
         if (!st.partitioner.empty() && st.partitioner.find("Murmur3") == std::string::npos)
@@ -1585,16 +1600,9 @@ static void compact_setup(const gpuc_job* job, CompactSetup& su, bool preread_fu
             su.col_fixed_h.push_back(-1);
         else throw std::runtime_error("unsupported column type " + ct);
     }
-    if (!su.stats[0].clustering_types.empty()) {
-        su.ck_type_str = su.stats[0].clustering_types[0];
-        if (su.ck_type_str == "org.apache.cassandra.db.marshal.LongType") su.ck_width = 8;
-        else if (su.ck_type_str == "org.apache.cassandra.db.marshal.Int32Type") su.ck_width = 4;
-        else if (su.ck_type_str == "org.apache.cassandra.db.marshal.UTF8Type" ||
-                 su.ck_type_str == "org.apache.cassandra.db.marshal.AsciiType" ||
-                 su.ck_type_str == "org.apache.cassandra.db.marshal.BytesType")
-            su.ck_width = -1;  // variable width, unsigned-lex comparator
-        else throw std::runtime_error("unsupported clustering type " + su.ck_type_str);
-    }
+    for (auto& t : su.stats[0].clustering_types) su.ck_widths.push_back(ck_type_width(t));
+    if (su.ck_widths.size() > 32)
+        throw std::runtime_error("at most 32 clustering columns supported");
     {
         std::vector<std::thread> th;
         std::vector<std::string> perr(k);
@@ -1817,12 +1825,15 @@ extern "C" int gpuc_verify(const char* input_base, int32_t device, char* error,
         d_cf.alloc(cfh.size() * 4);
         HIP_CHECK(hipMemcpyAsync(d_cf.p, cfh.data(), cfh.size() * 4, hipMemcpyHostToDevice, stream));
         sch.col_fixed = d_cf.as<int32_t>();
-        if (!st.clustering_types.empty()) {
-            const std::string& ck = st.clustering_types[0];
-            if (ck == "org.apache.cassandra.db.marshal.LongType") sch.ck_width = 8;
-            else if (ck == "org.apache.cassandra.db.marshal.Int32Type") sch.ck_width = 4;
-            else sch.ck_width = -1;
-        }
+        std::vector<int32_t> vckw;
+        for (auto& ct : st.clustering_types) vckw.push_back(ck_type_width(ct));
+        sch.n_ck = (uint32_t)vckw.size();
+        DevBuf d_vckw;
+        d_vckw.alloc(vckw.size() * 4 + 8);
+        if (sch.n_ck)
+            HIP_CHECK(hipMemcpyAsync(d_vckw.p, vckw.data(), vckw.size() * 4,
+                                     hipMemcpyHostToDevice, stream));
+        sch.ck_w = d_vckw.as<int32_t>();
         sch.column_index_size = 64 * 1024;
         SrcDesc2 src{};
         src.data = d_data.as<uint8_t>();
@@ -1915,6 +1926,9 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
             gp.key_len = spec->key_len ? spec->key_len : 8;
             gp.ck_text = spec->ck_text;
             gp.n_value_cols = spec->n_value_cols ? spec->n_value_cols : 1;
+            gp.ck_cols = spec->ck_cols;
+            if (gp.ck_cols > 2) throw std::runtime_error("ck_cols must be 0..2");
+            if (gp.ck_cols == 2 && gp.ck_text) throw std::runtime_error("ck_cols=2 with ck_text unsupported");
             gp.col_missing_pct = spec->col_missing_pct;
             if (gp.n_value_cols > 63) throw std::runtime_error("n_value_cols must be 1..63");
             if (gp.ck_text && (uint64_t)gp.clustering_rows * 16 >= 100000000ull)
@@ -1939,7 +1953,8 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
                                d_keys.as<uint8_t>());
             uint64_t total_rows = exscan_u64(d_prows.as<uint64_t>(), R, stream);
             UnfColsBuf rows;
-            rows.alloc(total_rows, gp.n_value_cols);
+            uint32_t gen_nck = gp.clustering_rows ? (gp.ck_cols ? gp.ck_cols : 1) : 0;
+            rows.alloc(total_rows, gp.n_value_cols, gen_nck);
             d_vals.alloc(total_rows * (uint64_t)gp.n_value_cols * spec->value_len);
             DevBuf d_ckarena;
             if (gp.ck_text) d_ckarena.alloc(total_rows * 16 + 16);
@@ -1968,7 +1983,14 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
                                 ? DELETION_TIME_EPOCH : (int64_t)(hs0.min_ldt_flip ^ 0x8000000000000000ULL);
             if (sp.hs.min_ldt == NO_DELETION_TIME) sp.hs.min_ldt = DELETION_TIME_EPOCH;
             sp.hs.min_ttl = 0;
-            sp.sch.ck_width = spec->clustering_rows ? (gp.ck_text ? -1 : 8) : 0;
+            sp.sch.n_ck = gen_nck;
+            std::vector<int32_t> ckw_h(gen_nck, gp.ck_text ? -1 : 8);
+            DevBuf d_ckw;
+            d_ckw.alloc(ckw_h.size() * 4 + 8);
+            if (gen_nck)
+                HIP_CHECK(hipMemcpyAsync(d_ckw.p, ckw_h.data(), ckw_h.size() * 4,
+                                         hipMemcpyHostToDevice, stream));
+            sp.sch.ck_w = d_ckw.as<int32_t>();
             sp.sch.n_cols = gp.n_value_cols;
             std::vector<int32_t> gcf(gp.n_value_cols, -1);  // val blobs
             DevBuf d_gcf;
@@ -1994,10 +2016,9 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
             write_sstable_device(opb, rows, R, sp, d_stats, d_tomb, tomb_cap, base,
                                  gp.key_len > 8 ? "org.apache.cassandra.db.marshal.BytesType"
                                                 : "org.apache.cassandra.db.marshal.LongType",
-                                 spec->clustering_rows
-                                     ? (gp.ck_text ? "org.apache.cassandra.db.marshal.UTF8Type"
-                                                   : "org.apache.cassandra.db.marshal.LongType")
-                                     : "",
+                                 std::vector<std::string>(
+                                     gen_nck, gp.ck_text ? "org.apache.cassandra.db.marshal.UTF8Type"
+                                                         : "org.apache.cassandra.db.marshal.LongType"),
                                  cols, stream);
         }
         HIP_CHECK(hipStreamDestroy(stream));

@@ -26,7 +26,8 @@ __device__ inline void tomb_push(OutStats* st, uint32_t* ldts, uint32_t cap, uin
 
 // schema/runtime constants shared by the general kernels
 struct SchemaParams {
-    int32_t ck_width;        // 0 = none; 4/8 fixed width; -1 variable (UTF8/Bytes)
+    uint32_t n_ck;           // clustering columns (0..32)
+    const int32_t* ck_w;     // per clustering column: 4/8 fixed or -1 variable
     uint32_t n_cols;         // regular columns (1..63; header superset order)
     const int32_t* col_fixed;  // per column: -1 variable else fixed width
     uint32_t column_index_size;  // promoted-index granularity (64 KiB default)
@@ -50,28 +51,44 @@ __device__ inline void ck_bytes(uint64_t ck, int width, uint8_t* out) {
     uint64_t v = ck ^ (1ULL << (8 * width - 1));
     for (int b = 0; b < width; b++) out[b] = (uint8_t)(v >> (8 * (width - 1 - b)));
 }
-// position compare: (ck, Kind.comparison); returns <0, 0, >0.
-// Fixed-width ck compares via the sortable u64; variable-width compares the
-// 8-byte prefix then byte-walks the full values (unsigned lex, shorter-first
-// on prefix equality — ClusteringComparator over UTF8/Bytes values).
-__device__ inline int pos_cmp(const UnfCols& in, uint64_t a, uint64_t b, int ck_width) {
-    if (ck_width != 0) {
-        uint64_t ca = in.ck[a], cb = in.ck[b];
-        if (ca != cb) return ca < cb ? -1 : 1;
-        if (ck_width < 0) {
-            uint32_t la = in.ck_len[a], lb = in.ck_len[b];
-            if (la > 8 || lb > 8) {
-                const uint8_t* pa = (const uint8_t*)in.ck_addr[a];
-                const uint8_t* pb = (const uint8_t*)in.ck_addr[b];
-                uint32_t n = la < lb ? la : lb;
-                for (uint32_t i = 8; i < n; i++)
-                    if (pa[i] != pb[i]) return pa[i] < pb[i] ? -1 : 1;
-            }
-            if (la != lb) return la < lb ? -1 : 1;
-        }
+// Kind.comparedToClustering (ClusteringPrefix.java): placement of a
+// SHORTER prefix (bound) against longer prefixes sharing its components
+__device__ inline int bk_cmp_to_clustering(uint8_t kk) {
+    const int tbl[8] = {-1, -1, -1, -1, 0, 1, 1, 1};
+    return tbl[kk];
+}
+// one clustering component (typed): fixed widths compare via the sortable
+// u64; variable widths compare prefix then byte-walk (unsigned lex,
+// shorter-first)
+__device__ inline int ck_cmp_component(const UnfCols& u, uint64_t ac, uint64_t bc, int32_t w) {
+    uint64_t pa = u.ck[ac], pb = u.ck[bc];
+    if (pa != pb) return pa < pb ? -1 : 1;
+    if (w >= 0) return 0;
+    uint32_t la = u.ck_len[ac], lb = u.ck_len[bc];
+    if (la > 8 || lb > 8) {
+        const uint8_t* a8 = (const uint8_t*)u.ck_addr[ac];
+        const uint8_t* b8 = (const uint8_t*)u.ck_addr[bc];
+        uint32_t n = la < lb ? la : lb;
+        for (uint32_t i = 8; i < n; i++)
+            if (a8[i] != b8[i]) return a8[i] < b8[i] ? -1 : 1;
     }
-    int c1 = bk_comparison(in.rkind[a]), c2 = bk_comparison(in.rkind[b]);
-    return c1 - c2;
+    return la == lb ? 0 : (la < lb ? -1 : 1);
+}
+// ClusteringComparator over full prefixes (compare_clustering_prefix in the
+// oracle; ClusteringComparator.compare in the reference): component-wise,
+// then size, then Kind.comparison
+__device__ inline int pos_cmp(const UnfCols& in, uint64_t a, uint64_t b, const SchemaParams& sch) {
+    if (sch.n_ck) {
+        uint32_t na = in.ck_count[a], nb = in.ck_count[b];
+        uint32_t mn = na < nb ? na : nb;
+        for (uint32_t c = 0; c < mn; c++) {
+            int r = ck_cmp_component(in, a * sch.n_ck + c, b * sch.n_ck + c, sch.ck_w[c]);
+            if (r) return r;
+        }
+        if (na != nb)
+            return na < nb ? bk_cmp_to_clustering(in.rkind[a]) : -bk_cmp_to_clustering(in.rkind[b]);
+    }
+    return bk_comparison(in.rkind[a]) - bk_comparison(in.rkind[b]);
 }
 
 // ---------------------------------------------------------------------------
@@ -140,17 +157,19 @@ __global__ void k_parse_count(const SrcDesc2* srcs, uint32_t n_srcs, uint32_t to
             uint8_t kind = base[pos++];
             uint32_t nv = ((uint32_t)base[pos] << 8) | base[pos + 1];
             pos += 2;
-            if (kind == BK_STATIC || nv > 1 || (nv == 1 && sp.ck_width == 0)) { atomicExch(error, 15ull); return; }
-            if (nv == 0 && sp.ck_width != 0) { atomicExch(error, 16ull); return; }  // 0-value bound unsupported
+            if (kind == BK_STATIC || nv > sp.n_ck || (nv && sp.n_ck == 0)) { atomicExch(error, 15ull); return; }
+            if (nv == 0 && sp.n_ck != 0) { atomicExch(error, 16ull); return; }  // 0-value bound unsupported
             if (nv) {
                 uint64_t hdr = uvint_get(base, &pos);
                 if (hdr) { atomicExch(error, 17ull); return; }
-                pos += sp.ck_width > 0 ? (uint64_t)sp.ck_width : uvint_get(base, &pos);
+                for (uint32_t c = 0; c < nv; c++)
+                    pos += sp.ck_w[c] > 0 ? (uint64_t)sp.ck_w[c] : uvint_get(base, &pos);
             }
-        } else if (sp.ck_width) {
+        } else if (sp.n_ck) {
             uint64_t hdr = uvint_get(base, &pos);
             if (hdr) { atomicExch(error, 17ull); return; }  // null/empty clustering unsupported
-            pos += sp.ck_width > 0 ? (uint64_t)sp.ck_width : uvint_get(base, &pos);
+            for (uint32_t c = 0; c < sp.n_ck; c++)
+                pos += sp.ck_w[c] > 0 ? (uint64_t)sp.ck_w[c] : uvint_get(base, &pos);
         }
         uint64_t size = uvint_get(base, &pos);
         uint64_t prev = uvint_get(base, &pos);
@@ -192,26 +211,25 @@ __global__ void k_parse_rows(const SrcDesc2* srcs, uint32_t n_srcs, uint32_t tot
             uint8_t kind = base[pos++];
             uint32_t nv = ((uint32_t)base[pos] << 8) | base[pos + 1];
             pos += 2;
-            uint64_t ck = 0, ckaddr = 0;
-            uint32_t cklen = 0;
-            if (nv) {
-                uvint_get(base, &pos);
-                if (sp.ck_width > 0) {
-                    ck = ck_sortable(base + pos, sp.ck_width);
-                    cklen = (uint32_t)sp.ck_width;
+            if (nv) uvint_get(base, &pos);  // 32-batch header (0: all non-null)
+            for (uint32_t c = 0; c < nv; c++) {
+                uint64_t oc = o * sp.n_ck + c;
+                uint32_t cklen;
+                if (sp.ck_w[c] > 0) {
+                    rc.ck[oc] = ck_sortable(base + pos, sp.ck_w[c]);
+                    cklen = (uint32_t)sp.ck_w[c];
                 } else {
                     cklen = (uint32_t)uvint_get(base, &pos);
-                    ck = ck_prefix_var(base + pos, cklen);
+                    rc.ck[oc] = ck_prefix_var(base + pos, cklen);
                 }
-                ckaddr = (uint64_t)(base + pos);
+                rc.ck_addr[oc] = (uint64_t)(base + pos);
+                rc.ck_len[oc] = cklen;
                 pos += cklen;
             }
             uvint_get(base, &pos);  // size
             uvint_get(base, &pos);  // prev
             rc.rkind[o] = kind;
-            rc.ck[o] = ck;
-            rc.ck_addr[o] = ckaddr;
-            rc.ck_len[o] = cklen;
+            rc.ck_count[o] = (uint8_t)nv;
             rc.flags[o] = 0;
             // deltas: boundary = end then start; bound = single deletion
             int64_t em = (int64_t)uvint_get(base, &pos) + sd.min_ts;
@@ -233,18 +251,19 @@ __global__ void k_parse_rows(const SrcDesc2* srcs, uint32_t n_srcs, uint32_t tot
             for (uint32_t c = 0; c < sp.n_cols; c++) rc.cell_flags[o * sp.n_cols + c] = 0;
         } else {
             // ---- row ----
-            uint64_t ck = 0, ckaddr = 0;
-            uint32_t cklen = 0;
-            if (sp.ck_width) {
-                uvint_get(base, &pos);
-                if (sp.ck_width > 0) {
-                    ck = ck_sortable(base + pos, sp.ck_width);
-                    cklen = (uint32_t)sp.ck_width;
+            if (sp.n_ck) uvint_get(base, &pos);  // 32-batch header
+            for (uint32_t c = 0; c < sp.n_ck; c++) {
+                uint64_t oc = o * sp.n_ck + c;
+                uint32_t cklen;
+                if (sp.ck_w[c] > 0) {
+                    rc.ck[oc] = ck_sortable(base + pos, sp.ck_w[c]);
+                    cklen = (uint32_t)sp.ck_w[c];
                 } else {
                     cklen = (uint32_t)uvint_get(base, &pos);
-                    ck = ck_prefix_var(base + pos, cklen);
+                    rc.ck[oc] = ck_prefix_var(base + pos, cklen);
                 }
-                ckaddr = (uint64_t)(base + pos);
+                rc.ck_addr[oc] = (uint64_t)(base + pos);
+                rc.ck_len[oc] = cklen;
                 pos += cklen;
             }
             uvint_get(base, &pos);  // size
@@ -299,9 +318,7 @@ __global__ void k_parse_rows(const SrcDesc2* srcs, uint32_t n_srcs, uint32_t tot
                 rc.val_len[oc] = vlen;
             }
             rc.rkind[o] = BK_CLUSTERING;
-            rc.ck[o] = ck;
-            rc.ck_addr[o] = ckaddr;
-            rc.ck_len[o] = cklen;
+            rc.ck_count[o] = (uint8_t)sp.n_ck;
             rc.flags[o] = pf;
             rc.live_ts[o] = lts;
             rc.live_ttl[o] = lttl;
@@ -458,20 +475,25 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
 
     uint64_t obase = out_base[g];
     uint32_t ocount = 0;
-    uint64_t cur_ck_addr = 0;
-    uint32_t cur_ck_len = 0;
+    uint64_t cur_fo = 0;  // row index whose clustering the next emit copies
     // cells_written: the row path fills out.cell_* at slot obase+ocount
     // BEFORE calling emit; markers pass false and get all-absent cells
     auto emit = [&](uint8_t kind, uint64_t ck, uint8_t flags, int64_t lts, int32_t lttl,
                     int64_t llet, int64_t rdm, uint32_t rdl, int64_t smf, uint32_t sld,
                     bool cells_written) {
+        (void)ck;
         uint64_t o = obase + ocount++;
         if (!cells_written)
             for (uint32_t c = 0; c < sp.n_cols; c++) out.cell_flags[o * sp.n_cols + c] = 0;
         out.rkind[o] = kind;
-        out.ck[o] = ck;
-        out.ck_addr[o] = cur_ck_addr;
-        out.ck_len[o] = cur_ck_len;
+        uint32_t nv = in.ck_count[cur_fo];
+        out.ck_count[o] = (uint8_t)nv;
+        for (uint32_t c = 0; c < nv; c++) {
+            uint64_t dc = o * sp.n_ck + c, sc = cur_fo * sp.n_ck + c;
+            out.ck[dc] = in.ck[sc];
+            out.ck_addr[dc] = in.ck_addr[sc];
+            out.ck_len[dc] = in.ck_len[sc];
+        }
         out.flags[o] = flags;
         out.live_ts[o] = lts;
         out.live_ttl[o] = lttl;
@@ -493,13 +515,12 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
             uint64_t o = mb[m] + mpos[m];
             if (first < 0) { first = (int)m; continue; }
             uint64_t f = mb[first] + mpos[first];
-            if (pos_cmp(in, o, f, sp.ck_width) < 0) first = (int)m;
+            if (pos_cmp(in, o, f, sp) < 0) first = (int)m;
         }
         if (first < 0) break;
         uint64_t fo = mb[first] + mpos[first];
-        uint64_t fck = in.ck[fo];
-        cur_ck_addr = in.ck_addr[fo];
-        cur_ck_len = in.ck_len[fo];
+        uint64_t fck = in.ck[fo * (sp.n_ck ? sp.n_ck : 1)];
+        cur_fo = fo;
         uint8_t fkind = in.rkind[fo];
         bool is_row = bk_comparison(fkind) == 2;
         // gather members at this position
@@ -509,7 +530,7 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
         for (uint32_t m = 0; m < k; m++) {
             if (mpos[m] >= mcnt[m]) continue;
             uint64_t o = mb[m] + mpos[m];
-            if (pos_cmp(in, o, fo, sp.ck_width) == 0) {
+            if (pos_cmp(in, o, fo, sp) == 0) {
                 members |= 1ULL << m;
                 nmem++;
                 lastm = (int)m;
@@ -878,11 +899,20 @@ __device__ inline uint32_t unf_body_size(const UnfCols& u, uint64_t o, const Ser
 
 // serialized size of a ClusteringPrefix (full serializer: kind byte [+ u16
 // size for bounds] + values-without-size) — IndexInfo first/last names
-__device__ inline uint32_t prefix_full_size(uint8_t kind, const SchemaParams& sch, uint32_t vlen) {
+__device__ inline uint32_t prefix_full_size(const UnfCols& u, uint64_t o,
+                                            const SchemaParams& sch) {
+    uint8_t kind = u.rkind[o];
+    uint32_t nv = u.ck_count[o];
     uint32_t s = 1;                       // kind byte
     if (kind != BK_CLUSTERING) s += 2;    // u16 value count
-    if (sch.ck_width > 0) s += 1 + sch.ck_width;           // header vint + fixed value
-    else if (sch.ck_width < 0) s += 1 + uvint_size(vlen) + vlen;  // header + vint len + bytes
+    if (nv) s += 1;                       // 32-batch header vint (all non-null)
+    for (uint32_t c = 0; c < nv; c++) {
+        if (sch.ck_w[c] > 0) s += sch.ck_w[c];
+        else {
+            uint32_t vlen = u.ck_len[o * sch.n_ck + c];
+            s += uvint_size(vlen) + vlen;
+        }
+    }
     return s;
 }
 __device__ inline uint32_t dt_ser_size(int64_t m, uint32_t l) {
@@ -934,25 +964,26 @@ __device__ uint64_t part_walk(const OutParts& op, const UnfCols& out, uint64_t g
         pos += 12;
     };
     auto emit_ck = [&](uint64_t o) {
-        if (sp.sch.ck_width == 0) return;
-        if (sp.sch.ck_width > 0) {
-            if (EMIT && lane == 0) {
-                out_data[data_off + pos] = 0;  // 32-batch header: one non-null value
-                ck_bytes(out.ck[o], sp.sch.ck_width, &out_data[data_off + pos + 1]);
+        uint32_t nv = out.ck_count[o];
+        if (nv == 0) return;
+        if (EMIT && lane == 0) out_data[data_off + pos] = 0;  // header: all non-null
+        pos += 1;
+        for (uint32_t c = 0; c < nv; c++) {
+            uint64_t oc = o * sp.sch.n_ck + c;
+            if (sp.sch.ck_w[c] > 0) {
+                if (EMIT && lane == 0) ck_bytes(out.ck[oc], sp.sch.ck_w[c], &out_data[data_off + pos]);
+                pos += sp.sch.ck_w[c];
+            } else {
+                uint32_t vlen = out.ck_len[oc];
+                if (EMIT && lane == 0) {
+                    uint8_t tmp[9];
+                    int n = uvint_put(tmp, vlen);
+                    for (int i = 0; i < n; i++) out_data[data_off + pos + i] = tmp[i];
+                    const uint8_t* src = (const uint8_t*)out.ck_addr[oc];
+                    for (uint32_t i = 0; i < vlen; i++) out_data[data_off + pos + n + i] = src[i];
+                }
+                pos += uvint_size(vlen) + vlen;
             }
-            pos += 1 + sp.sch.ck_width;
-        } else {
-            uint32_t vlen = out.ck_len[o];
-            if (EMIT && lane == 0) {
-                out_data[data_off + pos] = 0;
-                uint8_t tmp[9];
-                int n = uvint_put(tmp, vlen);
-                for (int i = 0; i < n; i++) out_data[data_off + pos + 1 + i] = tmp[i];
-                const uint8_t* src = (const uint8_t*)out.ck_addr[o];
-                for (uint32_t i = 0; i < vlen; i++)
-                    out_data[data_off + pos + 1 + n + i] = src[i];
-            }
-            pos += 1 + uvint_size(vlen) + vlen;
         }
     };
 
@@ -985,39 +1016,41 @@ __device__ uint64_t part_walk(const OutParts& op, const UnfCols& out, uint64_t g
 
     auto emit_prefix_idx = [&](uint64_t o, uint64_t q) -> uint64_t {
         uint8_t kk = out.rkind[o];
+        uint32_t nv = out.ck_count[o];
         if (EMIT && lane == 0) out_index[q] = kk;
         q++;
         if (kk != BK_CLUSTERING) {
-            if (EMIT && lane == 0) { out_index[q] = 0; out_index[q + 1] = (uint8_t)(sp.sch.ck_width ? 1 : 0); }
+            if (EMIT && lane == 0) { out_index[q] = 0; out_index[q + 1] = (uint8_t)nv; }
             q += 2;
         }
-        if (sp.sch.ck_width > 0) {
-            if (EMIT && lane == 0) {
-                out_index[q] = 0;
-                ck_bytes(out.ck[o], sp.sch.ck_width, &out_index[q + 1]);
+        if (nv) {
+            if (EMIT && lane == 0) out_index[q] = 0;
+            q += 1;
+            for (uint32_t c = 0; c < nv; c++) {
+                uint64_t oc = o * sp.sch.n_ck + c;
+                if (sp.sch.ck_w[c] > 0) {
+                    if (EMIT && lane == 0) ck_bytes(out.ck[oc], sp.sch.ck_w[c], &out_index[q]);
+                    q += sp.sch.ck_w[c];
+                } else {
+                    uint32_t vlen = out.ck_len[oc];
+                    if (EMIT && lane == 0) {
+                        uint8_t tmp[9];
+                        int n = uvint_put(tmp, vlen);
+                        for (int i = 0; i < n; i++) out_index[q + i] = tmp[i];
+                        const uint8_t* src = (const uint8_t*)out.ck_addr[oc];
+                        for (uint32_t i = 0; i < vlen; i++) out_index[q + n + i] = src[i];
+                    }
+                    q += uvint_size(vlen) + vlen;
+                }
             }
-            q += 1 + sp.sch.ck_width;
-        } else if (sp.sch.ck_width < 0) {
-            uint32_t vlen = out.ck_len[o];
-            if (EMIT && lane == 0) {
-                out_index[q] = 0;
-                uint8_t tmp[9];
-                int n = uvint_put(tmp, vlen);
-                for (int i = 0; i < n; i++) out_index[q + 1 + i] = tmp[i];
-                const uint8_t* src = (const uint8_t*)out.ck_addr[o];
-                for (uint32_t i = 0; i < vlen; i++) out_index[q + 1 + n + i] = src[i];
-            }
-            q += 1 + uvint_size(vlen) + vlen;
         }
         return q;
     };
     auto flush_block = [&](uint64_t end_pos) {
         uint64_t width = end_pos - block_start;
         bool has_open = !(open_m == INT64_MIN && open_l == LDT_NONE_U32);
-        uint64_t isz = prefix_full_size(out.rkind[block_first_o], sp.sch,
-                                        sp.sch.ck_width < 0 ? out.ck_len[block_first_o] : 0) +
-                       prefix_full_size(out.rkind[last_o], sp.sch,
-                                        sp.sch.ck_width < 0 ? out.ck_len[last_o] : 0) +
+        uint64_t isz = prefix_full_size(out, block_first_o, sp.sch) +
+                       prefix_full_size(out, last_o, sp.sch) +
                        uvint_size(block_start) + uvint_size(zigzag((int64_t)width - 65536)) + 1 +
                        (has_open ? dt_ser_size(open_m, open_l) : 0);
         if (EMIT && nblocks_hint > 1) {
@@ -1075,7 +1108,7 @@ __device__ uint64_t part_walk(const OutParts& op, const UnfCols& out, uint64_t g
             emit8(0x02);
             emit8(kind);
             emit8(0);
-            emit8(sp.sch.ck_width ? 1 : 0);
+            emit8(out.ck_count[o]);
             emit_ck(o);
             emit_uv(body + uvint_size(prev_sz));
             emit_uv(prev_sz);
@@ -1432,6 +1465,7 @@ struct GenParams2 {
     uint32_t clustering_rows, range_tomb_pct;
     uint32_t key_len;   // 8 (default) .. 255; bytes 8.. are gen2_key_salt(id, j)
     uint32_t ck_text;   // clustering values as UTF8 strings (oracle gen_ck_bytes)
+    uint32_t ck_cols;   // 0/1 = one clustering column; 2 = (bigint, bigint)
     uint32_t n_value_cols;    // regular columns val0..valN-1 (1..63)
     uint32_t col_missing_pct; // P(cell absent) per live row and column
     int64_t base_ts, base_ldt;
@@ -1487,7 +1521,10 @@ __global__ void k_gen_count(GenParams2 gp, const MRec* sorted, const uint64_t* i
         if (gp.range_tomb_pct && (splitmix64(gp.seed ^ 0xBEEFULL ^ id ^ ((uint64_t)gp.sst << 32)) % 100 < gp.range_tomb_pct)) {
             uint64_t r = splitmix64(gp.seed ^ 0xB00BULL ^ id ^ ((uint64_t)gp.sst << 32));
             uint32_t a = (uint32_t)(r % gp.clustering_rows);
-            if (a + 1 < gp.clustering_rows) cnt += 2;  // one open + one close bound
+            // 1-col: bounds sit between rows, so both are emitted only when a
+            // row follows the open position; 2-col prefix bounds always open
+            // before their ck0 group's first row
+            if (gp.ck_cols == 2 || a + 1 < gp.clustering_rows) cnt += 2;
         }
     }
     prow_count[i] = cnt;
@@ -1506,18 +1543,26 @@ __global__ void k_gen_fill2(GenParams2 gp, const MRec* sorted, const uint64_t* i
     uint64_t ob = row_base[i];
     op.row_base[i] = ob;
     uint32_t emitted = 0;
-    auto put_ck = [&](uint64_t o, int64_t ck, bool has_ck, bool is_row) {
-        if (!has_ck) { out.ck[o] = 0; out.ck_addr[o] = 0; out.ck_len[o] = 0; return; }
-        if (gp.ck_text) {
-            uint8_t* dst = ck_arena + o * 16;
-            uint32_t len = gen2_ck_text(ck, is_row, dst);
-            out.ck[o] = ck_prefix_var(dst, len);
-            out.ck_addr[o] = (uint64_t)dst;
-            out.ck_len[o] = len;
-        } else {
-            out.ck[o] = (uint64_t)ck ^ 0x8000000000000000ULL;
-            out.ck_addr[o] = 0;
-            out.ck_len[o] = 8;
+    const uint32_t NCK = gp.ck_cols ? gp.ck_cols : (gp.clustering_rows ? 1 : 0);
+    // ck encodes the full position; ck_cols==2 splits it as (ck/64, ck%64)
+    // scaled by 16 per component (oracle gen_ck2). nv < NCK = prefix bound.
+    auto put_ck = [&](uint64_t o, int64_t ck, bool has_ck, bool is_row, uint32_t nv) {
+        if (!has_ck || NCK == 0) { out.ck_count[o] = 0; return; }
+        out.ck_count[o] = (uint8_t)nv;
+        for (uint32_t c = 0; c < nv; c++) {
+            uint64_t oc = o * NCK + c;
+            int64_t comp = NCK == 2 ? (c == 0 ? (ck / 64) * 16 : (ck % 64)) : ck;
+            if (gp.ck_text) {
+                uint8_t* dst = ck_arena + oc * 16;
+                uint32_t len = gen2_ck_text(comp, is_row, dst);
+                out.ck[oc] = ck_prefix_var(dst, len);
+                out.ck_addr[oc] = (uint64_t)dst;
+                out.ck_len[oc] = len;
+            } else {
+                out.ck[oc] = (uint64_t)comp ^ 0x8000000000000000ULL;
+                out.ck_addr[oc] = 0;
+                out.ck_len[oc] = 8;
+            }
         }
     };
     const uint32_t NCV = gp.n_value_cols;
@@ -1525,7 +1570,7 @@ __global__ void k_gen_fill2(GenParams2 gp, const MRec* sorted, const uint64_t* i
         uint64_t o = ob + emitted++;
         for (uint32_t c = 0; c < NCV; c++) out.cell_flags[o * NCV + c] = 0;
         out.rkind[o] = kind;
-        put_ck(o, ck, true, false);
+        put_ck(o, ck, true, false, NCK == 2 ? 1 : NCK);  // 2-col mode: prefix bounds
         out.flags[o] = 0;
         out.live_ts[o] = NO_TIMESTAMP;
         out.live_ttl[o] = 0;
@@ -1544,7 +1589,7 @@ __global__ void k_gen_fill2(GenParams2 gp, const MRec* sorted, const uint64_t* i
                        uint32_t rowj) {
         uint64_t o = ob + emitted++;
         out.rkind[o] = BK_CLUSTERING;
-        put_ck(o, ck, has_ck, true);
+        put_ck(o, ck, has_ck, true, NCK);
         out.start_mfda[o] = INT64_MIN;
         out.start_ldt[o] = LDT_NONE_U32;
         out.live_ttl[o] = 0;
@@ -1599,22 +1644,31 @@ __global__ void k_gen_fill2(GenParams2 gp, const MRec* sorted, const uint64_t* i
             uint64_t r = splitmix64(gp.seed ^ 0xB00BULL ^ id ^ ((uint64_t)gp.sst << 32));
             uint32_t a = (uint32_t)(r % gp.clustering_rows);
             uint32_t b = a + 1 + (uint32_t)((r >> 32) % (gp.clustering_rows - a));
-            rlo = (int64_t)a * 16 + 8;
-            rhi = (int64_t)b * 16 + 8;
+            // 1-col: bounds between rows (a*16+8); 2-col: PREFIX bounds on the
+            // first component covering whole ck0 groups (oracle gen mirror)
+            rlo = gp.ck_cols == 2 ? (int64_t)(a / 4) * 64 : (int64_t)a * 16 + 8;
+            rhi = gp.ck_cols == 2 ? (int64_t)(b / 4) * 64 + 63 : (int64_t)b * 16 + 8;
             rts = gp.base_ts + (int64_t)(splitmix64(gp.seed ^ 0xAB1EULL ^ id ^ ((uint64_t)gp.sst << 48)) % 1000000000ULL);
             rldt = gen2_ldt(gp, id, 0xCC);
         }
         bool rt_open = false, rt_done = false;
         for (uint32_t j = 0; j < gp.clustering_rows; j++) {
-            int64_t ck = (int64_t)j * 16;
-            if (has_rt && !rt_open && !rt_done && ck > rlo) {
+            int64_t ck = gp.ck_cols == 2 ? (int64_t)(j / 4) * 64 + (int64_t)(j % 4)
+                                         : (int64_t)j * 16;
+            // 2-col mode: the START bound is a ck0-group PREFIX and sorts
+            // before the group's first row, so it opens on >= (1-col bounds
+            // sit between rows and open on >)
+            if (has_rt && !rt_open && !rt_done &&
+                (gp.ck_cols == 2 ? ck >= rlo : ck > rlo)) {
                 put_marker(BK_INCL_START, rlo, rts, rldt);
                 rt_open = true;
             }
             int64_t ts = gp.base_ts + (int64_t)(splitmix64(gp.seed ^ id * 31 ^ ((uint64_t)gp.sst << 48) ^ (uint64_t)(j + 1) * 0x9E37ULL) % 1000000000ULL);
             bool tomb = gp.tombstone_pct && (splitmix64(gp.seed ^ 0xDEADULL ^ id ^ ((uint64_t)gp.sst << 32) ^ (uint64_t)(j + 7) * 131) % 100 < gp.tombstone_pct);
             put_row(ck, true, ts, tomb, tomb ? gen2_ldt(gp, id * 1000 + j, 0xEE) : 0, j);
-            if (has_rt && rt_open && j + 1 < gp.clustering_rows && ((int64_t)(j + 1) * 16) > rhi) {
+            int64_t ck_next = gp.ck_cols == 2 ? (int64_t)((j + 1) / 4) * 64 + (int64_t)((j + 1) % 4)
+                                              : (int64_t)(j + 1) * 16;
+            if (has_rt && rt_open && j + 1 < gp.clustering_rows && ck_next > rhi) {
                 put_marker(BK_INCL_END, rhi, rts, rldt);
                 rt_open = false;
                 rt_done = true;

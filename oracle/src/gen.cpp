@@ -8,8 +8,10 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
     SSTable t;
     t.generation = g.first_generation + sst;
     t.header.key_type = g.key_len > 8 ? CqlType::BYTES : CqlType::LONG;  // pk bigint / blob
-    if (g.clustering_rows > 0)
-        t.header.clustering_types = {g.ck_text ? CqlType::UTF8 : CqlType::LONG};
+    if (g.clustering_rows > 0) {
+        uint32_t nck = g.ck_cols ? g.ck_cols : 1;
+        t.header.clustering_types.assign(nck, g.ck_text ? CqlType::UTF8 : CqlType::LONG);
+    }
     if (g.n_value_cols <= 1) {
         t.header.regular_cols = {{bytes{'v', 'a', 'l'}, CqlType::BYTES}};  // val blob
     } else {
@@ -55,14 +57,35 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
                 min_ts = std::min(min_ts, rts);
                 min_ldt_l = std::min<int64_t>(min_ldt_l, rldt);
             }
+            bool ck2 = g.ck_cols == 2;
+            if (ck2 && has_rt) {
+                // ck0-group PREFIX bounds (GPU mirror: k_gen_fill2 ck_cols==2)
+                uint64_t r2 = splitmix64(g.seed ^ 0xB00BULL ^ e.id ^ ((uint64_t)sst << 32));
+                uint32_t a2 = (uint32_t)(r2 % g.clustering_rows);
+                uint32_t b2 = a2 + 1 + (uint32_t)((r2 >> 32) % (g.clustering_rows - a2));
+                rlo = (int64_t)(a2 / 4) * 64;
+                rhi = (int64_t)(b2 / 4) * 64 + 63;
+            }
+            auto ck_vals = [&](int64_t ckpos, bool is_row) {
+                Clustering cv;
+                if (!ck2) {
+                    cv = {ClusterVal{ClusterVal::VALUE, gen_ck_bytes(g, ckpos, is_row)}};
+                } else if (is_row) {
+                    cv = {ClusterVal{ClusterVal::VALUE, gen_ck_bytes(g, (ckpos / 64) * 16, true)},
+                          ClusterVal{ClusterVal::VALUE, gen_ck_bytes(g, ckpos % 64, true)}};
+                } else {
+                    cv = {ClusterVal{ClusterVal::VALUE, gen_ck_bytes(g, (ckpos / 64) * 16, false)}};
+                }
+                return cv;
+            };
             bool rt_open = false, rt_done = false;
             for (uint32_t j = 0; j < g.clustering_rows; j++) {
-                int64_t ck = gen_ck(g, e.id, j);
-                if (has_rt && !rt_open && !rt_done && ck > rlo) {
+                int64_t ck = ck2 ? (int64_t)(j / 4) * 64 + (int64_t)(j % 4) : gen_ck(g, e.id, j);
+                if (has_rt && !rt_open && !rt_done && (ck2 ? ck >= rlo : ck > rlo)) {
                     Unfiltered u;
                     u.kind = Unfiltered::MARKER;
                     u.marker.kind = INCL_START;
-                    u.marker.values = {ClusterVal{ClusterVal::VALUE, gen_ck_bytes(g, rlo, false)}};
+                    u.marker.values = ck_vals(rlo, false);
                     u.marker.end_dt = DeletionTime{rts, rldt};
                     // marker sits at bound position rlo (before this row)
                     p.items.push_back(std::move(u));
@@ -71,7 +94,7 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
                 Unfiltered u;
                 u.kind = Unfiltered::ROW;
                 Row& r = u.row;
-                r.clustering = {ClusterVal{ClusterVal::VALUE, gen_ck_bytes(g, ck, true)}};
+                r.clustering = ck_vals(ck, true);
                 uint32_t ncols = g.n_value_cols ? g.n_value_cols : 1;
                 r.cells.resize(ncols);
                 int64_t ts = gen_row_ts(g, sst, e.id, j);
@@ -92,12 +115,13 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
                     }
                 }
                 p.items.push_back(std::move(u));
-                if (has_rt && rt_open && j + 1 < g.clustering_rows &&
-                    gen_ck(g, e.id, j + 1) > rhi) {
+                int64_t ck_next = ck2 ? (int64_t)((j + 1) / 4) * 64 + (int64_t)((j + 1) % 4)
+                                      : gen_ck(g, e.id, j + 1);
+                if (has_rt && rt_open && j + 1 < g.clustering_rows && ck_next > rhi) {
                     Unfiltered m;
                     m.kind = Unfiltered::MARKER;
                     m.marker.kind = INCL_END;
-                    m.marker.values = {ClusterVal{ClusterVal::VALUE, gen_ck_bytes(g, rhi, false)}};
+                    m.marker.values = ck_vals(rhi, false);
                     m.marker.end_dt = DeletionTime{rts, rldt};
                     p.items.push_back(std::move(m));
                     rt_open = false;
@@ -108,7 +132,7 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
                 Unfiltered m;
                 m.kind = Unfiltered::MARKER;
                 m.marker.kind = INCL_END;
-                m.marker.values = {ClusterVal{ClusterVal::VALUE, gen_ck_bytes(g, rhi, false)}};
+                m.marker.values = ck_vals(rhi, false);
                 m.marker.end_dt = DeletionTime{rts, rldt};
                 p.items.push_back(std::move(m));
             }

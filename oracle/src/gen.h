@@ -36,6 +36,10 @@ struct GenSpec {
     // order) plus, for ROW values only, (j %% 3) 'x' suffix bytes to
     // exercise variable width. GPU mirror: gen2_ck_text.
     uint32_t ck_text = 0;
+    // ck_cols=2: composite clustering (bigint ck0, bigint ck1); rows are
+    // (16*(j/4), j%4); range tombstones become ck0-PREFIX bounds covering
+    // whole ck0 groups (GPU mirror: GenParams2.ck_cols)
+    uint32_t ck_cols = 0;
     // partition key width in bytes, 8..255: first 8 = big-endian key id,
     // bytes 8.. = splitmix64(id ^ (0xC0FFEE5EED + j)) (GPU: gen2_key_salt).
     // key_len > 8 switches the declared key type LongType -> BytesType.

@@ -74,6 +74,21 @@ def test_generator_parity_wide(ca, oracle_bin, tmp_path):
         _assert_dirs_equal(f"{dg}/oa-{g}-big", f"{do}/oa-{g}-big")
 
 
+def test_generator_parity_composite_ck(ca, oracle_bin, tmp_path):
+    """Composite (2-column) clustering keys with ck0-prefix range-tombstone
+    bounds: write-path parity."""
+    dg, do = str(tmp_path / "gpu"), str(tmp_path / "cpu")
+    os.makedirs(dg), os.makedirs(do)
+    kw = dict(seed=104, n=2, rows=40, crows=48, vlen=300, overlap=30, tomb=10, rtomb=40,
+              ckcols=2)
+    ca.generate(dg, seed=104, n_sstables=2, rows_per_sstable=40, clustering_rows=48,
+                value_len=300, overlap_pct=30, tombstone_pct=10, range_tomb_pct=40,
+                ck_cols=2)
+    _oracle_gen(do, **kw)
+    for g in (1, 2):
+        _assert_dirs_equal(f"{dg}/oa-{g}-big", f"{do}/oa-{g}-big")
+
+
 def test_generator_parity_multi_column(ca, oracle_bin, tmp_path):
     """N regular columns + per-cell subset bitmaps: write-path parity."""
     dg, do = str(tmp_path / "gpu"), str(tmp_path / "cpu")
@@ -171,6 +186,18 @@ def test_generator_parity_long_keys(ca, oracle_bin, tmp_path):
     dict(name="multi_column_everything", gen=dict(seed=85, n=3, rows=30, crows=40, vlen=180,
                                                   overlap=30, tomb=12, rtomb=25, ncols=4,
                                                   colmiss=20, cktext=1, keylen=24), job={}),
+    dict(name="composite_ck", gen=dict(seed=101, n=3, rows=40, crows=48, vlen=250,
+                                       overlap=30, tomb=10, ckcols=2), job={}),
+    dict(name="composite_ck_range_tombs", gen=dict(seed=102, n=3, rows=40, crows=48,
+                                                   vlen=200, overlap=30, tomb=10, rtomb=40,
+                                                   ckcols=2), job={}),
+    dict(name="composite_ck_gc", gen=dict(seed=102, n=3, rows=40, crows=48, vlen=200,
+                                          overlap=30, tomb=10, rtomb=40, ckcols=2),
+         job=dict(gc_before=2000000000)),
+    dict(name="composite_ck_kitchen_sink", gen=dict(seed=103, n=3, rows=24, crows=40,
+                                                    vlen=180, overlap=30, tomb=12, rtomb=30,
+                                                    ckcols=2, ncols=3, colmiss=20, keylen=24),
+         job={}),
 ])
 def test_compaction_parity(ca, oracle_bin, tmp_path, case):
     d = str(tmp_path)

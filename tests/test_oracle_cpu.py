@@ -166,3 +166,15 @@ def test_multi_column_roundtrip_and_merge(oracle_bin, tmp_path):
     assert stats["partitions_out"] > 0
     r = oracle_run("roundtrip", f"{d}/oa-50-big")
     assert "FAIL" not in r.stdout, r.stdout
+
+
+def test_composite_clustering_roundtrip_and_merge(oracle_bin, tmp_path):
+    """Composite clustering (2 bigint columns) + ck0-prefix range tombstones."""
+    d = str(tmp_path)
+    _gen(d, n=3, rows=40, crows=48, vlen=150, overlap=30, tomb=10, rtomb=40,
+         ckcols=2, seed=105)
+    ins = [f"{d}/oa-{g}-big" for g in (1, 2, 3)]
+    stats = _compact(f"{d}/oa-50-big", ins)
+    assert stats["partitions_out"] > 0 and stats["rows_out"] > 0
+    r = oracle_run("roundtrip", f"{d}/oa-50-big")
+    assert "FAIL" not in r.stdout, r.stdout
