@@ -4,8 +4,9 @@
  * Drop-in boundary: replaces the hot loop of CompactionTask.runMayThrow
  * (reference: src/java/org/apache/cassandra/db/compaction/CompactionTask.java:184-236)
  * — scanners + CompactionIterator + CompactionAwareWriter — for eligible
- * tables (big-format `oa`, Murmur3Partitioner, any partition key, at most one
- * clustering column of fixed or variable width, 1..63 regular columns with
+ * tables (big-format `oa`, Murmur3Partitioner, any partition key, up to 32
+ * clustering columns of fixed (bigint/int) or variable (text/ascii/blob)
+ * width including prefix range-tombstone bounds, 1..63 regular columns with
  * cell subsets, LZ4 chunk compression; row, cell, partition and range
  * tombstones all supported). The Java host above the seam
  * (strategies, CompactionManager, LifecycleTransaction, metrics) is
