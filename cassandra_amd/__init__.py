@@ -82,6 +82,7 @@ class GpucGenSpec(ctypes.Structure):
         ("key_len", ctypes.c_uint32),
         ("ck_text", ctypes.c_uint32),
         ("ck_cols", ctypes.c_uint32),
+        ("static_pct", ctypes.c_uint32),
         ("n_value_cols", ctypes.c_uint32),
         ("col_missing_pct", ctypes.c_uint32),
         ("base_ts", ctypes.c_int64),
@@ -233,6 +234,7 @@ def generate(
     key_len=8,
     ck_text=False,
     ck_cols=0,
+    static_pct=0,
     n_value_cols=1,
     col_missing_pct=0,
     base_ts=1700000000000000,
@@ -256,6 +258,7 @@ def generate(
         key_len=key_len,
         ck_text=1 if ck_text else 0,
         ck_cols=ck_cols,
+        static_pct=static_pct,
         n_value_cols=n_value_cols,
         col_missing_pct=col_missing_pct,
         base_ts=base_ts,

@@ -111,6 +111,25 @@ __host__ __device__ inline bool bk_is_open(uint8_t k) {
     return k == BK_INCL_START || k == BK_EXCL_START || bk_is_boundary(k);
 }
 
+// one static row, stored at PARTITION level (SortedTablePartitionWriter
+// writes it between the partition deletion and the unfiltereds whenever the
+// header has static columns; it never interacts with the marker machinery).
+// Cells strided by n_static: index = part * n_static + c.
+struct StaticCols {
+    uint8_t* flags;       // PF_HAS_ROW | PF_LIVE_TS | PF_ROW_DEL (0 = empty row)
+    int64_t* live_ts;
+    int32_t* live_ttl;
+    int64_t* live_let;
+    int64_t* rdel_mfda;
+    uint32_t* rdel_ldt;
+    uint8_t* cell_flags;  // CELLF_* per static cell
+    int64_t* cell_ts;
+    uint32_t* cell_ldt;
+    int32_t* cell_ttl;
+    uint64_t* val_addr;
+    uint32_t* val_len;
+};
+
 // parsed per input-partition fields (partition level)
 struct ParsedCols {
     int64_t* pdel_mfda;
@@ -118,12 +137,14 @@ struct ParsedCols {
     uint32_t* row_count;  // unfiltereds in this partition
     uint64_t* row_base;   // start into the per-source-concatenated UnfCols
     uint64_t* key_addr;   // device address of the key bytes in input data
+    StaticCols st;        // static row per partition (when the schema has one)
 };
 
 // reconciled output partitions (also produced directly by the generator)
 struct OutParts {
     uint64_t* keypfx;    // big-endian zero-padded key prefix (debug/aux)
     uint64_t* key_addr;  // device address of the full key bytes
+    StaticCols st;       // merged static row per output partition
     uint16_t* klen;
     int64_t* pdel_mfda;
     uint32_t* pdel_ldt;

@@ -364,6 +364,7 @@ struct OutMeta {
     HeaderStats hs;
     std::string key_type;
     std::vector<std::string> ck_types;  // empty = no clustering columns
+    std::vector<std::pair<bytes, std::string>> static_cols;
     std::vector<std::pair<bytes, std::string>> regular_cols;
     // collected stats
     int64_t min_timestamp, max_timestamp, min_ldt, max_ldt;
@@ -437,7 +438,12 @@ static bytes serialize_statistics_out(const OutMeta& m) {
         put_type_str(header, m.key_type);
         put_uvint(header, m.ck_types.size());
         for (auto& t : m.ck_types) put_type_str(header, t);
-        put_uvint(header, 0);  // static columns
+        put_uvint(header, m.static_cols.size());
+        for (auto& [name, t] : m.static_cols) {
+            put_uvint(header, name.size());
+            header.insert(header.end(), name.begin(), name.end());
+            put_type_str(header, t);
+        }
         put_uvint(header, m.regular_cols.size());
         for (auto& [name, t] : m.regular_cols) {
             put_uvint(header, name.size());
@@ -668,14 +674,35 @@ static void ensure_crc_tables(hipStream_t stream) {
 }
 
 // allocate the OutParts SoA for n entries
+// allocate a StaticCols block for n partitions x n_static columns
+struct StaticColsBuf {
+    DevBuf flags, live_ts, live_ttl, live_let, rdel_mfda, rdel_ldt,
+        cell_flags, cell_ts, cell_ldt, cell_ttl, val_addr, val_len;
+    StaticCols st{};
+    void alloc(uint64_t n, uint32_t n_static) {
+        if (!n) n = 1;
+        uint64_t ns = n * (n_static ? n_static : 1);
+        flags.alloc(n); live_ts.alloc(n * 8); live_ttl.alloc(n * 4); live_let.alloc(n * 8);
+        rdel_mfda.alloc(n * 8); rdel_ldt.alloc(n * 4);
+        cell_flags.alloc(ns); cell_ts.alloc(ns * 8); cell_ldt.alloc(ns * 4);
+        cell_ttl.alloc(ns * 4); val_addr.alloc(ns * 8); val_len.alloc(ns * 4);
+        st = StaticCols{flags.as<uint8_t>(), live_ts.as<int64_t>(), live_ttl.as<int32_t>(),
+                        live_let.as<int64_t>(), rdel_mfda.as<int64_t>(), rdel_ldt.as<uint32_t>(),
+                        cell_flags.as<uint8_t>(), cell_ts.as<int64_t>(), cell_ldt.as<uint32_t>(),
+                        cell_ttl.as<int32_t>(), val_addr.as<uint64_t>(), val_len.as<uint32_t>()};
+    }
+};
+
 struct OutPartsBuf {
     DevBuf keypfx, key_addr, klen, pdel_mfda, pdel_ldt, row_base, row_count, keep;
+    StaticColsBuf stb;
     OutParts op{};
-    void alloc(uint64_t n) {
+    void alloc(uint64_t n, uint32_t n_static = 0) {
         keypfx.alloc(n * 8); key_addr.alloc(n * 8); klen.alloc(n * 2);
         pdel_mfda.alloc(n * 8); pdel_ldt.alloc(n * 4);
         row_base.alloc(n * 8); row_count.alloc(n * 4); keep.alloc(n);
-        op = OutParts{keypfx.as<uint64_t>(), key_addr.as<uint64_t>(), klen.as<uint16_t>(),
+        stb.alloc(n_static ? n : 1, n_static);
+        op = OutParts{keypfx.as<uint64_t>(), key_addr.as<uint64_t>(), stb.st, klen.as<uint16_t>(),
                       pdel_mfda.as<int64_t>(),
                       pdel_ldt.as<uint32_t>(), row_base.as<uint64_t>(), row_count.as<uint32_t>(),
                       keep.as<uint8_t>()};
@@ -725,6 +752,7 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
                                            const std::string& key_type,
                                            const std::vector<std::string>& ck_types,
                                            const std::vector<std::pair<bytes, std::string>>& regular_cols,
+                                           const std::vector<std::pair<bytes, std::string>>& static_cols,
                                            hipStream_t stream, int wslot = 0) {
     WriteDeviceOut w;
     TR("wsd: enter");
@@ -947,6 +975,7 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
         m.key_type = key_type;
         m.ck_types = ck_types;
         m.regular_cols = regular_cols;
+        m.static_cols = static_cols;
         bool no_ts = hst.min_ts_flip == 0xFFFFFFFFFFFFFFFFULL;
         m.min_timestamp = no_ts ? 0 : (int64_t)(hst.min_ts_flip ^ 0x8000000000000000ULL);
         m.max_timestamp = no_ts ? 0 : (int64_t)(hst.max_ts_flip ^ 0x8000000000000000ULL);
@@ -1111,6 +1140,7 @@ struct CompactSetup {
     std::vector<size_t> comp_file_sz;
     std::vector<int32_t> col_fixed_h;
     std::vector<int32_t> ck_widths;      // per clustering column
+    std::vector<int32_t> static_fixed_h; // per static column
     // unsharded fast path: whole-file Data.db reads started during index
     // parse (compact_one adopts them when its window covers the full file)
     mutable std::vector<std::thread> full_readers;
@@ -1293,6 +1323,13 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         HIP_CHECK(hipMemcpyAsync(d_col_fixed.p, col_fixed_h.data(), col_fixed_h.size() * 4,
                                  hipMemcpyHostToDevice, stream));
         sch.col_fixed = d_col_fixed.as<int32_t>();
+        sch.n_static = (uint32_t)su.static_fixed_h.size();
+        DevBuf d_static_fixed;
+        d_static_fixed.alloc(su.static_fixed_h.size() * 4 + 8);
+        if (sch.n_static)
+            HIP_CHECK(hipMemcpyAsync(d_static_fixed.p, su.static_fixed_h.data(),
+                                     su.static_fixed_h.size() * 4, hipMemcpyHostToDevice, stream));
+        sch.static_fixed = d_static_fixed.as<int32_t>();
         sch.column_index_size = 64 * 1024;
         DevBuf d_srcs, d_recs_a, d_recs_b, d_rows_in;
         d_srcs.alloc(srcs.size() * sizeof(SrcDesc2));
@@ -1310,6 +1347,9 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         p_rbase.alloc(total_parts * 8); pc.row_base = p_rbase.as<uint64_t>();
         DevBuf p_kaddr;
         p_kaddr.alloc(total_parts * 8); pc.key_addr = p_kaddr.as<uint64_t>();
+        StaticColsBuf p_static;
+        p_static.alloc(sch.n_static ? total_parts : 1, sch.n_static);
+        pc.st = p_static.st;
         {
             uint32_t blocks = (uint32_t)((total_parts + 255) / 256);
             hipLaunchKernelGGL(k_parse_count, dim3(blocks), dim3(256), 0, stream,
@@ -1397,7 +1437,7 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         uint64_t total_out_rows = exscan_u64(d_group_rows.as<uint64_t>(), n_groups, stream);
         TR("out rows scanned");
         OutPartsBuf opb;
-        opb.alloc(n_groups);
+        opb.alloc(n_groups, sch.n_static);
         UnfColsBuf out_rows;
         out_rows.alloc(total_out_rows, sch.n_cols, sch.n_ck);
         DevBuf d_stats, d_tomb;
@@ -1481,7 +1521,7 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         WriteDeviceOut w = write_sstable_device(opb, out_rows, n_groups, sp2, d_stats, d_tomb,
                                                 tomb_cap, out_base_str, stats[0].key_type,
                                                 stats[0].clustering_types, stats[0].regular_cols,
-                                                stream, wslot);
+                                                stats[0].static_cols, stream, wslot);
         TR("writer done");
         {
             OutStats hst;
@@ -1601,6 +1641,18 @@ static void compact_setup(const gpuc_job* job, CompactSetup& su, bool preread_fu
         else throw std::runtime_error("unsupported column type " + ct);
     }
     for (auto& t : su.stats[0].clustering_types) su.ck_widths.push_back(ck_type_width(t));
+    for (auto& [nm2, ct2] : su.stats[0].static_cols) {
+        (void)nm2;
+        if (ct2 == "org.apache.cassandra.db.marshal.LongType") su.static_fixed_h.push_back(8);
+        else if (ct2 == "org.apache.cassandra.db.marshal.Int32Type") su.static_fixed_h.push_back(4);
+        else if (ct2 == "org.apache.cassandra.db.marshal.BytesType" ||
+                 ct2 == "org.apache.cassandra.db.marshal.UTF8Type" ||
+                 ct2 == "org.apache.cassandra.db.marshal.AsciiType")
+            su.static_fixed_h.push_back(-1);
+        else throw std::runtime_error("unsupported static column type " + ct2);
+    }
+    if (su.static_fixed_h.size() > 63)
+        throw std::runtime_error("1..63 static columns supported");
     if (su.ck_widths.size() > 32)
         throw std::runtime_error("at most 32 clustering columns supported");
     {
@@ -1927,6 +1979,7 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
             gp.ck_text = spec->ck_text;
             gp.n_value_cols = spec->n_value_cols ? spec->n_value_cols : 1;
             gp.ck_cols = spec->ck_cols;
+            gp.static_pct = spec->static_pct;
             if (gp.ck_cols > 2) throw std::runtime_error("ck_cols must be 0..2");
             if (gp.ck_cols == 2 && gp.ck_text) throw std::runtime_error("ck_cols=2 with ck_text unsupported");
             gp.col_missing_pct = spec->col_missing_pct;
@@ -1946,7 +1999,7 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
                                d_a.as<MRec>(), d_ids.as<uint64_t>(), d_keys.as<uint8_t>());
             MRec* d_sorted = merge_sort_recs(d_a.as<MRec>(), d_b.as<MRec>(), R, stream, lut);
             OutPartsBuf opb;
-            opb.alloc(R);
+            opb.alloc(R, gp.static_pct ? 1 : 0);
             d_prows.alloc(R * 8);
             hipLaunchKernelGGL(k_gen_count, dim3(blocks), dim3(256), 0, stream, gp, d_sorted,
                                d_ids.as<uint64_t>(), R, opb.op, d_prows.as<uint64_t>(),
@@ -1956,11 +2009,13 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
             uint32_t gen_nck = gp.clustering_rows ? (gp.ck_cols ? gp.ck_cols : 1) : 0;
             rows.alloc(total_rows, gp.n_value_cols, gen_nck);
             d_vals.alloc(total_rows * (uint64_t)gp.n_value_cols * spec->value_len);
-            DevBuf d_ckarena;
+            DevBuf d_ckarena, d_svals;
             if (gp.ck_text) d_ckarena.alloc(total_rows * 16 + 16);
+            if (gp.static_pct) d_svals.alloc(R * (uint64_t)spec->value_len + 16);
             hipLaunchKernelGGL(k_gen_fill2, dim3(blocks), dim3(256), 0, stream, gp, d_sorted,
                                d_ids.as<uint64_t>(), R, opb.op, rows.uc, d_prows.as<uint64_t>(),
-                               d_vals.as<uint8_t>(), gp.ck_text ? d_ckarena.as<uint8_t>() : nullptr);
+                               d_vals.as<uint8_t>(), gp.ck_text ? d_ckarena.as<uint8_t>() : nullptr,
+                               gp.static_pct ? d_svals.as<uint8_t>() : nullptr);
             hipLaunchKernelGGL(k_gen_values2, dim3(blocks), dim3(256), 0, stream, gp, d_sorted,
                                d_ids.as<uint64_t>(), opb.op, rows.uc, R, d_vals.as<uint8_t>(),
                                d_prows.as<uint64_t>());
@@ -1991,6 +2046,12 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
                 HIP_CHECK(hipMemcpyAsync(d_ckw.p, ckw_h.data(), ckw_h.size() * 4,
                                          hipMemcpyHostToDevice, stream));
             sp.sch.ck_w = d_ckw.as<int32_t>();
+            sp.sch.n_static = gp.static_pct ? 1 : 0;
+            std::vector<int32_t> sfx_h(1, -1);
+            DevBuf d_sfx;
+            d_sfx.alloc(8);
+            HIP_CHECK(hipMemcpyAsync(d_sfx.p, sfx_h.data(), 4, hipMemcpyHostToDevice, stream));
+            sp.sch.static_fixed = d_sfx.as<int32_t>();
             sp.sch.n_cols = gp.n_value_cols;
             std::vector<int32_t> gcf(gp.n_value_cols, -1);  // val blobs
             DevBuf d_gcf;
@@ -2013,13 +2074,16 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
                                     "org.apache.cassandra.db.marshal.BytesType"});
                 }
             }
+            std::vector<std::pair<bytes, std::string>> scols;
+            if (gp.static_pct)
+                scols.push_back({bytes{'s', '0'}, "org.apache.cassandra.db.marshal.BytesType"});
             write_sstable_device(opb, rows, R, sp, d_stats, d_tomb, tomb_cap, base,
                                  gp.key_len > 8 ? "org.apache.cassandra.db.marshal.BytesType"
                                                 : "org.apache.cassandra.db.marshal.LongType",
                                  std::vector<std::string>(
                                      gen_nck, gp.ck_text ? "org.apache.cassandra.db.marshal.UTF8Type"
                                                          : "org.apache.cassandra.db.marshal.LongType"),
-                                 cols, stream);
+                                 cols, scols, stream);
         }
         HIP_CHECK(hipStreamDestroy(stream));
         return GPUC_OK;

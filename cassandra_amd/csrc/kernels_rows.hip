@@ -30,6 +30,8 @@ struct SchemaParams {
     const int32_t* ck_w;     // per clustering column: 4/8 fixed or -1 variable
     uint32_t n_cols;         // regular columns (1..63; header superset order)
     const int32_t* col_fixed;  // per column: -1 variable else fixed width
+    uint32_t n_static;       // static columns (0 = schema has no statics)
+    const int32_t* static_fixed;
     uint32_t column_index_size;  // promoted-index granularity (64 KiB default)
 };
 
@@ -146,6 +148,16 @@ __global__ void k_parse_count(const SrcDesc2* srcs, uint32_t n_srcs, uint32_t to
     pc.pdel_mfda[gi] = pdm;
     pc.pdel_ldt[gi] = pdl;
 
+    // static row slot (always present on disk when the schema has statics)
+    if (sp.n_static) {
+        uint8_t f = base[pos++];
+        if (!(f & 0x80) || base[pos] != 0x01) { atomicExch(error, 18ull); return; }
+        pos++;
+        uint64_t size = uvint_get(base, &pos);
+        uint64_t prev = uvint_get(base, &pos);
+        pos += size - uvint_size(prev);
+    }
+
     // skip-walk the unfiltereds
     uint32_t count = 0;
     while (true) {
@@ -199,6 +211,67 @@ __global__ void k_parse_rows(const SrcDesc2* srcs, uint32_t n_srcs, uint32_t tot
     uint32_t klen = ((uint32_t)base[pos] << 8) | base[pos + 1];
     pos += 2 + klen;
     pos += (base[pos] & 0x80) ? 1 : 12;
+
+    if (sp.n_static) {
+        uint8_t flags = base[pos++];
+        pos++;  // extended flags byte (IS_STATIC, validated in pass A)
+        uvint_get(base, &pos);  // size
+        uvint_get(base, &pos);  // prev
+        uint8_t pf = 0;
+        int64_t lts = NO_TIMESTAMP, llet = NO_DELETION_TIME, rdm = INT64_MIN;
+        int32_t lttl = 0;
+        uint32_t rdl = LDT_NONE_U32;
+        if (flags & 0x04) { pf |= PF_LIVE_TS; lts = (int64_t)uvint_get(base, &pos) + sd.min_ts; }
+        if (flags & 0x08) {
+            lttl = (int32_t)(uint32_t)uvint_get(base, &pos) + sd.min_ttl;
+            llet = (int64_t)(int32_t)(uint32_t)uvint_get(base, &pos) + sd.min_ldt;
+        }
+        if (flags & 0x10) {
+            pf |= PF_ROW_DEL;
+            rdm = (int64_t)uvint_get(base, &pos) + sd.min_ts;
+            rdl = ldt_u32((int64_t)(int32_t)(uint32_t)uvint_get(base, &pos) + sd.min_ldt);
+        }
+        uint64_t missing = 0;
+        if (!(flags & 0x20)) missing = uvint_get(base, &pos);
+        bool any_cell = false;
+        for (uint32_t c = 0; c < sp.n_static; c++) {
+            uint64_t oc = (uint64_t)gi * sp.n_static + c;
+            if (missing & (1ULL << c)) { pc.st.cell_flags[oc] = 0; continue; }
+            any_cell = true;
+            uint8_t cfl = CELLF_PRESENT;
+            uint8_t cf = base[pos++];
+            int64_t cts = (cf & 8) ? lts : (int64_t)uvint_get(base, &pos) + sd.min_ts;
+            bool dead = cf & 1, exp = cf & 2;
+            int64_t ldtl;
+            if (cf & 16) ldtl = llet;
+            else if (dead || exp) ldtl = (int64_t)(int32_t)(uint32_t)uvint_get(base, &pos) + sd.min_ldt;
+            else ldtl = NO_DELETION_TIME;
+            int32_t cttl = (cf & 16) ? lttl : (exp ? (int32_t)(uint32_t)uvint_get(base, &pos) + sd.min_ttl : 0);
+            if (exp) cfl |= CELLF_EXPIRING;
+            uint64_t vaddr = 0;
+            uint32_t vlen = 0;
+            if (!(cf & 4)) {
+                cfl |= CELLF_HAS_VALUE;
+                int32_t fw = sp.static_fixed[c];
+                vlen = fw >= 0 ? (uint32_t)fw : (uint32_t)uvint_get(base, &pos);
+                vaddr = (uint64_t)(base + pos);
+                pos += vlen;
+            }
+            pc.st.cell_flags[oc] = cfl;
+            pc.st.cell_ts[oc] = cts;
+            pc.st.cell_ldt[oc] = ldt_u32(ldtl);
+            pc.st.cell_ttl[oc] = cttl;
+            pc.st.val_addr[oc] = vaddr;
+            pc.st.val_len[oc] = vlen;
+        }
+        if ((pf & (PF_LIVE_TS | PF_ROW_DEL)) || any_cell) pf |= PF_HAS_ROW;
+        pc.st.flags[gi] = pf;
+        pc.st.live_ts[gi] = lts;
+        pc.st.live_ttl[gi] = lttl;
+        pc.st.live_let[gi] = llet;
+        pc.st.rdel_mfda[gi] = rdm;
+        pc.st.rdel_ldt[gi] = rdl;
+    }
 
     uint64_t out = pc.row_base[gi];
     uint32_t emitted = 0;
@@ -454,6 +527,177 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
         }
         adm = m2;
         adl = l2;
+    }
+
+    // ---- static row merge (Row.Merger over the versions' static rows with
+    // activeDeletion = the UNPURGED merged partition deletion) + purge ----
+    bool static_kept = false;
+    if (sp.n_static) {
+        const uint32_t NS = sp.n_static;
+        uint8_t of = 0;
+        int64_t lts = NO_TIMESTAMP, llet = NO_DELETION_TIME, rdm = INT64_MIN;
+        int32_t lttl = 0;
+        uint32_t rdl = LDT_NONE_U32;
+        uint64_t sslot = g;  // output partition index
+        bool any_cell = false;
+        // gather present versions
+        uint32_t nvers = 0;
+        uint32_t iver[GPUC_MAX_ARITY > 16 ? 64 : 64];
+        for (uint32_t m = 0; m < k; m++) {
+            uint32_t i = src_bases[recs[beg + m].src] + recs[beg + m].idx;
+            if (pc.st.flags[i] & PF_HAS_ROW) iver[nvers++] = i;
+        }
+        // k==1: UnfilteredRowIterators.merge of one iterator passes the
+        // static row through unfiltered (as the row path does); otherwise the
+        // Row.Merger single-version shortcut needs a live activeDeletion
+        if (nvers == 1 && (k == 1 || (adm == INT64_MIN && adl == LDT_NONE_U32))) {
+            uint32_t i = iver[0];
+            of = pc.st.flags[i];
+            lts = pc.st.live_ts[i]; lttl = pc.st.live_ttl[i]; llet = pc.st.live_let[i];
+            rdm = pc.st.rdel_mfda[i]; rdl = pc.st.rdel_ldt[i];
+            for (uint32_t c = 0; c < NS; c++) {
+                uint64_t dc = sslot * NS + c, sc = (uint64_t)i * NS + c;
+                uint8_t cfl = pc.st.cell_flags[sc];
+                op.st.cell_flags[dc] = cfl;
+                if (!(cfl & CELLF_PRESENT)) continue;
+                any_cell = true;
+                op.st.cell_ts[dc] = pc.st.cell_ts[sc];
+                op.st.cell_ldt[dc] = pc.st.cell_ldt[sc];
+                op.st.cell_ttl[dc] = pc.st.cell_ttl[sc];
+                op.st.val_addr[dc] = pc.st.val_addr[sc];
+                op.st.val_len[dc] = pc.st.val_len[sc];
+            }
+        } else if (nvers > 0) {
+            bool has_live = false;
+            for (uint32_t v = 0; v < nvers; v++) {
+                uint32_t i = iver[v];
+                uint8_t f = pc.st.flags[i];
+                if (f & PF_LIVE_TS) {
+                    int64_t t2 = pc.st.live_ts[i];
+                    int32_t ttl2 = pc.st.live_ttl[i];
+                    int64_t let2 = pc.st.live_let[i];
+                    bool sup;
+                    if (!has_live) sup = true;
+                    else if (t2 != lts) sup = t2 > lts;
+                    else {
+                        bool e1 = lttl == INT32_MAX, e2 = ttl2 == INT32_MAX;
+                        if (e1 != e2) sup = e2;
+                        else if ((lttl != 0) == (ttl2 != 0)) sup = let2 > llet;
+                        else sup = ttl2 != 0;
+                    }
+                    if (sup) { lts = t2; lttl = ttl2; llet = let2; has_live = true; }
+                }
+                if (f & PF_ROW_DEL)
+                    if (dt_sup(pc.st.rdel_mfda[i], pc.st.rdel_ldt[i], rdm, rdl)) { rdm = pc.st.rdel_mfda[i]; rdl = pc.st.rdel_ldt[i]; }
+            }
+            if (has_live) of |= PF_LIVE_TS;
+            int64_t am2 = adm;
+            uint32_t al2 = adl;
+            bool row_del_kept = false;
+            if (dt_sup(rdm, rdl, am2, al2)) { am2 = rdm; al2 = rdl; row_del_kept = true; }
+            if (!row_del_kept) { rdm = INT64_MIN; rdl = LDT_NONE_U32; }
+            else of |= PF_ROW_DEL;
+            if (has_live && lts <= am2) { of &= ~PF_LIVE_TS; lts = NO_TIMESTAMP; lttl = 0; llet = NO_DELETION_TIME; }
+            for (uint32_t c = 0; c < NS; c++) {
+                int64_t cts = NO_TIMESTAMP;
+                uint32_t cldt = LDT_NONE_U32;
+                int32_t cttl = 0;
+                uint64_t va = 0;
+                uint32_t vl = 0;
+                bool have_cell = false, cell_val = false, cell_exp = false;
+                for (uint32_t v = 0; v < nvers; v++) {
+                    uint64_t o = (uint64_t)iver[v] * NS + c;
+                    uint8_t f = pc.st.cell_flags[o];
+                    if (!(f & CELLF_PRESENT)) continue;
+                    int64_t ts2 = pc.st.cell_ts[o];
+                    if (ts2 <= am2) continue;
+                    if (!have_cell) {
+                        have_cell = true;
+                        cts = ts2; cldt = pc.st.cell_ldt[o]; cttl = pc.st.cell_ttl[o];
+                        va = pc.st.val_addr[o]; vl = pc.st.val_len[o];
+                        cell_val = f & CELLF_HAS_VALUE;
+                        cell_exp = f & CELLF_EXPIRING;
+                        continue;
+                    }
+                    bool take_right = false;
+                    uint32_t rl = pc.st.cell_ldt[o];
+                    bool l_dt = cldt != LDT_NONE_U32, r_dt = rl != LDT_NONE_U32;
+                    if (cts != ts2) take_right = ts2 > cts;
+                    else if (l_dt || r_dt) {
+                        if (l_dt != r_dt) take_right = r_dt;
+                        else {
+                            bool l_tomb = !cell_exp, r_tomb = !(f & CELLF_EXPIRING);
+                            if (l_tomb != r_tomb) take_right = r_tomb;
+                            else if (cldt != rl) take_right = ldt_long(rl) > ldt_long(cldt);
+                            else take_right = cmp_values(va, vl, pc.st.val_addr[o], pc.st.val_len[o]) < 0;
+                        }
+                    } else {
+                        take_right = cmp_values(va, vl, pc.st.val_addr[o], pc.st.val_len[o]) < 0;
+                    }
+                    if (take_right) {
+                        cts = ts2; cldt = rl; cttl = pc.st.cell_ttl[o];
+                        va = pc.st.val_addr[o]; vl = pc.st.val_len[o];
+                        cell_val = f & CELLF_HAS_VALUE;
+                        cell_exp = f & CELLF_EXPIRING;
+                    }
+                }
+                uint64_t dc = sslot * NS + c;
+                if (have_cell) {
+                    any_cell = true;
+                    op.st.cell_flags[dc] = CELLF_PRESENT | (cell_val ? CELLF_HAS_VALUE : 0) |
+                                           (cell_exp ? CELLF_EXPIRING : 0);
+                    op.st.cell_ts[dc] = cts;
+                    op.st.cell_ldt[dc] = cldt;
+                    op.st.cell_ttl[dc] = cttl;
+                    op.st.val_addr[dc] = va;
+                    op.st.val_len[dc] = vl;
+                } else {
+                    op.st.cell_flags[dc] = 0;
+                }
+            }
+            if ((of & (PF_LIVE_TS | PF_ROW_DEL)) || any_cell) of |= PF_HAS_ROW;
+            else of = 0;
+        } else {
+            for (uint32_t c = 0; c < NS; c++) op.st.cell_flags[sslot * NS + c] = 0;
+        }
+        // purge (BTreeRow.purge over the merged static row)
+        if (of & PF_HAS_ROW) {
+            if (of & PF_LIVE_TS) {
+                bool is_live = lttl == INT32_MAX ? false : (lttl != 0 ? pp.now_sec < llet : true);
+                if (!is_live && should_purge2(pp, token, lts, llet)) { of &= ~PF_LIVE_TS; lts = NO_TIMESTAMP; lttl = 0; llet = NO_DELETION_TIME; }
+            }
+            if ((of & PF_ROW_DEL) && rdm != INT64_MIN && should_purge2(pp, token, rdm, ldt_long(rdl))) { of &= ~PF_ROW_DEL; rdm = INT64_MIN; rdl = LDT_NONE_U32; }
+            any_cell = false;
+            for (uint32_t c = 0; c < NS; c++) {
+                uint64_t dc = sslot * NS + c;
+                uint8_t cfl = op.st.cell_flags[dc];
+                if (!(cfl & CELLF_PRESENT)) continue;
+                int64_t cts = op.st.cell_ts[dc];
+                uint32_t cldt = op.st.cell_ldt[dc];
+                int32_t cttl = op.st.cell_ttl[dc];
+                bool live_cell = cldt == LDT_NONE_U32 || (cttl != 0 && pp.now_sec < ldt_long(cldt));
+                if (!live_cell) {
+                    if (should_purge2(pp, token, cts, ldt_long(cldt))) { op.st.cell_flags[dc] = 0; continue; }
+                    if (cttl != 0) {
+                        int64_t nldt = ldt_long(cldt) - cttl;
+                        if (should_purge2(pp, token, cts, nldt)) { op.st.cell_flags[dc] = 0; continue; }
+                        op.st.cell_ldt[dc] = ldt_u32(nldt);
+                        op.st.cell_ttl[dc] = 0;
+                        op.st.cell_flags[dc] = CELLF_PRESENT;
+                        op.st.val_len[dc] = 0;
+                    }
+                }
+                any_cell = true;
+            }
+            if (!(of & (PF_LIVE_TS | PF_ROW_DEL)) && !any_cell) of = 0;
+        }
+        op.st.flags[g] = of;
+        op.st.live_ts[g] = lts;
+        op.st.live_ttl[g] = lttl;
+        op.st.live_let[g] = llet;
+        op.st.rdel_mfda[g] = rdm;
+        op.st.rdel_ldt[g] = rdl;
+        static_kept = (of & PF_HAS_ROW) != 0;
     }
 
     // marker-merger state (RangeTombstoneMarker.Merger)
@@ -801,7 +1045,7 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
     bool pdel_live = pdm == INT64_MIN && pdl == LDT_NONE_U32;
     op.pdel_mfda[g] = pdm;
     op.pdel_ldt[g] = pdl;
-    op.keep[g] = (!pdel_live || ocount > 0) ? 1 : 0;
+    op.keep[g] = (!pdel_live || ocount > 0 || static_kept) ? 1 : 0;
     if (op.keep[g]) {
         atomicMin(&st->first_group, (unsigned long long)g);
         atomicMax(&st->last_group, (unsigned long long)g);
@@ -820,8 +1064,10 @@ struct SerParams2 {
     SchemaParams sch;
 };
 
-// one cell: serialized body size + flags byte (Cell.Serializer)
-__device__ inline uint32_t cell_body_size(const UnfCols& u, uint64_t oc, const SerParams2& sp,
+// one cell: serialized body size + flags byte (Cell.Serializer);
+// COLS is UnfCols (regular cells) or StaticCols (same member names)
+template <typename COLS>
+__device__ inline uint32_t cell_body_size(const COLS& u, uint64_t oc, const SerParams2& sp,
                                           int32_t fixed, bool live, bool exp_live, uint64_t o,
                                           uint8_t* out_cf) {
     uint8_t f = u.cell_flags[oc];
@@ -897,6 +1143,91 @@ __device__ inline uint32_t unf_body_size(const UnfCols& u, uint64_t o, const Ser
     return body;
 }
 
+// the static row: size (always) and bytes (when out_data != nullptr) at
+// data_off+start. Written whenever the schema has statics — possibly as an
+// EMPTY static row (flags 0x80, ext 0x01, all-missing subset), exactly as
+// SortedTablePartitionWriter.addStaticRow does.
+__device__ inline uint64_t static_row_bytes(const OutParts& op, uint64_t g,
+                                            const SerParams2& sp, uint8_t* out_data,
+                                            uint64_t data_off, uint64_t start, int lane) {
+    const uint32_t NS = sp.sch.n_static;
+    const StaticCols& st = op.st;
+    uint8_t sof = st.flags[g];
+    bool live = sof & PF_LIVE_TS;
+    bool exp_live = live && st.live_ttl[g] != NO_TTL;
+    bool has_del = sof & PF_ROW_DEL;
+    uint64_t pmask = 0;
+    uint32_t present = 0;
+    for (uint32_t c = 0; c < NS; c++)
+        if (st.cell_flags[g * NS + c] & CELLF_PRESENT) { pmask |= 1ULL << c; present++; }
+    uint8_t rf = 0x80;
+    if (live) rf |= 0x04;
+    if (exp_live) rf |= 0x08;
+    if (has_del) rf |= 0x10;
+    if (present == NS) rf |= 0x20;
+    uint32_t body = 0;
+    if (live) body += uvint_size((uint64_t)(st.live_ts[g] - sp.hs.min_ts));
+    if (exp_live) body += uvint_size(sext32(st.live_ttl[g] - sp.hs.min_ttl)) +
+                          uvint_size(sext32(st.live_let[g] - sp.hs.min_ldt));
+    if (has_del) body += uvint_size((uint64_t)(st.rdel_mfda[g] - sp.hs.min_ts)) +
+                         uvint_size(sext32(ldt_long(st.rdel_ldt[g]) - sp.hs.min_ldt));
+    uint64_t miss = ((NS >= 64 ? ~0ULL : ((1ULL << NS) - 1)) & ~pmask);
+    if (!(rf & 0x20)) body += uvint_size(miss);
+    uint8_t cfb_arr[63];
+    for (uint32_t c = 0; c < NS; c++) {
+        if (!(pmask & (1ULL << c))) continue;
+        body += cell_body_size(st, g * NS + c, sp, sp.sch.static_fixed[c], live, exp_live, g,
+                               &cfb_arr[c]);
+    }
+    uint64_t total = 2 + uvint_size(body + uvint_size(0)) + uvint_size(0) + body;
+    if (!out_data) return total;
+    // ---- emit ----
+    uint64_t pos = start;
+    auto put8 = [&](uint8_t v) { if (lane == 0) out_data[data_off + pos] = v; pos++; };
+    auto put_uv = [&](uint64_t v) {
+        if (lane == 0) {
+            uint8_t tmp[9];
+            int n = uvint_put(tmp, v);
+            for (int i = 0; i < n; i++) out_data[data_off + pos + i] = tmp[i];
+        }
+        pos += uvint_size(v);
+    };
+    put8(rf);
+    put8(0x01);  // extended flags: IS_STATIC
+    put_uv(body + uvint_size(0));
+    put_uv(0);   // previousUnfilteredSize
+    if (live) put_uv((uint64_t)(st.live_ts[g] - sp.hs.min_ts));
+    if (exp_live) {
+        put_uv(sext32(st.live_ttl[g] - sp.hs.min_ttl));
+        put_uv(sext32(st.live_let[g] - sp.hs.min_ldt));
+    }
+    if (has_del) {
+        put_uv((uint64_t)(st.rdel_mfda[g] - sp.hs.min_ts));
+        put_uv(sext32(ldt_long(st.rdel_ldt[g]) - sp.hs.min_ldt));
+    }
+    if (!(rf & 0x20)) put_uv(miss);
+    for (uint32_t c = 0; c < NS; c++) {
+        if (!(pmask & (1ULL << c))) continue;
+        uint64_t oc = g * NS + c;
+        uint8_t cfb = cfb_arr[c];
+        put8(cfb);
+        if (!(cfb & 8)) put_uv((uint64_t)(st.cell_ts[oc] - sp.hs.min_ts));
+        bool deleted = cfb & 1, expiring = cfb & 2;
+        if ((deleted || expiring) && !(cfb & 16))
+            put_uv(sext32(ldt_long(st.cell_ldt[oc]) - sp.hs.min_ldt));
+        if (expiring && !(cfb & 16)) put_uv(sext32(st.cell_ttl[oc] - sp.hs.min_ttl));
+        if (!(cfb & 4)) {
+            uint32_t vlen = st.val_len[oc];
+            if (sp.sch.static_fixed[c] < 0) put_uv(vlen);
+            const uint8_t* src = (const uint8_t*)st.val_addr[oc];
+            for (uint32_t i = (uint32_t)lane; i < vlen; i += WAVE)
+                out_data[data_off + pos + i] = src[i];
+            pos += vlen;
+        }
+    }
+    return total;
+}
+
 // serialized size of a ClusteringPrefix (full serializer: kind byte [+ u16
 // size for bounds] + values-without-size) — IndexInfo first/last names
 __device__ inline uint32_t prefix_full_size(const UnfCols& u, uint64_t o,
@@ -938,7 +1269,10 @@ __device__ uint64_t part_walk(const OutParts& op, const UnfCols& out, uint64_t g
     uint32_t klen = op.klen[g];
     const uint8_t* keyb = (const uint8_t*)op.key_addr[g];
     bool pdel_live = op.pdel_mfda[g] == INT64_MIN && op.pdel_ldt[g] == LDT_NONE_U32;
-    uint64_t header_len = 2 + klen + (pdel_live ? 1 : 12);
+    uint64_t static_sz = sp.sch.n_static
+                             ? static_row_bytes(op, g, sp, nullptr, 0, 0, lane)
+                             : 0;
+    uint64_t header_len = 2 + klen + (pdel_live ? 1 : 12) + static_sz;
     uint64_t rb = op.row_base[g];
     uint32_t nrows = op.row_count[g];
 
@@ -1002,6 +1336,10 @@ __device__ uint64_t part_walk(const OutParts& op, const UnfCols& out, uint64_t g
     emit8((uint8_t)klen);
     for (uint32_t b = 0; b < klen; b++) emit8(keyb[b]);
     emit_dt(op.pdel_mfda[g], op.pdel_ldt[g]);
+    if (sp.sch.n_static) {
+        if (EMIT) static_row_bytes(op, g, sp, out_data, data_off, pos, lane);
+        pos += static_sz;
+    }
 
     // ---- walk unfiltereds ----
     uint64_t prev_start = 0;
@@ -1466,6 +1804,7 @@ struct GenParams2 {
     uint32_t key_len;   // 8 (default) .. 255; bytes 8.. are gen2_key_salt(id, j)
     uint32_t ck_text;   // clustering values as UTF8 strings (oracle gen_ck_bytes)
     uint32_t ck_cols;   // 0/1 = one clustering column; 2 = (bigint, bigint)
+    uint32_t static_pct;  // percent of wide partitions with a static row ("s0" blob)
     uint32_t n_value_cols;    // regular columns val0..valN-1 (1..63)
     uint32_t col_missing_pct; // P(cell absent) per live row and column
     int64_t base_ts, base_ldt;
@@ -1536,12 +1875,35 @@ __global__ void k_gen_count(GenParams2 gp, const MRec* sorted, const uint64_t* i
 
 __global__ void k_gen_fill2(GenParams2 gp, const MRec* sorted, const uint64_t* ids, uint64_t n,
                             OutParts op, UnfCols out, const uint64_t* row_base,
-                            uint8_t* values, uint8_t* ck_arena) {
+                            uint8_t* values, uint8_t* ck_arena, uint8_t* static_vals) {
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
     uint64_t id = ids[sorted[i].idx];
     uint64_t ob = row_base[i];
     op.row_base[i] = ob;
+    if (gp.static_pct) {
+        // oracle contract: gen_has_static / gen_static_ts; one blob cell
+        bool has = splitmix64(gp.seed ^ 0x57A71CULL ^ id ^ ((uint64_t)gp.sst << 32)) % 100 <
+                   gp.static_pct;
+        if (has) {
+            int64_t sts = gp.base_ts + (int64_t)(splitmix64(gp.seed ^ id * 977 ^ ((uint64_t)gp.sst << 48)) % 1000000000ULL);
+            op.st.flags[i] = PF_HAS_ROW | PF_LIVE_TS;
+            op.st.live_ts[i] = sts;
+            op.st.live_ttl[i] = 0;
+            op.st.live_let[i] = NO_DELETION_TIME;
+            op.st.rdel_mfda[i] = INT64_MIN;
+            op.st.rdel_ldt[i] = LDT_NONE_U32;
+            op.st.cell_flags[i] = CELLF_PRESENT | CELLF_HAS_VALUE;
+            op.st.cell_ts[i] = sts;
+            op.st.cell_ldt[i] = LDT_NONE_U32;
+            op.st.cell_ttl[i] = 0;
+            op.st.val_addr[i] = (uint64_t)(static_vals + i * (uint64_t)gp.value_len);
+            op.st.val_len[i] = gp.value_len;
+        } else {
+            op.st.flags[i] = 0;
+            op.st.cell_flags[i] = 0;
+        }
+    }
     uint32_t emitted = 0;
     const uint32_t NCK = gp.ck_cols ? gp.ck_cols : (gp.clustering_rows ? 1 : 0);
     // ck encodes the full position; ck_cols==2 splits it as (ck/64, ck%64)
@@ -1687,6 +2049,21 @@ __global__ void k_gen_values2(GenParams2 gp, const MRec* sorted, const uint64_t*
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
     uint64_t id = ids[sorted[i].idx];
+    if (gp.static_pct && (op.st.cell_flags[i] & CELLF_HAS_VALUE)) {
+        uint64_t vid = id ^ 0xABCDEF57ULL;
+        uint8_t* outp = (uint8_t*)op.st.val_addr[i];
+        uint64_t state = gp.seed ^ vid * 0x100000001B3ULL ^ ((uint64_t)gp.sst << 40);
+        uint64_t prev = splitmix64(state);
+        uint32_t nw = (gp.value_len + 7) / 8;
+        for (uint32_t w = 0; w < nw; w++) {
+            uint64_t r = splitmix64(state + 1 + w);
+            uint64_t word = (r % 100 < gp.value_repeat_pct && w > 0) ? prev : splitmix64(r);
+            prev = word;
+            uint32_t off = w * 8;
+            for (uint32_t b = 0; b < 8 && off + b < gp.value_len; b++)
+                outp[off + b] = (uint8_t)(word >> (8 * b));
+        }
+    }
     uint64_t rb = op.row_base[i];
     uint32_t cnt = op.row_count[i];
     uint32_t rowj = 0;

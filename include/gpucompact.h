@@ -127,6 +127,7 @@ typedef struct gpuc_gen_spec {
     uint32_t ck_text;           /* clustering values as UTF8 strings (variable width) */
     uint32_t ck_cols;           /* 0/1 = one clustering column; 2 = composite (bigint,bigint)
                                    with ck0-prefix range-tombstone bounds */
+    uint32_t static_pct;        /* percent of wide partitions with a static row */
     uint32_t n_value_cols;      /* regular columns val0..valN-1 (0 == 1 column "val") */
     uint32_t col_missing_pct;   /* P(cell absent) per live row and column */
     int64_t base_ts;

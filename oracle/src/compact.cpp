@@ -187,6 +187,20 @@ Partition merge_partition_versions(const std::vector<const Partition*>& versions
         if (!del.supersedes(p->del)) del = p->del;
     out.del = del;
 
+    // static rows merge like rows with activeDeletion = the merged partition
+    // deletion (no range tombstone can cover the static clustering)
+    if (h.has_static()) {
+        std::vector<const Row*> svs;
+        for (auto* p : versions)
+            if (!row_is_empty(p->static_row)) svs.push_back(&p->static_row);
+        if (!svs.empty()) {
+            Row merged_static;
+            if (row_merge(svs, del, h.static_cols.size(), merged_static))
+                out.static_row = std::move(merged_static);
+            out.static_row.static_flag = true;
+        }
+    }
+
     size_t k = versions.size();
     std::vector<size_t> pos(k, 0);
     MarkerMerger marker_merger(k, del);
@@ -303,6 +317,8 @@ bool purge_partition(Partition& p, int64_t now_sec, int64_t gc_before, bool neve
                      const std::vector<PurgeRange>& overlaps, bool enforce_strict_liveness) {
     Purger pg{now_sec, gc_before, never_purge, &overlaps, p.token};
     if (pg.should_purge(p.del)) p.del = DT_LIVE;
+    if (!row_is_empty(p.static_row) && !purge_row(p.static_row, pg, false))
+        p.static_row = Row{};
     std::vector<Unfiltered> kept;
     for (auto& u : p.items) {
         if (u.kind == Unfiltered::ROW) {
@@ -335,7 +351,7 @@ bool purge_partition(Partition& p, int64_t now_sec, int64_t gc_before, bool neve
         }
     }
     p.items = std::move(kept);
-    return !(p.del.live() && p.items.empty());
+    return !(p.del.live() && p.items.empty() && row_is_empty(p.static_row));
 }
 
 // ---------------------------------------------------------------------------

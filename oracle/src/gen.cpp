@@ -12,6 +12,8 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
         uint32_t nck = g.ck_cols ? g.ck_cols : 1;
         t.header.clustering_types.assign(nck, g.ck_text ? CqlType::UTF8 : CqlType::LONG);
     }
+    if (g.static_pct > 0)
+        t.header.static_cols = {{bytes{'s', '0'}, CqlType::BYTES}};
     if (g.n_value_cols <= 1) {
         t.header.regular_cols = {{bytes{'v', 'a', 'l'}, CqlType::BYTES}};  // val blob
     } else {
@@ -47,6 +49,18 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
             Partition p;
             p.key = e.key;
             p.token = e.token;
+            if (g.static_pct > 0 && gen_has_static(g, sst, e.id)) {
+                Row& sr = p.static_row;
+                sr.static_flag = true;
+                sr.cells.resize(1);
+                int64_t sts = gen_static_ts(g, sst, e.id);
+                min_ts = std::min(min_ts, sts);
+                sr.live.ts = sts;
+                Cell sc;
+                sc.ts = sts;
+                gen_value(g, sst, e.id ^ 0xABCDEF57ULL, sc.value);
+                sr.cells[0] = std::move(sc);
+            }
             bool has_rt = gen_has_range_tomb(g, sst, e.id);
             int64_t rlo = 0, rhi = 0, rts = 0;
             uint32_t rldt = 0;

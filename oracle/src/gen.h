@@ -40,6 +40,9 @@ struct GenSpec {
     // (16*(j/4), j%4); range tombstones become ck0-PREFIX bounds covering
     // whole ck0 groups (GPU mirror: GenParams2.ck_cols)
     uint32_t ck_cols = 0;
+    // static_pct: percent of partitions (wide mode) carrying a static row in
+    // one static column "s0" blob (GPU mirror: GenParams2.static_pct)
+    uint32_t static_pct = 0;
     // partition key width in bytes, 8..255: first 8 = big-endian key id,
     // bytes 8.. = splitmix64(id ^ (0xC0FFEE5EED + j)) (GPU: gen2_key_salt).
     // key_len > 8 switches the declared key type LongType -> BytesType.
@@ -126,6 +129,13 @@ inline int64_t gen_row_ts(const GenSpec& g, uint32_t sst, uint64_t key_id, uint3
 inline bool gen_row_is_tombstone(const GenSpec& g, uint32_t sst, uint64_t key_id, uint32_t j) {
     if (g.tombstone_pct == 0) return false;
     return splitmix64(g.seed ^ 0xDEADULL ^ key_id ^ ((uint64_t)sst << 32) ^ (uint64_t)(j + 7) * 131) % 100 < g.tombstone_pct;
+}
+inline bool gen_has_static(const GenSpec& g, uint32_t sst, uint64_t key_id) {
+    if (g.static_pct == 0) return false;
+    return splitmix64(g.seed ^ 0x57A71CULL ^ key_id ^ ((uint64_t)sst << 32)) % 100 < g.static_pct;
+}
+inline int64_t gen_static_ts(const GenSpec& g, uint32_t sst, uint64_t key_id) {
+    return g.base_ts + (int64_t)(splitmix64(g.seed ^ key_id * 977 ^ ((uint64_t)sst << 48)) % 1000000000ULL);
 }
 inline bool gen_has_range_tomb(const GenSpec& g, uint32_t sst, uint64_t key_id) {
     if (g.range_tomb_pct == 0) return false;

@@ -104,6 +104,7 @@ static GenSpec spec_from_kv(std::map<std::string, std::string>& kv) {
     g.key_len = (uint32_t)geti("keylen", g.key_len);
     g.ck_text = (uint32_t)geti("cktext", g.ck_text);
     g.ck_cols = (uint32_t)geti("ckcols", g.ck_cols);
+    g.static_pct = (uint32_t)geti("statics", g.static_pct);
     g.n_value_cols = (uint32_t)geti("ncols", g.n_value_cols);
     g.col_missing_pct = (uint32_t)geti("colmiss", g.col_missing_pct);
     g.first_generation = geti("gen0", g.first_generation);

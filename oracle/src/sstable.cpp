@@ -190,7 +190,7 @@ static void put_row_body(bytes& out, const Row& row, const Header& h, uint8_t fl
     if (flags & F_HAS_TIMESTAMP) h.w_ts(out, row.live.ts);
     if (flags & F_HAS_TTL) { h.w_ttl(out, row.live.ttl); h.w_ldt(out, row.live.let); }
     if (flags & F_HAS_DELETION) h.w_dt(out, row.del);
-    const auto& cols = h.regular_cols;  // statics unsupported in writer (asserted upstream)
+    const auto& cols = row.static_flag ? h.static_cols : h.regular_cols;
     if (!(flags & F_HAS_ALL_COLUMNS)) put_column_subset(out, row.cells, cols.size());
     for (size_t i = 0; i < cols.size(); i++)
         if (row.cells[i]) put_cell(out, *row.cells[i], cols[i].second, row.live, h);
@@ -204,7 +204,8 @@ static uint8_t row_flags(const Row& row, const Header& h) {
     if (!row.live.empty()) flags |= F_HAS_TIMESTAMP;
     if (row.live.expiring()) flags |= F_HAS_TTL;
     if (!row.del.live()) flags |= F_HAS_DELETION;
-    if (present == h.regular_cols.size()) flags |= F_HAS_ALL_COLUMNS;
+    if (present == (row.static_flag ? h.static_cols : h.regular_cols).size())
+        flags |= F_HAS_ALL_COLUMNS;
     return flags;
 }
 
@@ -297,7 +298,20 @@ void serialize_partition(const Partition& p, const Header& h, bytes& out, bytes&
     put_short_len_bytes(out, p.key);
     put_deletion_time(out, p.del);
     uint64_t header_len = out.size() - initial;
-    if (h.has_static()) throw std::runtime_error("static columns unsupported in writer");
+    if (h.has_static()) {
+        Row sr = p.static_row;
+        sr.static_flag = true;
+        sr.cells.resize(h.static_cols.size());
+        uint8_t flags = row_flags(sr, h) | F_EXTENSION;
+        out.push_back(flags);
+        out.push_back(XF_IS_STATIC);
+        bytes body;
+        put_row_body(body, sr, h, flags);
+        put_unsigned_vint(out, body.size() + unsigned_vint_size(0));
+        put_unsigned_vint(out, 0);  // previousUnfilteredSize of the static row
+        out.insert(out.end(), body.begin(), body.end());
+        header_len = out.size() - initial;
+    }
 
     std::vector<IndexInfoC> blocks;
     bool block_open = false;
@@ -885,7 +899,7 @@ static Partition read_partition(Reader& r, const Header& h) {
         if (!(flags & F_EXTENSION)) throw std::runtime_error("expected static row extension flag");
         uint8_t xflags = r.u8();
         if (!(xflags & XF_IS_STATIC)) throw std::runtime_error("expected static row");
-        throw std::runtime_error("static rows unsupported in round 1");
+        p.static_row = read_row(r, h, flags, true);
     }
     while (true) {
         uint8_t flags = r.u8();

@@ -97,6 +97,8 @@ enum BoundKind : uint8_t {
     CLUSTERING_K = 4, INCL_END_EXCL_START = 5, INCL_END = 6, EXCL_START = 7
 };
 
+struct Row;
+inline bool row_is_empty(const Row& r);
 struct Row {
     Clustering clustering;
     LivenessInfo live;
@@ -109,6 +111,12 @@ struct Row {
         return true;
     }
 };
+inline bool row_is_empty(const Row& r) {
+    if (!r.live.empty() || !r.del.live()) return false;
+    for (auto& c : r.cells)
+        if (c) return false;
+    return true;
+}
 
 struct Marker {
     BoundKind kind;
@@ -133,6 +141,7 @@ struct Partition {
     bytes key;
     int64_t token = 0;
     DeletionTime del;
+    Row static_row;   // static_flag=true when header.has_static(); may be empty
     std::vector<Unfiltered> items;
     void set_token() { token = murmur3_token(key.data(), key.size()); }
 };

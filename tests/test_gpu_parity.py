@@ -74,6 +74,19 @@ def test_generator_parity_wide(ca, oracle_bin, tmp_path):
         _assert_dirs_equal(f"{dg}/oa-{g}-big", f"{do}/oa-{g}-big")
 
 
+def test_generator_parity_static_rows(ca, oracle_bin, tmp_path):
+    """Static rows (one blob static column, SortedTablePartitionWriter slot
+    after the partition deletion): write-path parity."""
+    dg, do = str(tmp_path / "gpu"), str(tmp_path / "cpu")
+    os.makedirs(dg), os.makedirs(do)
+    kw = dict(seed=113, n=2, rows=40, crows=30, vlen=250, overlap=30, tomb=10, statics=60)
+    ca.generate(dg, seed=113, n_sstables=2, rows_per_sstable=40, clustering_rows=30,
+                value_len=250, overlap_pct=30, tombstone_pct=10, static_pct=60)
+    _oracle_gen(do, **kw)
+    for g in (1, 2):
+        _assert_dirs_equal(f"{dg}/oa-{g}-big", f"{do}/oa-{g}-big")
+
+
 def test_generator_parity_composite_ck(ca, oracle_bin, tmp_path):
     """Composite (2-column) clustering keys with ck0-prefix range-tombstone
     bounds: write-path parity."""
@@ -198,6 +211,15 @@ def test_generator_parity_long_keys(ca, oracle_bin, tmp_path):
                                                     vlen=180, overlap=30, tomb=12, rtomb=30,
                                                     ckcols=2, ncols=3, colmiss=20, keylen=24),
          job={}),
+    dict(name="static_rows", gen=dict(seed=111, n=3, rows=40, crows=30, vlen=200,
+                                      overlap=30, tomb=10, statics=50), job={}),
+    dict(name="static_rows_gc", gen=dict(seed=111, n=3, rows=40, crows=30, vlen=200,
+                                         overlap=30, tomb=10, statics=50),
+         job=dict(gc_before=2000000000)),
+    dict(name="static_rows_kitchen_sink", gen=dict(seed=112, n=3, rows=24, crows=36,
+                                                   vlen=150, overlap=30, tomb=12, rtomb=25,
+                                                   statics=40, ckcols=2, ncols=3, colmiss=20,
+                                                   keylen=24), job={}),
 ])
 def test_compaction_parity(ca, oracle_bin, tmp_path, case):
     d = str(tmp_path)

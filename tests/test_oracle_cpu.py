@@ -178,3 +178,14 @@ def test_composite_clustering_roundtrip_and_merge(oracle_bin, tmp_path):
     assert stats["partitions_out"] > 0 and stats["rows_out"] > 0
     r = oracle_run("roundtrip", f"{d}/oa-50-big")
     assert "FAIL" not in r.stdout, r.stdout
+
+
+def test_static_rows_roundtrip_and_merge(oracle_bin, tmp_path):
+    """Static rows: write -> reread -> merge -> gc purge."""
+    d = str(tmp_path)
+    _gen(d, n=3, rows=40, crows=30, vlen=150, overlap=30, tomb=10, statics=50, seed=114)
+    ins = [f"{d}/oa-{g}-big" for g in (1, 2, 3)]
+    stats = _compact(f"{d}/oa-50-big", ins)
+    assert stats["partitions_out"] > 0
+    r = oracle_run("roundtrip", f"{d}/oa-50-big")
+    assert "FAIL" not in r.stdout, r.stdout
