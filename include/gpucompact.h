@@ -6,9 +6,9 @@
  * — scanners + CompactionIterator + CompactionAwareWriter — for eligible
  * tables (big-format `oa`, Murmur3Partitioner, any partition key, up to 32
  * clustering columns of fixed (bigint/int) or variable (text/ascii/blob)
- * width including prefix range-tombstone bounds, 1..63 regular columns with
- * cell subsets, LZ4 chunk compression; row, cell, partition and range
- * tombstones all supported). The Java host above the seam
+ * width including prefix range-tombstone bounds, 1..63 regular and 1..63
+ * static columns with cell subsets, LZ4 chunk compression; row, cell,
+ * partition and range tombstones all supported). The Java host above the seam
  * (strategies, CompactionManager, LifecycleTransaction, metrics) is
  * unchanged and binds these entry points via JNI/Panama.
  *
