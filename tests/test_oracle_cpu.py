@@ -189,3 +189,23 @@ def test_static_rows_roundtrip_and_merge(oracle_bin, tmp_path):
     assert stats["partitions_out"] > 0
     r = oracle_run("roundtrip", f"{d}/oa-50-big")
     assert "FAIL" not in r.stdout, r.stdout
+
+
+def test_merge_semantics_laws(oracle_bin, tmp_path):
+    """Merge/GC semantics transcribed from the reference's own unit tests
+    (CompactionIteratorTest / UnfilteredRowIteratorsMergeTest /
+    CompactionsPurgeTest / Cells tie rules) asserted directly against the
+    oracle's merge_partition_versions/purge_partition."""
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    exe = str(tmp_path / "merge_laws")
+    subprocess.run(
+        ["g++", "-O2", "-std=c++17",
+         os.path.join(repo, "tests/native/merge_laws_test.cpp"),
+         os.path.join(repo, "oracle/src/sstable.cpp"),
+         os.path.join(repo, "oracle/src/compact.cpp"),
+         os.path.join(repo, "oracle/src/gen.cpp"),
+         "-o", exe, "-l:liblz4.so.1"],
+        check=True, capture_output=True)
+    r = subprocess.run([exe], capture_output=True, text=True)
+    assert r.returncode == 0, r.stdout
+    assert "merge laws OK" in r.stdout
