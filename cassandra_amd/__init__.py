@@ -92,6 +92,18 @@ class GpucGenSpec(ctypes.Structure):
     ]
 
 
+class GpucFlushRows(ctypes.Structure):
+    _fields_ = [
+        ("n_rows", ctypes.c_uint64),
+        ("keys", ctypes.POINTER(ctypes.POINTER(ctypes.c_uint8))),
+        ("key_lens", ctypes.POINTER(ctypes.c_uint16)),
+        ("timestamps", ctypes.POINTER(ctypes.c_int64)),
+        ("values", ctypes.POINTER(ctypes.POINTER(ctypes.c_uint8))),
+        ("value_lens", ctypes.POINTER(ctypes.c_uint32)),
+        ("del_ldts", ctypes.POINTER(ctypes.c_uint32)),
+    ]
+
+
 INT64_MIN = -(2**63)
 
 
@@ -124,10 +136,50 @@ def load_library(path=None):
     lib.gpuc_generate.restype = ctypes.c_int
     lib.gpuc_verify.argtypes = [ctypes.c_char_p, ctypes.c_int32, ctypes.c_char_p, ctypes.c_size_t]
     lib.gpuc_verify.restype = ctypes.c_int
+    lib.gpuc_flush.argtypes = [ctypes.POINTER(GpucFlushRows), ctypes.c_char_p, ctypes.c_int32,
+                               ctypes.c_char_p, ctypes.c_size_t]
+    lib.gpuc_flush.restype = ctypes.c_int
     lib.gpuc_version.restype = ctypes.c_char_p
     lib.gpuc_device_count.restype = ctypes.c_int
     _lib = lib
     return lib
+
+
+def flush(rows, output_base, device=0):
+    """Memtable-flush analog: rows = list of (key_bytes, timestamp, value_bytes)
+    for live rows or (key_bytes, timestamp, None, del_ldt) for row tombstones.
+    Keys must be unique; sorting happens on the GPU."""
+    lib = load_library()
+    n = len(rows)
+    keys = (ctypes.POINTER(ctypes.c_uint8) * n)()
+    klens = (ctypes.c_uint16 * n)()
+    tss = (ctypes.c_int64 * n)()
+    vals = (ctypes.POINTER(ctypes.c_uint8) * n)()
+    vlens = (ctypes.c_uint32 * n)()
+    dldts = (ctypes.c_uint32 * n)()
+    bufs = []
+    for i, r in enumerate(rows):
+        kb = ctypes.create_string_buffer(bytes(r[0]), len(r[0]))
+        bufs.append(kb)
+        keys[i] = ctypes.cast(kb, ctypes.POINTER(ctypes.c_uint8))
+        klens[i] = len(r[0])
+        tss[i] = r[1]
+        if r[2] is None:
+            vals[i] = None
+            vlens[i] = 0
+            dldts[i] = r[3]
+        else:
+            vb = ctypes.create_string_buffer(bytes(r[2]), len(r[2]))
+            bufs.append(vb)
+            vals[i] = ctypes.cast(vb, ctypes.POINTER(ctypes.c_uint8))
+            vlens[i] = len(r[2])
+            dldts[i] = 0xFFFFFFFF
+    fr = GpucFlushRows(n_rows=n, keys=keys, key_lens=klens, timestamps=tss,
+                       values=vals, value_lens=vlens, del_ldts=dldts)
+    err = ctypes.create_string_buffer(256)
+    rc = lib.gpuc_flush(ctypes.byref(fr), output_base.encode(), device, err, 256)
+    if rc != 0:
+        raise GpuCompactError(f"gpuc_flush rc={rc}: {err.value.decode(errors='replace')}")
 
 
 def verify(input_base, device=0):

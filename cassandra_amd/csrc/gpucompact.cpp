@@ -1950,6 +1950,138 @@ extern "C" int gpuc_verify(const char* input_base, int32_t device, char* error,
     }
 }
 
+// Memtable flush: unsorted unique-key rows from the host ->
+// token sort on device -> the shared serialize/compress/index/bloom kernels
+// -> one complete `oa` sstable. Schema: `pk blob PRIMARY KEY, val blob`
+// (one regular column; row tombstones via del_ldt != UINT32_MAX).
+extern "C" int gpuc_flush(const gpuc_flush_rows* rows, const char* output_base,
+                          int32_t device, char* error, size_t error_len) {
+    try {
+        if (gpuc_device_count() <= 0) { set_err(error, error_len, "no HIP device"); return GPUC_ERR_NO_GPU; }
+        if (!rows || rows->n_rows == 0) { set_err(error, error_len, "no rows"); return GPUC_ERR_UNSUPPORTED; }
+        HIP_CHECK(hipSetDevice(device));
+        hipStream_t stream;
+        HIP_CHECK(hipStreamCreate(&stream));
+        ensure_crc_tables(stream);
+        uint64_t n = rows->n_rows;
+        // flatten to arenas
+        std::vector<uint64_t> koff(n + 1, 0), voff(n + 1, 0);
+        for (uint64_t i = 0; i < n; i++) {
+            if (rows->key_lens[i] == 0) throw std::runtime_error("empty key");
+            koff[i + 1] = koff[i] + rows->key_lens[i];
+            voff[i + 1] = voff[i] + (rows->values[i] ? rows->value_lens[i] : 0);
+        }
+        std::vector<uint8_t> kbuf(koff[n]), vbuf(voff[n] ? voff[n] : 1);
+        std::vector<uint32_t> dldt(n);
+        for (uint64_t i = 0; i < n; i++) {
+            memcpy(kbuf.data() + koff[i], rows->keys[i], rows->key_lens[i]);
+            if (rows->values[i]) {
+                memcpy(vbuf.data() + voff[i], rows->values[i], rows->value_lens[i]);
+                dldt[i] = LDT_NONE_U32;
+            } else {
+                dldt[i] = rows->del_ldts[i];
+                if (dldt[i] == LDT_NONE_U32)
+                    throw std::runtime_error("row without value needs del_ldt");
+            }
+        }
+        DevBuf d_k, d_koff, d_v, d_voff, d_ts, d_dldt, d_a, d_b, d_err;
+        d_k.alloc(kbuf.size()); d_koff.alloc((n + 1) * 8);
+        d_v.alloc(vbuf.size()); d_voff.alloc((n + 1) * 8);
+        d_ts.alloc(n * 8); d_dldt.alloc(n * 4);
+        HIP_CHECK(hipMemcpyAsync(d_k.p, kbuf.data(), kbuf.size(), hipMemcpyHostToDevice, stream));
+        HIP_CHECK(hipMemcpyAsync(d_koff.p, koff.data(), (n + 1) * 8, hipMemcpyHostToDevice, stream));
+        HIP_CHECK(hipMemcpyAsync(d_v.p, vbuf.data(), vbuf.size(), hipMemcpyHostToDevice, stream));
+        HIP_CHECK(hipMemcpyAsync(d_voff.p, voff.data(), (n + 1) * 8, hipMemcpyHostToDevice, stream));
+        HIP_CHECK(hipMemcpyAsync(d_ts.p, rows->timestamps, n * 8, hipMemcpyHostToDevice, stream));
+        HIP_CHECK(hipMemcpyAsync(d_dldt.p, dldt.data(), n * 4, hipMemcpyHostToDevice, stream));
+        FlushParams fp{};
+        fp.keys = d_k.as<uint8_t>();
+        fp.key_off = d_koff.as<uint64_t>();
+        fp.ts = d_ts.as<int64_t>();
+        fp.vals = d_v.as<uint8_t>();
+        fp.val_off = d_voff.as<uint64_t>();
+        fp.del_ldt = d_dldt.as<uint32_t>();
+        fp.n = n;
+        d_a.alloc(n * sizeof(MRec));
+        d_b.alloc(n * sizeof(MRec));
+        uint32_t blocks = (uint32_t)((n + 255) / 256);
+        hipLaunchKernelGGL(k_flush_recs, dim3(blocks), dim3(256), 0, stream, fp, d_a.as<MRec>());
+        // exact-key comparator over the key arena (offsets pre-shifted by -2
+        // to cancel the short-length skip in mrec_key_bytes)
+        std::vector<uint64_t> koff_adj(n);
+        for (uint64_t i = 0; i < n; i++) koff_adj[i] = koff[i] - 2;
+        DevBuf d_koff_adj;
+        d_koff_adj.alloc(n * 8);
+        HIP_CHECK(hipMemcpyAsync(d_koff_adj.p, koff_adj.data(), n * 8, hipMemcpyHostToDevice, stream));
+        KeyLut lut{};
+        lut.base[0] = d_k.as<uint8_t>();
+        lut.pos[0] = d_koff_adj.as<uint64_t>();
+        lut.enabled = 1;
+        MRec* d_sorted = merge_sort_recs(d_a.as<MRec>(), d_b.as<MRec>(), n, stream, lut);
+        d_err.alloc(8);
+        HIP_CHECK(hipMemsetAsync(d_err.p, 0, 8, stream));
+        hipLaunchKernelGGL(k_flush_dupcheck, dim3(blocks), dim3(256), 0, stream, fp, d_sorted,
+                           d_err.as<unsigned long long>());
+        OutPartsBuf opb;
+        opb.alloc(n);
+        UnfColsBuf urows;
+        urows.alloc(n, 1, 0);
+        hipLaunchKernelGGL(k_flush_fill, dim3(blocks), dim3(256), 0, stream, fp, d_sorted,
+                           opb.op, urows.uc);
+        {
+            unsigned long long e = 0;
+            HIP_CHECK(hipStreamSynchronize(stream));
+            HIP_CHECK(hipMemcpy(&e, d_err.p, 8, hipMemcpyDeviceToHost));
+            if (e == 40) throw std::runtime_error("duplicate partition keys in flush input");
+        }
+        // header mins (SerializationHeader.make at flush: collect, then write)
+        DevBuf d_stats, d_tomb;
+        d_stats.alloc(sizeof(OutStats));
+        init_outstats(d_stats, stream);
+        uint32_t tomb_cap = (uint32_t)std::min<uint64_t>(n * 2 + 1024, 400000000ull);
+        d_tomb.alloc((uint64_t)tomb_cap * 4);
+        hipLaunchKernelGGL(k_collect_rows, dim3(blocks), dim3(256), 0, stream, opb.op,
+                           urows.uc, n, 1u, d_stats.as<OutStats>(), d_tomb.as<uint32_t>(), tomb_cap);
+        OutStats hs0;
+        HIP_CHECK(hipStreamSynchronize(stream));
+        HIP_CHECK(hipMemcpy(&hs0, d_stats.p, sizeof(OutStats), hipMemcpyDeviceToHost));
+        SerParams2 sp{};
+        sp.hs.min_ts = hs0.min_ts_flip == 0xFFFFFFFFFFFFFFFFULL
+                           ? TIMESTAMP_EPOCH : (int64_t)(hs0.min_ts_flip ^ 0x8000000000000000ULL);
+        sp.hs.min_ldt = hs0.min_ldt_flip == 0xFFFFFFFFFFFFFFFFULL
+                            ? DELETION_TIME_EPOCH : (int64_t)(hs0.min_ldt_flip ^ 0x8000000000000000ULL);
+        if (sp.hs.min_ldt == NO_DELETION_TIME) sp.hs.min_ldt = DELETION_TIME_EPOCH;
+        sp.hs.min_ttl = 0;
+        sp.sch.n_ck = 0;
+        DevBuf d_ckw;
+        d_ckw.alloc(8);
+        sp.sch.ck_w = d_ckw.as<int32_t>();
+        sp.sch.n_cols = 1;
+        std::vector<int32_t> cf{-1};
+        DevBuf d_cf;
+        d_cf.alloc(8);
+        HIP_CHECK(hipMemcpyAsync(d_cf.p, cf.data(), 4, hipMemcpyHostToDevice, stream));
+        sp.sch.col_fixed = d_cf.as<int32_t>();
+        sp.sch.n_static = 0;
+        DevBuf d_sf;
+        d_sf.alloc(8);
+        sp.sch.static_fixed = d_sf.as<int32_t>();
+        sp.sch.column_index_size = 64 * 1024;
+        init_outstats(d_stats, stream);
+        std::vector<std::pair<bytes, std::string>> cols = {
+            {bytes{'v', 'a', 'l'}, "org.apache.cassandra.db.marshal.BytesType"}};
+        write_sstable_device(opb, urows, n, sp, d_stats, d_tomb, tomb_cap, output_base,
+                             "org.apache.cassandra.db.marshal.BytesType",
+                             std::vector<std::string>{}, cols,
+                             std::vector<std::pair<bytes, std::string>>{}, stream);
+        HIP_CHECK(hipStreamDestroy(stream));
+        return GPUC_OK;
+    } catch (const std::exception& e) {
+        set_err(error, error_len, e.what());
+        return GPUC_ERR_INTERNAL;
+    }
+}
+
 extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* error, size_t error_len) {
     try {
         if (gpuc_device_count() <= 0) { set_err(error, error_len, "no HIP device"); return GPUC_ERR_NO_GPU; }

@@ -1765,6 +1765,85 @@ __global__ void k_collect_rows(OutParts op, UnfCols out, uint64_t n, uint32_t n_
 }
 
 // ---------------------------------------------------------------------------
+// flush path (Memtable -> sstable): host hands unsorted unique-key rows in
+// flat arenas; token+sort on device, then the shared writer kernels
+// ---------------------------------------------------------------------------
+struct FlushParams {
+    const uint8_t* keys;      // concatenated key bytes
+    const uint64_t* key_off;  // n+1 offsets into keys
+    const int64_t* ts;
+    const uint8_t* vals;      // concatenated value bytes
+    const uint64_t* val_off;  // n+1 offsets (del rows: off[i+1]==off[i])
+    const uint32_t* del_ldt;  // LDT_NONE_U32 = live row
+    uint64_t n;
+};
+
+__global__ void k_flush_recs(FlushParams fp, MRec* recs) {
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= fp.n) return;
+    const uint8_t* key = fp.keys + fp.key_off[i];
+    uint32_t klen = (uint32_t)(fp.key_off[i + 1] - fp.key_off[i]);
+    int64_t tok = murmur3_token(key, klen);
+    uint64_t pfx = 0;
+    uint32_t pb = klen < 8 ? klen : 8;
+    for (uint32_t b = 0; b < pb; b++) pfx |= (uint64_t)key[b] << (8 * (7 - b));
+    recs[i] = MRec{(uint64_t)tok ^ 0x8000000000000000ULL, pfx, (uint32_t)i, 0, (uint16_t)klen};
+}
+
+// uniqueness check: any adjacent equal (tok,pfx,klen) pair with equal bytes
+__global__ void k_flush_dupcheck(FlushParams fp, const MRec* recs, unsigned long long* error) {
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i == 0 || i >= fp.n) return;
+    const MRec &a = recs[i - 1], &b = recs[i];
+    if (a.tok != b.tok || a.pfx != b.pfx || a.klen != b.klen) return;
+    const uint8_t* ka = fp.keys + fp.key_off[a.idx];
+    const uint8_t* kb = fp.keys + fp.key_off[b.idx];
+    for (uint32_t x = 0; x < a.klen; x++)
+        if (ka[x] != kb[x]) return;
+    atomicExch(error, 40ull);
+}
+
+__global__ void k_flush_fill(FlushParams fp, const MRec* sorted, OutParts op, UnfCols out) {
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= fp.n) return;
+    uint32_t r = sorted[i].idx;
+    op.keypfx[i] = sorted[i].pfx;
+    op.key_addr[i] = (uint64_t)(fp.keys + fp.key_off[r]);
+    op.klen[i] = sorted[i].klen;
+    op.pdel_mfda[i] = INT64_MIN;
+    op.pdel_ldt[i] = LDT_NONE_U32;
+    op.row_base[i] = i;
+    op.row_count[i] = 1;
+    op.keep[i] = 1;
+    uint64_t o = i;
+    out.rkind[o] = BK_CLUSTERING;
+    out.ck_count[o] = 0;
+    out.start_mfda[o] = INT64_MIN;
+    out.start_ldt[o] = LDT_NONE_U32;
+    out.live_ttl[o] = 0;
+    out.live_let[o] = NO_DELETION_TIME;
+    bool tomb = fp.del_ldt[r] != LDT_NONE_U32;
+    if (tomb) {
+        out.flags[o] = PF_HAS_ROW | PF_ROW_DEL;
+        out.live_ts[o] = NO_TIMESTAMP;
+        out.rdel_mfda[o] = fp.ts[r];
+        out.rdel_ldt[o] = fp.del_ldt[r];
+        out.cell_flags[o] = 0;
+    } else {
+        out.flags[o] = PF_HAS_ROW | PF_LIVE_TS;
+        out.live_ts[o] = fp.ts[r];
+        out.rdel_mfda[o] = INT64_MIN;
+        out.rdel_ldt[o] = LDT_NONE_U32;
+        out.cell_flags[o] = CELLF_PRESENT | CELLF_HAS_VALUE;
+        out.cell_ts[o] = fp.ts[r];
+        out.cell_ldt[o] = LDT_NONE_U32;
+        out.cell_ttl[o] = 0;
+        out.val_addr[o] = (uint64_t)(fp.vals + fp.val_off[r]);
+        out.val_len[o] = (uint32_t)(fp.val_off[r + 1] - fp.val_off[r]);
+    }
+}
+
+// ---------------------------------------------------------------------------
 // verify epilogue (Verifier.java semantics: key order + bloom recheck)
 // ---------------------------------------------------------------------------
 // adjacent DecoratedKey order: every partition must sort strictly after its

@@ -138,6 +138,25 @@ typedef struct gpuc_gen_spec {
 
 int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* error, size_t error_len);
 
+/* Memtable flush (the write-path seed): n unsorted UNIQUE-key rows ->
+ * token-sorted on device -> one complete `oa` sstable under output_base.
+ * Schema: `pk blob PRIMARY KEY, val blob`. values[i] == NULL makes row i a
+ * row tombstone with local deletion time del_ldts[i] (else del_ldts ignored;
+ * pass UINT32_MAX). Replaces the flush serialize/compress path of
+ * Memtable.FlushablePartitionSet -> BigTableWriter. */
+typedef struct gpuc_flush_rows {
+    uint64_t n_rows;
+    const uint8_t* const* keys;
+    const uint16_t* key_lens;
+    const int64_t* timestamps;
+    const uint8_t* const* values;       /* NULL entry = row tombstone */
+    const uint32_t* value_lens;
+    const uint32_t* del_ldts;           /* UINT32_MAX = live */
+} gpuc_flush_rows;
+
+int gpuc_flush(const gpuc_flush_rows* rows, const char* output_base, int32_t device,
+               char* error, size_t error_len);
+
 /* Verify one sstable (Verifier.java / sstableverify --extended semantics):
  * CompressionInfo frame walk, per-chunk CRC32, full row-format walk,
  * strict DecoratedKey order, Digest.crc32 recomputation, bloom-filter

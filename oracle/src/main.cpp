@@ -5,6 +5,7 @@
 #include "gen.h"
 #include <chrono>
 #include <cstring>
+#include <algorithm>
 #include <map>
 #include <string>
 
@@ -128,6 +129,62 @@ static int cmd_gen(const std::string& outdir, std::map<std::string, std::string>
     return 0;
 }
 
+// flush rows from a text file: "keyhex ts valhex" | "keyhex ts T ldt"
+static int cmd_flush(const std::string& outbase, const std::string& rowfile) {
+    auto unhex = [](const std::string& s) {
+        bytes b;
+        for (size_t i = 0; i + 1 < s.size(); i += 2)
+            b.push_back((uint8_t)strtoul(s.substr(i, 2).c_str(), nullptr, 16));
+        return b;
+    };
+    SSTable t;
+    t.generation = 1;
+    t.header.key_type = CqlType::BYTES;
+    t.header.regular_cols = {{bytes{'v', 'a', 'l'}, CqlType::BYTES}};
+    FILE* f = fopen(rowfile.c_str(), "r");
+    if (!f) { fprintf(stderr, "cannot open %s\n", rowfile.c_str()); return 1; }
+    char k[4096], v[65536];
+    long long ts;
+    int64_t min_ts = INT64_MAX, min_ldt_l = INT64_MAX;
+    while (fscanf(f, "%4095s %lld %65535s", k, &ts, v) == 3) {
+        Partition p;
+        p.key = unhex(k);
+        p.set_token();
+        Unfiltered u;
+        u.kind = Unfiltered::ROW;
+        Row& r = u.row;
+        r.cells.resize(1);
+        min_ts = std::min(min_ts, (int64_t)ts);
+        if (v[0] == 'T') {
+            long long ldt;
+            if (fscanf(f, "%lld", &ldt) != 1) break;
+            r.del.mfda = ts;
+            r.del.ldt = (uint32_t)ldt;
+            min_ldt_l = std::min(min_ldt_l, (int64_t)ldt);
+        } else {
+            r.live.ts = ts;
+            Cell c;
+            c.ts = ts;
+            c.value = unhex(v);
+            r.cells[0] = std::move(c);
+        }
+        p.items.push_back(std::move(u));
+        t.parts.push_back(std::move(p));
+    }
+    fclose(f);
+    std::sort(t.parts.begin(), t.parts.end(), [](const Partition& a, const Partition& b) {
+        return compare_decorated_key(a.token, a.key.data(), a.key.size(), b.token,
+                                     b.key.data(), b.key.size()) < 0;
+    });
+    t.header.stats.min_ts = min_ts == INT64_MAX ? TIMESTAMP_EPOCH : min_ts;
+    t.header.stats.min_ldt = min_ldt_l == INT64_MAX ? DELETION_TIME_EPOCH : min_ldt_l;
+    t.header.stats.min_ttl = 0;
+    WriterOut w = write_sstable(t);
+    write_components(w, outbase);
+    printf("flushed %zu partitions\n", t.parts.size());
+    return 0;
+}
+
 static int cmd_compact(const std::string& outbase, std::vector<std::string>& inputs,
                        std::map<std::string, std::string>& kv) {
     using clk = std::chrono::steady_clock;
@@ -194,6 +251,7 @@ int main(int argc, char** argv) {
         if (cmd == "dump") return cmd_dump(pos.at(0));
         if (cmd == "roundtrip") return cmd_roundtrip(pos.at(0));
         if (cmd == "gen") return cmd_gen(pos.at(0), kv);
+        if (cmd == "flush") return cmd_flush(pos.at(0), pos.at(1));
         if (cmd == "compact") {
             std::string outbase = pos.at(0);
             std::vector<std::string> ins(pos.begin() + 1, pos.end());

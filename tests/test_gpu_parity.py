@@ -273,6 +273,38 @@ def test_sharded_output_parity_wide(ca, oracle_bin, tmp_path):
         _assert_dirs_equal(f"{d}/oa-{300+i}-big", f"{d}/oa-{200+i}-big")
 
 
+def test_flush_parity(ca, oracle_bin, tmp_path):
+    """gpuc_flush (memtable-flush analog): unsorted unique-key host rows ->
+    GPU token sort + writer == oracle flush of the same rows, byte for byte."""
+    import random
+    d = str(tmp_path)
+    rng = random.Random(117)
+    rows, lines = [], []
+    seen = set()
+    for i in range(3000):
+        klen = rng.choice([4, 8, 12, 24])
+        key = bytes(rng.randrange(256) for _ in range(klen))
+        if key in seen:
+            continue
+        seen.add(key)
+        ts = 1700000000000000 + rng.randrange(10**9)
+        if rng.random() < 0.15:
+            ldt = 1700000000 + rng.randrange(1000)
+            rows.append((key, ts, None, ldt))
+            lines.append(f"{key.hex()} {ts} T {ldt}")
+        else:
+            val = bytes(rng.randrange(256) for _ in range(rng.randrange(1, 400)))
+            rows.append((key, ts, val))
+            lines.append(f"{key.hex()} {ts} {val.hex()}")
+    with open(f"{d}/rows.txt", "w") as f:
+        f.write("\n".join(lines) + "\n")
+    subprocess.run([ORACLE, "flush", f"{d}/oa-1-big", f"{d}/rows.txt"],
+                   check=True, capture_output=True)
+    ca.flush(rows, f"{d}/oa-2-big")
+    _assert_dirs_equal(f"{d}/oa-1-big", f"{d}/oa-2-big")
+    ca.verify(f"{d}/oa-2-big")
+
+
 def test_verify_epilogue(ca, oracle_bin, tmp_path):
     """gpuc_verify passes on valid sstables (oracle- and GPU-written, all
     schema shapes) and fails on corruption of data, digest or filter."""
