@@ -35,6 +35,9 @@ class GpucJob(ctypes.Structure):
         ("token_lo", ctypes.c_int64),
         ("token_hi", ctypes.c_int64),
         ("device", ctypes.c_int32),
+        ("tombstone_source_bases", ctypes.POINTER(ctypes.c_char_p)),
+        ("n_tomb_sources", ctypes.c_int32),
+        ("cell_level_gc", ctypes.c_int32),
         ("n_output_shards", ctypes.c_int32),
     ]
 
@@ -211,6 +214,8 @@ def compact(
     token_range=None,
     device=0,
     n_output_shards=1,
+    tombstone_sources=None,
+    cell_level_gc=False,
 ):
     """One compaction task (mirrors CompactionTask.runMayThrow's hot loop).
 
@@ -240,6 +245,11 @@ def compact(
         job.token_lo, job.token_hi = token_range
     job.device = device
     job.n_output_shards = n_output_shards
+    if tombstone_sources:
+        tarr = (ctypes.c_char_p * len(tombstone_sources))(*[b.encode() for b in tombstone_sources])
+        job.tombstone_source_bases = tarr
+        job.n_tomb_sources = len(tombstone_sources)
+    job.cell_level_gc = 1 if cell_level_gc else 0
     res = GpucResult()
     rc = lib.gpuc_compact(ctypes.byref(job), ctypes.byref(res))
     if rc != 0:

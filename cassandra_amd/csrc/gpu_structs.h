@@ -144,6 +144,7 @@ struct ParsedCols {
 struct OutParts {
     uint64_t* keypfx;    // big-endian zero-padded key prefix (debug/aux)
     uint64_t* key_addr;  // device address of the full key bytes
+    int64_t* token;      // Murmur3 token (garbage-collect source matching)
     StaticCols st;       // merged static row per output partition
     uint16_t* klen;
     int64_t* pdel_mfda;

@@ -694,15 +694,16 @@ struct StaticColsBuf {
 };
 
 struct OutPartsBuf {
-    DevBuf keypfx, key_addr, klen, pdel_mfda, pdel_ldt, row_base, row_count, keep;
+    DevBuf keypfx, key_addr, token, klen, pdel_mfda, pdel_ldt, row_base, row_count, keep;
     StaticColsBuf stb;
     OutParts op{};
     void alloc(uint64_t n, uint32_t n_static = 0) {
-        keypfx.alloc(n * 8); key_addr.alloc(n * 8); klen.alloc(n * 2);
+        keypfx.alloc(n * 8); key_addr.alloc(n * 8); token.alloc(n * 8); klen.alloc(n * 2);
         pdel_mfda.alloc(n * 8); pdel_ldt.alloc(n * 4);
         row_base.alloc(n * 8); row_count.alloc(n * 4); keep.alloc(n);
         stb.alloc(n_static ? n : 1, n_static);
-        op = OutParts{keypfx.as<uint64_t>(), key_addr.as<uint64_t>(), stb.st, klen.as<uint16_t>(),
+        op = OutParts{keypfx.as<uint64_t>(), key_addr.as<uint64_t>(), token.as<int64_t>(),
+                      stb.st, klen.as<uint16_t>(),
                       pdel_mfda.as<int64_t>(),
                       pdel_ldt.as<uint32_t>(), row_base.as<uint64_t>(), row_count.as<uint32_t>(),
                       keep.as<uint8_t>()};
@@ -1129,7 +1130,8 @@ static int32_t ck_type_width(const std::string& t) {
 }
 
 struct CompactSetup {
-    int k = 0;
+    int k = 0;        // total ingested sstables (data + tombstone sources)
+    int k_data = 0;   // first k_data are the compacted data inputs
     std::vector<std::string> in_bases;
     std::vector<bytes> index_data;
     std::vector<HCompressionInfo> cinfos;
@@ -1391,27 +1393,33 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         lut.enabled = 1;
 
         // ---- merge (pairwise rounds over pre-sorted source runs) ----
+        // In garbage-collect mode the first k_data inputs are the compacted
+        // data; the rest are tombstone sources merged in a second pass.
+        int kd = su.k_data > 0 ? su.k_data : k;
+        bool gc_mode = kd < k;
+        uint64_t data_parts = 0;
+        for (int s = 0; s < kd; s++) data_parts += srcs[s].n_parts;
         std::vector<uint64_t> runs;
         runs.push_back(0);
-        for (int s = 0; s < k; s++) runs.push_back(runs.back() + srcs[s].n_parts);
+        for (int s = 0; s < kd; s++) runs.push_back(runs.back() + srcs[s].n_parts);
         MRec* d_sorted = merge_sorted_runs(d_recs_a.as<MRec>(), d_recs_b.as<MRec>(), runs, stream, lut);
         TR("merge issued");
 
         // ---- group heads + starts ----
         DevBuf d_head, d_gstart, d_ngroups;
-        d_head.alloc(total_parts * 8);
-        d_gstart.alloc(total_parts * 8);
+        d_head.alloc(data_parts * 8 + 8);
+        d_gstart.alloc(data_parts * 8 + 8);
         d_ngroups.alloc(8);
         {
-            uint32_t blocks = (uint32_t)((total_parts + 255) / 256);
+            uint32_t blocks = (uint32_t)((data_parts + 255) / 256);
             hipLaunchKernelGGL(k_group_heads, dim3(blocks), dim3(256), 0, stream, d_sorted,
-                               total_parts, d_head.as<uint64_t>(), lut);
+                               data_parts, d_head.as<uint64_t>(), lut);
         }
-        exscan_u64(d_head.as<uint64_t>(), total_parts, stream);
+        exscan_u64(d_head.as<uint64_t>(), data_parts, stream);
         {
-            uint32_t blocks = (uint32_t)((total_parts + 255) / 256);
+            uint32_t blocks = (uint32_t)((data_parts + 255) / 256);
             hipLaunchKernelGGL(k_group_starts2, dim3(blocks), dim3(256), 0, stream, d_sorted,
-                               total_parts, d_head.as<uint64_t>(), d_gstart.as<uint64_t>(),
+                               data_parts, d_head.as<uint64_t>(), d_gstart.as<uint64_t>(),
                                d_ngroups.as<uint64_t>(), lut);
         }
         uint64_t n_groups = 0;
@@ -1431,7 +1439,7 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         {
             uint32_t blocks = (uint32_t)((n_groups + 255) / 256);
             hipLaunchKernelGGL(k_group_row_sums, dim3(blocks), dim3(256), 0, stream, d_sorted,
-                               d_gstart.as<uint64_t>(), n_groups, total_parts,
+                               d_gstart.as<uint64_t>(), n_groups, data_parts,
                                d_srcbases.as<uint32_t>(), pc, d_group_rows.as<uint64_t>());
         }
         uint64_t total_out_rows = exscan_u64(d_group_rows.as<uint64_t>(), n_groups, stream);
@@ -1476,17 +1484,25 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         HIP_CHECK(hipEventCreate(&er0));
         HIP_CHECK(hipEventCreate(&er1));
         HIP_CHECK(hipEventRecord(er0, stream));
+        // GC mode: purge is DEFERRED until after the garbage filter
+        // (GarbageSkipper runs before the Purger in CompactionIterator)
+        PurgeParams2 pp_data = pp;
+        if (gc_mode) {
+            pp_data.gc_before = INT64_MIN;
+            pp_data.never_purge = 1;
+            pp_data.enforce_strict_liveness = 0;
+        }
         {
             uint32_t blocks = (uint32_t)((n_groups + 255) / 256);
             auto launch_rec = [&](auto kern) {
                 hipLaunchKernelGGL(kern, dim3(blocks), dim3(256), 0, stream, d_sorted,
-                                   d_gstart.as<uint64_t>(), n_groups, total_parts,
+                                   d_gstart.as<uint64_t>(), n_groups, data_parts,
                                    d_srcbases.as<uint32_t>(), pc, in_rows.uc, opb.op, out_rows.uc,
-                                   d_group_rows.as<uint64_t>(), sch, pp, d_stats.as<OutStats>(),
+                                   d_group_rows.as<uint64_t>(), sch, pp_data, d_stats.as<OutStats>(),
                                    d_error.as<unsigned long long>());
             };
-            if (k <= 8) launch_rec(k_reconcile_rows<8>);
-            else if (k <= 16) launch_rec(k_reconcile_rows<16>);
+            if (kd <= 8) launch_rec(k_reconcile_rows<8>);
+            else if (kd <= 16) launch_rec(k_reconcile_rows<16>);
             else launch_rec(k_reconcile_rows<64>);
         }
         HIP_CHECK(hipEventRecord(er1, stream));
@@ -1499,12 +1515,108 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
             if (err) throw std::runtime_error("GPU reconcile error code " + std::to_string(err));
         }
 
+        // ---- garbage collect: merge the tombstone sources, filter, purge ----
+        UnfColsBuf gc_rows;          // filtered data arena (replaces out_rows)
+        UnfColsBuf* rows_for_writer = &out_rows;
+        OutPartsBuf opb_t;
+        UnfColsBuf out_rows_t;
+        if (gc_mode && n_groups > 0) {
+            uint64_t src_parts = total_parts - data_parts;
+            uint64_t n_groups_t = 0;
+            if (src_parts > 0) {
+                std::vector<uint64_t> runs_t;
+                runs_t.push_back(0);
+                for (int s2 = kd; s2 < k; s2++) runs_t.push_back(runs_t.back() + srcs[s2].n_parts);
+                MRec* d_sorted_t = merge_sorted_runs(d_recs_a.as<MRec>() + data_parts,
+                                                     d_recs_b.as<MRec>() + data_parts, runs_t,
+                                                     stream, lut);
+                DevBuf d_head_t, d_gstart_t, d_ng_t;
+                d_head_t.alloc(src_parts * 8 + 8);
+                d_gstart_t.alloc(src_parts * 8 + 8);
+                d_ng_t.alloc(8);
+                uint32_t blocks_t = (uint32_t)((src_parts + 255) / 256);
+                hipLaunchKernelGGL(k_group_heads, dim3(blocks_t), dim3(256), 0, stream, d_sorted_t,
+                                   src_parts, d_head_t.as<uint64_t>(), lut);
+                exscan_u64(d_head_t.as<uint64_t>(), src_parts, stream);
+                hipLaunchKernelGGL(k_group_starts2, dim3(blocks_t), dim3(256), 0, stream, d_sorted_t,
+                                   src_parts, d_head_t.as<uint64_t>(), d_gstart_t.as<uint64_t>(),
+                                   d_ng_t.as<uint64_t>(), lut);
+                HIP_CHECK(hipStreamSynchronize(stream));
+                HIP_CHECK(hipMemcpy(&n_groups_t, d_ng_t.p, 8, hipMemcpyDeviceToHost));
+                DevBuf d_grows_t, d_stats_t;
+                d_grows_t.alloc(n_groups_t * 8 + 8);
+                {
+                    uint32_t blocks_g = (uint32_t)((n_groups_t + 255) / 256);
+                    hipLaunchKernelGGL(k_group_row_sums, dim3(blocks_g), dim3(256), 0, stream,
+                                       d_sorted_t, d_gstart_t.as<uint64_t>(), n_groups_t, src_parts,
+                                       d_srcbases.as<uint32_t>(), pc, d_grows_t.as<uint64_t>());
+                }
+                uint64_t t_out_rows = exscan_u64(d_grows_t.as<uint64_t>(), n_groups_t, stream);
+                opb_t.alloc(n_groups_t ? n_groups_t : 1, sch.n_static);
+                out_rows_t.alloc(t_out_rows, sch.n_cols, sch.n_ck);
+                d_stats_t.alloc(sizeof(OutStats));
+                init_outstats(d_stats_t, stream);
+                PurgeParams2 pp_src = pp_data;
+                pp_src.has_shard = 0;  // sources shadow regardless of the shard
+                {
+                    uint32_t blocks_g = (uint32_t)((n_groups_t + 255) / 256);
+                    auto launch_rec_t = [&](auto kern) {
+                        hipLaunchKernelGGL(kern, dim3(blocks_g), dim3(256), 0, stream, d_sorted_t,
+                                           d_gstart_t.as<uint64_t>(), n_groups_t, src_parts,
+                                           d_srcbases.as<uint32_t>(), pc, in_rows.uc, opb_t.op,
+                                           out_rows_t.uc, d_grows_t.as<uint64_t>(), sch, pp_src,
+                                           d_stats_t.as<OutStats>(),
+                                           d_error.as<unsigned long long>());
+                    };
+                    int ks = k - kd;
+                    if (ks <= 8) launch_rec_t(k_reconcile_rows<8>);
+                    else if (ks <= 16) launch_rec_t(k_reconcile_rows<16>);
+                    else launch_rec_t(k_reconcile_rows<64>);
+                }
+            }
+            // match + capacity scan + filter
+            DevBuf d_tidx, d_cap;
+            d_tidx.alloc(n_groups * 8);
+            d_cap.alloc(n_groups * 8);
+            {
+                uint32_t blocks_g = (uint32_t)((n_groups + 255) / 256);
+                hipLaunchKernelGGL(k_gc_match, dim3(blocks_g), dim3(256), 0, stream, opb.op,
+                                   n_groups, opb_t.op, n_groups_t, d_tidx.as<int64_t>(),
+                                   d_cap.as<uint64_t>());
+            }
+            uint64_t gc_total = exscan_u64(d_cap.as<uint64_t>(), n_groups, stream);
+            gc_rows.alloc(gc_total, sch.n_cols, sch.n_ck);
+            {
+                uint32_t blocks_g = (uint32_t)((n_groups + 255) / 256);
+                hipLaunchKernelGGL(k_garbage_filter, dim3(blocks_g), dim3(256), 0, stream, opb.op,
+                                   n_groups, opb_t.op, out_rows.uc, out_rows_t.uc, gc_rows.uc,
+                                   d_tidx.as<int64_t>(), d_cap.as<uint64_t>(), sch,
+                                   job->cell_level_gc ? 1 : 0);
+            }
+            // reset first/last group, then the deferred purge pass
+            HIP_CHECK(hipMemsetAsync((uint8_t*)d_stats.p + offsetof(OutStats, first_group), 0xFF, 8, stream));
+            HIP_CHECK(hipMemsetAsync((uint8_t*)d_stats.p + offsetof(OutStats, last_group), 0x00, 8, stream));
+            {
+                uint32_t blocks_g = (uint32_t)((n_groups + 255) / 256);
+                hipLaunchKernelGGL(k_purge_parts, dim3(blocks_g), dim3(256), 0, stream, opb.op,
+                                   gc_rows.uc, n_groups, sch, pp, d_stats.as<OutStats>());
+            }
+            HIP_CHECK(hipStreamSynchronize(stream));
+            {
+                unsigned long long err = 0;
+                HIP_CHECK(hipMemcpy(&err, d_error.p, 8, hipMemcpyDeviceToHost));
+                if (err) throw std::runtime_error("GPU gc filter error code " + std::to_string(err));
+            }
+            rows_for_writer = &gc_rows;
+            TR("garbage filtered");
+        }
+
         // ---- output header (SerializationHeader.make: desc-generation stats merge) ----
         SerParams2 sp2{};
         sp2.sch = sch;
         {
-            std::vector<int> order(k);
-            for (int s = 0; s < k; s++) order[s] = s;
+            std::vector<int> order(kd);
+            for (int s = 0; s < kd; s++) order[s] = s;
             std::stable_sort(order.begin(), order.end(),
                              [&](int a, int b) { return generations[a] > generations[b]; });
             int64_t min_ts = INT64_MAX, min_ldt = INT64_MAX;
@@ -1518,7 +1630,7 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
             sp2.hs.min_ldt = min_ldt == NO_DELETION_TIME ? DELETION_TIME_EPOCH : min_ldt;
             sp2.hs.min_ttl = min_ttl == INT32_MAX ? 0 : min_ttl;
         }
-        WriteDeviceOut w = write_sstable_device(opb, out_rows, n_groups, sp2, d_stats, d_tomb,
+        WriteDeviceOut w = write_sstable_device(opb, *rows_for_writer, n_groups, sp2, d_stats, d_tomb,
                                                 tomb_cap, out_base_str, stats[0].key_type,
                                                 stats[0].clustering_types, stats[0].regular_cols,
                                                 stats[0].static_cols, stream, wslot);
@@ -1573,8 +1685,9 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
 // ---------------------------------------------------------------------------
 
 static void compact_setup(const gpuc_job* job, CompactSetup& su, bool preread_full) {
-    int k = job->n_inputs;
+    int k = job->n_inputs + (job->n_tomb_sources > 0 ? job->n_tomb_sources : 0);
     su.k = k;
+    su.k_data = job->n_inputs;
     su.in_bases.resize(k);
     su.index_data.resize(k);
     su.cinfos.resize(k);
@@ -1589,7 +1702,8 @@ static void compact_setup(const gpuc_job* job, CompactSetup& su, bool preread_fu
         su.full_pin.resize(k, nullptr);
         su.full_readers.resize(k);
         for (int s = 0; s < k; s++) {
-            su.in_bases[s] = job->input_bases[s];
+            su.in_bases[s] = s < su.k_data ? job->input_bases[s]
+                                           : job->tombstone_source_bases[s - su.k_data];
             su.comp_file_sz[s] = file_size_of(su.in_bases[s] + "-Data.db");
             su.full_pin[s] = (uint8_t*)g_pin_in[0][s].get(su.comp_file_sz[s] ? su.comp_file_sz[s] : 1);
             if (!su.full_pin[s]) throw std::runtime_error("pinned alloc failed");
@@ -1603,7 +1717,8 @@ static void compact_setup(const gpuc_job* job, CompactSetup& su, bool preread_fu
         std::vector<std::thread> mth;
         std::vector<std::string> merr(k);
         for (int s = 0; s < k; s++) {
-            su.in_bases[s] = job->input_bases[s];
+            su.in_bases[s] = s < su.k_data ? job->input_bases[s]
+                                           : job->tombstone_source_bases[s - su.k_data];
             mth.emplace_back([&, s, preread_full]() {
                 try {
                     const std::string& base = su.in_bases[s];
@@ -1694,8 +1809,12 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
             ensure_crc_tables(s0);
             HIP_CHECK(hipStreamDestroy(s0));
         }
-        if (job->n_inputs < 1 || job->n_inputs > 64) {
-            set_err(res->error, sizeof(res->error), "n_inputs must be 1..64");
+        if (job->n_inputs < 1 || job->n_inputs + (job->n_tomb_sources > 0 ? job->n_tomb_sources : 0) > 64) {
+            set_err(res->error, sizeof(res->error), "n_inputs (+ tombstone sources) must be 1..64");
+            return GPUC_ERR_UNSUPPORTED;
+        }
+        if (job->n_tomb_sources > 0 && job->n_output_shards > 1) {
+            set_err(res->error, sizeof(res->error), "garbage collect + sharded outputs unsupported");
             return GPUC_ERR_UNSUPPORTED;
         }
         double t0 = wall();
@@ -1704,9 +1823,9 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
         compact_setup(job, su, S_pre == 1);
         res->ms_read_io = wall() - t0;
         TR("meta+index parsed");
-        for (int s = 0; s < su.k; s++)
+        for (int s = 0; s < su.k_data; s++)
             res->input_uncompressed_bytes += su.cinfos[s].data_len;
-        for (int s = 0; s < su.k; s++)
+        for (int s = 0; s < su.k_data; s++)
             res->partitions_in += su.positions[s].size() - 1;
 
         int S = S_pre;

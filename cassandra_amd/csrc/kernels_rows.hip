@@ -494,6 +494,7 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
     int64_t token = (int64_t)(r0.tok ^ 0x8000000000000000ULL);
     op.keypfx[g] = r0.pfx;
     op.key_addr[g] = pc.key_addr[src_bases[r0.src] + r0.idx];
+    op.token[g] = token;
     op.klen[g] = r0.klen;
     op.row_base[g] = out_base[g];
     op.row_count[g] = 0;
@@ -1765,6 +1766,533 @@ __global__ void k_collect_rows(OutParts op, UnfCols out, uint64_t n, uint32_t n_
 }
 
 // ---------------------------------------------------------------------------
+// garbage collect (nodetool garbagecollect / CompactionIterator.GarbageSkipper
+// CompactionIterator.java:401-635): drop data shadowed by tombstone SOURCES.
+// Two-phase: data and sources are merged separately (purge deferred), then
+// a thread-per-partition state machine filters the merged data stream
+// against the merged source stream, then k_purge_parts applies PurgeFunction.
+// ---------------------------------------------------------------------------
+__device__ inline bool dtp_sup(int64_t am, uint32_t al, int64_t bm, uint32_t bl) {
+    return am > bm || (am == bm && ldt_long(al) > ldt_long(bl));
+}
+// cross-container position compare (same rules as pos_cmp)
+__device__ inline int pos_cmp2(const UnfCols& A, uint64_t a, const UnfCols& B, uint64_t b,
+                               const SchemaParams& sch) {
+    if (sch.n_ck) {
+        uint32_t na = A.ck_count[a], nb = B.ck_count[b];
+        uint32_t mn = na < nb ? na : nb;
+        for (uint32_t c = 0; c < mn; c++) {
+            uint64_t ac = a * sch.n_ck + c, bc = b * sch.n_ck + c;
+            uint64_t pa = A.ck[ac], pb = B.ck[bc];
+            if (pa != pb) return pa < pb ? -1 : 1;
+            if (sch.ck_w[c] < 0) {
+                uint32_t la = A.ck_len[ac], lb = B.ck_len[bc];
+                if (la > 8 || lb > 8) {
+                    const uint8_t* a8 = (const uint8_t*)A.ck_addr[ac];
+                    const uint8_t* b8 = (const uint8_t*)B.ck_addr[bc];
+                    uint32_t n = la < lb ? la : lb;
+                    for (uint32_t i = 8; i < n; i++)
+                        if (a8[i] != b8[i]) return a8[i] < b8[i] ? -1 : 1;
+                }
+                if (la != lb) return la < lb ? -1 : 1;
+            }
+        }
+        if (na != nb)
+            return na < nb ? bk_cmp_to_clustering(A.rkind[a]) : -bk_cmp_to_clustering(B.rkind[b]);
+    }
+    return bk_comparison(A.rkind[a]) - bk_comparison(B.rkind[b]);
+}
+// Cells.reconcile(a, b) == b ? (the source cell wins)
+__device__ inline bool gc_cell_b_wins(int64_t ats, uint32_t aldt, bool aexp, uint64_t av,
+                                      uint32_t al, int64_t bts, uint32_t bldt, bool bexp,
+                                      uint64_t bv, uint32_t bl) {
+    if (ats != bts) return bts > ats;
+    bool a_dt = aldt != LDT_NONE_U32, b_dt = bldt != LDT_NONE_U32;
+    if (a_dt || b_dt) {
+        if (a_dt != b_dt) return b_dt;
+        bool a_tomb = !aexp, b_tomb = !bexp;
+        if (a_tomb != b_tomb) return b_tomb;
+        if (aldt != bldt) return ldt_long(bldt) > ldt_long(aldt);
+        return cmp_values(av, al, bv, bl) < 0;
+    }
+    return cmp_values(av, al, bv, bl) < 0;
+}
+
+// per data partition: match to a source partition by decorated key, produce
+// the filtered capacity (data rows + matched tomb rows for reissue headroom)
+__global__ void k_gc_match(OutParts dp, uint64_t nd, OutParts tp, uint64_t nt,
+                           int64_t* tomb_idx, uint64_t* cap) {
+    uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (g >= nd) return;
+    cap[g] = dp.row_count[g];
+    tomb_idx[g] = -1;
+    if (!dp.keep[g] || nt == 0) return;
+    int64_t tok = dp.token[g];
+    const uint8_t* key = (const uint8_t*)dp.key_addr[g];
+    uint32_t klen = dp.klen[g];
+    uint64_t lo = 0, hi = nt;
+    while (lo < hi) {  // lower_bound by (token, key bytes)
+        uint64_t mid = (lo + hi) >> 1;
+        int c = tp.token[mid] < tok ? -1 : tp.token[mid] > tok ? 1 : 0;
+        if (c == 0) {
+            const uint8_t* mk = (const uint8_t*)tp.key_addr[mid];
+            uint32_t ml = tp.klen[mid];
+            uint32_t n = ml < klen ? ml : klen;
+            for (uint32_t i = 0; i < n && c == 0; i++)
+                c = mk[i] < key[i] ? -1 : mk[i] > key[i] ? 1 : 0;
+            if (c == 0) c = ml < klen ? -1 : ml > klen ? 1 : 0;
+        }
+        if (c < 0) lo = mid + 1;
+        else hi = mid;
+    }
+    if (lo >= nt || tp.token[lo] != tok || tp.klen[lo] != klen) return;
+    const uint8_t* mk = (const uint8_t*)tp.key_addr[lo];
+    for (uint32_t i = 0; i < klen; i++)
+        if (mk[i] != key[i]) return;
+    tomb_idx[g] = (int64_t)lo;
+    cap[g] = dp.row_count[g] + tp.row_count[lo];
+}
+
+// BTreeRow.filter(all, active, false): drop liveness/cells/deletion shadowed
+// by `active`; returns row-non-empty. Operates on the OUTPUT arena row `o`.
+__device__ inline bool gc_row_filter_active(UnfCols& out, uint64_t o, int64_t am, uint32_t al,
+                                            const SchemaParams& sch) {
+    bool active_live = am == INT64_MIN && al == LDT_NONE_U32;
+    uint8_t f = out.flags[o];
+    bool may = !active_live && !dtp_sup(out.rdel_mfda[o], out.rdel_ldt[o], am, al);
+    bool any_cell = false;
+    if (may) {
+        if ((f & PF_LIVE_TS) && out.live_ts[o] <= am) {
+            f &= ~PF_LIVE_TS;
+            out.live_ts[o] = NO_TIMESTAMP;
+            out.live_ttl[o] = 0;
+            out.live_let[o] = NO_DELETION_TIME;
+        }
+        f &= ~PF_ROW_DEL;
+        out.rdel_mfda[o] = INT64_MIN;
+        out.rdel_ldt[o] = LDT_NONE_U32;
+        for (uint32_t c = 0; c < sch.n_cols; c++) {
+            uint64_t oc = o * sch.n_cols + c;
+            if (!(out.cell_flags[oc] & CELLF_PRESENT)) continue;
+            if (out.cell_ts[oc] <= am) out.cell_flags[oc] = 0;
+            else any_cell = true;
+        }
+    } else {
+        for (uint32_t c = 0; c < sch.n_cols; c++)
+            if (out.cell_flags[o * sch.n_cols + c] & CELLF_PRESENT) any_cell = true;
+    }
+    bool keep = (f & (PF_LIVE_TS | PF_ROW_DEL)) || any_cell;
+    out.flags[o] = keep ? (f | PF_HAS_ROW) : 0;
+    return keep;
+}
+
+// garbageFilterRow on the output arena row `o` vs tomb arena row `t`
+__device__ inline bool gc_filter_row_vs(UnfCols& out, uint64_t o, const UnfCols& tin, uint64_t t,
+                                        int64_t am, uint32_t al, bool cell_level,
+                                        const SchemaParams& sch) {
+    int64_t tdm = (tin.flags[t] & PF_ROW_DEL) ? tin.rdel_mfda[t] : INT64_MIN;
+    uint32_t tdl = (tin.flags[t] & PF_ROW_DEL) ? tin.rdel_ldt[t] : LDT_NONE_U32;
+    int64_t dm = am;
+    uint32_t dl = al;
+    if (dtp_sup(tdm, tdl, dm, dl)) { dm = tdm; dl = tdl; }
+    if (!cell_level) return gc_row_filter_active(out, o, dm, dl, sch);
+    // Rows.removeShadowedCells
+    uint8_t f = out.flags[o];
+    if ((f & PF_LIVE_TS) && out.live_ts[o] <= dm) {
+        f &= ~PF_LIVE_TS;
+        out.live_ts[o] = NO_TIMESTAMP;
+        out.live_ttl[o] = 0;
+        out.live_let[o] = NO_DELETION_TIME;
+    }
+    if ((f & PF_ROW_DEL) && dtp_sup(dm, dl, out.rdel_mfda[o], out.rdel_ldt[o])) {
+        f &= ~PF_ROW_DEL;
+        out.rdel_mfda[o] = INT64_MIN;
+        out.rdel_ldt[o] = LDT_NONE_U32;
+    }
+    bool any_cell = false;
+    for (uint32_t c = 0; c < sch.n_cols; c++) {
+        uint64_t oc = o * sch.n_cols + c, tc = t * sch.n_cols + c;
+        uint8_t cf = out.cell_flags[oc];
+        if (!(cf & CELLF_PRESENT)) continue;
+        if (out.cell_ts[oc] <= dm) { out.cell_flags[oc] = 0; continue; }
+        uint8_t tf = tin.cell_flags[tc];
+        if ((tf & CELLF_PRESENT) &&
+            gc_cell_b_wins(out.cell_ts[oc], out.cell_ldt[oc], cf & CELLF_EXPIRING,
+                           out.val_addr[oc], out.val_len[oc], tin.cell_ts[tc], tin.cell_ldt[tc],
+                           tf & CELLF_EXPIRING, tin.val_addr[tc], tin.val_len[tc])) {
+            out.cell_flags[oc] = 0;
+            continue;
+        }
+        any_cell = true;
+    }
+    bool keep = (f & (PF_LIVE_TS | PF_ROW_DEL)) || any_cell;
+    out.flags[o] = keep ? (f | PF_HAS_ROW) : 0;
+    return keep;
+}
+
+__device__ inline void gc_copy_unf(UnfCols& dst, uint64_t d, const UnfCols& src, uint64_t s,
+                                   const SchemaParams& sch) {
+    dst.rkind[d] = src.rkind[s];
+    dst.flags[d] = src.flags[s];
+    dst.live_ts[d] = src.live_ts[s];
+    dst.live_ttl[d] = src.live_ttl[s];
+    dst.live_let[d] = src.live_let[s];
+    dst.rdel_mfda[d] = src.rdel_mfda[s];
+    dst.rdel_ldt[d] = src.rdel_ldt[s];
+    dst.start_mfda[d] = src.start_mfda[s];
+    dst.start_ldt[d] = src.start_ldt[s];
+    dst.ck_count[d] = src.ck_count[s];
+    for (uint32_t c = 0; c < sch.n_ck; c++) {
+        dst.ck[d * sch.n_ck + c] = src.ck[s * sch.n_ck + c];
+        dst.ck_addr[d * sch.n_ck + c] = src.ck_addr[s * sch.n_ck + c];
+        dst.ck_len[d * sch.n_ck + c] = src.ck_len[s * sch.n_ck + c];
+    }
+    for (uint32_t c = 0; c < sch.n_cols; c++) {
+        uint64_t dc = d * sch.n_cols + c, sc = s * sch.n_cols + c;
+        dst.cell_flags[dc] = src.cell_flags[sc];
+        dst.cell_ts[dc] = src.cell_ts[sc];
+        dst.cell_ldt[dc] = src.cell_ldt[sc];
+        dst.cell_ttl[dc] = src.cell_ttl[sc];
+        dst.val_addr[dc] = src.val_addr[sc];
+        dst.val_len[dc] = src.val_len[sc];
+    }
+}
+
+__global__ void k_garbage_filter(OutParts dp, uint64_t nd, OutParts tp, UnfCols din,
+                                 UnfCols tin, UnfCols dout, const int64_t* tomb_idx,
+                                 const uint64_t* new_base, SchemaParams sch, int cell_level) {
+    uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (g >= nd) return;
+    uint64_t ob = new_base[g];
+    if (!dp.keep[g]) { dp.row_base[g] = ob; dp.row_count[g] = 0; return; }
+    int64_t tg = tomb_idx[g];
+    uint64_t db = dp.row_base[g];
+    uint32_t dn = dp.row_count[g];
+    if (tg < 0) {  // no shadow source for this key: copy through
+        for (uint32_t j = 0; j < dn; j++) gc_copy_unf(dout, ob + j, din, db + j, sch);
+        dp.row_base[g] = ob;
+        return;
+    }
+    uint64_t tb = tp.row_base[tg];
+    uint32_t tn = tp.row_count[tg];
+    // partition deletion: keep only if it supersedes the source's
+    int64_t pdm = tp.pdel_mfda[tg];
+    uint32_t pdl = tp.pdel_ldt[tg];
+    if (!dtp_sup(dp.pdel_mfda[g], dp.pdel_ldt[g], pdm, pdl)) {
+        dp.pdel_mfda[g] = INT64_MIN;
+        dp.pdel_ldt[g] = LDT_NONE_U32;
+    }
+    int64_t act_m = pdm;     // activeDeletionTime (starts at source partition deletion)
+    uint32_t act_l = pdl;
+    // static row vs source static row
+    if (sch.n_static && (dp.st.flags[g] & PF_HAS_ROW)) {
+        // reuse the row machinery on the partition-level static SoA via a
+        // small shim: treat static columns as "cells" of a 1-row UnfCols view
+        // (fields are laid out identically), simplest done inline:
+        int64_t sdm = act_m;
+        uint32_t sdl = act_l;
+        if (tp.st.flags[tg] & PF_ROW_DEL)
+            if (dtp_sup(tp.st.rdel_mfda[tg], tp.st.rdel_ldt[tg], sdm, sdl)) {
+                sdm = tp.st.rdel_mfda[tg];
+                sdl = tp.st.rdel_ldt[tg];
+            }
+        uint8_t f = dp.st.flags[g];
+        bool active_live = sdm == INT64_MIN && sdl == LDT_NONE_U32;
+        bool may = !active_live && !dtp_sup(dp.st.rdel_mfda[g], dp.st.rdel_ldt[g], sdm, sdl);
+        bool any_cell = false;
+        if (cell_level || may) {
+            int64_t lim = cell_level ? sdm : (may ? sdm : INT64_MIN);
+            if ((f & PF_LIVE_TS) && dp.st.live_ts[g] <= lim) {
+                f &= ~PF_LIVE_TS;
+                dp.st.live_ts[g] = NO_TIMESTAMP;
+            }
+            if (!cell_level) {
+                f &= ~PF_ROW_DEL;
+                dp.st.rdel_mfda[g] = INT64_MIN;
+                dp.st.rdel_ldt[g] = LDT_NONE_U32;
+            } else if ((f & PF_ROW_DEL) && dtp_sup(sdm, sdl, dp.st.rdel_mfda[g], dp.st.rdel_ldt[g])) {
+                f &= ~PF_ROW_DEL;
+                dp.st.rdel_mfda[g] = INT64_MIN;
+                dp.st.rdel_ldt[g] = LDT_NONE_U32;
+            }
+            for (uint32_t c = 0; c < sch.n_static; c++) {
+                uint64_t oc = g * sch.n_static + c, tc = (uint64_t)tg * sch.n_static + c;
+                uint8_t cf = dp.st.cell_flags[oc];
+                if (!(cf & CELLF_PRESENT)) continue;
+                if (dp.st.cell_ts[oc] <= lim) { dp.st.cell_flags[oc] = 0; continue; }
+                uint8_t tf = tp.st.cell_flags[tc];
+                if (cell_level && (tf & CELLF_PRESENT) &&
+                    gc_cell_b_wins(dp.st.cell_ts[oc], dp.st.cell_ldt[oc], cf & CELLF_EXPIRING,
+                                   dp.st.val_addr[oc], dp.st.val_len[oc], tp.st.cell_ts[tc],
+                                   tp.st.cell_ldt[tc], tf & CELLF_EXPIRING, tp.st.val_addr[tc],
+                                   tp.st.val_len[tc])) {
+                    dp.st.cell_flags[oc] = 0;
+                    continue;
+                }
+                any_cell = true;
+            }
+            dp.st.flags[g] = ((f & (PF_LIVE_TS | PF_ROW_DEL)) || any_cell) ? (f | PF_HAS_ROW) : 0;
+        }
+    }
+    // ---- two-cursor stream filter (GarbageSkippingUnfilteredRowIterator) ----
+    int64_t tomb_open_m = INT64_MIN, data_open_m = INT64_MIN, open_m = INT64_MIN;
+    uint32_t tomb_open_l = LDT_NONE_U32, data_open_l = LDT_NONE_U32, open_l = LDT_NONE_U32;
+    uint32_t di = 0, ti = 0, oc = 0;
+    auto upd_open = [&](const UnfCols& in, uint64_t o, int64_t* m, uint32_t* l) {
+        uint8_t kk = in.rkind[o];
+        if (bk_is_open(kk)) {
+            *m = bk_is_boundary(kk) ? in.start_mfda[o] : in.rdel_mfda[o];
+            *l = bk_is_boundary(kk) ? in.start_ldt[o] : in.rdel_ldt[o];
+        } else {
+            *m = INT64_MIN;
+            *l = LDT_NONE_U32;
+        }
+    };
+    while (di < dn) {
+        uint64_t d = db + di;
+        int cmp = ti >= tn ? -1 : pos_cmp2(din, d, tin, tb + ti, sch);
+        bool have = false;
+        uint64_t slot = ob + oc;
+        auto process_data_marker = [&]() {
+            upd_open(din, d, &data_open_m, &data_open_l);
+            bool before = open_m == INT64_MIN && open_l == LDT_NONE_U32;
+            bool after = !dtp_sup(data_open_m, data_open_l, act_m, act_l);
+            uint8_t kk = din.rkind[d];
+            if (!before && !after) {
+                gc_copy_unf(dout, slot, din, d, sch);
+                have = true;
+            } else if (!before && after) {
+                gc_copy_unf(dout, slot, din, d, sch);
+                dout.rkind[slot] = kk == BK_EXCL_END_INCL_START ? BK_EXCL_END
+                                   : kk == BK_INCL_END_EXCL_START ? BK_INCL_END : kk;
+                dout.start_mfda[slot] = INT64_MIN;
+                dout.start_ldt[slot] = LDT_NONE_U32;
+                have = true;
+            } else if (before && !after) {
+                gc_copy_unf(dout, slot, din, d, sch);
+                dout.rkind[slot] = kk == BK_EXCL_END_INCL_START ? BK_INCL_START
+                                   : kk == BK_INCL_END_EXCL_START ? BK_EXCL_START : kk;
+                // open-only bound: deletion = open_dt of the original
+                dout.rdel_mfda[slot] = bk_is_boundary(kk) ? din.start_mfda[d] : din.rdel_mfda[d];
+                dout.rdel_ldt[slot] = bk_is_boundary(kk) ? din.start_ldt[d] : din.rdel_ldt[d];
+                dout.start_mfda[slot] = INT64_MIN;
+                dout.start_ldt[slot] = LDT_NONE_U32;
+                have = true;
+            }
+        };
+        if (cmp < 0) {
+            if (din.rkind[d] == BK_CLUSTERING) {
+                gc_copy_unf(dout, slot, din, d, sch);
+                have = gc_row_filter_active(dout, slot, act_m, act_l, sch);
+            } else {
+                process_data_marker();
+            }
+        } else if (cmp == 0) {
+            uint64_t t = tb + ti;
+            if (din.rkind[d] == BK_CLUSTERING) {
+                gc_copy_unf(dout, slot, din, d, sch);
+                have = gc_filter_row_vs(dout, slot, tin, t, act_m, act_l, cell_level != 0, sch);
+            } else {
+                upd_open(tin, t, &tomb_open_m, &tomb_open_l);
+                act_m = pdm;
+                act_l = pdl;
+                if (dtp_sup(tomb_open_m, tomb_open_l, act_m, act_l)) { act_m = tomb_open_m; act_l = tomb_open_l; }
+                process_data_marker();
+            }
+        } else {
+            uint64_t t = tb + ti;
+            if (tin.rkind[t] != BK_CLUSTERING) {
+                upd_open(tin, t, &tomb_open_m, &tomb_open_l);
+                act_m = pdm;
+                act_l = pdl;
+                if (dtp_sup(tomb_open_m, tomb_open_l, act_m, act_l)) { act_m = tomb_open_m; act_l = tomb_open_l; }
+                bool before = open_m == INT64_MIN && open_l == LDT_NONE_U32;
+                bool after = !dtp_sup(data_open_m, data_open_l, act_m, act_l);
+                if (before && !after) {
+                    // reopen the data range at the inverted source close bound
+                    uint8_t tk = tin.rkind[t];
+                    uint8_t ck2 = tk == BK_EXCL_END_INCL_START ? BK_EXCL_END
+                                  : tk == BK_INCL_END_EXCL_START ? BK_INCL_END : tk;
+                    uint8_t inv = ck2 == BK_INCL_END ? BK_EXCL_START
+                                  : ck2 == BK_EXCL_END ? BK_INCL_START : ck2;
+                    gc_copy_unf(dout, slot, tin, t, sch);  // clustering position
+                    dout.flags[slot] = 0;
+                    dout.rkind[slot] = inv;
+                    dout.rdel_mfda[slot] = data_open_m;
+                    dout.rdel_ldt[slot] = data_open_l;
+                    dout.start_mfda[slot] = INT64_MIN;
+                    dout.start_ldt[slot] = LDT_NONE_U32;
+                    dout.live_ts[slot] = NO_TIMESTAMP;
+                    dout.live_ttl[slot] = 0;
+                    dout.live_let[slot] = NO_DELETION_TIME;
+                    for (uint32_t c = 0; c < sch.n_cols; c++)
+                        dout.cell_flags[slot * sch.n_cols + c] = 0;
+                    have = true;
+                }
+            }
+        }
+        if (have && dout.rkind[slot] != BK_CLUSTERING)
+            upd_open(dout, slot, &open_m, &open_l);
+        if (cmp <= 0) di++;
+        if (cmp >= 0) ti++;
+        if (have) oc++;
+    }
+    dp.row_base[g] = ob;
+    dp.row_count[g] = oc;
+}
+
+// PurgeFunction pass over an OutParts arena (the garbage-collect path defers
+// purge until after the filter; mirrors oracle purge_partition). Compacts
+// kept unfiltereds in place and refreshes keep/first/last bookkeeping.
+__global__ void k_purge_parts(OutParts op, UnfCols out, uint64_t n, SchemaParams sch,
+                              PurgeParams2 pp, OutStats* st) {
+    uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (g >= n) return;
+    if (!op.keep[g]) return;
+    int64_t token = op.token[g];
+    auto purge_dt = [&](int64_t m, uint32_t l) {
+        return m != INT64_MIN && should_purge2(pp, token, m, ldt_long(l));
+    };
+    // partition deletion
+    if (!(op.pdel_mfda[g] == INT64_MIN && op.pdel_ldt[g] == LDT_NONE_U32) &&
+        should_purge2(pp, token, op.pdel_mfda[g], ldt_long(op.pdel_ldt[g]))) {
+        op.pdel_mfda[g] = INT64_MIN;
+        op.pdel_ldt[g] = LDT_NONE_U32;
+    }
+    // static row
+    bool static_kept = false;
+    if (sch.n_static && (op.st.flags[g] & PF_HAS_ROW)) {
+        uint8_t f = op.st.flags[g];
+        if (f & PF_LIVE_TS) {
+            int32_t ttl = op.st.live_ttl[g];
+            int64_t let = op.st.live_let[g];
+            bool is_live = ttl == INT32_MAX ? false : (ttl != 0 ? pp.now_sec < let : true);
+            if (!is_live && should_purge2(pp, token, op.st.live_ts[g], let)) {
+                f &= ~PF_LIVE_TS;
+                op.st.live_ts[g] = NO_TIMESTAMP;
+                op.st.live_ttl[g] = 0;
+                op.st.live_let[g] = NO_DELETION_TIME;
+            }
+        }
+        if ((f & PF_ROW_DEL) && purge_dt(op.st.rdel_mfda[g], op.st.rdel_ldt[g])) {
+            f &= ~PF_ROW_DEL;
+            op.st.rdel_mfda[g] = INT64_MIN;
+            op.st.rdel_ldt[g] = LDT_NONE_U32;
+        }
+        bool any_cell = false;
+        for (uint32_t c = 0; c < sch.n_static; c++) {
+            uint64_t oc = g * sch.n_static + c;
+            uint8_t cf = op.st.cell_flags[oc];
+            if (!(cf & CELLF_PRESENT)) continue;
+            int64_t cts = op.st.cell_ts[oc];
+            uint32_t cldt = op.st.cell_ldt[oc];
+            int32_t cttl = op.st.cell_ttl[oc];
+            bool live_cell = cldt == LDT_NONE_U32 || (cttl != 0 && pp.now_sec < ldt_long(cldt));
+            if (!live_cell) {
+                if (should_purge2(pp, token, cts, ldt_long(cldt))) { op.st.cell_flags[oc] = 0; continue; }
+                if (cttl != 0) {
+                    int64_t nldt = ldt_long(cldt) - cttl;
+                    if (should_purge2(pp, token, cts, nldt)) { op.st.cell_flags[oc] = 0; continue; }
+                    op.st.cell_ldt[oc] = ldt_u32(nldt);
+                    op.st.cell_ttl[oc] = 0;
+                    op.st.cell_flags[oc] = CELLF_PRESENT;
+                    op.st.val_len[oc] = 0;
+                }
+            }
+            any_cell = true;
+        }
+        op.st.flags[g] = ((f & (PF_LIVE_TS | PF_ROW_DEL)) || any_cell) ? (f | PF_HAS_ROW) : 0;
+        static_kept = (op.st.flags[g] & PF_HAS_ROW) != 0;
+    }
+    // unfiltereds
+    uint64_t rb = op.row_base[g];
+    uint32_t nrows = op.row_count[g];
+    uint32_t w = 0;
+    for (uint32_t j = 0; j < nrows; j++) {
+        uint64_t o = rb + j;
+        uint8_t kk = out.rkind[o];
+        bool keep_u = false;
+        if (kk != BK_CLUSTERING) {
+            if (bk_is_boundary(kk)) {
+                bool pclose = purge_dt(out.rdel_mfda[o], out.rdel_ldt[o]);
+                bool popen = purge_dt(out.start_mfda[o], out.start_ldt[o]);
+                if (pclose && popen) keep_u = false;
+                else if (pclose) {
+                    out.rkind[o] = kk == BK_EXCL_END_INCL_START ? BK_INCL_START : BK_EXCL_START;
+                    out.rdel_mfda[o] = out.start_mfda[o];
+                    out.rdel_ldt[o] = out.start_ldt[o];
+                    out.start_mfda[o] = INT64_MIN;
+                    out.start_ldt[o] = LDT_NONE_U32;
+                    keep_u = true;
+                } else if (popen) {
+                    out.rkind[o] = kk == BK_EXCL_END_INCL_START ? BK_EXCL_END : BK_INCL_END;
+                    out.start_mfda[o] = INT64_MIN;
+                    out.start_ldt[o] = LDT_NONE_U32;
+                    keep_u = true;
+                } else keep_u = true;
+            } else {
+                keep_u = !purge_dt(out.rdel_mfda[o], out.rdel_ldt[o]);
+            }
+        } else {
+            uint8_t f = out.flags[o];
+            if (f & PF_LIVE_TS) {
+                int32_t ttl = out.live_ttl[o];
+                int64_t let = out.live_let[o];
+                bool is_live = ttl == INT32_MAX ? false : (ttl != 0 ? pp.now_sec < let : true);
+                if (!is_live && should_purge2(pp, token, out.live_ts[o], let)) {
+                    f &= ~PF_LIVE_TS;
+                    out.live_ts[o] = NO_TIMESTAMP;
+                    out.live_ttl[o] = 0;
+                    out.live_let[o] = NO_DELETION_TIME;
+                }
+            }
+            if ((f & PF_ROW_DEL) && purge_dt(out.rdel_mfda[o], out.rdel_ldt[o])) {
+                f &= ~PF_ROW_DEL;
+                out.rdel_mfda[o] = INT64_MIN;
+                out.rdel_ldt[o] = LDT_NONE_U32;
+            }
+            bool any_cell = false;
+            for (uint32_t c = 0; c < sch.n_cols; c++) {
+                uint64_t oc = o * sch.n_cols + c;
+                uint8_t cf = out.cell_flags[oc];
+                if (!(cf & CELLF_PRESENT)) continue;
+                int64_t cts = out.cell_ts[oc];
+                uint32_t cldt = out.cell_ldt[oc];
+                int32_t cttl = out.cell_ttl[oc];
+                bool live_cell = cldt == LDT_NONE_U32 || (cttl != 0 && pp.now_sec < ldt_long(cldt));
+                if (!live_cell) {
+                    if (should_purge2(pp, token, cts, ldt_long(cldt))) { out.cell_flags[oc] = 0; continue; }
+                    if (cttl != 0) {
+                        int64_t nldt = ldt_long(cldt) - cttl;
+                        if (should_purge2(pp, token, cts, nldt)) { out.cell_flags[oc] = 0; continue; }
+                        out.cell_ldt[oc] = ldt_u32(nldt);
+                        out.cell_ttl[oc] = 0;
+                        out.cell_flags[oc] = CELLF_PRESENT;
+                        out.val_len[oc] = 0;
+                    }
+                }
+                any_cell = true;
+            }
+            if (!(f & (PF_LIVE_TS | PF_ROW_DEL)) && !any_cell) f = 0;
+            else if (pp.enforce_strict_liveness && !(f & PF_LIVE_TS) && !(f & PF_ROW_DEL)) f = 0;
+            out.flags[o] = f ? (f | PF_HAS_ROW) : 0;
+            keep_u = (out.flags[o] & PF_HAS_ROW) != 0;
+        }
+        if (keep_u) {
+            if (w != j) gc_copy_unf(out, rb + w, out, o, sch);
+            w++;
+        }
+    }
+    op.row_count[g] = w;
+    bool pdel_live = op.pdel_mfda[g] == INT64_MIN && op.pdel_ldt[g] == LDT_NONE_U32;
+    op.keep[g] = (!pdel_live || w > 0 || static_kept) ? 1 : 0;
+    if (op.keep[g]) {
+        atomicMin(&st->first_group, (unsigned long long)g);
+        atomicMax(&st->last_group, (unsigned long long)g);
+    }
+}
+
+// ---------------------------------------------------------------------------
 // flush path (Memtable -> sstable): host hands unsorted unique-key rows in
 // flat arenas; token+sort on device, then the shared writer kernels
 // ---------------------------------------------------------------------------
@@ -1808,6 +2336,7 @@ __global__ void k_flush_fill(FlushParams fp, const MRec* sorted, OutParts op, Un
     if (i >= fp.n) return;
     uint32_t r = sorted[i].idx;
     op.keypfx[i] = sorted[i].pfx;
+    op.token[i] = (int64_t)(sorted[i].tok ^ 0x8000000000000000ULL);
     op.key_addr[i] = (uint64_t)(fp.keys + fp.key_off[r]);
     op.klen[i] = sorted[i].klen;
     op.pdel_mfda[i] = INT64_MIN;
@@ -1947,6 +2476,7 @@ __global__ void k_gen_count(GenParams2 gp, const MRec* sorted, const uint64_t* i
     }
     prow_count[i] = cnt;
     op.keypfx[i] = sorted[i].pfx;
+    op.token[i] = (int64_t)(sorted[i].tok ^ 0x8000000000000000ULL);
     op.key_addr[i] = (uint64_t)(keys + (uint64_t)sorted[i].idx * gp.key_len);
     op.klen[i] = (uint16_t)gp.key_len;
     op.keep[i] = 1;

@@ -78,6 +78,14 @@ typedef struct gpuc_job {
 
     int32_t device;             /* HIP device ordinal */
 
+    /* nodetool garbagecollect (CompactionIterator.GarbageSkipper): sstables
+       whose tombstones/cells REMOVE shadowed data from the compacted inputs
+       without being written out. cell_level_gc == TombstoneOption.CELL
+       (overwritten cells also removed); 0 == TombstoneOption.ROW. */
+    const char* const* tombstone_source_bases;
+    int32_t n_tomb_sources;
+    int32_t cell_level_gc;
+
     /* 0/1 = one output sstable. N>1: the job is split into N equal Murmur3
        token ranges; each produces its own complete output sstable
        (generation, generation+1, ...) — the UCS shard model / a splitting

@@ -313,6 +313,154 @@ static bool purge_row(Row& r, const Purger& pg, bool enforce_strict_liveness) {
     return !(r.live.empty() && r.del.live() && !any);  // empty row -> null
 }
 
+// ---------------------------------------------------------------------------
+// GarbageSkippingUnfilteredRowIterator (CompactionIterator.java:401-598):
+// remove data shadowed by the tombstone-source partition. Result merged with
+// tombSource == merge of dataSource and tombSource.
+// ---------------------------------------------------------------------------
+static DeletionTime dt_max(const DeletionTime& a, const DeletionTime& b) {
+    return a.supersedes(b) ? a : b;  // Ordering.natural over (mfda, ldt)
+}
+// BTreeRow.filter(ColumnFilter.all, activeDeletion, setActiveDeletionToRow=false)
+static bool row_filter_active(Row& r, const DeletionTime& active) {
+    bool may_shadow = !active.live() && !r.del.supersedes(active);
+    if (may_shadow) {
+        if (!r.live.empty() && r.live.ts <= active.mfda) r.live = LivenessInfo{};
+        r.del = DeletionTime{};  // shadowed row deletion is dropped
+        for (auto& c : r.cells)
+            if (c && c->ts <= active.mfda) c.reset();
+    }
+    return !row_is_empty(r);
+}
+// garbageFilterRow (CompactionIterator.java:544-556): row- or cell-level
+static bool garbage_filter_row(Row& data, const Row& tomb, const DeletionTime& active,
+                               bool cell_level) {
+    if (!cell_level) {
+        return row_filter_active(data, dt_max(tomb.del, active));
+    }
+    // Rows.removeShadowedCells (Rows.java:266-316), simple columns
+    DeletionTime deletion = dt_max(tomb.del, active);
+    if (!data.live.empty() && data.live.ts <= deletion.mfda) data.live = LivenessInfo{};
+    if (deletion.supersedes(data.del)) data.del = DeletionTime{};
+    for (size_t i = 0; i < data.cells.size(); i++) {
+        auto& a = data.cells[i];
+        if (!a) continue;
+        if (a->ts <= deletion.mfda) { a.reset(); continue; }
+        const Cell* b = i < tomb.cells.size() && tomb.cells[i] ? &*tomb.cells[i] : nullptr;
+        if (b && &reconcile_cells(*a, *b) == b) a.reset();  // overwritten by source
+    }
+    return !row_is_empty(data);
+}
+static BoundKind close_bound_kind(BoundKind k) {
+    return k == EXCL_END_INCL_START ? EXCL_END : k == INCL_END_EXCL_START ? INCL_END : k;
+}
+static BoundKind open_bound_kind(BoundKind k) {
+    return k == EXCL_END_INCL_START ? INCL_START : k == INCL_END_EXCL_START ? EXCL_START : k;
+}
+static BoundKind invert_bound_kind(BoundKind k) {
+    switch (k) {
+        case INCL_END: return EXCL_START;
+        case EXCL_END: return INCL_START;
+        case INCL_START: return EXCL_END;
+        case EXCL_START: return INCL_END;
+        default: return k;
+    }
+}
+static DeletionTime update_open_dt(const Unfiltered& u) {
+    return u.marker.open(false) ? u.marker.open_dt() : DeletionTime{};
+}
+
+void garbage_filter(Partition& data, const Partition& tomb, const Header& h, bool cell_level) {
+    DeletionTime partition_dt = tomb.del;
+    DeletionTime active = tomb.del;
+    DeletionTime tomb_open, data_open, open_dt;  // LIVE
+    if (!data.del.supersedes(tomb.del)) data.del = DeletionTime{};
+    if (h.has_static() && !row_is_empty(data.static_row))
+        if (!garbage_filter_row(data.static_row, tomb.static_row, active, cell_level))
+            data.static_row = Row{};
+    std::vector<Unfiltered> out;
+    size_t di = 0, ti = 0;
+    auto emit_bound = [&](BoundKind k, const Clustering& vals, const DeletionTime& dt) {
+        Unfiltered u;
+        u.kind = Unfiltered::MARKER;
+        u.marker.kind = k;
+        u.marker.values = vals;
+        u.marker.end_dt = dt;
+        return u;
+    };
+    while (di < data.items.size()) {
+        Unfiltered* next = nullptr;
+        Unfiltered produced;
+        bool have = false;
+        int cmp = ti >= tomb.items.size()
+                      ? -1
+                      : compare_unfiltered(h, data.items[di], tomb.items[ti]);
+        auto process_data_marker = [&]() -> bool {  // returns have
+            data_open = update_open_dt(data.items[di]);
+            bool before = open_dt.live();
+            bool after = !data_open.supersedes(active);
+            const Marker& m = data.items[di].marker;
+            if (!before && !after) { produced = data.items[di]; return true; }
+            if (!before && after) {
+                produced = emit_bound(close_bound_kind(m.kind), m.values, m.close_dt());
+                return true;
+            }
+            if (before && !after) {
+                produced = emit_bound(open_bound_kind(m.kind), m.values, m.open_dt());
+                return true;
+            }
+            return false;
+        };
+        if (cmp < 0) {
+            if (data.items[di].kind == Unfiltered::ROW) {
+                Row r = data.items[di].row;
+                if (row_filter_active(r, active)) {
+                    produced.kind = Unfiltered::ROW;
+                    produced.row = std::move(r);
+                    have = true;
+                }
+            } else {
+                have = process_data_marker();
+            }
+        } else if (cmp == 0) {
+            if (data.items[di].kind == Unfiltered::ROW) {
+                Row r = data.items[di].row;
+                if (garbage_filter_row(r, tomb.items[ti].row, active, cell_level)) {
+                    produced.kind = Unfiltered::ROW;
+                    produced.row = std::move(r);
+                    have = true;
+                }
+            } else {
+                tomb_open = update_open_dt(tomb.items[ti]);
+                active = dt_max(partition_dt, tomb_open);
+                have = process_data_marker();
+            }
+        } else {
+            if (tomb.items[ti].kind == Unfiltered::MARKER) {
+                tomb_open = update_open_dt(tomb.items[ti]);
+                active = dt_max(partition_dt, tomb_open);
+                bool before = open_dt.live();
+                bool after = !data_open.supersedes(active);
+                // a suppressing deletion ends inside an open data range:
+                // reopen the data range at the (inverted) tomb close bound
+                if (before && !after) {
+                    const Marker& tm = tomb.items[ti].marker;
+                    produced = emit_bound(invert_bound_kind(close_bound_kind(tm.kind)),
+                                          tm.values, data_open);
+                    have = true;
+                }
+            }
+        }
+        if (have && produced.kind == Unfiltered::MARKER)
+            open_dt = update_open_dt(produced);
+        if (cmp <= 0) di++;
+        if (cmp >= 0) ti++;
+        if (have) out.push_back(std::move(produced));
+        (void)next;
+    }
+    data.items = std::move(out);
+}
+
 bool purge_partition(Partition& p, int64_t now_sec, int64_t gc_before, bool never_purge,
                      const std::vector<PurgeRange>& overlaps, bool enforce_strict_liveness) {
     Purger pg{now_sec, gc_before, never_purge, &overlaps, p.token};
@@ -445,6 +593,24 @@ CompactionResult compact(const CompactionJob& job) {
             for (auto& u : v->items) if (u.kind == Unfiltered::ROW) res.rows_in++;
 
         Partition merged = merge_partition_versions(versions, res.out.header);
+        if (!job.tomb_sources.empty()) {
+            // GarbageSkipper.applyToPartition: merge the shadow sources for
+            // this key, then filter the merged data partition against them
+            std::vector<const Partition*> tvers;
+            for (auto& ts2 : job.tomb_sources) {
+                auto it = std::lower_bound(
+                    ts2.parts.begin(), ts2.parts.end(), merged,
+                    [](const Partition& a, const Partition& b) {
+                        return compare_decorated_key(a.token, a.key.data(), a.key.size(),
+                                                     b.token, b.key.data(), b.key.size()) < 0;
+                    });
+                if (it != ts2.parts.end() && it->key == merged.key) tvers.push_back(&*it);
+            }
+            if (!tvers.empty()) {
+                Partition tombm = merge_partition_versions(tvers, res.out.header);
+                garbage_filter(merged, tombm, res.out.header, job.cell_level_gc);
+            }
+        }
         if (purge_partition(merged, job.now_sec, job.gc_before, job.never_purge,
                             job.overlaps, job.enforce_strict_liveness)) {
             res.partitions_out++;

@@ -19,6 +19,11 @@ struct CompactionJob {
     std::vector<PurgeRange> overlaps;
     bool has_shard = false;
     int64_t shard_lo = INT64_MIN, shard_hi = INT64_MAX;  // inclusive token range filter
+    // nodetool garbagecollect: tombstone sources (other sstables) whose
+    // deletions/cells remove shadowed data WITHOUT being written out
+    // (CompactionIterator.GarbageSkipper); cell_level_gc == TombstoneOption.CELL
+    std::vector<SSTable> tomb_sources;
+    bool cell_level_gc = false;
 };
 
 struct CompactionResult {
@@ -38,5 +43,6 @@ bool purge_partition(Partition& p, int64_t now_sec, int64_t gc_before, bool neve
                      const std::vector<PurgeRange>& overlaps, bool enforce_strict_liveness);
 int compare_clustering_prefix(const Header& h, BoundKind ka, const Clustering& a,
                               BoundKind kb, const Clustering& b);
+void garbage_filter(Partition& data, const Partition& tomb, const Header& h, bool cell_level);
 
 }  // namespace oracle

@@ -106,6 +106,8 @@ static GenSpec spec_from_kv(std::map<std::string, std::string>& kv) {
     g.ck_text = (uint32_t)geti("cktext", g.ck_text);
     g.ck_cols = (uint32_t)geti("ckcols", g.ck_cols);
     g.static_pct = (uint32_t)geti("statics", g.static_pct);
+    g.base_ts = geti("ts0", g.base_ts);
+    g.base_ldt = geti("ldt0", g.base_ldt);
     g.n_value_cols = (uint32_t)geti("ncols", g.n_value_cols);
     g.col_missing_pct = (uint32_t)geti("colmiss", g.col_missing_pct);
     g.first_generation = geti("gen0", g.first_generation);
@@ -200,6 +202,18 @@ static int cmd_compact(const std::string& outbase, std::vector<std::string>& inp
     job.now_sec = kv.count("now") ? strtoll(kv["now"].c_str(), nullptr, 10) : 1800000000LL;
     job.gc_before = kv.count("gcbefore") ? strtoll(kv["gcbefore"].c_str(), nullptr, 10) : INT64_MIN;
     job.never_purge = kv.count("nevergc") && kv["nevergc"] == "1";
+    job.cell_level_gc = kv.count("cellgc") && kv["cellgc"] == "1";
+    if (kv.count("tombsrc")) {
+        // comma-separated sstable bases whose tombstones shadow the data
+        std::string s = kv["tombsrc"];
+        size_t p0 = 0;
+        while (p0 < s.size()) {
+            size_t c = s.find(',', p0);
+            if (c == std::string::npos) c = s.size();
+            job.tomb_sources.push_back(read_sstable(s.substr(p0, c - p0), true));
+            p0 = c + 1;
+        }
+    }
     if (kv.count("shard")) {
         auto s = kv["shard"];
         auto c = s.find(':');

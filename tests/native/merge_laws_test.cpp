@@ -195,6 +195,93 @@ int main() {
         CHECK(rows == 2 && markers == 2,
               "range tombstone [15,25]@2500 shadows row ck=20@2000, keeps 10 and 30");
     }
+    // ---- GarbageSkipper laws (CompactionIterator.java:401-598 / GcCompactionTest) ----
+    {
+        // source partition deletion shadows older data; the deletion itself
+        // is NOT copied to the output
+        Partition d = part(7), t = part(7);
+        d.items.push_back(urow(live_row(2000, bytes{'A'})));
+        t.del = DeletionTime{2500, 1600};
+        garbage_filter(d, t, h, false);
+        CHECK(d.items.empty() && d.del.live(),
+              "gc: source partition deletion removes shadowed data, is not copied");
+    }
+    {
+        // data newer than the source deletion survives
+        Partition d = part(7), t = part(7);
+        d.items.push_back(urow(live_row(3000, bytes{'A'})));
+        t.del = DeletionTime{2500, 1600};
+        garbage_filter(d, t, h, false);
+        CHECK(d.items.size() == 1, "gc: data newer than source deletion survives");
+    }
+    {
+        // source row deletion (row-level mode) removes the matching row only
+        Partition d = part(7), t = part(7);
+        d.items.push_back(urow(live_row(2000, bytes{'A'})));
+        t.items.push_back(urow(tomb_row(2500, 1600)));
+        garbage_filter(d, t, h, false);
+        CHECK(d.items.empty(), "gc row-level: source row deletion removes shadowed row");
+    }
+    {
+        // cell-level: a NEWER source cell overwrites (removes) the data cell;
+        // an older one does not
+        Partition d = part(7), t = part(7);
+        d.items.push_back(urow(live_row(2000, bytes{'A'})));
+        t.items.push_back(urow(live_row(2600, bytes{'B'})));
+        garbage_filter(d, t, h, true);
+        // the overwritten cell is removed; the pk liveness (row marker)
+        // survives (Rows.removeShadowedCells keeps undeleted liveness)
+        CHECK(d.items.size() == 1 && !d.items[0].row.cells[0] &&
+                  !d.items[0].row.live.empty(),
+              "gc cell-level: newer source cell removes data cell, keeps liveness");
+        Partition d2 = part(7), t2 = part(7);
+        d2.items.push_back(urow(live_row(2000, bytes{'A'})));
+        t2.items.push_back(urow(live_row(1500, bytes{'B'})));
+        garbage_filter(d2, t2, h, true);
+        CHECK(d2.items.size() == 1, "gc cell-level: older source cell keeps data cell");
+        // row-level mode ignores live source cells entirely
+        Partition d3 = part(7), t3 = part(7);
+        d3.items.push_back(urow(live_row(2000, bytes{'A'})));
+        t3.items.push_back(urow(live_row(2600, bytes{'B'})));
+        garbage_filter(d3, t3, h, false);
+        CHECK(d3.items.size() == 1, "gc row-level: live source cells do not shadow");
+    }
+    {
+        // source range tombstone removes covered data rows; the range itself
+        // is not copied
+        Header hc = simple_header();
+        hc.clustering_types = {CqlType::LONG};
+        auto be8 = [](int64_t v) {
+            bytes b(8);
+            for (int i = 0; i < 8; i++) b[i] = (uint8_t)((uint64_t)v >> (8 * (7 - i)));
+            return b;
+        };
+        auto ckrow = [&](int64_t ck, int64_t ts) {
+            Row r = live_row(ts, bytes{'V'});
+            r.clustering = {ClusterVal{ClusterVal::VALUE, be8(ck)}};
+            return urow(std::move(r));
+        };
+        Partition d = part(7), t = part(7);
+        d.items.push_back(ckrow(10, 2000));
+        d.items.push_back(ckrow(20, 2000));
+        d.items.push_back(ckrow(30, 2000));
+        Unfiltered o, c;
+        o.kind = Unfiltered::MARKER;
+        o.marker.kind = INCL_START;
+        o.marker.values = {ClusterVal{ClusterVal::VALUE, be8(15)}};
+        o.marker.end_dt = DeletionTime{2500, 1600};
+        c.kind = Unfiltered::MARKER;
+        c.marker.kind = INCL_END;
+        c.marker.values = {ClusterVal{ClusterVal::VALUE, be8(25)}};
+        c.marker.end_dt = DeletionTime{2500, 1600};
+        t.items.push_back(std::move(o));
+        t.items.push_back(std::move(c));
+        garbage_filter(d, t, hc, false);
+        int rows = 0, markers = 0;
+        for (auto& u : d.items) (u.kind == Unfiltered::ROW ? rows : markers)++;
+        CHECK(rows == 2 && markers == 0,
+              "gc: source range tombstone removes covered rows, copies nothing");
+    }
     printf(fails ? "FAILED %d\n" : "merge laws OK\n", fails);
     return fails ? 1 : 0;
 }

@@ -273,6 +273,40 @@ def test_sharded_output_parity_wide(ca, oracle_bin, tmp_path):
         _assert_dirs_equal(f"{d}/oa-{300+i}-big", f"{d}/oa-{200+i}-big")
 
 
+@pytest.mark.parametrize("mode", ["row", "cell"])
+def test_garbage_collect_parity(ca, oracle_bin, tmp_path, mode):
+    """nodetool garbagecollect (GarbageSkipper): tombstone SOURCES remove
+    shadowed data without being written; GPU output == oracle, byte for byte.
+    Sources are a later generation of the same key space (newer timestamps +
+    tombstones), so both partition-, row- and cell-level shadowing occur."""
+    d = str(tmp_path)
+    # data: seed A; sources: same key universe, later base_ts, heavy tombstones
+    _oracle_gen(d, seed=121, n=3, rows=2000, vlen=200, overlap=40, tomb=5)
+    os.makedirs(d + "/src")
+    _oracle_gen(d + "/src", seed=121, n=2, rows=1500, vlen=150, overlap=40, tomb=40,
+                pdel=10, ts0=1700000500000000)
+    ins = [f"{d}/oa-{g}-big" for g in (1, 2, 3)]
+    srcs = [f"{d}/src/oa-{g}-big" for g in (1, 2)]
+    cell = 1 if mode == "cell" else 0
+    _oracle_compact(f"{d}/oa-90-big", ins, tombsrc=",".join(srcs), cellgc=cell)
+    ca.compact(ins, f"{d}/oa-91-big", tombstone_sources=srcs, cell_level_gc=bool(cell))
+    _assert_dirs_equal(f"{d}/oa-90-big", f"{d}/oa-91-big")
+
+
+def test_garbage_collect_parity_wide(ca, oracle_bin, tmp_path):
+    """Garbage collect with clustering + range tombstones in the sources."""
+    d = str(tmp_path)
+    _oracle_gen(d, seed=122, n=2, rows=60, crows=50, vlen=200, overlap=40, tomb=5)
+    os.makedirs(d + "/src")
+    _oracle_gen(d + "/src", seed=122, n=2, rows=50, crows=50, vlen=150, overlap=40,
+                tomb=30, rtomb=50, ts0=1700000500000000)
+    ins = [f"{d}/oa-{g}-big" for g in (1, 2)]
+    srcs = [f"{d}/src/oa-{g}-big" for g in (1, 2)]
+    _oracle_compact(f"{d}/oa-90-big", ins, tombsrc=",".join(srcs))
+    ca.compact(ins, f"{d}/oa-91-big", tombstone_sources=srcs)
+    _assert_dirs_equal(f"{d}/oa-90-big", f"{d}/oa-91-big")
+
+
 def test_flush_parity(ca, oracle_bin, tmp_path):
     """gpuc_flush (memtable-flush analog): unsorted unique-key host rows ->
     GPU token sort + writer == oracle flush of the same rows, byte for byte."""
