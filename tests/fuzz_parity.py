@@ -1,0 +1,113 @@
+#!/usr/bin/env python3
+"""Randomized oracle-vs-GPU parity sweep (manual tool, GPU box only).
+
+Draws random generator/compaction configurations across the full supported
+schema space (key lengths, clustering shapes, column counts, tombstone mixes,
+gc/purge settings, shards, garbage-collect sources) and byte-compares every
+output component. Run: python tests/fuzz_parity.py [n_configs] [seed]
+"""
+import os
+import random
+import shutil
+import subprocess
+import sys
+import tempfile
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+sys.path.insert(0, REPO)
+ORACLE = os.path.join(REPO, "oracle", "bin", "oracle_tool")
+COMPONENTS = ["Data.db", "Index.db", "CompressionInfo.db", "Filter.db",
+              "Digest.crc32", "Statistics.db", "Summary.db", "TOC.txt"]
+
+
+def dirs_equal(a, b):
+    for c in COMPONENTS:
+        with open(f"{a}-{c}", "rb") as f1, open(f"{b}-{c}", "rb") as f2:
+            if f1.read() != f2.read():
+                return c
+    return None
+
+
+def main():
+    import cassandra_amd as ca
+    n_cfg = int(sys.argv[1]) if len(sys.argv) > 1 else 30
+    rng = random.Random(int(sys.argv[2]) if len(sys.argv) > 2 else 1234)
+    fails = 0
+    for t in range(n_cfg):
+        d = tempfile.mkdtemp(prefix="fuzz_")
+        wide = rng.random() < 0.6
+        gen = {
+            "seed": rng.randrange(1 << 30),
+            "n": rng.choice([1, 2, 3, 4, 6]),
+            "rows": rng.choice([50, 200, 800]) if wide else rng.choice([500, 2000, 5000]),
+            "vlen": rng.choice([0, 1, 17, 200, 900]),
+            "overlap": rng.choice([0, 10, 40, 100]),
+            "tomb": rng.choice([0, 10, 40]),
+            "vrep": rng.choice([0, 55, 97]),
+            "keylen": rng.choice([8, 8, 12, 24, 120]),
+        }
+        if gen["vlen"] == 0:
+            gen["vlen"] = 1
+        if wide:
+            gen["crows"] = rng.choice([5, 40, 200])
+            gen["rtomb"] = rng.choice([0, 30, 70])
+            shape = rng.choice(["long", "text", "two"])
+            if shape == "text":
+                gen["cktext"] = 1
+            elif shape == "two":
+                gen["ckcols"] = 2
+            if rng.random() < 0.4:
+                gen["statics"] = rng.choice([30, 80])
+        else:
+            gen["pdel"] = rng.choice([0, 5])
+        if rng.random() < 0.4:
+            gen["ncols"] = rng.choice([2, 4, 7])
+            gen["colmiss"] = rng.choice([0, 25, 60])
+        job, okw = {}, {}
+        if rng.random() < 0.4:
+            job["gc_before"] = 2000000000
+            okw["gcbefore"] = 2000000000
+        if rng.random() < 0.2:
+            job["never_purge"] = True
+            okw["nevergc"] = 1
+            okw.setdefault("gcbefore", 2000000000)
+            job.setdefault("gc_before", 2000000000)
+        use_gc_sources = not wide and rng.random() < 0.3 and "gc_before" not in job
+        args = [f"{k}={v}" for k, v in gen.items()]
+        subprocess.run([ORACLE, "gen", d, *args], check=True, capture_output=True)
+        ins = [f"{d}/oa-{g}-big" for g in range(1, gen["n"] + 1)]
+        if use_gc_sources:
+            os.makedirs(d + "/src")
+            sargs = dict(gen)
+            sargs["seed"] = gen["seed"]
+            sargs["tomb"] = 50
+            sargs["n"] = 2
+            sargs["ts0"] = 1700000500000000
+            subprocess.run([ORACLE, "gen", d + "/src",
+                            *[f"{k}={v}" for k, v in sargs.items()]],
+                           check=True, capture_output=True)
+            srcs = [f"{d}/src/oa-{g}-big" for g in (1, 2)]
+            okw["tombsrc"] = ",".join(srcs)
+            cell = rng.random() < 0.5
+            if cell:
+                okw["cellgc"] = 1
+            job["tombstone_sources"] = srcs
+            job["cell_level_gc"] = cell
+        oargs = [f"{k}={v}" for k, v in okw.items()]
+        subprocess.run([ORACLE, "compact", f"{d}/oa-90-big", *ins, *oargs],
+                       check=True, capture_output=True)
+        ca.compact(ins, f"{d}/oa-91-big", **job)
+        bad = dirs_equal(f"{d}/oa-90-big", f"{d}/oa-91-big")
+        if bad:
+            fails += 1
+            print(f"FAIL cfg {t}: component {bad}; gen={gen} job={job}")
+            print(f"  kept at {d}")
+        else:
+            print(f"ok  cfg {t}: gen={gen} gc_src={use_gc_sources}")
+            shutil.rmtree(d, ignore_errors=True)
+    print("FAILED" if fails else "ALL OK", f"({n_cfg} configs)")
+    return 1 if fails else 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())
