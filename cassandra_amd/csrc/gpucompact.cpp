@@ -796,7 +796,7 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
     {
         uint64_t blocks = (n_groups + 255) / 256;
         hipLaunchKernelGGL(k_index_sizes_rows, dim3((uint32_t)blocks), dim3(256), 0, stream,
-                           opb.op, n_groups, d_psize.as<uint64_t>(), d_nblocks.as<uint32_t>(),
+                           opb.op, n_groups, sp, d_psize.as<uint64_t>(), d_nblocks.as<uint32_t>(),
                            d_infsz.as<uint64_t>(), d_isize.as<uint64_t>());
     }
     uint64_t total_idx = exscan_u64(d_isize.as<uint64_t>(), n_groups, stream);

@@ -1621,7 +1621,8 @@ __global__ void k_sizes_rows(OutParts op, UnfCols out, uint64_t n, SerParams2 sp
 
 // index entry sizes — needs the scanned Data offsets (the position vint's
 // width depends on the partition's absolute position)
-__global__ void k_index_sizes_rows(OutParts op, uint64_t n, const uint64_t* data_off,
+__global__ void k_index_sizes_rows(OutParts op, uint64_t n, SerParams2 sp,
+                                   const uint64_t* data_off,
                                    const uint32_t* nblocks, const uint64_t* infos_size,
                                    uint64_t* isize) {
     uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -1629,7 +1630,11 @@ __global__ void k_index_sizes_rows(OutParts op, uint64_t n, const uint64_t* data
     if (!op.keep[g]) { isize[g] = 0; return; }
     uint32_t klen = op.klen[g];
     bool pdel_live = op.pdel_mfda[g] == INT64_MIN && op.pdel_ldt[g] == LDT_NONE_U32;
-    uint64_t header_len = 2 + klen + (pdel_live ? 1 : 12);
+    // header_len INCLUDES the static row (headerLength is taken after
+    // addStaticRow in SortedTablePartitionWriter) — keep in sync with
+    // part_walk's computation
+    uint64_t header_len = 2 + klen + (pdel_live ? 1 : 12) +
+                          (sp.sch.n_static ? static_row_bytes(op, g, sp, nullptr, 0, 0, 0) : 0);
     uint64_t e = 2 + klen + uvint_size(data_off[g]);
     uint32_t nb = nblocks[g];
     if (nb > 1) {

@@ -216,6 +216,10 @@ def test_generator_parity_long_keys(ca, oracle_bin, tmp_path):
     dict(name="static_rows_gc", gen=dict(seed=111, n=3, rows=40, crows=30, vlen=200,
                                          overlap=30, tomb=10, statics=50),
          job=dict(gc_before=2000000000)),
+    dict(name="static_rows_promoted_index", gen=dict(seed=98535134, n=6, rows=50, vlen=900,
+                                                      overlap=100, tomb=40, vrep=97, keylen=120,
+                                                      crows=200, rtomb=70, statics=80),
+         job=dict(gc_before=2000000000, never_purge=True)),
     dict(name="static_rows_kitchen_sink", gen=dict(seed=112, n=3, rows=24, crows=36,
                                                    vlen=150, overlap=30, tomb=12, rtomb=25,
                                                    statics=40, ckcols=2, ncols=3, colmiss=20,
