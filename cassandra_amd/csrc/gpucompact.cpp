@@ -110,25 +110,6 @@ static size_t file_size_of(const std::string& path) {
     fclose(f);
     return (size_t)n;
 }
-static void read_file_into(const std::string& path, uint8_t* dst, size_t n, int nthreads = 4) {
-    if (n == 0) return;
-    std::vector<std::thread> th;
-    size_t per = (n + nthreads - 1) / nthreads;
-    for (int t = 0; t < nthreads; t++) {
-        size_t off = (size_t)t * per;
-        if (off >= n) break;
-        size_t len = std::min(per, n - off);
-        th.emplace_back([&, off, len]() {
-            FILE* f = fopen(path.c_str(), "rb");
-            if (!f) return;
-            fseek(f, (long)off, SEEK_SET);
-            size_t rd = fread(dst + off, 1, len, f);
-            (void)rd;
-            fclose(f);
-        });
-    }
-    for (auto& x : th) x.join();
-}
 static void read_file_range(const std::string& path, uint8_t* dst, uint64_t off, size_t n,
                             int nthreads = 3) {
     if (n == 0) return;
@@ -1238,7 +1219,6 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         HIP_CHECK(hipMemsetAsync(d_error.p, 0, 8, stream));
         std::vector<hipEvent_t> ev_h2d(k);
         std::vector<std::vector<ChunkDesc>> chunks_all(k);  // kept alive past async copies
-        size_t n_chunks_total = 0;
         double ms_read_data = 0;
         std::vector<const uint8_t*> vbase(k, nullptr);  // virtual decompressed origin
         for (int s = 0; s < k; s++) {
@@ -1276,7 +1256,6 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
                 cd.out_len = (uint32_t)std::min<uint64_t>(CHUNK_LEN, ci.data_len - c * (uint64_t)CHUNK_LEN);
                 chunks.push_back(cd);
             }
-            n_chunks_total += chunks.size();
             d_chunks_s[s].alloc(chunks.size() * sizeof(ChunkDesc) + 16);
             if (!chunks.empty())
                 HIP_CHECK(hipMemcpyAsync(d_chunks_s[s].p, chunks.data(), chunks.size() * sizeof(ChunkDesc),

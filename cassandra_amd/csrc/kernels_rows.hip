@@ -1464,7 +1464,6 @@ __device__ uint64_t part_walk(const OutParts& op, const UnfCols& out, uint64_t g
             emit_ck(o);
             emit_uv(body + uvint_size(prev_sz));
             emit_uv(prev_sz);
-            uint8_t f = out.flags[o];
             if (rflags & 0x04) emit_uv((uint64_t)(out.live_ts[o] - sp.hs.min_ts));
             if (rflags & 0x08) {
                 emit_uv(sext32(out.live_ttl[o] - sp.hs.min_ttl));

@@ -93,17 +93,31 @@ def main():
                 okw["cellgc"] = 1
             job["tombstone_sources"] = srcs
             job["cell_level_gc"] = cell
+        use_shards = (not use_gc_sources) and rng.random() < 0.2
         oargs = [f"{k}={v}" for k, v in okw.items()]
-        subprocess.run([ORACLE, "compact", f"{d}/oa-90-big", *ins, *oargs],
-                       check=True, capture_output=True)
-        ca.compact(ins, f"{d}/oa-91-big", **job)
-        bad = dirs_equal(f"{d}/oa-90-big", f"{d}/oa-91-big")
+        if use_shards:
+            from cassandra_amd.sharding import split_token_range
+            S = rng.choice([2, 3])
+            ca.compact(ins, f"{d}/oa-91-big", n_output_shards=S, **job)
+            bad = None
+            for i in range(S):
+                lo, hi = split_token_range(S, i)
+                subprocess.run([ORACLE, "compact", f"{d}/oa-{90 + 10 * i}-big", *ins,
+                                f"shard={lo}:{hi}", *oargs], check=True, capture_output=True)
+                bad = bad or dirs_equal(f"{d}/oa-{90 + 10 * i}-big", f"{d}/oa-{91 + i}-big")
+        else:
+            subprocess.run([ORACLE, "compact", f"{d}/oa-90-big", *ins, *oargs],
+                           check=True, capture_output=True)
+            ca.compact(ins, f"{d}/oa-91-big", **job)
+            bad = dirs_equal(f"{d}/oa-90-big", f"{d}/oa-91-big")
+            if not bad and rng.random() < 0.3:
+                ca.verify(f"{d}/oa-91-big")
         if bad:
             fails += 1
             print(f"FAIL cfg {t}: component {bad}; gen={gen} job={job}")
             print(f"  kept at {d}")
         else:
-            print(f"ok  cfg {t}: gen={gen} gc_src={use_gc_sources}")
+            print(f"ok  cfg {t}: gen={gen} gc_src={use_gc_sources} shards={use_shards}")
             shutil.rmtree(d, ignore_errors=True)
     print("FAILED" if fails else "ALL OK", f"({n_cfg} configs)")
     return 1 if fails else 0

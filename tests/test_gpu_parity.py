@@ -341,6 +341,12 @@ def test_flush_parity(ca, oracle_bin, tmp_path):
     ca.flush(rows, f"{d}/oa-2-big")
     _assert_dirs_equal(f"{d}/oa-1-big", f"{d}/oa-2-big")
     ca.verify(f"{d}/oa-2-big")
+    # duplicate keys must be rejected loudly
+    try:
+        ca.flush([(b"same", 1, b"x"), (b"same", 2, b"y")], f"{d}/oa-9-big")
+        raise AssertionError("duplicate keys accepted")
+    except Exception as e:
+        assert "duplicate" in str(e), e
 
 
 def test_verify_epilogue(ca, oracle_bin, tmp_path):
