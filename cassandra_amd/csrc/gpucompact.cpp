@@ -1984,6 +1984,20 @@ extern "C" int gpuc_verify(const char* input_base, int32_t device, char* error,
             HIP_CHECK(hipMemcpyAsync(d_vckw.p, vckw.data(), vckw.size() * 4,
                                      hipMemcpyHostToDevice, stream));
         sch.ck_w = d_vckw.as<int32_t>();
+        std::vector<int32_t> vsf;
+        for (auto& [snm, sct] : st.static_cols) {
+            (void)snm;
+            if (sct == "org.apache.cassandra.db.marshal.LongType") vsf.push_back(8);
+            else if (sct == "org.apache.cassandra.db.marshal.Int32Type") vsf.push_back(4);
+            else vsf.push_back(-1);
+        }
+        sch.n_static = (uint32_t)vsf.size();
+        DevBuf d_vsf;
+        d_vsf.alloc(vsf.size() * 4 + 8);
+        if (sch.n_static)
+            HIP_CHECK(hipMemcpyAsync(d_vsf.p, vsf.data(), vsf.size() * 4,
+                                     hipMemcpyHostToDevice, stream));
+        sch.static_fixed = d_vsf.as<int32_t>();
         sch.column_index_size = 64 * 1024;
         SrcDesc2 src{};
         src.data = d_data.as<uint8_t>();

@@ -357,7 +357,7 @@ def test_verify_epilogue(ca, oracle_bin, tmp_path):
     os.makedirs(d + "/w"), os.makedirs(d + "/g")
     _oracle_gen(d, seed=95, n=1, rows=2000, vlen=300, overlap=0, tomb=10)
     _oracle_gen(d + "/w", seed=96, n=1, rows=30, crows=60, vlen=200, rtomb=30,
-                cktext=1, keylen=24, ncols=3, overlap=0)
+                cktext=1, keylen=24, ncols=3, overlap=0, statics=50)
     ca.generate(d + "/g", seed=97, n_sstables=1, rows_per_sstable=2000, value_len=256)
     for b in [f"{d}/oa-1-big", f"{d}/w/oa-1-big", f"{d}/g/oa-1-big"]:
         ca.verify(b)
