@@ -40,7 +40,7 @@ def main():
             "seed": rng.randrange(1 << 30),
             "n": rng.choice([1, 2, 3, 4, 6]),
             "rows": rng.choice([50, 200, 800]) if wide else rng.choice([500, 2000, 5000]),
-            "vlen": rng.choice([0, 1, 17, 200, 900]),
+            "vlen": rng.choice([0, 1, 17, 200, 900, 20000, 60000]),
             "overlap": rng.choice([0, 10, 40, 100]),
             "tomb": rng.choice([0, 10, 40]),
             "vrep": rng.choice([0, 55, 97]),
@@ -48,6 +48,10 @@ def main():
         }
         if gen["vlen"] == 0:
             gen["vlen"] = 1
+        if gen["vlen"] >= 20000:
+            gen["rows"] = min(gen["rows"], 200)
+            if wide:
+                gen["rows"] = 20
         if wide:
             gen["crows"] = rng.choice([5, 40, 200])
             gen["rtomb"] = rng.choice([0, 30, 70])
