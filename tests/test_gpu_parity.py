@@ -379,6 +379,23 @@ def test_verify_epilogue(ca, oracle_bin, tmp_path):
             assert "gpuc_verify" in str(e), e
 
 
+def test_compact_rejects_corrupt_input(ca, oracle_bin, tmp_path):
+    """A flipped byte in an input chunk must fail the compaction loudly
+    (CompressedChunkReader CRC semantics) — no silent bad output."""
+    d = str(tmp_path)
+    _oracle_gen(d, seed=131, n=2, rows=1000, vlen=300, overlap=20)
+    with open(f"{d}/oa-1-big-Data.db", "r+b") as f:
+        f.seek(os.path.getsize(f"{d}/oa-1-big-Data.db") // 3)
+        b0 = f.read(1)
+        f.seek(-1, 1)
+        f.write(bytes([b0[0] ^ 0x40]))
+    try:
+        ca.compact([f"{d}/oa-1-big", f"{d}/oa-2-big"], f"{d}/oa-9-big")
+        raise AssertionError("corrupt input accepted")
+    except Exception as e:
+        assert "gpuc_compact" in str(e), e
+
+
 def test_purge_overlap_table_parity(ca, oracle_bin, tmp_path):
     """gcBefore purge gated by the token-interval min-timestamp table."""
     d = str(tmp_path)
