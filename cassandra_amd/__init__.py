@@ -34,6 +34,9 @@ class GpucJob(ctypes.Structure):
         ("has_token_range", ctypes.c_int32),
         ("token_lo", ctypes.c_int64),
         ("token_hi", ctypes.c_int64),
+        ("keep_ranges", ctypes.POINTER(GpucPurgeRange)),
+        ("n_keep_ranges", ctypes.c_int32),
+        ("invert_ranges", ctypes.c_int32),
         ("device", ctypes.c_int32),
         ("tombstone_source_bases", ctypes.POINTER(ctypes.c_char_p)),
         ("n_tomb_sources", ctypes.c_int32),
@@ -216,6 +219,8 @@ def compact(
     n_output_shards=1,
     tombstone_sources=None,
     cell_level_gc=False,
+    keep_ranges=None,
+    invert_ranges=False,
 ):
     """One compaction task (mirrors CompactionTask.runMayThrow's hot loop).
 
@@ -243,6 +248,13 @@ def compact(
     if token_range is not None:
         job.has_token_range = 1
         job.token_lo, job.token_hi = token_range
+    if keep_ranges:
+        kr = (GpucPurgeRange * len(keep_ranges))()
+        for i, (lo, hi) in enumerate(keep_ranges):
+            kr[i].token_lo, kr[i].token_hi = lo, hi
+        job.keep_ranges = kr
+        job.n_keep_ranges = len(keep_ranges)
+        job.invert_ranges = 1 if invert_ranges else 0
     job.device = device
     job.n_output_shards = n_output_shards
     if tombstone_sources:

@@ -1442,6 +1442,22 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         pp.has_shard = job->has_token_range;
         pp.shard_lo = job->token_lo;
         pp.shard_hi = job->token_hi;
+        DevBuf d_kr_lo, d_kr_hi;
+        if (job->n_keep_ranges > 0) {
+            std::vector<int64_t> klo(job->n_keep_ranges), khi(job->n_keep_ranges);
+            for (int i = 0; i < job->n_keep_ranges; i++) {
+                klo[i] = job->keep_ranges[i].token_lo;
+                khi[i] = job->keep_ranges[i].token_hi;
+            }
+            d_kr_lo.alloc(job->n_keep_ranges * 8);
+            d_kr_hi.alloc(job->n_keep_ranges * 8);
+            HIP_CHECK(hipMemcpyAsync(d_kr_lo.p, klo.data(), job->n_keep_ranges * 8, hipMemcpyHostToDevice, stream));
+            HIP_CHECK(hipMemcpyAsync(d_kr_hi.p, khi.data(), job->n_keep_ranges * 8, hipMemcpyHostToDevice, stream));
+            pp.kr_lo = d_kr_lo.as<int64_t>();
+            pp.kr_hi = d_kr_hi.as<int64_t>();
+            pp.n_keep_ranges = job->n_keep_ranges;
+            pp.invert_ranges = job->invert_ranges;
+        }
         if (job->n_overlaps > 0) {
             std::vector<int64_t> lo(job->n_overlaps), hi(job->n_overlaps), ts(job->n_overlaps);
             for (int i = 0; i < job->n_overlaps; i++) {

@@ -443,7 +443,25 @@ struct PurgeParams2 {
     int32_t n_overlaps;
     int32_t has_shard;
     int64_t shard_lo, shard_hi;
+    // multi-range keep filter (anticompaction: CompactionManager.antiCompactGroup
+    // splits data by repaired ranges — one pass keeps tokens IN the ranges,
+    // the inverted pass keeps the complement)
+    const int64_t* kr_lo;
+    const int64_t* kr_hi;
+    int32_t n_keep_ranges;
+    int32_t invert_ranges;
 };
+
+__device__ inline bool token_kept(const PurgeParams2& pp, int64_t token) {
+    if (pp.has_shard && (token < pp.shard_lo || token > pp.shard_hi)) return false;
+    if (pp.n_keep_ranges) {
+        bool in = false;
+        for (int i = 0; i < pp.n_keep_ranges && !in; i++)
+            in = token >= pp.kr_lo[i] && token <= pp.kr_hi[i];
+        if (in == (pp.invert_ranges != 0)) return false;
+    }
+    return true;
+}
 
 __device__ inline bool purge_eval2(const PurgeParams2& pp, int64_t token, int64_t ts) {
     int64_t min_ts = INT64_MAX;
@@ -498,7 +516,7 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
     op.klen[g] = r0.klen;
     op.row_base[g] = out_base[g];
     op.row_count[g] = 0;
-    if (pp.has_shard && (token < pp.shard_lo || token > pp.shard_hi)) { op.keep[g] = 0; return; }
+    if (!token_kept(pp, token)) { op.keep[g] = 0; return; }
 
     // member streams
     uint64_t mb[MA];

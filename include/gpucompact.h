@@ -76,6 +76,14 @@ typedef struct gpuc_job {
     int64_t token_lo;           /* inclusive */
     int64_t token_hi;           /* inclusive */
 
+    /* anticompaction split (CompactionManager.antiCompactGroup / RepairFinishedCompactionTask):
+       keep only partitions whose token falls in one of these inclusive ranges
+       (invert_ranges=1 keeps the complement). One anticompaction = two calls:
+       ranges -> the repaired output, ranges+invert -> the unrepaired one. */
+    const gpuc_purge_range* keep_ranges;  /* min_timestamp field unused */
+    int32_t n_keep_ranges;
+    int32_t invert_ranges;
+
     int32_t device;             /* HIP device ordinal */
 
     /* nodetool garbagecollect (CompactionIterator.GarbageSkipper): sstables

@@ -561,10 +561,18 @@ CompactionResult compact(const CompactionJob& job) {
     while (true) {
         int min_src = -1;
         for (size_t i = 0; i < k; i++) {
-            // shard filter: skip partitions outside our token range
-            while (pos[i] < job.inputs[i].parts.size() && job.has_shard) {
+            // shard / anticompaction-range filter: skip non-kept partitions
+            while (pos[i] < job.inputs[i].parts.size() &&
+                   (job.has_shard || !job.keep_ranges.empty())) {
                 int64_t tok = job.inputs[i].parts[pos[i]].token;
-                if (tok < job.shard_lo || tok > job.shard_hi) pos[i]++;
+                bool drop = job.has_shard && (tok < job.shard_lo || tok > job.shard_hi);
+                if (!drop && !job.keep_ranges.empty()) {
+                    bool in = false;
+                    for (auto& r : job.keep_ranges)
+                        if (tok >= r.tok_lo && tok <= r.tok_hi) { in = true; break; }
+                    drop = (in == job.invert_ranges);
+                }
+                if (drop) pos[i]++;
                 else break;
             }
             if (pos[i] >= job.inputs[i].parts.size()) continue;

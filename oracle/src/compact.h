@@ -24,6 +24,9 @@ struct CompactionJob {
     // (CompactionIterator.GarbageSkipper); cell_level_gc == TombstoneOption.CELL
     std::vector<SSTable> tomb_sources;
     bool cell_level_gc = false;
+    // anticompaction split: keep tokens inside (or, inverted, outside) ranges
+    std::vector<PurgeRange> keep_ranges;  // min_ts unused
+    bool invert_ranges = false;
 };
 
 struct CompactionResult {

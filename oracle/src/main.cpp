@@ -214,6 +214,23 @@ static int cmd_compact(const std::string& outbase, std::vector<std::string>& inp
             p0 = c + 1;
         }
     }
+    if (kv.count("ranges")) {
+        // "lo:hi,lo:hi" inclusive keep-ranges; invertranges=1 keeps the complement
+        std::string s = kv["ranges"];
+        size_t p0 = 0;
+        while (p0 < s.size()) {
+            size_t cm = s.find(',', p0);
+            if (cm == std::string::npos) cm = s.size();
+            std::string one = s.substr(p0, cm - p0);
+            size_t c2 = one.find(':');
+            PurgeRange r{};
+            r.tok_lo = strtoll(one.substr(0, c2).c_str(), nullptr, 10);
+            r.tok_hi = strtoll(one.substr(c2 + 1).c_str(), nullptr, 10);
+            job.keep_ranges.push_back(r);
+            p0 = cm + 1;
+        }
+        job.invert_ranges = kv.count("invertranges") && kv["invertranges"] == "1";
+    }
     if (kv.count("shard")) {
         auto s = kv["shard"];
         auto c = s.find(':');

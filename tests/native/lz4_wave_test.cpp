@@ -154,16 +154,23 @@ int main() {
     printf(fails ? "FAILED %d of %d\n" : "lz4 wave OK (%d cases)\n", fails ? fails : cases, cases);
     if (fails) return 1;
 
-    // ---- A/B throughput: LDS-staged vs global-chunk, value-like content ----
-    {
+    // ---- A/B throughput: LDS-staged vs global-chunk, content sweep ----
+    for (int content = 0; content < 4; content++) {
         const uint32_t NC = 16384;  // 256 MiB of chunks
         std::vector<uint8_t> big((uint64_t)NC * CHUNK_LEN);
         for (uint64_t i = 0; i < big.size(); i += 8) {
-            // ~55% repeat words (bench-like compressibility)
             uint64_t r = sm(i);
-            uint64_t w = (r % 100 < 55 && i) ? *(uint64_t*)&big[i - 8] : sm(r);
+            uint64_t w;
+            switch (content) {
+                case 0: w = (r % 100 < 55 && i) ? *(uint64_t*)&big[i - 8] : sm(r); break;
+                case 1: w = sm(r); break;        // incompressible
+                case 2: w = 0; break;            // all zeros (1 long match)
+                default: w = (r % 100 < 97 && i) ? *(uint64_t*)&big[i - 8] : sm(r); break;
+            }
             memcpy(&big[i], &w, 8);
         }
+        printf("content=%d (%s)\n", content,
+               content == 0 ? "vrep55" : content == 1 ? "random" : content == 2 ? "zeros" : "vrep97");
         uint8_t *d_big, *d_slots2;
         uint32_t *d_cs, *d_crc2;
         hipMalloc(&d_big, big.size());
@@ -194,6 +201,7 @@ int main() {
             printf("compress %s: %.1f ms for 3x256MiB = %.2f GB/s in\n",
                    variant == 0 ? "LDS-staged " : "global-src", ms, 3.0 * big.size() / (ms / 1e3) / 1e9);
         }
+        hipFree(d_big); hipFree(d_slots2); hipFree(d_cs); hipFree(d_crc2);
     }
     return 0;
 }

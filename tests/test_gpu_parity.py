@@ -379,6 +379,32 @@ def test_verify_epilogue(ca, oracle_bin, tmp_path):
             assert "gpuc_verify" in str(e), e
 
 
+def test_anticompaction_split_parity(ca, oracle_bin, tmp_path):
+    """Anticompaction (antiCompactGroup): one pass keeps tokens inside the
+    repaired ranges, the inverted pass keeps the complement; together they
+    partition the data. Both outputs byte-identical to the oracle."""
+    d = str(tmp_path)
+    _oracle_gen(d, seed=141, n=3, rows=2000, vlen=250, overlap=30, tomb=10)
+    ins = [f"{d}/oa-{g}-big" for g in (1, 2, 3)]
+    ranges = [(-2**62, 0), (2**61, 2**62)]
+    rs = ",".join(f"{lo}:{hi}" for lo, hi in ranges)
+    _oracle_compact(f"{d}/oa-90-big", ins, ranges=rs)
+    _oracle_compact(f"{d}/oa-92-big", ins, ranges=rs, invertranges=1)
+    ca.compact(ins, f"{d}/oa-91-big", keep_ranges=ranges)
+    ca.compact(ins, f"{d}/oa-93-big", keep_ranges=ranges, invert_ranges=True)
+    _assert_dirs_equal(f"{d}/oa-90-big", f"{d}/oa-91-big")
+    _assert_dirs_equal(f"{d}/oa-92-big", f"{d}/oa-93-big")
+    # the two halves partition the rows
+    import json
+    a = json.loads(subprocess.run([ORACLE, "compact", f"{d}/oa-94-big", *ins],
+                                  capture_output=True, text=True, check=True).stdout.splitlines()[-1])
+    pr = 0
+    for base in (f"{d}/oa-90-big", f"{d}/oa-92-big"):
+        out = subprocess.run([ORACLE, "dump", base], capture_output=True, text=True, check=True)
+        pr += int(out.stdout.splitlines()[1].split("partitions=")[1].split()[0])
+    assert pr == a["partitions_out"], (pr, a["partitions_out"])
+
+
 def test_compact_rejects_corrupt_input(ca, oracle_bin, tmp_path):
     """A flipped byte in an input chunk must fail the compaction loudly
     (CompressedChunkReader CRC semantics) — no silent bad output."""
