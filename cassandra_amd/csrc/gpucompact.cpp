@@ -291,6 +291,8 @@ static void parse_index_positions(const bytes& ib, uint64_t data_len,
             uint64_t pos = r.uvint();
             uint64_t promoted = r.uvint();
             r.skip(promoted);
+            if (pos >= data_len || (!positions.empty() && pos <= positions.back()))
+                throw std::runtime_error("Index.db positions not increasing/in range");
             positions.push_back(pos);
         }
         positions.push_back(data_len);
@@ -1275,6 +1277,18 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         TR("ingest issued");
         HIP_CHECK(hipEventRecord(e1, stream));  // e0..e1: H2D+decompress pipeline (overlapped)
         HIP_CHECK(hipEventRecord(e2, stream));
+
+        // corrupt inputs must fail cleanly, not fault the GPU: the parse
+        // walk trusts decompressed bytes, so the chunk-CRC verdict is checked
+        // BEFORE any parse kernel touches them
+        {
+            unsigned long long err = 0;
+            HIP_CHECK(hipStreamSynchronize(stream));
+            HIP_CHECK(hipMemcpy(&err, d_error.p, 8, hipMemcpyDeviceToHost));
+            if (err)
+                throw std::runtime_error("input chunk decompress/CRC failed, code " +
+                                         std::to_string(err));
+        }
 
         // ---- parse (pass A: count + partition meta; pass B: row decode) ----
         uint64_t total_parts = 0;
