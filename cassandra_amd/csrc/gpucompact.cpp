@@ -1990,6 +1990,13 @@ extern "C" int gpuc_verify(const char* input_base, int32_t device, char* error,
             hipLaunchKernelGGL(k_lz4_decompress_wave, dim3((uint32_t)chunks.size()), dim3(WAVE), 0,
                                stream, d_chunks.as<ChunkDesc>(), (uint32_t)chunks.size(), 1,
                                d_error.as<unsigned long long>(), (const uint32_t*)g_crc256);
+        {
+            unsigned long long e = 0;
+            HIP_CHECK(hipStreamSynchronize(stream));
+            HIP_CHECK(hipMemcpy(&e, d_error.p, 8, hipMemcpyDeviceToHost));
+            if (e == 1) throw std::runtime_error("chunk CRC mismatch");
+            if (e) throw std::runtime_error("chunk decompress failed, code " + std::to_string(e));
+        }
         SchemaParams sch{};
         std::vector<int32_t> cfh;
         for (auto& [nm, ct] : st.regular_cols) {
