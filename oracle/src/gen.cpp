@@ -38,11 +38,6 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
     });
 
     int64_t min_ts = INT64_MAX, min_ldt_l = INT64_MAX;
-    auto be8 = [](int64_t v) {
-        bytes b(8);
-        for (int i = 0; i < 8; i++) b[i] = (uint8_t)((uint64_t)v >> (8 * (7 - i)));
-        return b;
-    };
     for (const Ent& e : ents) {
         if (g.clustering_rows > 0) {
             // ---- wide partition: clustering rows + optional range tombstone ----
