@@ -142,6 +142,10 @@ def load_library(path=None):
     lib.gpuc_generate.restype = ctypes.c_int
     lib.gpuc_verify.argtypes = [ctypes.c_char_p, ctypes.c_int32, ctypes.c_char_p, ctypes.c_size_t]
     lib.gpuc_verify.restype = ctypes.c_int
+    lib.gpuc_scrub.argtypes = [ctypes.c_char_p, ctypes.c_char_p, ctypes.c_int32,
+                               ctypes.POINTER(ctypes.c_uint64), ctypes.POINTER(ctypes.c_uint64),
+                               ctypes.c_char_p, ctypes.c_size_t]
+    lib.gpuc_scrub.restype = ctypes.c_int
     lib.gpuc_flush.argtypes = [ctypes.POINTER(GpucFlushRows), ctypes.c_char_p, ctypes.c_int32,
                                ctypes.c_char_p, ctypes.c_size_t]
     lib.gpuc_flush.restype = ctypes.c_int
@@ -186,6 +190,20 @@ def flush(rows, output_base, device=0):
     rc = lib.gpuc_flush(ctypes.byref(fr), output_base.encode(), device, err, 256)
     if rc != 0:
         raise GpuCompactError(f"gpuc_flush rc={rc}: {err.value.decode(errors='replace')}")
+
+
+def scrub(input_base, output_base, device=0):
+    """sstablescrub equivalent: salvage partitions untouched by corrupt
+    chunks into a clean sstable. Returns (kept, dropped)."""
+    lib = load_library()
+    kept = ctypes.c_uint64()
+    dropped = ctypes.c_uint64()
+    err = ctypes.create_string_buffer(256)
+    rc = lib.gpuc_scrub(input_base.encode(), output_base.encode(), device,
+                        ctypes.byref(kept), ctypes.byref(dropped), err, 256)
+    if rc != 0:
+        raise GpuCompactError(f"gpuc_scrub rc={rc}: {err.value.decode(errors='replace')}")
+    return kept.value, dropped.value
 
 
 def verify(input_base, device=0):

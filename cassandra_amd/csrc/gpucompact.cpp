@@ -2099,6 +2099,230 @@ extern "C" int gpuc_verify(const char* input_base, int32_t device, char* error,
     }
 }
 
+// Scrub (SortedTableScrubber semantics, this implementation's recovery
+// granularity — mirrored by oracle::scrub_sstable): salvage partitions whose
+// byte range touches only CRC/decode-clean chunks, rewrite them through the
+// standard writer; report kept/dropped counts.
+extern "C" int gpuc_scrub(const char* input_base, const char* output_base, int32_t device,
+                          uint64_t* kept, uint64_t* dropped, char* error, size_t error_len) {
+    try {
+        if (gpuc_device_count() <= 0) { set_err(error, error_len, "no HIP device"); return GPUC_ERR_NO_GPU; }
+        HIP_CHECK(hipSetDevice(device));
+        hipStream_t stream;
+        HIP_CHECK(hipStreamCreate(&stream));
+        ensure_crc_tables(stream);
+        std::string base = input_base;
+        bytes index_data = read_file(base + "-Index.db");
+        HCompressionInfo ci = parse_compression_info(read_file(base + "-CompressionInfo.db"));
+        HStatistics st = parse_statistics(read_file(base + "-Statistics.db"));
+        size_t comp_sz = file_size_of(base + "-Data.db");
+        bytes comp = read_file(base + "-Data.db");
+        std::vector<uint64_t> positions;
+        std::string perr;
+        parse_index_positions(index_data, ci.data_len, positions, nullptr, perr);
+        if (!perr.empty()) throw std::runtime_error("Index.db: " + perr);
+        uint64_t n_parts = positions.size() - 1;
+        size_t n_chunks = ci.offsets.size();
+
+        // decompress with per-chunk bad flags
+        DevBuf d_comp, d_data, d_chunksb, d_error, d_bad;
+        d_comp.alloc(comp.size());
+        HIP_CHECK(hipMemcpyAsync(d_comp.p, comp.data(), comp.size(), hipMemcpyHostToDevice, stream));
+        d_data.alloc(ci.data_len + 16);
+        d_bad.alloc(n_chunks + 8);
+        HIP_CHECK(hipMemsetAsync(d_bad.p, 0, n_chunks + 8, stream));
+        d_error.alloc(8);
+        HIP_CHECK(hipMemsetAsync(d_error.p, 0, 8, stream));
+        std::vector<ChunkDesc> chunks;
+        for (size_t c = 0; c < n_chunks; c++) {
+            uint64_t off = ci.offsets[c];
+            uint64_t end = c + 1 < n_chunks ? ci.offsets[c + 1] : comp_sz;
+            ChunkDesc cd;
+            cd.comp = d_comp.as<uint8_t>() + off;
+            cd.out = d_data.as<uint8_t>() + c * (uint64_t)CHUNK_LEN;
+            cd.comp_len = (uint32_t)(end - off - 4);
+            cd.out_len = (uint32_t)std::min<uint64_t>(CHUNK_LEN, ci.data_len - c * (uint64_t)CHUNK_LEN);
+            chunks.push_back(cd);
+        }
+        d_chunksb.alloc(chunks.size() * sizeof(ChunkDesc) + 16);
+        HIP_CHECK(hipMemcpyAsync(d_chunksb.p, chunks.data(), chunks.size() * sizeof(ChunkDesc),
+                                 hipMemcpyHostToDevice, stream));
+        if (!chunks.empty())
+            hipLaunchKernelGGL(k_lz4_decompress_wave, dim3((uint32_t)chunks.size()), dim3(WAVE), 0,
+                               stream, d_chunksb.as<ChunkDesc>(), (uint32_t)chunks.size(), 1,
+                               d_error.as<unsigned long long>(), (const uint32_t*)g_crc256,
+                               d_bad.as<uint8_t>());
+        std::vector<uint8_t> bad(n_chunks);
+        HIP_CHECK(hipStreamSynchronize(stream));
+        if (n_chunks)
+            HIP_CHECK(hipMemcpy(bad.data(), d_bad.p, n_chunks, hipMemcpyDeviceToHost));
+
+        // kept partitions: all overlapped chunks clean
+        std::vector<uint64_t> kpos, kend;
+        for (uint64_t i = 0; i < n_parts; i++) {
+            uint64_t c0 = positions[i] / CHUNK_LEN;
+            uint64_t c1 = (positions[i + 1] + CHUNK_LEN - 1) / CHUNK_LEN;
+            bool ok = true;
+            for (uint64_t c = c0; c < c1 && c < n_chunks; c++)
+                if (bad[c]) ok = false;
+            if (ok) {
+                kpos.push_back(positions[i]);
+                kend.push_back(positions[i + 1]);
+            }
+        }
+        uint64_t n_kept = kpos.size();
+        if (kept) *kept = n_kept;
+        if (dropped) *dropped = n_parts - n_kept;
+
+        // schema (same resolution as compaction)
+        SchemaParams sch{};
+        std::vector<int32_t> cfh, vckw, vsf;
+        for (auto& [nm, ct] : st.regular_cols) { (void)nm; cfh.push_back(ck_type_width(ct)); }
+        for (auto& ct : st.clustering_types) vckw.push_back(ck_type_width(ct));
+        for (auto& [nm, ct] : st.static_cols) { (void)nm; vsf.push_back(ck_type_width(ct)); }
+        sch.n_cols = (uint32_t)cfh.size();
+        sch.n_ck = (uint32_t)vckw.size();
+        sch.n_static = (uint32_t)vsf.size();
+        DevBuf d_cf, d_ckw, d_sf;
+        d_cf.alloc(cfh.size() * 4 + 8);
+        if (!cfh.empty()) HIP_CHECK(hipMemcpyAsync(d_cf.p, cfh.data(), cfh.size() * 4, hipMemcpyHostToDevice, stream));
+        d_ckw.alloc(vckw.size() * 4 + 8);
+        if (!vckw.empty()) HIP_CHECK(hipMemcpyAsync(d_ckw.p, vckw.data(), vckw.size() * 4, hipMemcpyHostToDevice, stream));
+        d_sf.alloc(vsf.size() * 4 + 8);
+        if (!vsf.empty()) HIP_CHECK(hipMemcpyAsync(d_sf.p, vsf.data(), vsf.size() * 4, hipMemcpyHostToDevice, stream));
+        sch.col_fixed = d_cf.as<int32_t>();
+        sch.ck_w = d_ckw.as<int32_t>();
+        sch.static_fixed = d_sf.as<int32_t>();
+        sch.column_index_size = 64 * 1024;
+
+        // parse kept partitions (explicit ends: gaps where partitions dropped)
+        DevBuf d_kpos, d_kend, d_recs, d_rows_in;
+        d_kpos.alloc(n_kept * 8 + 8);
+        d_kend.alloc(n_kept * 8 + 8);
+        if (n_kept) {
+            HIP_CHECK(hipMemcpyAsync(d_kpos.p, kpos.data(), n_kept * 8, hipMemcpyHostToDevice, stream));
+            HIP_CHECK(hipMemcpyAsync(d_kend.p, kend.data(), n_kept * 8, hipMemcpyHostToDevice, stream));
+        }
+        SrcDesc2 src{};
+        src.data = d_data.as<uint8_t>();
+        src.part_pos = d_kpos.as<uint64_t>();
+        src.part_end = d_kend.as<uint64_t>();
+        src.n_parts = (uint32_t)n_kept;
+        src.min_ts = st.hdr_min_ts;
+        src.min_ldt = st.hdr_min_ldt;
+        src.min_ttl = st.hdr_min_ttl;
+        src.rec_base = 0;
+        DevBuf d_src;
+        d_src.alloc(sizeof(src));
+        HIP_CHECK(hipMemcpyAsync(d_src.p, &src, sizeof(src), hipMemcpyHostToDevice, stream));
+        d_recs.alloc(n_kept * sizeof(MRec) + 32);
+        d_rows_in.alloc(8);
+        HIP_CHECK(hipMemsetAsync(d_rows_in.p, 0, 8, stream));
+        DevBuf p_pdm, p_pdl, p_rcnt, p_rbase, p_kaddr;
+        ParsedCols pc{};
+        p_pdm.alloc(n_kept * 8 + 8); pc.pdel_mfda = p_pdm.as<int64_t>();
+        p_pdl.alloc(n_kept * 4 + 8); pc.pdel_ldt = p_pdl.as<uint32_t>();
+        p_rcnt.alloc(n_kept * 4 + 8); pc.row_count = p_rcnt.as<uint32_t>();
+        p_rbase.alloc(n_kept * 8 + 8); pc.row_base = p_rbase.as<uint64_t>();
+        p_kaddr.alloc(n_kept * 8 + 8); pc.key_addr = p_kaddr.as<uint64_t>();
+        StaticColsBuf p_static;
+        p_static.alloc(sch.n_static ? n_kept : 1, sch.n_static);
+        pc.st = p_static.st;
+        uint32_t blocks = (uint32_t)((n_kept + 255) / 256);
+        if (n_kept) {
+            hipLaunchKernelGGL(k_parse_count, dim3(blocks), dim3(256), 0, stream,
+                               d_src.as<SrcDesc2>(), 1u, (uint32_t)n_kept, d_recs.as<MRec>(), pc,
+                               sch, d_error.as<unsigned long long>());
+            hipLaunchKernelGGL(k_widen_u32, dim3(blocks), dim3(256), 0, stream, pc.row_count,
+                               pc.row_base, n_kept);
+        }
+        uint64_t total_rows = n_kept ? exscan_u64(pc.row_base, n_kept, stream) : 0;
+        UnfColsBuf in_rows;
+        in_rows.alloc(total_rows, sch.n_cols, sch.n_ck);
+        if (n_kept)
+            hipLaunchKernelGGL(k_parse_rows, dim3(blocks), dim3(256), 0, stream,
+                               d_src.as<SrcDesc2>(), 1u, (uint32_t)n_kept, pc, in_rows.uc, sch,
+                               d_error.as<unsigned long long>(), d_rows_in.as<unsigned long long>());
+        {
+            unsigned long long e = 0;
+            HIP_CHECK(hipStreamSynchronize(stream));
+            HIP_CHECK(hipMemcpy(&e, d_error.p, 8, hipMemcpyDeviceToHost));
+            if (e) throw std::runtime_error("scrub parse failed, code " + std::to_string(e));
+        }
+        // single source: records are already in DecoratedKey order
+        KeyLut lut{};
+        lut.base[0] = d_data.as<uint8_t>();
+        lut.pos[0] = d_kpos.as<uint64_t>();
+        lut.enabled = 1;
+        DevBuf d_head, d_gstart, d_ng;
+        d_head.alloc(n_kept * 8 + 8);
+        d_gstart.alloc(n_kept * 8 + 8);
+        d_ng.alloc(8);
+        HIP_CHECK(hipMemsetAsync(d_ng.p, 0, 8, stream));
+        if (n_kept) {
+            hipLaunchKernelGGL(k_group_heads, dim3(blocks), dim3(256), 0, stream, d_recs.as<MRec>(),
+                               n_kept, d_head.as<uint64_t>(), lut);
+            exscan_u64(d_head.as<uint64_t>(), n_kept, stream);
+            hipLaunchKernelGGL(k_group_starts2, dim3(blocks), dim3(256), 0, stream,
+                               d_recs.as<MRec>(), n_kept, d_head.as<uint64_t>(),
+                               d_gstart.as<uint64_t>(), d_ng.as<uint64_t>(), lut);
+        }
+        uint64_t n_groups = 0;
+        HIP_CHECK(hipStreamSynchronize(stream));
+        HIP_CHECK(hipMemcpy(&n_groups, d_ng.p, 8, hipMemcpyDeviceToHost));
+        DevBuf d_srcbases, d_grows, d_stats, d_tomb;
+        uint32_t zero = 0;
+        d_srcbases.alloc(8);
+        HIP_CHECK(hipMemcpyAsync(d_srcbases.p, &zero, 4, hipMemcpyHostToDevice, stream));
+        d_grows.alloc(n_groups * 8 + 8);
+        if (n_groups) {
+            uint32_t gblocks = (uint32_t)((n_groups + 255) / 256);
+            hipLaunchKernelGGL(k_group_row_sums, dim3(gblocks), dim3(256), 0, stream,
+                               d_recs.as<MRec>(), d_gstart.as<uint64_t>(), n_groups, n_kept,
+                               d_srcbases.as<uint32_t>(), pc, d_grows.as<uint64_t>());
+        }
+        uint64_t out_total = n_groups ? exscan_u64(d_grows.as<uint64_t>(), n_groups, stream) : 0;
+        OutPartsBuf opb;
+        opb.alloc(n_groups ? n_groups : 1, sch.n_static);
+        UnfColsBuf out_rows;
+        out_rows.alloc(out_total, sch.n_cols, sch.n_ck);
+        d_stats.alloc(sizeof(OutStats));
+        init_outstats(d_stats, stream);
+        uint32_t tomb_cap = (uint32_t)std::min<uint64_t>(out_total * 2 + n_groups + 1024, 400000000ull);
+        d_tomb.alloc((uint64_t)tomb_cap * 4);
+        PurgeParams2 pp{};
+        pp.gc_before = INT64_MIN;
+        pp.never_purge = 1;
+        if (n_groups) {
+            uint32_t gblocks = (uint32_t)((n_groups + 255) / 256);
+            hipLaunchKernelGGL(k_reconcile_rows<8>, dim3(gblocks), dim3(256), 0, stream,
+                               d_recs.as<MRec>(), d_gstart.as<uint64_t>(), n_groups, n_kept,
+                               d_srcbases.as<uint32_t>(), pc, in_rows.uc, opb.op, out_rows.uc,
+                               d_grows.as<uint64_t>(), sch, pp, d_stats.as<OutStats>(),
+                               d_error.as<unsigned long long>());
+        }
+        HIP_CHECK(hipStreamSynchronize(stream));
+        {
+            unsigned long long e = 0;
+            HIP_CHECK(hipMemcpy(&e, d_error.p, 8, hipMemcpyDeviceToHost));
+            if (e) throw std::runtime_error("scrub reconcile failed, code " + std::to_string(e));
+        }
+        SerParams2 sp2{};
+        sp2.sch = sch;
+        sp2.hs.min_ts = st.min_timestamp == NO_TIMESTAMP ? TIMESTAMP_EPOCH : st.min_timestamp;
+        sp2.hs.min_ldt = st.min_ldt == NO_DELETION_TIME ? DELETION_TIME_EPOCH : st.min_ldt;
+        sp2.hs.min_ttl = st.min_ttl == INT32_MAX ? 0 : st.min_ttl;
+        write_sstable_device(opb, out_rows, n_groups, sp2, d_stats, d_tomb, tomb_cap, output_base,
+                             st.key_type, st.clustering_types, st.regular_cols, st.static_cols,
+                             stream, 0);
+        HIP_CHECK(hipStreamDestroy(stream));
+        return GPUC_OK;
+    } catch (const std::exception& e) {
+        set_err(error, error_len, e.what());
+        return GPUC_ERR_INTERNAL;
+    }
+}
+
 // Memtable flush: unsorted unique-key rows from the host ->
 // token sort on device -> the shared serialize/compress/index/bloom kernels
 // -> one complete `oa` sstable. Schema: `pk blob PRIMARY KEY, val blob`

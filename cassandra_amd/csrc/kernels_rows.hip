@@ -100,6 +100,7 @@ __device__ inline int pos_cmp(const UnfCols& in, uint64_t a, uint64_t b, const S
 struct SrcDesc2 {
     const uint8_t* data;
     const uint64_t* part_pos;  // n_parts+1
+    const uint64_t* part_end;  // scrub: explicit ends (else part_pos[li+1])
     uint32_t n_parts;
     int64_t min_ts, min_ldt;
     int32_t min_ttl;
@@ -117,7 +118,7 @@ __global__ void k_parse_count(const SrcDesc2* srcs, uint32_t n_srcs, uint32_t to
     uint32_t li = gi - sd.rec_base;
     const uint8_t* base = sd.data;
     uint64_t pos = sd.part_pos[li];
-    uint64_t end = sd.part_pos[li + 1];
+    uint64_t end = sd.part_end ? sd.part_end[li] : sd.part_pos[li + 1];
 
     uint32_t klen = ((uint32_t)base[pos] << 8) | base[pos + 1];
     pos += 2;

@@ -358,7 +358,8 @@ __global__ void __launch_bounds__(WAVE) k_lz4_compress_wave_t(const uint8_t* dat
 __global__ void __launch_bounds__(WAVE) k_lz4_decompress_wave(const ChunkDesc* chunks, uint32_t n,
                                                               int verify_crc,
                                                               unsigned long long* error,
-                                                              const uint32_t* crc_table) {
+                                                              const uint32_t* crc_table,
+                                                              uint8_t* bad_chunks = nullptr) {
     // compressed bytes are read through L1/L2 (all-lane same-address reads
     // broadcast; no LDS staging -> higher occupancy, same trade as the
     // compressor's global-src variant). crc_table is the 8x256 sliced set.
@@ -366,7 +367,7 @@ __global__ void __launch_bounds__(WAVE) k_lz4_decompress_wave(const ChunkDesc* c
     if (c >= n) return;
     int lane = threadIdx.x;
     ChunkDesc ch = chunks[c];
-    if (ch.comp_len > LZ4_SLOT) { if (lane == 0) atomicExch(error, 9ull); return; }
+    if (ch.comp_len > LZ4_SLOT) { if (lane == 0) { if (bad_chunks) bad_chunks[c] = 1; else atomicExch(error, 9ull); } return; }
     const uint8_t* s_comp = ch.comp;
     if (verify_crc && lane == 0) {
         // CRC32 of the compressed bytes, slicing-by-8 (zlib technique)
@@ -386,10 +387,10 @@ __global__ void __launch_bounds__(WAVE) k_lz4_decompress_wave(const ChunkDesc* c
         crc = ~crc;
         uint32_t stored = ((uint32_t)ch.comp[ch.comp_len] << 24) | ((uint32_t)ch.comp[ch.comp_len + 1] << 16) |
                           ((uint32_t)ch.comp[ch.comp_len + 2] << 8) | ch.comp[ch.comp_len + 3];
-        if (crc != stored) atomicExch(error, 1ull);
+        if (crc != stored) { if (bad_chunks) bad_chunks[c] = 1; else atomicExch(error, 1ull); }
     }
     uint32_t hdr = s_comp[0] | (s_comp[1] << 8) | (s_comp[2] << 16) | ((uint32_t)s_comp[3] << 24);
-    if (hdr != ch.out_len) { if (lane == 0) atomicExch(error, 2ull); return; }
+    if (hdr != ch.out_len) { if (lane == 0) { if (bad_chunks) bad_chunks[c] = 1; else atomicExch(error, 2ull); } return; }
     // uniform decode on all lanes
     uint32_t ip = 4, iend = ch.comp_len;
     uint32_t opos = 0, olen = ch.out_len;
@@ -401,7 +402,7 @@ __global__ void __launch_bounds__(WAVE) k_lz4_decompress_wave(const ChunkDesc* c
             uint32_t s;
             do { s = s_comp[ip++]; lit += s; } while (s == 255);
         }
-        if (opos + lit > olen || ip + lit > iend) { if (lane == 0) atomicExch(error, 3ull); return; }
+        if (opos + lit > olen || ip + lit > iend) { if (lane == 0) { if (bad_chunks) bad_chunks[c] = 1; else atomicExch(error, 3ull); } return; }
         for (uint32_t i = lane; i < lit; i += WAVE) out[opos + i] = s_comp[ip + i];
         opos += lit; ip += lit;
         if (ip >= iend) break;
@@ -412,7 +413,7 @@ __global__ void __launch_bounds__(WAVE) k_lz4_decompress_wave(const ChunkDesc* c
             uint32_t s;
             do { s = s_comp[ip++]; ml += s; } while (s == 255);
         }
-        if (off == 0 || opos < off || opos + ml > olen) { if (lane == 0) atomicExch(error, 4ull); return; }
+        if (off == 0 || opos < off || opos + ml > olen) { if (lane == 0) { if (bad_chunks) bad_chunks[c] = 1; else atomicExch(error, 4ull); } return; }
         const uint8_t* src = out + opos - off;
         if (ml <= off) {
             for (uint32_t i = lane; i < ml; i += WAVE) out[opos + i] = src[i];
@@ -424,7 +425,7 @@ __global__ void __launch_bounds__(WAVE) k_lz4_decompress_wave(const ChunkDesc* c
         }
         opos += ml;
     }
-    if (opos != olen && lane == 0) atomicExch(error, 5ull);
+    if (opos != olen && lane == 0) { if (bad_chunks) bad_chunks[c] = 1; else atomicExch(error, 5ull); }
 }
 
 }  // namespace gpuc

@@ -173,6 +173,13 @@ typedef struct gpuc_flush_rows {
 int gpuc_flush(const gpuc_flush_rows* rows, const char* output_base, int32_t device,
                char* error, size_t error_len);
 
+/* Scrub one sstable (SortedTableScrubber): salvage every partition whose
+ * byte range touches only CRC/decode-clean 16 KiB chunks and rewrite them as
+ * a clean sstable under output_base (recovery granularity documented in
+ * DESIGN.md; identical to the oracle's). kept/dropped may be NULL. */
+int gpuc_scrub(const char* input_base, const char* output_base, int32_t device,
+               uint64_t* kept, uint64_t* dropped, char* error, size_t error_len);
+
 /* Verify one sstable (Verifier.java / sstableverify --extended semantics):
  * CompressionInfo frame walk, per-chunk CRC32, full row-format walk,
  * strict DecoratedKey order, Digest.crc32 recomputation, bloom-filter

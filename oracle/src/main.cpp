@@ -187,6 +187,13 @@ static int cmd_flush(const std::string& outbase, const std::string& rowfile) {
     return 0;
 }
 
+static int cmd_scrub(const std::string& outbase, const std::string& inbase) {
+    ScrubResult sr = scrub_sstable(inbase, outbase);
+    printf("{\"partitions_kept\": %llu, \"partitions_dropped\": %llu}\n",
+           (unsigned long long)sr.kept, (unsigned long long)sr.dropped);
+    return 0;
+}
+
 static int cmd_compact(const std::string& outbase, std::vector<std::string>& inputs,
                        std::map<std::string, std::string>& kv) {
     using clk = std::chrono::steady_clock;
@@ -283,6 +290,7 @@ int main(int argc, char** argv) {
         if (cmd == "roundtrip") return cmd_roundtrip(pos.at(0));
         if (cmd == "gen") return cmd_gen(pos.at(0), kv);
         if (cmd == "flush") return cmd_flush(pos.at(0), pos.at(1));
+        if (cmd == "scrub") return cmd_scrub(pos.at(0), pos.at(1));
         if (cmd == "compact") {
             std::string outbase = pos.at(0);
             std::vector<std::string> ins(pos.begin() + 1, pos.end());

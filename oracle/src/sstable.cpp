@@ -925,6 +925,78 @@ static Partition read_partition(Reader& r, const Header& h) {
     return p;
 }
 
+ScrubResult scrub_sstable(const std::string& inbase, const std::string& outbase) {
+    bytes ci_b = read_file(inbase + "-CompressionInfo.db");
+    bytes data_b = read_file(inbase + "-Data.db");
+    bytes stats_b = read_file(inbase + "-Statistics.db");
+    bytes index_b = read_file(inbase + "-Index.db");
+    CompressionInfo ci = read_compression_info(ci_b);
+    StatisticsFile sf = read_statistics(stats_b);
+    // Index.db -> partition positions
+    std::vector<uint64_t> positions;
+    {
+        Reader r(index_b);
+        while (!r.eof()) {
+            uint16_t klen = r.be16();
+            r.skip(klen);
+            positions.push_back(read_unsigned_vint(r));
+            uint64_t promoted = read_unsigned_vint(r);
+            r.skip(promoted);
+        }
+        positions.push_back(ci.data_len);
+    }
+    // tolerant decompress with per-chunk bad flags
+    size_t n_chunks = ci.offsets.size();
+    std::vector<uint8_t> bad(n_chunks, 0);
+    bytes raw(ci.data_len, 0);
+    for (size_t i = 0; i < n_chunks; i++) {
+        uint64_t off = ci.offsets[i];
+        uint64_t end = (i + 1 < n_chunks) ? ci.offsets[i + 1] : data_b.size();
+        uint64_t want = std::min<uint64_t>(ci.params.chunk_len, ci.data_len - i * (uint64_t)ci.params.chunk_len);
+        if (end < off + 8) { bad[i] = 1; continue; }
+        uint64_t comp_len = end - off - 4;
+        uint32_t crc_stored = 0;
+        for (int k = 0; k < 4; k++) crc_stored = (crc_stored << 8) | data_b[off + comp_len + k];
+        if (crc_stored != crc32(data_b.data() + off, comp_len)) { bad[i] = 1; continue; }
+        uint32_t ulen = data_b[off] | (data_b[off + 1] << 8) | (data_b[off + 2] << 16) |
+                        ((uint32_t)data_b[off + 3] << 24);
+        if (ulen != want) { bad[i] = 1; continue; }
+        int got = LZ4_decompress_safe((const char*)data_b.data() + off + 4,
+                                      (char*)raw.data() + i * (uint64_t)ci.params.chunk_len,
+                                      (int)(comp_len - 4), (int)ulen);
+        if (got != (int)ulen) bad[i] = 1;
+    }
+    SSTable t;
+    t.generation = 1;
+    t.comp = ci.params;
+    t.header = sf.header;
+    t.header.stats.min_ts = sf.mins.min_timestamp == INT64_MIN ? TIMESTAMP_EPOCH : sf.mins.min_timestamp;
+    t.header.stats.min_ldt = sf.mins.min_ldt == INT64_MAX ? DELETION_TIME_EPOCH : sf.mins.min_ldt;
+    t.header.stats.min_ttl = sf.mins.min_ttl == INT32_MAX ? 0 : sf.mins.min_ttl;
+    uint64_t dropped = 0;
+    for (size_t i = 0; i + 1 < positions.size(); i++) {
+        uint64_t c0 = positions[i] / ci.params.chunk_len;
+        uint64_t c1 = (positions[i + 1] + ci.params.chunk_len - 1) / ci.params.chunk_len;
+        bool ok = true;
+        for (uint64_t c = c0; c < c1 && c < n_chunks; c++)
+            if (bad[c]) ok = false;
+        if (!ok) { dropped++; continue; }
+        bytes slice(raw.begin() + positions[i], raw.begin() + positions[i + 1]);
+        Reader r(slice);
+        Partition p = read_partition(r, t.header);
+        p.token = sf.partitioner == Partitioner::MURMUR3
+                      ? murmur3_token(p.key.data(), p.key.size()) : 0;
+        t.parts.push_back(std::move(p));
+    }
+    WriterOut w = write_sstable(t);
+    write_components(w, outbase);
+    ScrubResult sr;
+    sr.kept = t.parts.size();
+    sr.dropped = dropped;
+    return sr;
+}
+
+
 SSTable read_sstable(const std::string& base, bool keep_raw) {
     SSTable t;
     bytes ci_b = read_file(base + "-CompressionInfo.db");

@@ -214,6 +214,14 @@ struct SSTable {
 
 // ---- reader ----
 // base = path prefix like "/dir/oa-1-big" (components appended as "-Data.db"...)
+// scrub: salvage partitions untouched by corrupt chunks, rewrite clean.
+// Recovery granularity (this implementation's contract, mirrored by the GPU):
+// a partition is dropped iff any 16 KiB chunk overlapping its byte range
+// fails CRC/decode; the rest is rewritten with header mins from the input
+// Statistics (as a 1-input compaction would take them).
+struct ScrubResult { uint64_t kept = 0, dropped = 0; };
+ScrubResult scrub_sstable(const std::string& inbase, const std::string& outbase);
+
 SSTable read_sstable(const std::string& base, bool keep_raw = false);
 
 // ---- writer ----

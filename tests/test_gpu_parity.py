@@ -405,6 +405,39 @@ def test_anticompaction_split_parity(ca, oracle_bin, tmp_path):
     assert pr == a["partitions_out"], (pr, a["partitions_out"])
 
 
+def test_scrub_parity(ca, oracle_bin, tmp_path):
+    """gpuc_scrub (SortedTableScrubber): flip a byte mid-Data.db, scrub on
+    GPU and in the oracle -> identical kept/dropped counts and byte-identical
+    scrubbed sstables; the scrubbed output passes gpuc_verify. Also: scrub of
+    a clean sstable keeps every partition; statics/composite schema covered."""
+    import json
+    d = str(tmp_path)
+    _oracle_gen(d, seed=171, n=1, rows=3000, vlen=300, overlap=0, tomb=10)
+    os.makedirs(d + "/w")
+    _oracle_gen(d + "/w", seed=172, n=1, rows=60, crows=50, vlen=200, rtomb=30,
+                cktext=1, keylen=20, ncols=3, overlap=0, statics=40)
+    for sub, base in [("", f"{d}/oa-1-big"), ("/w", f"{d}/w/oa-1-big")]:
+        with open(base + "-Data.db", "r+b") as f:
+            f.seek(os.path.getsize(base + "-Data.db") // 2)
+            b0 = f.read(1)
+            f.seek(-1, 1)
+            f.write(bytes([b0[0] ^ 0x10]))
+        out = subprocess.run([ORACLE, "scrub", f"{d}{sub}/oa-80-big", base],
+                             capture_output=True, text=True, check=True)
+        ores = json.loads(out.stdout.splitlines()[-1])
+        kept, dropped = ca.scrub(base, f"{d}{sub}/oa-81-big")
+        assert dropped > 0, "corruption did not drop any partition"
+        assert (kept, dropped) == (ores["partitions_kept"], ores["partitions_dropped"])
+        _assert_dirs_equal(f"{d}{sub}/oa-80-big", f"{d}{sub}/oa-81-big")
+        ca.verify(f"{d}{sub}/oa-81-big")
+    # clean input -> everything kept, output verifies
+    os.makedirs(d + "/c")
+    _oracle_gen(d + "/c", seed=173, n=1, rows=800, vlen=150, overlap=0)
+    kept, dropped = ca.scrub(f"{d}/c/oa-1-big", f"{d}/c/oa-81-big")
+    assert dropped == 0 and kept > 0
+    ca.verify(f"{d}/c/oa-81-big")
+
+
 def test_compact_rejects_corrupt_input(ca, oracle_bin, tmp_path):
     """A flipped byte in an input chunk must fail the compaction loudly
     (CompressedChunkReader CRC semantics) — no silent bad output."""
