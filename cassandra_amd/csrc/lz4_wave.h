@@ -257,18 +257,33 @@ __device__ inline int lz4_wave_compress(const uint8_t* __restrict__ s_chunk, int
                 }
                 if (lane == 0) { dst[op] = (uint8_t)offv; dst[op + 1] = (uint8_t)(offv >> 8); }
                 op += 2;
-                // match length from ip+4 / match+4 up to matchlimit
+                // match length from ip+4 / match+4 up to matchlimit.
+                // dword-per-lane: 256 bytes per ballot round trip (the loop is
+                // serial — each round's ballot gates the next — so bytes per
+                // round trip is the lever). Per-byte semantics identical to
+                // the scalar loop: a byte at/after matchlimit is a mismatch.
                 int mc = 0;
                 {
                     int offb = 0;
                     while (true) {
-                        int pi = ip + 4 + offb + lane;
-                        bool eqb = pi < matchlimit && s_chunk[pi] == s_chunk[match + 4 + offb + lane];
-                        uint64_t ne = wave_ballot(!eqb);
-                        int run = ne ? (int)__ffsll((long long)ne) - 1 : WAVE;
-                        mc = offb + run;
-                        if (run < WAVE) break;
-                        offb += WAVE;
+                        int pi = ip + 4 + offb + 4 * lane;
+                        int navail = matchlimit - pi;
+                        uint32_t aa = 0, bb = 0;
+                        if (navail > 0) {
+                            aa = lds_read32(s_chunk, (uint32_t)pi);
+                            bb = lds_read32(s_chunk, (uint32_t)(match + 4 + offb + 4 * lane));
+                        }
+                        uint32_t x = aa ^ bb;
+                        int eq4 = navail <= 0 ? 0
+                                              : (x == 0 ? 4 : (__ffs((int)x) - 1) >> 3);
+                        if (eq4 > navail) eq4 = navail;
+                        uint64_t ne = wave_ballot(eq4 < 4);
+                        if (ne) {
+                            int fl = (int)__ffsll((long long)ne) - 1;
+                            mc = offb + 4 * fl + __shfl(eq4, fl);
+                            break;
+                        }
+                        offb += 4 * WAVE;
                     }
                 }
                 ip += mc + LZ4M_MINMATCH;
