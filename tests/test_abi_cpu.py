@@ -10,8 +10,15 @@ from conftest import REPO
 
 
 def test_library_exports(product_lib):
+    """Every function include/gpucompact.h declares must be exported."""
+    import re
     lib = ctypes.CDLL(product_lib)
-    for sym in ["gpuc_compact", "gpuc_generate", "gpuc_version", "gpuc_device_count"]:
+    hdr = open(os.path.join(REPO, "include", "gpucompact.h")).read()
+    syms = set(re.findall(r"^\s*(?:int|const char\*|int32_t)\s+(gpuc_\w+)\s*\(",
+                          hdr, re.M))
+    assert {"gpuc_compact", "gpuc_generate", "gpuc_verify", "gpuc_flush",
+            "gpuc_scrub", "gpuc_version", "gpuc_device_count"} <= syms, syms
+    for sym in sorted(syms):
         assert hasattr(lib, sym), f"missing export {sym}"
     lib.gpuc_version.restype = ctypes.c_char_p
     assert b"gfx950" in lib.gpuc_version()

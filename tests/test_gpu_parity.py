@@ -438,6 +438,36 @@ def test_scrub_parity(ca, oracle_bin, tmp_path):
     ca.verify(f"{d}/c/oa-81-big")
 
 
+def test_empty_outputs(ca, oracle_bin, tmp_path):
+    """Degenerate outputs stay byte-identical: (1) a compaction whose every
+    row is purged writes the oracle's empty sstable; (2) a scrub where every
+    chunk is corrupt keeps 0 partitions and writes the same empty table."""
+    import json
+    d = str(tmp_path)
+    _oracle_gen(d, seed=5, n=1, rows=200, vlen=50, tomb=100, overlap=0,
+                ts0=1000000, ldt0=1000)
+    _oracle_compact(f"{d}/oa-90-big", [f"{d}/oa-1-big"], gcbefore=2000000000)
+    ca.compact([f"{d}/oa-1-big"], f"{d}/oa-91-big", gc_before=2000000000)
+    _assert_dirs_equal(f"{d}/oa-90-big", f"{d}/oa-91-big")
+    os.makedirs(d + "/s")
+    _oracle_gen(d + "/s", seed=6, n=1, rows=50, vlen=40, overlap=0)
+    base = f"{d}/s/oa-1-big"
+    sz = os.path.getsize(base + "-Data.db")
+    with open(base + "-Data.db", "r+b") as f:
+        for off in range(5, sz, 4000):
+            f.seek(off)
+            b0 = f.read(1)
+            f.seek(-1, 1)
+            f.write(bytes([b0[0] ^ 0xFF]))
+    out = subprocess.run([ORACLE, "scrub", f"{d}/s/oa-80-big", base],
+                         capture_output=True, text=True, check=True)
+    ores = json.loads(out.stdout.splitlines()[-1])
+    kept, dropped = ca.scrub(base, f"{d}/s/oa-81-big")
+    assert kept == 0 and (kept, dropped) == (ores["partitions_kept"],
+                                             ores["partitions_dropped"])
+    _assert_dirs_equal(f"{d}/s/oa-80-big", f"{d}/s/oa-81-big")
+
+
 def test_compact_rejects_corrupt_input(ca, oracle_bin, tmp_path):
     """A flipped byte in an input chunk must fail the compaction loudly
     (CompressedChunkReader CRC semantics) — no silent bad output."""
