@@ -177,12 +177,17 @@ __device__ inline int lz4_wave_compress(const uint8_t* __restrict__ s_chunk, int
                     }
                 }
                 // exact match test: in-window predecessor position if any,
-                // else the pre-window table entry. The predecessor's position
-                // is closed-form per lane (no divergent cross-lane read).
-                uint32_t cand_pos = pred >= 0 ? (uint32_t)(P0 + lz4_adv_sum(k0, k0 + pred))
-                                              : (uint32_t)t_l;
-                uint32_t cand_val = pred >= 0 ? (valid ? lds_read32(s_chunk, cand_pos) : 0)
-                                              : spec_cand;
+                // else the pre-window table entry. The predecessor lane
+                // already holds both its probe position AND its probe dword
+                // in registers — shfl them instead of re-deriving the
+                // position and re-loading the dword (removes a dependent
+                // global load from the per-window critical chain; the shfl
+                // index is per-lane, which CDNA's ds_bpermute supports).
+                int pred_idx = pred >= 0 ? pred : 0;
+                uint32_t pred_pos = (uint32_t)__shfl(p_l, pred_idx);
+                uint32_t pred_val = (uint32_t)__shfl((int)v_l, pred_idx);
+                uint32_t cand_pos = pred >= 0 ? pred_pos : (uint32_t)t_l;
+                uint32_t cand_val = pred >= 0 ? pred_val : spec_cand;
                 bool m_l = valid && cand_val == v_l;
                 uint64_t abort_mask = wave_ballot(!valid);
                 int first_abort = abort_mask ? (int)__ffsll((long long)abort_mask) - 1 : WAVE;
@@ -202,7 +207,7 @@ __device__ inline int lz4_wave_compress(const uint8_t* __restrict__ s_chunk, int
                     if (lane <= commit_hi) vt[h_l] = (uint16_t)p_l;
                 }
                 if (have_match) {
-                    ip = P0 + lz4_adv_sum(k0, k0 + first_event);
+                    ip = __shfl(p_l, first_event);  // == P0 + adv_sum(k0, k0+first_event)
                     match = (int)(uint32_t)__shfl((int)cand_pos, first_event);
                     found = true;
                 } else if (first_abort < WAVE) {
