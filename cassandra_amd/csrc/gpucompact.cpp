@@ -1266,7 +1266,7 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
             HIP_CHECK(hipEventRecord(ev_h2d[s], copy_stream));
             HIP_CHECK(hipStreamWaitEvent(stream, ev_h2d[s], 0));
             if (!chunks.empty())
-                hipLaunchKernelGGL(k_lz4_decompress_wave, dim3((uint32_t)chunks.size()), dim3(WAVE), 0,
+                hipLaunchKernelGGL(k_lz4_decompress_wave, dim3(lz4_decomp_grid((uint32_t)chunks.size(), 1)), dim3(WAVE), 0,
                                    stream, d_chunks_s[s].as<ChunkDesc>(), (uint32_t)chunks.size(), 1,
                                    d_error.as<unsigned long long>(), (const uint32_t*)g_crc256);
         }
@@ -1987,7 +1987,7 @@ extern "C" int gpuc_verify(const char* input_base, int32_t device, char* error,
         d_error.alloc(8);
         HIP_CHECK(hipMemsetAsync(d_error.p, 0, 8, stream));
         if (!chunks.empty())
-            hipLaunchKernelGGL(k_lz4_decompress_wave, dim3((uint32_t)chunks.size()), dim3(WAVE), 0,
+            hipLaunchKernelGGL(k_lz4_decompress_wave, dim3(lz4_decomp_grid((uint32_t)chunks.size(), 1)), dim3(WAVE), 0,
                                stream, d_chunks.as<ChunkDesc>(), (uint32_t)chunks.size(), 1,
                                d_error.as<unsigned long long>(), (const uint32_t*)g_crc256);
         {
@@ -2148,7 +2148,7 @@ extern "C" int gpuc_scrub(const char* input_base, const char* output_base, int32
         HIP_CHECK(hipMemcpyAsync(d_chunksb.p, chunks.data(), chunks.size() * sizeof(ChunkDesc),
                                  hipMemcpyHostToDevice, stream));
         if (!chunks.empty())
-            hipLaunchKernelGGL(k_lz4_decompress_wave, dim3((uint32_t)chunks.size()), dim3(WAVE), 0,
+            hipLaunchKernelGGL(k_lz4_decompress_wave, dim3(lz4_decomp_grid((uint32_t)chunks.size(), 1)), dim3(WAVE), 0,
                                stream, d_chunksb.as<ChunkDesc>(), (uint32_t)chunks.size(), 1,
                                d_error.as<unsigned long long>(), (const uint32_t*)g_crc256,
                                d_bad.as<uint8_t>());

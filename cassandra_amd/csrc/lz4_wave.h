@@ -375,6 +375,13 @@ __global__ void __launch_bounds__(WAVE) k_lz4_compress_wave_t(const uint8_t* dat
 // wave-cooperative decompress: comp chunk staged in LDS; the token chain runs
 // redundantly on all lanes (uniform); copies are wave-parallel.
 // ---------------------------------------------------------------------------
+// grid helper: with verify_crc the CRC work runs in extra trailing blocks
+// (one chunk per LANE — the byte-serial CRC would otherwise hold all 64
+// lanes of each decode block hostage behind one lane's loop)
+static inline uint32_t lz4_decomp_grid(uint32_t n, int verify_crc) {
+    return verify_crc ? n + (n + WAVE - 1) / WAVE : n;
+}
+
 __global__ void __launch_bounds__(WAVE) k_lz4_decompress_wave(const ChunkDesc* chunks, uint32_t n,
                                                               int verify_crc,
                                                               unsigned long long* error,
@@ -384,13 +391,16 @@ __global__ void __launch_bounds__(WAVE) k_lz4_decompress_wave(const ChunkDesc* c
     // broadcast; no LDS staging -> higher occupancy, same trade as the
     // compressor's global-src variant). crc_table is the 8x256 sliced set.
     uint32_t c = blockIdx.x;
-    if (c >= n) return;
     int lane = threadIdx.x;
-    ChunkDesc ch = chunks[c];
-    if (ch.comp_len > LZ4_SLOT) { if (lane == 0) { if (bad_chunks) bad_chunks[c] = 1; else atomicExch(error, 9ull); } return; }
-    const uint8_t* s_comp = ch.comp;
-    if (verify_crc && lane == 0) {
-        // CRC32 of the compressed bytes, slicing-by-8 (zlib technique)
+    if (c >= n) {
+        // CRC block: lane l checks chunk (c-n)*WAVE + l (launched only when
+        // verify_crc; see lz4_decomp_grid). Each lane walks its own chunk
+        // serially — consecutive iterations reuse the lane's cachelines.
+        uint32_t ci = (c - n) * WAVE + (uint32_t)lane;
+        if (ci >= n) return;
+        ChunkDesc ch = chunks[ci];
+        if (ch.comp_len > LZ4_SLOT) return;  // decode block flags it
+        const uint8_t* s_comp = ch.comp;
         uint32_t crc = 0xFFFFFFFFu;
         uint32_t i = 0;
         for (; i + 8 <= ch.comp_len; i += 8) {
@@ -405,10 +415,14 @@ __global__ void __launch_bounds__(WAVE) k_lz4_decompress_wave(const ChunkDesc* c
         }
         for (; i < ch.comp_len; i++) crc = crc_table[(crc ^ s_comp[i]) & 0xFF] ^ (crc >> 8);
         crc = ~crc;
-        uint32_t stored = ((uint32_t)ch.comp[ch.comp_len] << 24) | ((uint32_t)ch.comp[ch.comp_len + 1] << 16) |
-                          ((uint32_t)ch.comp[ch.comp_len + 2] << 8) | ch.comp[ch.comp_len + 3];
-        if (crc != stored) { if (bad_chunks) bad_chunks[c] = 1; else atomicExch(error, 1ull); }
+        uint32_t stored = ((uint32_t)s_comp[ch.comp_len] << 24) | ((uint32_t)s_comp[ch.comp_len + 1] << 16) |
+                          ((uint32_t)s_comp[ch.comp_len + 2] << 8) | s_comp[ch.comp_len + 3];
+        if (crc != stored) { if (bad_chunks) bad_chunks[ci] = 1; else atomicExch(error, 1ull); }
+        return;
     }
+    ChunkDesc ch = chunks[c];
+    if (ch.comp_len > LZ4_SLOT) { if (lane == 0) { if (bad_chunks) bad_chunks[c] = 1; else atomicExch(error, 9ull); } return; }
+    const uint8_t* s_comp = ch.comp;
     uint32_t hdr = s_comp[0] | (s_comp[1] << 8) | (s_comp[2] << 16) | ((uint32_t)s_comp[3] << 24);
     if (hdr != ch.out_len) { if (lane == 0) { if (bad_chunks) bad_chunks[c] = 1; else atomicExch(error, 2ull); } return; }
     // uniform decode on all lanes
@@ -423,7 +437,16 @@ __global__ void __launch_bounds__(WAVE) k_lz4_decompress_wave(const ChunkDesc* c
             do { s = s_comp[ip++]; lit += s; } while (s == 255);
         }
         if (opos + lit > olen || ip + lit > iend) { if (lane == 0) { if (bad_chunks) bad_chunks[c] = 1; else atomicExch(error, 3ull); } return; }
-        for (uint32_t i = lane; i < lit; i += WAVE) out[opos + i] = s_comp[ip + i];
+        for (uint32_t i = 4u * lane; i < lit; i += 4u * WAVE) {
+            uint32_t nb = lit - i;
+            if (nb >= 4) {
+                uint32_t v;
+                memcpy(&v, s_comp + ip + i, 4);
+                memcpy(out + opos + i, &v, 4);
+            } else {
+                for (uint32_t j = 0; j < nb; j++) out[opos + i + j] = s_comp[ip + i + j];
+            }
+        }
         opos += lit; ip += lit;
         if (ip >= iend) break;
         uint32_t off = s_comp[ip] | (s_comp[ip + 1] << 8);
@@ -435,8 +458,26 @@ __global__ void __launch_bounds__(WAVE) k_lz4_decompress_wave(const ChunkDesc* c
         }
         if (off == 0 || opos < off || opos + ml > olen) { if (lane == 0) { if (bad_chunks) bad_chunks[c] = 1; else atomicExch(error, 4ull); } return; }
         const uint8_t* src = out + opos - off;
-        if (ml <= off) {
-            for (uint32_t i = lane; i < ml; i += WAVE) out[opos + i] = src[i];
+        if (off == 1) {
+            // byte run: broadcast, dword stores
+            uint32_t b4 = 0x01010101u * src[0];
+            for (uint32_t i = 4u * lane; i < ml; i += 4u * WAVE) {
+                uint32_t nb = ml - i;
+                if (nb >= 4) memcpy(out + opos + i, &b4, 4);
+                else for (uint32_t j = 0; j < nb; j++) out[opos + i + j] = (uint8_t)b4;
+            }
+        } else if (ml <= off) {
+            // disjoint regions: dword copy
+            for (uint32_t i = 4u * lane; i < ml; i += 4u * WAVE) {
+                uint32_t nb = ml - i;
+                if (nb >= 4) {
+                    uint32_t v;
+                    memcpy(&v, src + i, 4);
+                    memcpy(out + opos + i, &v, 4);
+                } else {
+                    for (uint32_t j = 0; j < nb; j++) out[opos + i + j] = src[i + j];
+                }
+            }
         } else {
             // overlapped copy (ml > off) == periodic repetition of the last
             // `off` bytes; the modulo form avoids reading bytes this same

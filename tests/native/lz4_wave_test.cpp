@@ -135,7 +135,7 @@ int main() {
                 ChunkDesc* d_cd;
                 hipMalloc(&d_cd, sizeof(cd));
                 hipMemcpy(d_cd, &cd, sizeof(cd), hipMemcpyHostToDevice);
-                hipLaunchKernelGGL(k_lz4_decompress_wave, dim3(1), dim3(64), 0, 0, d_cd, 1, 1, d_err, d_tab);
+                hipLaunchKernelGGL(k_lz4_decompress_wave, dim3(lz4_decomp_grid(1, 1)), dim3(64), 0, 0, d_cd, 1, 1, d_err, d_tab);
                 hipDeviceSynchronize();
                 unsigned long long err;
                 hipMemcpy(&err, d_err, 8, hipMemcpyDeviceToHost);
