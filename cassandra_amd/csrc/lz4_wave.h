@@ -82,6 +82,22 @@ __device__ inline int lz4_adv_sum(int a, int b) {
 }
 __device__ inline int lz4_adv(int m) { return m == 0 ? 1 : (63 + m) >> 6; }
 
+// wave-parallel byte-exact copy, dword per lane (src/dst any alignment,
+// regions must not overlap)
+__device__ inline void wave_copy(uint8_t* __restrict__ dst, const uint8_t* __restrict__ src,
+                                 int nbytes, int lane) {
+    for (int i = 4 * lane; i < nbytes; i += 4 * WAVE) {
+        int nb = nbytes - i;
+        if (nb >= 4) {
+            uint32_t v;
+            memcpy(&v, src + i, 4);
+            memcpy(dst + i, &v, 4);
+        } else {
+            for (int j = 0; j < nb; j++) dst[i + j] = src[i + j];
+        }
+    }
+}
+
 template <bool STAGE_LDS = true>
 __device__ inline int lz4_wave_compress(const uint8_t* __restrict__ s_chunk, int srcSize,
                                         uint8_t* __restrict__ dst, uint16_t* __restrict__ s_table,
@@ -106,7 +122,7 @@ __device__ inline int lz4_wave_compress(const uint8_t* __restrict__ s_chunk, int
             if (lane == 0) dst[op] = (uint8_t)(lastRun << LZ4M_ML_BITS);
             op++;
         }
-        for (int i = lane; i < lastRun; i += WAVE) dst[op + i] = s_chunk[anchor + i];
+        wave_copy(dst + op, s_chunk + anchor, lastRun, lane);
         op += lastRun;
     };
 
@@ -250,7 +266,7 @@ __device__ inline int lz4_wave_compress(const uint8_t* __restrict__ s_chunk, int
             } else {
                 token = (uint8_t)(lit << LZ4M_ML_BITS);
             }
-            for (int i = lane; i < lit; i += WAVE) dst[op + i] = s_chunk[anchor + i];
+            wave_copy(dst + op, s_chunk + anchor, lit, lane);
             op += lit;
 
             // ================= offset + match length =================
