@@ -213,7 +213,16 @@ __global__ void k_chunk_gather(const uint8_t* slots, const uint32_t* csize, cons
     uint32_t sz = csize[c];
     const uint8_t* src = slots + (uint64_t)c * LZ4_SLOT;
     uint8_t* dst = out + foff[c] + (uint64_t)c * 4;  // +4 per preceding chunk CRC
-    for (uint32_t i = lane; i < sz; i += WAVE) dst[i] = src[i];
+    for (uint32_t i = 4u * lane; i < sz; i += 4u * WAVE) {
+        uint32_t nb = sz - i;
+        if (nb >= 4) {
+            uint32_t v;
+            memcpy(&v, src + i, 4);
+            memcpy(dst + i, &v, 4);
+        } else {
+            for (uint32_t j = 0; j < nb; j++) dst[i + j] = src[i + j];
+        }
+    }
     if (lane == 0) {
         uint32_t crc = ccrc[c];
         dst[sz] = (uint8_t)(crc >> 24); dst[sz + 1] = (uint8_t)(crc >> 16);

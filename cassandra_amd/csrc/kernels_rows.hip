@@ -1239,9 +1239,8 @@ __device__ inline uint64_t static_row_bytes(const OutParts& op, uint64_t g,
         if (!(cfb & 4)) {
             uint32_t vlen = st.val_len[oc];
             if (sp.sch.static_fixed[c] < 0) put_uv(vlen);
-            const uint8_t* src = (const uint8_t*)st.val_addr[oc];
-            for (uint32_t i = (uint32_t)lane; i < vlen; i += WAVE)
-                out_data[data_off + pos + i] = src[i];
+            wave_copy(out_data + data_off + pos, (const uint8_t*)st.val_addr[oc],
+                      (int)vlen, lane);
             pos += vlen;
         }
     }
@@ -1513,11 +1512,9 @@ __device__ uint64_t part_walk(const OutParts& op, const UnfCols& out, uint64_t g
                 if (!(cfb & 4)) {
                     uint32_t vlen = out.val_len[oc];
                     if (sp.sch.col_fixed[c] < 0) emit_uv(vlen);
-                    if (EMIT) {
-                        const uint8_t* src = (const uint8_t*)out.val_addr[oc];
-                        for (uint32_t i = (uint32_t)lane; i < vlen; i += WAVE)
-                            out_data[data_off + pos + i] = src[i];
-                    }
+                    if (EMIT)
+                        wave_copy(out_data + data_off + pos,
+                                  (const uint8_t*)out.val_addr[oc], (int)vlen, lane);
                     pos += vlen;
                 }
             }
