@@ -438,6 +438,20 @@ def test_scrub_parity(ca, oracle_bin, tmp_path):
     ca.verify(f"{d}/c/oa-81-big")
 
 
+def test_compaction_associativity_gpu(ca, oracle_bin, tmp_path):
+    """GPU analog of the oracle associativity law: re-compacting a GPU
+    compaction output with a third sstable equals the one-shot three-way
+    compaction, byte-for-byte (levelled/tiered strategies re-compact
+    outputs, so intermediate-output header mins must propagate exactly)."""
+    d = str(tmp_path)
+    _oracle_gen(d, seed=311, n=3, rows=1500, vlen=200, overlap=30, tomb=15)
+    ins = [f"{d}/oa-{g}-big" for g in (1, 2, 3)]
+    ca.compact(ins[:2], f"{d}/oa-50-big")
+    ca.compact([f"{d}/oa-50-big", ins[2]], f"{d}/oa-60-big")
+    ca.compact(ins, f"{d}/oa-70-big")
+    _assert_dirs_equal(f"{d}/oa-60-big", f"{d}/oa-70-big")
+
+
 def test_empty_outputs(ca, oracle_bin, tmp_path):
     """Degenerate outputs stay byte-identical: (1) a compaction whose every
     row is purged writes the oracle's empty sstable; (2) a scrub where every

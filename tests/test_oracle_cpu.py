@@ -209,3 +209,33 @@ def test_merge_semantics_laws(oracle_bin, tmp_path):
     r = subprocess.run([exe], capture_output=True, text=True)
     assert r.returncode == 0, r.stdout
     assert "merge laws OK" in r.stdout
+
+
+def test_compaction_associativity(oracle_bin, tmp_path):
+    """Multi-level compaction converges: compact(compact(A,B),C) is
+    byte-identical to compact(A,B,C) (no purge). This is what lets levelled/
+    tiered strategies re-compact outputs; it exercises EncodingStats min
+    propagation through intermediate outputs (a wrong header min changes
+    every vint delta downstream)."""
+    import subprocess
+    COMPONENTS = ["Data.db", "Index.db", "CompressionInfo.db", "Filter.db",
+                  "Digest.crc32", "Statistics.db", "Summary.db", "TOC.txt"]
+    d = str(tmp_path)
+    for seed, gen in [(42, "n=3 rows=800 vlen=120 overlap=30 tomb=15"),
+                      (91, "n=3 rows=40 crows=60 vlen=150 overlap=40 rtomb=40 "
+                           "cktext=1 statics=50 ncols=3 colmiss=20 keylen=20")]:
+        sub = f"{d}/{seed}"
+        os.makedirs(sub)
+        subprocess.run([ORACLE, "gen", sub, f"seed={seed}", *gen.split()],
+                       check=True, capture_output=True)
+        ins = [f"{sub}/oa-{g}-big" for g in (1, 2, 3)]
+        subprocess.run([ORACLE, "compact", f"{sub}/oa-50-big", ins[0], ins[1]],
+                       check=True, capture_output=True)
+        subprocess.run([ORACLE, "compact", f"{sub}/oa-60-big",
+                        f"{sub}/oa-50-big", ins[2]], check=True, capture_output=True)
+        subprocess.run([ORACLE, "compact", f"{sub}/oa-70-big", *ins],
+                       check=True, capture_output=True)
+        for c in COMPONENTS:
+            a = open(f"{sub}/oa-60-big-{c}", "rb").read()
+            b = open(f"{sub}/oa-70-big-{c}", "rb").read()
+            assert a == b, f"associativity broken in {c} (seed {seed})"
