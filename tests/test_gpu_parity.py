@@ -439,17 +439,27 @@ def test_scrub_parity(ca, oracle_bin, tmp_path):
 
 
 def test_compaction_associativity_gpu(ca, oracle_bin, tmp_path):
-    """GPU analog of the oracle associativity law: re-compacting a GPU
-    compaction output with a third sstable equals the one-shot three-way
-    compaction, byte-for-byte (levelled/tiered strategies re-compact
-    outputs, so intermediate-output header mins must propagate exactly)."""
+    """Multi-level compaction on GPU: (1) with key overlap, re-compacting a
+    compaction OUTPUT (a parity case the generator never produces: its
+    StatsMetadata mins can exceed its header mins) stays byte-identical to
+    the oracle doing the same chain; (2) with disjoint keys the chain is
+    also byte-associative: compact(compact(A,B),C) == compact(A,B,C)."""
     d = str(tmp_path)
     _oracle_gen(d, seed=311, n=3, rows=1500, vlen=200, overlap=30, tomb=15)
     ins = [f"{d}/oa-{g}-big" for g in (1, 2, 3)]
     ca.compact(ins[:2], f"{d}/oa-50-big")
-    ca.compact([f"{d}/oa-50-big", ins[2]], f"{d}/oa-60-big")
-    ca.compact(ins, f"{d}/oa-70-big")
-    _assert_dirs_equal(f"{d}/oa-60-big", f"{d}/oa-70-big")
+    _oracle_compact(f"{d}/oa-80-big", ins[:2])
+    _assert_dirs_equal(f"{d}/oa-50-big", f"{d}/oa-80-big")
+    ca.compact([f"{d}/oa-80-big", ins[2]], f"{d}/oa-60-big")
+    _oracle_compact(f"{d}/oa-81-big", [f"{d}/oa-80-big", ins[2]])
+    _assert_dirs_equal(f"{d}/oa-60-big", f"{d}/oa-81-big")
+    os.makedirs(d + "/dj")
+    _oracle_gen(d + "/dj", seed=313, n=3, rows=1200, vlen=150, overlap=0, tomb=15)
+    dins = [f"{d}/dj/oa-{g}-big" for g in (1, 2, 3)]
+    ca.compact(dins[:2], f"{d}/dj/oa-50-big")
+    ca.compact([f"{d}/dj/oa-50-big", dins[2]], f"{d}/dj/oa-60-big")
+    ca.compact(dins, f"{d}/dj/oa-70-big")
+    _assert_dirs_equal(f"{d}/dj/oa-60-big", f"{d}/dj/oa-70-big")
 
 
 def test_empty_outputs(ca, oracle_bin, tmp_path):

@@ -213,16 +213,20 @@ def test_merge_semantics_laws(oracle_bin, tmp_path):
 
 def test_compaction_associativity(oracle_bin, tmp_path):
     """Multi-level compaction converges: compact(compact(A,B),C) is
-    byte-identical to compact(A,B,C) (no purge). This is what lets levelled/
-    tiered strategies re-compact outputs; it exercises EncodingStats min
-    propagation through intermediate outputs (a wrong header min changes
-    every vint delta downstream)."""
+    byte-identical to compact(A,B,C) when no rows are dropped (disjoint
+    keys, no purge) — it exercises EncodingStats min propagation through
+    intermediate outputs (a wrong header min changes every vint delta
+    downstream). With key overlap the law holds only LOGICALLY, not at
+    byte level: dropping shadowed data can raise the intermediate's
+    StatsMetadata mins, which legitimately changes downstream header
+    deltas (the reference behaves the same — SerializationHeader.make
+    reads the inputs' stats)."""
     import subprocess
     COMPONENTS = ["Data.db", "Index.db", "CompressionInfo.db", "Filter.db",
                   "Digest.crc32", "Statistics.db", "Summary.db", "TOC.txt"]
     d = str(tmp_path)
-    for seed, gen in [(42, "n=3 rows=800 vlen=120 overlap=30 tomb=15"),
-                      (91, "n=3 rows=40 crows=60 vlen=150 overlap=40 rtomb=40 "
+    for seed, gen in [(42, "n=3 rows=800 vlen=120 overlap=0 tomb=15"),
+                      (91, "n=3 rows=40 crows=60 vlen=150 overlap=0 rtomb=40 "
                            "cktext=1 statics=50 ncols=3 colmiss=20 keylen=20")]:
         sub = f"{d}/{seed}"
         os.makedirs(sub)
