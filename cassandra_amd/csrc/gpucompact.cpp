@@ -1127,12 +1127,18 @@ struct CompactSetup {
     std::vector<int32_t> ck_widths;      // per clustering column
     std::vector<int32_t> static_fixed_h; // per static column
     // unsharded fast path: whole-file Data.db reads started during index
-    // parse (compact_one adopts them when its window covers the full file)
-    mutable std::vector<std::thread> full_readers;
+    // parse as ordered STRIPES per source (compact_one adopts them when its
+    // window covers the full file and overlaps H2D with the remaining reads)
+    struct StripeRead {
+        mutable std::vector<std::thread> th;  // one per stripe, in file order
+        std::vector<uint64_t> off, len;
+    };
+    mutable std::vector<StripeRead> full_stripes;
     std::vector<uint8_t*> full_pin;
     ~CompactSetup() {
-        for (auto& t : full_readers)
-            if (t.joinable()) t.join();
+        for (auto& sr : full_stripes)
+            for (auto& t : sr.th)
+                if (t.joinable()) t.join();
     }
 };
 
@@ -1225,9 +1231,6 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         std::vector<const uint8_t*> vbase(k, nullptr);  // virtual decompressed origin
         for (int s = 0; s < k; s++) {
             double tr = wall();
-            if (adopt) su.full_readers[s].join();
-            else data_readers[s].join();  // in-order wait; all reads run concurrently
-            ms_read_data = wall() - tr + ms_read_data;
             auto& ci = cinfos[s];
             uint64_t n_chunks_file = ci.offsets.size();
             uint64_t c_lo = win_chunk_lo[s];
@@ -1235,9 +1238,33 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
             if (c_hi > n_chunks_file) c_hi = n_chunks_file;
             uint64_t n_wchunks = c_hi - c_lo;
             d_comp[s].alloc(comp_sz[s] ? comp_sz[s] : 1);
-            if (comp_sz[s])
-                HIP_CHECK(hipMemcpyAsync(d_comp[s].p, comp_pin[s], comp_sz[s],
-                                         hipMemcpyHostToDevice, copy_stream));
+            if (adopt && win_comp_lo[s] == 0 && comp_sz[s] == su.comp_file_sz[s]) {
+                // stripe-progressive H2D: copy each stripe as its read lands
+                // (whole-file window only; stripe offsets are file-absolute)
+                auto& sr = su.full_stripes[s];
+                for (size_t t = 0; t < sr.th.size(); t++) {
+                    if (sr.th[t].joinable()) sr.th[t].join();
+                    if (sr.len[t])
+                        HIP_CHECK(hipMemcpyAsync(d_comp[s].as<uint8_t>() + sr.off[t],
+                                                 comp_pin[s] + sr.off[t], sr.len[t],
+                                                 hipMemcpyHostToDevice, copy_stream));
+                }
+            } else if (adopt) {
+                // token-restricted window over a preread file: wait for all
+                // stripes, then one window-sized copy
+                auto& sr = su.full_stripes[s];
+                for (auto& t : sr.th)
+                    if (t.joinable()) t.join();
+                if (comp_sz[s])
+                    HIP_CHECK(hipMemcpyAsync(d_comp[s].p, comp_pin[s], comp_sz[s],
+                                             hipMemcpyHostToDevice, copy_stream));
+            } else {
+                data_readers[s].join();  // in-order wait; all reads run concurrently
+                if (comp_sz[s])
+                    HIP_CHECK(hipMemcpyAsync(d_comp[s].p, comp_pin[s], comp_sz[s],
+                                             hipMemcpyHostToDevice, copy_stream));
+            }
+            ms_read_data = wall() - tr + ms_read_data;
             d_data[s].alloc(n_wchunks * (uint64_t)CHUNK_LEN + 16);
             // positions stay ABSOLUTE: the decompressed window is addressed
             // through a virtual origin so partition offsets need no rewrite
@@ -1709,17 +1736,26 @@ static void compact_setup(const gpuc_job* job, CompactSetup& su, bool preread_fu
     // index decode run while they stream into the pinned arenas
     if (preread_full) {
         su.full_pin.resize(k, nullptr);
-        su.full_readers.resize(k);
+        su.full_stripes.resize(k);
         for (int s = 0; s < k; s++) {
             su.in_bases[s] = s < su.k_data ? job->input_bases[s]
                                            : job->tombstone_source_bases[s - su.k_data];
             su.comp_file_sz[s] = file_size_of(su.in_bases[s] + "-Data.db");
             su.full_pin[s] = (uint8_t*)g_pin_in[0][s].get(su.comp_file_sz[s] ? su.comp_file_sz[s] : 1);
             if (!su.full_pin[s]) throw std::runtime_error("pinned alloc failed");
-            su.full_readers[s] = std::thread([&su, s] {
-                read_file_range(su.in_bases[s] + "-Data.db", su.full_pin[s], 0,
-                                su.comp_file_sz[s], 4);
-            });
+            auto& sr = su.full_stripes[s];
+            const int NS_STRIPE = 6;
+            uint64_t per = (su.comp_file_sz[s] + NS_STRIPE - 1) / NS_STRIPE;
+            for (int t = 0; t < NS_STRIPE; t++) {
+                uint64_t o = (uint64_t)t * per;
+                if (o >= su.comp_file_sz[s]) break;
+                uint64_t len = std::min<uint64_t>(per, su.comp_file_sz[s] - o);
+                sr.off.push_back(o);
+                sr.len.push_back(len);
+                sr.th.emplace_back([&su, s, o, len] {
+                    read_file_range(su.in_bases[s] + "-Data.db", su.full_pin[s] + o, o, len, 1);
+                });
+            }
         }
     }
     {
