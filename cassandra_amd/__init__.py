@@ -42,6 +42,7 @@ class GpucJob(ctypes.Structure):
         ("n_tomb_sources", ctypes.c_int32),
         ("cell_level_gc", ctypes.c_int32),
         ("n_output_shards", ctypes.c_int32),
+        ("cancel_flag", ctypes.POINTER(ctypes.c_int32)),
     ]
 
 
@@ -235,6 +236,7 @@ def compact(
     token_range=None,
     device=0,
     n_output_shards=1,
+    cancel_flag=None,
     tombstone_sources=None,
     cell_level_gc=False,
     keep_ranges=None,
@@ -275,6 +277,8 @@ def compact(
         job.invert_ranges = 1 if invert_ranges else 0
     job.device = device
     job.n_output_shards = n_output_shards
+    if cancel_flag is not None:
+        job.cancel_flag = cancel_flag   # ctypes pointer to c_int32; poll-based cancel
     if tombstone_sources:
         tarr = (ctypes.c_char_p * len(tombstone_sources))(*[b.encode() for b in tombstone_sources])
         job.tombstone_source_bases = tarr

@@ -1147,6 +1147,15 @@ struct CompactSetup {
 // range in the unsharded case). wslot selects the pinned-arena set; shards
 // run two at a time so front-phase kernels of one shard overlap the other
 // shard's LDS-bound compress (they use disjoint CU resources).
+struct cancelled_error : std::runtime_error {
+    cancelled_error() : std::runtime_error("compaction cancelled (cancel_flag set)") {}
+};
+// cooperative cancellation poll (CompactionIterator.isStopRequested):
+// called between pipeline phases of a task
+static inline void check_cancel(const gpuc_job* job) {
+    if (job->cancel_flag && *job->cancel_flag) throw cancelled_error();
+}
+
 static void compact_one(const gpuc_job* job, const CompactSetup& su,
                         const std::vector<std::pair<uint32_t, uint32_t>>& pr,
                         const std::string& out_base_str, int wslot,
@@ -1214,6 +1223,7 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
             res->ms_read_io += wall() - t0;
         }
         TR("window resolved");
+        check_cancel(job);
 
         // ---- H2D + decompress (pipelined per sstable) ----
         hipStream_t copy_stream;
@@ -1395,6 +1405,7 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
             unsigned long long err = 0;
             HIP_CHECK(hipStreamSynchronize(stream));
             TR("parse synced");
+        check_cancel(job);
             HIP_CHECK(hipMemcpy(&err, d_error.p, 8, hipMemcpyDeviceToHost));
             if (err) throw std::runtime_error("GPU decode/parse error code " + std::to_string(err));
             uint64_t rows_in_local = 0;
@@ -1446,6 +1457,7 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         HIP_CHECK(hipStreamSynchronize(stream));
         HIP_CHECK(hipMemcpy(&n_groups, d_ngroups.p, 8, hipMemcpyDeviceToHost));
         TR("groups known");
+        check_cancel(job);
 
         // ---- reconcile + purge ----
         DevBuf d_srcbases, d_group_rows;
@@ -1939,10 +1951,17 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
             w0.join();
             w1.join();
             for (auto& e : werr)
-                if (!e.empty()) throw std::runtime_error(e);
+                if (!e.empty()) {
+                    if (e.find("cancel_flag set") != std::string::npos) throw cancelled_error();
+                    throw std::runtime_error(e);
+                }
         }
         res->ms_total = wall() - t_start_all;
         return GPUC_OK;
+    } catch (const cancelled_error& e) {
+        set_err(res->error, sizeof(res->error), e.what());
+        res->ms_total = wall() - t_start_all;
+        return GPUC_ERR_CANCELLED;
     } catch (const std::exception& e) {
         set_err(res->error, sizeof(res->error), e.what());
         res->ms_total = wall() - t_start_all;

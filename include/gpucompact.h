@@ -47,6 +47,7 @@ extern "C" {
 #define GPUC_ERR_NO_GPU 4
 #define GPUC_ERR_HIP 5
 #define GPUC_ERR_INTERNAL 6
+#define GPUC_ERR_CANCELLED 7     /* cancel_flag observed set (CompactionInterruptedException) */
 
 typedef struct gpuc_purge_range {
     int64_t token_lo;   /* inclusive Murmur3 token bounds */
@@ -100,6 +101,14 @@ typedef struct gpuc_job {
        compaction writer. Two ranges are processed concurrently so front-phase
        kernels overlap the LDS-bound compressor. */
     int32_t n_output_shards;
+
+    /* optional cooperative cancellation (CompactionIterator.isStopRequested,
+       CompactionIterator.java:721-740): when non-NULL, polled between
+       pipeline phases; a non-zero value aborts the task with
+       GPUC_ERR_CANCELLED. Output files may be partially written — the host
+       discards them exactly as it discards tmp files of an interrupted Java
+       compaction (LifecycleTransaction abort). */
+    const volatile int32_t* cancel_flag;
 } gpuc_job;
 
 typedef struct gpuc_result {

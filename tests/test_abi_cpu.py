@@ -35,5 +35,6 @@ def test_no_gpu_fails_loudly(product_lib):
 def test_header_matches_wrapper():
     hdr = open(os.path.join(REPO, "include", "gpucompact.h")).read()
     for field in ["input_uncompressed_bytes", "merged_counts", "dominant_kernel",
-                  "token_lo", "n_overlaps", "gc_before"]:
+                  "token_lo", "n_overlaps", "gc_before", "cancel_flag",
+                  "GPUC_ERR_CANCELLED"]:
         assert field in hdr

@@ -462,6 +462,26 @@ def test_compaction_associativity_gpu(ca, oracle_bin, tmp_path):
     _assert_dirs_equal(f"{d}/dj/oa-60-big", f"{d}/dj/oa-70-big")
 
 
+def test_cancellation(ca, oracle_bin, tmp_path):
+    """Cooperative cancel (CompactionIterator.isStopRequested): a set
+    cancel_flag aborts the task with GPUC_ERR_CANCELLED; a zero flag is
+    inert and the output still matches the oracle byte-for-byte."""
+    import ctypes
+    d = str(tmp_path)
+    _oracle_gen(d, seed=401, n=2, rows=800, vlen=100, overlap=20)
+    ins = [f"{d}/oa-1-big", f"{d}/oa-2-big"]
+    flag = ctypes.c_int32(1)
+    try:
+        ca.compact(ins, f"{d}/oa-90-big", cancel_flag=ctypes.byref(flag))
+        raise AssertionError("cancelled compaction did not fail")
+    except ca.GpuCompactError as e:
+        assert "rc=7" in str(e) and "cancel" in str(e), e
+    flag.value = 0
+    ca.compact(ins, f"{d}/oa-91-big", cancel_flag=ctypes.byref(flag))
+    _oracle_compact(f"{d}/oa-92-big", ins)
+    _assert_dirs_equal(f"{d}/oa-92-big", f"{d}/oa-91-big")
+
+
 def test_empty_outputs(ca, oracle_bin, tmp_path):
     """Degenerate outputs stay byte-identical: (1) a compaction whose every
     row is purged writes the oracle's empty sstable; (2) a scrub where every
