@@ -278,7 +278,9 @@ def compact(
     job.device = device
     job.n_output_shards = n_output_shards
     if cancel_flag is not None:
-        job.cancel_flag = cancel_flag   # ctypes pointer to c_int32; poll-based cancel
+        if isinstance(cancel_flag, ctypes.c_int32):
+            cancel_flag = ctypes.pointer(cancel_flag)
+        job.cancel_flag = ctypes.cast(cancel_flag, ctypes.POINTER(ctypes.c_int32))
     if tombstone_sources:
         tarr = (ctypes.c_char_p * len(tombstone_sources))(*[b.encode() for b in tombstone_sources])
         job.tombstone_source_bases = tarr

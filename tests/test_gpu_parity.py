@@ -472,12 +472,12 @@ def test_cancellation(ca, oracle_bin, tmp_path):
     ins = [f"{d}/oa-1-big", f"{d}/oa-2-big"]
     flag = ctypes.c_int32(1)
     try:
-        ca.compact(ins, f"{d}/oa-90-big", cancel_flag=ctypes.byref(flag))
+        ca.compact(ins, f"{d}/oa-90-big", cancel_flag=ctypes.pointer(flag))
         raise AssertionError("cancelled compaction did not fail")
     except ca.GpuCompactError as e:
         assert "rc=7" in str(e) and "cancel" in str(e), e
     flag.value = 0
-    ca.compact(ins, f"{d}/oa-91-big", cancel_flag=ctypes.byref(flag))
+    ca.compact(ins, f"{d}/oa-91-big", cancel_flag=ctypes.pointer(flag))
     _oracle_compact(f"{d}/oa-92-big", ins)
     _assert_dirs_equal(f"{d}/oa-92-big", f"{d}/oa-91-big")
 
