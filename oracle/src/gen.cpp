@@ -6,6 +6,7 @@ namespace oracle {
 
 SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
     SSTable t;
+    if (g.snappy) t.comp.algo = Compressor::SNAPPY;
     t.generation = g.first_generation + sst;
     t.header.key_type = g.key_len > 8 ? CqlType::BYTES : CqlType::LONG;  // pk bigint / blob
     if (g.clustering_rows > 0) {

@@ -3,6 +3,7 @@
 // of the serializers cited per function.
 #include "sstable.h"
 #include "lz4_ref.h"
+#include "snappy_ref.h"
 #include <algorithm>
 #include <cassert>
 #include <cmath>
@@ -392,20 +393,33 @@ void serialize_partition(const Partition& p, const Header& h, bytes& out, bytes&
 // LZ4Compressor.java:118-134, ChecksumWriter.java:62-104)
 // ---------------------------------------------------------------------------
 ChunkedOut chunk_compress(const bytes& raw, const CompressionParams& cp) {
-    if (cp.algo != Compressor::LZ4) throw std::runtime_error("only LZ4 in round 1");
+    if (cp.algo != Compressor::LZ4 && cp.algo != Compressor::SNAPPY)
+        throw std::runtime_error("unsupported compressor");
     ChunkedOut co;
     size_t nchunks = (raw.size() + cp.chunk_len - 1) / cp.chunk_len;
-    bytes tmp(4 + LZ4_compressBound(cp.chunk_len));
+    bytes tmp(4 + std::max<size_t>(LZ4_compressBound(cp.chunk_len),
+                                   snappy_ref_max_compressed_length(cp.chunk_len)));
     for (size_t i = 0; i < nchunks; i++) {
         size_t off = i * cp.chunk_len;
         size_t len = std::min((size_t)cp.chunk_len, raw.size() - off);
+        uint32_t total;
+        if (cp.algo == Compressor::SNAPPY) {
+            // SnappyCompressor.compress (SnappyCompressor.java:82-86): the
+            // chunk payload is one raw snappy block (snappy carries its own
+            // leading uncompressed-length varint; no extra header)
+            size_t csz = tmp.size();
+            if (!snappy_ref_compress((const char*)raw.data() + off, len, (char*)tmp.data(), &csz))
+                throw std::runtime_error("snappy compress failed");
+            total = (uint32_t)csz;
+        } else {
         // 4-byte LITTLE-endian uncompressed length + raw LZ4 block
         tmp[0] = (uint8_t)len; tmp[1] = (uint8_t)(len >> 8);
         tmp[2] = (uint8_t)(len >> 16); tmp[3] = (uint8_t)(len >> 24);
         int csz = LZ4_compress_default((const char*)raw.data() + off, (char*)tmp.data() + 4,
                                        (int)len, (int)tmp.size() - 4);
         if (csz <= 0) throw std::runtime_error("LZ4_compress_default failed");
-        uint32_t total = (uint32_t)csz + 4;
+        total = (uint32_t)csz + 4;
+        }
         if (total >= cp.max_compressed_len)
             throw std::runtime_error("store-uncompressed fallback unsupported (default params never hit it)");
         co.offsets.push_back(co.file.size());
@@ -816,7 +830,16 @@ static bytes decompress_data(const bytes& file, const CompressionInfo& ci) {
                                           (int)(comp_len - 4), (int)ulen);
             if (got != (int)ulen) throw std::runtime_error("LZ4 decode failed");
         } else {
-            throw std::runtime_error("snappy read unsupported in round 1");
+            // raw snappy block: decoded length from its own varint header
+            size_t ulen = 0;
+            if (!snappy_ref_uncompressed_length((const char*)file.data() + off, comp_len, &ulen) ||
+                ulen != want)
+                throw std::runtime_error("snappy chunk length mismatch");
+            size_t prev = out.size();
+            out.resize(prev + ulen);
+            if (!snappy_ref_uncompress((const char*)file.data() + off, comp_len,
+                                       (char*)out.data() + prev))
+                throw std::runtime_error("snappy decode failed");
         }
     }
     if (out.size() != ci.data_len) throw std::runtime_error("data length mismatch");
@@ -958,6 +981,15 @@ ScrubResult scrub_sstable(const std::string& inbase, const std::string& outbase)
         uint32_t crc_stored = 0;
         for (int k = 0; k < 4; k++) crc_stored = (crc_stored << 8) | data_b[off + comp_len + k];
         if (crc_stored != crc32(data_b.data() + off, comp_len)) { bad[i] = 1; continue; }
+        if (ci.params.algo == Compressor::SNAPPY) {
+            size_t ulen = 0;
+            if (!snappy_ref_uncompressed_length((const char*)data_b.data() + off, comp_len, &ulen) ||
+                ulen != want ||
+                !snappy_ref_uncompress((const char*)data_b.data() + off, comp_len,
+                                       (char*)raw.data() + i * (uint64_t)ci.params.chunk_len))
+                bad[i] = 1;
+            continue;
+        }
         uint32_t ulen = data_b[off] | (data_b[off + 1] << 8) | (data_b[off + 2] << 16) |
                         ((uint32_t)data_b[off + 3] << 24);
         if (ulen != want) { bad[i] = 1; continue; }

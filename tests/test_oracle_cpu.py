@@ -243,3 +243,46 @@ def test_compaction_associativity(oracle_bin, tmp_path):
             a = open(f"{sub}/oa-60-big-{c}", "rb").read()
             b = open(f"{sub}/oa-70-big-{c}", "rb").read()
             assert a == b, f"associativity broken in {c} (seed {seed})"
+
+
+def test_snappy_sstables(oracle_bin, tmp_path):
+    """C3 shape (SnappyCompressor chunks): gen -> compact -> dump round-trips,
+    output bytes are stable, and scrub recovers around a corrupt chunk. The
+    chunk payload is one raw snappy block (SnappyCompressor.java:82-86, no
+    LZ4-style 4-byte length header); compressed bytes are pinned to the
+    container's libsnappy 1.1.8 (BASELINE.md caveat: the reference bundles
+    1.1.10; the format is stable and the reference's own CompressorTest pins
+    decompress-equality only)."""
+    import json
+    import subprocess
+    d = str(tmp_path)
+    subprocess.run([ORACLE, "gen", d, "seed=7", "n=3", "rows=1500", "vlen=400",
+                    "overlap=25", "tomb=10", "snappy=1"], check=True, capture_output=True)
+    ins = [f"{d}/oa-{g}-big" for g in (1, 2, 3)]
+    subprocess.run([ORACLE, "compact", f"{d}/oa-90-big", *ins], check=True,
+                   capture_output=True)
+    subprocess.run([ORACLE, "compact", f"{d}/oa-95-big", *ins], check=True,
+                   capture_output=True)
+    for c in ["Data.db", "CompressionInfo.db", "Digest.crc32", "Statistics.db"]:
+        assert open(f"{d}/oa-90-big-{c}", "rb").read() ==                open(f"{d}/oa-95-big-{c}", "rb").read(), c
+    out = subprocess.run([ORACLE, "dump", f"{d}/oa-90-big"], capture_output=True,
+                         text=True, check=True)
+    assert "partitions=3375" in out.stdout
+    assert b"SnappyCompressor" in open(f"{d}/oa-90-big-CompressionInfo.db", "rb").read()
+    # snappy payload has no LZ4 4-byte LE length header: first chunk must not
+    # start with (chunk_len & 0xFF)-style little-endian 16384
+    first = open(f"{d}/oa-90-big-Data.db", "rb").read(4)
+    assert first != b"\x00\x40\x00\x00"
+    # scrub around a flipped byte
+    with open(f"{d}/oa-1-big-Data.db", "r+b") as f:
+        f.seek(os.path.getsize(f"{d}/oa-1-big-Data.db") // 2)
+        b0 = f.read(1)
+        f.seek(-1, 1)
+        f.write(bytes([b0[0] ^ 0x20]))
+    out = subprocess.run([ORACLE, "scrub", f"{d}/oa-80-big", f"{d}/oa-1-big"],
+                         capture_output=True, text=True, check=True)
+    sr = json.loads(out.stdout.splitlines()[-1])
+    assert sr["partitions_dropped"] > 0 and sr["partitions_kept"] > 0
+    out = subprocess.run([ORACLE, "dump", f"{d}/oa-80-big"], capture_output=True,
+                         text=True, check=True)
+    assert f"partitions={sr['partitions_kept']}" in out.stdout
