@@ -15,22 +15,28 @@
 #include <cstdint>
 #include <cstring>
 
+#if defined(__HIPCC__)
+#define SNP_HD __host__ __device__
+#else
+#define SNP_HD
+#endif
+
 namespace gpuc {
 
 constexpr uint32_t SNP_BLOCK_LOG = 16;              // kBlockLog (64 KiB fragments)
 constexpr uint32_t SNP_MAX_TABLE = 1u << 14;        // kMaxHashTableSize entries
 constexpr uint32_t SNP_INPUT_MARGIN = 15;           // kInputMarginBytes
 
-inline uint32_t snp_load32(const uint8_t* p) { uint32_t v; memcpy(&v, p, 4); return v; }
-inline uint64_t snp_load64(const uint8_t* p) { uint64_t v; memcpy(&v, p, 8); return v; }
+SNP_HD inline uint32_t snp_load32(const uint8_t* p) { uint32_t v; memcpy(&v, p, 4); return v; }
+SNP_HD inline uint64_t snp_load64(const uint8_t* p) { uint64_t v; memcpy(&v, p, 8); return v; }
 
-inline uint32_t snp_hash(uint32_t bytes, int shift) {
+SNP_HD inline uint32_t snp_hash(uint32_t bytes, int shift) {
     return (bytes * 0x1e35a7bdu) >> shift;
 }
 
 // smallest power of two >= max(16, min(input, kMaxHashTableSize)) — snappy.cc
 // Alloc/WorkingMemory::GetHashTable semantics
-inline uint32_t snp_table_size(uint32_t input_size) {
+SNP_HD inline uint32_t snp_table_size(uint32_t input_size) {
     uint32_t htsize = 256;
     while (htsize < SNP_MAX_TABLE && htsize < input_size) htsize <<= 1;
     return htsize;
