@@ -179,5 +179,6 @@ struct OutStats {
 };
 
 constexpr uint32_t LZ4_SLOT = 16480;  // > 4 + LZ4_compressBound(16384), 16B aligned
+constexpr uint32_t SNP_SLOT = 19456;  // > snappy_max_compressed_length(16384)=19146, 16B aligned
 
 }  // namespace gpuc

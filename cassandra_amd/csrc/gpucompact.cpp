@@ -187,6 +187,7 @@ struct HCompressionInfo {
     uint32_t chunk_len = 16384;
     uint32_t max_compressed = 0x7FFFFFFF;
     uint64_t data_len = 0;
+    bool snappy = false;  // SnappyCompressor (else LZ4Compressor)
     std::vector<uint64_t> offsets;
 };
 static HCompressionInfo parse_compression_info(const bytes& b) {
@@ -194,10 +195,12 @@ static HCompressionInfo parse_compression_info(const bytes& b) {
     uint16_t nlen = r.be16();
     bytes name = r.take(nlen);
     std::string algo((char*)name.data(), name.size());
-    if (algo != "LZ4Compressor") throw std::runtime_error("unsupported compressor " + algo + " (GPU path: LZ4 only)");
+    if (algo != "LZ4Compressor" && algo != "SnappyCompressor")
+        throw std::runtime_error("unsupported compressor " + algo + " (GPU path: LZ4/Snappy)");
     uint32_t opts = r.be32();
     for (uint32_t i = 0; i < opts; i++) { r.take(r.be16()); r.take(r.be16()); }
     HCompressionInfo ci;
+    ci.snappy = algo == "SnappyCompressor";
     ci.chunk_len = r.be32();
     ci.max_compressed = r.be32();
     ci.data_len = r.be64();
@@ -631,6 +634,9 @@ static void init_outstats(DevBuf& d_stats, hipStream_t stream) {
 static bool g_crc_tables_ready = false;
 static void* g_crc256 = nullptr;  // 8x256 sliced CRC tables (first 256 = base)
 static std::mutex g_crc_init_mu;
+static uint32_t* g_snp_off = nullptr;   // snappy probe-offset table (device)
+static uint32_t g_snp_off_n = 0;
+
 static void ensure_crc_tables(hipStream_t stream) {
     std::lock_guard<std::mutex> crc_g(g_crc_init_mu);
     if (g_crc_tables_ready) return;
@@ -642,6 +648,19 @@ static void ensure_crc_tables(hipStream_t stream) {
     crc32_make_table8(tab8);
     HIP_CHECK(hipMalloc(&g_crc256, sizeof(tab8)));
     HIP_CHECK(hipMemcpyAsync(g_crc256, tab8, sizeof(tab8), hipMemcpyHostToDevice, stream));
+    {   // snappy probe offsets (snappy.cc skip heuristic: skip=32, inc=skip>>5)
+        std::vector<uint32_t> off{0};
+        uint32_t skip = 32, o = 0;
+        while (o < (1u << 16) + 128) {
+            uint32_t inc = skip >> 5;
+            skip += inc;
+            o += inc;
+            off.push_back(o);
+        }
+        g_snp_off_n = (uint32_t)off.size();
+        HIP_CHECK(hipMalloc(&g_snp_off, off.size() * 4));
+        HIP_CHECK(hipMemcpy(g_snp_off, off.data(), off.size() * 4, hipMemcpyHostToDevice));
+    }
     // verify the LDS same-address write-order rule the wave compressor relies on
     unsigned int* d_probe;
     HIP_CHECK(hipMalloc(&d_probe, 8));
@@ -737,8 +756,10 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
                                            const std::vector<std::string>& ck_types,
                                            const std::vector<std::pair<bytes, std::string>>& regular_cols,
                                            const std::vector<std::pair<bytes, std::string>>& static_cols,
-                                           hipStream_t stream, int wslot = 0) {
+                                           hipStream_t stream, int wslot = 0,
+                                           bool snappy_out = false) {
     WriteDeviceOut w;
+    const uint32_t SLOT_STRIDE = snappy_out ? SNP_SLOT : LZ4_SLOT;
     TR("wsd: enter");
     static const std::vector<int64_t> ps_off_h = est_hist_offsets(155);
     static const std::vector<int64_t> ch_off_h = est_hist_offsets(118);
@@ -824,7 +845,7 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
     const uint32_t SLAB = 32768;  // 512 MiB uncompressed per slab
     uint32_t n_slabs = n_chunks ? (n_chunks + SLAB - 1) / SLAB : 0;
     DevBuf d_slots, d_csize, d_ccrc;
-    d_slots.alloc((uint64_t)n_chunks * LZ4_SLOT + 16);
+    d_slots.alloc((uint64_t)n_chunks * SLOT_STRIDE + 16);
     d_csize.alloc((uint64_t)n_chunks * 4 + 16);
     d_ccrc.alloc((uint64_t)n_chunks * 4 + 16);
     hipStream_t cstream;
@@ -841,7 +862,15 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
         uint32_t cb = i * SLAB, m = std::min(SLAB, n_chunks - cb);
         // all kernel arguments shift uniformly per chunk, so a slab launch is
         // just base-offset pointers with a local chunk count
-        hipLaunchKernelGGL(k_lz4_compress_wave_t<false>, dim3(m), dim3(WAVE), 0, stream,
+        if (snappy_out)
+            hipLaunchKernelGGL(k_snappy_compress_chunks, dim3(m), dim3(WAVE), 0, stream,
+                               d_out_data.as<uint8_t>() + (uint64_t)cb * CHUNK_LEN,
+                               total_unc - (uint64_t)cb * CHUNK_LEN,
+                               d_slots.as<uint8_t>() + (uint64_t)cb * SNP_SLOT,
+                               d_csize.as<uint32_t>() + cb, d_ccrc.as<uint32_t>() + cb, m,
+                               (const uint32_t*)g_crc256, g_snp_off, g_snp_off_n);
+        else
+            hipLaunchKernelGGL(k_lz4_compress_wave_t<false>, dim3(m), dim3(WAVE), 0, stream,
                            d_out_data.as<uint8_t>() + (uint64_t)cb * CHUNK_LEN,
                            total_unc - (uint64_t)cb * CHUNK_LEN,
                            d_slots.as<uint8_t>() + (uint64_t)cb * LZ4_SLOT,
@@ -859,7 +888,7 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
         if (!f) throw std::runtime_error("cannot create " + data_path);
         fclose(f);
     }
-    const uint64_t worst_slab = (uint64_t)SLAB * (LZ4_SLOT + 4);
+    const uint64_t worst_slab = (uint64_t)SLAB * ((uint64_t)SLOT_STRIDE + 4);
     const int NSLOTS = 3;
     uint8_t* h_slab0 = (uint8_t*)g_pin_out[wslot][0].get(worst_slab * NSLOTS);
     if (!h_slab0) throw std::runtime_error("pinned out alloc failed");
@@ -891,9 +920,10 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
         HIP_CHECK(hipMemcpyAsync(d_foff[i & 1].p, fo.data(), (uint64_t)m * 8,
                                  hipMemcpyHostToDevice, cstream));
         hipLaunchKernelGGL(k_chunk_gather, dim3(m), dim3(WAVE), 0, cstream,
-                           d_slots.as<uint8_t>() + (uint64_t)cb * LZ4_SLOT,
+                           d_slots.as<uint8_t>() + (uint64_t)cb * SLOT_STRIDE,
                            d_csize.as<uint32_t>() + cb, d_ccrc.as<uint32_t>() + cb,
-                           d_foff[i & 1].as<uint64_t>(), d_gat[i & 1].as<uint8_t>(), m);
+                           d_foff[i & 1].as<uint64_t>(), d_gat[i & 1].as<uint8_t>(), m,
+                           SLOT_STRIDE);
         int slot = (int)(i % NSLOTS);
         if (wfut[slot].valid()) wfut[slot].get();  // pinned buffer free again
         uint8_t* hbuf = h_slab0 + (uint64_t)slot * worst_slab;
@@ -1034,7 +1064,7 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
         }
         {
             bytes ci;
-            std::string algo = "LZ4Compressor";
+            std::string algo = snappy_out ? "SnappyCompressor" : "LZ4Compressor";
             put_be16(ci, (uint16_t)algo.size());
             ci.insert(ci.end(), algo.begin(), algo.end());
             put_be32(ci, 0);
@@ -1302,10 +1332,18 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
             HIP_CHECK(hipEventCreate(&ev_h2d[s]));
             HIP_CHECK(hipEventRecord(ev_h2d[s], copy_stream));
             HIP_CHECK(hipStreamWaitEvent(stream, ev_h2d[s], 0));
-            if (!chunks.empty())
-                hipLaunchKernelGGL(k_lz4_decompress_wave, dim3(lz4_decomp_grid((uint32_t)chunks.size(), 1)), dim3(WAVE), 0,
-                                   stream, d_chunks_s[s].as<ChunkDesc>(), (uint32_t)chunks.size(), 1,
-                                   d_error.as<unsigned long long>(), (const uint32_t*)g_crc256);
+            if (!chunks.empty()) {
+                if (ci.snappy)
+                    hipLaunchKernelGGL(k_snappy_decompress_chunks,
+                                       dim3(lz4_decomp_grid((uint32_t)chunks.size(), 1)), dim3(WAVE), 0,
+                                       stream, d_chunks_s[s].as<ChunkDesc>(), (uint32_t)chunks.size(), 1,
+                                       d_error.as<unsigned long long>(), (const uint32_t*)g_crc256,
+                                       (uint8_t*)nullptr);
+                else
+                    hipLaunchKernelGGL(k_lz4_decompress_wave, dim3(lz4_decomp_grid((uint32_t)chunks.size(), 1)), dim3(WAVE), 0,
+                                       stream, d_chunks_s[s].as<ChunkDesc>(), (uint32_t)chunks.size(), 1,
+                                       d_error.as<unsigned long long>(), (const uint32_t*)g_crc256);
+            }
         }
         {
             std::lock_guard<std::mutex> g(res_mu);
@@ -1681,7 +1719,8 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         WriteDeviceOut w = write_sstable_device(opb, *rows_for_writer, n_groups, sp2, d_stats, d_tomb,
                                                 tomb_cap, out_base_str, stats[0].key_type,
                                                 stats[0].clustering_types, stats[0].regular_cols,
-                                                stats[0].static_cols, stream, wslot);
+                                                stats[0].static_cols, stream, wslot,
+                                                su.cinfos[0].snappy);
         TR("writer done");
         {
             OutStats hst;
@@ -1838,6 +1877,9 @@ static void compact_setup(const gpuc_job* job, CompactSetup& su, bool preread_fu
         for (int s = 0; s < k; s++)
             if (!perr[s].empty()) throw std::runtime_error("Index.db parse: " + perr[s]);
     }
+    for (int s = 1; s < k; s++)
+        if (su.cinfos[s].snappy != su.cinfos[0].snappy)
+            throw std::runtime_error("mixed compressors across inputs unsupported");
 }
 
 // Murmur3 token of the key in Index.db entry at byte offset `off`
@@ -2042,7 +2084,14 @@ extern "C" int gpuc_verify(const char* input_base, int32_t device, char* error,
         d_error.alloc(8);
         HIP_CHECK(hipMemsetAsync(d_error.p, 0, 8, stream));
         if (!chunks.empty())
-            hipLaunchKernelGGL(k_lz4_decompress_wave, dim3(lz4_decomp_grid((uint32_t)chunks.size(), 1)), dim3(WAVE), 0,
+            if (ci.snappy)
+                hipLaunchKernelGGL(k_snappy_decompress_chunks,
+                                   dim3(lz4_decomp_grid((uint32_t)chunks.size(), 1)), dim3(WAVE), 0,
+                                   stream, d_chunks.as<ChunkDesc>(), (uint32_t)chunks.size(), 1,
+                                   d_error.as<unsigned long long>(), (const uint32_t*)g_crc256,
+                                   (uint8_t*)nullptr);
+            else
+                hipLaunchKernelGGL(k_lz4_decompress_wave, dim3(lz4_decomp_grid((uint32_t)chunks.size(), 1)), dim3(WAVE), 0,
                                stream, d_chunks.as<ChunkDesc>(), (uint32_t)chunks.size(), 1,
                                d_error.as<unsigned long long>(), (const uint32_t*)g_crc256);
         {
@@ -2203,7 +2252,14 @@ extern "C" int gpuc_scrub(const char* input_base, const char* output_base, int32
         HIP_CHECK(hipMemcpyAsync(d_chunksb.p, chunks.data(), chunks.size() * sizeof(ChunkDesc),
                                  hipMemcpyHostToDevice, stream));
         if (!chunks.empty())
-            hipLaunchKernelGGL(k_lz4_decompress_wave, dim3(lz4_decomp_grid((uint32_t)chunks.size(), 1)), dim3(WAVE), 0,
+            if (ci.snappy)
+                hipLaunchKernelGGL(k_snappy_decompress_chunks,
+                                   dim3(lz4_decomp_grid((uint32_t)chunks.size(), 1)), dim3(WAVE), 0,
+                                   stream, d_chunksb.as<ChunkDesc>(), (uint32_t)chunks.size(), 1,
+                                   d_error.as<unsigned long long>(), (const uint32_t*)g_crc256,
+                                   d_bad.as<uint8_t>());
+            else
+                hipLaunchKernelGGL(k_lz4_decompress_wave, dim3(lz4_decomp_grid((uint32_t)chunks.size(), 1)), dim3(WAVE), 0,
                                stream, d_chunksb.as<ChunkDesc>(), (uint32_t)chunks.size(), 1,
                                d_error.as<unsigned long long>(), (const uint32_t*)g_crc256,
                                d_bad.as<uint8_t>());
@@ -2369,7 +2425,7 @@ extern "C" int gpuc_scrub(const char* input_base, const char* output_base, int32
         sp2.hs.min_ttl = st.min_ttl == INT32_MAX ? 0 : st.min_ttl;
         write_sstable_device(opb, out_rows, n_groups, sp2, d_stats, d_tomb, tomb_cap, output_base,
                              st.key_type, st.clustering_types, st.regular_cols, st.static_cols,
-                             stream, 0);
+                             stream, 0, ci.snappy);
         HIP_CHECK(hipStreamDestroy(stream));
         return GPUC_OK;
     } catch (const std::exception& e) {

@@ -206,12 +206,13 @@ namespace gpuc {
 // ---------------------------------------------------------------------------
 // gather compressed chunks into the final contiguous Data.db image
 __global__ void k_chunk_gather(const uint8_t* slots, const uint32_t* csize, const uint32_t* ccrc,
-                               const uint64_t* foff, uint8_t* out, uint32_t n_chunks) {
+                               const uint64_t* foff, uint8_t* out, uint32_t n_chunks,
+                               uint32_t slot_stride = LZ4_SLOT) {
     uint32_t c = blockIdx.x;
     if (c >= n_chunks) return;
     int lane = threadIdx.x;
     uint32_t sz = csize[c];
-    const uint8_t* src = slots + (uint64_t)c * LZ4_SLOT;
+    const uint8_t* src = slots + (uint64_t)c * slot_stride;
     uint8_t* dst = out + foff[c] + (uint64_t)c * 4;  // +4 per preceding chunk CRC
     for (uint32_t i = 4u * lane; i < sz; i += 4u * WAVE) {
         uint32_t nb = sz - i;
@@ -283,3 +284,4 @@ __global__ void k_merge_uniform(const MRec* in, MRec* out, uint64_t n, uint64_t 
 }  // namespace gpuc
 
 #include "lz4_wave.h"
+#include "snappy_wave.h"

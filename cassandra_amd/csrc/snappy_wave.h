@@ -350,4 +350,162 @@ __global__ void __launch_bounds__(WAVE) k_snappy_decompress_wave(const SnpChunk*
     if (opos != olen) fail(25);
 }
 
+// ---------------------------------------------------------------------------
+// product-pipeline kernels: same slot/csize/ccrc contract as the LZ4 pair in
+// lz4_wave.h (slot stride SNP_SLOT; payload = varint + fragment, CRC32 over
+// the payload; CompressedSequentialWriter framing is codec-agnostic)
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(WAVE) k_snappy_compress_chunks(
+    const uint8_t* data, uint64_t data_len, uint8_t* slots, uint32_t* csize, uint32_t* ccrc,
+    uint32_t n_chunks, const uint32_t* crc_table, const uint32_t* OFF, uint32_t off_n) {
+    __shared__ uint16_t s_table[SNP_MAX_TABLE];
+    uint32_t c = blockIdx.x;
+    if (c >= n_chunks) return;
+    int lane = threadIdx.x;
+    uint64_t off = (uint64_t)c * CHUNK_LEN;
+    uint32_t len = (uint32_t)min((uint64_t)CHUNK_LEN, data_len - off);
+    uint8_t* dst = slots + (uint64_t)c * SNP_SLOT;
+    int csz = snp_wave_compress(data + off, len, dst, s_table, snp_table_size(len), OFF, off_n,
+                                lane);
+    if (lane == 0) {
+        csize[c] = (uint32_t)csz;
+        uint32_t crc = 0xFFFFFFFFu;
+        for (int i = 0; i < csz; i++) crc = crc_table[(crc ^ dst[i]) & 0xFF] ^ (crc >> 8);
+        ccrc[c] = ~crc;
+    }
+}
+
+// decompress on the ingest ChunkDesc contract (mirror of k_lz4_decompress_wave
+// incl. the trailing per-lane CRC blocks; grid via lz4_decomp_grid)
+__global__ void __launch_bounds__(WAVE) k_snappy_decompress_chunks(const ChunkDesc* chunks,
+                                                                   uint32_t n, int verify_crc,
+                                                                   unsigned long long* error,
+                                                                   const uint32_t* crc_table,
+                                                                   uint8_t* bad_chunks = nullptr) {
+    uint32_t c = blockIdx.x;
+    int lane = threadIdx.x;
+    if (c >= n) {
+        uint32_t ci = (c - n) * WAVE + (uint32_t)lane;
+        if (ci >= n) return;
+        ChunkDesc ch = chunks[ci];
+        if (ch.comp_len > SNP_SLOT) return;  // decode block flags it
+        const uint8_t* s_comp = ch.comp;
+        uint32_t crc = 0xFFFFFFFFu;
+        uint32_t i = 0;
+        for (; i + 8 <= ch.comp_len; i += 8) {
+            uint32_t lo, hi;
+            memcpy(&lo, s_comp + i, 4);
+            memcpy(&hi, s_comp + i + 4, 4);
+            lo ^= crc;
+            crc = crc_table[7 * 256 + (lo & 0xFF)] ^ crc_table[6 * 256 + ((lo >> 8) & 0xFF)] ^
+                  crc_table[5 * 256 + ((lo >> 16) & 0xFF)] ^ crc_table[4 * 256 + (lo >> 24)] ^
+                  crc_table[3 * 256 + (hi & 0xFF)] ^ crc_table[2 * 256 + ((hi >> 8) & 0xFF)] ^
+                  crc_table[1 * 256 + ((hi >> 16) & 0xFF)] ^ crc_table[0 * 256 + (hi >> 24)];
+        }
+        for (; i < ch.comp_len; i++) crc = crc_table[(crc ^ s_comp[i]) & 0xFF] ^ (crc >> 8);
+        crc = ~crc;
+        uint32_t stored = ((uint32_t)s_comp[ch.comp_len] << 24) |
+                          ((uint32_t)s_comp[ch.comp_len + 1] << 16) |
+                          ((uint32_t)s_comp[ch.comp_len + 2] << 8) | s_comp[ch.comp_len + 3];
+        if (crc != stored) { if (bad_chunks) bad_chunks[ci] = 1; else atomicExch(error, 1ull); }
+        return;
+    }
+    (void)verify_crc;
+    ChunkDesc ch = chunks[c];
+    if (ch.comp_len > SNP_SLOT) { if (lane == 0) { if (bad_chunks) bad_chunks[c] = 1; else atomicExch(error, 9ull); } return; }
+    const uint8_t* in = ch.comp;
+    uint8_t* out = ch.out;
+    uint32_t ip = 0, iend = ch.comp_len, olen = ch.out_len;
+    auto fail = [&](unsigned long long code) {
+        if (lane == 0) {
+            if (bad_chunks) bad_chunks[c] = 1;
+            else atomicExch(error, code);
+        }
+    };
+    uint32_t hdr = 0;
+    {
+        int sh = 0;
+        while (true) {
+            if (ip >= iend || sh > 28) { fail(2); return; }
+            uint8_t b = in[ip++];
+            hdr |= (uint32_t)(b & 0x7f) << sh;
+            if (!(b & 0x80)) break;
+            sh += 7;
+        }
+    }
+    if (hdr != olen) { fail(2); return; }
+    uint32_t opos = 0;
+    while (ip < iend) {
+        uint8_t tag = in[ip++];
+        uint32_t kind = tag & 3;
+        if (kind == 0) {
+            uint32_t len = (tag >> 2) + 1;
+            if (len > 60) {
+                uint32_t nb = len - 60;
+                if (ip + nb > iend) { fail(3); return; }
+                len = 0;
+                for (uint32_t j = 0; j < nb; j++) len |= (uint32_t)in[ip + j] << (8 * j);
+                len += 1;
+                ip += nb;
+            }
+            if (ip + len > iend || opos + len > olen) { fail(3); return; }
+            for (uint32_t i = 4 * (uint32_t)lane; i < len; i += 4 * WAVE) {
+                uint32_t nb = len - i;
+                if (nb >= 4) {
+                    uint32_t v;
+                    memcpy(&v, in + ip + i, 4);
+                    memcpy(out + opos + i, &v, 4);
+                } else {
+                    for (uint32_t j = 0; j < nb; j++) out[opos + i + j] = in[ip + i + j];
+                }
+            }
+            ip += len;
+            opos += len;
+        } else {
+            uint32_t len, offset;
+            if (kind == 1) {
+                len = ((tag >> 2) & 7) + 4;
+                if (ip >= iend) { fail(4); return; }
+                offset = ((uint32_t)(tag >> 5) << 8) | in[ip++];
+            } else if (kind == 2) {
+                len = (tag >> 2) + 1;
+                if (ip + 2 > iend) { fail(4); return; }
+                offset = in[ip] | ((uint32_t)in[ip + 1] << 8);
+                ip += 2;
+            } else {
+                len = (tag >> 2) + 1;
+                if (ip + 4 > iend) { fail(4); return; }
+                memcpy(&offset, in + ip, 4);
+                ip += 4;
+            }
+            if (offset == 0 || offset > opos || opos + len > olen) { fail(4); return; }
+            const uint8_t* src = out + opos - offset;
+            if (offset == 1) {
+                uint32_t b4 = 0x01010101u * src[0];
+                for (uint32_t i = 4 * (uint32_t)lane; i < len; i += 4 * WAVE) {
+                    uint32_t nb = len - i;
+                    if (nb >= 4) memcpy(out + opos + i, &b4, 4);
+                    else for (uint32_t j = 0; j < nb; j++) out[opos + i + j] = (uint8_t)b4;
+                }
+            } else if (len <= offset) {
+                for (uint32_t i = 4 * (uint32_t)lane; i < len; i += 4 * WAVE) {
+                    uint32_t nb = len - i;
+                    if (nb >= 4) {
+                        uint32_t v;
+                        memcpy(&v, src + i, 4);
+                        memcpy(out + opos + i, &v, 4);
+                    } else {
+                        for (uint32_t j = 0; j < nb; j++) out[opos + i + j] = src[i + j];
+                    }
+                }
+            } else {
+                for (uint32_t i = (uint32_t)lane; i < len; i += WAVE)
+                    out[opos + i] = src[i % offset];
+            }
+            opos += len;
+        }
+    }
+    if (opos != olen) fail(5);
+}
+
 }  // namespace gpuc

@@ -462,6 +462,42 @@ def test_compaction_associativity_gpu(ca, oracle_bin, tmp_path):
     _assert_dirs_equal(f"{d}/dj/oa-60-big", f"{d}/dj/oa-70-big")
 
 
+def test_snappy_pipeline(ca, oracle_bin, tmp_path):
+    """C3 codec end-to-end on GPU: SnappyCompressor sstables (chunk payload =
+    one raw snappy block, SnappyCompressor.java:82-86; wave kernels proven
+    bit-exact vs the 1.1.8 restatement in profiles/r01_snappy_kernels.txt)
+    compact, verify and scrub byte-identically to the oracle."""
+    import json
+    d = str(tmp_path)
+    _oracle_gen(d, seed=7, n=3, rows=1500, vlen=400, overlap=25, tomb=10, snappy=1)
+    ins = [f"{d}/oa-{g}-big" for g in (1, 2, 3)]
+    _oracle_compact(f"{d}/oa-90-big", ins)
+    ca.compact(ins, f"{d}/oa-91-big")
+    _assert_dirs_equal(f"{d}/oa-90-big", f"{d}/oa-91-big")
+    ca.verify(f"{d}/oa-91-big")
+    # wide partitions + statics through the same codec
+    os.makedirs(d + "/w")
+    _oracle_gen(d + "/w", seed=8, n=2, rows=40, crows=50, vlen=200, rtomb=30,
+                cktext=1, statics=40, ncols=3, overlap=20, snappy=1)
+    wins = [f"{d}/w/oa-{g}-big" for g in (1, 2)]
+    _oracle_compact(f"{d}/w/oa-90-big", wins)
+    ca.compact(wins, f"{d}/w/oa-91-big")
+    _assert_dirs_equal(f"{d}/w/oa-90-big", f"{d}/w/oa-91-big")
+    # scrub parity on a corrupt snappy chunk
+    with open(f"{d}/oa-1-big-Data.db", "r+b") as f:
+        f.seek(os.path.getsize(f"{d}/oa-1-big-Data.db") // 2)
+        b0 = f.read(1)
+        f.seek(-1, 1)
+        f.write(bytes([b0[0] ^ 0x11]))
+    out = subprocess.run([ORACLE, "scrub", f"{d}/oa-80-big", f"{d}/oa-1-big"],
+                         capture_output=True, text=True, check=True)
+    ores = json.loads(out.stdout.splitlines()[-1])
+    kept, dropped = ca.scrub(f"{d}/oa-1-big", f"{d}/oa-81-big")
+    assert dropped > 0 and (kept, dropped) == (ores["partitions_kept"],
+                                               ores["partitions_dropped"])
+    _assert_dirs_equal(f"{d}/oa-80-big", f"{d}/oa-81-big")
+
+
 def test_cancellation(ca, oracle_bin, tmp_path):
     """Cooperative cancel (CompactionIterator.isStopRequested): a set
     cancel_flag aborts the task with GPUC_ERR_CANCELLED; a zero flag is
