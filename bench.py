@@ -87,6 +87,8 @@ def main():
                     help="n_output_shards per job (UCS-style sharded outputs; "
                          "2 shards run concurrently inside the library)")
     ap.add_argument("--no-cpu-baseline", action="store_true")
+    ap.add_argument("--snappy", action="store_true",
+                    help="C3 codec: SnappyCompressor chunks end to end")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -116,7 +118,8 @@ def main():
                 value_len=args.vlen, value_repeat_pct=55, device=device,
                 clustering_rows=args.clustering_rows,
                 tombstone_pct=args.tombstone_pct,
-                range_tomb_pct=args.range_tomb_pct)
+                range_tomb_pct=args.range_tomb_pct,
+                snappy=args.snappy)
     t_gen = time.time() - t_gen
     bases = [os.path.join(d, f"oa-{g}-big") for g in range(1, args.sstables + 1)]
     input_compressed = sum(os.path.getsize(b + "-Data.db") for b in bases)
@@ -233,6 +236,8 @@ def main():
                          f"({args.clustering_rows} clustering rows/partition, "
                          f"{args.tombstone_pct}% row + {args.range_tomb_pct}% range tombstones)"
                          if args.clustering_rows else
+                         f"C3: {args.sstables}x sstables, Snappy 16KiB chunks"
+                         if args.snappy else
                          "C2: 8x2GiB sstables, 10% key overlap, ~1KiB values, LZ4 16KiB chunks"),
             "sstables": args.sstables,
             "rows_per_sstable": args.rows,

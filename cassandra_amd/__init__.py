@@ -363,6 +363,7 @@ def generate(
         base_ldt=base_ldt,
         first_generation=first_generation,
         device=device,
+        snappy=1 if snappy else 0,
     )
     err = ctypes.create_string_buffer(256)
     rc = lib.gpuc_generate(ctypes.byref(spec), out_dir.encode(), err, 256)

@@ -2699,7 +2699,7 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
                                  std::vector<std::string>(
                                      gen_nck, gp.ck_text ? "org.apache.cassandra.db.marshal.UTF8Type"
                                                          : "org.apache.cassandra.db.marshal.LongType"),
-                                 cols, scols, stream);
+                                 cols, scols, stream, 0, spec->snappy != 0);
         }
         HIP_CHECK(hipStreamDestroy(stream));
         return GPUC_OK;

@@ -159,6 +159,7 @@ typedef struct gpuc_gen_spec {
     int64_t base_ldt;
     uint64_t first_generation;
     int32_t device;
+    int32_t snappy;             /* 1: SnappyCompressor chunks (C3 shape) */
 } gpuc_gen_spec;
 
 int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* error, size_t error_len);
