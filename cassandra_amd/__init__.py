@@ -96,6 +96,7 @@ class GpucGenSpec(ctypes.Structure):
         ("base_ldt", ctypes.c_int64),
         ("first_generation", ctypes.c_uint64),
         ("device", ctypes.c_int32),
+        ("snappy", ctypes.c_int32),
     ]
 
 
@@ -319,6 +320,7 @@ def compact(
 
 def generate(
     out_dir,
+    snappy=False,
     seed=42,
     n_sstables=4,
     rows_per_sstable=1000,
