@@ -46,6 +46,8 @@ def main():
             "vrep": rng.choice([0, 55, 97]),
             "keylen": rng.choice([8, 8, 12, 24, 120]),
         }
+        if rng.random() < 0.25:
+            gen["snappy"] = 1
         if gen["vlen"] == 0:
             gen["vlen"] = 1
         if gen["vlen"] >= 20000:
