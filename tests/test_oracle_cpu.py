@@ -211,6 +211,27 @@ def test_merge_semantics_laws(oracle_bin, tmp_path):
     assert "merge laws OK" in r.stdout
 
 
+def test_gc_dsl_golden_vectors(tmp_path):
+    """The reference's own GarbageSkipper golden vectors
+    (CompactionIteratorTest.java:93-176, the UnfilteredRowsGenerator DSL
+    cases incl. boundary folding and the equivalence law) transcribed
+    verbatim and run against the oracle merge -> garbage_filter -> purge
+    pipeline."""
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    exe = str(tmp_path / "gc_dsl")
+    subprocess.run(
+        ["g++", "-O2", "-std=c++17",
+         os.path.join(repo, "tests/native/gc_dsl_test.cpp"),
+         os.path.join(repo, "oracle/src/sstable.cpp"),
+         os.path.join(repo, "oracle/src/compact.cpp"),
+         os.path.join(repo, "oracle/src/gen.cpp"),
+         "-o", exe, "-l:liblz4.so.1", "-ldl"],
+        check=True, capture_output=True)
+    r = subprocess.run([exe], capture_output=True, text=True)
+    assert r.returncode == 0, r.stdout
+    assert "all OK" in r.stdout
+
+
 def test_compaction_associativity(oracle_bin, tmp_path):
     """Multi-level compaction converges: compact(compact(A,B),C) is
     byte-identical to compact(A,B,C) when no rows are dropped (disjoint
