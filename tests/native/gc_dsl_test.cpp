@@ -186,6 +186,57 @@ static void equivalence(const std::vector<std::string>& inputs,
     CHECK(items_equal(lhs, rhs), msg.c_str());
 }
 
+
+static uint64_t gc_rnd(uint64_t& st) {
+    st ^= st << 13;
+    st ^= st >> 7;
+    st ^= st << 17;
+    return st;
+}
+
+// random VALID unfiltered stream in the DSL (normalized rows: deletion
+// strictly below liveness — see the note at the call site)
+static std::string gen_stream(uint64_t seed) {
+    uint64_t st = seed * 2654435761u + 1;
+    std::string out;
+    int pos = (int)(gc_rnd(st) % 5);
+    bool open = false;
+    int64_t open_t = 0;
+    for (int i = 0; i < 12 && pos < 200; i++) {
+        int what = (int)(gc_rnd(st) % 3);
+        int64_t t = 100 + (int)(gc_rnd(st) % 100);
+        char buf[64];
+        if (what == 0 && !open) {
+            snprintf(buf, sizeof buf, "%d<%s[%lld] ", pos, gc_rnd(st) & 1 ? "=" : "",
+                     (long long)t);
+            open = true;
+            open_t = t;
+        } else if (what == 1 && open) {
+            snprintf(buf, sizeof buf, "[%lld]<%s%d ", (long long)open_t,
+                     gc_rnd(st) & 1 ? "=" : "", pos);
+            open = false;
+        } else {
+            if (gc_rnd(st) % 4 == 0 && t > 101)
+                snprintf(buf, sizeof buf, "%d[%lld D%lld] ", pos, (long long)t,
+                         (long long)(100 + (int)(gc_rnd(st) % (t - 101))));
+            else
+                snprintf(buf, sizeof buf, "%d[%lld] ", pos, (long long)t);
+        }
+        std::string tokstr(buf);
+        size_t sp = tokstr.find(" D");
+        if (sp != std::string::npos) tokstr.erase(sp, 1);
+        out += tokstr;
+        pos += 1 + (int)(gc_rnd(st) % 9);
+    }
+    if (open) {
+        char buf[64];
+        snprintf(buf, sizeof buf, "[%lld]<%d ", (long long)open_t, pos);
+        out += buf;
+    }
+    if (!out.empty()) out.pop_back();
+    return out;
+}
+
 int main() {
     Header h = dsl_header();
     struct CountCase {
@@ -241,59 +292,6 @@ int main() {
     // through the same pipeline; the equivalence law and output validity
     // (strictly ordered, well-formed marker nesting) must hold for every seed
     {
-        auto rnd = [](uint64_t& st) {
-            st ^= st << 13; st ^= st >> 7; st ^= st << 17;
-            return st;
-        };
-        auto gen_stream = [&](uint64_t seed) {
-            uint64_t st = seed * 2654435761u + 1;
-            std::string out;
-            int pos = (int)(rnd(st) % 5);
-            bool open = false;
-            int64_t open_t = 0;
-            for (int i = 0; i < 12 && pos < 200; i++) {
-                int what = (int)(rnd(st) % 3);
-                int64_t t = 100 + (int)(rnd(st) % 100);
-                char buf[64];
-                if (what == 0 && !open) {
-                    snprintf(buf, sizeof buf, "%d<%s[%lld] ", pos, rnd(st) & 1 ? "=" : "",
-                             (long long)t);
-                    open = true;
-                    open_t = t;
-                } else if (what == 1 && open) {
-                    snprintf(buf, sizeof buf, "[%lld]<%s%d ", (long long)open_t,
-                             rnd(st) & 1 ? "=" : "", pos);
-                    open = false;
-                } else {
-                    // row deletion strictly below the liveness timestamp:
-                    // rows whose own deletion shadows their liveness are
-                    // "denormalized" inputs the reference's generator never
-                    // produces (UnfilteredRowsGenerator.generateSource emits
-                    // liveness-only rows), and for them the equivalence law
-                    // does not hold in the reference either (a k-way merge
-                    // normalizes the shadowed liveness away; a single-version
-                    // passthrough keeps it verbatim)
-                    if (rnd(st) % 4 == 0 && t > 101)
-                        snprintf(buf, sizeof buf, "%d[%lld D%lld] ", pos, (long long)t,
-                                 (long long)(100 + (int)(rnd(st) % (t - 101))));
-                    else
-                        snprintf(buf, sizeof buf, "%d[%lld] ", pos, (long long)t);
-                }
-                // the row-with-deletion form has no space in the DSL
-                std::string tokstr(buf);
-                size_t sp = tokstr.find(" D");
-                if (sp != std::string::npos) tokstr.erase(sp, 1);
-                out += tokstr;
-                pos += 1 + (int)(rnd(st) % 9);
-            }
-            if (open) {
-                char buf[64];
-                snprintf(buf, sizeof buf, "[%lld]<%d ", (long long)open_t, pos);
-                out += buf;
-            }
-            if (!out.empty()) out.pop_back();
-            return out;
-        };
         auto valid = [&](const Partition& p) {
             bool open = false;
             for (auto& u : p.items) {
@@ -335,6 +333,95 @@ int main() {
         snprintf(msg, sizeof msg, "random GC equivalence sweep: 300 seeds (%d bad)", bad);
         CHECK(bad == 0, msg);
         fails += bad ? 1 : 0;
+    }
+    // first-principles merge check (UnfilteredRowIteratorsMergeTest
+    // testTombstoneMerge essence): merge N random streams, then verify the
+    // output against a pointwise timeline oracle — at every sampled
+    // clustering position (integers and midpoints), the merged active
+    // deletion equals the max over sources, and each row survives with
+    // max-liveness iff not shadowed (DeletionTime.deletes: mfda >= ts)
+    {
+        // active deletion of one partition at query q (in half-steps: q=2*ck
+        // is the exact position, q=2*ck+1 is between ck and ck+1)
+        auto active_at = [](const Partition& p, int q) {
+            int64_t best = INT64_MIN;
+            if (!p.del.live()) best = p.del.mfda;
+            int64_t cur = INT64_MIN;
+            for (auto& u : p.items) {
+                if (u.kind != Unfiltered::MARKER) continue;
+                const Marker& mk = u.marker;
+                long long ck = 0;
+                for (int i = 0; i < 8; i++) ck = (ck << 8) | mk.values[0].v[i];
+                int mp = (int)(2 * ck);
+                // evaluate BEFORE considering the marker if q < its bound;
+                // bound position in half-steps per kind
+                int bq;  // first q the post-marker state applies to
+                if (mk.kind == INCL_START) bq = mp;
+                else if (mk.kind == EXCL_START) bq = mp + 1;
+                else if (mk.kind == INCL_END) bq = mp + 1;       // closes after ck
+                else if (mk.kind == EXCL_END) bq = mp;           // closes before ck
+                else if (mk.kind == EXCL_END_INCL_START) bq = mp;
+                else bq = mp + 1;                                // INCL_END_EXCL_START
+                if (q < bq) break;
+                if (mk.boundary()) cur = mk.start_dt.mfda;
+                else if (mk.open(false)) cur = mk.end_dt.mfda;
+                else cur = INT64_MIN;
+            }
+            return std::max(best, cur);
+        };
+        int bad = 0;
+        for (uint64_t seed = 1; seed <= 300 && bad < 5; seed++) {
+            std::vector<std::string> in{gen_stream(seed * 7 + 1), gen_stream(seed * 7 + 2),
+                                        gen_stream(seed * 7 + 3)};
+            std::vector<Partition> ps = parse_all(in);
+            Partition m = merge_all(ps, {}, h);
+            for (int q = 0; q <= 2 * 210 && bad < 5; q++) {
+                int64_t want = INT64_MIN;
+                for (auto& p : ps) want = std::max(want, active_at(p, q));
+                int64_t got = active_at(m, q);
+                if (want != got) {
+                    printf("FAIL seed %llu q=%d timeline: want %lld got %lld\n",
+                           (unsigned long long)seed, q, (long long)want, (long long)got);
+                    bad++;
+                }
+            }
+            // row survival + liveness
+            for (int ck = 0; ck <= 210 && bad < 5; ck++) {
+                int64_t maxts = INT64_MIN, maxdel = INT64_MIN;
+                for (auto& p : ps)
+                    for (auto& u : p.items)
+                        if (u.kind == Unfiltered::ROW) {
+                            long long c = 0;
+                            for (int i = 0; i < 8; i++) c = (c << 8) | u.row.clustering[0].v[i];
+                            if (c != ck) continue;
+                            maxts = std::max(maxts, u.row.live.ts);
+                            if (!u.row.del.live()) maxdel = std::max(maxdel, u.row.del.mfda);
+                        }
+                int64_t act = INT64_MIN;
+                for (auto& p : ps) act = std::max(act, active_at(p, 2 * ck));
+                bool expect = (maxts != INT64_MIN && maxts > act) || maxdel > act;
+                const Unfiltered* got = nullptr;
+                for (auto& u : m.items)
+                    if (u.kind == Unfiltered::ROW) {
+                        long long c = 0;
+                        for (int i = 0; i < 8; i++) c = (c << 8) | u.row.clustering[0].v[i];
+                        if (c == ck) got = &u;
+                    }
+                if (expect != (got != nullptr)) {
+                    printf("FAIL seed %llu ck=%d row survival: want %d\n",
+                           (unsigned long long)seed, ck, (int)expect);
+                    bad++;
+                } else if (got && maxts > act && got->row.live.ts != maxts) {
+                    printf("FAIL seed %llu ck=%d liveness: want %lld got %lld\n",
+                           (unsigned long long)seed, ck, (long long)maxts,
+                           (long long)got->row.live.ts);
+                    bad++;
+                }
+            }
+        }
+        char msg[96];
+        snprintf(msg, sizeof msg, "pointwise merge timeline oracle: 300 seeds (%d bad)", bad);
+        CHECK(bad == 0, msg);
     }
     printf(fails ? "GC DSL vectors: %d FAILURES\n" : "GC DSL vectors: all OK\n", fails);
     return fails ? 1 : 0;
