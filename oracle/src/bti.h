@@ -1,0 +1,50 @@
+// ORACLE — BTI (Big Trie-Indexed, version `da`) index READER: trie node
+// decoding per io/tries/TrieNode.java (ordinals TrieNode.java:945-962, node
+// layouts per BtiFormat.md "Trie nodes") and the Partitions.db / Rows.db
+// container layouts (BtiFormat.md "Partition index"/"Row index",
+// PartitionIndex.java, RowIndexReader.java). Round-2 scaffolding for
+// SURVEY §8(f) rank 2; the writer restatement comes later (docs/bti_notes.md).
+#pragma once
+#include "sstable.h"
+
+namespace oracle {
+
+struct BtiEntry {
+    bytes prefix;        // unique byte-comparable prefix stored in the trie
+    uint8_t hash = 0;    // present iff pb >= 8 (Cassandra 5 files)
+    bool has_hash = false;
+    int64_t idxpos = 0;  // >= 0: Rows.db position; < 0: ~pos in (uncompressed) Data.db
+};
+
+struct BtiPartitionsFile {
+    bytes first_key, last_key;  // short-length-prefixed keys from the footer
+    uint64_t key_count = 0;
+    uint64_t root_pos = 0;
+    std::vector<BtiEntry> entries;  // DFS order == byte order
+};
+
+// parse + fully enumerate a -Partitions.db file
+BtiPartitionsFile read_bti_partitions(const bytes& file);
+
+// one row-index block payload (Rows.db trie leaf)
+struct BtiRowIndexEntry {
+    bytes prefix;
+    uint64_t offset = 0;        // offset within the partition
+    bool has_open = false;      // pb >= 8: deletion active at block start
+    DeletionTime open_dt;
+};
+
+struct BtiRowIndexBlock {
+    bytes partition_key;
+    uint64_t data_pos = 0;      // partition position in (uncompressed) Data.db
+    uint64_t root_pos = 0;
+    uint64_t row_count = 0;
+    DeletionTime partition_del;
+    std::vector<BtiRowIndexEntry> entries;
+};
+
+// parse the row index whose FOOTER ends the structure rooted at `index_pos`
+// (the position a Partitions.db payload with idxpos >= 0 points at)
+BtiRowIndexBlock read_bti_row_index(const bytes& file, uint64_t index_pos);
+
+}  // namespace oracle

@@ -3,6 +3,7 @@
 #include "sstable.h"
 #include "compact.h"
 #include "gen.h"
+#include "bti.h"
 #include <chrono>
 #include <cstring>
 #include <algorithm>
@@ -289,6 +290,37 @@ int main(int argc, char** argv) {
         if (cmd == "selftest") return cmd_selftest();
         if (cmd == "dump") return cmd_dump(pos.at(0));
         if (cmd == "roundtrip") return cmd_roundtrip(pos.at(0));
+        if (cmd == "btidump") {
+            // BTI (da) Partitions.db [+ Rows.db] dump — reader scaffolding
+            bytes pf = read_file(pos.at(0) + "-Partitions.db");
+            BtiPartitionsFile bp = read_bti_partitions(pf);
+            printf("{\"key_count\": %llu, \"root\": %llu, \"entries\": [",
+                   (unsigned long long)bp.key_count, (unsigned long long)bp.root_pos);
+            bytes rf;
+            try { rf = read_file(pos.at(0) + "-Rows.db"); } catch (...) {}
+            for (size_t i = 0; i < bp.entries.size(); i++) {
+                auto& e = bp.entries[i];
+                std::string pfx;
+                for (uint8_t b : e.prefix) { char t[4]; snprintf(t, 4, "%02x", b); pfx += t; }
+                printf("%s{\"prefix\": \"%s\", \"hash\": %d, ", i ? ", " : "",
+                       pfx.c_str(), e.has_hash ? e.hash : -1);
+                if (e.idxpos < 0) {
+                    printf("\"data_pos\": %lld}", (long long)~e.idxpos);
+                } else {
+                    printf("\"rowindex_pos\": %lld", (long long)e.idxpos);
+                    if (!rf.empty()) {
+                        BtiRowIndexBlock rb = read_bti_row_index(rf, (uint64_t)e.idxpos);
+                        printf(", \"data_pos\": %llu, \"rows\": %llu, \"blocks\": %zu",
+                               (unsigned long long)rb.data_pos,
+                               (unsigned long long)rb.row_count, rb.entries.size());
+                    }
+                    printf("}");
+                }
+            }
+            printf("], \"first_key_len\": %zu, \"last_key_len\": %zu}\n",
+                   bp.first_key.size(), bp.last_key.size());
+            return 0;
+        }
         if (cmd == "gen") return cmd_gen(pos.at(0), kv);
         if (cmd == "flush") return cmd_flush(pos.at(0), pos.at(1));
         if (cmd == "scrub") return cmd_scrub(pos.at(0), pos.at(1));

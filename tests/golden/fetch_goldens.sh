@@ -11,4 +11,11 @@ for t in legacy_oa_simple legacy_oa_clust; do
   mkdir -p "$DST/$t"
   cp "$SRC/$t"/oa-1-big-* "$DST/$t/"
 done
-chmod -R u+w "$DST"/legacy_oa_*
+# BTI (version da) fixtures, same provenance (read-verified by the
+# reference's LegacySSTableTest): anchors for the round-2 BTI reader/writer
+SRCD=/root/reference/test/data/legacy-sstables/da/legacy_tables
+for t in legacy_da_simple legacy_da_clust; do
+  mkdir -p "$DST/$t"
+  cp "$SRCD/$t"/da-1-bti-* "$DST/$t/"
+done
+chmod -R u+w "$DST"/legacy_oa_* "$DST"/legacy_da_*
