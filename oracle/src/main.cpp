@@ -310,7 +310,7 @@ int main(int argc, char** argv) {
                     printf("\"rowindex_pos\": %lld", (long long)e.idxpos);
                     if (!rf.empty()) {
                         BtiRowIndexBlock rb = read_bti_row_index(rf, (uint64_t)e.idxpos);
-                        printf(", \"data_pos\": %llu, \"rows\": %llu, \"blocks\": %zu",
+                        printf(", \"data_pos\": %llu, \"index_blocks\": %llu, \"trie_payloads\": %zu",
                                (unsigned long long)rb.data_pos,
                                (unsigned long long)rb.row_count, rb.entries.size());
                     }

@@ -334,7 +334,15 @@ def test_bti_partition_index_reader(oracle_bin):
         # byte-comparable prefixes: 0x40 component marker + key byte '0'..'4'
         assert [e["prefix"] for e in d["entries"]] ==                ["4030", "4031", "4032", "4033", "4034"]
         if want_rows:
-            assert all("rowindex_pos" in e and e["blocks"] > 0 for e in d["entries"])
+            assert all("rowindex_pos" in e and e["trie_payloads"] > 0
+                       for e in d["entries"])
+        # BTI shares the BIG data format (BtiFormat.md): the big-format reader
+        # must decode the da Data.db + Statistics completely
+        out = subprocess.run([ORACLE, "dump", fix], capture_output=True,
+                             text=True, check=True)
+        assert "partitions=5" in out.stdout
+        if want_rows:
+            assert "items=50" in out.stdout
         # decompress Data.db (LZ4 chunk framing) and check keys at positions
         ci = open(fix + "-CompressionInfo.db", "rb").read()
         nlen = struct.unpack(">H", ci[:2])[0]
