@@ -1,5 +1,7 @@
 #include "bti.h"
+#include <algorithm>
 #include <functional>
+#include <memory>
 #include <stdexcept>
 
 namespace oracle {
@@ -191,6 +193,402 @@ BtiRowIndexBlock read_bti_row_index(const bytes& file, uint64_t index_pos) {
         out.entries.push_back(std::move(e));
     });
     return out;
+}
+
+// ---------------------------------------------------------------------------
+// WRITER (IncrementalTrieWriterPageAware restatement; 4096-byte pages)
+// ---------------------------------------------------------------------------
+namespace btiw {
+
+constexpr int PAGE = 4096;
+
+struct WNode {
+    int transition = 0;
+    std::vector<std::unique_ptr<WNode>> children;  // ascending transition
+    bool has_payload = false;
+    uint8_t hash_bits = 0;
+    int64_t idxpos = 0;
+    int branch_size = -1;
+    int node_size = -1;
+    bool oop_children = true;   // BaseNode default (forced true pre-complete)
+    bool oop_in_branch = false;
+    int64_t file_pos = -1;
+};
+
+struct Dest {
+    bytes buf;
+    int64_t position() const { return (int64_t)buf.size(); }
+    int bytes_left_in_page() const { return PAGE - (int)(buf.size() % PAGE); }
+    void pad_to_page() { while (buf.size() % PAGE) buf.push_back(0); }
+    void u8(uint8_t b) { buf.push_back(b); }
+    void be(uint64_t v, int n) {
+        for (int i = n - 1; i >= 0; i--) buf.push_back((uint8_t)(v >> (8 * i)));
+    }
+};
+
+// SizedInts.nonZeroSize: significant bits + sign, rounded up to bytes
+static int sized_int_size(int64_t v) {
+    uint64_t u = v < 0 ? ~(uint64_t)v : (uint64_t)v;
+    int bits = 0;
+    while (u >> bits) bits++;
+    return (bits + 1 + 7) / 8;
+}
+
+static int payload_size(const WNode& n) {
+    // PartitionIndexSerializer.sizeofNode: 1 hash byte + SizedInts bytes
+    return n.has_payload ? 1 + sized_int_size(n.idxpos) : 0;
+}
+
+// max (most negative) position delta per Node.maxPositionDelta
+static int64_t max_position_delta(const WNode& n, int64_t node_pos) {
+    if (!n.oop_children)
+        return -(int64_t)(n.branch_size - n.children[0]->branch_size);
+    int64_t min_placed = 0, min_unplaced = 1;
+    for (auto& c : n.children) {
+        if (c->file_pos != -1) min_placed = std::min(min_placed, c->file_pos - node_pos);
+        else if (min_unplaced > 0) min_unplaced = -(int64_t)(n.branch_size - c->branch_size);
+    }
+    return std::min(min_placed, min_unplaced);
+}
+
+// node-type machinery (TrieNode.java). type ids = ordinals.
+struct TypeInfo {
+    int ordinal;
+    int bytes_per_pointer;  // 0 == fractional (SNP4: 4-bit, SNP12/SPARSE_12/DENSE_12: 12-bit)
+};
+
+static int fits_bits(int ordinal) {
+    switch (ordinal) {
+        case 1: return 4;    // SINGLE_NOPAYLOAD_4
+        case 2: return 8;    // SINGLE_8
+        case 3: return 12;   // SINGLE_NOPAYLOAD_12
+        case 4: return 16;   // SINGLE_16
+        case 5: return 8;    // SPARSE_8
+        case 6: return 12;   // SPARSE_12
+        case 7: return 16;   // SPARSE_16
+        case 8: return 24;   // SPARSE_24
+        case 9: return 40;   // SPARSE_40
+        case 10: return 12;  // DENSE_12
+        case 11: return 16;
+        case 12: return 24;
+        case 13: return 32;
+        case 14: return 40;
+        case 15: return 64;  // LONG_DENSE
+    }
+    return 64;
+}
+static bool type_fits(int ordinal, uint64_t dist) {
+    int b = fits_bits(ordinal);
+    return b >= 64 || dist < (1ull << b);
+}
+
+static int child_span(const WNode& n) {
+    return n.children.back()->transition - n.children.front()->transition + 1;
+}
+
+// TrieNode.sizeofNode per type (excl. payload)
+static int type_sizeof(int ordinal, const WNode& n) {
+    int cc = (int)n.children.size();
+    switch (ordinal) {
+        case 0: return 1;
+        case 1: return 2;
+        case 2: return 3;
+        case 3: return 3;
+        case 4: return 4;
+        case 5: return 2 + cc * 2;
+        case 6: return 2 + cc + (cc * 3 + 1) / 2;
+        case 7: return 2 + cc * 3;
+        case 8: return 2 + cc * 4;
+        case 9: return 2 + cc * 6;
+        case 10: return 3 + (child_span(n) * 3 + 1) / 2;
+        case 11: return 3 + child_span(n) * 2;
+        case 12: return 3 + child_span(n) * 3;
+        case 13: return 3 + child_span(n) * 4;
+        case 14: return 3 + child_span(n) * 5;
+        default: return 3 + child_span(n) * 8;
+    }
+}
+
+// TrieNode.typeFor (TrieNode.java:157-180)
+static const int SINGLES[8] = {1, 2, 3, 4, 12, 13, 14, 15};
+static const int SPARSES[8] = {5, 5, 6, 7, 8, 9, 9, 15};
+static const int DENSES[8] = {10, 10, 10, 11, 12, 13, 14, 15};
+
+static int type_for(const WNode& n, int64_t node_pos) {
+    int cc = (int)n.children.size();
+    if (cc == 0) return 0;
+    int64_t delta = max_position_delta(n, node_pos);
+    int idx = 0;
+    while (!type_fits(SINGLES[idx], (uint64_t)(-delta))) idx++;
+    if (cc == 1) {
+        // fractional singles cannot carry a payload
+        if (n.has_payload && (SINGLES[idx] == 1 || SINGLES[idx] == 3)) idx++;
+        return SINGLES[idx];
+    }
+    int sparse = SPARSES[idx], dense = DENSES[idx];
+    return type_sizeof(sparse, n) < type_sizeof(dense, n) ? sparse : dense;
+}
+
+// serializer.sizeofNode (PartitionIndex.PartitionIndexSerializer)
+static int sizeof_node(const WNode& n, int64_t node_pos) {
+    return type_sizeof(type_for(n, node_pos), n) + payload_size(n);
+}
+
+// TrieNode.serialize per type + payload (PartitionIndexSerializer.write)
+static void write_node_bytes(Dest& d, const WNode& n, int64_t node_pos) {
+    int t = type_for(n, node_pos);
+    int pb = 0;
+    if (n.has_payload) pb = 8 + (sized_int_size(n.idxpos) - 1);
+    auto dist = [&](const WNode& c) { return (uint64_t)(node_pos - c.file_pos); };
+    switch (t) {
+        case 0:
+            d.u8((uint8_t)(0 << 4 | pb));
+            break;
+        case 1:
+            d.u8((uint8_t)(1 << 4 | (int)dist(*n.children[0])));
+            d.u8((uint8_t)n.children[0]->transition);
+            break;
+        case 2:
+            d.u8((uint8_t)(2 << 4 | pb));
+            d.u8((uint8_t)n.children[0]->transition);
+            d.be(dist(*n.children[0]), 1);
+            break;
+        case 3: {
+            uint64_t v = dist(*n.children[0]);
+            d.u8((uint8_t)(3 << 4 | (int)(v >> 8)));
+            d.u8((uint8_t)(v & 0xFF));
+            d.u8((uint8_t)n.children[0]->transition);
+            break;
+        }
+        case 4:
+            d.u8((uint8_t)(4 << 4 | pb));
+            d.u8((uint8_t)n.children[0]->transition);
+            d.be(dist(*n.children[0]), 2);
+            break;
+        case 5: case 7: case 8: case 9: {
+            int w = t == 5 ? 1 : t == 7 ? 2 : t == 8 ? 3 : 5;
+            d.u8((uint8_t)(t << 4 | pb));
+            d.u8((uint8_t)n.children.size());
+            for (auto& c : n.children) d.u8((uint8_t)c->transition);
+            for (auto& c : n.children) d.be(dist(*c), w);
+            break;
+        }
+        case 6: {  // SPARSE_12: packed 12-bit
+            d.u8((uint8_t)(6 << 4 | pb));
+            d.u8((uint8_t)n.children.size());
+            for (auto& c : n.children) d.u8((uint8_t)c->transition);
+            uint32_t carry = 0;
+            bool half = false;
+            for (auto& c : n.children) {
+                uint32_t v = (uint32_t)dist(*c);
+                if (!half) { d.u8((uint8_t)(v >> 4)); carry = v & 0xF; half = true; }
+                else { d.u8((uint8_t)(carry << 4 | (v >> 8))); d.u8((uint8_t)(v & 0xFF)); half = false; }
+            }
+            if (half) d.u8((uint8_t)(carry << 4));
+            break;
+        }
+        case 10: {  // DENSE_12
+            d.u8((uint8_t)(10 << 4 | pb));
+            int start = n.children.front()->transition, span = child_span(n);
+            d.u8((uint8_t)start);
+            d.u8((uint8_t)(span - 1));
+            size_t ci = 0;
+            uint32_t carry = 0;
+            bool half = false;
+            for (int i = 0; i < span; i++) {
+                uint32_t v = 0;
+                if (ci < n.children.size() && n.children[ci]->transition == start + i)
+                    v = (uint32_t)dist(*n.children[ci++]);
+                if (!half) { d.u8((uint8_t)(v >> 4)); carry = v & 0xF; half = true; }
+                else { d.u8((uint8_t)(carry << 4 | (v >> 8))); d.u8((uint8_t)(v & 0xFF)); half = false; }
+            }
+            if (half) d.u8((uint8_t)(carry << 4));
+            break;
+        }
+        case 11: case 12: case 13: case 14: case 15: {
+            int w = t == 11 ? 2 : t == 12 ? 3 : t == 13 ? 4 : t == 14 ? 5 : 8;
+            d.u8((uint8_t)(t << 4 | pb));
+            int start = n.children.front()->transition, span = child_span(n);
+            d.u8((uint8_t)start);
+            d.u8((uint8_t)(span - 1));
+            size_t ci = 0;
+            for (int i = 0; i < span; i++) {
+                uint64_t v = 0;
+                if (ci < n.children.size() && n.children[ci]->transition == start + i)
+                    v = dist(*n.children[ci++]);
+                d.be(v, w);
+            }
+            break;
+        }
+    }
+    if (n.has_payload) {
+        d.u8(n.hash_bits);
+        d.be((uint64_t)n.idxpos, sized_int_size(n.idxpos));
+    }
+}
+
+static int recalc_total_size(WNode& n, int64_t node_pos);
+static void layout_children(Dest& d, WNode& n);
+
+// recursive in-page write (IncrementalTrieWriterPageAware.write)
+static int64_t write_rec(Dest& d, WNode& n) {
+    int64_t node_pos = d.position();
+    for (auto& c : n.children)
+        if (c->file_pos == -1) c->file_pos = write_rec(d, *c);
+    node_pos += n.branch_size;
+    write_node_bytes(d, n, node_pos);
+    return node_pos;
+}
+
+static int recalc_total_size(WNode& n, int64_t node_pos) {
+    if (n.oop_in_branch) {
+        int sz = 0;
+        for (auto& c : n.children) sz += recalc_total_size(*c, node_pos + sz);
+        n.branch_size = sz;
+    }
+    if (n.oop_children || n.oop_in_branch)
+        n.node_size = sizeof_node(n, node_pos + n.branch_size);
+    return n.branch_size + n.node_size;
+}
+
+static void layout_children(Dest& d, WNode& n) {
+    // NavigableSet ordered by (branch+node size, transition); pick the
+    // largest that fits the current page, else pad and take the largest
+    std::vector<WNode*> pending;
+    for (auto& c : n.children)
+        if (c->file_pos == -1) pending.push_back(c.get());
+    auto cmp = [](WNode* a, WNode* b) {
+        int sa = a->branch_size + a->node_size, sb = b->branch_size + b->node_size;
+        if (sa != sb) return sa < sb;
+        return a->transition < b->transition;
+    };
+    std::sort(pending.begin(), pending.end(), cmp);
+    int bytes_left = d.bytes_left_in_page();
+    while (!pending.empty()) {
+        // largest with branch+node <= bytes_left
+        int i = (int)pending.size() - 1;
+        while (i >= 0 && pending[i]->branch_size + pending[i]->node_size > bytes_left) i--;
+        WNode* child;
+        if (i < 0) {
+            d.pad_to_page();
+            bytes_left = PAGE;
+            child = pending.back();
+            pending.pop_back();
+        } else {
+            child = pending[i];
+            pending.erase(pending.begin() + i);
+        }
+        if (child->oop_children || child->oop_in_branch) {
+            int actual = recalc_total_size(*child, d.position());
+            if (actual > bytes_left) {
+                if (bytes_left == PAGE) {
+                    layout_children(d, *child);
+                    bytes_left = d.bytes_left_in_page();
+                }
+                // put back with the new size
+                pending.push_back(child);
+                std::sort(pending.begin(), pending.end(), cmp);
+                continue;
+            }
+        }
+        child->file_pos = write_rec(d, *child);
+        bytes_left = d.bytes_left_in_page();
+    }
+    n.branch_size = 0;
+    n.oop_children = true;
+    n.oop_in_branch = false;
+    n.node_size = sizeof_node(n, d.position());
+}
+
+// bottom-up completion (IncrementalTrieWriterPageAware.complete), post-order
+static void complete_rec(Dest& d, WNode& n) {
+    for (auto& c : n.children) complete_rec(d, *c);
+    int branch = 0;
+    for (auto& c : n.children) branch += c->branch_size + c->node_size;
+    n.branch_size = branch;
+    int node_size = sizeof_node(n, d.position());
+    if (node_size + branch < PAGE) {
+        n.node_size = node_size;
+        n.oop_children = false;
+        n.oop_in_branch = false;
+        for (auto& c : n.children) {
+            if (c->file_pos != -1) n.oop_children = true;
+            else if (c->oop_children || c->oop_in_branch) n.oop_in_branch = true;
+        }
+        return;
+    }
+    layout_children(d, n);
+}
+
+}  // namespace btiw
+
+bytes write_bti_partitions(const std::vector<BtiKeyEntry>& entries) {
+    using namespace btiw;
+    // PartitionIndexBuilder.addEntry: store each key cut to
+    // max(diffPoint(prev,cur), diffPoint(cur,next)) bytes
+    auto diff_point = [](const bytes& a, const bytes& b) {
+        size_t i = 0;
+        while (i < a.size() && i < b.size() && a[i] == b[i]) i++;
+        return (int)i + 1;
+    };
+    WNode root;
+    auto insert = [&](const bytes& bc, int cut, const BtiKeyEntry& e) {
+        WNode* n = &root;
+        for (int i = 0; i < cut && i < (int)bc.size(); i++) {
+            if (n->children.empty() || n->children.back()->transition != bc[i]) {
+                n->children.push_back(std::make_unique<WNode>());
+                n->children.back()->transition = bc[i];
+            }
+            n = n->children.back().get();
+        }
+        n->has_payload = true;
+        n->hash_bits = e.hash_bits;
+        n->idxpos = e.idxpos;
+    };
+    for (size_t i = 0; i < entries.size(); i++) {
+        int dp_prev = i ? diff_point(entries[i - 1].byte_comparable, entries[i].byte_comparable) : 0;
+        int dp_next = i + 1 < entries.size()
+                          ? diff_point(entries[i].byte_comparable, entries[i + 1].byte_comparable)
+                          : 0;
+        insert(entries[i].byte_comparable, std::max(dp_prev, dp_next), entries[i]);
+    }
+
+    Dest d;
+    // performCompletion
+    complete_rec(d, root);
+    int actual = recalc_total_size(root, d.position());
+    int bytes_left = d.bytes_left_in_page();
+    if (actual > bytes_left) {
+        if (actual <= PAGE) {
+            d.pad_to_page();
+            bytes_left = PAGE;
+            actual = recalc_total_size(root, d.position());
+        }
+        if (actual > bytes_left) {
+            layout_children(d, root);
+            if (root.node_size > d.bytes_left_in_page()) {
+                d.pad_to_page();
+                recalc_total_size(root, d.position());
+            }
+        }
+    }
+    int64_t root_pos = write_rec(d, root);
+
+    // PartitionIndexBuilder.complete footer
+    int64_t first_key_pos = d.position();
+    if (!entries.empty()) {
+        d.be(entries.front().raw_key.size(), 2);
+        for (uint8_t b : entries.front().raw_key) d.u8(b);
+        d.be(entries.back().raw_key.size(), 2);
+        for (uint8_t b : entries.back().raw_key) d.u8(b);
+    } else {
+        d.be(0, 2);
+        d.be(0, 2);
+    }
+    d.be((uint64_t)first_key_pos, 8);
+    d.be(entries.size(), 8);
+    d.be((uint64_t)root_pos, 8);
+    return d.buf;
 }
 
 }  // namespace oracle

@@ -47,4 +47,23 @@ struct BtiRowIndexBlock {
 // (the position a Partitions.db payload with idxpos >= 0 points at)
 BtiRowIndexBlock read_bti_row_index(const bytes& file, uint64_t index_pos);
 
+// ---------------------------------------------------------------------------
+// BTI Partitions.db WRITER restatement: IncrementalTrieWriterPageAware
+// (io/tries/IncrementalTrieWriterPageAware.java) + TrieNode type selection
+// (TrieNode.java:157-180) + PartitionIndexBuilder key-cutting and footer
+// (PartitionIndexBuilder.java:130-183) + the PartitionIndex payload
+// serializer (PartitionIndex.java:111-141). Byte-exactness pinned against
+// the reference's legacy_da fixtures and round-tripped through the reader.
+// ---------------------------------------------------------------------------
+struct BtiKeyEntry {
+    bytes byte_comparable;  // full byte-ordered representation of the key
+    bytes raw_key;          // the key bytes for the footer (first/last)
+    uint8_t hash_bits;      // DecoratedKey.filterHashLowerBits
+    int64_t idxpos;         // >=0 row index pos; <0 = ~data_pos
+};
+
+// builds a complete -Partitions.db image (trie + keys + footer) from entries
+// in key order
+bytes write_bti_partitions(const std::vector<BtiKeyEntry>& entries);
+
 }  // namespace oracle
