@@ -522,6 +522,41 @@ static void complete_rec(Dest& d, WNode& n) {
 
 }  // namespace btiw
 
+// escaped component emission (ByteSource.AbstractEscaper)
+static void bc_escape(bytes& out, const bytes& data) {
+    size_t i = 0;
+    while (i < data.size()) {
+        if (data[i] != 0) {
+            out.push_back(data[i++]);
+            continue;
+        }
+        out.push_back(0x00);  // ESCAPE
+        i++;
+        while (i < data.size() && data[i] == 0) {
+            out.push_back(0xFE);  // ESCAPED_0_CONT
+            i++;
+        }
+        if (i < data.size()) {
+            out.push_back(0xFF);  // ESCAPED_0_DONE, then the non-zero byte
+        } else {
+            out.push_back(0xFE);  // zeros at end: CONT and stop
+            return;
+        }
+    }
+    out.push_back(0x00);  // trailing ESCAPE after non-zero-ending data
+}
+
+bytes bti_byte_comparable_m3(int64_t token, const bytes& key) {
+    bytes out;
+    out.push_back(0x40);  // NEXT_COMPONENT
+    uint64_t t = (uint64_t)token ^ (1ull << 63);
+    for (int i = 7; i >= 0; i--) out.push_back((uint8_t)(t >> (8 * i)));
+    out.push_back(0x40);  // NEXT_COMPONENT
+    bc_escape(out, key);
+    out.push_back(0x38);  // TERMINATOR
+    return out;
+}
+
 bytes write_bti_partitions(const std::vector<BtiKeyEntry>& entries) {
     using namespace btiw;
     // PartitionIndexBuilder.addEntry: store each key cut to

@@ -66,4 +66,11 @@ struct BtiKeyEntry {
 // in key order
 bytes write_bti_partitions(const std::vector<BtiKeyEntry>& entries);
 
+// OSS50 byte-comparable encoding of a Murmur3 DecoratedKey
+// (DecoratedKey.asComparableBytes: Multi(NEXT_COMPONENT-prefixed components,
+// TERMINATOR) over [ByteSource.of(token) = 8 BE bytes sign-flipped,
+// escaped key bytes]; AbstractEscaper: 0x00 runs -> 0x00 FE* [FF byte |
+// FE-at-end], non-zero tail -> trailing 0x00)
+bytes bti_byte_comparable_m3(int64_t token, const bytes& key);
+
 }  // namespace oracle
