@@ -178,18 +178,23 @@ BtiRowIndexBlock read_bti_row_index(const bytes& file, uint64_t index_pos) {
     walk(file, out.root_pos, prefix, [&](const bytes& pf, int pb, uint64_t ppos) {
         BtiRowIndexEntry e;
         e.prefix = pf;
+        e.pb = pb;
         int obytes = pb & 7;
         e.offset = be(file, ppos, obytes);
+        size_t plen = obytes;
         if (pb >= 8) {
             e.has_open = true;
             uint8_t flags = file.at(ppos + obytes);
             if (flags & 0x80) {
                 e.open_dt = DeletionTime{};  // LIVE
+                plen += 1;
             } else {
                 e.open_dt.mfda = (int64_t)be(file, ppos + obytes, 8);
                 e.open_dt.ldt = (uint32_t)be(file, ppos + obytes + 8, 4);
+                plen += 12;
             }
         }
+        e.raw_payload = bytes(file.begin() + ppos, file.begin() + ppos + plen);
         out.entries.push_back(std::move(e));
     });
     return out;
@@ -205,9 +210,8 @@ constexpr int PAGE = 4096;
 struct WNode {
     int transition = 0;
     std::vector<std::unique_ptr<WNode>> children;  // ascending transition
-    bool has_payload = false;
-    uint8_t hash_bits = 0;
-    int64_t idxpos = 0;
+    int pb = 0;           // 4 payload bits (0 == no payload)
+    bytes payload;        // serialized payload bytes following the node
     int branch_size = -1;
     int node_size = -1;
     bool oop_children = true;   // BaseNode default (forced true pre-complete)
@@ -234,10 +238,7 @@ static int sized_int_size(int64_t v) {
     return (bits + 1 + 7) / 8;
 }
 
-static int payload_size(const WNode& n) {
-    // PartitionIndexSerializer.sizeofNode: 1 hash byte + SizedInts bytes
-    return n.has_payload ? 1 + sized_int_size(n.idxpos) : 0;
-}
+static int payload_size(const WNode& n) { return (int)n.payload.size(); }
 
 // max (most negative) position delta per Node.maxPositionDelta
 static int64_t max_position_delta(const WNode& n, int64_t node_pos) {
@@ -322,7 +323,7 @@ static int type_for(const WNode& n, int64_t node_pos) {
     while (!type_fits(SINGLES[idx], (uint64_t)(-delta))) idx++;
     if (cc == 1) {
         // fractional singles cannot carry a payload
-        if (n.has_payload && (SINGLES[idx] == 1 || SINGLES[idx] == 3)) idx++;
+        if (n.pb != 0 && (SINGLES[idx] == 1 || SINGLES[idx] == 3)) idx++;
         return SINGLES[idx];
     }
     int sparse = SPARSES[idx], dense = DENSES[idx];
@@ -337,8 +338,7 @@ static int sizeof_node(const WNode& n, int64_t node_pos) {
 // TrieNode.serialize per type + payload (PartitionIndexSerializer.write)
 static void write_node_bytes(Dest& d, const WNode& n, int64_t node_pos) {
     int t = type_for(n, node_pos);
-    int pb = 0;
-    if (n.has_payload) pb = 8 + (sized_int_size(n.idxpos) - 1);
+    int pb = n.pb;
     auto dist = [&](const WNode& c) { return (uint64_t)(node_pos - c.file_pos); };
     switch (t) {
         case 0:
@@ -421,10 +421,7 @@ static void write_node_bytes(Dest& d, const WNode& n, int64_t node_pos) {
             break;
         }
     }
-    if (n.has_payload) {
-        d.u8(n.hash_bits);
-        d.be((uint64_t)n.idxpos, sized_int_size(n.idxpos));
-    }
+    for (uint8_t b : n.payload) d.u8(b);
 }
 
 static int recalc_total_size(WNode& n, int64_t node_pos);
@@ -500,6 +497,8 @@ static void layout_children(Dest& d, WNode& n) {
     n.node_size = sizeof_node(n, d.position());
 }
 
+static int64_t complete_and_write(Dest& d, WNode& root);
+
 // bottom-up completion (IncrementalTrieWriterPageAware.complete), post-order
 static void complete_rec(Dest& d, WNode& n) {
     for (auto& c : n.children) complete_rec(d, *c);
@@ -518,6 +517,28 @@ static void complete_rec(Dest& d, WNode& n) {
         return;
     }
     layout_children(d, n);
+}
+
+// performCompletion + final root write (IncrementalTrieWriterPageAware)
+static int64_t complete_and_write(Dest& d, WNode& root) {
+    complete_rec(d, root);
+    int actual = recalc_total_size(root, d.position());
+    int bytes_left = d.bytes_left_in_page();
+    if (actual > bytes_left) {
+        if (actual <= PAGE) {
+            d.pad_to_page();
+            bytes_left = PAGE;
+            actual = recalc_total_size(root, d.position());
+        }
+        if (actual > bytes_left) {
+            layout_children(d, root);
+            if (root.node_size > d.bytes_left_in_page()) {
+                d.pad_to_page();
+                recalc_total_size(root, d.position());
+            }
+        }
+    }
+    return write_rec(d, root);
 }
 
 }  // namespace btiw
@@ -544,6 +565,43 @@ static void bc_escape(bytes& out, const bytes& data) {
         }
     }
     out.push_back(0x00);  // trailing ESCAPE after non-zero-ending data
+}
+
+uint64_t append_bti_row_index(bytes& file, const BtiRowIndexBlockSpec& spec) {
+    using namespace btiw;
+    WNode root;
+    for (auto& e : spec.entries) {
+        WNode* n = &root;
+        for (uint8_t b : e.prefix) {
+            if (n->children.empty() || n->children.back()->transition != b) {
+                n->children.push_back(std::make_unique<WNode>());
+                n->children.back()->transition = b;
+            }
+            n = n->children.back().get();
+        }
+        n->pb = e.pb;
+        n->payload = e.payload;
+    }
+    Dest d;
+    d.buf = std::move(file);
+    int64_t root_pos = complete_and_write(d, root);
+    // TrieIndexEntry footer (TrieIndexEntry.serialize; root delta is
+    // relative to the entry start = position after the short-length key)
+    uint64_t index_pos = (uint64_t)d.position();
+    d.be(spec.partition_key.size(), 2);
+    for (uint8_t b : spec.partition_key) d.u8(b);
+    int64_t base = d.position();
+    put_unsigned_vint(d.buf, spec.data_pos);
+    put_vint(d.buf, root_pos - base);
+    put_unsigned_vint(d.buf, spec.block_count);
+    if (spec.partition_del.live()) {
+        d.u8(0x80);  // compact DeletionTime: LIVE flag
+    } else {
+        d.be((uint64_t)spec.partition_del.mfda, 8);
+        d.be(spec.partition_del.ldt, 4);
+    }
+    file = std::move(d.buf);
+    return index_pos;
 }
 
 bytes bti_byte_comparable_m3(int64_t token, const bytes& key) {
@@ -576,9 +634,12 @@ bytes write_bti_partitions(const std::vector<BtiKeyEntry>& entries) {
             }
             n = n->children.back().get();
         }
-        n->has_payload = true;
-        n->hash_bits = e.hash_bits;
-        n->idxpos = e.idxpos;
+        int sz = sized_int_size(e.idxpos);
+        n->pb = 8 + (sz - 1);  // PartitionIndexSerializer: always with hash
+        n->payload.clear();
+        n->payload.push_back(e.hash_bits);
+        for (int b = sz - 1; b >= 0; b--)
+            n->payload.push_back((uint8_t)((uint64_t)e.idxpos >> (8 * b)));
     };
     for (size_t i = 0; i < entries.size(); i++) {
         int dp_prev = i ? diff_point(entries[i - 1].byte_comparable, entries[i].byte_comparable) : 0;
@@ -589,25 +650,7 @@ bytes write_bti_partitions(const std::vector<BtiKeyEntry>& entries) {
     }
 
     Dest d;
-    // performCompletion
-    complete_rec(d, root);
-    int actual = recalc_total_size(root, d.position());
-    int bytes_left = d.bytes_left_in_page();
-    if (actual > bytes_left) {
-        if (actual <= PAGE) {
-            d.pad_to_page();
-            bytes_left = PAGE;
-            actual = recalc_total_size(root, d.position());
-        }
-        if (actual > bytes_left) {
-            layout_children(d, root);
-            if (root.node_size > d.bytes_left_in_page()) {
-                d.pad_to_page();
-                recalc_total_size(root, d.position());
-            }
-        }
-    }
-    int64_t root_pos = write_rec(d, root);
+    int64_t root_pos = btiw::complete_and_write(d, root);
 
     // PartitionIndexBuilder.complete footer
     int64_t first_key_pos = d.position();

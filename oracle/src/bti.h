@@ -32,6 +32,8 @@ struct BtiRowIndexEntry {
     uint64_t offset = 0;        // offset within the partition
     bool has_open = false;      // pb >= 8: deletion active at block start
     DeletionTime open_dt;
+    int pb = 0;                 // raw payload bits + bytes (for regeneration)
+    bytes raw_payload;
 };
 
 struct BtiRowIndexBlock {
@@ -65,6 +67,25 @@ struct BtiKeyEntry {
 // builds a complete -Partitions.db image (trie + keys + footer) from entries
 // in key order
 bytes write_bti_partitions(const std::vector<BtiKeyEntry>& entries);
+
+// Rows.db writer: one row-index structure (trie + TrieIndexEntry footer)
+// appended to `file` (page arithmetic is file-absolute). Entries carry the
+// pre-cut separator prefixes and raw payloads (pb = offset-bytes |
+// 8-if-open-deletion; payload = SizedInts offset [+ compact DeletionTime]).
+// Returns the footer position (what a Partitions.db idxpos >= 0 points at).
+struct BtiRowIndexBlockSpec {
+    struct Entry {
+        bytes prefix;
+        int pb;
+        bytes payload;
+    };
+    std::vector<Entry> entries;
+    bytes partition_key;
+    uint64_t data_pos = 0;
+    uint64_t block_count = 0;
+    DeletionTime partition_del;
+};
+uint64_t append_bti_row_index(bytes& file, const BtiRowIndexBlockSpec& spec);
 
 // OSS50 byte-comparable encoding of a Murmur3 DecoratedKey
 // (DecoratedKey.asComparableBytes: Multi(NEXT_COMPONENT-prefixed components,

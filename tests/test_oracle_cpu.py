@@ -391,6 +391,7 @@ def test_bti_partition_index_writer(tmp_path):
          os.path.join(repo, "oracle/src/gen.cpp"),
          "-o", exe, "-l:liblz4.so.1", "-ldl"],
         check=True, capture_output=True)
-    r = subprocess.run([exe, fix], capture_output=True, text=True)
+    clust = os.path.join(repo, "tests", "golden", "legacy_da_clust", "da-1-bti")
+    r = subprocess.run([exe, fix, clust], capture_output=True, text=True)
     assert r.returncode == 0, r.stdout + r.stderr
     assert "all OK" in r.stdout
