@@ -567,6 +567,88 @@ static void bc_escape(bytes& out, const bytes& data) {
     out.push_back(0x00);  // trailing ESCAPE after non-zero-ending data
 }
 
+// ByteSource.variableLengthInteger (ByteSource.java:219-…)
+static void bc_varint(bytes& out, int64_t value64) {
+    uint64_t v = (uint64_t)value64;
+    uint64_t neg = (uint64_t)(value64 >> 63);  // all-ones for negative
+    v ^= neg;
+    int bits = 64;
+    while (bits > 1 && !((v | 1) >> (bits - 1))) bits--;  // 64 - clz(v|1)
+    int nbytes = bits / 7 + 1;
+    if (nbytes >= 9) {
+        out.push_back((uint8_t)(neg ? 0x00 : 0xFF));
+        uint64_t val = (v | 0x8000000000000000ull) ^ neg;
+        for (int i = 7; i >= 0; i--) out.push_back((uint8_t)(val >> (8 * i)));
+    } else {
+        uint64_t mask = ((uint64_t)(int64_t)-0x100 >> nbytes) & 0xFF;
+        int pos = nbytes * 8;
+        uint64_t val = (v | (mask << (pos - 8))) ^ neg;
+        for (int i = nbytes - 1; i >= 0; i--) out.push_back((uint8_t)(val >> (8 * i)));
+    }
+}
+
+bytes bti_byte_comparable_clustering(const Clustering& c,
+                                     const std::vector<CqlType>& types, BoundKind kind) {
+    bytes out;
+    for (size_t i = 0; i < c.size(); i++) {
+        out.push_back(0x40);  // NEXT_COMPONENT
+        switch (types.at(i)) {
+            case CqlType::LONG: {
+                if (c[i].v.size() != 8) throw std::runtime_error("bad LONG clustering");
+                int64_t v = 0;
+                for (int b = 0; b < 8; b++) v = (v << 8) | c[i].v[b];
+                bc_varint(out, v);
+                break;
+            }
+            case CqlType::INT32: {
+                if (c[i].v.size() != 4) throw std::runtime_error("bad INT clustering");
+                int32_t v = 0;
+                for (int b = 0; b < 4; b++) v = (v << 8) | c[i].v[b];
+                bc_varint(out, (int64_t)v);
+                break;
+            }
+            default:
+                bc_escape(out, c[i].v);
+        }
+    }
+    switch (kind) {
+        case CLUSTERING_K: out.push_back(0x40); break;
+        case INCL_END: case EXCL_START: out.push_back(0x60); break;
+        case EXCL_END: case INCL_START: out.push_back(0x20); break;
+        default: throw std::runtime_error("unsupported bound kind for byte-comparable");
+    }
+    return out;
+}
+
+bytes bti_separator_gt(const bytes& prev, const bytes& cur) {
+    bytes out;
+    size_t i = 0;
+    while (i < prev.size() && i < cur.size() && prev[i] == cur[i]) {
+        out.push_back(cur[i]);
+        i++;
+    }
+    // prev must be strictly less; emit cur's differing byte (or cur ended ==
+    // prev prefix case cannot happen for valid prefix-free inputs)
+    if (i < cur.size()) out.push_back(cur[i]);
+    return out;
+}
+
+bytes bti_nudge(const bytes& value, size_t nudge_at) {
+    bytes out;
+    size_t i = 0;
+    for (; i <= nudge_at && i < value.size(); i++) out.push_back(value[i]);
+    // increment at nudge_at; 0xFF spills rightward (emit and nudge next)
+    while (!out.empty() && out.size() - 1 >= nudge_at && out.back() == 0xFF) {
+        nudge_at++;
+        if (out.size() - 1 < nudge_at) {
+            if (i < value.size()) out.push_back(value[i++]);
+            else break;
+        } else break;
+    }
+    if (!out.empty() && out.back() != 0xFF) out.back()++;
+    return out;
+}
+
 uint64_t append_bti_row_index(bytes& file, const BtiRowIndexBlockSpec& spec) {
     using namespace btiw;
     WNode root;

@@ -87,6 +87,23 @@ struct BtiRowIndexBlockSpec {
 };
 uint64_t append_bti_row_index(bytes& file, const BtiRowIndexBlockSpec& spec);
 
+// OSS50 byte-comparable encoding of a ClusteringPrefix
+// (ClusteringComparator.ByteComparableClustering: per component
+// NEXT_COMPONENT 0x40 + the type's OSS50 encoding; kind terminator —
+// CLUSTERING_K 0x40, INCL_END/EXCL_START 0x60, EXCL_END/INCL_START 0x20).
+// Types supported: UTF8/ASCII/BYTES (escaped) and LONG/INT
+// (ByteSource.variableLengthInteger).
+bytes bti_byte_comparable_clustering(const Clustering& c,
+                                     const std::vector<CqlType>& types, BoundKind kind);
+
+// ByteComparable.separatorGt(prev, cur): shortest byte string s with
+// prev < s <= cur (common prefix + cur's first differing byte)
+bytes bti_separator_gt(const bytes& prev, const bytes& cur);
+
+// RowIndexWriter.nudge(value, nudgeAt): value's bytes through index nudgeAt
+// with the byte at nudgeAt incremented (0xFF spills to the next position)
+bytes bti_nudge(const bytes& value, size_t nudge_at);
+
 // OSS50 byte-comparable encoding of a Murmur3 DecoratedKey
 // (DecoratedKey.asComparableBytes: Multi(NEXT_COMPONENT-prefixed components,
 // TERMINATOR) over [ByteSource.of(token) = 8 BE bytes sign-flipped,

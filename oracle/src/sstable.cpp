@@ -911,7 +911,8 @@ static StatisticsFile read_statistics(const bytes& b) {
     return sf;
 }
 
-static Partition read_partition(Reader& r, const Header& h) {
+static Partition read_partition(Reader& r, const Header& h,
+                                std::vector<uint64_t>* item_offsets = nullptr) {
     Partition p;
     uint16_t klen = r.be16();
     p.key = r.take(klen);
@@ -925,8 +926,10 @@ static Partition read_partition(Reader& r, const Header& h) {
         p.static_row = read_row(r, h, flags, true);
     }
     while (true) {
+        uint64_t item_pos = r.pos;
         uint8_t flags = r.u8();
         if (flags & F_END_OF_PARTITION) break;
+        if (item_offsets) item_offsets->push_back(item_pos);
         Unfiltered u;
         if (flags & F_IS_MARKER) {
             u.kind = Unfiltered::MARKER;
@@ -1028,6 +1031,13 @@ ScrubResult scrub_sstable(const std::string& inbase, const std::string& outbase)
     return sr;
 }
 
+
+Partition read_one_partition(const bytes& raw, uint64_t offset, const Header& h,
+                             std::vector<uint64_t>* item_offsets) {
+    Reader r(raw);
+    r.skip(offset);
+    return read_partition(r, h, item_offsets);
+}
 
 SSTable read_sstable(const std::string& base, bool keep_raw) {
     SSTable t;

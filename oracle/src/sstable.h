@@ -223,6 +223,10 @@ struct ScrubResult { uint64_t kept = 0, dropped = 0; };
 ScrubResult scrub_sstable(const std::string& inbase, const std::string& outbase);
 
 SSTable read_sstable(const std::string& base, bool keep_raw = false);
+// parse one partition at `offset` of a decompressed Data.db image; optionally
+// records each unfiltered's absolute byte offset (test/BTI-index tooling)
+Partition read_one_partition(const bytes& raw, uint64_t offset, const Header& h,
+                             std::vector<uint64_t>* item_offsets = nullptr);
 
 // ---- writer ----
 struct WriterOut {
