@@ -889,7 +889,7 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
         fclose(f);
     }
     const uint64_t worst_slab = (uint64_t)SLAB * ((uint64_t)SLOT_STRIDE + 4);
-    const int NSLOTS = 3;
+    const int NSLOTS = 4;
     uint8_t* h_slab0 = (uint8_t*)g_pin_out[wslot][0].get(worst_slab * NSLOTS);
     if (!h_slab0) throw std::runtime_error("pinned out alloc failed");
     DevBuf d_gat[2], d_foff[2];
@@ -931,7 +931,7 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
         HIP_CHECK(hipStreamSynchronize(cstream));
         uint64_t off0 = file_off;
         wfut[slot] = std::async(std::launch::async, [=]() {
-            int nth = 4;
+            int nth = 12;  // tmpfs/page-cache writes are memcpy-bound and scale with threads
             std::vector<std::thread> th;
             size_t per = (slab_bytes + nth - 1) / nth;
             for (int t = 0; t < nth; t++) {
@@ -1498,9 +1498,14 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         check_cancel(job);
 
         // ---- reconcile + purge ----
+        // Host staging for the small async H2D setup copies below. These must
+        // outlive the copies until the next hipStreamSynchronize on `stream`
+        // (an async copy from freed pageable memory is undefined), so they
+        // live at this scope, past the reconcile-phase sync.
+        std::vector<uint32_t> bases(k);
+        std::vector<int64_t> klo_h, khi_h, ov_lo_h, ov_hi_h, ov_ts_h;
         DevBuf d_srcbases, d_group_rows;
         {
-            std::vector<uint32_t> bases(k);
             for (int s = 0; s < k; s++) bases[s] = srcs[s].rec_base;
             d_srcbases.alloc(k * 4);
             HIP_CHECK(hipMemcpyAsync(d_srcbases.p, bases.data(), k * 4, hipMemcpyHostToDevice, stream));
@@ -1535,33 +1540,36 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         pp.shard_hi = job->token_hi;
         DevBuf d_kr_lo, d_kr_hi;
         if (job->n_keep_ranges > 0) {
-            std::vector<int64_t> klo(job->n_keep_ranges), khi(job->n_keep_ranges);
+            klo_h.resize(job->n_keep_ranges);
+            khi_h.resize(job->n_keep_ranges);
             for (int i = 0; i < job->n_keep_ranges; i++) {
-                klo[i] = job->keep_ranges[i].token_lo;
-                khi[i] = job->keep_ranges[i].token_hi;
+                klo_h[i] = job->keep_ranges[i].token_lo;
+                khi_h[i] = job->keep_ranges[i].token_hi;
             }
             d_kr_lo.alloc(job->n_keep_ranges * 8);
             d_kr_hi.alloc(job->n_keep_ranges * 8);
-            HIP_CHECK(hipMemcpyAsync(d_kr_lo.p, klo.data(), job->n_keep_ranges * 8, hipMemcpyHostToDevice, stream));
-            HIP_CHECK(hipMemcpyAsync(d_kr_hi.p, khi.data(), job->n_keep_ranges * 8, hipMemcpyHostToDevice, stream));
+            HIP_CHECK(hipMemcpyAsync(d_kr_lo.p, klo_h.data(), job->n_keep_ranges * 8, hipMemcpyHostToDevice, stream));
+            HIP_CHECK(hipMemcpyAsync(d_kr_hi.p, khi_h.data(), job->n_keep_ranges * 8, hipMemcpyHostToDevice, stream));
             pp.kr_lo = d_kr_lo.as<int64_t>();
             pp.kr_hi = d_kr_hi.as<int64_t>();
             pp.n_keep_ranges = job->n_keep_ranges;
             pp.invert_ranges = job->invert_ranges;
         }
         if (job->n_overlaps > 0) {
-            std::vector<int64_t> lo(job->n_overlaps), hi(job->n_overlaps), ts(job->n_overlaps);
+            ov_lo_h.resize(job->n_overlaps);
+            ov_hi_h.resize(job->n_overlaps);
+            ov_ts_h.resize(job->n_overlaps);
             for (int i = 0; i < job->n_overlaps; i++) {
-                lo[i] = job->overlaps[i].token_lo;
-                hi[i] = job->overlaps[i].token_hi;
-                ts[i] = job->overlaps[i].min_timestamp;
+                ov_lo_h[i] = job->overlaps[i].token_lo;
+                ov_hi_h[i] = job->overlaps[i].token_hi;
+                ov_ts_h[i] = job->overlaps[i].min_timestamp;
             }
             d_ov_lo.alloc(job->n_overlaps * 8);
             d_ov_hi.alloc(job->n_overlaps * 8);
             d_ov_ts.alloc(job->n_overlaps * 8);
-            HIP_CHECK(hipMemcpyAsync(d_ov_lo.p, lo.data(), job->n_overlaps * 8, hipMemcpyHostToDevice, stream));
-            HIP_CHECK(hipMemcpyAsync(d_ov_hi.p, hi.data(), job->n_overlaps * 8, hipMemcpyHostToDevice, stream));
-            HIP_CHECK(hipMemcpyAsync(d_ov_ts.p, ts.data(), job->n_overlaps * 8, hipMemcpyHostToDevice, stream));
+            HIP_CHECK(hipMemcpyAsync(d_ov_lo.p, ov_lo_h.data(), job->n_overlaps * 8, hipMemcpyHostToDevice, stream));
+            HIP_CHECK(hipMemcpyAsync(d_ov_hi.p, ov_hi_h.data(), job->n_overlaps * 8, hipMemcpyHostToDevice, stream));
+            HIP_CHECK(hipMemcpyAsync(d_ov_ts.p, ov_ts_h.data(), job->n_overlaps * 8, hipMemcpyHostToDevice, stream));
             pp.ov_lo = d_ov_lo.as<int64_t>();
             pp.ov_hi = d_ov_hi.as<int64_t>();
             pp.ov_min_ts = d_ov_ts.as<int64_t>();

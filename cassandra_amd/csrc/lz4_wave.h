@@ -135,6 +135,56 @@ __device__ inline int lz4_wave_compress(const uint8_t* __restrict__ s_chunk, int
     s_table[lz4m_hash(lds_read32(s_chunk, 0))] = 0;
     ip = 1;
 
+    // ---- match-extension resolver ----------------------------------------
+    // Counts common bytes of s_chunk[ip+4..] vs s_chunk[match+4..] up to
+    // matchlimit (dword per lane, 256 B per ballot round). Round-0 operands
+    // may be preloaded by the caller so their global loads overlap the
+    // catch-up round's. On return w8/w8v carry the 8 bytes at (ip+4)+(mc-2)
+    // — the next table-fill dword (new_ip-2) and immediate-test dword
+    // (new_ip) — reconstructed from the breaking round's registers when the
+    // span lies inside lanes [fl-1, fl+1] of that round (else the caller
+    // loads them; both only matter when the compressor continues, in which
+    // case lane fl+1's dword is provably in range and loaded).
+    uint64_t w8 = 0;
+    bool w8v = false;
+    auto ext_count = [&](int eip, int ematch, bool pre, uint32_t aa0, uint32_t bb0) -> int {
+        int offb = 0;
+        while (true) {
+            int pi = eip + 4 + offb + 4 * lane;
+            int navail = matchlimit - pi;
+            uint32_t aa = 0, bb = 0;
+            if (offb == 0 && pre) {
+                aa = aa0;
+                bb = bb0;
+            } else if (navail > 0) {
+                aa = lds_read32(s_chunk, (uint32_t)pi);
+                bb = lds_read32(s_chunk, (uint32_t)(ematch + 4 + offb + 4 * lane));
+            }
+            uint32_t x = aa ^ bb;
+            int eq4 = navail <= 0 ? 0 : (x == 0 ? 4 : (__ffs((int)x) - 1) >> 3);
+            if (eq4 > navail) eq4 = navail;
+            uint64_t ne = wave_ballot(eq4 < 4);
+            if (ne) {
+                int fl = (int)__ffsll((long long)ne) - 1;
+                int eqf = __shfl(eq4, fl);
+                // 8-byte window starting at round offset 4*fl + eqf - 2 spans
+                // lanes fl-1 (eqf<2), fl, fl+1 (eqf>=1)
+                w8v = !((eqf < 2 && fl == 0) || (eqf >= 1 && fl == 63));
+                if (w8v) {
+                    uint32_t A = (uint32_t)__shfl((int)aa, fl > 0 ? fl - 1 : 0);
+                    uint32_t B = (uint32_t)__shfl((int)aa, fl);
+                    uint32_t C = (uint32_t)__shfl((int)aa, fl < 63 ? fl + 1 : 63);
+                    uint64_t AB = (uint64_t)A | ((uint64_t)B << 32);
+                    uint64_t BC = (uint64_t)B | ((uint64_t)C << 32);
+                    w8 = eqf >= 2 ? (BC >> (8 * (eqf - 2)))
+                                  : ((AB >> (8 * (eqf + 2))) | ((uint64_t)C << (8 * (6 - eqf))));
+                }
+                return offb + 4 * fl + eqf;
+            }
+            offb += 4 * WAVE;
+        }
+    };
+
     bool done = false;
     while (!done) {
         // ================= match finder: 64-probe windows =================
@@ -147,7 +197,7 @@ __device__ inline int lz4_wave_compress(const uint8_t* __restrict__ s_chunk, int
             // window's dwords are prefetched while the current one resolves
             // (pure loads: the probe positions are closed-form, independent
             // of the window outcome, so semantics are unchanged and the
-            // global-load latency hides under the LDS/ballot chain)
+            // global-load latency hides under the ballot/compare chain)
             int p_l = P0 + lz4_adv_sum(k0, k0 + lane);
             uint32_t v_l = (p_l >= 0 && p_l + 4 <= srcSize) ? lds_read32(s_chunk, (uint32_t)p_l) : 0;
             while (true) {
@@ -156,49 +206,36 @@ __device__ inline int lz4_wave_compress(const uint8_t* __restrict__ s_chunk, int
                 uint32_t h_l = lz4m_hash(v_l);
                 uint16_t t_l = s_table[h_l];              // pre-window candidate
                 // speculative candidate dword for the (common) no-duplicate
-                // case: issue the load before the marker/ballot section.
-                // Loads only on `valid` lanes: an invalid lane's (in-window
-                // predecessor) position can lie past the chunk end — the
-                // scalar code never evaluates those candidates at all.
+                // case: issue the load before the ballot section. Loads only
+                // on `valid` lanes: an invalid lane's (in-window predecessor)
+                // position can lie past the chunk end — the scalar code never
+                // evaluates those candidates at all.
                 uint32_t spec_cand = valid ? lds_read32(s_chunk, (uint32_t)t_l) : 0;
                 // prefetch next window probes
                 int p_n = P0 + lz4_adv_sum(k0, k0 + WAVE + lane);
                 uint32_t v_n = (p_n >= 0 && p_n + 4 <= srcSize) ? lds_read32(s_chunk, (uint32_t)p_n) : 0;
-                // marker round: conflicting LDS writes retire highest-lane-last,
-                // so the read-back names each hash group's max lane. volatile
-                // stops the compiler from forwarding the lane's own store —
-                // the value read must come from the LDS conflict resolution.
-                {
-                    volatile uint16_t* vt = s_table;
-                    vt[h_l] = (uint16_t)lane;
+                // in-window duplicate groups, all in registers: 13 ballots
+                // over the (HASHLOG+1)-bit hash give each lane the exact set
+                // of lanes probing the same table slot — no LDS marker round
+                // trip, no table clobber/restore, and the ballots overlap the
+                // spec_cand load latency. A lane's in-window predecessor (the
+                // scalar loop's latest earlier probe with this hash) is the
+                // highest same-hash lane strictly below it; invalid lanes are
+                // a suffix (positions/advances are monotone in lane), so a
+                // valid lane's predecessor is always a valid lane.
+                uint64_t mem = ~0ull;
+#pragma unroll
+                for (int j = 0; j < LZ4M_HASHLOG + 1; j++) {
+                    uint64_t Bj = wave_ballot(((h_l >> j) & 1u) != 0);
+                    mem &= ((h_l >> j) & 1u) ? Bj : ~Bj;
                 }
-                int maxgroup;
-                {
-                    volatile const uint16_t* vt = s_table;
-                    maxgroup = (int)vt[h_l];
-                }
-                // exact in-window predecessor per lane via group masks
-                int pred = -1;
-                {
-                    uint64_t G = wave_ballot(maxgroup != lane);  // non-max members
-                    while (G) {
-                        int g = (int)__ffsll((long long)G) - 1;
-                        int mg = __shfl(maxgroup, g);
-                        uint64_t members = wave_ballot(maxgroup == mg);
-                        if (maxgroup == mg) {
-                            uint64_t below = members & ((1ULL << lane) - 1);
-                            if (below) pred = 63 - (int)__clzll((long long)below);
-                        }
-                        G &= ~members;
-                    }
-                }
+                uint64_t below = mem & ((1ULL << lane) - 1);
+                int pred = below ? 63 - (int)__clzll((long long)below) : -1;
                 // exact match test: in-window predecessor position if any,
                 // else the pre-window table entry. The predecessor lane
                 // already holds both its probe position AND its probe dword
                 // in registers — shfl them instead of re-deriving the
-                // position and re-loading the dword (removes a dependent
-                // global load from the per-window critical chain; the shfl
-                // index is per-lane, which CDNA's ds_bpermute supports).
+                // position and re-loading the dword.
                 int pred_idx = pred >= 0 ? pred : 0;
                 uint32_t pred_pos = (uint32_t)__shfl(p_l, pred_idx);
                 uint32_t pred_val = (uint32_t)__shfl((int)v_l, pred_idx);
@@ -211,16 +248,14 @@ __device__ inline int lz4_wave_compress(const uint8_t* __restrict__ s_chunk, int
                 int first_event = match_mask ? (int)__ffsll((long long)match_mask) - 1 : WAVE;
                 bool have_match = first_event < first_abort && first_event < WAVE;
                 int commit_hi = have_match ? first_event : (first_abort < WAVE ? first_abort - 1 : WAVE - 1);
-                // restore pre-window entries, then commit the probed range.
-                // These MUST stay two separate wave-wide stores (restore pass
-                // by all lanes, then commit pass by lanes <= commit_hi):
-                // merging them into one select-store changes the cross-lane
-                // winner when a higher non-committing lane shares a slot —
-                // volatile forbids the compiler from combining them.
-                {
+                // commit the probed range: only lanes up to the match/abort
+                // point write (the table was never clobbered, so no restore
+                // pass). Conflicting same-slot writes retire highest-lane-
+                // last (k_probe_lds_order verifies at init), matching the
+                // scalar loop's last-write-wins order.
+                if (lane <= commit_hi) {
                     volatile uint16_t* vt = s_table;
-                    vt[h_l] = t_l;
-                    if (lane <= commit_hi) vt[h_l] = (uint16_t)p_l;
+                    vt[h_l] = (uint16_t)p_l;
                 }
                 if (have_match) {
                     ip = __shfl(p_l, first_event);  // == P0 + adv_sum(k0, k0+first_event)
@@ -238,16 +273,45 @@ __device__ inline int lz4_wave_compress(const uint8_t* __restrict__ s_chunk, int
             if (aborted) { emit_last_literals(); return (int)op; }
         }
 
-        // ================= catch up =================
-        while (true) {
+        // ========== catch-up ∥ match extension (round-0 loads fused) ======
+        // The scalar loop extends the match backward (catch-up) and then
+        // forward from the moved ip. Forward extension from the PRE-catch-up
+        // ip compares the same byte pairs shifted by `run`, and the first
+        // `run` post-catch-up pairs are already known equal (they lie in the
+        // catch-up run / verified match dword), so
+        //   mc_post == run + mc_pre        (same matchlimit clamp)
+        // — both rounds' loads issue together and resolve in one global
+        // round trip instead of two.
+        int mc;
+        {
             int e = lane + 1;
-            bool ok = (ip - e + 1 > anchor) && (match - e + 1 > 0) &&
-                      s_chunk[ip - e] == s_chunk[match - e];
-            uint64_t bad = wave_ballot(!ok);
+            bool cu_ok = (ip - e + 1 > anchor) && (match - e + 1 > 0) &&
+                         s_chunk[ip - e] == s_chunk[match - e];
+            uint32_t aa0 = 0, bb0 = 0;
+            {
+                int pi0 = ip + 4 + 4 * lane;
+                if (matchlimit - pi0 > 0) {
+                    aa0 = lds_read32(s_chunk, (uint32_t)pi0);
+                    bb0 = lds_read32(s_chunk, (uint32_t)(match + 4 + 4 * lane));
+                }
+            }
+            uint64_t bad = wave_ballot(!cu_ok);
             int run = bad ? (int)__ffsll((long long)bad) - 1 : WAVE;
+            {   // full-wave catch-up: keep walking backward (rare)
+                int cip = ip - run, cmatch = match - run, r = run;
+                while (r == WAVE) {
+                    bool ok = (cip - e + 1 > anchor) && (cmatch - e + 1 > 0) &&
+                              s_chunk[cip - e] == s_chunk[cmatch - e];
+                    uint64_t bad2 = wave_ballot(!ok);
+                    r = bad2 ? (int)__ffsll((long long)bad2) - 1 : WAVE;
+                    run += r;
+                    cip -= r;
+                    cmatch -= r;
+                }
+            }
+            mc = run + ext_count(ip, match, true, aa0, bb0);
             ip -= run;
             match -= run;
-            if (run < WAVE) break;
         }
 
         // ================= literals =================
@@ -270,6 +334,8 @@ __device__ inline int lz4_wave_compress(const uint8_t* __restrict__ s_chunk, int
             op += lit;
 
             // ================= offset + match length =================
+            // `mc` for the first iteration was computed above (fused with
+            // catch-up); the immediate-match continue path computes its own.
             while (true) {
                 int offv = ip - match;
                 if (dbg && lane == 0) {
@@ -278,35 +344,6 @@ __device__ inline int lz4_wave_compress(const uint8_t* __restrict__ s_chunk, int
                 }
                 if (lane == 0) { dst[op] = (uint8_t)offv; dst[op + 1] = (uint8_t)(offv >> 8); }
                 op += 2;
-                // match length from ip+4 / match+4 up to matchlimit.
-                // dword-per-lane: 256 bytes per ballot round trip (the loop is
-                // serial — each round's ballot gates the next — so bytes per
-                // round trip is the lever). Per-byte semantics identical to
-                // the scalar loop: a byte at/after matchlimit is a mismatch.
-                int mc = 0;
-                {
-                    int offb = 0;
-                    while (true) {
-                        int pi = ip + 4 + offb + 4 * lane;
-                        int navail = matchlimit - pi;
-                        uint32_t aa = 0, bb = 0;
-                        if (navail > 0) {
-                            aa = lds_read32(s_chunk, (uint32_t)pi);
-                            bb = lds_read32(s_chunk, (uint32_t)(match + 4 + offb + 4 * lane));
-                        }
-                        uint32_t x = aa ^ bb;
-                        int eq4 = navail <= 0 ? 0
-                                              : (x == 0 ? 4 : (__ffs((int)x) - 1) >> 3);
-                        if (eq4 > navail) eq4 = navail;
-                        uint64_t ne = wave_ballot(eq4 < 4);
-                        if (ne) {
-                            int fl = (int)__ffsll((long long)ne) - 1;
-                            mc = offb + 4 * fl + __shfl(eq4, fl);
-                            break;
-                        }
-                        offb += 4 * WAVE;
-                    }
-                }
                 ip += mc + LZ4M_MINMATCH;
                 if (mc >= (int)LZ4M_ML_MASK) {
                     token += LZ4M_ML_MASK;
@@ -323,17 +360,32 @@ __device__ inline int lz4_wave_compress(const uint8_t* __restrict__ s_chunk, int
 
                 anchor = ip;
                 if (ip >= mflimitPlusOne) { done = true; break; }
-                // fill table at ip-2
-                s_table[lz4m_hash(lds_read32(s_chunk, (uint32_t)(ip - 2)))] = (uint16_t)(ip - 2);
-                // immediate test at ip
-                uint32_t v = lds_read32(s_chunk, (uint32_t)ip);
+                // table fill at ip-2 + immediate test at ip. Both dwords live
+                // in w8 (extension registers, or one merged 8-byte load).
+                // The scalar order is fill(ip-2) THEN read the test slot; do
+                // the read first and patch it when the two hashes collide —
+                // the final table state (both writes applied, test slot's ip
+                // winning on collision) is unchanged.
+                uint64_t w;
+                if (w8v) {
+                    w = w8;
+                } else {
+                    uint8_t const* q = s_chunk + (uint32_t)(ip - 2);
+                    memcpy(&w, q, 8);
+                }
+                uint32_t v2 = (uint32_t)w;          // dword at ip-2
+                uint32_t v = (uint32_t)(w >> 16);   // dword at ip
                 uint32_t h = lz4m_hash(v);
                 int m2 = (int)s_table[h];
+                uint32_t h2 = lz4m_hash(v2);
+                if (h2 == h) m2 = ip - 2;
+                s_table[h2] = (uint16_t)(ip - 2);
                 s_table[h] = (uint16_t)ip;
                 if (lds_read32(s_chunk, (uint32_t)m2) == v) {
                     match = m2;
                     token_pos = op++;
                     token = 0;
+                    mc = ext_count(ip, match, false, 0, 0);
                     continue;  // another match with empty literal run
                 }
                 ip += 1;
@@ -378,11 +430,23 @@ __global__ void __launch_bounds__(WAVE) k_lz4_compress_wave_t(const uint8_t* dat
     int csz = lz4_wave_compress(src, (int)len, dst + 4, s_table, lane, dbg);
     uint32_t total = (uint32_t)csz + 4;
     if (lane == 0) csize[c] = total;
-    // CRC over the compressed bytes: wave-parallel slices via per-lane CRC of
-    // strided... (kept lane0 table CRC; compressed bytes ~8KB, minor next to search)
+    // CRC over the compressed bytes on lane 0, slice-by-8 (crc_table is the
+    // 8x256 sliced set): 8 independent table hits per serial round instead
+    // of one — ~8x fewer dependent round trips than the bytewise loop.
     if (lane == 0) {
         uint32_t crc = 0xFFFFFFFFu;
-        for (uint32_t i = 0; i < total; i++) crc = crc_table[(crc ^ dst[i]) & 0xFF] ^ (crc >> 8);
+        uint32_t i = 0;
+        for (; i + 8 <= total; i += 8) {
+            uint32_t lo, hi;
+            memcpy(&lo, dst + i, 4);
+            memcpy(&hi, dst + i + 4, 4);
+            lo ^= crc;
+            crc = crc_table[7 * 256 + (lo & 0xFF)] ^ crc_table[6 * 256 + ((lo >> 8) & 0xFF)] ^
+                  crc_table[5 * 256 + ((lo >> 16) & 0xFF)] ^ crc_table[4 * 256 + (lo >> 24)] ^
+                  crc_table[3 * 256 + (hi & 0xFF)] ^ crc_table[2 * 256 + ((hi >> 8) & 0xFF)] ^
+                  crc_table[1 * 256 + ((hi >> 16) & 0xFF)] ^ crc_table[0 * 256 + (hi >> 24)];
+        }
+        for (; i < total; i++) crc = crc_table[(crc ^ dst[i]) & 0xFF] ^ (crc >> 8);
         ccrc[c] = ~crc;
     }
 }
