@@ -498,6 +498,25 @@ def test_snappy_pipeline(ca, oracle_bin, tmp_path):
     _assert_dirs_equal(f"{d}/oa-80-big", f"{d}/oa-81-big")
 
 
+def test_snappy_generate_parity(ca, oracle_bin, tmp_path):
+    """GPU snappy WRITE path (generate(snappy=True)) == oracle generator,
+    then compact+verify of the result (SnappyCompressor.java:82-86 chunk
+    framing through the product writer kernels). Promoted from the round-1
+    manual check (VERDICT round-2 item 1)."""
+    dg, do = str(tmp_path / "gpu"), str(tmp_path / "cpu")
+    os.makedirs(dg), os.makedirs(do)
+    _oracle_gen(do, seed=11, n=2, rows=2000, vlen=300, overlap=15, tomb=10, snappy=1)
+    ca.generate(dg, seed=11, n_sstables=2, rows_per_sstable=2000, value_len=300,
+                overlap_pct=15, tombstone_pct=10, snappy=True)
+    for g in (1, 2):
+        _assert_dirs_equal(f"{dg}/oa-{g}-big", f"{do}/oa-{g}-big")
+    out = f"{dg}/oa-90-big"
+    ca.compact([f"{dg}/oa-1-big", f"{dg}/oa-2-big"], out)
+    _oracle_compact(f"{do}/oa-90-big", [f"{do}/oa-1-big", f"{do}/oa-2-big"])
+    _assert_dirs_equal(out, f"{do}/oa-90-big")
+    ca.verify(out)
+
+
 def test_cancellation(ca, oracle_bin, tmp_path):
     """Cooperative cancel (CompactionIterator.isStopRequested): a set
     cancel_flag aborts the task with GPUC_ERR_CANCELLED; a zero flag is
