@@ -907,12 +907,19 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
         t_drain0 = tsd.tv_sec * 1e3 + tsd.tv_nsec / 1e6;
     }
     uint64_t file_off = 0;
+    double tw_comp = 0, tw_d2h = 0, tw_wfut = 0;  // drain wait breakdown (GPUC_TRACE)
+    auto wallm = []() {
+        struct timespec t; clock_gettime(CLOCK_MONOTONIC, &t);
+        return t.tv_sec * 1e3 + t.tv_nsec / 1e6;
+    };
     for (uint32_t i = 0; i < n_slabs; i++) {
         uint32_t cb = i * SLAB, m = std::min(SLAB, n_chunks - cb);
         HIP_CHECK(hipStreamWaitEvent(cstream, ev_c[i], 0));
         HIP_CHECK(hipMemcpyAsync(cs.data() + cb, d_csize.as<uint32_t>() + cb, (uint64_t)m * 4,
                                  hipMemcpyDeviceToHost, cstream));
+        double tq0 = wallm();
         HIP_CHECK(hipStreamSynchronize(cstream));
+        tw_comp += wallm() - tq0;
         auto& fo = foff_h[i & 1];
         uint64_t acc = 0;
         for (uint32_t j = 0; j < m; j++) { fo[j] = acc; acc += cs[cb + j]; }
@@ -925,10 +932,14 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
                            d_foff[i & 1].as<uint64_t>(), d_gat[i & 1].as<uint8_t>(), m,
                            SLOT_STRIDE);
         int slot = (int)(i % NSLOTS);
+        double tq1 = wallm();
         if (wfut[slot].valid()) wfut[slot].get();  // pinned buffer free again
+        tw_wfut += wallm() - tq1;
         uint8_t* hbuf = h_slab0 + (uint64_t)slot * worst_slab;
         HIP_CHECK(hipMemcpyAsync(hbuf, d_gat[i & 1].p, slab_bytes, hipMemcpyDeviceToHost, cstream));
+        double tq2 = wallm();
         HIP_CHECK(hipStreamSynchronize(cstream));
+        tw_d2h += wallm() - tq2;
         uint64_t off0 = file_off;
         wfut[slot] = std::async(std::launch::async, [=]() {
             int nth = 12;  // tmpfs/page-cache writes are memcpy-bound and scale with threads
@@ -954,8 +965,15 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
         });
         file_off += slab_bytes;
     }
-    for (int sfin = 0; sfin < NSLOTS; sfin++)
-        if (wfut[sfin].valid()) wfut[sfin].get();
+    {
+        double tq3 = wallm();
+        for (int sfin = 0; sfin < NSLOTS; sfin++)
+            if (wfut[sfin].valid()) wfut[sfin].get();
+        tw_wfut += wallm() - tq3;
+    }
+    if (g_trace)
+        fprintf(stderr, "[drain] slabs=%u wait_compress=%.0fms wait_d2h=%.0fms wait_write=%.0fms\n",
+                n_slabs, tw_comp, tw_d2h, tw_wfut);
     {
         struct timespec tsd; clock_gettime(CLOCK_MONOTONIC, &tsd);
         t_drain1 = tsd.tv_sec * 1e3 + tsd.tv_nsec / 1e6;
@@ -1160,15 +1178,21 @@ struct CompactSetup {
     // parse as ordered STRIPES per source (compact_one adopts them when its
     // window covers the full file and overlaps H2D with the remaining reads)
     struct StripeRead {
-        mutable std::vector<std::thread> th;  // one per stripe, in file order
+        mutable std::vector<std::future<void>> th;  // one per stripe, in file order
         std::vector<uint64_t> off, len;
     };
     mutable std::vector<StripeRead> full_stripes;
+    // file-ordered reader pool: stripes are queued (file 0 first) so the
+    // FIRST sstable's bytes land early and its H2D+decompress pipeline
+    // starts ~1/k into the read instead of after all files finish together
+    mutable std::vector<std::thread> read_pool;
     std::vector<uint8_t*> full_pin;
     ~CompactSetup() {
         for (auto& sr : full_stripes)
             for (auto& t : sr.th)
-                if (t.joinable()) t.join();
+                if (t.valid()) t.wait();
+        for (auto& t : read_pool)
+            if (t.joinable()) t.join();
     }
 };
 
@@ -1283,7 +1307,7 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
                 // (whole-file window only; stripe offsets are file-absolute)
                 auto& sr = su.full_stripes[s];
                 for (size_t t = 0; t < sr.th.size(); t++) {
-                    if (sr.th[t].joinable()) sr.th[t].join();
+                    if (sr.th[t].valid()) sr.th[t].wait();
                     if (sr.len[t])
                         HIP_CHECK(hipMemcpyAsync(d_comp[s].as<uint8_t>() + sr.off[t],
                                                  comp_pin[s] + sr.off[t], sr.len[t],
@@ -1294,7 +1318,7 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
                 // stripes, then one window-sized copy
                 auto& sr = su.full_stripes[s];
                 for (auto& t : sr.th)
-                    if (t.joinable()) t.join();
+                    if (t.valid()) t.wait();
                 if (comp_sz[s])
                     HIP_CHECK(hipMemcpyAsync(d_comp[s].p, comp_pin[s], comp_sz[s],
                                              hipMemcpyHostToDevice, copy_stream));
@@ -1796,6 +1820,8 @@ static void compact_setup(const gpuc_job* job, CompactSetup& su, bool preread_fu
     if (preread_full) {
         su.full_pin.resize(k, nullptr);
         su.full_stripes.resize(k);
+        struct RTask { int s; uint64_t off, len; std::promise<void> done; };
+        auto tasks = std::make_shared<std::vector<RTask>>();
         for (int s = 0; s < k; s++) {
             su.in_bases[s] = s < su.k_data ? job->input_bases[s]
                                            : job->tombstone_source_bases[s - su.k_data];
@@ -1811,10 +1837,25 @@ static void compact_setup(const gpuc_job* job, CompactSetup& su, bool preread_fu
                 uint64_t len = std::min<uint64_t>(per, su.comp_file_sz[s] - o);
                 sr.off.push_back(o);
                 sr.len.push_back(len);
-                sr.th.emplace_back([&su, s, o, len] {
-                    read_file_range(su.in_bases[s] + "-Data.db", su.full_pin[s] + o, o, len, 1);
-                });
+                tasks->push_back(RTask{s, o, len, std::promise<void>()});
+                sr.th.push_back(tasks->back().done.get_future());
             }
+        }
+        // pool drains tasks in queue order (file-major): early files complete
+        // first and the consumer's per-stripe waits overlap the later reads
+        auto next = std::make_shared<std::atomic<size_t>>(0);
+        int nworkers = (int)std::min<size_t>(12, tasks->size());
+        for (int wkr = 0; wkr < nworkers; wkr++) {
+            su.read_pool.emplace_back([&su, tasks, next] {
+                for (;;) {
+                    size_t i = next->fetch_add(1);
+                    if (i >= tasks->size()) return;
+                    RTask& rt = (*tasks)[i];
+                    read_file_range(su.in_bases[rt.s] + "-Data.db", su.full_pin[rt.s] + rt.off,
+                                    rt.off, rt.len, 1);
+                    rt.done.set_value();
+                }
+            });
         }
     }
     {

@@ -185,6 +185,12 @@ __device__ inline int lz4_wave_compress(const uint8_t* __restrict__ s_chunk, int
         }
     };
 
+    // cross-sequence prefetch: the sequence tail issues the next window's
+    // round-0 probe loads before its immediate-test chain resolves (pure
+    // loads at closed-form positions — semantics unchanged)
+    int pre_p = -1;
+    uint32_t pre_v = 0;
+    bool pre_ok = false;
     bool done = false;
     while (!done) {
         // ================= match finder: 64-probe windows =================
@@ -193,13 +199,17 @@ __device__ inline int lz4_wave_compress(const uint8_t* __restrict__ s_chunk, int
             int k0 = 0;           // probe index of this window's lane 0
             int P0 = ip;          // its position
             bool found = false, aborted = false;
-            // preload this window's probe dwords; inside the loop the NEXT
-            // window's dwords are prefetched while the current one resolves
-            // (pure loads: the probe positions are closed-form, independent
-            // of the window outcome, so semantics are unchanged and the
-            // global-load latency hides under the ballot/compare chain)
+            // round-0 probe dwords: taken from the sequence tail's prefetch
+            // when it targeted this ip; inside the loop the NEXT window's
+            // dwords are prefetched while the current one resolves
             int p_l = P0 + lz4_adv_sum(k0, k0 + lane);
-            uint32_t v_l = (p_l >= 0 && p_l + 4 <= srcSize) ? lds_read32(s_chunk, (uint32_t)p_l) : 0;
+            uint32_t v_l;
+            if (pre_ok && pre_p == p_l) {
+                v_l = pre_v;
+            } else {
+                v_l = (p_l >= 0 && p_l + 4 <= srcSize) ? lds_read32(s_chunk, (uint32_t)p_l) : 0;
+            }
+            pre_ok = false;
             while (true) {
                 // scalar loop aborts probe m when q_m + adv_m > mflimitPlusOne
                 bool valid = (p_l + lz4_adv(k0 + lane)) <= mflimitPlusOne;
@@ -360,6 +370,13 @@ __device__ inline int lz4_wave_compress(const uint8_t* __restrict__ s_chunk, int
 
                 anchor = ip;
                 if (ip >= mflimitPlusOne) { done = true; break; }
+                // prefetch the next window's round-0 probes (at ip+1, where
+                // the finder restarts if the immediate test misses) so their
+                // loads resolve under the immediate-test chain; discarded
+                // harmlessly when the test hits
+                pre_p = (ip + 1) + lz4_adv_sum(0, lane);
+                pre_v = (pre_p + 4 <= srcSize) ? lds_read32(s_chunk, (uint32_t)pre_p) : 0;
+                pre_ok = true;
                 // table fill at ip-2 + immediate test at ip. Both dwords live
                 // in w8 (extension registers, or one merged 8-byte load).
                 // The scalar order is fill(ip-2) THEN read the test slot; do
