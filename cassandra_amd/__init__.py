@@ -97,6 +97,7 @@ class GpucGenSpec(ctypes.Structure):
         ("first_generation", ctypes.c_uint64),
         ("device", ctypes.c_int32),
         ("snappy", ctypes.c_int32),
+        ("ttl_pct", ctypes.c_uint32),
     ]
 
 
@@ -337,6 +338,7 @@ def generate(
     static_pct=0,
     n_value_cols=1,
     col_missing_pct=0,
+    ttl_pct=0,
     base_ts=1700000000000000,
     base_ldt=1700000000,
     first_generation=1,
@@ -366,6 +368,7 @@ def generate(
         first_generation=first_generation,
         device=device,
         snappy=1 if snappy else 0,
+        ttl_pct=ttl_pct,
     )
     err = ctypes.create_string_buffer(256)
     rc = lib.gpuc_generate(ctypes.byref(spec), out_dir.encode(), err, 256)

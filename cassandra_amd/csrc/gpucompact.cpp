@@ -2648,6 +2648,7 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
             if (gp.ck_cols > 2) throw std::runtime_error("ck_cols must be 0..2");
             if (gp.ck_cols == 2 && gp.ck_text) throw std::runtime_error("ck_cols=2 with ck_text unsupported");
             gp.col_missing_pct = spec->col_missing_pct;
+            gp.ttl_pct = spec->ttl_pct;
             if (gp.n_value_cols > 63) throw std::runtime_error("n_value_cols must be 1..63");
             if (gp.ck_text && (uint64_t)gp.clustering_rows * 16 >= 100000000ull)
                 throw std::runtime_error("ck_text needs clustering_rows*16 < 1e8 (8-digit order)");
@@ -2702,7 +2703,7 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
             sp.hs.min_ldt = hs0.min_ldt_flip == 0xFFFFFFFFFFFFFFFFULL
                                 ? DELETION_TIME_EPOCH : (int64_t)(hs0.min_ldt_flip ^ 0x8000000000000000ULL);
             if (sp.hs.min_ldt == NO_DELETION_TIME) sp.hs.min_ldt = DELETION_TIME_EPOCH;
-            sp.hs.min_ttl = 0;
+            sp.hs.min_ttl = hs0.min_ttl == 0xFFFFFFFFu ? 0 : (int32_t)hs0.min_ttl;
             sp.sch.n_ck = gen_nck;
             std::vector<int32_t> ckw_h(gen_nck, gp.ck_text ? -1 : 8);
             DevBuf d_ckw;

@@ -2435,6 +2435,7 @@ struct GenParams2 {
     uint32_t static_pct;  // percent of wide partitions with a static row ("s0" blob)
     uint32_t n_value_cols;    // regular columns val0..valN-1 (1..63)
     uint32_t col_missing_pct; // P(cell absent) per live row and column
+    uint32_t ttl_pct;         // P(live row written expiring) — oracle gen_row_expiring
     int64_t base_ts, base_ldt;
 };
 
@@ -2592,6 +2593,20 @@ __global__ void k_gen_fill2(GenParams2 gp, const MRec* sorted, const uint64_t* i
             out.rdel_ldt[o] = tomb_ldt;
             for (uint32_t c = 0; c < NCV; c++) out.cell_flags[o * NCV + c] = 0;
         } else {
+            // oracle contract: gen_row_expiring / gen_ttl / gen_let —
+            // expiring rows carry ExpiringLivenessInfo and their cells
+            // ttl + localDeletionTime == localExpirationTime
+            bool expg = gp.ttl_pct &&
+                        splitmix64(gp.seed ^ 0x771E771EULL ^ id ^ ((uint64_t)gp.sst << 32) ^
+                                   (uint64_t)(rowj + 3) * 101) % 100 < gp.ttl_pct;
+            int32_t ettl = 0;
+            int64_t elet = NO_DELETION_TIME;
+            if (expg) {
+                ettl = (int32_t)(60 + splitmix64(id ^ 0x77AA11ULL ^ (uint64_t)(rowj + 1) * 131) % 86400);
+                elet = gp.base_ldt + (int64_t)(splitmix64(id ^ 0x1E7E1E7EULL ^ (uint64_t)(rowj + 1) * 17) % 2000);
+                out.live_ttl[o] = ettl;
+                out.live_let[o] = elet;
+            }
             out.flags[o] = PF_HAS_ROW | PF_LIVE_TS;
             out.live_ts[o] = ts;
             out.rdel_mfda[o] = INT64_MIN;
@@ -2605,8 +2620,8 @@ __global__ void k_gen_fill2(GenParams2 gp, const MRec* sorted, const uint64_t* i
                 if (miss) { out.cell_flags[oc] = 0; continue; }
                 out.cell_flags[oc] = CELLF_PRESENT | CELLF_HAS_VALUE;
                 out.cell_ts[oc] = ts;
-                out.cell_ldt[oc] = LDT_NONE_U32;
-                out.cell_ttl[oc] = 0;
+                out.cell_ldt[oc] = expg ? (uint32_t)elet : LDT_NONE_U32;
+                out.cell_ttl[oc] = expg ? ettl : 0;
                 out.val_addr[oc] = (uint64_t)(values + oc * (uint64_t)gp.value_len);
                 out.val_len[oc] = gp.value_len;
             }

@@ -160,6 +160,9 @@ typedef struct gpuc_gen_spec {
     uint64_t first_generation;
     int32_t device;
     int32_t snappy;             /* 1: SnappyCompressor chunks (C3 shape) */
+    uint32_t ttl_pct;           /* percent of live rows written with expiring
+                                   liveness/cells (LivenessInfo.java:67,
+                                   AbstractCell.java:53-76) */
 } gpuc_gen_spec;
 
 int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* error, size_t error_len);

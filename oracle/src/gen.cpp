@@ -39,6 +39,7 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
     });
 
     int64_t min_ts = INT64_MAX, min_ldt_l = INT64_MAX;
+    int32_t min_ttl = INT32_MAX;
     for (const Ent& e : ents) {
         if (g.clustering_rows > 0) {
             // ---- wide partition: clustering rows + optional range tombstone ----
@@ -115,11 +116,22 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
                     min_ldt_l = std::min<int64_t>(min_ldt_l, r.del.ldt);
                 } else {
                     r.live.ts = ts;
+                    bool expg = gen_row_expiring(g, sst, e.id, j);
+                    if (expg) {
+                        r.live.ttl = gen_ttl(g, e.id, j);
+                        r.live.let = gen_let(g, e.id, j);
+                        min_ldt_l = std::min(min_ldt_l, r.live.let);
+                        min_ttl = std::min(min_ttl, r.live.ttl);
+                    }
                     uint64_t seed_id = e.id ^ ((uint64_t)j << 52) ^ j;
                     for (uint32_t cc = 0; cc < ncols; cc++) {
                         if (gen_col_missing(g, sst, e.id, j, cc)) continue;
                         Cell cell;
                         cell.ts = ts;
+                        if (expg) {
+                            cell.ttl = r.live.ttl;
+                            cell.ldt = ldt_to_u32(r.live.let);
+                        }
                         gen_value(g, sst, gen_col_value_id(seed_id, cc), cell.value);
                         r.cells[cc] = std::move(cell);
                     }
@@ -172,10 +184,21 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
             min_ldt_l = std::min<int64_t>(min_ldt_l, r.del.ldt);
         } else {
             r.live.ts = ts;
+            bool expg = gen_row_expiring(g, sst, e.id, 0);
+            if (expg) {
+                r.live.ttl = gen_ttl(g, e.id, 0);
+                r.live.let = gen_let(g, e.id, 0);
+                min_ldt_l = std::min(min_ldt_l, r.live.let);
+                min_ttl = std::min(min_ttl, r.live.ttl);
+            }
             for (uint32_t cc = 0; cc < ncols; cc++) {
                 if (gen_col_missing(g, sst, e.id, 0, cc)) continue;
                 Cell c;
                 c.ts = ts;
+                if (expg) {
+                    c.ttl = r.live.ttl;
+                    c.ldt = ldt_to_u32(r.live.let);
+                }
                 gen_value(g, sst, gen_col_value_id(e.id, cc), c.value);
                 r.cells[cc] = std::move(c);
             }
@@ -187,7 +210,7 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
     // EncodingStats.java:78-89)
     t.header.stats.min_ts = min_ts == INT64_MAX ? TIMESTAMP_EPOCH : min_ts;
     t.header.stats.min_ldt = min_ldt_l == INT64_MAX ? DELETION_TIME_EPOCH : min_ldt_l;
-    t.header.stats.min_ttl = 0;
+    t.header.stats.min_ttl = min_ttl == INT32_MAX ? 0 : min_ttl;
     return t;
 }
 

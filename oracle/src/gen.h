@@ -53,6 +53,11 @@ struct GenSpec {
     // (row, column). GPU mirror: GenParams2.n_value_cols/col_missing_pct.
     uint32_t n_value_cols = 1;
     uint32_t col_missing_pct = 0;
+    // ttl_pct: percent of LIVE rows written with EXPIRING liveness/cells
+    // (LivenessInfo.java:67 ExpiringLivenessInfo; cells carry ttl +
+    // localDeletionTime == localExpirationTime, AbstractCell.java:53-76).
+    // GPU mirror: GenParams2.ttl_pct.
+    uint32_t ttl_pct = 0;
 
     uint64_t stride() const { return rows_per_sstable * (100 - overlap_pct) / 100; }
     uint64_t universe() const {
@@ -130,6 +135,21 @@ inline int64_t gen_row_ts(const GenSpec& g, uint32_t sst, uint64_t key_id, uint3
 inline bool gen_row_is_tombstone(const GenSpec& g, uint32_t sst, uint64_t key_id, uint32_t j) {
     if (g.tombstone_pct == 0) return false;
     return splitmix64(g.seed ^ 0xDEADULL ^ key_id ^ ((uint64_t)sst << 32) ^ (uint64_t)(j + 7) * 131) % 100 < g.tombstone_pct;
+}
+// expiring-row derivations (shared contract; GPU mirror in k_gen_fill2)
+inline bool gen_row_expiring(const GenSpec& g, uint32_t sst, uint64_t key_id, uint32_t j) {
+    if (g.ttl_pct == 0) return false;
+    return splitmix64(g.seed ^ 0x771E771EULL ^ key_id ^ ((uint64_t)sst << 32) ^
+                      (uint64_t)(j + 3) * 101) % 100 < g.ttl_pct;
+}
+inline int32_t gen_ttl(const GenSpec& g, uint64_t key_id, uint32_t j) {
+    (void)g;
+    return (int32_t)(60 + splitmix64(key_id ^ 0x77AA11ULL ^ (uint64_t)(j + 1) * 131) % 86400);
+}
+// localExpirationTime: seconds near base_ldt so tests can place `now` on
+// either side of expiry
+inline int64_t gen_let(const GenSpec& g, uint64_t key_id, uint32_t j) {
+    return g.base_ldt + (int64_t)(splitmix64(key_id ^ 0x1E7E1E7EULL ^ (uint64_t)(j + 1) * 17) % 2000);
 }
 inline bool gen_has_static(const GenSpec& g, uint32_t sst, uint64_t key_id) {
     if (g.static_pct == 0) return false;

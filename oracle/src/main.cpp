@@ -113,6 +113,7 @@ static GenSpec spec_from_kv(std::map<std::string, std::string>& kv) {
     g.col_missing_pct = (uint32_t)geti("colmiss", g.col_missing_pct);
     g.first_generation = geti("gen0", g.first_generation);
     g.snappy = (uint32_t)geti("snappy", g.snappy);
+    g.ttl_pct = (uint32_t)geti("ttl", g.ttl_pct);
     return g;
 }
 

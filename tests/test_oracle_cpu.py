@@ -395,3 +395,26 @@ def test_bti_partition_index_writer(tmp_path):
     r = subprocess.run([exe, fix, clust], capture_output=True, text=True)
     assert r.returncode == 0, r.stdout + r.stderr
     assert "all OK" in r.stdout
+
+
+def test_ttl_generator_roundtrip(oracle_bin, tmp_path):
+    """ttl_pct rows write ExpiringLivenessInfo + expiring cells
+    (LivenessInfo.java:67, AbstractCell.java:53-76) and the components
+    re-serialize byte-identically; compaction across the expiry boundary
+    converts then purges (Cells.java purge chain)."""
+    import json
+    d = str(tmp_path)
+    oracle_run("gen", d, "seed=51", "n=2", "rows=500", "vlen=100", "overlap=20",
+               "tomb=10", "ttl=35")
+    for g in (1, 2):
+        r = oracle_run("roundtrip", f"{d}/oa-{g}-big")
+        assert "MISMATCH" not in r.stdout and "OK  Data.db" in r.stdout, r.stdout
+    outs = []
+    for i, extra in enumerate((["now=1699999999"],
+                               ["now=1800000000"],
+                               ["now=1800000000", "gcbefore=1800000000"])):
+        r = oracle_run("compact", f"{d}/oa-{60+i}-big", f"{d}/oa-1-big", f"{d}/oa-2-big", *extra)
+        outs.append(json.loads(r.stdout.strip().splitlines()[-1])["rows_out"])
+    # live-expiring keeps every merged row; expired-unpurgeable converts but
+    # keeps (tombstones retained at gcBefore=MIN); expired+purgeable drops
+    assert outs[0] == outs[1] and outs[2] < outs[1], outs
