@@ -74,7 +74,8 @@ static const Cell& reconcile_cells(const Cell& left, const Cell& right) {
 // Row.Merger.merge (Row.java:730-781) — simple columns only
 // ---------------------------------------------------------------------------
 static bool row_merge(const std::vector<const Row*>& versions, const DeletionTime& active_deletion,
-                      size_t ncols, Row& out) {
+                      const std::vector<std::pair<bytes, CqlType>>& cols, Row& out) {
+    const size_t ncols = cols.size();
     // returns false if merged row is null (fully shadowed/empty)
     int present = 0;
     const Row* last = nullptr;
@@ -102,8 +103,54 @@ static bool row_merge(const std::vector<const Row*>& versions, const DeletionTim
     out.live = info;
     out.del = row_del;
     out.cells.assign(ncols, std::nullopt);
+    bool any_cpx_col = false;
+    for (auto& cp : cols) any_cpx_col |= is_complex_type(cp.second);
+    if (any_cpx_col) out.complex.assign(ncols, std::nullopt);
     bool any_cell = false;
     for (size_t ci = 0; ci < ncols; ci++) {
+        if (is_complex_type(cols[ci].second)) {
+            // ColumnDataReducer complex branch (Row.java:851-884)
+            std::vector<const ComplexData*> cds;
+            for (auto* r : versions)
+                if (r && ci < r->complex.size() && r->complex[ci]) cds.push_back(&*r->complex[ci]);
+            if (cds.empty()) continue;
+            DeletionTime complex_del;  // LIVE
+            for (auto* cd : cds)
+                if (cd->del.supersedes(complex_del)) complex_del = cd->del;
+            ComplexData out_cd;
+            DeletionTime cell_active = active;
+            if (complex_del.supersedes(active)) {
+                cell_active = complex_del;
+                out_cd.del = complex_del;  // addComplexDeletion
+            }
+            // k-way path merge with CellReducer (Row.java:893-909): per path,
+            // versions reduce in source order, cells deleted by cell_active skip
+            std::vector<size_t> cpos(cds.size(), 0);
+            while (true) {
+                const bytes* minp = nullptr;
+                for (size_t i = 0; i < cds.size(); i++) {
+                    if (cpos[i] >= cds[i]->cells.size()) continue;
+                    const bytes& pth = cds[i]->cells[cpos[i]].path;
+                    if (!minp || compare_cell_path(pth, *minp) < 0) minp = &pth;
+                }
+                if (!minp) break;
+                const Cell* merged = nullptr;
+                for (size_t i = 0; i < cds.size(); i++) {
+                    if (cpos[i] >= cds[i]->cells.size()) continue;
+                    const Cell& cell = cds[i]->cells[cpos[i]];
+                    if (compare_cell_path(cell.path, *minp) != 0) continue;
+                    cpos[i]++;
+                    if (cell_active.deletes(cell.ts)) continue;
+                    merged = merged == nullptr ? &cell : &reconcile_cells(*merged, cell);
+                }
+                if (merged) out_cd.cells.push_back(*merged);
+            }
+            // Builder.build: live deletion + no cells -> null column
+            if (out_cd.del.live() && out_cd.cells.empty()) continue;
+            out.complex[ci] = std::move(out_cd);
+            any_cell = true;
+            continue;
+        }
         const Cell* merged = nullptr;
         for (auto* r : versions) {  // version order == source order (reduce call order)
             if (!r || ci >= r->cells.size() || !r->cells[ci]) continue;
@@ -195,7 +242,7 @@ Partition merge_partition_versions(const std::vector<const Partition*>& versions
             if (!row_is_empty(p->static_row)) svs.push_back(&p->static_row);
         if (!svs.empty()) {
             Row merged_static;
-            if (row_merge(svs, del, h.static_cols.size(), merged_static))
+            if (row_merge(svs, del, h.static_cols, merged_static))
                 out.static_row = std::move(merged_static);
             out.static_row.static_flag = true;
         }
@@ -204,7 +251,7 @@ Partition merge_partition_versions(const std::vector<const Partition*>& versions
     size_t k = versions.size();
     std::vector<size_t> pos(k, 0);
     MarkerMerger marker_merger(k, del);
-    size_t ncols = h.regular_cols.size();
+    const auto& rcols = h.regular_cols;
     while (true) {
         // find min position among streams
         int min_src = -1;
@@ -232,7 +279,7 @@ Partition merge_partition_versions(const std::vector<const Partition*>& versions
         }
         if (is_row) {
             Row merged;
-            if (row_merge({row_versions.begin(), row_versions.end()}, marker_merger.active_deletion(), ncols, merged)) {
+            if (row_merge({row_versions.begin(), row_versions.end()}, marker_merger.active_deletion(), rcols, merged)) {
                 Unfiltered u;
                 u.kind = Unfiltered::ROW;
                 u.row = std::move(merged);
@@ -309,6 +356,20 @@ static bool purge_row(Row& r, const Purger& pg, bool enforce_strict_liveness) {
         if (!oc) continue;
         oc = purge_cell(*oc, pg);
         if (oc) any = true;
+    }
+    for (auto& ocd : r.complex) {
+        if (!ocd) continue;
+        // ComplexColumnData.purge (ComplexColumnData.java:212-216): purge the
+        // complex deletion, purge each cell, null the column when empty
+        if (pg.should_purge(ocd->del)) ocd->del = DeletionTime{};
+        std::vector<Cell> kept;
+        for (const Cell& c : ocd->cells) {
+            auto p2 = purge_cell(c, pg);
+            if (p2) { p2->path = c.path; kept.push_back(std::move(*p2)); }
+        }
+        ocd->cells = std::move(kept);
+        if (ocd->del.live() && ocd->cells.empty()) ocd.reset();
+        else any = true;
     }
     return !(r.live.empty() && r.del.live() && !any);  // empty row -> null
 }
@@ -551,6 +612,11 @@ static Header make_output_header(const std::vector<SSTable>& inputs) {
 // top-level compaction (CompactionTask.runMayThrow hot loop semantics)
 // ---------------------------------------------------------------------------
 CompactionResult compact(const CompactionJob& job) {
+    if (!job.tomb_sources.empty() && !job.inputs.empty())
+        for (auto& cp : job.inputs[0].header.regular_cols)
+            if (is_complex_type(cp.second))
+                throw std::runtime_error(
+                    "garbage-collect mode with complex columns unsupported");
     CompactionResult res;
     res.out.header = make_output_header(job.inputs);
     res.out.comp = job.inputs.empty() ? CompressionParams{} : job.inputs[0].comp;

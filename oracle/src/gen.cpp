@@ -23,6 +23,8 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
             t.header.regular_cols.push_back({bytes(nm.begin(), nm.end()), CqlType::BYTES});
         }
     }
+    if (g.complex_pct > 0)  // one complex column, named to sort last
+        t.header.regular_cols.push_back({bytes{'z', 'm'}, CqlType::MAP_BB});
 
     struct Ent { int64_t token; bytes key; uint64_t id; };
     std::vector<Ent> ents;
@@ -40,6 +42,37 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
 
     int64_t min_ts = INT64_MAX, min_ldt_l = INT64_MAX;
     int32_t min_ttl = INT32_MAX;
+    // complex column "zm": dedup+sorted map cells, optional complexDeletion
+    auto put_complex = [&](Row& r, uint32_t sst2, uint64_t id, uint32_t j, int64_t ts) {
+        if (!gen_has_complex(g, sst2, id, j)) return;
+        size_t ci = t.header.regular_cols.size() - 1;
+        r.cells.resize(t.header.regular_cols.size());
+        r.complex.resize(t.header.regular_cols.size());
+        ComplexData cd;
+        if (gen_has_cpx_del(g, sst2, id, j)) {
+            cd.del.mfda = ts - 1;
+            cd.del.ldt = gen_ldt(g, id * 5 + j, 0xCD);
+            min_ts = std::min(min_ts, cd.del.mfda);
+            min_ldt_l = std::min<int64_t>(min_ldt_l, cd.del.ldt);
+        }
+        uint32_t nc = gen_cpx_count(g, id, j);
+        std::vector<uint32_t> pv;
+        for (uint32_t e = 0; e < nc; e++) pv.push_back(gen_cpx_path_val(g, sst2, id, j, e));
+        std::sort(pv.begin(), pv.end());
+        pv.erase(std::unique(pv.begin(), pv.end()), pv.end());
+        for (uint32_t p4 : pv) {
+            Cell c;
+            c.ts = ts - (int64_t)(p4 % 3);
+            min_ts = std::min(min_ts, c.ts);
+            c.path.resize(4);
+            for (int b = 0; b < 4; b++) c.path[b] = (uint8_t)(p4 >> (8 * (3 - b)));
+            uint64_t w = gen_cpx_value_word(g, sst2, id, j, p4);
+            c.value.resize(8);
+            for (int b = 0; b < 8; b++) c.value[b] = (uint8_t)(w >> (8 * b));
+            cd.cells.push_back(std::move(c));
+        }
+        r.complex[ci] = std::move(cd);
+    };
     for (const Ent& e : ents) {
         if (g.clustering_rows > 0) {
             // ---- wide partition: clustering rows + optional range tombstone ----
@@ -107,7 +140,7 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
                 Row& r = u.row;
                 r.clustering = ck_vals(ck, true);
                 uint32_t ncols = g.n_value_cols ? g.n_value_cols : 1;
-                r.cells.resize(ncols);
+                r.cells.resize(t.header.regular_cols.size());
                 int64_t ts = gen_row_ts(g, sst, e.id, j);
                 min_ts = std::min(min_ts, ts);
                 if (gen_row_is_tombstone(g, sst, e.id, j)) {
@@ -135,6 +168,7 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
                         gen_value(g, sst, gen_col_value_id(seed_id, cc), cell.value);
                         r.cells[cc] = std::move(cell);
                     }
+                    put_complex(r, sst, e.id, j, ts);
                 }
                 p.items.push_back(std::move(u));
                 int64_t ck_next = ck2 ? (int64_t)((j + 1) / 4) * 64 + (int64_t)((j + 1) % 4)
@@ -177,7 +211,7 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
         u.kind = Unfiltered::ROW;
         Row& r = u.row;
         uint32_t ncols = g.n_value_cols ? g.n_value_cols : 1;
-        r.cells.resize(ncols);
+        r.cells.resize(t.header.regular_cols.size());
         if (gen_is_tombstone(g, sst, e.id)) {
             r.del.mfda = ts;
             r.del.ldt = gen_ldt(g, e.id, 0xEE);
@@ -202,6 +236,7 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
                 gen_value(g, sst, gen_col_value_id(e.id, cc), c.value);
                 r.cells[cc] = std::move(c);
             }
+            put_complex(r, sst, e.id, 0, ts);
         }
         p.items.push_back(std::move(u));
         t.parts.push_back(std::move(p));

@@ -53,6 +53,14 @@ struct GenSpec {
     // (row, column). GPU mirror: GenParams2.n_value_cols/col_missing_pct.
     uint32_t n_value_cols = 1;
     uint32_t col_missing_pct = 0;
+    // complex_pct: percent of LIVE rows carrying cells in ONE complex column
+    // "zm" map<blob,blob> (MapType(BytesType,BytesType)) appended after the
+    // simple value columns ('z' keeps it last in name order). Paths are drawn
+    // from a small space so versions of a key collide and exercise per-path
+    // reconcile; complex_del_pct of those rows also carry a complexDeletion
+    // (shadowing ts-older cells on merge). GPU mirror: GenParams2.complex_pct.
+    uint32_t complex_pct = 0;
+    uint32_t complex_del_pct = 0;
     // ttl_pct: percent of LIVE rows written with EXPIRING liveness/cells
     // (LivenessInfo.java:67 ExpiringLivenessInfo; cells carry ttl +
     // localDeletionTime == localExpirationTime, AbstractCell.java:53-76).
@@ -150,6 +158,32 @@ inline int32_t gen_ttl(const GenSpec& g, uint64_t key_id, uint32_t j) {
 // either side of expiry
 inline int64_t gen_let(const GenSpec& g, uint64_t key_id, uint32_t j) {
     return g.base_ldt + (int64_t)(splitmix64(key_id ^ 0x1E7E1E7EULL ^ (uint64_t)(j + 1) * 17) % 2000);
+}
+// complex-column derivations (GPU mirror: k_gen_fill2 / k_gen_values2)
+inline bool gen_has_complex(const GenSpec& g, uint32_t sst, uint64_t key_id, uint32_t j) {
+    if (g.complex_pct == 0) return false;
+    return splitmix64(g.seed ^ 0xC0113C71ULL ^ key_id ^ ((uint64_t)sst << 32) ^
+                      (uint64_t)(j + 5) * 157) % 100 < g.complex_pct;
+}
+inline bool gen_has_cpx_del(const GenSpec& g, uint32_t sst, uint64_t key_id, uint32_t j) {
+    if (g.complex_del_pct == 0) return false;
+    return splitmix64(g.seed ^ 0xCDE1CDE1ULL ^ key_id ^ ((uint64_t)sst << 32) ^
+                      (uint64_t)(j + 2) * 211) % 100 < g.complex_del_pct;
+}
+// candidate cell count 1..4 BEFORE path dedup (paths from a 40-value space,
+// so intra-row duplicates collapse and cross-sstable versions collide)
+inline uint32_t gen_cpx_count(const GenSpec& g, uint64_t key_id, uint32_t j) {
+    (void)g;
+    return 1 + (uint32_t)(splitmix64(key_id ^ 0xCE11C07ULL ^ (uint64_t)(j + 1) * 19) % 4);
+}
+inline uint32_t gen_cpx_path_val(const GenSpec& g, uint32_t sst, uint64_t key_id, uint32_t j, uint32_t e) {
+    (void)g;
+    return (uint32_t)(splitmix64(key_id ^ 0x9A7B9A7BULL ^ ((uint64_t)sst << 24) ^
+                                 (uint64_t)(j + 1) * 23 ^ (uint64_t)(e + 1) * 71) % 40);
+}
+inline uint64_t gen_cpx_value_word(const GenSpec& g, uint32_t sst, uint64_t key_id, uint32_t j, uint32_t pathv) {
+    return splitmix64(g.seed ^ key_id * 131 ^ ((uint64_t)sst << 40) ^ (uint64_t)(j + 1) * 29 ^
+                      (uint64_t)(pathv + 1) * 389);
 }
 inline bool gen_has_static(const GenSpec& g, uint32_t sst, uint64_t key_id) {
     if (g.static_pct == 0) return false;

@@ -114,6 +114,8 @@ static GenSpec spec_from_kv(std::map<std::string, std::string>& kv) {
     g.first_generation = geti("gen0", g.first_generation);
     g.snappy = (uint32_t)geti("snappy", g.snappy);
     g.ttl_pct = (uint32_t)geti("ttl", g.ttl_pct);
+    g.complex_pct = (uint32_t)geti("cpx", g.complex_pct);
+    g.complex_del_pct = (uint32_t)geti("cpxdel", g.complex_del_pct);
     return g;
 }
 

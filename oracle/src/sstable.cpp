@@ -20,11 +20,16 @@ const char* cql_type_name(CqlType t) {
         case CqlType::ASCII: return "org.apache.cassandra.db.marshal.AsciiType";
         case CqlType::LONG: return "org.apache.cassandra.db.marshal.LongType";
         case CqlType::INT32: return "org.apache.cassandra.db.marshal.Int32Type";
+        case CqlType::MAP_BB:
+            return "org.apache.cassandra.db.marshal.MapType"
+                   "(org.apache.cassandra.db.marshal.BytesType,"
+                   "org.apache.cassandra.db.marshal.BytesType)";
     }
     return "?";
 }
 CqlType cql_type_from_name(const std::string& n) {
-    for (CqlType t : {CqlType::BYTES, CqlType::UTF8, CqlType::ASCII, CqlType::LONG, CqlType::INT32})
+    for (CqlType t : {CqlType::BYTES, CqlType::UTF8, CqlType::ASCII, CqlType::LONG, CqlType::INT32,
+                      CqlType::MAP_BB})
         if (n == cql_type_name(t)) return t;
     throw std::runtime_error("oracle: unsupported AbstractType " + n);
 }
@@ -135,7 +140,7 @@ enum : uint8_t {
 
 // ---- cell serialize (Cell.java:268-306) ----
 static void put_cell(bytes& out, const Cell& c, CqlType type, const LivenessInfo& row_live,
-                     const Header& h) {
+                     const Header& h, bool complex = false) {
     bool has_value = !c.value.empty();
     bool is_deleted = c.tombstone();
     bool is_expiring = c.expiring();
@@ -152,9 +157,14 @@ static void put_cell(bytes& out, const Cell& c, CqlType type, const LivenessInfo
     if (!use_row_ts) h.w_ts(out, c.ts);
     if ((is_deleted || is_expiring) && !use_row_ttl) h.w_ldt(out, ldt_to_long(c.ldt));
     if (is_expiring && !use_row_ttl) h.w_ttl(out, c.ttl);
-    if (has_value) put_typed_value(out, type, c.value);
+    if (complex) {  // CellPath: vint length + bytes (CollectionType.java:401-404)
+        put_unsigned_vint(out, c.path.size());
+        out.insert(out.end(), c.path.begin(), c.path.end());
+    }
+    if (has_value) put_typed_value(out, complex ? CqlType::BYTES : type, c.value);
 }
-static Cell read_cell(Reader& r, CqlType type, const LivenessInfo& row_live, const Header& h) {
+static Cell read_cell(Reader& r, CqlType type, const LivenessInfo& row_live, const Header& h,
+                      bool complex = false) {
     uint8_t flags = r.u8();
     bool has_value = !(flags & CF_HAS_EMPTY_VALUE);
     bool is_deleted = flags & CF_IS_DELETED;
@@ -167,18 +177,21 @@ static Cell read_cell(Reader& r, CqlType type, const LivenessInfo& row_live, con
                               : ((is_deleted || is_expiring) ? h.r_ldt(r) : NO_DELETION_TIME);
     c.ttl = use_row_ttl ? row_live.ttl : (is_expiring ? h.r_ttl(r) : NO_TTL);
     c.ldt = ldt_to_u32(ldt);
-    if (has_value) c.value = read_typed_value(r, type);
+    if (complex) {
+        uint64_t plen = read_unsigned_vint(r);
+        c.path = r.take((size_t)plen);
+    }
+    if (has_value) c.value = read_typed_value(r, complex ? CqlType::BYTES : type);
     return c;
 }
 
 // ---- columns subset (Columns.java:503-608) ----
-static void put_column_subset(bytes& out, const std::vector<std::optional<Cell>>& cells,
-                              size_t superset_count) {
+static void put_column_subset(bytes& out, const Row& row, size_t superset_count) {
     // encodeBitmap: bit i set == superset column i MISSING (Columns.java:586-608)
     if (superset_count >= 64) throw std::runtime_error("large column subsets unsupported");
     uint64_t bitmap = 0;
     for (size_t i = 0; i < superset_count; i++)
-        if (!cells[i]) bitmap |= 1ULL << i;
+        if (!row.col_present(i)) bitmap |= 1ULL << i;
     put_unsigned_vint(out, bitmap);
 }
 static uint64_t read_column_subset_bitmap(Reader& r, size_t superset_count) {
@@ -192,21 +205,33 @@ static void put_row_body(bytes& out, const Row& row, const Header& h, uint8_t fl
     if (flags & F_HAS_TTL) { h.w_ttl(out, row.live.ttl); h.w_ldt(out, row.live.let); }
     if (flags & F_HAS_DELETION) h.w_dt(out, row.del);
     const auto& cols = row.static_flag ? h.static_cols : h.regular_cols;
-    if (!(flags & F_HAS_ALL_COLUMNS)) put_column_subset(out, row.cells, cols.size());
-    for (size_t i = 0; i < cols.size(); i++)
-        if (row.cells[i]) put_cell(out, *row.cells[i], cols[i].second, row.live, h);
+    if (!(flags & F_HAS_ALL_COLUMNS)) put_column_subset(out, row, cols.size());
+    for (size_t i = 0; i < cols.size(); i++) {
+        if (is_complex_type(cols[i].second)) {
+            // UnfilteredSerializer.writeComplexColumn (UnfilteredSerializer.java:271-280)
+            if (i >= row.complex.size() || !row.complex[i]) continue;
+            const ComplexData& cd = *row.complex[i];
+            if (flags & F_HAS_COMPLEX_DELETION) h.w_dt(out, cd.del);
+            put_unsigned_vint(out, cd.cells.size());
+            for (const Cell& c : cd.cells) put_cell(out, c, cols[i].second, row.live, h, true);
+        } else if (row.cells[i]) {
+            put_cell(out, *row.cells[i], cols[i].second, row.live, h);
+        }
+    }
 }
 
 // row flag computation (UnfilteredSerializer.java:151-185)
 static uint8_t row_flags(const Row& row, const Header& h) {
     uint8_t flags = 0;
     size_t present = 0;
-    for (auto& c : row.cells) if (c) present++;
+    size_t ncols = (row.static_flag ? h.static_cols : h.regular_cols).size();
+    for (size_t i = 0; i < ncols; i++) if (row.col_present(i)) present++;
     if (!row.live.empty()) flags |= F_HAS_TIMESTAMP;
     if (row.live.expiring()) flags |= F_HAS_TTL;
     if (!row.del.live()) flags |= F_HAS_DELETION;
-    if (present == (row.static_flag ? h.static_cols : h.regular_cols).size())
-        flags |= F_HAS_ALL_COLUMNS;
+    if (present == ncols) flags |= F_HAS_ALL_COLUMNS;
+    for (auto& cd : row.complex)
+        if (cd && !cd->del.live()) { flags |= F_HAS_COMPLEX_DELETION; break; }
     return flags;
 }
 
@@ -257,11 +282,23 @@ static Row read_row(Reader& r, const Header& h, uint8_t flags, bool is_static_ro
     const auto& cols = is_static_row ? h.static_cols : h.regular_cols;
     uint64_t missing = 0;
     if (!(flags & F_HAS_ALL_COLUMNS)) missing = read_column_subset_bitmap(r, cols.size());
-    if (flags & F_HAS_COMPLEX_DELETION) throw std::runtime_error("complex columns unsupported");
     row.cells.resize(cols.size());
+    bool any_cpx = false;
+    for (auto& cp : cols) any_cpx |= is_complex_type(cp.second);
+    if (any_cpx) row.complex.resize(cols.size());
     for (size_t i = 0; i < cols.size(); i++) {
         if (missing & (1ULL << i)) continue;
-        row.cells[i] = read_cell(r, cols[i].second, row.live, h);
+        if (is_complex_type(cols[i].second)) {
+            ComplexData cd;
+            if (flags & F_HAS_COMPLEX_DELETION) cd.del = h.r_dt(r);
+            uint64_t n = read_unsigned_vint(r);
+            cd.cells.reserve(n);
+            for (uint64_t ci = 0; ci < n; ci++)
+                cd.cells.push_back(read_cell(r, cols[i].second, row.live, h, true));
+            row.complex[i] = std::move(cd);
+        } else {
+            row.cells[i] = read_cell(r, cols[i].second, row.live, h);
+        }
     }
     return row;
 }
@@ -688,7 +725,7 @@ WriterOut write_sstable(const SSTable& t) {
     Bloom bloom = make_bloom(t.parts.size(), 0.01);
     StatsComponentInput st;
     st.mins = StatsMins{};
-    uint64_t total_cells = 0, total_rows = 0;
+    uint64_t total_cells = 0, total_rows = 0, total_cols_set = 0;
     bool has_partition_deletions = false;
     for (const Partition& p : t.parts) {
         size_t before = data_raw.size();
@@ -700,7 +737,12 @@ WriterOut write_sstable(const SSTable& t) {
         for (auto& u : p.items) {
             if (u.kind != Unfiltered::ROW) continue;
             total_rows++;
-            for (auto& c : u.row.cells) if (c) { cells++; }
+            // Rows.collectStats: totalColumnsSet counts COLUMNS (a complex
+            // column once when it has >=1 cell); the per-partition histogram
+            // counts CELLS (each complex cell) — Rows.java:60-82
+            for (auto& c : u.row.cells) if (c) { cells++; total_cols_set++; }
+            for (auto& cd : u.row.complex)
+                if (cd && !cd->cells.empty()) { cells += cd->cells.size(); total_cols_set++; }
         }
         total_cells += cells;
         st.cells_per_partition.add(cells);
@@ -729,12 +771,23 @@ WriterOut write_sstable(const SSTable& t) {
                 else if (c->expiring()) { upd_ldt(ldt_to_long(c->ldt)); upd_ttl(c->ttl); }
                 else upd_ldt(NO_DELETION_TIME);
             }
+            for (auto& cd : r.complex) {
+                if (!cd) continue;
+                // MetadataCollector.update(complexDeletion) — Rows.java:75
+                if (!cd->del.live()) { upd_ts(cd->del.mfda); upd_ldt(ldt_to_long(cd->del.ldt)); upd_tomb(cd->del.ldt); }
+                for (const Cell& c : cd->cells) {
+                    upd_ts(c.ts);
+                    if (c.tombstone()) { upd_ldt(ldt_to_long(c.ldt)); upd_tomb(c.ldt); }
+                    else if (c.expiring()) { upd_ldt(ldt_to_long(c.ldt)); upd_ttl(c.ttl); }
+                    else upd_ldt(NO_DELETION_TIME);
+                }
+            }
         }
     }
     if (st.mins.min_timestamp == INT64_MAX) { st.mins.min_timestamp = 0; st.mins.max_timestamp = 0; }
     if (st.mins.min_ttl == INT32_MAX) st.mins.min_ttl = 0;
     st.mins.total_rows = total_rows;
-    st.mins.total_columns_set = total_cells;
+    st.mins.total_columns_set = total_cols_set;
     st.clustering_count = t.header.clustering_types.size();
     if (st.clustering_count) st.clustering_type = t.header.clustering_types[0];
     if (!t.parts.empty()) { st.first_key = t.parts.front().key; st.last_key = t.parts.back().key; }

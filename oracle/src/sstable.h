@@ -27,7 +27,11 @@ inline int64_t ldt_to_long(uint32_t u) { return u == LDT_NONE_U32 ? NO_DELETION_
 inline uint32_t ldt_to_u32(int64_t l) { return l == NO_DELETION_TIME ? LDT_NONE_U32 : (uint32_t)l; }
 
 // ---- types (db/marshal/*) ----
-enum class CqlType : uint8_t { BYTES, UTF8, ASCII, LONG, INT32 };
+enum class CqlType : uint8_t { BYTES, UTF8, ASCII, LONG, INT32, MAP_BB };
+// MAP_BB == MapType(BytesType,BytesType): the one COMPLEX column type the
+// engine carries (ColumnMetadata.isComplex; db/rows/ComplexColumnData.java:47).
+// Cell paths are the map keys, compared as BytesType (unsigned lexicographic).
+inline bool is_complex_type(CqlType t) { return t == CqlType::MAP_BB; }
 const char* cql_type_name(CqlType t);
 CqlType cql_type_from_name(const std::string& java_name);
 inline int fixed_len(CqlType t) {
@@ -74,6 +78,7 @@ struct Cell {
     uint32_t ldt = LDT_NONE_U32;  // localDeletionTime (u32 encoding)
     int32_t ttl = NO_TTL;
     bytes value;                  // empty + !has_value -> HAS_EMPTY_VALUE
+    bytes path;                   // CellPath (complex columns only): map key
     bool tombstone() const { return ldt != LDT_NONE_U32 && ttl == NO_TTL; }
     bool expiring() const { return ttl != NO_TTL; }
     // AbstractCell.isLive(nowInSec)
@@ -97,6 +102,22 @@ enum BoundKind : uint8_t {
     CLUSTERING_K = 4, INCL_END_EXCL_START = 5, INCL_END = 6, EXCL_START = 7
 };
 
+// CellPath comparator: map keys compare as the key type (BytesType here)
+inline int compare_cell_path(const bytes& a, const bytes& b) {
+    size_t n = a.size() < b.size() ? a.size() : b.size();
+    int c = n ? __builtin_memcmp(a.data(), b.data(), n) : 0;
+    if (c) return c;
+    return a.size() == b.size() ? 0 : (a.size() < b.size() ? -1 : 1);
+}
+
+// db/rows/ComplexColumnData.java:47 — a complex column's deletion + its
+// path-ordered cells. A ComplexData with live deletion and no cells is
+// represented as absent (ComplexColumnData.update returns null then).
+struct ComplexData {
+    DeletionTime del;          // complexDeletion
+    std::vector<Cell> cells;   // ordered by compare_cell_path
+};
+
 struct Row;
 inline bool row_is_empty(const Row& r);
 struct Row {
@@ -105,15 +126,25 @@ struct Row {
     DeletionTime del;
     bool static_flag = false;
     std::vector<std::optional<Cell>> cells;  // index == header regular-column index
+    // complex column data, same indexing (empty vector when the schema has
+    // no complex columns; entries at simple-column indices stay nullopt)
+    std::vector<std::optional<ComplexData>> complex;
+    bool col_present(size_t i) const {
+        if (i < cells.size() && cells[i]) return true;
+        return i < complex.size() && complex[i].has_value();
+    }
     bool empty_row() const {
         if (!live.empty() || !del.live()) return false;
         for (auto& c : cells) if (c) return false;
+        for (auto& c : complex) if (c) return false;
         return true;
     }
 };
 inline bool row_is_empty(const Row& r) {
     if (!r.live.empty() || !r.del.live()) return false;
     for (auto& c : r.cells)
+        if (c) return false;
+    for (auto& c : r.complex)
         if (c) return false;
     return true;
 }
