@@ -98,6 +98,8 @@ class GpucGenSpec(ctypes.Structure):
         ("device", ctypes.c_int32),
         ("snappy", ctypes.c_int32),
         ("ttl_pct", ctypes.c_uint32),
+        ("complex_pct", ctypes.c_uint32),
+        ("complex_del_pct", ctypes.c_uint32),
     ]
 
 
@@ -339,6 +341,8 @@ def generate(
     n_value_cols=1,
     col_missing_pct=0,
     ttl_pct=0,
+    complex_pct=0,
+    complex_del_pct=0,
     base_ts=1700000000000000,
     base_ldt=1700000000,
     first_generation=1,
@@ -369,6 +373,8 @@ def generate(
         device=device,
         snappy=1 if snappy else 0,
         ttl_pct=ttl_pct,
+        complex_pct=complex_pct,
+        complex_del_pct=complex_del_pct,
     )
     err = ctypes.create_string_buffer(256)
     rc = lib.gpuc_generate(ctypes.byref(spec), out_dir.encode(), err, 256)

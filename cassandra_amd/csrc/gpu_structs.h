@@ -90,8 +90,27 @@ struct UnfCols {
     uint64_t* ck_addr;
     uint32_t* ck_len;
     uint8_t* ck_count;
+    // ONE complex column (MapType(BytesType,BytesType)), the LAST regular
+    // column when SchemaParams.n_cpx == 1 (ComplexColumnData.java:47): a
+    // per-row complexDeletion + a [cpx_start, cpx_start+cpx_count) segment
+    // of path-ordered cells in the CpxCells arena. PF_HAS_CPX marks column
+    // presence (deletion-only columns have cpx_count == 0).
+    int64_t* cpx_del_mfda;
+    uint32_t* cpx_del_ldt;
+    uint64_t* cpx_start;
+    uint32_t* cpx_count;
+    struct CpxCells {
+        int64_t* ts;
+        uint32_t* ldt;
+        int32_t* ttl;
+        uint8_t* flags;       // CELLF_*
+        uint64_t* path_addr;  // CellPath bytes (map key)
+        uint32_t* path_len;
+        uint64_t* val_addr;
+        uint32_t* val_len;
+    } cpx;
 };
-enum : uint8_t { PF_HAS_ROW = 1, PF_LIVE_TS = 2, PF_ROW_DEL = 4 };
+enum : uint8_t { PF_HAS_ROW = 1, PF_LIVE_TS = 2, PF_ROW_DEL = 4, PF_HAS_CPX = 8 };
 // per-cell flags (cell_flags array)
 enum : uint8_t { CELLF_PRESENT = 1, CELLF_HAS_VALUE = 2, CELLF_EXPIRING = 4 };
 // ClusteringPrefix.Kind ordinals (ClusteringPrefix.java:65-85)
@@ -137,6 +156,7 @@ struct ParsedCols {
     uint32_t* row_count;  // unfiltereds in this partition
     uint64_t* row_base;   // start into the per-source-concatenated UnfCols
     uint64_t* key_addr;   // device address of the key bytes in input data
+    uint32_t* cpx_total;  // complex cells in this partition (arena sizing pass)
     StaticCols st;        // static row per partition (when the schema has one)
 };
 
@@ -176,6 +196,7 @@ struct OutStats {
     unsigned long long cells_hist[119];              // EstimatedHistogram(118)
     unsigned long long tomb_count;                   // appended ldt list counter
     unsigned long long error;                        // parse/feature error flag
+    unsigned long long cpx_alloc;                    // output cpx-cell arena bump
 };
 
 constexpr uint32_t LZ4_SLOT = 16480;  // > 4 + LZ4_compressBound(16384), 16B aligned

@@ -717,8 +717,21 @@ struct UnfColsBuf {
     DevBuf ck, rkind, flags, live_ts, live_ttl, live_let, rdel_mfda, rdel_ldt,
         start_mfda, start_ldt, cell_ts, cell_ldt, cell_ttl, val_addr, val_len, ck_addr, ck_len,
         cell_flags, ck_count;
+    DevBuf cpx_del_mfda, cpx_del_ldt, cpx_start, cpx_count;          // per row
+    DevBuf cx_ts, cx_ldt, cx_ttl, cx_flags, cx_pa, cx_pl, cx_va, cx_vl;  // cell arena
     UnfCols uc{};
-    void alloc(uint64_t n, uint32_t n_cols, uint32_t n_ck) {
+    // cell arena sized AFTER the counting pass (arena capacity = total
+    // complex cells); per-row arrays are allocated with the rows
+    void alloc_cpx_arena(uint64_t cells) {
+        if (!cells) cells = 1;
+        cx_ts.alloc(cells * 8); cx_ldt.alloc(cells * 4); cx_ttl.alloc(cells * 4);
+        cx_flags.alloc(cells); cx_pa.alloc(cells * 8); cx_pl.alloc(cells * 4);
+        cx_va.alloc(cells * 8); cx_vl.alloc(cells * 4);
+        uc.cpx = UnfCols::CpxCells{cx_ts.as<int64_t>(), cx_ldt.as<uint32_t>(), cx_ttl.as<int32_t>(),
+                                   cx_flags.as<uint8_t>(), cx_pa.as<uint64_t>(), cx_pl.as<uint32_t>(),
+                                   cx_va.as<uint64_t>(), cx_vl.as<uint32_t>()};
+    }
+    void alloc(uint64_t n, uint32_t n_cols, uint32_t n_ck, uint32_t n_cpx = 0) {
         if (!n) n = 1;
         uint64_t nc = n * n_cols;
         uint64_t nk = n * (n_ck ? n_ck : 1);
@@ -728,6 +741,10 @@ struct UnfColsBuf {
         cell_ts.alloc(nc * 8); cell_ldt.alloc(nc * 4); cell_ttl.alloc(nc * 4);
         val_addr.alloc(nc * 8); val_len.alloc(nc * 4); cell_flags.alloc(nc);
         ck_addr.alloc(nk * 8); ck_len.alloc(nk * 4); ck_count.alloc(n);
+        if (n_cpx) {
+            cpx_del_mfda.alloc(n * 8); cpx_del_ldt.alloc(n * 4);
+            cpx_start.alloc(n * 8); cpx_count.alloc(n * 4);
+        }
         uc = UnfCols{ck.as<uint64_t>(), rkind.as<uint8_t>(), flags.as<uint8_t>(),
                      live_ts.as<int64_t>(), live_ttl.as<int32_t>(), live_let.as<int64_t>(),
                      rdel_mfda.as<int64_t>(), rdel_ldt.as<uint32_t>(), start_mfda.as<int64_t>(),
@@ -735,6 +752,12 @@ struct UnfColsBuf {
                      cell_ttl.as<int32_t>(), val_addr.as<uint64_t>(), val_len.as<uint32_t>(),
                      cell_flags.as<uint8_t>(),
                      ck_addr.as<uint64_t>(), ck_len.as<uint32_t>(), ck_count.as<uint8_t>()};
+        if (n_cpx) {
+            uc.cpx_del_mfda = cpx_del_mfda.as<int64_t>();
+            uc.cpx_del_ldt = cpx_del_ldt.as<uint32_t>();
+            uc.cpx_start = cpx_start.as<uint64_t>();
+            uc.cpx_count = cpx_count.as<uint32_t>();
+        }
     }
 };
 
@@ -1171,7 +1194,8 @@ struct CompactSetup {
     std::vector<std::vector<uint64_t>> positions;   // n_parts+1 absolute offsets
     std::vector<std::vector<uint64_t>> entry_offs;  // Index.db entry byte offsets
     std::vector<size_t> comp_file_sz;
-    std::vector<int32_t> col_fixed_h;
+    std::vector<int32_t> col_fixed_h;    // SIMPLE regular columns only
+    uint32_t n_cpx = 0;                  // one complex (map<blob,blob>) column, last
     std::vector<int32_t> ck_widths;      // per clustering column
     std::vector<int32_t> static_fixed_h; // per static column
     // unsharded fast path: whole-file Data.db reads started during index
@@ -1412,6 +1436,7 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
                                      hipMemcpyHostToDevice, stream));
         sch.ck_w = d_ck_w.as<int32_t>();
         sch.n_cols = (uint32_t)col_fixed_h.size();
+        sch.n_cpx = su.n_cpx;
         DevBuf d_col_fixed;
         d_col_fixed.alloc(col_fixed_h.size() * 4);
         HIP_CHECK(hipMemcpyAsync(d_col_fixed.p, col_fixed_h.data(), col_fixed_h.size() * 4,
@@ -1454,13 +1479,31 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         }
         uint64_t total_in_rows = exscan_u64(pc.row_base, total_parts, stream);
         UnfColsBuf in_rows;
-        in_rows.alloc(total_in_rows, sch.n_cols, sch.n_ck);
+        in_rows.alloc(total_in_rows, sch.n_cols, sch.n_ck, sch.n_cpx);
+        DevBuf p_cpxtot, p_cpxbase;
+        uint64_t total_in_cpx = 0;
         {
             uint32_t blocks = (uint32_t)((total_parts + 255) / 256);
+            if (sch.n_cpx) {
+                // pass B1 (COUNT): full walk, per-partition complex-cell totals
+                p_cpxtot.alloc(total_parts * 4 + 8);
+                pc.cpx_total = p_cpxtot.as<uint32_t>();
+                hipLaunchKernelGGL(k_parse_rows, dim3(blocks), dim3(256), 0, stream,
+                                   d_srcs.as<SrcDesc2>(), (uint32_t)k, (uint32_t)total_parts, pc,
+                                   in_rows.uc, sch, d_error.as<unsigned long long>(),
+                                   d_rows_in.as<unsigned long long>(), nullptr);
+                HIP_CHECK(hipMemsetAsync(d_rows_in.p, 0, 8, stream));  // recounted by pass B2
+                p_cpxbase.alloc(total_parts * 8 + 8);
+                hipLaunchKernelGGL(k_widen_u32, dim3(blocks), dim3(256), 0, stream,
+                                   pc.cpx_total, p_cpxbase.as<uint64_t>(), total_parts);
+                total_in_cpx = exscan_u64(p_cpxbase.as<uint64_t>(), total_parts, stream);
+                in_rows.alloc_cpx_arena(total_in_cpx);
+            }
             hipLaunchKernelGGL(k_parse_rows, dim3(blocks), dim3(256), 0, stream,
                                d_srcs.as<SrcDesc2>(), (uint32_t)k, (uint32_t)total_parts, pc,
                                in_rows.uc, sch, d_error.as<unsigned long long>(),
-                               d_rows_in.as<unsigned long long>());
+                               d_rows_in.as<unsigned long long>(),
+                               sch.n_cpx ? p_cpxbase.as<uint64_t>() : nullptr);
         }
         HIP_CHECK(hipEventRecord(e3, stream));
         {
@@ -1490,6 +1533,8 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         // data; the rest are tombstone sources merged in a second pass.
         int kd = su.k_data > 0 ? su.k_data : k;
         bool gc_mode = kd < k;
+        if (gc_mode && su.n_cpx)
+            throw std::runtime_error("garbage-collect mode with complex columns unsupported");
         uint64_t data_parts = 0;
         for (int s = 0; s < kd; s++) data_parts += srcs[s].n_parts;
         std::vector<uint64_t> runs;
@@ -1546,7 +1591,8 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         OutPartsBuf opb;
         opb.alloc(n_groups, sch.n_static);
         UnfColsBuf out_rows;
-        out_rows.alloc(total_out_rows, sch.n_cols, sch.n_ck);
+        out_rows.alloc(total_out_rows, sch.n_cols, sch.n_ck, sch.n_cpx);
+        if (sch.n_cpx) out_rows.alloc_cpx_arena(total_in_cpx);
         DevBuf d_stats, d_tomb;
         d_stats.alloc(sizeof(OutStats));
         init_outstats(d_stats, stream);
@@ -1890,15 +1936,27 @@ static void compact_setup(const gpuc_job* job, CompactSetup& su, bool preread_fu
         size_t a = name.find('-'), b2 = name.find('-', a + 1);
         su.generations[s] = std::stoull(name.substr(a + 1, b2 - a - 1));
     }
-    for (auto& [nm, ct] : su.stats[0].regular_cols) {
-        (void)nm;
-        if (ct == "org.apache.cassandra.db.marshal.LongType") su.col_fixed_h.push_back(8);
-        else if (ct == "org.apache.cassandra.db.marshal.Int32Type") su.col_fixed_h.push_back(4);
-        else if (ct == "org.apache.cassandra.db.marshal.BytesType" ||
-                 ct == "org.apache.cassandra.db.marshal.UTF8Type" ||
-                 ct == "org.apache.cassandra.db.marshal.AsciiType")
-            su.col_fixed_h.push_back(-1);
-        else throw std::runtime_error("unsupported column type " + ct);
+    {
+        const auto& rcols = su.stats[0].regular_cols;
+        for (size_t ci = 0; ci < rcols.size(); ci++) {
+            const std::string& ct = rcols[ci].second;
+            if (ct == "org.apache.cassandra.db.marshal.LongType") su.col_fixed_h.push_back(8);
+            else if (ct == "org.apache.cassandra.db.marshal.Int32Type") su.col_fixed_h.push_back(4);
+            else if (ct == "org.apache.cassandra.db.marshal.BytesType" ||
+                     ct == "org.apache.cassandra.db.marshal.UTF8Type" ||
+                     ct == "org.apache.cassandra.db.marshal.AsciiType")
+                su.col_fixed_h.push_back(-1);
+            else if (ct == "org.apache.cassandra.db.marshal.MapType(org.apache.cassandra.db.marshal.BytesType,org.apache.cassandra.db.marshal.BytesType)") {
+                // one complex column, and it must be the LAST regular column
+                // (engine layout constraint; the generator names it 'zm' so
+                // name order puts it last)
+                if (su.n_cpx || ci + 1 != rcols.size())
+                    throw std::runtime_error("at most one complex column, as the last regular column");
+                su.n_cpx = 1;
+            } else {
+                throw std::runtime_error("unsupported column type " + ct);
+            }
+        }
     }
     for (auto& t : su.stats[0].clustering_types) su.ck_widths.push_back(ck_type_width(t));
     for (auto& [nm2, ct2] : su.stats[0].static_cols) {
@@ -2156,11 +2214,12 @@ extern "C" int gpuc_verify(const char* input_base, int32_t device, char* error,
             (void)nm;
             if (ct == "org.apache.cassandra.db.marshal.LongType") cfh.push_back(8);
             else if (ct == "org.apache.cassandra.db.marshal.Int32Type") cfh.push_back(4);
+            else if (ct == "org.apache.cassandra.db.marshal.MapType(org.apache.cassandra.db.marshal.BytesType,org.apache.cassandra.db.marshal.BytesType)") sch.n_cpx = 1;  // last regular column
             else cfh.push_back(-1);
         }
-        if (cfh.empty()) cfh.push_back(-1);
-        sch.n_cols = (uint32_t)st.regular_cols.size();
-        if (!sch.n_cols) sch.n_cols = 1;
+        if (cfh.empty() && !sch.n_cpx) cfh.push_back(-1);
+        sch.n_cols = (uint32_t)cfh.size();
+        if (!sch.n_cols && !sch.n_cpx) sch.n_cols = 1;
         DevBuf d_cf;
         d_cf.alloc(cfh.size() * 4);
         HIP_CHECK(hipMemcpyAsync(d_cf.p, cfh.data(), cfh.size() * 4, hipMemcpyHostToDevice, stream));
@@ -2337,7 +2396,11 @@ extern "C" int gpuc_scrub(const char* input_base, const char* output_base, int32
         // schema (same resolution as compaction)
         SchemaParams sch{};
         std::vector<int32_t> cfh, vckw, vsf;
-        for (auto& [nm, ct] : st.regular_cols) { (void)nm; cfh.push_back(ck_type_width(ct)); }
+        for (auto& [nm, ct] : st.regular_cols) {
+            (void)nm;
+            if (ct == "org.apache.cassandra.db.marshal.MapType(org.apache.cassandra.db.marshal.BytesType,org.apache.cassandra.db.marshal.BytesType)") sch.n_cpx = 1;  // last regular column
+            else cfh.push_back(ck_type_width(ct));
+        }
         for (auto& ct : st.clustering_types) vckw.push_back(ck_type_width(ct));
         for (auto& [nm, ct] : st.static_cols) { (void)nm; vsf.push_back(ck_type_width(ct)); }
         sch.n_cols = (uint32_t)cfh.size();
@@ -2398,7 +2461,7 @@ extern "C" int gpuc_scrub(const char* input_base, const char* output_base, int32
         }
         uint64_t total_rows = n_kept ? exscan_u64(pc.row_base, n_kept, stream) : 0;
         UnfColsBuf in_rows;
-        in_rows.alloc(total_rows, sch.n_cols, sch.n_ck);
+        in_rows.alloc(total_rows, sch.n_cols, sch.n_ck, sch.n_cpx);
         if (n_kept)
             hipLaunchKernelGGL(k_parse_rows, dim3(blocks), dim3(256), 0, stream,
                                d_src.as<SrcDesc2>(), 1u, (uint32_t)n_kept, pc, in_rows.uc, sch,
@@ -2649,6 +2712,8 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
             if (gp.ck_cols == 2 && gp.ck_text) throw std::runtime_error("ck_cols=2 with ck_text unsupported");
             gp.col_missing_pct = spec->col_missing_pct;
             gp.ttl_pct = spec->ttl_pct;
+            gp.complex_pct = spec->complex_pct;
+            gp.complex_del_pct = spec->complex_del_pct;
             if (gp.n_value_cols > 63) throw std::runtime_error("n_value_cols must be 1..63");
             if (gp.ck_text && (uint64_t)gp.clustering_rows * 16 >= 100000000ull)
                 throw std::runtime_error("ck_text needs clustering_rows*16 < 1e8 (8-digit order)");
@@ -2667,13 +2732,21 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
             OutPartsBuf opb;
             opb.alloc(R, gp.static_pct ? 1 : 0);
             d_prows.alloc(R * 8);
+            DevBuf d_cpxcnt, d_cpxbytes;
+            if (gp.complex_pct) d_cpxcnt.alloc(R * 8 + 8);
             hipLaunchKernelGGL(k_gen_count, dim3(blocks), dim3(256), 0, stream, gp, d_sorted,
                                d_ids.as<uint64_t>(), R, opb.op, d_prows.as<uint64_t>(),
-                               d_keys.as<uint8_t>());
+                               d_keys.as<uint8_t>(),
+                               gp.complex_pct ? d_cpxcnt.as<uint64_t>() : nullptr);
+            uint64_t total_cpx = gp.complex_pct ? exscan_u64(d_cpxcnt.as<uint64_t>(), R, stream) : 0;
             uint64_t total_rows = exscan_u64(d_prows.as<uint64_t>(), R, stream);
             UnfColsBuf rows;
             uint32_t gen_nck = gp.clustering_rows ? (gp.ck_cols ? gp.ck_cols : 1) : 0;
-            rows.alloc(total_rows, gp.n_value_cols, gen_nck);
+            rows.alloc(total_rows, gp.n_value_cols, gen_nck, gp.complex_pct ? 1 : 0);
+            if (gp.complex_pct) {
+                rows.alloc_cpx_arena(total_cpx);
+                d_cpxbytes.alloc(total_cpx * 12 + 16);  // 4B path + 8B value per cell
+            }
             d_vals.alloc(total_rows * (uint64_t)gp.n_value_cols * spec->value_len);
             DevBuf d_ckarena, d_svals;
             if (gp.ck_text) d_ckarena.alloc(total_rows * 16 + 16);
@@ -2681,7 +2754,9 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
             hipLaunchKernelGGL(k_gen_fill2, dim3(blocks), dim3(256), 0, stream, gp, d_sorted,
                                d_ids.as<uint64_t>(), R, opb.op, rows.uc, d_prows.as<uint64_t>(),
                                d_vals.as<uint8_t>(), gp.ck_text ? d_ckarena.as<uint8_t>() : nullptr,
-                               gp.static_pct ? d_svals.as<uint8_t>() : nullptr);
+                               gp.static_pct ? d_svals.as<uint8_t>() : nullptr,
+                               gp.complex_pct ? d_cpxbytes.as<uint8_t>() : nullptr,
+                               gp.complex_pct ? d_cpxcnt.as<uint64_t>() : nullptr);
             hipLaunchKernelGGL(k_gen_values2, dim3(blocks), dim3(256), 0, stream, gp, d_sorted,
                                d_ids.as<uint64_t>(), opb.op, rows.uc, R, d_vals.as<uint8_t>(),
                                d_prows.as<uint64_t>());
@@ -2719,6 +2794,7 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
             HIP_CHECK(hipMemcpyAsync(d_sfx.p, sfx_h.data(), 4, hipMemcpyHostToDevice, stream));
             sp.sch.static_fixed = d_sfx.as<int32_t>();
             sp.sch.n_cols = gp.n_value_cols;
+            sp.sch.n_cpx = gp.complex_pct ? 1 : 0;
             std::vector<int32_t> gcf(gp.n_value_cols, -1);  // val blobs
             DevBuf d_gcf;
             d_gcf.alloc(gcf.size() * 4);
@@ -2740,6 +2816,8 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
                                     "org.apache.cassandra.db.marshal.BytesType"});
                 }
             }
+            if (gp.complex_pct)
+                cols.push_back({bytes{'z', 'm'}, "org.apache.cassandra.db.marshal.MapType(org.apache.cassandra.db.marshal.BytesType,org.apache.cassandra.db.marshal.BytesType)"});
             std::vector<std::pair<bytes, std::string>> scols;
             if (gp.static_pct)
                 scols.push_back({bytes{'s', '0'}, "org.apache.cassandra.db.marshal.BytesType"});

@@ -28,8 +28,10 @@ __device__ inline void tomb_push(OutStats* st, uint32_t* ldts, uint32_t cap, uin
 struct SchemaParams {
     uint32_t n_ck;           // clustering columns (0..32)
     const int32_t* ck_w;     // per clustering column: 4/8 fixed or -1 variable
-    uint32_t n_cols;         // regular columns (1..63; header superset order)
-    const int32_t* col_fixed;  // per column: -1 variable else fixed width
+    uint32_t n_cols;         // SIMPLE regular columns (1..63; header order)
+    const int32_t* col_fixed;  // per simple column: -1 variable else fixed width
+    uint32_t n_cpx;          // 0 or 1: one complex column AFTER the simple ones
+                             // (subset bitmap covers n_cols + n_cpx bits)
     uint32_t n_static;       // static columns (0 = schema has no statics)
     const int32_t* static_fixed;
     uint32_t column_index_size;  // promoted-index granularity (64 KiB default)
@@ -164,7 +166,7 @@ __global__ void k_parse_count(const SrcDesc2* srcs, uint32_t n_srcs, uint32_t to
     while (true) {
         uint8_t flags = base[pos++];
         if (flags & 0x01) break;  // END_OF_PARTITION
-        if (flags & 0xC0) { atomicExch(error, 12ull); return; }  // complex/extension
+        if ((flags & 0x80) || ((flags & 0x40) && !sp.n_cpx)) { atomicExch(error, 12ull); return; }  // extension / unexpected complex
         if (flags & 0x02) {
             // marker: kind, u16 size, values
             uint8_t kind = base[pos++];
@@ -197,9 +199,13 @@ __global__ void k_parse_count(const SrcDesc2* srcs, uint32_t n_srcs, uint32_t to
 // ---------------------------------------------------------------------------
 // parse pass B: full decode into the UnfCols arena (row_base prescanned)
 // ---------------------------------------------------------------------------
+// cpx_base: per-partition bases into the complex-cell arena (exscan of
+// pc.cpx_total). nullptr = COUNT mode: walk everything, fill all non-arena
+// outputs, and write pc.cpx_total[gi] for the sizing exscan.
 __global__ void k_parse_rows(const SrcDesc2* srcs, uint32_t n_srcs, uint32_t total,
                              ParsedCols pc, UnfCols rc, SchemaParams sp,
-                             unsigned long long* error, unsigned long long* rows_in) {
+                             unsigned long long* error, unsigned long long* rows_in,
+                             const uint64_t* cpx_base = nullptr) {
     uint32_t gi = blockIdx.x * blockDim.x + threadIdx.x;
     if (gi >= total) return;
     uint32_t s = 0;
@@ -276,6 +282,8 @@ __global__ void k_parse_rows(const SrcDesc2* srcs, uint32_t n_srcs, uint32_t tot
 
     uint64_t out = pc.row_base[gi];
     uint32_t emitted = 0;
+    uint64_t cpx_cur = (sp.n_cpx && cpx_base) ? cpx_base[gi] : 0;
+    uint32_t cpx_seen = 0;
     while (true) {
         uint8_t flags = base[pos++];
         if (flags & 0x01) break;
@@ -323,6 +331,12 @@ __global__ void k_parse_rows(const SrcDesc2* srcs, uint32_t n_srcs, uint32_t tot
             rc.live_ttl[o] = 0;
             rc.live_let[o] = NO_DELETION_TIME;
             for (uint32_t c = 0; c < sp.n_cols; c++) rc.cell_flags[o * sp.n_cols + c] = 0;
+            if (sp.n_cpx) {
+                rc.cpx_del_mfda[o] = INT64_MIN;
+                rc.cpx_del_ldt[o] = LDT_NONE_U32;
+                rc.cpx_start[o] = cpx_cur;
+                rc.cpx_count[o] = 0;
+            }
         } else {
             // ---- row ----
             if (sp.n_ck) uvint_get(base, &pos);  // 32-batch header
@@ -391,6 +405,56 @@ __global__ void k_parse_rows(const SrcDesc2* srcs, uint32_t n_srcs, uint32_t tot
                 rc.val_addr[oc] = vaddr;
                 rc.val_len[oc] = vlen;
             }
+            // ---- complex column (bit sp.n_cols of the subset bitmap) ----
+            if (sp.n_cpx) {
+                rc.cpx_del_mfda[o] = INT64_MIN;
+                rc.cpx_del_ldt[o] = LDT_NONE_U32;
+                rc.cpx_count[o] = 0;
+                rc.cpx_start[o] = cpx_cur;
+                if (!(missing & (1ULL << sp.n_cols))) {
+                    pf |= PF_HAS_CPX;
+                    if (flags & 0x40) {  // HAS_COMPLEX_DELETION: per-column deletion
+                        rc.cpx_del_mfda[o] = (int64_t)uvint_get(base, &pos) + sd.min_ts;
+                        rc.cpx_del_ldt[o] = ldt_u32((int64_t)(int32_t)(uint32_t)uvint_get(base, &pos) + sd.min_ldt);
+                    }
+                    uint32_t ncc = (uint32_t)uvint_get(base, &pos);
+                    rc.cpx_count[o] = ncc;
+                    for (uint32_t e = 0; e < ncc; e++) {
+                        uint8_t cf = base[pos++];
+                        int64_t cts = (cf & 8) ? lts : (int64_t)uvint_get(base, &pos) + sd.min_ts;
+                        bool dead = cf & 1, exp = cf & 2;
+                        int64_t ldtl;
+                        if (cf & 16) ldtl = llet;
+                        else if (dead || exp) ldtl = (int64_t)(int32_t)(uint32_t)uvint_get(base, &pos) + sd.min_ldt;
+                        else ldtl = NO_DELETION_TIME;
+                        int32_t cttl = (cf & 16) ? lttl : (exp ? (int32_t)(uint32_t)uvint_get(base, &pos) + sd.min_ttl : 0);
+                        uint32_t plen = (uint32_t)uvint_get(base, &pos);  // CellPath vint+bytes
+                        uint64_t paddr = (uint64_t)(base + pos);
+                        pos += plen;
+                        uint64_t vaddr = 0;
+                        uint32_t vlen = 0;
+                        if (!(cf & 4)) {  // value: BytesType (map value), vint length
+                            vlen = (uint32_t)uvint_get(base, &pos);
+                            vaddr = (uint64_t)(base + pos);
+                            pos += vlen;
+                        }
+                        if (cpx_base) {
+                            uint64_t xe = cpx_cur + e;
+                            rc.cpx.ts[xe] = cts;
+                            rc.cpx.ldt[xe] = ldt_u32(ldtl);
+                            rc.cpx.ttl[xe] = cttl;
+                            rc.cpx.flags[xe] = CELLF_PRESENT | ((cf & 4) ? 0 : CELLF_HAS_VALUE) |
+                                               (exp ? CELLF_EXPIRING : 0);
+                            rc.cpx.path_addr[xe] = paddr;
+                            rc.cpx.path_len[xe] = plen;
+                            rc.cpx.val_addr[xe] = vaddr;
+                            rc.cpx.val_len[xe] = vlen;
+                        }
+                    }
+                    cpx_cur += ncc;
+                    cpx_seen += ncc;
+                }
+            }
             rc.rkind[o] = BK_CLUSTERING;
             rc.ck_count[o] = (uint8_t)sp.n_ck;
             rc.flags[o] = pf;
@@ -405,6 +469,7 @@ __global__ void k_parse_rows(const SrcDesc2* srcs, uint32_t n_srcs, uint32_t tot
         }
         emitted++;
     }
+    if (sp.n_cpx && pc.cpx_total) pc.cpx_total[gi] = cpx_seen;
     (void)error;
 }
 
@@ -824,6 +889,11 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
             const uint32_t NC = sp.n_cols;
             uint64_t oslot = obase + ocount;  // cells written here pre-emit
             bool any_cell = false;
+            // merged complex column (ColumnDataReducer complex branch)
+            int64_t xdm = INT64_MIN;
+            uint32_t xdl = LDT_NONE_U32;
+            uint64_t xstart = 0;
+            uint32_t xcnt = 0;
             if (k == 1 || (nmem == 1 && active_live)) {
                 uint64_t o = mb[lastm] + mpos[lastm];
                 of = in.flags[o];
@@ -840,6 +910,24 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
                     out.cell_ttl[ocx] = in.cell_ttl[ic];
                     out.val_addr[ocx] = in.val_addr[ic];
                     out.val_len[ocx] = in.val_len[ic];
+                }
+                if (sp.n_cpx && (of & PF_HAS_CPX)) {
+                    xdm = in.cpx_del_mfda[o];
+                    xdl = in.cpx_del_ldt[o];
+                    uint32_t nc2 = in.cpx_count[o];
+                    xstart = atomicAdd(&st->cpx_alloc, (unsigned long long)nc2);
+                    uint64_t si = in.cpx_start[o];
+                    for (uint32_t e2 = 0; e2 < nc2; e2++) {
+                        out.cpx.ts[xstart + e2] = in.cpx.ts[si + e2];
+                        out.cpx.ldt[xstart + e2] = in.cpx.ldt[si + e2];
+                        out.cpx.ttl[xstart + e2] = in.cpx.ttl[si + e2];
+                        out.cpx.flags[xstart + e2] = in.cpx.flags[si + e2];
+                        out.cpx.path_addr[xstart + e2] = in.cpx.path_addr[si + e2];
+                        out.cpx.path_len[xstart + e2] = in.cpx.path_len[si + e2];
+                        out.cpx.val_addr[xstart + e2] = in.cpx.val_addr[si + e2];
+                        out.cpx.val_len[xstart + e2] = in.cpx.val_len[si + e2];
+                    }
+                    xcnt = nc2;
                 }
             } else {
                 bool has_live = false;
@@ -932,6 +1020,121 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
                         out.cell_flags[ocx] = 0;
                     }
                 }
+                if (sp.n_cpx) {
+                    // complexDeletion = max over versions; supersedes(active)
+                    // decides both the cell filter and whether it is kept
+                    // (Row.java:857-874)
+                    bool anyv = false;
+                    for (uint32_t m = 0; m < k; m++) {
+                        if (!(members & (1ULL << m))) continue;
+                        uint64_t o = mb[m] + mpos[m];
+                        if (!(in.flags[o] & PF_HAS_CPX)) continue;
+                        anyv = true;
+                        if (dt_sup(in.cpx_del_mfda[o], in.cpx_del_ldt[o], xdm, xdl)) {
+                            xdm = in.cpx_del_mfda[o];
+                            xdl = in.cpx_del_ldt[o];
+                        }
+                    }
+                    if (anyv) {
+                        int64_t cam = am2;
+                        uint32_t cal = al2;
+                        bool keep_del = dt_sup(xdm, xdl, am2, al2);
+                        if (keep_del) { cam = xdm; cal = xdl; }
+                        else { xdm = INT64_MIN; xdl = LDT_NONE_U32; }
+                        uint32_t ppos[MA], pcnt[MA];
+                        uint64_t pbase[MA];
+                        uint32_t nv2 = 0, totc = 0;
+                        for (uint32_t m = 0; m < k; m++) {
+                            if (!(members & (1ULL << m))) continue;
+                            uint64_t o = mb[m] + mpos[m];
+                            if (!(in.flags[o] & PF_HAS_CPX)) continue;
+                            pbase[nv2] = in.cpx_start[o];
+                            pcnt[nv2] = in.cpx_count[o];
+                            ppos[nv2] = 0;
+                            totc += in.cpx_count[o];
+                            nv2++;
+                        }
+                        xstart = atomicAdd(&st->cpx_alloc, (unsigned long long)totc);
+                        auto cmp_path = [&](uint64_t a, uint64_t b) -> int {
+                            return cmp_values(in.cpx.path_addr[a], in.cpx.path_len[a],
+                                              in.cpx.path_addr[b], in.cpx.path_len[b]);
+                        };
+                        while (true) {
+                            int fi = -1;
+                            for (uint32_t m = 0; m < nv2; m++) {
+                                if (ppos[m] >= pcnt[m]) continue;
+                                if (fi < 0 || cmp_path(pbase[m] + ppos[m], pbase[fi] + ppos[fi]) < 0)
+                                    fi = (int)m;
+                            }
+                            if (fi < 0) break;
+                            uint64_t fxe = pbase[fi] + ppos[fi];
+                            // CellReducer over same-path cells in version order
+                            bool have = false;
+                            int64_t cts = NO_TIMESTAMP;
+                            uint32_t cldt = LDT_NONE_U32;
+                            int32_t cttl = 0;
+                            uint64_t va = 0, pa = 0;
+                            uint32_t vl = 0, pl = 0;
+                            bool cell_val = false, cell_exp = false;
+                            for (uint32_t m = 0; m < nv2; m++) {
+                                if (ppos[m] >= pcnt[m]) continue;
+                                uint64_t xe = pbase[m] + ppos[m];
+                                if (cmp_path(xe, fxe) != 0) continue;
+                                ppos[m]++;
+                                int64_t ts2 = in.cpx.ts[xe];
+                                if (cam != INT64_MIN && ts2 <= cam) continue;  // activeDeletion.deletes
+                                uint8_t f2 = in.cpx.flags[xe];
+                                if (!have) {
+                                    have = true;
+                                    cts = ts2; cldt = in.cpx.ldt[xe]; cttl = in.cpx.ttl[xe];
+                                    va = in.cpx.val_addr[xe]; vl = in.cpx.val_len[xe];
+                                    pa = in.cpx.path_addr[xe]; pl = in.cpx.path_len[xe];
+                                    cell_val = f2 & CELLF_HAS_VALUE;
+                                    cell_exp = f2 & CELLF_EXPIRING;
+                                    continue;
+                                }
+                                bool take_right = false;
+                                uint32_t rl = in.cpx.ldt[xe];
+                                bool l_dt = cldt != LDT_NONE_U32, r_dt = rl != LDT_NONE_U32;
+                                if (cts != ts2) take_right = ts2 > cts;
+                                else if (l_dt || r_dt) {
+                                    if (l_dt != r_dt) take_right = r_dt;
+                                    else {
+                                        bool l_tomb = !cell_exp, r_tomb = !(f2 & CELLF_EXPIRING);
+                                        if (l_tomb != r_tomb) take_right = r_tomb;
+                                        else if (cldt != rl) take_right = ldt_long(rl) > ldt_long(cldt);
+                                        else take_right = cmp_values(va, vl, in.cpx.val_addr[xe], in.cpx.val_len[xe]) < 0;
+                                    }
+                                } else {
+                                    take_right = cmp_values(va, vl, in.cpx.val_addr[xe], in.cpx.val_len[xe]) < 0;
+                                }
+                                if (take_right) {
+                                    cts = ts2; cldt = rl; cttl = in.cpx.ttl[xe];
+                                    va = in.cpx.val_addr[xe]; vl = in.cpx.val_len[xe];
+                                    cell_val = f2 & CELLF_HAS_VALUE;
+                                    cell_exp = f2 & CELLF_EXPIRING;
+                                }
+                            }
+                            if (have) {
+                                uint64_t xe2 = xstart + xcnt++;
+                                out.cpx.ts[xe2] = cts;
+                                out.cpx.ldt[xe2] = cldt;
+                                out.cpx.ttl[xe2] = cttl;
+                                out.cpx.flags[xe2] = CELLF_PRESENT | (cell_val ? CELLF_HAS_VALUE : 0) |
+                                                     (cell_exp ? CELLF_EXPIRING : 0);
+                                out.cpx.path_addr[xe2] = pa;
+                                out.cpx.path_len[xe2] = pl;
+                                out.cpx.val_addr[xe2] = va;
+                                out.cpx.val_len[xe2] = vl;
+                            }
+                        }
+                        // Builder.build: live deletion + no cells -> null column
+                        if (xcnt || xdm != INT64_MIN || xdl != LDT_NONE_U32) {
+                            of |= PF_HAS_CPX;
+                            any_cell = true;
+                        }
+                    }
+                }
                 if ((of & (PF_LIVE_TS | PF_ROW_DEL)) || any_cell) of |= PF_HAS_ROW;
                 else of = 0;
             }
@@ -964,12 +1167,59 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
                     }
                     any_cell = true;
                 }
+                if (sp.n_cpx && (of & PF_HAS_CPX)) {
+                    // ComplexColumnData.purge (ComplexColumnData.java:212-216)
+                    if (xdm != INT64_MIN && should_purge2(pp, token, xdm, ldt_long(xdl))) {
+                        xdm = INT64_MIN;
+                        xdl = LDT_NONE_U32;
+                    }
+                    uint32_t w = 0;
+                    for (uint32_t e2 = 0; e2 < xcnt; e2++) {
+                        uint64_t xe = xstart + e2;
+                        int64_t cts = out.cpx.ts[xe];
+                        uint32_t cldt = out.cpx.ldt[xe];
+                        int32_t cttl = out.cpx.ttl[xe];
+                        uint8_t cfl = out.cpx.flags[xe];
+                        uint32_t vl = out.cpx.val_len[xe];
+                        bool live_cell = cldt == LDT_NONE_U32 || (cttl != 0 && pp.now_sec < ldt_long(cldt));
+                        if (!live_cell) {
+                            if (should_purge2(pp, token, cts, ldt_long(cldt))) continue;
+                            if (cttl != 0) {
+                                int64_t nldt = ldt_long(cldt) - cttl;
+                                if (should_purge2(pp, token, cts, nldt)) continue;
+                                cldt = ldt_u32(nldt);
+                                cttl = 0;
+                                cfl = CELLF_PRESENT;  // expired -> tombstone, value dropped
+                                vl = 0;
+                            }
+                        }
+                        uint64_t xo = xstart + w++;
+                        out.cpx.ts[xo] = cts;
+                        out.cpx.ldt[xo] = cldt;
+                        out.cpx.ttl[xo] = cttl;
+                        out.cpx.flags[xo] = cfl;
+                        out.cpx.path_addr[xo] = out.cpx.path_addr[xe];
+                        out.cpx.path_len[xo] = out.cpx.path_len[xe];
+                        out.cpx.val_addr[xo] = out.cpx.val_addr[xe];
+                        out.cpx.val_len[xo] = vl;
+                    }
+                    xcnt = w;
+                    if (xcnt == 0 && xdm == INT64_MIN && xdl == LDT_NONE_U32) of &= ~PF_HAS_CPX;
+                    else any_cell = true;
+                }
                 if (!(of & (PF_LIVE_TS | PF_ROW_DEL)) && !any_cell) of = 0;
                 else if (pp.enforce_strict_liveness && !(of & PF_LIVE_TS) && !(of & PF_ROW_DEL)) of = 0;
             }
-            if (of & PF_HAS_ROW)
+            if (of & PF_HAS_ROW) {
+                if (sp.n_cpx) {
+                    out.cpx_del_mfda[oslot] = xdm;
+                    out.cpx_del_ldt[oslot] = xdl;
+                    out.cpx_start[oslot] = xstart;
+                    out.cpx_count[oslot] = (of & PF_HAS_CPX) ? xcnt : 0;
+                }
                 emit(BK_CLUSTERING, fck, of, lts, lttl, llet, rdm, rdl, INT64_MIN, LDT_NONE_U32,
                      true);
+            }
         } else if (k == 1) {
             // single-version partition: UnfilteredRowIterators.merge of one
             // iterator returns it unchanged — markers pass through as-is and
@@ -1115,6 +1365,38 @@ __device__ inline uint32_t cell_body_size(const COLS& u, uint64_t oc, const SerP
     return body;
 }
 
+// complex-cell body size (Cell.Serializer + CellPath vint+bytes; value is
+// the map VALUE type == BytesType, so always vint-length)
+__device__ inline uint32_t cpx_cell_body_size(const UnfCols& u, uint64_t xe, const SerParams2& sp,
+                                              bool live, bool exp_live, uint64_t o,
+                                              uint8_t* out_cf) {
+    uint8_t f = u.cpx.flags[xe];
+    int64_t cts = u.cpx.ts[xe];
+    uint32_t cldt = u.cpx.ldt[xe];
+    int32_t cttl = u.cpx.ttl[xe];
+    uint32_t vlen = u.cpx.val_len[xe];
+    bool has_value = vlen > 0 && (f & CELLF_HAS_VALUE);
+    bool deleted = cldt != LDT_NONE_U32 && cttl == NO_TTL;
+    bool expiring = cttl != NO_TTL;
+    bool use_row_ts = live && cts == u.live_ts[o];
+    bool use_row_ttl = expiring && exp_live && cttl == u.live_ttl[o] && ldt_long(cldt) == u.live_let[o];
+    uint8_t cflags = 0;
+    if (!has_value) cflags |= 4;
+    if (deleted) cflags |= 1;
+    else if (expiring) cflags |= 2;
+    if (use_row_ts) cflags |= 8;
+    if (use_row_ttl) cflags |= 16;
+    uint32_t body = 1;
+    if (!use_row_ts) body += uvint_size((uint64_t)(cts - sp.hs.min_ts));
+    if ((deleted || expiring) && !use_row_ttl) body += uvint_size(sext32(ldt_long(cldt) - sp.hs.min_ldt));
+    if (expiring && !use_row_ttl) body += uvint_size(sext32(cttl - sp.hs.min_ttl));
+    uint32_t plen = u.cpx.path_len[xe];
+    body += uvint_size(plen) + plen;
+    if (has_value) body += uvint_size(vlen) + vlen;
+    if (out_cf) *out_cf = cflags;
+    return body;
+}
+
 // row/marker body size + flag bytes (mirrors the oracle serializers)
 __device__ inline uint32_t unf_body_size(const UnfCols& u, uint64_t o, const SerParams2& sp,
                                          uint8_t* out_flags, uint8_t* out_cflags) {
@@ -1145,18 +1427,34 @@ __device__ inline uint32_t unf_body_size(const UnfCols& u, uint64_t o, const Ser
         body += uvint_size((uint64_t)(u.rdel_mfda[o] - sp.hs.min_ts));
         body += uvint_size(sext32(ldt_long(u.rdel_ldt[o]) - sp.hs.min_ldt));
     }
-    // columns subset + per-cell sizes (cell_flags_byte shared with EMIT)
+    // columns subset + per-cell sizes (cell_flags_byte shared with EMIT);
+    // the superset covers simple columns + the complex column (bit n_cols)
     uint64_t present_mask = 0;
     uint32_t present = 0;
+    const uint32_t NSUP = sp.sch.n_cols + sp.sch.n_cpx;
+    bool has_cpx = sp.sch.n_cpx && (f & PF_HAS_CPX);
     for (uint32_t c = 0; c < sp.sch.n_cols; c++)
         if (u.cell_flags[o * sp.sch.n_cols + c] & CELLF_PRESENT) { present_mask |= 1ULL << c; present++; }
-    if (present == sp.sch.n_cols) rflags |= 0x20;
-    else body += uvint_size(((1ULL << sp.sch.n_cols) - 1) & ~present_mask);
+    if (has_cpx) { present_mask |= 1ULL << sp.sch.n_cols; present++; }
+    if (present == NSUP) rflags |= 0x20;
+    else body += uvint_size(((1ULL << NSUP) - 1) & ~present_mask);
     for (uint32_t c = 0; c < sp.sch.n_cols; c++) {
         uint64_t oc = o * sp.sch.n_cols + c;
         if (!(u.cell_flags[oc] & CELLF_PRESENT)) continue;
         uint8_t cflags;
         body += cell_body_size(u, oc, sp, sp.sch.col_fixed[c], live, exp_live, o, &cflags);
+    }
+    if (has_cpx) {
+        bool del_live = u.cpx_del_mfda[o] == INT64_MIN && u.cpx_del_ldt[o] == LDT_NONE_U32;
+        if (!del_live) rflags |= 0x40;  // HAS_COMPLEX_DELETION
+        if (rflags & 0x40)
+            body += uvint_size((uint64_t)(u.cpx_del_mfda[o] - sp.hs.min_ts)) +
+                    uvint_size(sext32(ldt_long(u.cpx_del_ldt[o]) - sp.hs.min_ldt));
+        uint32_t nc2 = u.cpx_count[o];
+        body += uvint_size(nc2);
+        uint64_t si = u.cpx_start[o];
+        for (uint32_t e2 = 0; e2 < nc2; e2++)
+            body += cpx_cell_body_size(u, si + e2, sp, live, exp_live, o, nullptr);
     }
     if (out_flags) *out_flags = rflags;
     if (out_cflags) *out_cflags = 0;
@@ -1493,9 +1791,11 @@ __device__ uint64_t part_walk(const OutParts& op, const UnfCols& out, uint64_t g
             }
             if (!(rflags & 0x20)) {
                 uint64_t present_mask = 0;
+                const uint32_t NSUP = sp.sch.n_cols + sp.sch.n_cpx;
                 for (uint32_t c = 0; c < sp.sch.n_cols; c++)
                     if (out.cell_flags[o * sp.sch.n_cols + c] & CELLF_PRESENT) present_mask |= 1ULL << c;
-                emit_uv(((1ULL << sp.sch.n_cols) - 1) & ~present_mask);
+                if (sp.sch.n_cpx && (out.flags[o] & PF_HAS_CPX)) present_mask |= 1ULL << sp.sch.n_cols;
+                emit_uv(((1ULL << NSUP) - 1) & ~present_mask);
             }
             bool live = rflags & 0x04, exp_live = rflags & 0x08;
             for (uint32_t c = 0; c < sp.sch.n_cols; c++) {
@@ -1516,6 +1816,41 @@ __device__ uint64_t part_walk(const OutParts& op, const UnfCols& out, uint64_t g
                         wave_copy(out_data + data_off + pos,
                                   (const uint8_t*)out.val_addr[oc], (int)vlen, lane);
                     pos += vlen;
+                }
+            }
+            // ---- complex column (UnfilteredSerializer.writeComplexColumn) ----
+            if (sp.sch.n_cpx && (out.flags[o] & PF_HAS_CPX)) {
+                if (rflags & 0x40) {
+                    emit_uv((uint64_t)(out.cpx_del_mfda[o] - sp.hs.min_ts));
+                    emit_uv(sext32(ldt_long(out.cpx_del_ldt[o]) - sp.hs.min_ldt));
+                }
+                uint32_t nc2 = out.cpx_count[o];
+                emit_uv(nc2);
+                uint64_t si = out.cpx_start[o];
+                for (uint32_t e2 = 0; e2 < nc2; e2++) {
+                    uint64_t xe = si + e2;
+                    uint8_t cfb;
+                    cpx_cell_body_size(out, xe, sp, live, exp_live, o, &cfb);
+                    emit8(cfb);
+                    if (!(cfb & 8)) emit_uv((uint64_t)(out.cpx.ts[xe] - sp.hs.min_ts));
+                    bool deleted = cfb & 1, expiring = cfb & 2;
+                    if ((deleted || expiring) && !(cfb & 16))
+                        emit_uv(sext32(ldt_long(out.cpx.ldt[xe]) - sp.hs.min_ldt));
+                    if (expiring && !(cfb & 16)) emit_uv(sext32(out.cpx.ttl[xe] - sp.hs.min_ttl));
+                    uint32_t plen = out.cpx.path_len[xe];
+                    emit_uv(plen);
+                    if (EMIT)
+                        wave_copy(out_data + data_off + pos,
+                                  (const uint8_t*)out.cpx.path_addr[xe], (int)plen, lane);
+                    pos += plen;
+                    if (!(cfb & 4)) {
+                        uint32_t vlen = out.cpx.val_len[xe];
+                        emit_uv(vlen);
+                        if (EMIT)
+                            wave_copy(out_data + data_off + pos,
+                                      (const uint8_t*)out.cpx.val_addr[xe], (int)vlen, lane);
+                        pos += vlen;
+                    }
                 }
             }
         }
@@ -1619,9 +1954,12 @@ __global__ void k_sizes_rows(OutParts op, UnfCols out, uint64_t n, SerParams2 sp
             while (lo < hi) { int mid = (lo + hi) >> 1; if ((uint64_t)ps_hist_off[mid] < s) lo = mid + 1; else hi = mid; }
             atomicAdd(&sh_ps[lo], 1u);
             uint64_t cells = 0;
-            for (uint32_t j = 0; j < op.row_count[g]; j++)
+            for (uint32_t j = 0; j < op.row_count[g]; j++) {
+                uint64_t ro = op.row_base[g] + j;
                 for (uint32_t c = 0; c < sp.sch.n_cols; c++)
-                    if (out.cell_flags[(op.row_base[g] + j) * sp.sch.n_cols + c] & CELLF_PRESENT) cells++;
+                    if (out.cell_flags[ro * sp.sch.n_cols + c] & CELLF_PRESENT) cells++;
+                if (sp.sch.n_cpx && (out.flags[ro] & PF_HAS_CPX)) cells += out.cpx_count[ro];
+            }
             lo = 0; hi = ch_hist_n;
             while (lo < hi) { int mid = (lo + hi) >> 1; if ((uint64_t)ch_hist_off[mid] < cells) lo = mid + 1; else hi = mid; }
             atomicAdd(&sh_ch[lo], 1u);
@@ -1765,6 +2103,33 @@ __global__ void k_collect_rows(OutParts op, UnfCols out, uint64_t n, uint32_t n_
                     atomicMin(&sh_minttl, (unsigned)cttl);
                     atomicMax(&sh_maxttl, (unsigned)cttl);
                 } else lldt(NO_DELETION_TIME);
+            }
+            if (of & PF_HAS_CPX) {
+                // MetadataCollector.update(complexDeletion) + per-cell stats;
+                // totalColumnsSet counts the COLUMN once when it has cells
+                // (Rows.java:66-82)
+                int64_t xdm = out.cpx_del_mfda[o];
+                uint32_t xdl = out.cpx_del_ldt[o];
+                if (!(xdm == INT64_MIN && xdl == LDT_NONE_U32)) {
+                    lts(xdm);
+                    lldt(ldt_long(xdl));
+                    tomb_push(st, tomb_ldts, tomb_cap, xdl);
+                }
+                uint32_t nc2 = out.cpx_count[o];
+                if (nc2) atomicAdd(&sh_cells, 1ull);
+                uint64_t si = out.cpx_start[o];
+                for (uint32_t e2 = 0; e2 < nc2; e2++) {
+                    uint64_t xe = si + e2;
+                    lts(out.cpx.ts[xe]);
+                    uint32_t cldt = out.cpx.ldt[xe];
+                    int32_t cttl = out.cpx.ttl[xe];
+                    if (cldt != LDT_NONE_U32 && cttl == 0) { lldt(ldt_long(cldt)); tomb_push(st, tomb_ldts, tomb_cap, cldt); }
+                    else if (cttl != 0) {
+                        lldt(ldt_long(cldt));
+                        atomicMin(&sh_minttl, (unsigned)cttl);
+                        atomicMax(&sh_maxttl, (unsigned)cttl);
+                    } else lldt(NO_DELETION_TIME);
+                }
             }
         }
         atomicMin(&st->first_group, (unsigned long long)g);
@@ -2436,6 +2801,8 @@ struct GenParams2 {
     uint32_t n_value_cols;    // regular columns val0..valN-1 (1..63)
     uint32_t col_missing_pct; // P(cell absent) per live row and column
     uint32_t ttl_pct;         // P(live row written expiring) — oracle gen_row_expiring
+    uint32_t complex_pct;     // P(live row carries 'zm' map cells) — gen_has_complex
+    uint32_t complex_del_pct; // P(those rows also carry a complexDeletion)
     int64_t base_ts, base_ldt;
 };
 
@@ -2474,12 +2841,60 @@ __device__ inline uint32_t gen2_ldt(const GenParams2& gp, uint64_t key_id, uint6
     return (uint32_t)(gp.base_ldt + (int64_t)(splitmix64(key_id ^ salt) % 1000));
 }
 
+// complex-column derivations (oracle gen.h: gen_has_complex/gen_cpx_*):
+// candidate paths from a 40-value space, dedup+sorted per row
+__device__ inline bool gen2_has_cpx(const GenParams2& gp, uint64_t id, uint32_t j) {
+    if (gp.complex_pct == 0) return false;
+    return splitmix64(gp.seed ^ 0xC0113C71ULL ^ id ^ ((uint64_t)gp.sst << 32) ^
+                      (uint64_t)(j + 5) * 157) % 100 < gp.complex_pct;
+}
+__device__ inline bool gen2_has_cpx_del(const GenParams2& gp, uint64_t id, uint32_t j) {
+    if (gp.complex_del_pct == 0) return false;
+    return splitmix64(gp.seed ^ 0xCDE1CDE1ULL ^ id ^ ((uint64_t)gp.sst << 32) ^
+                      (uint64_t)(j + 2) * 211) % 100 < gp.complex_del_pct;
+}
+__device__ inline uint32_t gen2_cpx_paths(const GenParams2& gp, uint64_t id, uint32_t j,
+                                          uint32_t* pv) {
+    uint32_t nc = 1 + (uint32_t)(splitmix64(id ^ 0xCE11C07ULL ^ (uint64_t)(j + 1) * 19) % 4);
+    for (uint32_t e = 0; e < nc; e++)
+        pv[e] = (uint32_t)(splitmix64(id ^ 0x9A7B9A7BULL ^ ((uint64_t)gp.sst << 24) ^
+                                      (uint64_t)(j + 1) * 23 ^ (uint64_t)(e + 1) * 71) % 40);
+    // insertion sort + unique over <=4 values
+    for (uint32_t a = 1; a < nc; a++) {
+        uint32_t v = pv[a];
+        int b = (int)a - 1;
+        while (b >= 0 && pv[b] > v) { pv[b + 1] = pv[b]; b--; }
+        pv[b + 1] = v;
+    }
+    uint32_t w = nc ? 1 : 0;
+    for (uint32_t a = 1; a < nc; a++)
+        if (pv[a] != pv[a - 1]) pv[w++] = pv[a];
+    return w;
+}
+
 // per-partition unfiltered count (upper bound used for layout: exact)
 __global__ void k_gen_count(GenParams2 gp, const MRec* sorted, const uint64_t* ids, uint64_t n,
-                            OutParts op, uint64_t* prow_count, const uint8_t* keys) {
+                            OutParts op, uint64_t* prow_count, const uint8_t* keys,
+                            uint64_t* cpx_count = nullptr) {
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
     uint64_t id = ids[sorted[i].idx];
+    if (cpx_count) {
+        uint64_t xc = 0;
+        uint32_t pv[4];
+        uint32_t nrows = gp.clustering_rows ? gp.clustering_rows : 1;
+        bool pdel = gp.clustering_rows == 0 && gp.partition_del_pct &&
+                    (splitmix64(gp.seed ^ 0xFEEDULL ^ id ^ ((uint64_t)gp.sst << 32)) % 100 < gp.partition_del_pct);
+        if (!pdel)
+            for (uint32_t j = 0; j < nrows; j++) {
+                bool tomb = gp.tombstone_pct &&
+                            (splitmix64(gp.seed ^ 0xDEADULL ^ id ^ ((uint64_t)gp.sst << 32) ^
+                                        (gp.clustering_rows ? (uint64_t)(j + 7) * 131 : 0)) % 100 <
+                             gp.tombstone_pct);
+                if (!tomb && gen2_has_cpx(gp, id, j)) xc += gen2_cpx_paths(gp, id, j, pv);
+            }
+        cpx_count[i] = xc;
+    }
     uint32_t cnt;
     if (gp.clustering_rows == 0) {
         bool pdel = gp.partition_del_pct && (splitmix64(gp.seed ^ 0xFEEDULL ^ id ^ ((uint64_t)gp.sst << 32)) % 100 < gp.partition_del_pct);
@@ -2505,12 +2920,14 @@ __global__ void k_gen_count(GenParams2 gp, const MRec* sorted, const uint64_t* i
 
 __global__ void k_gen_fill2(GenParams2 gp, const MRec* sorted, const uint64_t* ids, uint64_t n,
                             OutParts op, UnfCols out, const uint64_t* row_base,
-                            uint8_t* values, uint8_t* ck_arena, uint8_t* static_vals) {
+                            uint8_t* values, uint8_t* ck_arena, uint8_t* static_vals,
+                            uint8_t* cpx_bytes = nullptr, const uint64_t* cpx_base = nullptr) {
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
     uint64_t id = ids[sorted[i].idx];
     uint64_t ob = row_base[i];
     op.row_base[i] = ob;
+    uint64_t cpx_cur = cpx_base ? cpx_base[i] : 0;
     if (gp.static_pct) {
         // oracle contract: gen_has_static / gen_static_ts; one blob cell
         bool has = splitmix64(gp.seed ^ 0x57A71CULL ^ id ^ ((uint64_t)gp.sst << 32)) % 100 <
@@ -2563,6 +2980,12 @@ __global__ void k_gen_fill2(GenParams2 gp, const MRec* sorted, const uint64_t* i
         for (uint32_t c = 0; c < NCV; c++) out.cell_flags[o * NCV + c] = 0;
         out.rkind[o] = kind;
         put_ck(o, ck, true, false, NCK == 2 ? 1 : NCK);  // 2-col mode: prefix bounds
+        if (gp.complex_pct) {
+            out.cpx_del_mfda[o] = INT64_MIN;
+            out.cpx_del_ldt[o] = LDT_NONE_U32;
+            out.cpx_start[o] = cpx_cur;
+            out.cpx_count[o] = 0;
+        }
         out.flags[o] = 0;
         out.live_ts[o] = NO_TIMESTAMP;
         out.live_ttl[o] = 0;
@@ -2586,6 +3009,12 @@ __global__ void k_gen_fill2(GenParams2 gp, const MRec* sorted, const uint64_t* i
         out.start_ldt[o] = LDT_NONE_U32;
         out.live_ttl[o] = 0;
         out.live_let[o] = NO_DELETION_TIME;
+        if (gp.complex_pct) {
+            out.cpx_del_mfda[o] = INT64_MIN;
+            out.cpx_del_ldt[o] = LDT_NONE_U32;
+            out.cpx_start[o] = cpx_cur;
+            out.cpx_count[o] = 0;
+        }
         if (tomb) {
             out.flags[o] = PF_HAS_ROW | PF_ROW_DEL;
             out.live_ts[o] = NO_TIMESTAMP;
@@ -2624,6 +3053,35 @@ __global__ void k_gen_fill2(GenParams2 gp, const MRec* sorted, const uint64_t* i
                 out.cell_ttl[oc] = expg ? ettl : 0;
                 out.val_addr[oc] = (uint64_t)(values + oc * (uint64_t)gp.value_len);
                 out.val_len[oc] = gp.value_len;
+            }
+            // complex column 'zm': dedup+sorted map cells (oracle put_complex)
+            if (gp.complex_pct && gen2_has_cpx(gp, id, rowj)) {
+                out.flags[o] |= PF_HAS_CPX;
+                if (gen2_has_cpx_del(gp, id, rowj)) {
+                    out.cpx_del_mfda[o] = ts - 1;
+                    out.cpx_del_ldt[o] = gen2_ldt(gp, id * 5 + rowj, 0xCD);
+                }
+                uint32_t pv[4];
+                uint32_t nc2 = gen2_cpx_paths(gp, id, rowj, pv);
+                out.cpx_count[o] = nc2;
+                for (uint32_t e = 0; e < nc2; e++) {
+                    uint64_t xe = cpx_cur + e;
+                    uint8_t* b = cpx_bytes + xe * 12;
+                    uint32_t p4 = pv[e];
+                    for (int bi = 0; bi < 4; bi++) b[bi] = (uint8_t)(p4 >> (8 * (3 - bi)));
+                    uint64_t w = splitmix64(gp.seed ^ id * 131 ^ ((uint64_t)gp.sst << 40) ^
+                                            (uint64_t)(rowj + 1) * 29 ^ (uint64_t)(p4 + 1) * 389);
+                    for (int bi = 0; bi < 8; bi++) b[4 + bi] = (uint8_t)(w >> (8 * bi));
+                    out.cpx.ts[xe] = ts - (int64_t)(p4 % 3);
+                    out.cpx.ldt[xe] = LDT_NONE_U32;
+                    out.cpx.ttl[xe] = 0;
+                    out.cpx.flags[xe] = CELLF_PRESENT | CELLF_HAS_VALUE;
+                    out.cpx.path_addr[xe] = (uint64_t)b;
+                    out.cpx.path_len[xe] = 4;
+                    out.cpx.val_addr[xe] = (uint64_t)(b + 4);
+                    out.cpx.val_len[xe] = 8;
+                }
+                cpx_cur += nc2;
             }
         }
     };

@@ -163,6 +163,11 @@ typedef struct gpuc_gen_spec {
     uint32_t ttl_pct;           /* percent of live rows written with expiring
                                    liveness/cells (LivenessInfo.java:67,
                                    AbstractCell.java:53-76) */
+    uint32_t complex_pct;       /* percent of live rows with cells in one complex
+                                   column 'zm' map<blob,blob> (last regular column;
+                                   ComplexColumnData.java:47) */
+    uint32_t complex_del_pct;   /* percent of those rows also carrying a
+                                   complexDeletion */
 } gpuc_gen_spec;
 
 int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* error, size_t error_len);
