@@ -965,7 +965,7 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
         tw_d2h += wallm() - tq2;
         uint64_t off0 = file_off;
         wfut[slot] = std::async(std::launch::async, [=]() {
-            int nth = 12;  // tmpfs/page-cache writes are memcpy-bound and scale with threads
+            int nth = 4;  // page-cache pwrites peak at ~4 threads (tools/wbench.c matrix)
             std::vector<std::thread> th;
             size_t per = (slab_bytes + nth - 1) / nth;
             for (int t = 0; t < nth; t++) {
