@@ -111,29 +111,19 @@ __device__ inline int snp_wave_compress(const uint8_t* __restrict__ s, uint32_t 
                     uint32_t h_l = snp_hash(v_l, shift);
                     uint16_t t_l = s_table[h_l];
                     uint32_t spec_cand = valid ? rd32g((int)t_l) : 0;
-                    // marker round (volatile: must observe LDS conflict rule)
-                    {
-                        volatile uint16_t* vt = s_table;
-                        vt[h_l] = (uint16_t)lane;
-                    }
-                    int maxgroup;
-                    {
-                        volatile const uint16_t* vt = s_table;
-                        maxgroup = (int)vt[h_l];
-                    }
+                    // in-window duplicate groups via hash-bit ballots, all in
+                    // registers (same transform as the LZ4 kernel): no LDS
+                    // marker round trip, no table clobber/restore. hash_bits
+                    // = 32 - shift (snappy's table is input-size dependent).
                     int pred = -1;
                     {
-                        uint64_t G = __ballot(maxgroup != lane);
-                        while (G) {
-                            int g = (int)__ffsll((long long)G) - 1;
-                            int mg = __shfl(maxgroup, g);
-                            uint64_t members = __ballot(maxgroup == mg);
-                            if (maxgroup == mg) {
-                                uint64_t below = members & ((1ULL << lane) - 1);
-                                if (below) pred = 63 - (int)__clzll((long long)below);
-                            }
-                            G &= ~members;
+                        uint64_t mem = ~0ull;
+                        for (int j = 0; j < 32 - shift; j++) {
+                            uint64_t Bj = __ballot(((h_l >> j) & 1u) != 0);
+                            mem &= ((h_l >> j) & 1u) ? Bj : ~Bj;
                         }
+                        uint64_t below = mem & ((1ULL << lane) - 1);
+                        if (below) pred = 63 - (int)__clzll((long long)below);
                     }
                     int pred_idx = pred >= 0 ? pred : 0;
                     int pred_pos = __shfl(p_l, pred_idx);
@@ -148,10 +138,12 @@ __device__ inline int snp_wave_compress(const uint8_t* __restrict__ s, uint32_t 
                     bool have_match = first_event < first_abort && first_event < WAVE;
                     int commit_hi = have_match ? first_event
                                                : (first_abort < WAVE ? first_abort - 1 : WAVE - 1);
-                    {
+                    // commit only (the table was never clobbered): conflicting
+                    // same-slot stores retire highest-lane-last == the scalar
+                    // loop's last-write-wins order
+                    if (lane <= commit_hi) {
                         volatile uint16_t* vt = s_table;
-                        vt[h_l] = t_l;
-                        if (lane <= commit_hi) vt[h_l] = (uint16_t)p_l;
+                        vt[h_l] = (uint16_t)p_l;
                     }
                     if (have_match) {
                         ip = __shfl(p_l, first_event);
@@ -369,8 +361,21 @@ __global__ void __launch_bounds__(WAVE) k_snappy_compress_chunks(
                                 lane);
     if (lane == 0) {
         csize[c] = (uint32_t)csz;
+        // slice-by-8 (crc_table is the 8x256 sliced set): 8 independent
+        // table hits per serial round instead of one
         uint32_t crc = 0xFFFFFFFFu;
-        for (int i = 0; i < csz; i++) crc = crc_table[(crc ^ dst[i]) & 0xFF] ^ (crc >> 8);
+        int i = 0;
+        for (; i + 8 <= csz; i += 8) {
+            uint32_t lo, hi;
+            memcpy(&lo, dst + i, 4);
+            memcpy(&hi, dst + i + 4, 4);
+            lo ^= crc;
+            crc = crc_table[7 * 256 + (lo & 0xFF)] ^ crc_table[6 * 256 + ((lo >> 8) & 0xFF)] ^
+                  crc_table[5 * 256 + ((lo >> 16) & 0xFF)] ^ crc_table[4 * 256 + (lo >> 24)] ^
+                  crc_table[3 * 256 + (hi & 0xFF)] ^ crc_table[2 * 256 + ((hi >> 8) & 0xFF)] ^
+                  crc_table[1 * 256 + ((hi >> 16) & 0xFF)] ^ crc_table[0 * 256 + (hi >> 24)];
+        }
+        for (; i < csz; i++) crc = crc_table[(crc ^ dst[i]) & 0xFF] ^ (crc >> 8);
         ccrc[c] = ~crc;
     }
 }

@@ -3,6 +3,8 @@
 // decompress must round-trip model-compressed chunks. Plus an A/B content
 // throughput sweep. Build with hipcc; needs an MI355X.
 #include <hip/hip_runtime.h>
+#include "../../cassandra_amd/csrc/codec.h"
+#include "../../cassandra_amd/csrc/gpu_structs.h"
 #include "../../cassandra_amd/csrc/snappy_wave.h"
 #include "snappy_sim.h"
 
