@@ -5,7 +5,7 @@
 #include <hip/hip_runtime.h>
 #include "../../cassandra_amd/csrc/codec.h"
 #include "../../cassandra_amd/csrc/gpu_structs.h"
-#include "../../cassandra_amd/csrc/snappy_wave.h"
+#include "../../cassandra_amd/csrc/kernels.hip"
 #include "snappy_sim.h"
 
 #include <cstdio>
