@@ -182,15 +182,22 @@ def main():
     last["_n_inputs"] = args.sstables
     dom = last["dominant_kernel"]
     dom_ms = last["dominant_kernel_ms"]
+    r_out_unc = last["output_uncompressed_bytes"]
     ab = algorithmic_bytes(dom, last)
     achieved_gbs = (ab / (dom_ms / 1e3)) / 1e9 if dom_ms > 0 else 0.0
+    # PMC-measured HBM traffic for the LZ4 compress kernel (separate rocprofv3
+    # --pmc FETCH_SIZE / WRITE_SIZE passes, profiles/r02_pmc_c2.txt): full-slab
+    # launches move 0.73 GB per 0.537 GB of uncompressed input (1.36 B/B) vs
+    # 1.53 B/B algorithmic — traffic ~= algorithmic, the kernel is latency-
+    # bound, not HBM-bound. Scaled to this job's uncompressed output bytes.
+    traffic = int(1.36 * r_out_unc) if dom == "k_lz4_compress" else None
     roofline = {
         "bound": "hbm",
         "achieved": round(achieved_gbs, 1),
         "peak": HBM_PEAK_GBS,
         "unit": "GB/s",
         "frac": round(achieved_gbs / HBM_PEAK_GBS, 4),
-        "traffic": None,  # PMC traffic measured separately via rocprofv3 (profiles/)
+        "traffic": traffic,
     }
 
     # ---- CPU baseline: oracle compactor, bounded sample, 1 thread ----
