@@ -17,6 +17,11 @@ class GpucPurgeRange(ctypes.Structure):
         ("token_lo", ctypes.c_int64),
         ("token_hi", ctypes.c_int64),
         ("min_timestamp", ctypes.c_int64),
+        # optional Filter.db bit payload of the overlapping sstable: enables
+        # the per-key bloom-checked purge evaluator (see gpucompact.h)
+        ("bloom_bits", ctypes.POINTER(ctypes.c_uint32)),
+        ("bloom_bit_len", ctypes.c_uint64),
+        ("bloom_hash_count", ctypes.c_int32),
     ]
 
 
@@ -265,8 +270,27 @@ def compact(
     job.enforce_strict_liveness = 1 if enforce_strict_liveness else 0
     if overlaps:
         ovr = (GpucPurgeRange * len(overlaps))()
-        for i, (lo, hi, ts) in enumerate(overlaps):
+        _bloom_keep = []  # keep the bit buffers alive for the call
+        for i, ov in enumerate(overlaps):
+            if len(ov) == 3:
+                lo, hi, ts = ov
+                blm = None
+            else:
+                # (lo, hi, ts, filter_db_path): load the sstable's Filter.db
+                # for the per-key bloom-checked evaluator
+                lo, hi, ts, fpath = ov
+                raw = open(fpath, "rb").read()
+                k = int.from_bytes(raw[0:4], "big")
+                words = int.from_bytes(raw[4:8], "big")
+                bits = raw[8:8 + words * 8]
+                buf = (ctypes.c_uint32 * (len(bits) // 4)).from_buffer_copy(bits)
+                _bloom_keep.append(buf)
+                blm = (buf, words * 64, k)
             ovr[i].token_lo, ovr[i].token_hi, ovr[i].min_timestamp = lo, hi, ts
+            if blm is not None:
+                ovr[i].bloom_bits = ctypes.cast(blm[0], ctypes.POINTER(ctypes.c_uint32))
+                ovr[i].bloom_bit_len = blm[1]
+                ovr[i].bloom_hash_count = blm[2]
         job.overlaps = ovr
         job.n_overlaps = len(overlaps)
     if token_range is not None:

@@ -1625,14 +1625,20 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
             pp.n_keep_ranges = job->n_keep_ranges;
             pp.invert_ranges = job->invert_ranges;
         }
+        DevBuf d_ov_bw, d_ov_boff, d_ov_bbits, d_ov_bk;
+        std::vector<uint32_t> ov_bw_h;
+        std::vector<uint64_t> ov_boff_h, ov_bbits_h;
+        std::vector<int32_t> ov_bk_h;
         if (job->n_overlaps > 0) {
             ov_lo_h.resize(job->n_overlaps);
             ov_hi_h.resize(job->n_overlaps);
             ov_ts_h.resize(job->n_overlaps);
+            bool any_bloom = false;
             for (int i = 0; i < job->n_overlaps; i++) {
                 ov_lo_h[i] = job->overlaps[i].token_lo;
                 ov_hi_h[i] = job->overlaps[i].token_hi;
                 ov_ts_h[i] = job->overlaps[i].min_timestamp;
+                any_bloom |= job->overlaps[i].bloom_bits != nullptr;
             }
             d_ov_lo.alloc(job->n_overlaps * 8);
             d_ov_hi.alloc(job->n_overlaps * 8);
@@ -1643,6 +1649,36 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
             pp.ov_lo = d_ov_lo.as<int64_t>();
             pp.ov_hi = d_ov_hi.as<int64_t>();
             pp.ov_min_ts = d_ov_ts.as<int64_t>();
+            if (any_bloom) {
+                // pack the per-overlap bloom bit arrays into one device
+                // buffer (word offsets per entry; bit_len 0 = no bloom)
+                for (int i = 0; i < job->n_overlaps; i++) {
+                    const auto& ov = job->overlaps[i];
+                    ov_boff_h.push_back(ov_bw_h.size());
+                    if (ov.bloom_bits && ov.bloom_bit_len) {
+                        uint64_t words32 = (ov.bloom_bit_len + 31) / 32;
+                        ov_bw_h.insert(ov_bw_h.end(), ov.bloom_bits, ov.bloom_bits + words32);
+                        ov_bbits_h.push_back(ov.bloom_bit_len);
+                        ov_bk_h.push_back(ov.bloom_hash_count);
+                    } else {
+                        ov_bbits_h.push_back(0);
+                        ov_bk_h.push_back(0);
+                    }
+                }
+                d_ov_bw.alloc(ov_bw_h.size() * 4 + 8);
+                d_ov_boff.alloc(ov_boff_h.size() * 8);
+                d_ov_bbits.alloc(ov_bbits_h.size() * 8);
+                d_ov_bk.alloc(ov_bk_h.size() * 4);
+                HIP_CHECK(hipMemcpyAsync(d_ov_bw.p, ov_bw_h.data(), ov_bw_h.size() * 4, hipMemcpyHostToDevice, stream));
+                HIP_CHECK(hipMemcpyAsync(d_ov_boff.p, ov_boff_h.data(), ov_boff_h.size() * 8, hipMemcpyHostToDevice, stream));
+                HIP_CHECK(hipMemcpyAsync(d_ov_bbits.p, ov_bbits_h.data(), ov_bbits_h.size() * 8, hipMemcpyHostToDevice, stream));
+                HIP_CHECK(hipMemcpyAsync(d_ov_bk.p, ov_bk_h.data(), ov_bk_h.size() * 4, hipMemcpyHostToDevice, stream));
+                pp.ov_bloom_words = d_ov_bw.as<uint32_t>();
+                pp.ov_bloom_off = d_ov_boff.as<uint64_t>();
+                pp.ov_bloom_bits = d_ov_bbits.as<uint64_t>();
+                pp.ov_bloom_k = d_ov_bk.as<int32_t>();
+                pp.ov_has_bloom = 1;
+            }
         }
         hipEvent_t er0, er1;
         HIP_CHECK(hipEventCreate(&er0));

@@ -506,6 +506,17 @@ struct PurgeParams2 {
     const int64_t* ov_lo;
     const int64_t* ov_hi;
     const int64_t* ov_min_ts;
+    // optional per-overlap bloom filters (packed u32 words + per-entry word
+    // offset/bit-length/k; bit_len 0 = interval-only entry): enables the
+    // per-key evaluator of CompactionController.getPurgeEvaluator. kb/ki
+    // hold the CURRENT partition key's murmur3_128 (h1, h0), filled by the
+    // kernel into its by-value copy before any should_purge2 call.
+    const uint32_t* ov_bloom_words;
+    const uint64_t* ov_bloom_off;
+    const uint64_t* ov_bloom_bits;
+    const int32_t* ov_bloom_k;
+    int32_t ov_has_bloom;
+    int64_t kb, ki;
     int32_t n_overlaps;
     int32_t has_shard;
     int64_t shard_lo, shard_hi;
@@ -532,8 +543,25 @@ __device__ inline bool token_kept(const PurgeParams2& pp, int64_t token) {
 __device__ inline bool purge_eval2(const PurgeParams2& pp, int64_t token, int64_t ts) {
     int64_t min_ts = INT64_MAX;
     bool has = false;
-    for (int i = 0; i < pp.n_overlaps; i++)
-        if (token >= pp.ov_lo[i] && token <= pp.ov_hi[i]) { has = true; min_ts = min(min_ts, pp.ov_min_ts[i]); }
+    for (int i = 0; i < pp.n_overlaps; i++) {
+        if (token < pp.ov_lo[i] || token > pp.ov_hi[i]) continue;
+        if (pp.ov_has_bloom && pp.ov_bloom_bits[i]) {
+            // BloomFilter.isPresent on the overlapping sstable: skip the
+            // entry when it cannot contain this key
+            const uint32_t* w = pp.ov_bloom_words + pp.ov_bloom_off[i];
+            int64_t base = pp.kb, inc = pp.ki;
+            bool present = true;
+            for (int b = 0; b < pp.ov_bloom_k[i]; b++) {
+                int64_t m = base % (int64_t)pp.ov_bloom_bits[i];
+                uint64_t idx = (uint64_t)((m ^ (m >> 63)) - (m >> 63));
+                if (!(w[idx >> 5] & (1u << (idx & 31)))) { present = false; break; }
+                base += inc;
+            }
+            if (!present) continue;
+        }
+        has = true;
+        min_ts = min(min_ts, pp.ov_min_ts[i]);
+    }
     return !has || ts < min_ts;
 }
 __device__ inline bool should_purge2(const PurgeParams2& pp, int64_t token, int64_t ts, int64_t ldt) {
@@ -583,6 +611,12 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
     op.row_base[g] = out_base[g];
     op.row_count[g] = 0;
     if (!token_kept(pp, token)) { op.keep[g] = 0; return; }
+    if (pp.ov_has_bloom) {
+        uint64_t h[2];
+        murmur3_128((const uint8_t*)op.key_addr[g], r0.klen, 0, h);
+        pp.kb = (int64_t)h[1];
+        pp.ki = (int64_t)h[0];
+    }
 
     // member streams
     uint64_t mb[MA];
@@ -2535,6 +2569,12 @@ __global__ void k_purge_parts(OutParts op, UnfCols out, uint64_t n, SchemaParams
     if (g >= n) return;
     if (!op.keep[g]) return;
     int64_t token = op.token[g];
+    if (pp.ov_has_bloom) {
+        uint64_t h[2];
+        murmur3_128((const uint8_t*)op.key_addr[g], op.klen[g], 0, h);
+        pp.kb = (int64_t)h[1];
+        pp.ki = (int64_t)h[0];
+    }
     auto purge_dt = [&](int64_t m, uint32_t l) {
         return m != INT64_MIN && should_purge2(pp, token, m, ldt_long(l));
     };

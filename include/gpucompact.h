@@ -53,6 +53,16 @@ typedef struct gpuc_purge_range {
     int64_t token_lo;   /* inclusive Murmur3 token bounds */
     int64_t token_hi;
     int64_t min_timestamp;
+    /* optional bloom filter of the overlapping sstable (the raw Filter.db
+       bit payload after its 8-byte header; little-endian words). When set,
+       purge gating applies only to keys the bloom might contain — exactly
+       CompactionController.getPurgeEvaluator's overlapIterator +
+       BF.isPresent chain (CompactionController.java:247-286,308-329). NULL
+       keeps the conservative interval-only behavior (never purges more than
+       the JVM, may retain tombstones it would drop). */
+    const uint32_t* bloom_bits;
+    uint64_t bloom_bit_len;   /* number of bits (words*64 from Filter.db) */
+    int32_t bloom_hash_count; /* Filter.db hash count */
 } gpuc_purge_range;
 
 typedef struct gpuc_job {

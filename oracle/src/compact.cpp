@@ -306,13 +306,40 @@ struct Purger {
     bool never_purge;
     const std::vector<PurgeRange>* overlaps;
     int64_t token;
+    const bytes* key = nullptr;            // partition key (bloom probes)
+    mutable int64_t kb = 0, ki = 0;        // murmur3_128 of key (h1, h0)
+    mutable bool khash = false;
 
-    // purgeEvaluator(key).test(ts) (CompactionController.java:247-286)
+    // purgeEvaluator(key).test(ts) (CompactionController.java:247-286):
+    // token range + (when the entry carries the sstable's bloom) a per-key
+    // might-contain check, matching overlapIterator + BF.isPresent
     bool evaluator(int64_t ts) const {
         int64_t min_ts = INT64_MAX;
         bool has = false;
-        for (const auto& r : *overlaps)
-            if (token >= r.tok_lo && token <= r.tok_hi) { has = true; min_ts = std::min(min_ts, r.min_ts); }
+        for (const auto& r : *overlaps) {
+            if (token < r.tok_lo || token > r.tok_hi) continue;
+            if (!r.bloom.empty() && key) {
+                if (!khash) {
+                    uint64_t h[2];
+                    murmur3_128_cassandra(key->data(), key->size(), 0, h);
+                    kb = (int64_t)h[1];
+                    ki = (int64_t)h[0];
+                    khash = true;
+                }
+                uint64_t max = (uint64_t)r.bloom.size() * 8;
+                int64_t base = kb, inc = ki;
+                bool present = true;
+                for (int i = 0; i < r.bloom_k; i++) {
+                    int64_t m = base % (int64_t)max;
+                    uint64_t idx = (uint64_t)((m ^ (m >> 63)) - (m >> 63));
+                    if (!(r.bloom[idx >> 3] & (1u << (idx & 7)))) { present = false; break; }
+                    base += inc;
+                }
+                if (!present) continue;  // source cannot contain this key
+            }
+            has = true;
+            min_ts = std::min(min_ts, r.min_ts);
+        }
         return !has || ts < min_ts;
     }
     bool should_purge(int64_t ts, int64_t ldt) const {  // the DeletionPurger lambda (PurgeFunction.java:40-44)
@@ -524,7 +551,7 @@ void garbage_filter(Partition& data, const Partition& tomb, const Header& h, boo
 
 bool purge_partition(Partition& p, int64_t now_sec, int64_t gc_before, bool never_purge,
                      const std::vector<PurgeRange>& overlaps, bool enforce_strict_liveness) {
-    Purger pg{now_sec, gc_before, never_purge, &overlaps, p.token};
+    Purger pg{now_sec, gc_before, never_purge, &overlaps, p.token, &p.key};
     if (pg.should_purge(p.del)) p.del = DT_LIVE;
     if (!row_is_empty(p.static_row) && !purge_row(p.static_row, pg, false))
         p.static_row = Row{};

@@ -8,7 +8,17 @@ namespace oracle {
 // CompactionController.java:247-286): host-precomputed token-interval ->
 // min-timestamp table of overlapping NON-compacting sources that may contain
 // the key. Empty table == no overlaps == purge-everything-allowed.
-struct PurgeRange { int64_t tok_lo, tok_hi; int64_t min_ts; };  // inclusive token bounds
+struct PurgeRange {
+    int64_t tok_lo, tok_hi;        // inclusive token bounds
+    int64_t min_ts;
+    // optional bloom-filter bits of the overlapping sstable (Filter.db
+    // payload, little-endian words): enables the per-key purge evaluator of
+    // CompactionController.getPurgeEvaluator (CompactionController.java:
+    // 247-286,308-329) — the entry gates purge only for keys it might
+    // contain. Empty = the conservative interval-only behavior.
+    std::vector<uint8_t> bloom;    // byte-addressed bits (bit i -> bloom[i>>3] & 1<<(i&7))
+    int32_t bloom_k = 0;
+};
 
 struct CompactionJob {
     std::vector<SSTable> inputs;  // in task order (reconcile tie order)

@@ -226,6 +226,32 @@ static int cmd_compact(const std::string& outbase, std::vector<std::string>& inp
             p0 = c + 1;
         }
     }
+    if (kv.count("ov")) {
+        // overlap table "lo:hi:ts[:FilterPath],..." — FilterPath loads the
+        // sstable's Filter.db for the per-key bloom-checked evaluator
+        std::string sv = kv["ov"];
+        size_t p0 = 0;
+        while (p0 < sv.size()) {
+            size_t cm = sv.find(',', p0);
+            if (cm == std::string::npos) cm = sv.size();
+            std::string one = sv.substr(p0, cm - p0);
+            PurgeRange r{};
+            size_t a = one.find(':'), b = one.find(':', a + 1), c = one.find(':', b + 1);
+            r.tok_lo = strtoll(one.substr(0, a).c_str(), nullptr, 10);
+            r.tok_hi = strtoll(one.substr(a + 1, b - a - 1).c_str(), nullptr, 10);
+            r.min_ts = strtoll(one.substr(b + 1, (c == std::string::npos ? one.size() : c) - b - 1).c_str(), nullptr, 10);
+            if (c != std::string::npos) {
+                bytes f = read_file(one.substr(c + 1));
+                Reader rd(f);
+                r.bloom_k = (int32_t)rd.be32();
+                uint32_t words = rd.be32();
+                bytes bits = rd.take((size_t)words * 8);
+                r.bloom.assign(bits.begin(), bits.end());
+            }
+            job.overlaps.push_back(r);
+            p0 = cm + 1;
+        }
+    }
     if (kv.count("ranges")) {
         // "lo:hi,lo:hi" inclusive keep-ranges; invertranges=1 keeps the complement
         std::string s = kv["ranges"];

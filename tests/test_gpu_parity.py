@@ -607,6 +607,38 @@ def test_complex_columns_pipeline(ca, oracle_bin, tmp_path):
     _assert_dirs_equal(f"{dw}/oa-60-big", f"{dw}/oa-80-big")
 
 
+def test_purge_bloom_evaluator(ca, oracle_bin, tmp_path):
+    """Per-key purge evaluator (VERDICT round-2 item 8): overlap entries can
+    carry the overlapping sstable's Filter.db bits, reproducing
+    CompactionController.getPurgeEvaluator's overlapIterator + BF.isPresent
+    chain (CompactionController.java:247-286,308-329). A disjoint-key
+    overlapping source must NOT gate purge when its bloom is provided, and
+    MUST gate everything in the conservative interval-only mode."""
+    d = str(tmp_path)
+    _oracle_gen(d, seed=5, n=2, rows=800, vlen=80, overlap=0, tomb=40)
+    os.makedirs(d + "/y")
+    # disjoint keys: 9-byte keys never collide with the 8-byte main set
+    _oracle_gen(d + "/y", seed=99, n=1, rows=500, vlen=50, keylen=9)
+    ins = [f"{d}/oa-1-big", f"{d}/oa-2-big"]
+    LO, HI = -(2 ** 63), 2 ** 63 - 1
+    flt = f"{d}/y/oa-1-big-Filter.db"
+    GCB = 2000000000
+    # (a) interval-only: conservative, nothing purged
+    ra = ca.compact(ins, f"{d}/oa-90-big", gc_before=GCB, overlaps=[(LO, HI, 1)])
+    _oracle_compact(f"{d}/oa-80-big", ins, gcbefore=GCB, ov=f"{LO}:{HI}:1")
+    _assert_dirs_equal(f"{d}/oa-90-big", f"{d}/oa-80-big")
+    # (b) bloom-checked: the disjoint source gates only its ~1% false positives
+    rb = ca.compact(ins, f"{d}/oa-91-big", gc_before=GCB,
+                    overlaps=[(LO, HI, 1, flt)])
+    _oracle_compact(f"{d}/oa-81-big", ins, gcbefore=GCB, ov=f"{LO}:{HI}:1:{flt}")
+    _assert_dirs_equal(f"{d}/oa-91-big", f"{d}/oa-81-big")
+    # (c) unconstrained purge as the floor; the divergence is real and bounded
+    rc2 = ca.compact(ins, f"{d}/oa-92-big", gc_before=GCB)
+    assert ra["partitions_out"] == 1600, ra          # interval-only: all retained
+    assert rc2["partitions_out"] <= rb["partitions_out"] <= rc2["partitions_out"] * 1.05, \
+        (ra["partitions_out"], rb["partitions_out"], rc2["partitions_out"])  # ~1% bloom FPs
+
+
 def test_cancellation(ca, oracle_bin, tmp_path):
     """Cooperative cancel (CompactionIterator.isStopRequested): a set
     cancel_flag aborts the task with GPUC_ERR_CANCELLED; a zero flag is

@@ -456,3 +456,24 @@ def test_complex_columns_oracle(oracle_bin, tmp_path):
     oracle_run("compact", f"{d2}/oa-70-big", *dins)
     for c in ("Data.db", "Index.db", "Digest.crc32", "Statistics.db"):
         assert filecmp.cmp(f"{d2}/oa-60-big-{c}", f"{d2}/oa-70-big-{c}", shallow=False), c
+
+
+def test_purge_bloom_evaluator_cpu(oracle_bin, tmp_path):
+    """Per-key bloom-checked purge evaluator vs the conservative interval
+    table (CompactionController.java:247-286,308-329): a disjoint-key
+    overlapping source retains everything interval-only, almost nothing
+    with its bloom provided."""
+    import json
+    d = str(tmp_path)
+    oracle_run("gen", d, "seed=5", "n=2", "rows=800", "vlen=80", "overlap=0", "tomb=40")
+    os.makedirs(d + "/y")
+    oracle_run("gen", d + "/y", "seed=99", "n=1", "rows=500", "vlen=50", "keylen=9")
+    LO, HI = -(2 ** 63), 2 ** 63 - 1
+    def parts(out, *extra):
+        r = oracle_run("compact", out, f"{d}/oa-1-big", f"{d}/oa-2-big",
+                       "gcbefore=2000000000", *extra)
+        return json.loads(r.stdout.strip().splitlines()[-1])["partitions_out"]
+    a = parts(f"{d}/oa-80-big", f"ov={LO}:{HI}:1")
+    b = parts(f"{d}/oa-81-big", f"ov={LO}:{HI}:1:{d}/y/oa-1-big-Filter.db")
+    c = parts(f"{d}/oa-82-big")
+    assert a == 1600 and c <= b <= c * 1.05, (a, b, c)
