@@ -14,6 +14,7 @@
 #include <cstdio>
 #include <cstring>
 #include <map>
+#include <set>
 #include <memory>
 #include <mutex>
 #include <stdexcept>
@@ -331,14 +332,49 @@ static void put_est_hist(bytes& o, const std::vector<int64_t>& off, const uint64
         put_be64(o, buckets[i]);
     }
 }
-static bytes hll_stub() {
+// HyperLogLogPlus(13, 25) of the partition-key hash2_64 values (clearspring
+// stream-lib; MetadataCollector.java:180-183). SPARSE while under the
+// 0.75*m threshold with no flagged (low-bits-zero) key: (h >>> 39) << 1,
+// delta varints over the sorted set — this encoding byte-reproduces the
+// reference's own oa fixtures (oracle test_compaction_hll_fixture_pin).
+// NORMAL otherwise: 2^13 five-bit registers, six per big-endian 32-bit word.
+static bytes hll_bytes(const std::vector<uint64_t>& key_hashes) {
+    constexpr int P = 13, SP = 25;
     bytes h;
     put_be32(h, (uint32_t)-2);
     auto pv = [&](uint32_t v) { while (v >= 0x80) { h.push_back((uint8_t)(v | 0x80)); v >>= 7; } h.push_back((uint8_t)v); };
-    pv(13); pv(25); pv(0);
-    uint32_t reg_ints = (uint32_t)((((1 << 13) * 5) + 31) / 32);
-    pv(reg_ints * 4);
-    h.insert(h.end(), reg_ints * 4, 0);
+    pv(P); pv(SP);
+    std::set<uint32_t> sparse;
+    bool flagged = false;
+    const uint32_t threshold = (uint32_t)((1u << P) * 3 / 4);
+    for (uint64_t kh : key_hashes) {
+        uint32_t sidx = (uint32_t)(kh >> (64 - SP));
+        if ((sidx & ((1u << (SP - P)) - 1)) == 0) { flagged = true; break; }
+        sparse.insert(sidx << 1);
+        if (sparse.size() > threshold) break;
+    }
+    if (!flagged && sparse.size() <= threshold && key_hashes.size() <= threshold) {
+        pv(1 /*SPARSE*/);
+        pv((uint32_t)sparse.size());
+        uint32_t prev = 0;
+        for (uint32_t v : sparse) { pv(v - prev); prev = v; }
+    } else {
+        pv(0 /*NORMAL*/);
+        std::vector<uint8_t> regs(1u << P, 0);
+        for (uint64_t kh : key_hashes) {
+            uint32_t idx = (uint32_t)(kh >> (64 - P));
+            uint64_t w = (kh << P) | (1ull << (P - 1));
+            uint8_t rho = (uint8_t)(__builtin_clzll(w) + 1);
+            if (rho > regs[idx]) regs[idx] = rho;
+        }
+        uint32_t bits = (1u << P) / 6;
+        uint32_t reg_ints = (bits % 32 == 0) ? bits : bits + 1;  // RegisterSet.getSizeForCount
+        std::vector<uint32_t> M(reg_ints, 0);
+        for (uint32_t i = 0; i < (1u << P); i++)
+            M[i / 6] |= (uint32_t)regs[i] << (5 * (i % 6));
+        pv(reg_ints * 4);
+        for (uint32_t wv : M) put_be32(h, wv);
+    }
     bytes o;
     put_be32(o, (uint32_t)h.size());
     o.insert(o.end(), h.begin(), h.end());
@@ -362,6 +398,7 @@ struct OutMeta {
     double compression_ratio;
     uint64_t part_size_hist[156];
     uint64_t cells_hist[119];
+    std::vector<uint64_t> key_hashes;  // hash2_64 per kept partition (HLL)
 };
 
 static bytes serialize_statistics_out(const OutMeta& m) {
@@ -375,7 +412,7 @@ static bytes serialize_statistics_out(const OutMeta& m) {
         memcpy(&fp, &fpv, 8);
         put_be64(validation, fp);
     }
-    bytes compaction = hll_stub();
+    bytes compaction = hll_bytes(m.key_hashes);
     bytes stats;
     {
         static const std::vector<int64_t> ps_off = est_hist_offsets(155);
@@ -1026,6 +1063,26 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
                 throw std::runtime_error("tombstone list overflow (internal cap)");
         }
         OutMeta m{};
+        {
+            // per-partition key hashes for the COMPACTION HLL (device-hashed,
+            // filtered to kept partitions on host)
+            DevBuf d_kh;
+            d_kh.alloc(n_groups * 8 + 8);
+            uint32_t blocks = (uint32_t)((n_groups + 255) / 256);
+            if (n_groups)
+                hipLaunchKernelGGL(k_key_hash2, dim3(blocks), dim3(256), 0, stream, opb.op,
+                                   n_groups, d_kh.as<uint64_t>());
+            std::vector<uint64_t> kh(n_groups);
+            std::vector<uint8_t> keep(n_groups);
+            if (n_groups) {
+                HIP_CHECK(hipStreamSynchronize(stream));
+                HIP_CHECK(hipMemcpy(kh.data(), d_kh.p, n_groups * 8, hipMemcpyDeviceToHost));
+                HIP_CHECK(hipMemcpy(keep.data(), opb.op.keep, n_groups, hipMemcpyDeviceToHost));
+            }
+            m.key_hashes.reserve(n_groups);
+            for (uint64_t g2 = 0; g2 < n_groups; g2++)
+                if (keep[g2]) m.key_hashes.push_back(kh[g2]);
+        }
         m.hs = sp.hs;
         m.key_type = key_type;
         m.ck_types = ck_types;

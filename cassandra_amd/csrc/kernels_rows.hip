@@ -473,6 +473,43 @@ __global__ void k_parse_rows(const SrcDesc2* srcs, uint32_t n_srcs, uint32_t tot
     (void)error;
 }
 
+// MurmurHash.hash2_64 (utils/MurmurHash.java:96-150), the COMPACTION-HLL
+// key hash (MetadataCollector.java:180-183); tail bytes sign-extended
+// exactly as the reference's java bytes are. Oracle mirror:
+// murmur2_64_cassandra (fixture-pinned).
+__device__ inline uint64_t murmur2_64(const uint8_t* key, uint32_t length) {
+    const uint64_t m = 0xc6a4a7935bd1e995ULL;
+    const int r = 47;
+    uint64_t h = m * (uint64_t)length;
+    uint32_t nl = length >> 3;
+    for (uint32_t i = 0; i < nl; i++) {
+        uint64_t k;
+        memcpy(&k, key + i * 8, 8);
+        k *= m;
+        k ^= k >> r;
+        k *= m;
+        h ^= k;
+        h *= m;
+    }
+    uint32_t rem = length & 7;
+    if (rem) {
+        const uint8_t* t = key + length - rem;
+        for (uint32_t b = rem; b-- > 0;)
+            h ^= (uint64_t)(int64_t)(int8_t)t[b] << (8 * b);
+        h *= m;
+    }
+    h ^= h >> r;
+    h *= m;
+    h ^= h >> r;
+    return h;
+}
+
+__global__ void k_key_hash2(OutParts op, uint64_t n, uint64_t* out) {
+    uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (g >= n) return;
+    out[g] = op.keep[g] ? murmur2_64((const uint8_t*)op.key_addr[g], op.klen[g]) : 0;
+}
+
 __global__ void k_widen_u32(const uint32_t* in, uint64_t* out, uint64_t n) {
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i < n) out[i] = in[i];

@@ -77,15 +77,42 @@ static int cmd_roundtrip(const std::string& base) {
         check("Digest.crc32", bytes(d.begin(), d.end()), read_file(base + "-Digest.crc32"));
     }
     {
-        // Filter.db: recompute bloom from keys
+        // Filter.db: recompute bloom from keys; Statistics COMPACTION blob:
+        // the HLL is real now (sparse fixture-pinned), compare its slice
         SSTable tmp;
         tmp.header = t.header;
         tmp.comp = t.comp;
         tmp.parts = t.parts;
         WriterOut w = write_sstable(tmp);
         check("Filter.db", w.filter, read_file(base + "-Filter.db"));
+        auto comp_slice = [](const bytes& st) -> bytes {
+            if (st.size() < 4) return {};
+            Reader r(st);
+            uint32_t n = r.be32();
+            r.be32();  // toc CRC32 (checksummed metadata v2)
+            uint32_t start = 0, end2 = 0;
+            for (uint32_t i = 0; i < n; i++) {
+                uint32_t typ = r.be32();
+                uint32_t off = r.be32();
+                if (typ == 1) start = off;
+                if (typ == 2) end2 = off;
+            }
+            if (!start || end2 <= start + 4) return {};
+            return bytes(st.begin() + start, st.begin() + end2 - 4);  // strip component CRC
+        };
+        check("Stats[COMPACTION]", comp_slice(w.statistics), comp_slice(t.raw_statistics));
     }
     return rc;
+}
+
+// rewrite: read an sstable and write all components to a new base (HLL /
+// writer-path inspection without compaction semantics)
+static int cmd_rewrite(const std::string& inbase, const std::string& outbase) {
+    SSTable t = read_sstable(inbase, true);
+    WriterOut w = write_sstable(t);
+    write_components(w, outbase);
+    printf("rewrote %zu partitions\n", t.parts.size());
+    return 0;
 }
 
 static GenSpec spec_from_kv(std::map<std::string, std::string>& kv) {
@@ -319,6 +346,7 @@ int main(int argc, char** argv) {
         if (cmd == "selftest") return cmd_selftest();
         if (cmd == "dump") return cmd_dump(pos.at(0));
         if (cmd == "roundtrip") return cmd_roundtrip(pos.at(0));
+        if (cmd == "rewrite") return cmd_rewrite(pos.at(0), pos.at(1));
         if (cmd == "btidump") {
             // BTI (da) Partitions.db [+ Rows.db] dump — reader scaffolding
             bytes pf = read_file(pos.at(0) + "-Partitions.db");

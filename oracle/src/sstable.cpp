@@ -2,6 +2,7 @@
 // Reader/writer for Cassandra big-format `oa` sstables; bit-exact restatement
 // of the serializers cited per function.
 #include "sstable.h"
+#include <set>
 #include "lz4_ref.h"
 #include "snappy_ref.h"
 #include <algorithm>
@@ -620,6 +621,7 @@ struct StatsComponentInput {
     StatsMins mins;
     double compression_ratio = -1;
     bytes first_key, last_key;
+    std::vector<uint64_t> key_hashes;  // hash2_64 per partition key (HLL)
     std::map<uint32_t, uint32_t> tombstone_hist;  // ldt(seconds)->count, ≤100 bins
     double token_space_coverage = 0;
     size_t clustering_count = 0;
@@ -666,17 +668,53 @@ static bytes serialize_stats_component(const StatsComponentInput& s) {
     return out;
 }
 
-// minimal valid-shaped HyperLogLogPlus blob (clearspring stream-lib 2.5.2 normal
-// format, p=13 sp=25 per MetadataCollector). NOT parity-pinned (no JVM here);
-// only structural validity is attempted. COMPACTION component.
-static bytes serialize_compaction_component() {
+// HyperLogLogPlus of the partition-key hash2_64 values (clearspring
+// stream-lib 2.5.x, p=13 sp=25 — MetadataCollector's `new HyperLogLogPlus
+// (13, 25)`), COMPACTION component. SPARSE (format 1) while the distinct
+// sparse-index set stays under the 0.75*m conversion threshold and no key
+// needs the flagged rho encoding (a 2^-(sp-p) event): entries are
+// (hash >>> (64-sp)) << 1, serialized as delta varints over the sorted set
+// — byte-pinned against the reference's own oa fixture. Otherwise NORMAL
+// (format 0): 2^p five-bit registers, six per 32-bit word
+// (RegisterSet.REGISTER_SIZE=5, LOG2_BITS_PER_WORD=6), register =
+// max(nlz((h << p) | 1 << (p-1)) + 1), words big-endian.
+static bytes serialize_compaction_component(const std::vector<uint64_t>& key_hashes) {
+    constexpr int P = 13, SP = 25;
     bytes hll;
     put_be32(hll, (uint32_t)-2);  // -VERSION
     auto put_varint = [&](uint32_t v) { while (v >= 0x80) { hll.push_back((uint8_t)(v | 0x80)); v >>= 7; } hll.push_back((uint8_t)v); };
-    put_varint(13); put_varint(25); put_varint(0 /*NORMAL*/);
-    uint32_t reg_ints = (uint32_t)((((1 << 13) * 5) + 31) / 32);  // RegisterSet word count
-    put_varint(reg_ints * 4);
-    for (uint32_t i = 0; i < reg_ints * 4; i++) hll.push_back(0);
+    put_varint(P); put_varint(SP);
+    std::set<uint32_t> sparse;
+    bool flagged = false;
+    const uint32_t threshold = (uint32_t)((1u << P) * 3 / 4);
+    for (uint64_t h : key_hashes) {
+        uint32_t sidx = (uint32_t)(h >> (64 - SP));
+        if ((sidx & ((1u << (SP - P)) - 1)) == 0) { flagged = true; break; }
+        sparse.insert(sidx << 1);
+        if (sparse.size() > threshold) break;
+    }
+    if (!flagged && sparse.size() <= threshold && key_hashes.size() <= threshold) {
+        put_varint(1 /*SPARSE*/);
+        put_varint((uint32_t)sparse.size());
+        uint32_t prev = 0;
+        for (uint32_t v : sparse) { put_varint(v - prev); prev = v; }
+    } else {
+        put_varint(0 /*NORMAL*/);
+        std::vector<uint8_t> regs(1u << P, 0);
+        for (uint64_t h : key_hashes) {
+            uint32_t idx = (uint32_t)(h >> (64 - P));
+            uint64_t w = (h << P) | (1ull << (P - 1));
+            uint8_t rho = (uint8_t)(__builtin_clzll(w) + 1);
+            if (rho > regs[idx]) regs[idx] = rho;
+        }
+        uint32_t bits = (1u << P) / 6;
+        uint32_t reg_ints = (bits % 32 == 0) ? bits : bits + 1;  // RegisterSet.getSizeForCount
+        std::vector<uint32_t> M(reg_ints, 0);
+        for (uint32_t i = 0; i < (1u << P); i++)
+            M[i / 6] |= (uint32_t)regs[i] << (5 * (i % 6));
+        put_varint(reg_ints * 4);
+        for (uint32_t wv : M) put_be32(hll, wv);
+    }
     bytes out;
     put_be32(out, (uint32_t)hll.size());
     out.insert(out.end(), hll.begin(), hll.end());
@@ -689,7 +727,7 @@ static bytes serialize_statistics(const Header& h, const StatsComponentInput& s)
     bytes validation;
     put_utf(validation, "org.apache.cassandra.dht.Murmur3Partitioner");
     uint64_t fp; double fpv = 0.01; memcpy(&fp, &fpv, 8); put_be64(validation, fp);
-    bytes compaction = serialize_compaction_component();
+    bytes compaction = serialize_compaction_component(s.key_hashes);
     bytes stats = serialize_stats_component(s);
     bytes header = serialize_header_component(h);
     const bytes* comps[4] = {&validation, &compaction, &stats, &header};
@@ -791,6 +829,9 @@ WriterOut write_sstable(const SSTable& t) {
     st.clustering_count = t.header.clustering_types.size();
     if (st.clustering_count) st.clustering_type = t.header.clustering_types[0];
     if (!t.parts.empty()) { st.first_key = t.parts.front().key; st.last_key = t.parts.back().key; }
+    st.key_hashes.reserve(t.parts.size());
+    for (const Partition& p : t.parts)
+        st.key_hashes.push_back(murmur2_64_cassandra(p.key.data(), p.key.size(), 0));
 
     ChunkedOut co = chunk_compress(data_raw, t.comp);
     // compressionRatio = compressedSize/uncompressedSize where compressedSize

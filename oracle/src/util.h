@@ -107,6 +107,37 @@ inline uint64_t fmix_(uint64_t k) {
     k ^= k >> 33; k *= 0xc4ceb9fe1a85ec53ULL;
     k ^= k >> 33; return k;
 }
+// MurmurHash.hash2_64 (utils/MurmurHash.java:96-150) — the partition-key
+// hash feeding the COMPACTION HyperLogLogPlus (MetadataCollector.java:180-183).
+// NOTE the reference's tail bytes are SIGN-EXTENDED (java byte, no & 0xff):
+// transcribed exactly, fixture-pinned (test_compaction_hll_fixture_pin).
+inline uint64_t murmur2_64_cassandra(const uint8_t* key, size_t length, uint64_t seed) {
+    const uint64_t m = 0xc6a4a7935bd1e995ULL;
+    const int r = 47;
+    uint64_t h = (seed & 0xffffffffULL) ^ (m * (uint64_t)length);
+    size_t nl = length >> 3;
+    for (size_t i = 0; i < nl; i++) {
+        uint64_t k;
+        memcpy(&k, key + i * 8, 8);  // little-endian load
+        k *= m;
+        k ^= k >> r;
+        k *= m;
+        h ^= k;
+        h *= m;
+    }
+    size_t rem = length & 7;
+    if (rem) {
+        const uint8_t* t = key + length - rem;
+        for (size_t b = rem; b-- > 0;)
+            h ^= (uint64_t)(int64_t)(int8_t)t[b] << (8 * b);
+        h *= m;
+    }
+    h ^= h >> r;
+    h *= m;
+    h ^= h >> r;
+    return h;
+}
+
 inline void murmur3_128_cassandra(const uint8_t* key, size_t length, uint64_t seed,
                                   uint64_t out[2]) {
     const size_t nblocks = length >> 4;
