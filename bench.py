@@ -93,6 +93,11 @@ def main():
     ap.add_argument("--no-cpu-baseline", action="store_true")
     ap.add_argument("--snappy", action="store_true",
                     help="C3 codec: SnappyCompressor chunks end to end")
+    ap.add_argument("--dry-run", action="store_true",
+                    help="CPU rehearsal of the exact rank path (init, device "
+                         "mapping, barriers, MAX/SUM reductions, JSON emit) with "
+                         "generate/compact mocked — so the first real multi-GPU "
+                         "run is not the first execution of this code")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -105,8 +110,28 @@ def main():
         tdist.init_process_group("gloo")  # control-plane only; data path has no collectives
         dist = tdist
 
-    build_if_needed()
-    import cassandra_amd as ca
+    if args.dry_run:
+        class _MockCa:
+            @staticmethod
+            def device_count():
+                return max(1, world)
+            @staticmethod
+            def generate(*a, **k):
+                pass
+            @staticmethod
+            def compact(bases, out, device=0, n_output_shards=1):
+                os.makedirs(os.path.dirname(out), exist_ok=True)
+                return {"input_uncompressed_bytes": 1 << 20,
+                        "output_uncompressed_bytes": 1 << 20,
+                        "output_compressed_bytes": 1 << 19,
+                        "partitions_in": 100, "partitions_out": 100,
+                        "dominant_kernel": "k_lz4_compress",
+                        "dominant_kernel_ms": 1.0, "dominant_kernel_launches": 1,
+                        "ms": {"total": 1.0}}
+        ca = _MockCa()
+    else:
+        build_if_needed()
+        import cassandra_amd as ca
     if ca.device_count() < 1:
         print(json.dumps({"error": "no GPU visible"}))
         return 1
@@ -126,7 +151,7 @@ def main():
                 snappy=args.snappy)
     t_gen = time.time() - t_gen
     bases = [os.path.join(d, f"oa-{g}-big") for g in range(1, args.sstables + 1)]
-    input_compressed = sum(os.path.getsize(b + "-Data.db") for b in bases)
+    input_compressed = 0 if args.dry_run else sum(os.path.getsize(b + "-Data.db") for b in bases)
 
     import threading
     cleaners = []
@@ -202,7 +227,7 @@ def main():
 
     # ---- CPU baseline: oracle compactor, bounded sample, 1 thread ----
     cpu_baseline = None
-    if world == 1 and not args.no_cpu_baseline:
+    if world == 1 and not args.no_cpu_baseline and not args.dry_run:
         ob = oracle_bin()
         sd = os.path.join(args.dir, "cpu_sample")
         shutil.rmtree(sd, ignore_errors=True)
