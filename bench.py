@@ -274,6 +274,9 @@ def main():
                          if args.clustering_rows else
                          f"C3: {args.sstables}x sstables, Snappy 16KiB chunks"
                          if args.snappy else
+                         f"C5: full major compaction, {args.sstables}x sstables, "
+                         f"LZ4, {args.shards} token shards (vnode ranges)"
+                         if args.shards > 1 else
                          "C2: 8x2GiB sstables, 10% key overlap, ~1KiB values, LZ4 16KiB chunks"),
             "sstables": args.sstables,
             "rows_per_sstable": args.rows,
