@@ -2128,7 +2128,7 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
             res->partitions_in += su.positions[s].size() - 1;
 
         int S = S_pre;
-        if (S > 64) throw std::runtime_error("n_output_shards must be <= 64");
+        if (S > 1024) throw std::runtime_error("n_output_shards must be <= 1024");
         std::mutex res_mu;
         if (S == 1) {
             std::vector<std::pair<uint32_t, uint32_t>> pr(su.k);

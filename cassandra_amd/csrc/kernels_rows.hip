@@ -660,7 +660,9 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
     uint32_t mcnt[MA], mpos[MA];
     int64_t pdm = INT64_MIN;
     uint32_t pdl = LDT_NONE_U32;
-    for (uint32_t m = 0; m < k; m++) {
+    #pragma unroll
+        for (uint32_t m = 0; m < (uint32_t)MA; m++) {
+            if (m >= k) break;
         uint32_t i = src_bases[recs[beg + m].src] + recs[beg + m].idx;
         mb[m] = pc.row_base[i];
         mcnt[m] = pc.row_count[i];
@@ -677,7 +679,9 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
         // recompute un-purged merged value (pdm/pdl may have been purged above)
         int64_t m2 = INT64_MIN;
         uint32_t l2 = LDT_NONE_U32;
-        for (uint32_t m = 0; m < k; m++) {
+        #pragma unroll
+        for (uint32_t m = 0; m < (uint32_t)MA; m++) {
+            if (m >= k) break;
             uint32_t i = src_bases[recs[beg + m].src] + recs[beg + m].idx;
             if (!dt_sup(m2, l2, pc.pdel_mfda[i], pc.pdel_ldt[i])) { m2 = pc.pdel_mfda[i]; l2 = pc.pdel_ldt[i]; }
         }
@@ -699,7 +703,9 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
         // gather present versions
         uint32_t nvers = 0;
         uint32_t iver[GPUC_MAX_ARITY > 16 ? 64 : 64];
-        for (uint32_t m = 0; m < k; m++) {
+        #pragma unroll
+        for (uint32_t m = 0; m < (uint32_t)MA; m++) {
+            if (m >= k) break;
             uint32_t i = src_bases[recs[beg + m].src] + recs[beg + m].idx;
             if (pc.st.flags[i] & PF_HAS_ROW) iver[nvers++] = i;
         }
@@ -910,7 +916,9 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
     while (true) {
         // find min position
         int first = -1;
-        for (uint32_t m = 0; m < k; m++) {
+        #pragma unroll
+        for (uint32_t m = 0; m < (uint32_t)MA; m++) {
+            if (m >= k) break;
             if (mpos[m] >= mcnt[m]) continue;
             uint64_t o = mb[m] + mpos[m];
             if (first < 0) { first = (int)m; continue; }
@@ -927,7 +935,9 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
         uint64_t members = 0;
         uint32_t nmem = 0;
         int lastm = -1;
-        for (uint32_t m = 0; m < k; m++) {
+        #pragma unroll
+        for (uint32_t m = 0; m < (uint32_t)MA; m++) {
+            if (m >= k) break;
             if (mpos[m] >= mcnt[m]) continue;
             uint64_t o = mb[m] + mpos[m];
             if (pos_cmp(in, o, fo, sp) == 0) {
@@ -1002,7 +1012,9 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
                 }
             } else {
                 bool has_live = false;
-                for (uint32_t m = 0; m < k; m++) {
+                #pragma unroll
+        for (uint32_t m = 0; m < (uint32_t)MA; m++) {
+            if (m >= k) break;
                     if (!(members & (1ULL << m))) continue;
                     uint64_t o = mb[m] + mpos[m];
                     uint8_t f = in.flags[o];
@@ -1040,7 +1052,9 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
                     uint64_t va = 0;
                     uint32_t vl = 0;
                     bool have_cell = false, cell_val = false, cell_exp = false;
-                    for (uint32_t m = 0; m < k; m++) {
+                    #pragma unroll
+        for (uint32_t m = 0; m < (uint32_t)MA; m++) {
+            if (m >= k) break;
                         if (!(members & (1ULL << m))) continue;
                         uint64_t o = (mb[m] + mpos[m]) * NC + c;
                         uint8_t f = in.cell_flags[o];
@@ -1096,7 +1110,9 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
                     // decides both the cell filter and whether it is kept
                     // (Row.java:857-874)
                     bool anyv = false;
-                    for (uint32_t m = 0; m < k; m++) {
+                    #pragma unroll
+        for (uint32_t m = 0; m < (uint32_t)MA; m++) {
+            if (m >= k) break;
                         if (!(members & (1ULL << m))) continue;
                         uint64_t o = mb[m] + mpos[m];
                         if (!(in.flags[o] & PF_HAS_CPX)) continue;
@@ -1115,7 +1131,9 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
                         uint32_t ppos[MA], pcnt[MA];
                         uint64_t pbase[MA];
                         uint32_t nv2 = 0, totc = 0;
-                        for (uint32_t m = 0; m < k; m++) {
+                        #pragma unroll
+        for (uint32_t m = 0; m < (uint32_t)MA; m++) {
+            if (m >= k) break;
                             if (!(members & (1ULL << m))) continue;
                             uint64_t o = mb[m] + mpos[m];
                             if (!(in.flags[o] & PF_HAS_CPX)) continue;
@@ -1132,7 +1150,9 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
                         };
                         while (true) {
                             int fi = -1;
-                            for (uint32_t m = 0; m < nv2; m++) {
+                            #pragma unroll
+        for (uint32_t m = 0; m < (uint32_t)MA; m++) {
+            if (m >= nv2) break;
                                 if (ppos[m] >= pcnt[m]) continue;
                                 if (fi < 0 || cmp_path(pbase[m] + ppos[m], pbase[fi] + ppos[fi]) < 0)
                                     fi = (int)m;
@@ -1147,7 +1167,9 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
                             uint64_t va = 0, pa = 0;
                             uint32_t vl = 0, pl = 0;
                             bool cell_val = false, cell_exp = false;
-                            for (uint32_t m = 0; m < nv2; m++) {
+                            #pragma unroll
+        for (uint32_t m = 0; m < (uint32_t)MA; m++) {
+            if (m >= nv2) break;
                                 if (ppos[m] >= pcnt[m]) continue;
                                 uint64_t xe = pbase[m] + ppos[m];
                                 if (cmp_path(xe, fxe) != 0) continue;
@@ -1322,7 +1344,9 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
             int64_t prev_m;
             uint32_t prev_l;
             merged_open(&prev_m, &prev_l);
-            for (uint32_t m = 0; m < k; m++) {
+            #pragma unroll
+        for (uint32_t m = 0; m < (uint32_t)MA; m++) {
+            if (m >= k) break;
                 if (!(members & (1ULL << m))) continue;
                 uint64_t o = mb[m] + mpos[m];
                 uint8_t kd = in.rkind[o];
