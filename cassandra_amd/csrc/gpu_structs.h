@@ -172,6 +172,7 @@ struct OutParts {
     uint64_t* row_base;  // into the output UnfCols arena
     uint32_t* row_count; // merged+purged unfiltereds
     uint8_t* keep;       // 0 = dropped (purged empty / out of shard)
+    uint8_t* merged_k;   // versions merged per group (histogram fed by k_merged_hist)
 };
 
 // EncodingStats bases of the OUTPUT header (SerializationHeader.make)

@@ -736,16 +736,19 @@ struct OutPartsBuf {
     DevBuf keypfx, key_addr, token, klen, pdel_mfda, pdel_ldt, row_base, row_count, keep;
     StaticColsBuf stb;
     OutParts op{};
+    DevBuf merged_k;
     void alloc(uint64_t n, uint32_t n_static = 0) {
         keypfx.alloc(n * 8); key_addr.alloc(n * 8); token.alloc(n * 8); klen.alloc(n * 2);
         pdel_mfda.alloc(n * 8); pdel_ldt.alloc(n * 4);
         row_base.alloc(n * 8); row_count.alloc(n * 4); keep.alloc(n);
+        merged_k.alloc(n);
+        HIP_CHECK(hipMemset(merged_k.p, 0, n));
         stb.alloc(n_static ? n : 1, n_static);
         op = OutParts{keypfx.as<uint64_t>(), key_addr.as<uint64_t>(), token.as<int64_t>(),
                       stb.st, klen.as<uint16_t>(),
                       pdel_mfda.as<int64_t>(),
                       pdel_ldt.as<uint32_t>(), row_base.as<uint64_t>(), row_count.as<uint32_t>(),
-                      keep.as<uint8_t>()};
+                      keep.as<uint8_t>(), merged_k.as<uint8_t>()};
     }
 };
 

@@ -510,6 +510,18 @@ __global__ void k_key_hash2(OutParts op, uint64_t n, uint64_t* out) {
     out[g] = op.keep[g] ? murmur2_64((const uint8_t*)op.key_addr[g], op.klen[g]) : 0;
 }
 
+// block-reduced merged-arity histogram (fed from op.merged_k)
+__global__ void k_merged_hist(const uint8_t* merged_k, uint64_t n, OutStats* st) {
+    __shared__ unsigned int sh[64];
+    for (int i = threadIdx.x; i < 64; i += blockDim.x) sh[i] = 0;
+    __syncthreads();
+    uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (g < n && merged_k[g]) atomicAdd(&sh[merged_k[g] - 1], 1u);
+    __syncthreads();
+    for (int i = threadIdx.x; i < 64; i += blockDim.x)
+        if (sh[i]) atomicAdd(&st->merged_counts[i], (unsigned long long)sh[i]);
+}
+
 __global__ void k_widen_u32(const uint32_t* in, uint64_t* out, uint64_t n) {
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i < n) out[i] = in[i];
@@ -636,7 +648,10 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
     uint64_t beg = group_start[g];
     uint64_t endi = g + 1 < n_groups ? group_start[g + 1] : n_recs;
     uint32_t k = (uint32_t)(endi - beg);
-    atomicAdd(&st->merged_counts[k > 64 ? 63 : k - 1], 1ull);
+    // merged-arity histogram: a direct atomicAdd here serializes millions of
+    // threads on one address (~150 ms at C2's 14.9M single-row groups);
+    // store the arity and let k_merged_hist block-reduce it instead
+    op.merged_k[g] = (uint8_t)(k > 64 ? 64 : k);
     if (k > (uint32_t)MA) { atomicExch(error, 20ull); return; }
 
     const MRec r0 = recs[beg];
@@ -914,19 +929,18 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
     auto purge_dt = [&](int64_t m, uint32_t l) { return m != INT64_MIN && should_purge2(pp, token, m, ldt_long(l)); };
 
     while (true) {
-        // find min position
+        // find min position; track the winner's VALUE so the MA-sized
+        // stream arrays never see a runtime index (keeps them in registers)
         int first = -1;
+        uint64_t fo = 0;
         #pragma unroll
         for (uint32_t m = 0; m < (uint32_t)MA; m++) {
             if (m >= k) break;
             if (mpos[m] >= mcnt[m]) continue;
             uint64_t o = mb[m] + mpos[m];
-            if (first < 0) { first = (int)m; continue; }
-            uint64_t f = mb[first] + mpos[first];
-            if (pos_cmp(in, o, f, sp) < 0) first = (int)m;
+            if (first < 0 || pos_cmp(in, o, fo, sp) < 0) { first = (int)m; fo = o; }
         }
         if (first < 0) break;
-        uint64_t fo = mb[first] + mpos[first];
         uint64_t fck = in.ck[fo * (sp.n_ck ? sp.n_ck : 1)];
         cur_fo = fo;
         uint8_t fkind = in.rkind[fo];
@@ -935,6 +949,7 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
         uint64_t members = 0;
         uint32_t nmem = 0;
         int lastm = -1;
+        uint64_t last_o = 0;
         #pragma unroll
         for (uint32_t m = 0; m < (uint32_t)MA; m++) {
             if (m >= k) break;
@@ -944,6 +959,7 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
                 members |= 1ULL << m;
                 nmem++;
                 lastm = (int)m;
+                last_o = o;
             }
         }
 
@@ -976,7 +992,7 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
             uint64_t xstart = 0;
             uint32_t xcnt = 0;
             if (k == 1 || (nmem == 1 && active_live)) {
-                uint64_t o = mb[lastm] + mpos[lastm];
+                uint64_t o = last_o;
                 of = in.flags[o];
                 lts = in.live_ts[o]; lttl = in.live_ttl[o]; llet = in.live_let[o];
                 rdm = in.rdel_mfda[o]; rdl = in.rdel_ldt[o];
@@ -1150,15 +1166,15 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
                         };
                         while (true) {
                             int fi = -1;
+                            uint64_t fxe = 0;
                             #pragma unroll
         for (uint32_t m = 0; m < (uint32_t)MA; m++) {
             if (m >= nv2) break;
                                 if (ppos[m] >= pcnt[m]) continue;
-                                if (fi < 0 || cmp_path(pbase[m] + ppos[m], pbase[fi] + ppos[fi]) < 0)
-                                    fi = (int)m;
+                                uint64_t xe0 = pbase[m] + ppos[m];
+                                if (fi < 0 || cmp_path(xe0, fxe) < 0) { fi = (int)m; fxe = xe0; }
                             }
                             if (fi < 0) break;
-                            uint64_t fxe = pbase[fi] + ppos[fi];
                             // CellReducer over same-path cells in version order
                             bool have = false;
                             int64_t cts = NO_TIMESTAMP;
