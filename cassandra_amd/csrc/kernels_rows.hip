@@ -284,6 +284,7 @@ __global__ void k_parse_rows(const SrcDesc2* srcs, uint32_t n_srcs, uint32_t tot
     uint32_t emitted = 0;
     uint64_t cpx_cur = (sp.n_cpx && cpx_base) ? cpx_base[gi] : 0;
     uint32_t cpx_seen = 0;
+    unsigned long long rows_local = 0;
     while (true) {
         uint8_t flags = base[pos++];
         if (flags & 0x01) break;
@@ -465,10 +466,11 @@ __global__ void k_parse_rows(const SrcDesc2* srcs, uint32_t n_srcs, uint32_t tot
             rc.rdel_ldt[o] = rdl;
             rc.start_mfda[o] = INT64_MIN;
             rc.start_ldt[o] = LDT_NONE_U32;
-            atomicAdd(rows_in, 1ull);
+            rows_local++;
         }
         emitted++;
     }
+    if (rows_local) atomicAdd(rows_in, rows_local);  // one atomic per partition
     if (sp.n_cpx && pc.cpx_total) pc.cpx_total[gi] = cpx_seen;
     (void)error;
 }
