@@ -886,16 +886,49 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
     d_out_index.alloc(total_idx);
     d_bloom.alloc(words * 8);
     HIP_CHECK(hipMemsetAsync(d_bloom.p, 0, words * 8, stream));
+    // serialize in byte-balanced SEGMENTS on a side stream so the compress
+    // slabs below start as soon as their byte range is fully written (the
+    // output bytes are offset-addressed, so the split cannot change them)
+    hipStream_t sstream;
+    HIP_CHECK(hipStreamCreate(&sstream));
+    const int NSEG = n_groups >= 4096 && total_unc > (512ull << 20) ? 8 : 1;
+    std::vector<uint64_t> seg_g(NSEG + 1), seg_end_byte(NSEG);
+    if (NSEG > 1) {
+        // d_psize holds the exscanned per-group data offsets
+        std::vector<uint64_t> h_off(n_groups);
+        HIP_CHECK(hipMemcpy(h_off.data(), d_psize.p, n_groups * 8, hipMemcpyDeviceToHost));
+        seg_g[0] = 0;
+        for (int j = 1; j < NSEG; j++) {
+            uint64_t target = total_unc / NSEG * j;
+            seg_g[j] = std::lower_bound(h_off.begin(), h_off.end(), target) - h_off.begin();
+        }
+        seg_g[NSEG] = n_groups;
+        for (int j = 0; j < NSEG; j++)
+            seg_end_byte[j] = j + 1 < NSEG ? h_off[seg_g[j + 1]] : total_unc;
+    } else {
+        seg_g[0] = 0;
+        seg_g[NSEG] = n_groups;
+        seg_end_byte[NSEG - 1] = total_unc;
+    }
+    std::vector<hipEvent_t> ev_seg(NSEG);
     {
         uint32_t waves_per_block = 4;
-        uint64_t blocks = (n_groups + waves_per_block - 1) / waves_per_block;
-        hipLaunchKernelGGL(k_serialize_rows, dim3((uint32_t)blocks), dim3(WAVE * waves_per_block),
-                           0, stream, opb.op, rows.uc, n_groups, sp, d_psize.as<uint64_t>(),
-                           d_isize.as<uint64_t>(), d_nblocks.as<uint32_t>(), d_infsz.as<uint64_t>(),
-                           d_out_data.as<uint8_t>(), d_out_index.as<uint8_t>(),
-                           d_bloom.as<uint32_t>(), words * 64, bs.k);
+        HIP_CHECK(hipStreamWaitEvent(sstream, ev1, 0));
+        for (int j = 0; j < NSEG; j++) {
+            uint64_t gs = seg_g[j], ge = seg_g[j + 1];
+            uint64_t blocks = (std::max<uint64_t>(ge - gs, 1) + waves_per_block - 1) / waves_per_block;
+            if (ge > gs)
+                hipLaunchKernelGGL(k_serialize_rows, dim3((uint32_t)blocks),
+                                   dim3(WAVE * waves_per_block), 0, sstream, opb.op, rows.uc, ge,
+                                   sp, d_psize.as<uint64_t>(), d_isize.as<uint64_t>(),
+                                   d_nblocks.as<uint32_t>(), d_infsz.as<uint64_t>(),
+                                   d_out_data.as<uint8_t>(), d_out_index.as<uint8_t>(),
+                                   d_bloom.as<uint32_t>(), words * 64, bs.k, gs);
+            HIP_CHECK(hipEventCreate(&ev_seg[j]));
+            HIP_CHECK(hipEventRecord(ev_seg[j], sstream));
+        }
     }
-    HIP_CHECK(hipEventRecord(ev2, stream));
+    HIP_CHECK(hipEventRecord(ev2, sstream));
 
     // ---- compress + gather + D2H + write: slab-pipelined ----
     // Compress launches for all slabs are enqueued on `stream` back to back;
@@ -921,8 +954,13 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
     HIP_CHECK(hipMemcpyAsync(h_index, d_out_index.p, total_idx, hipMemcpyDeviceToHost, cstream));
 
     std::vector<hipEvent_t> ev_c(n_slabs);
+    int next_seg = 0;
     for (uint32_t i = 0; i < n_slabs; i++) {
         uint32_t cb = i * SLAB, m = std::min(SLAB, n_chunks - cb);
+        // gate this slab on the serialize segment that completes its bytes
+        uint64_t slab_end = std::min<uint64_t>((uint64_t)(cb + m) * CHUNK_LEN, total_unc);
+        while (next_seg < NSEG && seg_end_byte[next_seg] < slab_end) next_seg++;
+        HIP_CHECK(hipStreamWaitEvent(stream, next_seg < NSEG ? ev_seg[next_seg] : ev2, 0));
         // all kernel arguments shift uniformly per chunk, so a slab launch is
         // just base-offset pointers with a local chunk count
         if (snappy_out)
@@ -1199,6 +1237,9 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
         w.ms_d2h = t_drain1 - t_drain0;  // slab drain wall (overlaps compress)
     }
     for (uint32_t i = 0; i < n_slabs; i++) (void)hipEventDestroy(ev_c[i]);
+    HIP_CHECK(hipStreamSynchronize(sstream));
+    for (int j = 0; j < NSEG; j++) (void)hipEventDestroy(ev_seg[j]);
+    HIP_CHECK(hipStreamDestroy(sstream));
     HIP_CHECK(hipStreamDestroy(cstream));
     w.uncompressed_len = total_unc;
     float t01, t12, t23;

@@ -2118,8 +2118,9 @@ __global__ void k_serialize_rows(OutParts op, UnfCols out, uint64_t n, SerParams
                                  const uint64_t* data_off, const uint64_t* idx_off,
                                  const uint32_t* nblocks, const uint64_t* infos_size,
                                  uint8_t* out_data, uint8_t* out_index,
-                                 uint32_t* bloom_bits, uint64_t bloom_bitlen, int32_t bloom_k) {
-    uint64_t g = blockIdx.x * (blockDim.x / WAVE) + (threadIdx.x / WAVE);
+                                 uint32_t* bloom_bits, uint64_t bloom_bitlen, int32_t bloom_k,
+                                 uint64_t g0 = 0) {
+    uint64_t g = g0 + blockIdx.x * (blockDim.x / WAVE) + (threadIdx.x / WAVE);
     if (g >= n || !op.keep[g]) return;
     int lane = threadIdx.x & (WAVE - 1);
     part_walk<true>(op, out, g, sp, data_off[g], idx_off[g], infos_size[g], nblocks[g],
