@@ -891,7 +891,10 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
     // output bytes are offset-addressed, so the split cannot change them)
     hipStream_t sstream;
     HIP_CHECK(hipStreamCreate(&sstream));
-    const int NSEG = n_groups >= 4096 && total_unc > (512ull << 20) ? 8 : 1;
+    int NSEG = 1;  // >1 overlaps serialize under compress; no win measured on
+                   // write-wall-bound boxes, kept behind GPUC_NSEG for fast ones
+    if (const char* e = getenv("GPUC_NSEG")) NSEG = std::max(1, std::min(64, atoi(e)));
+    if (n_groups < 4096) NSEG = 1;
     std::vector<uint64_t> seg_g(NSEG + 1), seg_end_byte(NSEG);
     if (NSEG > 1) {
         // d_psize holds the exscanned per-group data offsets
