@@ -686,6 +686,16 @@ uint64_t append_bti_row_index(bytes& file, const BtiRowIndexBlockSpec& spec) {
     return index_pos;
 }
 
+// ByteOrderedPartitioner keys: single escaped component (the fixture's
+// legacy_da tables; prefixes `40 <key bytes>` pin this form)
+bytes bti_byte_comparable_bop(const bytes& key) {
+    bytes out;
+    out.push_back(0x40);  // NEXT_COMPONENT
+    bc_escape(out, key);
+    out.push_back(0x38);  // TERMINATOR
+    return out;
+}
+
 bytes bti_byte_comparable_m3(int64_t token, const bytes& key) {
     bytes out;
     out.push_back(0x40);  // NEXT_COMPONENT

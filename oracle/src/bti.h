@@ -111,4 +111,7 @@ bytes bti_nudge(const bytes& value, size_t nudge_at);
 // FE-at-end], non-zero tail -> trailing 0x00)
 bytes bti_byte_comparable_m3(int64_t token, const bytes& key);
 
+// ByteOrderedPartitioner form: single escaped key component (legacy_da fixtures)
+bytes bti_byte_comparable_bop(const bytes& key);
+
 }  // namespace oracle

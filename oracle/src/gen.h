@@ -26,6 +26,7 @@ struct GenSpec {
     int64_t base_ldt = 1700000000LL;        // seconds
     uint64_t first_generation = 1;
     uint32_t snappy = 0;             // 1: SnappyCompressor chunks (C3 shape)
+    uint32_t bti = 0;                // 1: write the `da` (trie-indexed) component set
     // wide-partition (C4) shape: clustering_rows > 0 makes each partition hold
     // clustering_rows rows keyed by a LongType clustering column; tombstone_pct
     // then applies per ROW, and range_tomb_pct partitions get one range

@@ -48,13 +48,24 @@ static int cmd_roundtrip(const std::string& base) {
     // (test/conf/cassandra.yaml:24). Accept either.
     bytes data, index;
     for (auto& p : t.parts) serialize_partition(p, t.header, data, index, COLUMN_INDEX_SIZE);
-    {
+    if (!t.bti) {
         bytes want_idx = read_file(base + "-Index.db");
         if (index != want_idx) {
             bytes d2, i2;
             for (auto& p : t.parts) serialize_partition(p, t.header, d2, i2, 4096);
             if (i2 == want_idx) index = std::move(i2);
         }
+    } else {
+        // column_index_size is config, not recorded: try the 64 KiB default,
+        // fall back to the fixtures' 4 KiB test setting on mismatch
+        bytes want_rows = read_file(base + "-Rows.db");
+        SSTable probe;
+        probe.header = t.header;
+        probe.comp = t.comp;
+        probe.parts = t.parts;
+        probe.partitioner = t.partitioner;
+        probe.column_index_size = COLUMN_INDEX_SIZE;
+        if (write_sstable(probe, true).rows_db != want_rows) t.column_index_size = 4096;
     }
     int rc = 0;
     auto check = [&](const char* what, const bytes& got, const bytes& want) {
@@ -70,7 +81,7 @@ static int cmd_roundtrip(const std::string& base) {
     check("Data.db", co.file, read_file(base + "-Data.db"));
     check("CompressionInfo.db", make_compression_info(t.comp, data.size(), co.offsets),
           read_file(base + "-CompressionInfo.db"));
-    check("Index.db", index, read_file(base + "-Index.db"));
+    if (!t.bti) check("Index.db", index, read_file(base + "-Index.db"));
     if (getenv("ORACLE_DUMP_INDEX")) write_file(std::string(getenv("ORACLE_DUMP_INDEX")), index);
     {
         std::string d = std::to_string(crc32(co.file.data(), co.file.size()));
@@ -83,8 +94,14 @@ static int cmd_roundtrip(const std::string& base) {
         tmp.header = t.header;
         tmp.comp = t.comp;
         tmp.parts = t.parts;
-        WriterOut w = write_sstable(tmp);
+        tmp.partitioner = t.partitioner;
+        tmp.column_index_size = t.column_index_size;
+        WriterOut w = write_sstable(tmp, t.bti);
         check("Filter.db", w.filter, read_file(base + "-Filter.db"));
+        if (t.bti) {
+            check("Partitions.db", w.partitions_db, read_file(base + "-Partitions.db"));
+            check("Rows.db", w.rows_db, read_file(base + "-Rows.db"));
+        }
         auto comp_slice = [](const bytes& st) -> bytes {
             if (st.size() < 4) return {};
             Reader r(st);
@@ -109,7 +126,8 @@ static int cmd_roundtrip(const std::string& base) {
 // writer-path inspection without compaction semantics)
 static int cmd_rewrite(const std::string& inbase, const std::string& outbase) {
     SSTable t = read_sstable(inbase, true);
-    WriterOut w = write_sstable(t);
+    if (const char* e = getenv("ORACLE_CIS")) t.column_index_size = (uint32_t)atoi(e);
+    WriterOut w = write_sstable(t, t.bti);
     write_components(w, outbase);
     printf("rewrote %zu partitions\n", t.parts.size());
     return 0;
@@ -141,6 +159,7 @@ static GenSpec spec_from_kv(std::map<std::string, std::string>& kv) {
     g.first_generation = geti("gen0", g.first_generation);
     g.snappy = (uint32_t)geti("snappy", g.snappy);
     g.ttl_pct = (uint32_t)geti("ttl", g.ttl_pct);
+    g.bti = (uint32_t)geti("bti", g.bti);
     g.complex_pct = (uint32_t)geti("cpx", g.complex_pct);
     g.complex_del_pct = (uint32_t)geti("cpxdel", g.complex_del_pct);
     return g;
@@ -151,8 +170,9 @@ static int cmd_gen(const std::string& outdir, std::map<std::string, std::string>
     uint64_t total_unc = 0;
     for (uint32_t s = 0; s < g.n_sstables; s++) {
         SSTable t = generate_sstable(g, s);
-        WriterOut w = write_sstable(t);
-        std::string base = outdir + "/oa-" + std::to_string(t.generation) + "-big";
+        WriterOut w = write_sstable(t, g.bti != 0);
+        std::string base = g.bti ? outdir + "/da-" + std::to_string(t.generation) + "-bti"
+                                 : outdir + "/oa-" + std::to_string(t.generation) + "-big";
         write_components(w, base);
         total_unc += w.uncompressed_data_len;
         printf("wrote %s: parts=%llu uncompressed=%llu compressed=%zu\n", base.c_str(),
@@ -306,7 +326,11 @@ static int cmd_compact(const std::string& outbase, std::vector<std::string>& inp
     auto t1 = clk::now();
     CompactionResult res = compact(job);
     auto t2 = clk::now();
-    WriterOut w = write_sstable(res.out);
+    // output format follows the inputs (a da compaction emits da)
+    bool out_bti = !job.inputs.empty() && job.inputs[0].bti;
+    if (const char* e = getenv("ORACLE_CIS"))
+        res.out.column_index_size = (uint32_t)atoi(e);
+    WriterOut w = write_sstable(res.out, out_bti);
     write_components(w, outbase);
     auto t3 = clk::now();
     double read_s = std::chrono::duration<double>(t1 - t0).count();

@@ -2,6 +2,7 @@
 // Reader/writer for Cassandra big-format `oa` sstables; bit-exact restatement
 // of the serializers cited per function.
 #include "sstable.h"
+#include "bti.h"
 #include <set>
 #include "lz4_ref.h"
 #include "snappy_ref.h"
@@ -332,7 +333,13 @@ static void put_index_info(bytes& out, const IndexInfoC& ii, const Header& h) {
     if (ii.has_open) put_deletion_time(out, ii.end_open_marker);
 }
 
+void serialize_partition_ex(const Partition& p, const Header& h, bytes& out, bytes& index_out,
+                            uint32_t column_index_size, std::vector<IndexInfoC>* out_blocks);
 void serialize_partition(const Partition& p, const Header& h, bytes& out, bytes& index_out, uint32_t column_index_size) {
+    serialize_partition_ex(p, h, out, index_out, column_index_size, nullptr);
+}
+void serialize_partition_ex(const Partition& p, const Header& h, bytes& out, bytes& index_out,
+                            uint32_t column_index_size, std::vector<IndexInfoC>* out_blocks) {
     size_t initial = out.size();
     put_short_len_bytes(out, p.key);
     put_deletion_time(out, p.del);
@@ -398,6 +405,7 @@ void serialize_partition(const Partition& p, const Header& h, bytes& out, bytes&
     }
     // NOTE on has_open of a CLOSED block: IndexInfo.endOpenMarker is the open marker
     // state at the END of the block (BigFormatPartitionWriter.addIndexBlock:128-134).
+    if (out_blocks) *out_blocks = blocks;
 
     // Index.db entry (BigTableWriter.IndexWriter.append:266-279)
     put_short_len_bytes(index_out, p.key);
@@ -757,8 +765,96 @@ static bytes serialize_statistics(const Header& h, const StatsComponentInput& s)
 // ---------------------------------------------------------------------------
 // whole-sstable writer
 // ---------------------------------------------------------------------------
-WriterOut write_sstable(const SSTable& t) {
+// ---------------------------------------------------------------------------
+// BTI (`da`) index assembly: Partitions.db + Rows.db from the SAME
+// column_index_size blocks as the big promoted index (rowIndexBlockSize ==
+// column_index_size, BtiFormatPartitionWriter.java:52). Row-index entry b
+// carries (block b's start offset within the partition, the deletion open at
+// its start == block b-1's end-open); a partition gets a row index only when
+// it has >1 blocks (BtiFormatPartitionWriter.finish:100-109); the final
+// entry is nudge(prevMax, |common(prevMax, prevSep)|) with payload
+// (partition length - 1 == the END_OF_PARTITION byte's offset, LIVE) —
+// pinned against the legacy_da fixtures (cmd_roundtrip Stats/Partitions/
+// Rows comparisons).
+static void build_bti_index(const SSTable& t, const std::vector<uint64_t>& part_pos,
+                            const std::vector<std::vector<IndexInfoC>>& part_blocks,
+                            uint64_t data_len, WriterOut& w) {
+    bytes rows;
+    std::vector<BtiKeyEntry> pes;
+    auto enc = [&](BoundKind k, const Clustering& c) {
+        return bti_byte_comparable_clustering(c, t.header.clustering_types, k);
+    };
+    auto payload = [](uint64_t off, bool has_open, const DeletionTime& od, int* pb, bytes* pay) {
+        // SizedInts.nonZeroSize: SIGNED size — the leading bit stays clear
+        int ob = 1;
+        while (off >> (8 * ob - 1)) ob++;
+        *pb = ob | (has_open ? 8 : 0);
+        for (int b = ob - 1; b >= 0; b--) pay->push_back((uint8_t)(off >> (8 * b)));
+        if (has_open) {
+            if (od.live()) {
+                pay->push_back(0x80);
+            } else {
+                for (int b = 7; b >= 0; b--) pay->push_back((uint8_t)((uint64_t)od.mfda >> (8 * b)));
+                for (int b = 3; b >= 0; b--) pay->push_back((uint8_t)(od.ldt >> (8 * b)));
+            }
+        }
+    };
+    for (size_t i = 0; i < t.parts.size(); i++) {
+        const Partition& p = t.parts[i];
+        uint64_t part_end = (i + 1 < t.parts.size() ? part_pos[i + 1] : data_len) - part_pos[i];
+        const auto& blocks = part_blocks[i];
+        int64_t idxpos;
+        if (blocks.size() > 1) {
+            BtiRowIndexBlockSpec spec;
+            spec.partition_key = p.key;
+            spec.data_pos = part_pos[i];
+            spec.block_count = blocks.size();
+            spec.partition_del = p.del;
+            bytes prev_sep, prev_max;
+            for (size_t b = 0; b < blocks.size(); b++) {
+                bytes key;
+                if (b > 0)
+                    key = bti_separator_gt(enc(blocks[b - 1].last_kind, blocks[b - 1].last),
+                                           enc(blocks[b].first_kind, blocks[b].first));
+                bool has_od = b > 0 && blocks[b - 1].has_open;
+                DeletionTime od = has_od ? blocks[b - 1].end_open_marker : DT_LIVE;
+                int pb;
+                bytes pay;
+                payload(blocks[b].offset, has_od, od, &pb, &pay);
+                spec.entries.push_back({key, pb, pay});
+                prev_sep = key;
+                prev_max = enc(blocks[b].last_kind, blocks[b].last);
+            }
+            size_t cm = 0;
+            while (cm < prev_max.size() && cm < prev_sep.size() && prev_max[cm] == prev_sep[cm]) cm++;
+            int pb;
+            bytes pay;
+            payload(part_end - 1, false, DT_LIVE, &pb, &pay);
+            spec.entries.push_back({bti_nudge(prev_max, cm), pb, pay});
+            idxpos = (int64_t)append_bti_row_index(rows, spec);
+        } else {
+            idxpos = ~(int64_t)part_pos[i];
+        }
+        BtiKeyEntry e;
+        e.byte_comparable = t.partitioner == Partitioner::MURMUR3
+                                ? bti_byte_comparable_m3(p.token, p.key)
+                                : bti_byte_comparable_bop(p.key);
+        e.raw_key = p.key;
+        uint64_t h2[2];
+        murmur3_128_cassandra(p.key.data(), p.key.size(), 0, h2);
+        e.hash_bits = (uint8_t)h2[1];  // DecoratedKey.filterHashLowerBits (IFilter.java:34-40)
+        e.idxpos = idxpos;
+        pes.push_back(e);
+    }
+    w.rows_db = std::move(rows);
+    w.partitions_db = write_bti_partitions(pes);
+}
+
+WriterOut write_sstable(const SSTable& t, bool bti) {
     WriterOut w;
+    w.bti = bti;
+    std::vector<uint64_t> part_pos;
+    std::vector<std::vector<IndexInfoC>> part_blocks;
     bytes data_raw;
     Bloom bloom = make_bloom(t.parts.size(), 0.01);
     StatsComponentInput st;
@@ -767,7 +863,10 @@ WriterOut write_sstable(const SSTable& t) {
     bool has_partition_deletions = false;
     for (const Partition& p : t.parts) {
         size_t before = data_raw.size();
-        serialize_partition(p, t.header, data_raw, w.index_db, t.column_index_size);
+        part_pos.push_back(before);
+        part_blocks.emplace_back();
+        serialize_partition_ex(p, t.header, data_raw, w.index_db, t.column_index_size,
+                               bti ? &part_blocks.back() : nullptr);
         bloom.add(p.key);
         st.partition_size.add(data_raw.size() - before);
         uint64_t cells = 0;
@@ -856,9 +955,11 @@ WriterOut write_sstable(const SSTable& t) {
         put_be32(s, 0);                         // entries (unsampled placeholder)
         put_be64(s, 8);                         // offheap size of entries section
         put_be32(s, 128); put_be32(s, 0);       // samplingLevel? sizeAtFullSampling?
-        std::string toc = "Data.db\nStatistics.db\nDigest.crc32\nTOC.txt\nCompressionInfo.db\nFilter.db\nIndex.db\nSummary.db\n";
+        std::string toc = w.bti ? "Data.db\nStatistics.db\nDigest.crc32\nTOC.txt\nCompressionInfo.db\nFilter.db\nPartitions.db\nRows.db\n"
+                                : "Data.db\nStatistics.db\nDigest.crc32\nTOC.txt\nCompressionInfo.db\nFilter.db\nIndex.db\nSummary.db\n";
         w.toc.assign(toc.begin(), toc.end());
     }
+    if (bti) build_bti_index(t, part_pos, part_blocks, data_raw.size(), w);
     w.uncompressed_data_len = data_raw.size();
     w.partition_count = t.parts.size();
     return w;
@@ -866,12 +967,17 @@ WriterOut write_sstable(const SSTable& t) {
 
 void write_components(const WriterOut& w, const std::string& base) {
     write_file(base + "-Data.db", w.data_db);
-    write_file(base + "-Index.db", w.index_db);
+    if (w.bti) {
+        write_file(base + "-Partitions.db", w.partitions_db);
+        write_file(base + "-Rows.db", w.rows_db);
+    } else {
+        write_file(base + "-Index.db", w.index_db);
+        write_file(base + "-Summary.db", w.summary);
+    }
     write_file(base + "-CompressionInfo.db", w.compression_info);
     write_file(base + "-Filter.db", w.filter);
     write_file(base + "-Digest.crc32", w.digest);
     write_file(base + "-Statistics.db", w.statistics);
-    write_file(base + "-Summary.db", w.summary);
     write_file(base + "-TOC.txt", w.toc);
 }
 
@@ -1133,8 +1239,15 @@ Partition read_one_partition(const bytes& raw, uint64_t offset, const Header& h,
     return read_partition(r, h, item_offsets);
 }
 
+static bool file_exists_(const std::string& p2) {
+    FILE* f = fopen(p2.c_str(), "rb");
+    if (f) fclose(f);
+    return f != nullptr;
+}
+
 SSTable read_sstable(const std::string& base, bool keep_raw) {
     SSTable t;
+    t.bti = file_exists_(base + "-Partitions.db") && !file_exists_(base + "-Index.db");
     bytes ci_b = read_file(base + "-CompressionInfo.db");
     bytes data_b = read_file(base + "-Data.db");
     bytes stats_b = read_file(base + "-Statistics.db");

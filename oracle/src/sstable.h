@@ -241,6 +241,7 @@ struct SSTable {
     // raw images kept by the reader for byte-level comparisons in tests
     bytes raw_data_uncompressed;
     bytes raw_statistics;   // whole Statistics.db as read
+    bool bti = false;       // `da` (Big Trie-Indexed) component set
 };
 
 // ---- reader ----
@@ -262,17 +263,24 @@ Partition read_one_partition(const bytes& raw, uint64_t offset, const Header& h,
 // ---- writer ----
 struct WriterOut {
     bytes data_db, index_db, compression_info, filter, digest, statistics, summary, toc;
+    bytes partitions_db, rows_db;  // BTI (`da`) index components
+    bool bti = false;
     uint64_t uncompressed_data_len = 0;
     uint64_t partition_count = 0;
 };
 // Serializes with sstable.header (stats deltas), comp params; bloom fp 0.01.
-WriterOut write_sstable(const SSTable& t);
+// bti=true additionally assembles Partitions.db/Rows.db (BtiFormat.md) from
+// the SAME column_index_size blocks the big promoted index uses
+// (BtiFormatPartitionWriter.java:52 — rowIndexBlockSize == column_index_size)
+// and write_components emits the `da` component set instead of Index/Summary.
+WriterOut write_sstable(const SSTable& t, bool bti = false);
 void write_components(const WriterOut& w, const std::string& base);
 
 // serialize one partition into `out` (Data.db stream) and append its Index.db
 // entry to `index_out`. Exposed for round-trip tests.
 void serialize_partition(const Partition& p, const Header& h, bytes& out, bytes& index_out,
                          uint32_t column_index_size = COLUMN_INDEX_SIZE);
+struct IndexInfoC;  // promoted-index block (defined in sstable.cpp)
 
 // LZ4 chunk framing: compress `raw` into chunks (Data.db bytes), offsets, etc.
 struct ChunkedOut { bytes file; std::vector<uint64_t> offsets; };
