@@ -105,6 +105,7 @@ class GpucGenSpec(ctypes.Structure):
         ("ttl_pct", ctypes.c_uint32),
         ("complex_pct", ctypes.c_uint32),
         ("complex_del_pct", ctypes.c_uint32),
+        ("bti", ctypes.c_int32),
     ]
 
 
@@ -367,6 +368,7 @@ def generate(
     ttl_pct=0,
     complex_pct=0,
     complex_del_pct=0,
+    bti=False,
     base_ts=1700000000000000,
     base_ldt=1700000000,
     first_generation=1,
@@ -399,6 +401,7 @@ def generate(
         ttl_pct=ttl_pct,
         complex_pct=complex_pct,
         complex_del_pct=complex_del_pct,
+        bti=1 if bti else 0,
     )
     err = ctypes.create_string_buffer(256)
     rc = lib.gpuc_generate(ctypes.byref(spec), out_dir.encode(), err, 256)

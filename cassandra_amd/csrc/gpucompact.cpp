@@ -7,6 +7,7 @@
 // fallback — gpuc_compact fails with GPUC_ERR_NO_GPU when no device exists.
 #include "kernels.hip"
 #include "kernels_rows.hip"
+#include "bti_core.h"
 #include "../../include/gpucompact.h"
 
 #include <algorithm>
@@ -811,6 +812,183 @@ struct WriteDeviceOut {
 // shared device->files writer: sizes/scan -> serialize -> compress -> gather ->
 // D2H -> write all components. `st` is the device OutStats already filled by
 // reconcile (or gen); meta_* give header/schema info.
+// BTI (`da`) index post-pass: Partitions.db + Rows.db built on the host
+// from the big-format Index.db IMAGE the writer kernels already produced
+// (the image carries keys, data positions and the promoted-index blocks —
+// firstName/lastName clusterings, offsets, end-open markers — and those
+// blocks ARE the bti row-index blocks: rowIndexBlockSize ==
+// column_index_size, BtiFormatPartitionWriter.java:52). Byte layout pinned
+// against the reference's legacy_da fixtures through the oracle
+// (write_sstable(bti)) and the GPU parity suite.
+static void build_bti_from_index_image(const uint8_t* idx, uint64_t idx_len, uint64_t data_total,
+                                       const std::vector<int32_t>& ck_w,
+                                       bti::bytes* partitions_out, bti::bytes* rows_out) {
+    using bti::bytes;
+    uint64_t p = 0;
+    bytes rows_file;
+    std::vector<bti::BtiKeyEntry> pes;
+    std::vector<bti::BtiRowIndexBlockSpec> pending_specs;
+    std::vector<size_t> pending_idx;
+    std::vector<uint64_t> part_positions;
+    auto uvint = [&]() -> uint64_t {
+        uint8_t first = idx[p++];
+        int extra = 0;
+        uint8_t x = first;
+        while (x & 0x80) { extra++; x <<= 1; }
+        uint64_t r = first & (uint8_t)(0xFFu >> extra);
+        for (int i = 0; i < extra; i++) r = (r << 8) | idx[p++];
+        return r;
+    };
+    auto svint = [&]() -> int64_t {
+        uint64_t u = uvint();
+        return (int64_t)(u >> 1) ^ -(int64_t)(u & 1);
+    };
+    struct Blk {
+        bytes first_bc, last_bc;
+        uint64_t offset = 0;
+        bool has_open = false;
+        int64_t open_m = 0;
+        uint32_t open_l = 0;
+    };
+    auto parse_prefix = [&]() -> bytes {
+        uint8_t kind = idx[p++];
+        uint32_t nv;
+        if (kind != 4) {
+            nv = ((uint32_t)idx[p] << 8) | idx[p + 1];
+            p += 2;
+        } else {
+            nv = (uint32_t)ck_w.size();
+        }
+        std::vector<std::pair<const uint8_t*, uint32_t>> comps;
+        std::vector<int32_t> widths;
+        if (nv) {
+            (void)uvint();  // 32-batch null header (0: all present)
+            for (uint32_t c = 0; c < nv; c++) {
+                uint32_t len = ck_w[c] > 0 ? (uint32_t)ck_w[c] : (uint32_t)uvint();
+                comps.push_back({idx + p, len});
+                widths.push_back(ck_w[c]);
+                p += len;
+            }
+        }
+        return bti::byte_comparable_clustering(comps, widths, kind);
+    };
+    while (p < idx_len) {
+        uint32_t klen = ((uint32_t)idx[p] << 8) | idx[p + 1];
+        p += 2;
+        const uint8_t* key = idx + p;
+        p += klen;
+        uint64_t pos = uvint();
+        uint64_t promoted = uvint();
+        int64_t idxpos = ~(int64_t)pos;
+        if (promoted > 0) {
+            uint64_t pend = p + promoted;
+            (void)uvint();  // headerLength
+            bool pdel_live = idx[p] == 0x80;
+            int64_t pdm = 0;
+            uint32_t pdl = 0;
+            if (pdel_live) {
+                p += 1;
+            } else {
+                uint64_t m2 = 0;
+                for (int i = 0; i < 8; i++) m2 = (m2 << 8) | idx[p + i];
+                pdm = (int64_t)m2;
+                pdl = ((uint32_t)idx[p + 8] << 24) | ((uint32_t)idx[p + 9] << 16) |
+                      ((uint32_t)idx[p + 10] << 8) | idx[p + 11];
+                p += 12;
+            }
+            uint64_t nblocks = uvint();
+            std::vector<Blk> blocks;
+            for (uint64_t b = 0; b < nblocks; b++) {
+                Blk bk;
+                bk.first_bc = parse_prefix();
+                bk.last_bc = parse_prefix();
+                bk.offset = uvint();
+                (void)svint();  // width - 64KiB (unused here)
+                bk.has_open = idx[p++] != 0;
+                if (bk.has_open) {
+                    if (idx[p] == 0x80) {
+                        bk.has_open = false;
+                        p += 1;
+                    } else {
+                        uint64_t m3v = 0;
+                        for (int i = 0; i < 8; i++) m3v = (m3v << 8) | idx[p + i];
+                        bk.open_m = (int64_t)m3v;
+                        bk.open_l = ((uint32_t)idx[p + 8] << 24) | ((uint32_t)idx[p + 9] << 16) |
+                                    ((uint32_t)idx[p + 10] << 8) | idx[p + 11];
+                        p += 12;
+                    }
+                }
+                blocks.push_back(std::move(bk));
+            }
+            p = pend;  // skip the block-offsets u32 table
+            bti::BtiRowIndexBlockSpec spec;
+            spec.partition_key.assign(key, key + klen);
+            spec.data_pos = pos;
+            spec.block_count = blocks.size();
+            spec.pdel_live = pdel_live;
+            spec.pdel_mfda = pdm;
+            spec.pdel_ldt = pdl;
+            bytes prev_sep, prev_max;
+            for (size_t b = 0; b < blocks.size(); b++) {
+                bytes bkey;
+                if (b > 0) bkey = bti::bti_separator_gt(blocks[b - 1].last_bc, blocks[b].first_bc);
+                bool has_od = b > 0 && blocks[b - 1].has_open;
+                int ob = 1;
+                while (blocks[b].offset >> (8 * ob - 1)) ob++;  // signed SizedInts
+                bti::BtiRowIndexBlockSpec::Entry e;
+                e.prefix = bkey;
+                e.pb = ob | (has_od ? 8 : 0);
+                for (int i = ob - 1; i >= 0; i--)
+                    e.payload.push_back((uint8_t)(blocks[b].offset >> (8 * i)));
+                if (has_od) {
+                    for (int i = 7; i >= 0; i--)
+                        e.payload.push_back((uint8_t)((uint64_t)blocks[b - 1].open_m >> (8 * i)));
+                    for (int i = 3; i >= 0; i--)
+                        e.payload.push_back((uint8_t)(blocks[b - 1].open_l >> (8 * i)));
+                }
+                spec.entries.push_back(std::move(e));
+                prev_sep = bkey;
+                prev_max = blocks[b].last_bc;
+            }
+            // final nudge entry; its payload (partition length - 1, the
+            // END_OF_PARTITION byte's offset) is patched once the next
+            // partition's position is known
+            size_t cm = 0;
+            while (cm < prev_max.size() && cm < prev_sep.size() && prev_max[cm] == prev_sep[cm]) cm++;
+            bti::BtiRowIndexBlockSpec::Entry fin;
+            fin.prefix = bti::bti_nudge(prev_max, cm);
+            spec.entries.push_back(std::move(fin));
+            pending_specs.push_back(std::move(spec));
+            pending_idx.push_back(pes.size());
+        }
+        bti::BtiKeyEntry e;
+        int64_t tok = murmur3_token(key, klen);
+        e.byte_comparable = bti::bti_byte_comparable_m3(tok, bytes(key, key + klen));
+        e.raw_key.assign(key, key + klen);
+        uint64_t h2[2];
+        murmur3_128(key, klen, 0, h2);
+        e.hash_bits = (uint8_t)h2[1];  // DecoratedKey.filterHashLowerBits
+        e.idxpos = idxpos;
+        pes.push_back(std::move(e));
+        part_positions.push_back(pos);
+    }
+    for (size_t i = 0; i < pending_specs.size(); i++) {
+        auto& spec = pending_specs[i];
+        size_t pi = pending_idx[i];
+        uint64_t next_pos = pi + 1 < part_positions.size() ? part_positions[pi + 1] : data_total;
+        uint64_t endoff = next_pos - spec.data_pos - 1;
+        auto& fin = spec.entries.back();
+        int ob = 1;
+        while (endoff >> (8 * ob - 1)) ob++;
+        fin.pb = ob;
+        fin.payload.clear();
+        for (int b = ob - 1; b >= 0; b--) fin.payload.push_back((uint8_t)(endoff >> (8 * b)));
+        pes[pi].idxpos = (int64_t)bti::append_bti_row_index(rows_file, spec);
+    }
+    *partitions_out = bti::write_bti_partitions(pes);
+    *rows_out = std::move(rows_file);
+}
+
 static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfColsBuf& rows,
                                            uint64_t n_groups, SerParams2 sp, DevBuf& d_stats,
                                            DevBuf& d_tomb, uint32_t tomb_cap,
@@ -820,7 +998,7 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
                                            const std::vector<std::pair<bytes, std::string>>& regular_cols,
                                            const std::vector<std::pair<bytes, std::string>>& static_cols,
                                            hipStream_t stream, int wslot = 0,
-                                           bool snappy_out = false) {
+                                           bool snappy_out = false, bool bti_out = false) {
     WriteDeviceOut w;
     const uint32_t SLOT_STRIDE = snappy_out ? SNP_SLOT : LZ4_SLOT;
     TR("wsd: enter");
@@ -1196,7 +1374,20 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
         TR("wsd: digest done");
         struct timespec ts0, ts1;
         clock_gettime(CLOCK_MONOTONIC, &ts0);
-        write_file_parallel(out_base + "-Index.db", h_index, total_idx, 4);
+        if (bti_out) {
+            bti::bytes parts_img, rows_img;
+            std::vector<int32_t> ckw_h2;
+            for (auto& t2 : ck_types)
+                ckw_h2.push_back(t2 == "org.apache.cassandra.db.marshal.LongType" ? 8
+                                 : t2 == "org.apache.cassandra.db.marshal.Int32Type" ? 4
+                                                                                     : -1);
+            build_bti_from_index_image(h_index, total_idx, total_unc, ckw_h2,
+                                       &parts_img, &rows_img);
+            write_file(out_base + "-Partitions.db", parts_img.data(), parts_img.size());
+            write_file(out_base + "-Rows.db", rows_img.data(), rows_img.size());
+        } else {
+            write_file_parallel(out_base + "-Index.db", h_index, total_idx, 4);
+        }
         {
             bytes f;
             put_be32(f, (uint32_t)bs.k);
@@ -1227,10 +1418,14 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
             write_file(out_base + "-Statistics.db", st.data(), st.size());
         }
         {
-            bytes s;
-            put_be32(s, 128); put_be32(s, 0); put_be64(s, 8); put_be32(s, 128); put_be32(s, 0);
-            write_file(out_base + "-Summary.db", s.data(), s.size());
-            std::string toc = "Data.db\nStatistics.db\nDigest.crc32\nTOC.txt\nCompressionInfo.db\nFilter.db\nIndex.db\nSummary.db\n";
+            if (!bti_out) {
+                bytes s;
+                put_be32(s, 128); put_be32(s, 0); put_be64(s, 8); put_be32(s, 128); put_be32(s, 0);
+                write_file(out_base + "-Summary.db", s.data(), s.size());
+            }
+            std::string toc = bti_out
+                ? "Data.db\nStatistics.db\nDigest.crc32\nTOC.txt\nCompressionInfo.db\nFilter.db\nPartitions.db\nRows.db\n"
+                : "Data.db\nStatistics.db\nDigest.crc32\nTOC.txt\nCompressionInfo.db\nFilter.db\nIndex.db\nSummary.db\n";
             write_file(out_base + "-TOC.txt", (const uint8_t*)toc.data(), toc.size());
         }
         clock_gettime(CLOCK_MONOTONIC, &ts1);
@@ -1300,6 +1495,7 @@ struct CompactSetup {
     std::vector<size_t> comp_file_sz;
     std::vector<int32_t> col_fixed_h;    // SIMPLE regular columns only
     uint32_t n_cpx = 0;                  // one complex (map<blob,blob>) column, last
+    bool bti = false;                    // inputs are `da` (trie-indexed)
     std::vector<int32_t> ck_widths;      // per clustering column
     std::vector<int32_t> static_fixed_h; // per static column
     // unsharded fast path: whole-file Data.db reads started during index
@@ -1639,6 +1835,8 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         bool gc_mode = kd < k;
         if (gc_mode && su.n_cpx)
             throw std::runtime_error("garbage-collect mode with complex columns unsupported");
+        if (su.bti && gc_mode)
+            throw std::runtime_error("garbage-collect mode with bti (da) inputs unsupported");
         uint64_t data_parts = 0;
         for (int s = 0; s < kd; s++) data_parts += srcs[s].n_parts;
         std::vector<uint64_t> runs;
@@ -1938,7 +2136,7 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
                                                 tomb_cap, out_base_str, stats[0].key_type,
                                                 stats[0].clustering_types, stats[0].regular_cols,
                                                 stats[0].static_cols, stream, wslot,
-                                                su.cinfos[0].snappy);
+                                                su.cinfos[0].snappy, su.bti);
         TR("writer done");
         {
             OutStats hst;
@@ -1992,6 +2190,15 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
 static void compact_setup(const gpuc_job* job, CompactSetup& su, bool preread_full) {
     int k = job->n_inputs + (job->n_tomb_sources > 0 ? job->n_tomb_sources : 0);
     su.k = k;
+    {
+        auto exists = [](const std::string& p2) {
+            FILE* f = fopen(p2.c_str(), "rb");
+            if (f) fclose(f);
+            return f != nullptr;
+        };
+        std::string b0 = job->input_bases[0];
+        su.bti = exists(b0 + "-Partitions.db") && !exists(b0 + "-Index.db");
+    }
     su.k_data = job->n_inputs;
     su.in_bases.resize(k);
     su.index_data.resize(k);
@@ -2053,7 +2260,25 @@ static void compact_setup(const gpuc_job* job, CompactSetup& su, bool preread_fu
             mth.emplace_back([&, s, preread_full]() {
                 try {
                     const std::string& base = su.in_bases[s];
-                    su.index_data[s] = read_file(base + "-Index.db");
+                    if (su.bti) {
+                        // `da` inputs: partition positions come from the
+                        // Partitions.db trie (payload < 0 == ~data_pos;
+                        // >= 0 == Rows.db TrieIndexEntry footer, whose
+                        // dataStartPosition leads). DFS order == key order
+                        // == data order.
+                        bytes pf = read_file(base + "-Partitions.db");
+                        bytes rf = read_file(base + "-Rows.db");
+                        bti::bytes pfb(pf.begin(), pf.end());
+                        bti::bytes rfb(rf.begin(), rf.end());
+                        auto bp = bti::read_bti_partitions(pfb);
+                        auto& pos = su.positions[s];
+                        pos.reserve(bp.entries.size() + 1);
+                        for (auto& e : bp.entries)
+                            pos.push_back(e.idxpos < 0 ? (uint64_t)~e.idxpos
+                                                       : bti::row_index_data_pos(rfb, (uint64_t)e.idxpos));
+                    } else {
+                        su.index_data[s] = read_file(base + "-Index.db");
+                    }
                     su.cinfos[s] = parse_compression_info(read_file(base + "-CompressionInfo.db"));
                     su.stats[s] = parse_statistics(read_file(base + "-Statistics.db"));
                     if (!preread_full) su.comp_file_sz[s] = file_size_of(base + "-Data.db");
@@ -2113,7 +2338,10 @@ static void compact_setup(const gpuc_job* job, CompactSetup& su, bool preread_fu
         throw std::runtime_error("1..63 static columns supported");
     if (su.ck_widths.size() > 32)
         throw std::runtime_error("at most 32 clustering columns supported");
-    {
+    if (su.bti) {
+        // positions collected from the tries above; append the end sentinel
+        for (int s = 0; s < k; s++) su.positions[s].push_back(su.cinfos[s].data_len);
+    } else {
         std::vector<std::thread> th;
         std::vector<std::string> perr(k);
         for (int s = 0; s < k; s++)
@@ -2176,6 +2404,8 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
 
         int S = S_pre;
         if (S > 1024) throw std::runtime_error("n_output_shards must be <= 1024");
+        if (su.bti && (S > 1 || job->has_token_range))
+            throw std::runtime_error("token sharding of bti (da) inputs unsupported this round");
         std::mutex res_mu;
         if (S == 1) {
             std::vector<std::pair<uint32_t, uint32_t>> pr(su.k);
@@ -2945,7 +3175,9 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
             // reset and recollect so the writer sees fresh stats (collect ran
             // once above only to derive the header mins)
             init_outstats(d_stats, stream);
-            std::string base = std::string(dir) + "/oa-" + std::to_string(spec->first_generation + s) + "-big";
+            std::string base = spec->bti
+                ? std::string(dir) + "/da-" + std::to_string(spec->first_generation + s) + "-bti"
+                : std::string(dir) + "/oa-" + std::to_string(spec->first_generation + s) + "-big";
             std::vector<std::pair<bytes, std::string>> cols;
             if (gp.n_value_cols == 1) {
                 cols.push_back({bytes{'v', 'a', 'l'}, "org.apache.cassandra.db.marshal.BytesType"});
@@ -2967,7 +3199,7 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
                                  std::vector<std::string>(
                                      gen_nck, gp.ck_text ? "org.apache.cassandra.db.marshal.UTF8Type"
                                                          : "org.apache.cassandra.db.marshal.LongType"),
-                                 cols, scols, stream, 0, spec->snappy != 0);
+                                 cols, scols, stream, 0, spec->snappy != 0, spec->bti != 0);
         }
         HIP_CHECK(hipStreamDestroy(stream));
         return GPUC_OK;

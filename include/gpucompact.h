@@ -178,6 +178,8 @@ typedef struct gpuc_gen_spec {
                                    ComplexColumnData.java:47) */
     uint32_t complex_del_pct;   /* percent of those rows also carrying a
                                    complexDeletion */
+    int32_t bti;                /* 1: write the `da` (trie-indexed) component
+                                   set (Partitions.db/Rows.db, BtiFormat.md) */
 } gpuc_gen_spec;
 
 int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* error, size_t error_len);

@@ -202,6 +202,7 @@ def test_merge_semantics_laws(oracle_bin, tmp_path):
         ["g++", "-O2", "-std=c++17",
          os.path.join(repo, "tests/native/merge_laws_test.cpp"),
          os.path.join(repo, "oracle/src/sstable.cpp"),
+         os.path.join(repo, "oracle/src/bti.cpp"),
          os.path.join(repo, "oracle/src/compact.cpp"),
          os.path.join(repo, "oracle/src/gen.cpp"),
          "-o", exe, "-l:liblz4.so.1"],
@@ -223,6 +224,7 @@ def test_gc_dsl_golden_vectors(tmp_path):
         ["g++", "-O2", "-std=c++17",
          os.path.join(repo, "tests/native/gc_dsl_test.cpp"),
          os.path.join(repo, "oracle/src/sstable.cpp"),
+         os.path.join(repo, "oracle/src/bti.cpp"),
          os.path.join(repo, "oracle/src/compact.cpp"),
          os.path.join(repo, "oracle/src/gen.cpp"),
          "-o", exe, "-l:liblz4.so.1", "-ldl"],
