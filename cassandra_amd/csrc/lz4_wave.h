@@ -472,11 +472,15 @@ __global__ void __launch_bounds__(WAVE) k_lz4_compress_wave_t(const uint8_t* dat
 // wave-cooperative decompress: comp chunk staged in LDS; the token chain runs
 // redundantly on all lanes (uniform); copies are wave-parallel.
 // ---------------------------------------------------------------------------
-// grid helper: with verify_crc the CRC work runs in extra trailing blocks
-// (one chunk per LANE — the byte-serial CRC would otherwise hold all 64
-// lanes of each decode block hostage behind one lane's loop)
+// grid helper: decode runs TWO chunks per wave (one per half-wave: the
+// token chain is a serial pointer chase, so two divergent 32-lane decodes
+// overlap their dependent-load latencies at no occupancy cost); with
+// verify_crc the CRC work runs in extra trailing blocks (one chunk per
+// LANE — the byte-serial CRC would otherwise hold all lanes of a decode
+// block hostage behind one lane's loop)
+__host__ __device__ static inline uint32_t lz4_decomp_blocks(uint32_t n) { return (n + 1) / 2; }
 static inline uint32_t lz4_decomp_grid(uint32_t n, int verify_crc) {
-    return verify_crc ? n + (n + WAVE - 1) / WAVE : n;
+    return lz4_decomp_blocks(n) + (verify_crc ? (n + WAVE - 1) / WAVE : 0);
 }
 
 __global__ void __launch_bounds__(WAVE) k_lz4_decompress_wave(const ChunkDesc* chunks, uint32_t n,
@@ -487,13 +491,15 @@ __global__ void __launch_bounds__(WAVE) k_lz4_decompress_wave(const ChunkDesc* c
     // compressed bytes are read through L1/L2 (all-lane same-address reads
     // broadcast; no LDS staging -> higher occupancy, same trade as the
     // compressor's global-src variant). crc_table is the 8x256 sliced set.
-    uint32_t c = blockIdx.x;
-    int lane = threadIdx.x;
-    if (c >= n) {
-        // CRC block: lane l checks chunk (c-n)*WAVE + l (launched only when
+    const uint32_t DB = lz4_decomp_blocks(n);
+    uint32_t c = blockIdx.x * 2 + (threadIdx.x >> 5);  // chunk per half-wave
+    int lane = threadIdx.x & 31;                       // lane within the half
+    constexpr int HW = 32;
+    if (blockIdx.x >= DB) {
+        // CRC block: lane l checks chunk (b-DB)*WAVE + l (launched only when
         // verify_crc; see lz4_decomp_grid). Each lane walks its own chunk
         // serially — consecutive iterations reuse the lane's cachelines.
-        uint32_t ci = (c - n) * WAVE + (uint32_t)lane;
+        uint32_t ci = (blockIdx.x - DB) * WAVE + (uint32_t)threadIdx.x;
         if (ci >= n) return;
         ChunkDesc ch = chunks[ci];
         if (ch.comp_len > LZ4_SLOT) return;  // decode block flags it
@@ -517,6 +523,7 @@ __global__ void __launch_bounds__(WAVE) k_lz4_decompress_wave(const ChunkDesc* c
         if (crc != stored) { if (bad_chunks) bad_chunks[ci] = 1; else atomicExch(error, 1ull); }
         return;
     }
+    if (c >= n) return;  // odd tail: second half-wave has no chunk
     ChunkDesc ch = chunks[c];
     if (ch.comp_len > LZ4_SLOT) { if (lane == 0) { if (bad_chunks) bad_chunks[c] = 1; else atomicExch(error, 9ull); } return; }
     const uint8_t* s_comp = ch.comp;
@@ -534,7 +541,7 @@ __global__ void __launch_bounds__(WAVE) k_lz4_decompress_wave(const ChunkDesc* c
             do { s = s_comp[ip++]; lit += s; } while (s == 255);
         }
         if (opos + lit > olen || ip + lit > iend) { if (lane == 0) { if (bad_chunks) bad_chunks[c] = 1; else atomicExch(error, 3ull); } return; }
-        for (uint32_t i = 4u * lane; i < lit; i += 4u * WAVE) {
+        for (uint32_t i = 4u * lane; i < lit; i += 4u * HW) {
             uint32_t nb = lit - i;
             if (nb >= 4) {
                 uint32_t v;
@@ -558,14 +565,14 @@ __global__ void __launch_bounds__(WAVE) k_lz4_decompress_wave(const ChunkDesc* c
         if (off == 1) {
             // byte run: broadcast, dword stores
             uint32_t b4 = 0x01010101u * src[0];
-            for (uint32_t i = 4u * lane; i < ml; i += 4u * WAVE) {
+            for (uint32_t i = 4u * lane; i < ml; i += 4u * HW) {
                 uint32_t nb = ml - i;
                 if (nb >= 4) memcpy(out + opos + i, &b4, 4);
                 else for (uint32_t j = 0; j < nb; j++) out[opos + i + j] = (uint8_t)b4;
             }
         } else if (ml <= off) {
             // disjoint regions: dword copy
-            for (uint32_t i = 4u * lane; i < ml; i += 4u * WAVE) {
+            for (uint32_t i = 4u * lane; i < ml; i += 4u * HW) {
                 uint32_t nb = ml - i;
                 if (nb >= 4) {
                     uint32_t v;
@@ -579,7 +586,7 @@ __global__ void __launch_bounds__(WAVE) k_lz4_decompress_wave(const ChunkDesc* c
             // overlapped copy (ml > off) == periodic repetition of the last
             // `off` bytes; the modulo form avoids reading bytes this same
             // copy has not written yet
-            for (uint32_t i = lane; i < ml; i += WAVE) out[opos + i] = src[i % off];
+            for (uint32_t i = lane; i < ml; i += HW) out[opos + i] = src[i % off];
         }
         opos += ml;
     }

@@ -387,10 +387,15 @@ __global__ void __launch_bounds__(WAVE) k_snappy_decompress_chunks(const ChunkDe
                                                                    unsigned long long* error,
                                                                    const uint32_t* crc_table,
                                                                    uint8_t* bad_chunks = nullptr) {
-    uint32_t c = blockIdx.x;
-    int lane = threadIdx.x;
-    if (c >= n) {
-        uint32_t ci = (c - n) * WAVE + (uint32_t)lane;
+    // grid shared with the LZ4 decoder (lz4_decomp_grid): TWO chunks per
+    // wave (one per half-wave — overlaps the serial tag-chain latencies),
+    // CRC verification in trailing one-chunk-per-LANE blocks
+    const uint32_t DB = lz4_decomp_blocks(n);
+    uint32_t c = blockIdx.x * 2 + (threadIdx.x >> 5);
+    int lane = threadIdx.x & 31;
+    constexpr int HW = 32;
+    if (blockIdx.x >= DB) {
+        uint32_t ci = (blockIdx.x - DB) * WAVE + (uint32_t)threadIdx.x;
         if (ci >= n) return;
         ChunkDesc ch = chunks[ci];
         if (ch.comp_len > SNP_SLOT) return;  // decode block flags it
@@ -416,6 +421,7 @@ __global__ void __launch_bounds__(WAVE) k_snappy_decompress_chunks(const ChunkDe
         return;
     }
     (void)verify_crc;
+    if (c >= n) return;  // odd tail half-wave
     ChunkDesc ch = chunks[c];
     if (ch.comp_len > SNP_SLOT) { if (lane == 0) { if (bad_chunks) bad_chunks[c] = 1; else atomicExch(error, 9ull); } return; }
     const uint8_t* in = ch.comp;
@@ -454,7 +460,7 @@ __global__ void __launch_bounds__(WAVE) k_snappy_decompress_chunks(const ChunkDe
                 ip += nb;
             }
             if (ip + len > iend || opos + len > olen) { fail(3); return; }
-            for (uint32_t i = 4 * (uint32_t)lane; i < len; i += 4 * WAVE) {
+            for (uint32_t i = 4 * (uint32_t)lane; i < len; i += 4 * HW) {
                 uint32_t nb = len - i;
                 if (nb >= 4) {
                     uint32_t v;
@@ -487,13 +493,13 @@ __global__ void __launch_bounds__(WAVE) k_snappy_decompress_chunks(const ChunkDe
             const uint8_t* src = out + opos - offset;
             if (offset == 1) {
                 uint32_t b4 = 0x01010101u * src[0];
-                for (uint32_t i = 4 * (uint32_t)lane; i < len; i += 4 * WAVE) {
+                for (uint32_t i = 4 * (uint32_t)lane; i < len; i += 4 * HW) {
                     uint32_t nb = len - i;
                     if (nb >= 4) memcpy(out + opos + i, &b4, 4);
                     else for (uint32_t j = 0; j < nb; j++) out[opos + i + j] = (uint8_t)b4;
                 }
             } else if (len <= offset) {
-                for (uint32_t i = 4 * (uint32_t)lane; i < len; i += 4 * WAVE) {
+                for (uint32_t i = 4 * (uint32_t)lane; i < len; i += 4 * HW) {
                     uint32_t nb = len - i;
                     if (nb >= 4) {
                         uint32_t v;
@@ -504,7 +510,7 @@ __global__ void __launch_bounds__(WAVE) k_snappy_decompress_chunks(const ChunkDe
                     }
                 }
             } else {
-                for (uint32_t i = (uint32_t)lane; i < len; i += WAVE)
+                for (uint32_t i = (uint32_t)lane; i < len; i += HW)
                     out[opos + i] = src[i % offset];
             }
             opos += len;
