@@ -677,6 +677,57 @@ def test_bti_da_pipeline(ca, oracle_bin, tmp_path):
         _assert_dirs_equal(f"{dg}/da-{g}-bti", f"{do}/da-{g}-bti", BTI_COMPONENTS)
 
 
+def test_cross_feature_combinations(ca, oracle_bin, tmp_path):
+    """Round-2 features composed: (a) complex columns + TTL + expiry-crossing
+    purge in one table; (b) complex columns in a da (BTI) sstable set;
+    (c) snappy + wide partitions + TTL; (d) complex + overlap bloom purge —
+    each byte-identical to the oracle."""
+    INT64_MIN = -(2 ** 63)
+    d = str(tmp_path)
+    # (a) complex + ttl + purge across expiry
+    _oracle_gen(d, seed=91, n=3, rows=900, vlen=120, overlap=25, tomb=10,
+                cpx=40, cpxdel=25, ttl=30)
+    ins = [f"{d}/oa-{g}-big" for g in (1, 2, 3)]
+    ca.compact(ins, f"{d}/oa-60-big", now_sec=1800000000, gc_before=1700001000)
+    _oracle_compact(f"{d}/oa-80-big", ins, now=1800000000, gcbefore=1700001000)
+    _assert_dirs_equal(f"{d}/oa-60-big", f"{d}/oa-80-big")
+    # (b) complex columns inside da sstables
+    db = d + "/b"
+    os.makedirs(db)
+    _oracle_gen(db, seed=92, n=2, rows=700, vlen=100, overlap=20, tomb=10,
+                cpx=45, cpxdel=25, bti=1)
+    bins = [f"{db}/da-{g}-bti" for g in (1, 2)]
+    ca.compact(bins, f"{db}/da-60-bti")
+    _oracle_compact(f"{db}/da-80-bti", bins)
+    _assert_dirs_equal(f"{db}/da-60-bti", f"{db}/da-80-bti",
+                       ["Data.db", "CompressionInfo.db", "Filter.db", "Digest.crc32",
+                        "Statistics.db", "Partitions.db", "Rows.db", "TOC.txt"])
+    # (c) snappy + wide + ttl
+    dc = d + "/c"
+    os.makedirs(dc)
+    _oracle_gen(dc, seed=93, n=2, rows=30, crows=60, vlen=150, rtomb=25,
+                tomb=10, overlap=20, ttl=35, snappy=1)
+    cins = [f"{dc}/oa-{g}-big" for g in (1, 2)]
+    ca.compact(cins, f"{dc}/oa-60-big", now_sec=1800000000, gc_before=INT64_MIN)
+    _oracle_compact(f"{dc}/oa-80-big", cins, now=1800000000)
+    _assert_dirs_equal(f"{dc}/oa-60-big", f"{dc}/oa-80-big")
+    # (d) complex + per-key bloom-gated purge
+    dd = d + "/d"
+    os.makedirs(dd)
+    _oracle_gen(dd, seed=94, n=2, rows=600, vlen=80, overlap=0, tomb=40,
+                cpx=40, cpxdel=30)
+    os.makedirs(dd + "/y")
+    _oracle_gen(dd + "/y", seed=95, n=1, rows=300, vlen=50, keylen=9)
+    dins = [f"{dd}/oa-1-big", f"{dd}/oa-2-big"]
+    LO, HI = INT64_MIN, 2 ** 63 - 1
+    flt = f"{dd}/y/oa-1-big-Filter.db"
+    ca.compact(dins, f"{dd}/oa-60-big", gc_before=2000000000,
+               overlaps=[(LO, HI, 1, flt)])
+    _oracle_compact(f"{dd}/oa-80-big", dins, gcbefore=2000000000,
+                    ov=f"{LO}:{HI}:1:{flt}")
+    _assert_dirs_equal(f"{dd}/oa-60-big", f"{dd}/oa-80-big")
+
+
 def test_cancellation(ca, oracle_bin, tmp_path):
     """Cooperative cancel (CompactionIterator.isStopRequested): a set
     cancel_flag aborts the task with GPUC_ERR_CANCELLED; a zero flag is
