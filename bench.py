@@ -138,7 +138,19 @@ def main():
     device = local_rank % max(1, ca.device_count())
 
     # ---- per-rank inputs (GPU-generated, seed disjoint per rank) ----
-    d = os.path.join(args.dir, f"r{rank}")
+    # Disk budget at high rank counts: 8 ranks x (8.6 GB inputs + up to
+    # 2 x 8.6 GB in-flight outputs) would overflow the pool boxes' ~81 GB
+    # /tmp overlay. With >=4 ranks, park the READ-ONLY inputs on /dev/shm
+    # (tmpfs reads are fine; only its write path is slow) and keep outputs
+    # on the page-cache-backed dir; deletes join before the next write.
+    in_root = args.dir
+    if world >= 4 and args.dir.startswith("/tmp") and os.path.isdir("/dev/shm"):
+        in_root = "/dev/shm/gpuc_bench_in"
+    d = os.path.join(in_root, f"r{rank}")
+    dout = os.path.join(args.dir, f"r{rank}")
+    if dout != d:
+        shutil.rmtree(dout, ignore_errors=True)
+        os.makedirs(dout, exist_ok=True)
     shutil.rmtree(d, ignore_errors=True)
     os.makedirs(d, exist_ok=True)
     t_gen = time.time()
@@ -157,15 +169,19 @@ def main():
     cleaners = []
 
     def one_step(i, prev=[None]):
-        out = os.path.join(d, f"out-{i}", "oa-100-big")
+        out = os.path.join(dout, f"out-{i}", "oa-100-big")
         os.makedirs(os.path.dirname(out), exist_ok=True)
         r = ca.compact(bases, out, device=device, n_output_shards=args.shards)
         # previous step's output is deleted in the background (bounded disk,
-        # no serial rmtree inside the measured path)
+        # no serial rmtree inside the measured path); at >=4 ranks the delete
+        # must land before the NEXT step's write or /tmp overflows
         if prev[0]:
             t = threading.Thread(target=shutil.rmtree, args=(prev[0],), kwargs={"ignore_errors": True})
             t.start()
-            cleaners.append(t)
+            if world >= 4:
+                t.join()
+            else:
+                cleaners.append(t)
         prev[0] = os.path.dirname(out)
         return r
 
