@@ -71,6 +71,148 @@ static const Cell& reconcile_cells(const Cell& left, const Cell& right) {
 }
 
 // ---------------------------------------------------------------------------
+// CounterContext (db/context/CounterContext.java): header of global/local
+// element flags + (CounterId[16], clock i64, count i64) shards in unsigned
+// id order. merge() transcribed pairwise (compare(): global beats all with
+// (clock, count) ties; local+local sums (DISJOINT); remote by the legacy-
+// aware clock rule then count). Cell format fixture-pinned by the
+// reference's legacy_oa_*_counter sstables.
+// ---------------------------------------------------------------------------
+struct CtxShard {
+    uint8_t id[16];
+    int64_t clock, count;
+    bool global = false, local = false;
+};
+static std::vector<CtxShard> ctx_parse(const bytes& v) {
+    std::vector<CtxShard> out;
+    if (v.size() < 2) return out;
+    int16_t hdr = (int16_t)((v[0] << 8) | v[1]);
+    int n_flagged = hdr < 0 ? -hdr : hdr;
+    size_t body = 2 + (size_t)n_flagged * 2;
+    size_t nsh = (v.size() - body) / 32;
+    std::vector<int16_t> elts(n_flagged);
+    for (int i = 0; i < n_flagged; i++) elts[i] = (int16_t)((v[2 + 2 * i] << 8) | v[3 + 2 * i]);
+    for (size_t i = 0; i < nsh; i++) {
+        CtxShard sh;
+        memcpy(sh.id, v.data() + body + i * 32, 16);
+        uint64_t c1 = 0, c2 = 0;
+        for (int b = 0; b < 8; b++) c1 = (c1 << 8) | v[body + i * 32 + 16 + b];
+        for (int b = 0; b < 8; b++) c2 = (c2 << 8) | v[body + i * 32 + 24 + b];
+        sh.clock = (int64_t)c1;
+        sh.count = (int64_t)c2;
+        for (int16_t e : elts) {
+            if (hdr >= 0 && e == (int16_t)(i + INT16_MIN)) sh.global = true;
+            if (e == (int16_t)i && hdr >= 0) sh.local = true;
+            if (hdr < 0 && e == (int16_t)i) sh.local = true;  // pre-2.1 header
+        }
+        out.push_back(sh);
+    }
+    return out;
+}
+static bytes ctx_serialize(const std::vector<CtxShard>& shards) {
+    bytes out;
+    std::vector<int16_t> elts;
+    for (size_t i = 0; i < shards.size(); i++) {
+        if (shards[i].global) elts.push_back((int16_t)(i + INT16_MIN));
+        else if (shards[i].local) elts.push_back((int16_t)i);
+    }
+    out.push_back((uint8_t)(elts.size() >> 8));
+    out.push_back((uint8_t)elts.size());
+    for (int16_t e : elts) {
+        out.push_back((uint8_t)((uint16_t)e >> 8));
+        out.push_back((uint8_t)e);
+    }
+    for (auto& sh : shards) {
+        out.insert(out.end(), sh.id, sh.id + 16);
+        for (int b = 7; b >= 0; b--) out.push_back((uint8_t)((uint64_t)sh.clock >> (8 * b)));
+        for (int b = 7; b >= 0; b--) out.push_back((uint8_t)((uint64_t)sh.count >> (8 * b)));
+    }
+    return out;
+}
+// compare(): -1 LESS, 0 EQUAL, +1 GREATER, 2 DISJOINT (CounterContext.java:~300)
+static int ctx_compare(const CtxShard& l, const CtxShard& r) {
+    if (l.global || r.global) {
+        if (l.global && r.global) {
+            if (l.clock == r.clock)
+                return l.count > r.count ? 1 : l.count == r.count ? 0 : -1;
+            return l.clock > r.clock ? 1 : -1;
+        }
+        return l.global ? 1 : -1;
+    }
+    if (l.local || r.local) {
+        if (l.local && r.local) return 2;
+        return l.local ? 1 : -1;
+    }
+    if (l.clock == r.clock)
+        return l.count > r.count ? 1 : l.count == r.count ? 0 : -1;
+    if ((l.clock >= 0 && r.clock > 0 && l.clock >= r.clock) ||
+        (l.clock < 0 && (r.clock > 0 || l.clock < r.clock)))
+        return 1;
+    return -1;
+}
+static bytes ctx_merge(const bytes& lv, const bytes& rv) {
+    auto L = ctx_parse(lv), R = ctx_parse(rv);
+    std::vector<CtxShard> out;
+    size_t i = 0, j = 0;
+    bool lsup = true, rsup = true;
+    while (i < L.size() && j < R.size()) {
+        int c = memcmp(L[i].id, R[j].id, 16);
+        if (c == 0) {
+            int rel = ctx_compare(L[i], R[j]);
+            if (rel == 2) {  // DISJOINT: two local shards sum
+                CtxShard m = L[i];
+                m.clock += R[j].clock;
+                m.count += R[j].count;
+                out.push_back(m);
+                lsup = rsup = false;
+            } else if (rel > 0) {
+                out.push_back(L[i]);
+                rsup = false;
+            } else {
+                out.push_back(R[j]);
+                if (rel < 0) lsup = false;
+            }
+            i++; j++;
+        } else if (c < 0) {
+            out.push_back(L[i++]);
+            rsup = false;
+        } else {
+            out.push_back(R[j++]);
+            lsup = false;
+        }
+    }
+    if (i < L.size()) rsup = false;
+    if (j < R.size()) lsup = false;
+    while (i < L.size()) out.push_back(L[i++]);
+    while (j < R.size()) out.push_back(R[j++]);
+    // superset early-return keeps the side's exact bytes (merge():~)
+    if (lsup) return lv;
+    if (rsup) return rv;
+    return ctx_serialize(out);
+}
+// Cells.resolveCounter (Cells.java): tombstone beats any live counter; empty
+// values lose; else context merge with ts = max
+static Cell counter_reconcile(const Cell& l, const Cell& r) {
+    bool lt = l.ldt != LDT_NONE_U32, rt = r.ldt != LDT_NONE_U32;
+    if (lt || rt) {
+        if (lt != rt) return lt ? l : r;
+        // two tombstones: regular rules (both land in resolveRegular)
+        return reconcile_cells(l, r);
+    }
+    bool le = l.value.empty(), re2 = r.value.empty();
+    if (le || re2) {
+        if (le != re2) return le ? r : l;
+        return l.ts > r.ts ? l : r;
+    }
+    Cell out = l;
+    out.value = ctx_merge(l.value, r.value);
+    out.ts = std::max(l.ts, r.ts);
+    out.ldt = LDT_NONE_U32;
+    out.ttl = NO_TTL;
+    return out;
+}
+
+// ---------------------------------------------------------------------------
 // Row.Merger.merge (Row.java:730-781) — simple columns only
 // ---------------------------------------------------------------------------
 static bool row_merge(const std::vector<const Row*>& versions, const DeletionTime& active_deletion,
@@ -149,6 +291,19 @@ static bool row_merge(const std::vector<const Row*>& versions, const DeletionTim
             if (out_cd.del.live() && out_cd.cells.empty()) continue;
             out.complex[ci] = std::move(out_cd);
             any_cell = true;
+            continue;
+        }
+        if (is_counter_type(cols[ci].second)) {
+            // Cells.resolveCounter chain (value-semantics: merged contexts
+            // allocate fresh bytes)
+            std::optional<Cell> mc;
+            for (auto* r : versions) {
+                if (!r || ci >= r->cells.size() || !r->cells[ci]) continue;
+                const Cell& cell = *r->cells[ci];
+                if (active.deletes(cell.ts)) continue;
+                mc = mc ? counter_reconcile(*mc, cell) : cell;
+            }
+            if (mc) { out.cells[ci] = std::move(*mc); any_cell = true; }
             continue;
         }
         const Cell* merged = nullptr;
@@ -641,9 +796,9 @@ static Header make_output_header(const std::vector<SSTable>& inputs) {
 CompactionResult compact(const CompactionJob& job) {
     if (!job.tomb_sources.empty() && !job.inputs.empty())
         for (auto& cp : job.inputs[0].header.regular_cols)
-            if (is_complex_type(cp.second))
+            if (is_complex_type(cp.second) || is_counter_type(cp.second))
                 throw std::runtime_error(
-                    "garbage-collect mode with complex columns unsupported");
+                    "garbage-collect mode with complex/counter columns unsupported");
     CompactionResult res;
     res.out.header = make_output_header(job.inputs);
     res.out.comp = job.inputs.empty() ? CompressionParams{} : job.inputs[0].comp;

@@ -25,6 +25,8 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
     }
     if (g.complex_pct > 0)  // one complex column, named to sort last
         t.header.regular_cols.push_back({bytes{'z', 'm'}, CqlType::MAP_BB});
+    if (g.counter)  // counter tables: the single counter column replaces the blobs
+        t.header.regular_cols = {{bytes{'c', 'n', 't'}, CqlType::COUNTER}};
 
     struct Ent { int64_t token; bytes key; uint64_t id; };
     std::vector<Ent> ents;
@@ -42,6 +44,42 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
 
     int64_t min_ts = INT64_MAX, min_ldt_l = INT64_MAX;
     int32_t min_ttl = INT32_MAX;
+    // counter context for (sst, key): shards from the sorted 8-id pool
+    auto make_ctx = [&](uint32_t sst2, uint64_t id) {
+        struct P { uint8_t id[16]; uint32_t idx; };
+        static std::vector<P> pool = [] {
+            std::vector<P> v(8);
+            for (uint32_t i = 0; i < 8; i++) { gen_counter_id(i, v[i].id); v[i].idx = i; }
+            std::sort(v.begin(), v.end(), [](const P& a, const P& b) {
+                return memcmp(a.id, b.id, 16) < 0;
+            });
+            return v;
+        }();
+        bytes body;
+        std::vector<int16_t> elts;
+        int shard_no = 0;
+        for (auto& pp2 : pool) {
+            if (!gen_ctr_present(g, sst2, id, pp2.idx)) continue;
+            int role = gen_ctr_role(pp2.idx);
+            if (role == 0) elts.push_back((int16_t)(shard_no + INT16_MIN));
+            else if (role == 1) elts.push_back((int16_t)shard_no);
+            body.insert(body.end(), pp2.id, pp2.id + 16);
+            int64_t ck2 = gen_ctr_clock(g, sst2, id, pp2.idx);
+            int64_t cn = gen_ctr_count(g, sst2, id, pp2.idx);
+            for (int b = 7; b >= 0; b--) body.push_back((uint8_t)((uint64_t)ck2 >> (8 * b)));
+            for (int b = 7; b >= 0; b--) body.push_back((uint8_t)((uint64_t)cn >> (8 * b)));
+            shard_no++;
+        }
+        bytes out;
+        out.push_back((uint8_t)(elts.size() >> 8));
+        out.push_back((uint8_t)elts.size());
+        for (int16_t e2 : elts) {
+            out.push_back((uint8_t)((uint16_t)e2 >> 8));
+            out.push_back((uint8_t)e2);
+        }
+        out.insert(out.end(), body.begin(), body.end());
+        return out;
+    };
     // complex column "zm": dedup+sorted map cells, optional complexDeletion
     auto put_complex = [&](Row& r, uint32_t sst2, uint64_t id, uint32_t j, int64_t ts) {
         if (!gen_has_complex(g, sst2, id, j)) return;
@@ -165,7 +203,8 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
                             cell.ttl = r.live.ttl;
                             cell.ldt = ldt_to_u32(r.live.let);
                         }
-                        gen_value(g, sst, gen_col_value_id(seed_id, cc), cell.value);
+                        if (g.counter) cell.value = make_ctx(sst, e.id ^ ((uint64_t)j << 20));
+                        else gen_value(g, sst, gen_col_value_id(seed_id, cc), cell.value);
                         r.cells[cc] = std::move(cell);
                     }
                     put_complex(r, sst, e.id, j, ts);
@@ -233,7 +272,8 @@ SSTable generate_sstable(const GenSpec& g, uint32_t sst) {
                     c.ttl = r.live.ttl;
                     c.ldt = ldt_to_u32(r.live.let);
                 }
-                gen_value(g, sst, gen_col_value_id(e.id, cc), c.value);
+                if (g.counter) c.value = make_ctx(sst, e.id);
+                else gen_value(g, sst, gen_col_value_id(e.id, cc), c.value);
                 r.cells[cc] = std::move(c);
             }
             put_complex(r, sst, e.id, 0, ts);

@@ -62,6 +62,13 @@ struct GenSpec {
     // (shadowing ts-older cells on merge). GPU mirror: GenParams2.complex_pct.
     uint32_t complex_pct = 0;
     uint32_t complex_del_pct = 0;
+    // counter: 1 = the schema becomes ONE counter column "cnt"
+    // (CounterColumnType; counter tables hold only counter columns). Cell
+    // values are synthetic CounterContexts drawn from an 8-id pool with
+    // global/local/remote roles chosen to exercise every compare() branch
+    // (global clock ties with differing counts, local+local sums, remote
+    // clock rules incl. negative legacy clocks). GPU mirror: GenParams2.
+    uint32_t counter = 0;
     // ttl_pct: percent of LIVE rows written with EXPIRING liveness/cells
     // (LivenessInfo.java:67 ExpiringLivenessInfo; cells carry ttl +
     // localDeletionTime == localExpirationTime, AbstractCell.java:53-76).
@@ -185,6 +192,40 @@ inline uint32_t gen_cpx_path_val(const GenSpec& g, uint32_t sst, uint64_t key_id
 inline uint64_t gen_cpx_value_word(const GenSpec& g, uint32_t sst, uint64_t key_id, uint32_t j, uint32_t pathv) {
     return splitmix64(g.seed ^ key_id * 131 ^ ((uint64_t)sst << 40) ^ (uint64_t)(j + 1) * 29 ^
                       (uint64_t)(pathv + 1) * 389);
+}
+// counter-context derivations (roles fixed per id; presence/clock/count per
+// (sstable, key, id) so versions of a key collide on every role pairing)
+inline void gen_counter_id(uint32_t idx, uint8_t out[16]) {
+    uint64_t a = splitmix64(0xC0C0C0C0ULL + idx), b = splitmix64(0xF00DF00DULL + idx);
+    for (int i = 0; i < 8; i++) out[i] = (uint8_t)(a >> (8 * (7 - i)));
+    for (int i = 0; i < 8; i++) out[8 + i] = (uint8_t)(b >> (8 * (7 - i)));
+}
+inline bool gen_ctr_present(const GenSpec& g, uint32_t sst, uint64_t key_id, uint32_t idx) {
+    return splitmix64(g.seed ^ 0xC717C717ULL ^ key_id ^ ((uint64_t)sst << 32) ^
+                      (uint64_t)(idx + 1) * 37) % 100 < 55;
+}
+// role per id: 0=global, 1=local, 2=remote
+inline int gen_ctr_role(uint32_t idx) { return idx % 3; }
+inline int64_t gen_ctr_clock(const GenSpec& g, uint32_t sst, uint64_t key_id, uint32_t idx) {
+    int role = gen_ctr_role(idx);
+    uint64_t r = splitmix64(g.seed ^ 0xC10CULL ^ key_id ^ (uint64_t)(idx + 1) * 131 ^
+                            ((uint64_t)sst << 40));
+    if (role == 0) {
+        // globals: half the keys share the clock across sstables (the
+        // equal-clock, differing-count self-heal branch), half differ
+        uint64_t base = splitmix64(g.seed ^ 0x610BULL ^ key_id ^ idx);
+        return (int64_t)(1000 + (key_id % 2 ? base % 50 : (base + sst) % 50));
+    }
+    if (role == 1) return (int64_t)(1 + r % 5);  // locals sum
+    if (idx == 8 - 1 && (key_id % 5) == 0) return -(int64_t)(1 + r % 90);  // legacy negative
+    // remote clocks are STRICTLY positive (clock 0 hits the asymmetric
+    // no-branch case of CounterContext.compare and is never produced by
+    // live Cassandra: clocks start at 1)
+    return (int64_t)(1 + r % 99);
+}
+inline int64_t gen_ctr_count(const GenSpec& g, uint32_t sst, uint64_t key_id, uint32_t idx) {
+    return (int64_t)(splitmix64(g.seed ^ 0xC0117ULL ^ key_id ^ (uint64_t)(idx + 1) * 17 ^
+                                ((uint64_t)sst << 24)) % 1000) - 100;
 }
 inline bool gen_has_static(const GenSpec& g, uint32_t sst, uint64_t key_id) {
     if (g.static_pct == 0) return false;

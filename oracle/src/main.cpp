@@ -160,6 +160,7 @@ static GenSpec spec_from_kv(std::map<std::string, std::string>& kv) {
     g.snappy = (uint32_t)geti("snappy", g.snappy);
     g.ttl_pct = (uint32_t)geti("ttl", g.ttl_pct);
     g.bti = (uint32_t)geti("bti", g.bti);
+    g.counter = (uint32_t)geti("counter", g.counter);
     g.complex_pct = (uint32_t)geti("cpx", g.complex_pct);
     g.complex_del_pct = (uint32_t)geti("cpxdel", g.complex_del_pct);
     return g;

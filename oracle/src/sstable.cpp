@@ -26,12 +26,13 @@ const char* cql_type_name(CqlType t) {
             return "org.apache.cassandra.db.marshal.MapType"
                    "(org.apache.cassandra.db.marshal.BytesType,"
                    "org.apache.cassandra.db.marshal.BytesType)";
+        case CqlType::COUNTER: return "org.apache.cassandra.db.marshal.CounterColumnType";
     }
     return "?";
 }
 CqlType cql_type_from_name(const std::string& n) {
     for (CqlType t : {CqlType::BYTES, CqlType::UTF8, CqlType::ASCII, CqlType::LONG, CqlType::INT32,
-                      CqlType::MAP_BB})
+                      CqlType::MAP_BB, CqlType::COUNTER})
         if (n == cql_type_name(t)) return t;
     throw std::runtime_error("oracle: unsupported AbstractType " + n);
 }

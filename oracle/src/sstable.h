@@ -27,7 +27,11 @@ inline int64_t ldt_to_long(uint32_t u) { return u == LDT_NONE_U32 ? NO_DELETION_
 inline uint32_t ldt_to_u32(int64_t l) { return l == NO_DELETION_TIME ? LDT_NONE_U32 : (uint32_t)l; }
 
 // ---- types (db/marshal/*) ----
-enum class CqlType : uint8_t { BYTES, UTF8, ASCII, LONG, INT32, MAP_BB };
+enum class CqlType : uint8_t { BYTES, UTF8, ASCII, LONG, INT32, MAP_BB, COUNTER };
+// CounterColumnType cells: value = a counter CONTEXT (CounterContext.java:
+// header of global/local flags + (CounterId, clock, count) shards); variable
+// width on the wire like BytesType, but reconciled by context merge.
+inline bool is_counter_type(CqlType t) { return t == CqlType::COUNTER; }
 // MAP_BB == MapType(BytesType,BytesType): the one COMPLEX column type the
 // engine carries (ColumnMetadata.isComplex; db/rows/ComplexColumnData.java:47).
 // Cell paths are the map keys, compared as BytesType (unsigned lexicographic).

@@ -19,3 +19,8 @@ for t in legacy_da_simple legacy_da_clust; do
   cp "$SRCD/$t"/da-1-bti-* "$DST/$t/"
 done
 chmod -R u+w "$DST"/legacy_oa_* "$DST"/legacy_da_*
+# counter-table fixtures (same provenance): anchors for round-2 counter support
+for t in legacy_oa_simple_counter legacy_oa_clust_counter; do
+  mkdir -p "$DST/$t"
+  cp "$SRC/$t"/oa-1-big-* "$DST/$t/"
+done

@@ -528,3 +528,28 @@ def test_bti_da_writer(oracle_bin, tmp_path):
     oracle_run("compact", f"{d}/da-90-bti", *[f"{d}/da-{g}-bti" for g in (1, 2, 3)])
     assert os.path.exists(f"{d}/da-90-bti-Partitions.db")
     assert "FAIL" not in oracle_run("roundtrip", f"{d}/da-90-bti").stdout
+
+
+def test_counter_columns_oracle(oracle_bin, tmp_path):
+    """Counter columns (CounterColumnType): the reference's own oa counter
+    fixtures roundtrip byte-identically (cell format pinned); generated
+    counter tables (8-id shard pool covering every CounterContext.compare
+    branch: global clock ties w/ differing counts, local+local sums, remote
+    clock rules incl. legacy negative clocks) compact deterministically —
+    input-order invariant — and outputs roundtrip (Cells.resolveCounter +
+    CounterContext.merge, CounterContext.java)."""
+    import filecmp
+    for fx in ("legacy_oa_simple_counter", "legacy_oa_clust_counter"):
+        r = oracle_run("roundtrip", os.path.join(GOLDEN, fx, "oa-1-big"))
+        assert "FAIL" not in r.stdout, (fx, r.stdout)
+    d = str(tmp_path)
+    oracle_run("gen", d, "seed=99", "n=3", "rows=500", "overlap=30", "tomb=10",
+               "counter=1")
+    ins = [f"{d}/oa-{g}-big" for g in (1, 2, 3)]
+    for b in ins:
+        assert "FAIL" not in oracle_run("roundtrip", b).stdout
+    oracle_run("compact", f"{d}/oa-90-big", *ins)
+    oracle_run("compact", f"{d}/oa-91-big", *reversed(ins))
+    for c in ("Data.db", "Index.db", "Digest.crc32"):
+        assert filecmp.cmp(f"{d}/oa-90-big-{c}", f"{d}/oa-91-big-{c}", shallow=False), c
+    assert "FAIL" not in oracle_run("roundtrip", f"{d}/oa-90-big").stdout
