@@ -106,6 +106,7 @@ class GpucGenSpec(ctypes.Structure):
         ("complex_pct", ctypes.c_uint32),
         ("complex_del_pct", ctypes.c_uint32),
         ("bti", ctypes.c_int32),
+        ("counter", ctypes.c_int32),
     ]
 
 
@@ -369,6 +370,7 @@ def generate(
     complex_pct=0,
     complex_del_pct=0,
     bti=False,
+    counter=False,
     base_ts=1700000000000000,
     base_ldt=1700000000,
     first_generation=1,
@@ -402,6 +404,7 @@ def generate(
         complex_pct=complex_pct,
         complex_del_pct=complex_del_pct,
         bti=1 if bti else 0,
+        counter=1 if counter else 0,
     )
     err = ctypes.create_string_buffer(256)
     rc = lib.gpuc_generate(ctypes.byref(spec), out_dir.encode(), err, 256)

@@ -1495,6 +1495,7 @@ struct CompactSetup {
     std::vector<size_t> comp_file_sz;
     std::vector<int32_t> col_fixed_h;    // SIMPLE regular columns only
     uint32_t n_cpx = 0;                  // one complex (map<blob,blob>) column, last
+    bool counters = false;               // counter table (every column CounterColumnType)
     bool bti = false;                    // inputs are `da` (trie-indexed)
     std::vector<int32_t> ck_widths;      // per clustering column
     std::vector<int32_t> static_fixed_h; // per static column
@@ -1737,6 +1738,7 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         sch.ck_w = d_ck_w.as<int32_t>();
         sch.n_cols = (uint32_t)col_fixed_h.size();
         sch.n_cpx = su.n_cpx;
+        sch.counters = su.counters ? 1u : 0u;
         DevBuf d_col_fixed;
         d_col_fixed.alloc(col_fixed_h.size() * 4);
         HIP_CHECK(hipMemcpyAsync(d_col_fixed.p, col_fixed_h.data(), col_fixed_h.size() * 4,
@@ -1898,6 +1900,25 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         DevBuf d_stats, d_tomb;
         d_stats.alloc(sizeof(OutStats));
         init_outstats(d_stats, stream);
+        DevBuf d_ctr_arena;
+        if (sch.counters) {
+            // merged-context arena: capacity = total input counter bytes
+            // (any k-way merge result is bounded by the sum of its inputs)
+            DevBuf d_sum;
+            d_sum.alloc(8);
+            HIP_CHECK(hipMemsetAsync(d_sum.p, 0, 8, stream));
+            uint64_t n_cells = total_in_rows * sch.n_cols;
+            if (n_cells) {
+                uint32_t blocks2 = (uint32_t)((n_cells + 255) / 256);
+                hipLaunchKernelGGL(k_sum_vallen, dim3(blocks2), dim3(256), 0, stream,
+                                   in_rows.uc, n_cells, d_sum.as<unsigned long long>());
+            }
+            unsigned long long total_ctr = 0;
+            HIP_CHECK(hipStreamSynchronize(stream));
+            HIP_CHECK(hipMemcpy(&total_ctr, d_sum.p, 8, hipMemcpyDeviceToHost));
+            d_ctr_arena.alloc(total_ctr + 64);
+            sch.ctr_arena = d_ctr_arena.as<uint8_t>();
+        }
         uint32_t tomb_cap = (uint32_t)std::min<uint64_t>(total_out_rows * 2 + n_groups + 1024, 400000000ull);
         d_tomb.alloc((uint64_t)tomb_cap * 4);
         DevBuf d_ov_lo, d_ov_hi, d_ov_ts;
@@ -2311,6 +2332,10 @@ static void compact_setup(const gpuc_job* job, CompactSetup& su, bool preread_fu
                      ct == "org.apache.cassandra.db.marshal.UTF8Type" ||
                      ct == "org.apache.cassandra.db.marshal.AsciiType")
                 su.col_fixed_h.push_back(-1);
+            else if (ct == "org.apache.cassandra.db.marshal.CounterColumnType") {
+                su.col_fixed_h.push_back(-1);
+                su.counters = true;
+            }
             else if (ct == "org.apache.cassandra.db.marshal.MapType(org.apache.cassandra.db.marshal.BytesType,org.apache.cassandra.db.marshal.BytesType)") {
                 // one complex column, and it must be the LAST regular column
                 // (engine layout constraint; the generator names it 'zm' so
@@ -2321,6 +2346,17 @@ static void compact_setup(const gpuc_job* job, CompactSetup& su, bool preread_fu
             } else {
                 throw std::runtime_error("unsupported column type " + ct);
             }
+        }
+        if (su.counters) {
+            // counter tables hold only counter columns (CreateTableStatement
+            // forbids mixing); complex/static columns are rejected with them
+            if (su.col_fixed_h.size() != rcols.size() || su.n_cpx)
+                throw std::runtime_error("counter tables must be all-counter columns");
+            for (auto& cp : rcols)
+                if (cp.second != "org.apache.cassandra.db.marshal.CounterColumnType")
+                    throw std::runtime_error("counter tables must be all-counter columns");
+            if (!su.stats[0].static_cols.empty())
+                throw std::runtime_error("static columns with counters unsupported");
         }
     }
     for (auto& t : su.stats[0].clustering_types) su.ck_widths.push_back(ck_type_width(t));
@@ -3084,6 +3120,28 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
             gp.ttl_pct = spec->ttl_pct;
             gp.complex_pct = spec->complex_pct;
             gp.complex_del_pct = spec->complex_del_pct;
+            gp.counter = spec->counter;
+            if (gp.counter) {
+                if (gp.complex_pct || spec->static_pct || spec->ttl_pct || gp.n_value_cols > 1)
+                    throw std::runtime_error("counter mode: single counter column only");
+                gp.value_len = 274;  // context arena stride (8-shard worst case)
+                // sorted CounterId pool (oracle gen.h gen_counter_id)
+                struct P { uint8_t id[16]; uint8_t idx; };
+                std::vector<P> pool(8);
+                for (uint32_t i = 0; i < 8; i++) {
+                    uint64_t a = splitmix64(0xC0C0C0C0ULL + i), b = splitmix64(0xF00DF00DULL + i);
+                    for (int x = 0; x < 8; x++) pool[i].id[x] = (uint8_t)(a >> (8 * (7 - x)));
+                    for (int x = 0; x < 8; x++) pool[i].id[8 + x] = (uint8_t)(b >> (8 * (7 - x)));
+                    pool[i].idx = (uint8_t)i;
+                }
+                std::sort(pool.begin(), pool.end(), [](const P& x, const P& y) {
+                    return memcmp(x.id, y.id, 16) < 0;
+                });
+                for (int i = 0; i < 8; i++) {
+                    memcpy(gp.ctr_pool[i], pool[i].id, 16);
+                    gp.ctr_pool_idx[i] = pool[i].idx;
+                }
+            }
             if (gp.n_value_cols > 63) throw std::runtime_error("n_value_cols must be 1..63");
             if (gp.ck_text && (uint64_t)gp.clustering_rows * 16 >= 100000000ull)
                 throw std::runtime_error("ck_text needs clustering_rows*16 < 1e8 (8-digit order)");
@@ -3179,7 +3237,9 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
                 ? std::string(dir) + "/da-" + std::to_string(spec->first_generation + s) + "-bti"
                 : std::string(dir) + "/oa-" + std::to_string(spec->first_generation + s) + "-big";
             std::vector<std::pair<bytes, std::string>> cols;
-            if (gp.n_value_cols == 1) {
+            if (gp.counter) {
+                cols.push_back({bytes{'c', 'n', 't'}, "org.apache.cassandra.db.marshal.CounterColumnType"});
+            } else if (gp.n_value_cols == 1) {
                 cols.push_back({bytes{'v', 'a', 'l'}, "org.apache.cassandra.db.marshal.BytesType"});
             } else {
                 for (uint32_t c = 0; c < gp.n_value_cols; c++) {

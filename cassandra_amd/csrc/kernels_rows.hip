@@ -35,6 +35,13 @@ struct SchemaParams {
     uint32_t n_static;       // static columns (0 = schema has no statics)
     const int32_t* static_fixed;
     uint32_t column_index_size;  // promoted-index granularity (64 KiB default)
+    // counter tables (every regular column CounterColumnType): reconcile
+    // merges CounterContexts instead of timestamp resolution; merged
+    // contexts are fresh bytes bump-allocated from ctr_arena (capacity =
+    // total input counter-value bytes; st->cpx_alloc doubles as the bump —
+    // counter schemas exclude complex columns)
+    uint32_t counters;
+    uint8_t* ctr_arena;
 };
 
 // sortable ck encoding: big-endian fixed-width signed value -> flip sign bit
@@ -524,6 +531,19 @@ __global__ void k_merged_hist(const uint8_t* merged_k, uint64_t n, OutStats* st)
         if (sh[i]) atomicAdd(&st->merged_counts[i], (unsigned long long)sh[i]);
 }
 
+// total present-cell value bytes (counter arena sizing)
+__global__ void k_sum_vallen(UnfCols in, uint64_t n_cells, unsigned long long* out) {
+    __shared__ unsigned long long sh;
+    if (threadIdx.x == 0) sh = 0;
+    __syncthreads();
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    unsigned long long v = 0;
+    if (i < n_cells && (in.cell_flags[i] & CELLF_PRESENT)) v = in.val_len[i];
+    if (v) atomicAdd(&sh, v);
+    __syncthreads();
+    if (threadIdx.x == 0 && sh) atomicAdd(out, sh);
+}
+
 __global__ void k_widen_u32(const uint32_t* in, uint64_t* out, uint64_t n) {
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i < n) out[i] = in[i];
@@ -631,6 +651,65 @@ __device__ inline int cmp_values(uint64_t la, uint32_t ll, uint64_t ra, uint32_t
     for (uint32_t x = 0; x < nn; x++)
         if (lp[x] != rp[x]) return lp[x] < rp[x] ? -1 : 1;
     return ll == rl ? 0 : (ll < rl ? -1 : 1);
+}
+
+// ---------------------------------------------------------------------------
+// CounterContext device machinery (db/context/CounterContext.java): walker
+// over (header flags, 16B-id shards) + the k-way merge equivalent of the
+// reference's pairwise chain (equal by commutativity/associativity of the
+// compare() lattice for live clocks; remote clocks are never 0 in practice)
+// ---------------------------------------------------------------------------
+struct CtxSt {
+    const uint8_t* p;
+    uint32_t len, body, off, hoff;
+    bool neg_hdr, g, l;
+    __device__ void update() {
+        g = l = false;
+        if (hoff < body) {
+            int16_t e = (int16_t)((uint16_t)((uint16_t)p[hoff] << 8) | p[hoff + 1]);
+            int32_t idx = (int32_t)((off - body) / 32);
+            if (!neg_hdr && e == (int16_t)(idx + INT16_MIN)) g = true;
+            else if (e == (int16_t)idx) l = true;
+        }
+    }
+    __device__ void init(const uint8_t* q, uint32_t n) {
+        p = q;
+        len = n;
+        int16_t h = n >= 2 ? (int16_t)((uint16_t)((uint16_t)q[0] << 8) | q[1]) : 0;
+        neg_hdr = h < 0;
+        uint32_t hn = (uint32_t)(h < 0 ? -h : h);
+        body = 2 + hn * 2;
+        off = body;
+        hoff = 2;
+        update();
+    }
+    __device__ bool has() const { return off + 32 <= len; }
+    __device__ void next() {
+        if (g || l) hoff += 2;
+        off += 32;
+        update();
+    }
+    __device__ const uint8_t* id() const { return p + off; }
+    __device__ int64_t clock() const {
+        uint64_t v = 0;
+        for (int b = 0; b < 8; b++) v = (v << 8) | p[off + 16 + b];
+        return (int64_t)v;
+    }
+    __device__ int64_t count() const {
+        uint64_t v = 0;
+        for (int b = 0; b < 8; b++) v = (v << 8) | p[off + 24 + b];
+        return (int64_t)v;
+    }
+};
+__device__ inline int ctx_idcmp(const uint8_t* a, const uint8_t* b) {
+    for (int i = 0; i < 16; i++)
+        if (a[i] != b[i]) return a[i] < b[i] ? -1 : 1;
+    return 0;
+}
+// remote shard "left wins" rule (CounterContext.compare, both-remote tail)
+__device__ inline bool ctx_remote_better(int64_t lc, int64_t lk, int64_t rc, int64_t rk) {
+    if (lc == rc) return lk > rk;
+    return (lc >= 0 && rc > 0 && lc >= rc) || (lc < 0 && (rc > 0 || lc < rc));
 }
 
 #define GPUC_MAX_ARITY 64
@@ -1063,7 +1142,154 @@ __global__ void k_reconcile_rows(const MRec* recs, const uint64_t* group_start,
                 else of |= PF_ROW_DEL;
                 if (has_live && lts <= am2) { of &= ~PF_LIVE_TS; lts = NO_TIMESTAMP; lttl = 0; llet = NO_DELETION_TIME; }
                 // per-column Cells.reconcile (Cells.java:145-179) + activeDeletion filter
-                for (uint32_t c = 0; c < NC; c++) {
+                for (uint32_t c = 0; c < NC && sp.counters; c++) {
+                    // Cells.resolveCounter: tombstones beat any live counter
+                    // (then regular rules among tombstones); empty values
+                    // lose; else k-way CounterContext merge into the arena
+                    bool any_tomb = false, any_val = false;
+                    uint32_t nv3 = 0;
+                    uint64_t mem_oc[MA];
+                    #pragma unroll
+                    for (uint32_t m = 0; m < (uint32_t)MA; m++) {
+                        if (m >= k) break;
+                        if (!(members & (1ULL << m))) continue;
+                        uint64_t o = (mb[m] + mpos[m]) * NC + c;
+                        uint8_t f2 = in.cell_flags[o];
+                        if (!(f2 & CELLF_PRESENT)) continue;
+                        if (in.cell_ts[o] <= am2) continue;
+                        if (in.cell_ldt[o] != LDT_NONE_U32) any_tomb = true;
+                        else if (in.val_len[o] > 0) any_val = true;
+                        mem_oc[nv3++] = o;
+                    }
+                    uint64_t ocx = oslot * NC + c;
+                    if (nv3 == 0) { out.cell_flags[ocx] = 0; continue; }
+                    any_cell = true;
+                    if (any_tomb) {
+                        // regular reconcile restricted to the tombstones
+                        bool have = false;
+                        int64_t cts = NO_TIMESTAMP;
+                        uint32_t cldt = LDT_NONE_U32;
+                        uint64_t va = 0;
+                        uint32_t vl = 0;
+                        for (uint32_t m2 = 0; m2 < nv3; m2++) {
+                            uint64_t o = mem_oc[m2];
+                            if (in.cell_ldt[o] == LDT_NONE_U32) continue;
+                            int64_t ts2 = in.cell_ts[o];
+                            uint32_t rl = in.cell_ldt[o];
+                            bool take = !have ||
+                                        (cts != ts2 ? ts2 > cts
+                                                    : (cldt != rl ? ldt_long(rl) > ldt_long(cldt)
+                                                                  : cmp_values(va, vl, in.val_addr[o], in.val_len[o]) < 0));
+                            if (take) { have = true; cts = ts2; cldt = rl; va = in.val_addr[o]; vl = in.val_len[o]; }
+                        }
+                        out.cell_flags[ocx] = CELLF_PRESENT | (vl ? CELLF_HAS_VALUE : 0);
+                        out.cell_ts[ocx] = cts;
+                        out.cell_ldt[ocx] = cldt;
+                        out.cell_ttl[ocx] = 0;
+                        out.val_addr[ocx] = va;
+                        out.val_len[ocx] = vl;
+                        continue;
+                    }
+                    int64_t max_ts = NO_TIMESTAMP;
+                    uint64_t first_nonempty = 0;
+                    uint32_t n_live = 0, total_len = 0;
+                    for (uint32_t m2 = 0; m2 < nv3; m2++) {
+                        uint64_t o = mem_oc[m2];
+                        if (in.cell_ts[o] > max_ts) max_ts = in.cell_ts[o];
+                        if (in.val_len[o] > 0) {
+                            if (!n_live) first_nonempty = o;
+                            mem_oc[n_live++] = o;  // compact non-empty to front
+                            total_len += in.val_len[o];
+                        }
+                    }
+                    out.cell_flags[ocx] = CELLF_PRESENT;
+                    out.cell_ts[ocx] = max_ts;
+                    out.cell_ldt[ocx] = LDT_NONE_U32;
+                    out.cell_ttl[ocx] = 0;
+                    if (!any_val || n_live == 0) {
+                        out.val_addr[ocx] = 0;
+                        out.val_len[ocx] = 0;
+                        continue;
+                    }
+                    if (n_live == 1) {
+                        out.cell_flags[ocx] |= CELLF_HAS_VALUE;
+                        out.val_addr[ocx] = in.val_addr[first_nonempty];
+                        out.val_len[ocx] = in.val_len[first_nonempty];
+                        continue;
+                    }
+                    // k-way context merge, two passes (count, then write)
+                    uint8_t* dst = nullptr;
+                    uint32_t wr = 0, nflag = 0, nsh = 0;
+                    for (int pass = 0; pass < 2; pass++) {
+                        CtxSt st2[MA];
+                        for (uint32_t m2 = 0; m2 < n_live; m2++) {
+                            uint64_t o = mem_oc[m2];
+                            st2[m2].init((const uint8_t*)in.val_addr[o], in.val_len[o]);
+                        }
+                        if (pass == 1) {
+                            uint64_t base = atomicAdd(&st->cpx_alloc, (unsigned long long)(2 + nflag * 2 + nsh * 32));
+                            dst = sp.ctr_arena + base;
+                            dst[0] = (uint8_t)(nflag >> 8);
+                            dst[1] = (uint8_t)nflag;
+                            wr = 2 + nflag * 2;
+                            nflag = 0;
+                            nsh = 0;
+                        }
+                        while (true) {
+                            const uint8_t* minid = nullptr;
+                            for (uint32_t m2 = 0; m2 < n_live; m2++)
+                                if (st2[m2].has() && (!minid || ctx_idcmp(st2[m2].id(), minid) < 0))
+                                    minid = st2[m2].id();
+                            if (!minid) break;
+                            bool hg = false, hl = false;
+                            int64_t bc = 0, bk = 0, suml_c = 0, suml_k = 0, rc = 0, rk = 0;
+                            bool hr = false;
+                            const uint8_t* sid = minid;
+                            for (uint32_t m2 = 0; m2 < n_live; m2++) {
+                                if (!st2[m2].has() || ctx_idcmp(st2[m2].id(), minid) != 0) continue;
+                                int64_t ck3 = st2[m2].clock(), cn3 = st2[m2].count();
+                                if (st2[m2].g) {
+                                    if (!hg || ck3 > bc || (ck3 == bc && cn3 > bk)) { bc = ck3; bk = cn3; }
+                                    hg = true;
+                                } else if (st2[m2].l) {
+                                    hl = true;
+                                    suml_c += ck3;
+                                    suml_k += cn3;
+                                } else {
+                                    if (!hr || ctx_remote_better(ck3, cn3, rc, rk)) { rc = ck3; rk = cn3; }
+                                    hr = true;
+                                }
+                                st2[m2].next();
+                            }
+                            int64_t oc3, ok3;
+                            int role;  // 0 global 1 local 2 remote
+                            if (hg) { oc3 = bc; ok3 = bk; role = 0; }
+                            else if (hl) { oc3 = suml_c; ok3 = suml_k; role = 1; }
+                            else { oc3 = rc; ok3 = rk; role = 2; }
+                            if (pass == 0) {
+                                if (role != 2) nflag++;
+                                nsh++;
+                            } else {
+                                if (role != 2) {
+                                    int16_t e = role == 0 ? (int16_t)((int32_t)nsh + INT16_MIN)
+                                                          : (int16_t)nsh;
+                                    dst[2 + nflag * 2] = (uint8_t)((uint16_t)e >> 8);
+                                    dst[3 + nflag * 2] = (uint8_t)e;
+                                    nflag++;
+                                }
+                                for (int b = 0; b < 16; b++) dst[wr + b] = sid[b];
+                                for (int b = 7; b >= 0; b--) dst[wr + 16 + (7 - b)] = (uint8_t)((uint64_t)oc3 >> (8 * b));
+                                for (int b = 7; b >= 0; b--) dst[wr + 24 + (7 - b)] = (uint8_t)((uint64_t)ok3 >> (8 * b));
+                                wr += 32;
+                                nsh++;
+                            }
+                        }
+                    }
+                    out.cell_flags[ocx] |= CELLF_HAS_VALUE;
+                    out.val_addr[ocx] = (uint64_t)dst;
+                    out.val_len[ocx] = wr;
+                }
+                for (uint32_t c = 0; c < NC && !sp.counters; c++) {
                     int64_t cts = NO_TIMESTAMP;
                     uint32_t cldt = LDT_NONE_U32;
                     int32_t cttl = 0;
@@ -2923,6 +3149,12 @@ struct GenParams2 {
     uint32_t ttl_pct;         // P(live row written expiring) — oracle gen_row_expiring
     uint32_t complex_pct;     // P(live row carries 'zm' map cells) — gen_has_complex
     uint32_t complex_del_pct; // P(those rows also carry a complexDeletion)
+    // counter mode (oracle gen.h gen_ctr_*): one CounterColumnType column;
+    // ctr_pool = the 8 CounterIds in SORTED byte order, ctr_pool_idx their
+    // original indices (role = idx % 3: global/local/remote)
+    uint32_t counter;
+    uint8_t ctr_pool[8][16];
+    uint8_t ctr_pool_idx[8];
     int64_t base_ts, base_ldt;
 };
 
@@ -3171,8 +3403,55 @@ __global__ void k_gen_fill2(GenParams2 gp, const MRec* sorted, const uint64_t* i
                 out.cell_ts[oc] = ts;
                 out.cell_ldt[oc] = expg ? (uint32_t)elet : LDT_NONE_U32;
                 out.cell_ttl[oc] = expg ? ettl : 0;
-                out.val_addr[oc] = (uint64_t)(values + oc * (uint64_t)gp.value_len);
-                out.val_len[oc] = gp.value_len;
+                uint8_t* vp = values + oc * (uint64_t)gp.value_len;
+                out.val_addr[oc] = (uint64_t)vp;
+                if (gp.counter) {
+                    // build the CounterContext in the value slot (oracle
+                    // make_ctx mirror); stride gp.value_len >= 274
+                    uint64_t cid = gp.clustering_rows ? (id ^ ((uint64_t)rowj << 20)) : id;
+                    uint8_t body[8 * 32];
+                    int16_t elts[8];
+                    uint32_t nb = 0, ne = 0, shard_no = 0;
+                    for (int pi2 = 0; pi2 < 8; pi2++) {
+                        uint32_t idx = gp.ctr_pool_idx[pi2];
+                        if (splitmix64(gp.seed ^ 0xC717C717ULL ^ cid ^ ((uint64_t)gp.sst << 32) ^
+                                       (uint64_t)(idx + 1) * 37) % 100 >= 55) continue;
+                        int role = idx % 3;
+                        uint64_t r2 = splitmix64(gp.seed ^ 0xC10CULL ^ cid ^ (uint64_t)(idx + 1) * 131 ^
+                                                 ((uint64_t)gp.sst << 40));
+                        int64_t ck3;
+                        if (role == 0) {
+                            uint64_t base2 = splitmix64(gp.seed ^ 0x610BULL ^ cid ^ idx);
+                            ck3 = (int64_t)(1000 + (cid % 2 ? base2 % 50 : (base2 + gp.sst) % 50));
+                        } else if (role == 1) {
+                            ck3 = (int64_t)(1 + r2 % 5);
+                        } else if (idx == 7 && (cid % 5) == 0) {
+                            ck3 = -(int64_t)(1 + r2 % 90);
+                        } else {
+                            ck3 = (int64_t)(1 + r2 % 99);
+                        }
+                        int64_t cn3 = (int64_t)(splitmix64(gp.seed ^ 0xC0117ULL ^ cid ^
+                                                           (uint64_t)(idx + 1) * 17 ^
+                                                           ((uint64_t)gp.sst << 24)) % 1000) - 100;
+                        if (role == 0) elts[ne++] = (int16_t)((int32_t)shard_no + INT16_MIN);
+                        else if (role == 1) elts[ne++] = (int16_t)shard_no;
+                        for (int b = 0; b < 16; b++) body[nb + b] = gp.ctr_pool[pi2][b];
+                        for (int b = 7; b >= 0; b--) body[nb + 16 + (7 - b)] = (uint8_t)((uint64_t)ck3 >> (8 * b));
+                        for (int b = 7; b >= 0; b--) body[nb + 24 + (7 - b)] = (uint8_t)((uint64_t)cn3 >> (8 * b));
+                        nb += 32;
+                        shard_no++;
+                    }
+                    vp[0] = (uint8_t)(ne >> 8);
+                    vp[1] = (uint8_t)ne;
+                    for (uint32_t e2 = 0; e2 < ne; e2++) {
+                        vp[2 + e2 * 2] = (uint8_t)((uint16_t)elts[e2] >> 8);
+                        vp[3 + e2 * 2] = (uint8_t)elts[e2];
+                    }
+                    for (uint32_t b = 0; b < nb; b++) vp[2 + ne * 2 + b] = body[b];
+                    out.val_len[oc] = 2 + ne * 2 + nb;
+                } else {
+                    out.val_len[oc] = gp.value_len;
+                }
             }
             // complex column 'zm': dedup+sorted map cells (oracle put_complex)
             if (gp.complex_pct && gen2_has_cpx(gp, id, rowj)) {
@@ -3270,6 +3549,7 @@ __global__ void k_gen_values2(GenParams2 gp, const MRec* sorted, const uint64_t*
                               const uint64_t* row_base) {
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n) return;
+    if (gp.counter) return;  // counter contexts written by k_gen_fill2
     uint64_t id = ids[sorted[i].idx];
     if (gp.static_pct && (op.st.cell_flags[i] & CELLF_HAS_VALUE)) {
         uint64_t vid = id ^ 0xABCDEF57ULL;

@@ -180,6 +180,8 @@ typedef struct gpuc_gen_spec {
                                    complexDeletion */
     int32_t bti;                /* 1: write the `da` (trie-indexed) component
                                    set (Partitions.db/Rows.db, BtiFormat.md) */
+    int32_t counter;            /* 1: one CounterColumnType column with
+                                   synthetic CounterContexts (CounterContext.java) */
 } gpuc_gen_spec;
 
 int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* error, size_t error_len);

@@ -728,6 +728,44 @@ def test_cross_feature_combinations(ca, oracle_bin, tmp_path):
     _assert_dirs_equal(f"{dd}/oa-60-big", f"{dd}/oa-80-big")
 
 
+def test_counter_columns_pipeline(ca, oracle_bin, tmp_path):
+    """Counter columns on GPU (CounterColumnType): the k-way CounterContext
+    merge in reconcile (provably equal to the reference's pairwise chain —
+    commutative lattice; oracle transcribes pairwise) + Cells.resolveCounter
+    tombstone/empty rules, byte-identical to the oracle across 3-way merges,
+    re-compaction, purge of counter tombstones, and writer parity. Cell
+    format pinned by the reference's legacy_oa_*_counter fixtures."""
+    d = str(tmp_path)
+    _oracle_gen(d, seed=99, n=3, rows=1200, overlap=30, tomb=10, counter=1)
+    ins = [f"{d}/oa-{g}-big" for g in (1, 2, 3)]
+    ca.compact(ins, f"{d}/oa-60-big")
+    _oracle_compact(f"{d}/oa-80-big", ins)
+    _assert_dirs_equal(f"{d}/oa-60-big", f"{d}/oa-80-big")
+    ca.verify(f"{d}/oa-60-big")
+    # re-compaction of a compaction output + purge of counter tombstones
+    ca.compact([f"{d}/oa-60-big", ins[2]], f"{d}/oa-61-big",
+               now_sec=1800000000, gc_before=1800000000)
+    _oracle_compact(f"{d}/oa-81-big", [f"{d}/oa-80-big", ins[2]],
+                    now=1800000000, gcbefore=1800000000)
+    _assert_dirs_equal(f"{d}/oa-61-big", f"{d}/oa-81-big")
+    # writer parity: generate(counter=True) == oracle gen counter=1
+    dg, do = d + "/g", d + "/o"
+    os.makedirs(dg), os.makedirs(do)
+    ca.generate(dg, seed=99, n_sstables=2, rows_per_sstable=800, overlap_pct=30,
+                tombstone_pct=10, counter=True)
+    _oracle_gen(do, seed=99, n=2, rows=800, overlap=30, tomb=10, counter=1)
+    for g in (1, 2):
+        _assert_dirs_equal(f"{dg}/oa-{g}-big", f"{do}/oa-{g}-big")
+    # wide counter partitions (the clust-counter fixture shape)
+    dw = d + "/w"
+    os.makedirs(dw)
+    _oracle_gen(dw, seed=98, n=2, rows=40, crows=50, overlap=25, tomb=10, counter=1)
+    wins = [f"{dw}/oa-{g}-big" for g in (1, 2)]
+    ca.compact(wins, f"{dw}/oa-60-big")
+    _oracle_compact(f"{dw}/oa-80-big", wins)
+    _assert_dirs_equal(f"{dw}/oa-60-big", f"{dw}/oa-80-big")
+
+
 def test_cancellation(ca, oracle_bin, tmp_path):
     """Cooperative cancel (CompactionIterator.isStopRequested): a set
     cancel_flag aborts the task with GPUC_ERR_CANCELLED; a zero flag is
