@@ -347,6 +347,35 @@ def compact(
     }
 
 
+def validate(input_bases, out_path, now_sec=1800000000, gc_before=INT64_MIN,
+             overlaps=None, device=0):
+    """VALIDATION compaction (CompactionManager.doValidationCompaction +
+    Validator.rowHash): merge+purge, then per-partition repair digests
+    written to out_path as (token i64 BE + 32-byte hash) records."""
+    lib = load_library()
+    job = GpucJob()
+    arr = (ctypes.c_char_p * len(input_bases))(*[b.encode() for b in input_bases])
+    job.input_bases = arr
+    job.n_inputs = len(input_bases)
+    job.output_base = b"/unused"
+    job.now_sec = now_sec
+    job.gc_before = gc_before
+    if overlaps:
+        ovr = (GpucPurgeRange * len(overlaps))()
+        for i, (lo, hi, ts) in enumerate(overlaps):
+            ovr[i].token_lo, ovr[i].token_hi, ovr[i].min_timestamp = lo, hi, ts
+        job.overlaps = ovr
+        job.n_overlaps = len(overlaps)
+    job.device = device
+    job.n_output_shards = 1
+    n = ctypes.c_uint64(0)
+    err = ctypes.create_string_buffer(256)
+    rc = lib.gpuc_validate(ctypes.byref(job), out_path.encode(), ctypes.byref(n), err, 256)
+    if rc != 0:
+        raise GpuCompactError(f"gpuc_validate rc={rc}: {err.value.decode(errors='replace')}")
+    return n.value
+
+
 def generate(
     out_dir,
     snappy=False,

@@ -227,4 +227,81 @@ GPUC_HD inline uint64_t feistel_perm(uint64_t seed, uint64_t universe, uint64_t 
     return x;
 }
 
+
+// Streaming MurmurHash3 x64_128 with integer seed (Guava Murmur3_128Hasher
+// semantics: h1 = h2 = seed; finalize XORs total length; output h1||h2
+// little-endian). The repair Validator's digest is
+// concat(murmur3_128(1000), murmur3_128(2000)) over the same byte stream
+// (db/Digest.java:53-59).
+struct M3Stream {
+    uint64_t h1, h2, len;
+    uint8_t buf[16];
+    uint32_t n;
+    GPUC_HD void init(int64_t seed) {
+        h1 = (uint64_t)seed;
+        h2 = (uint64_t)seed;
+        len = 0;
+        n = 0;
+    }
+    GPUC_HD static uint64_t rotl(uint64_t x, int r) { return (x << r) | (x >> (64 - r)); }
+    GPUC_HD void block(const uint8_t* p) {
+        const uint64_t c1 = 0x87c37b91114253d5ULL, c2 = 0x4cf5ad432745937fULL;
+        uint64_t k1, k2;
+        memcpy(&k1, p, 8);
+        memcpy(&k2, p + 8, 8);
+        k1 *= c1; k1 = rotl(k1, 31); k1 *= c2; h1 ^= k1;
+        h1 = rotl(h1, 27); h1 += h2; h1 = h1 * 5 + 0x52dce729;
+        k2 *= c2; k2 = rotl(k2, 33); k2 *= c1; h2 ^= k2;
+        h2 = rotl(h2, 31); h2 += h1; h2 = h2 * 5 + 0x38495ab5;
+    }
+    GPUC_HD void put(const uint8_t* p, uint64_t m) {
+        len += m;
+        if (n) {
+            while (m && n < 16) { buf[n++] = *p++; m--; }
+            if (n == 16) { block(buf); n = 0; }
+        }
+        while (m >= 16) { block(p); p += 16; m -= 16; }
+        while (m) { buf[n++] = *p++; m--; }
+    }
+    GPUC_HD void put_u8(uint8_t v) { put(&v, 1); }
+    GPUC_HD void put_i32be(int32_t v) {  // Digest.updateWithInt
+        uint8_t b[4];
+        for (int i = 0; i < 4; i++) b[i] = (uint8_t)((uint32_t)v >> (8 * (3 - i)));
+        put(b, 4);
+    }
+    GPUC_HD void put_i64be(int64_t v) {  // Digest.updateWithLong
+        uint8_t b[8];
+        for (int i = 0; i < 8; i++) b[i] = (uint8_t)((uint64_t)v >> (8 * (7 - i)));
+        put(b, 8);
+    }
+    GPUC_HD void put_bool(bool v) { put_u8(v ? 0 : 1); }  // updateWithBoolean INVERTS
+    GPUC_HD static uint64_t fmix(uint64_t k) {
+        k ^= k >> 33;
+        k *= 0xff51afd7ed558ccdULL;
+        k ^= k >> 33;
+        k *= 0xc4ceb9fe1a85ec53ULL;
+        k ^= k >> 33;
+        return k;
+    }
+    GPUC_HD void final16(uint8_t out[16]) {
+        const uint64_t c1 = 0x87c37b91114253d5ULL, c2 = 0x4cf5ad432745937fULL;
+        uint64_t k1 = 0, k2 = 0;
+        if (n > 8) {
+            for (uint32_t i = n; i > 8;) { i--; k2 = (k2 << 8) | buf[i]; }
+            k2 *= c2; k2 = rotl(k2, 33); k2 *= c1; h2 ^= k2;
+        }
+        if (n > 0) {
+            uint32_t m = n > 8 ? 8 : n;
+            for (uint32_t i = m; i > 0;) { i--; k1 = (k1 << 8) | buf[i]; }
+            k1 *= c1; k1 = rotl(k1, 31); k1 *= c2; h1 ^= k1;
+        }
+        h1 ^= len; h2 ^= len;
+        h1 += h2; h2 += h1;
+        h1 = fmix(h1); h2 = fmix(h2);
+        h1 += h2; h2 += h1;
+        memcpy(out, &h1, 8);       // little-endian out (Guava asBytes)
+        memcpy(out + 8, &h2, 8);
+    }
+};
+
 }  // namespace gpuc

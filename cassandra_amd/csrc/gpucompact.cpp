@@ -91,6 +91,9 @@ static PinnedSlot g_pin_in[2][64], g_pin_out[2][4];
 
 // GPUC_TRACE=1: wall-clock checkpoints on stderr (host-gap hunting)
 static bool g_trace = getenv("GPUC_TRACE") != nullptr;
+// validation mode: set by gpuc_validate around its gpuc_compact-shaped run
+static thread_local const char* g_validate_out = nullptr;
+static thread_local uint64_t g_validate_count = 0;
 static double g_trace_t0 = 0;
 static double trace_wall() {
     struct timespec ts; clock_gettime(CLOCK_MONOTONIC, &ts);
@@ -2153,6 +2156,66 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
             sp2.hs.min_ldt = min_ldt == NO_DELETION_TIME ? DELETION_TIME_EPOCH : min_ldt;
             sp2.hs.min_ttl = min_ttl == INT32_MAX ? 0 : min_ttl;
         }
+        if (g_validate_out) {
+            // VALIDATION epilogue: per-partition repair digests instead of a
+            // written sstable (k_validate_digest; Validator.rowHash)
+            bytes names;
+            std::vector<uint32_t> noff;
+            for (auto& [nm, t2] : stats[0].regular_cols) {
+                (void)t2;
+                noff.push_back((uint32_t)names.size());
+                names.insert(names.end(), nm.begin(), nm.end());
+            }
+            for (auto& [nm, t2] : stats[0].static_cols) {
+                (void)t2;
+                noff.push_back((uint32_t)names.size());
+                names.insert(names.end(), nm.begin(), nm.end());
+            }
+            noff.push_back((uint32_t)names.size());
+            DevBuf d_names, d_noff, d_hash;
+            d_names.alloc(names.size() + 8);
+            d_noff.alloc(noff.size() * 4);
+            d_hash.alloc(n_groups * 32 + 32);
+            HIP_CHECK(hipMemcpyAsync(d_names.p, names.data(), names.size(), hipMemcpyHostToDevice, stream));
+            HIP_CHECK(hipMemcpyAsync(d_noff.p, noff.data(), noff.size() * 4, hipMemcpyHostToDevice, stream));
+            ValidateParams vpar{};
+            vpar.names = d_names.as<uint8_t>();
+            vpar.name_off = d_noff.as<uint32_t>();
+            vpar.n_reg = (uint32_t)stats[0].regular_cols.size();
+            vpar.n_static = (uint32_t)stats[0].static_cols.size();
+            vpar.counters = sch.counters;
+            vpar.n_cpx = sch.n_cpx;
+            vpar.n_ck = sch.n_ck;
+            vpar.ck_w = sch.ck_w;
+            {
+                uint32_t blocks2 = (uint32_t)((n_groups + 255) / 256);
+                hipLaunchKernelGGL(k_validate_digest, dim3(blocks2), dim3(256), 0, stream, opb.op,
+                                   rows_for_writer->uc, n_groups, vpar, d_hash.as<uint8_t>());
+            }
+            std::vector<uint8_t> h_hash(n_groups * 32);
+            std::vector<int64_t> h_tok(n_groups);
+            std::vector<uint8_t> h_keep(n_groups);
+            HIP_CHECK(hipStreamSynchronize(stream));
+            HIP_CHECK(hipMemcpy(h_hash.data(), d_hash.p, n_groups * 32, hipMemcpyDeviceToHost));
+            HIP_CHECK(hipMemcpy(h_tok.data(), opb.op.token, n_groups * 8, hipMemcpyDeviceToHost));
+            HIP_CHECK(hipMemcpy(h_keep.data(), opb.op.keep, n_groups, hipMemcpyDeviceToHost));
+            bytes outb;
+            uint64_t cnt = 0;
+            for (uint64_t g2 = 0; g2 < n_groups; g2++) {
+                if (!h_keep[g2]) continue;
+                for (int b2 = 7; b2 >= 0; b2--) outb.push_back((uint8_t)((uint64_t)h_tok[g2] >> (8 * b2)));
+                outb.insert(outb.end(), h_hash.begin() + g2 * 32, h_hash.begin() + g2 * 32 + 32);
+                cnt++;
+            }
+            write_file(g_validate_out, outb.data(), outb.size());
+            g_validate_count = cnt;
+            {
+                std::lock_guard<std::mutex> gl(res_mu);
+                res->partitions_out += cnt;
+            }
+            HIP_CHECK(hipStreamDestroy(stream));
+            return;
+        }
         WriteDeviceOut w = write_sstable_device(opb, *rows_for_writer, n_groups, sp2, d_stats, d_tomb,
                                                 tomb_cap, out_base_str, stats[0].key_type,
                                                 stats[0].clustering_types, stats[0].regular_cols,
@@ -3082,6 +3145,22 @@ extern "C" int gpuc_flush(const gpuc_flush_rows* rows, const char* output_base,
         set_err(error, error_len, e.what());
         return GPUC_ERR_INTERNAL;
     }
+}
+
+extern "C" int gpuc_validate(const gpuc_job* job, const char* out_path, uint64_t* n_partitions,
+                             char* error, size_t error_len) {
+    if (job->n_output_shards > 1 || job->n_tomb_sources > 0 || job->n_keep_ranges > 0) {
+        set_err(error, error_len, "validation: sharding/tomb-source/keep-range modes unsupported");
+        return GPUC_ERR_UNSUPPORTED;
+    }
+    gpuc_result res{};
+    g_validate_out = out_path;
+    g_validate_count = 0;
+    int rc = gpuc_compact(job, &res);
+    g_validate_out = nullptr;
+    if (rc == GPUC_OK && n_partitions) *n_partitions = g_validate_count;
+    if (rc != GPUC_OK) set_err(error, error_len, res.error);
+    return rc;
 }
 
 extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* error, size_t error_len) {

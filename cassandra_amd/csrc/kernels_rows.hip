@@ -531,6 +531,137 @@ __global__ void k_merged_hist(const uint8_t* merged_k, uint64_t n, OutStats* st)
         if (sh[i]) atomicAdd(&st->merged_counts[i], (unsigned long long)sh[i]);
 }
 
+// ---------------------------------------------------------------------------
+// repair-validation digests (Validator.rowHash + UnfilteredRowIterators/
+// Rows/Cells/DeletionTime/ClusteringPrefix .digest): per kept output
+// partition, concat(murmur3_128(1000), murmur3_128(2000)) over the field
+// stream — big-endian ints/longs, inverted booleans, counter contexts
+// body-only, cell localDeletionTime excluded. Oracle mirror:
+// validator_digest (oracle/src/compact.cpp).
+// ---------------------------------------------------------------------------
+struct ValidateParams {
+    const uint8_t* names;      // packed regular-column then static-column names
+    const uint32_t* name_off;  // n_reg + n_static + 1 offsets
+    uint32_t n_reg, n_static;
+    uint32_t counters, n_cpx;
+    uint32_t n_ck;
+    const int32_t* ck_w;
+};
+
+__global__ void k_validate_digest(OutParts op, UnfCols out, uint64_t n, ValidateParams vp,
+                                  uint8_t* hashes) {
+    uint64_t g = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (g >= n || !op.keep[g]) return;
+    M3Stream a, b;
+    a.init(1000);
+    b.init(2000);
+    auto put = [&](const uint8_t* d, uint64_t m) { a.put(d, m); b.put(d, m); };
+    auto u8 = [&](uint8_t v) { a.put_u8(v); b.put_u8(v); };
+    auto i32 = [&](int32_t v) { a.put_i32be(v); b.put_i32be(v); };
+    auto i64 = [&](int64_t v) { a.put_i64be(v); b.put_i64be(v); };
+    auto bl = [&](bool v) { a.put_bool(v); b.put_bool(v); };
+    auto put_ck_comp = [&](uint64_t oc) {
+        if (out.ck_addr[oc]) {
+            put((const uint8_t*)out.ck_addr[oc], out.ck_len[oc]);
+        } else {
+            int w = vp.ck_w[oc % (vp.n_ck ? vp.n_ck : 1)];
+            uint64_t v = out.ck[oc] ^ (1ull << (8 * w - 1));
+            uint8_t bts[8];
+            for (int i = 0; i < w; i++) bts[i] = (uint8_t)(v >> (8 * (w - 1 - i)));
+            put(bts, w);
+        }
+    };
+    auto put_cell_fields = [&](int64_t ts, int32_t ttl, const uint8_t* val, uint32_t vlen,
+                               bool counter_cell, const uint8_t* path, uint32_t plen) {
+        if (counter_cell) {
+            if (vlen >= 2) {
+                int16_t hn = (int16_t)((uint16_t)((uint16_t)val[0] << 8) | val[1]);
+                uint32_t hl = 2 + (uint32_t)(hn < 0 ? -hn : hn) * 2;
+                if (vlen > hl) put(val + hl, vlen - hl);
+            }
+        } else {
+            put(val, vlen);
+        }
+        i64(ts);
+        i32(ttl);
+        bl(counter_cell);
+        if (path) put(path, plen);
+    };
+    // partition key + deletion + column names + reverse flag
+    put((const uint8_t*)op.key_addr[g], op.klen[g]);
+    i64(op.pdel_mfda[g]);
+    for (uint32_t c = 0; c < vp.n_reg; c++)
+        put(vp.names + vp.name_off[c], vp.name_off[c + 1] - vp.name_off[c]);
+    bool static_present = vp.n_static && (op.st.flags[g] & PF_HAS_ROW);
+    if (static_present)
+        for (uint32_t c = vp.n_reg; c < vp.n_reg + vp.n_static; c++)
+            put(vp.names + vp.name_off[c], vp.name_off[c + 1] - vp.name_off[c]);
+    bl(false);  // isReverseOrder
+    // static row (EMPTY_STATIC_ROW when absent)
+    {
+        u8(0);
+        u8(3);  // STATIC kind
+        if (static_present) {
+            i64(op.st.rdel_mfda[g]);
+            bl(false);
+            i64((op.st.flags[g] & PF_LIVE_TS) ? op.st.live_ts[g] : NO_TIMESTAMP);
+            for (uint32_t c = 0; c < vp.n_static; c++) {
+                uint64_t oc = g * vp.n_static + c;
+                if (!(op.st.cell_flags[oc] & CELLF_PRESENT)) continue;
+                put_cell_fields(op.st.cell_ts[oc], op.st.cell_ttl[oc],
+                                (const uint8_t*)op.st.val_addr[oc], op.st.val_len[oc], false,
+                                nullptr, 0);
+            }
+        } else {
+            i64(INT64_MIN);
+            bl(false);
+            i64(NO_TIMESTAMP);
+        }
+    }
+    // unfiltereds
+    uint64_t rb = op.row_base[g];
+    for (uint32_t j = 0; j < op.row_count[g]; j++) {
+        uint64_t o = rb + j;
+        uint8_t kind = out.rkind[o];
+        if (kind == BK_CLUSTERING) {
+            u8(0);
+            for (uint32_t c = 0; c < vp.n_ck; c++) put_ck_comp(o * vp.n_ck + c);
+            u8(4);
+            uint8_t f = out.flags[o];
+            i64((f & PF_ROW_DEL) ? out.rdel_mfda[o] : INT64_MIN);
+            bl(false);
+            i64((f & PF_LIVE_TS) ? out.live_ts[o] : NO_TIMESTAMP);
+            for (uint32_t c = 0; c < vp.n_reg - vp.n_cpx; c++) {
+                uint64_t oc = o * (vp.n_reg - vp.n_cpx) + c;
+                if (!(out.cell_flags[oc] & CELLF_PRESENT)) continue;
+                put_cell_fields(out.cell_ts[oc], out.cell_ttl[oc],
+                                (const uint8_t*)out.val_addr[oc], out.val_len[oc],
+                                vp.counters && out.cell_ldt[oc] == LDT_NONE_U32, nullptr, 0);
+            }
+            if (vp.n_cpx && (f & PF_HAS_CPX)) {
+                bool del_live = out.cpx_del_mfda[o] == INT64_MIN && out.cpx_del_ldt[o] == LDT_NONE_U32;
+                if (!del_live) i64(out.cpx_del_mfda[o]);
+                uint64_t si = out.cpx_start[o];
+                for (uint32_t e = 0; e < out.cpx_count[o]; e++) {
+                    uint64_t xe = si + e;
+                    put_cell_fields(out.cpx.ts[xe], out.cpx.ttl[xe],
+                                    (const uint8_t*)out.cpx.val_addr[xe], out.cpx.val_len[xe],
+                                    false, (const uint8_t*)out.cpx.path_addr[xe],
+                                    out.cpx.path_len[xe]);
+                }
+            }
+        } else {
+            u8(1);  // RANGE_TOMBSTONE_MARKER
+            for (uint32_t c = 0; c < out.ck_count[o]; c++) put_ck_comp(o * vp.n_ck + c);
+            u8(kind);
+            i64(out.rdel_mfda[o]);
+            if (bk_is_boundary(kind)) i64(out.start_mfda[o]);
+        }
+    }
+    a.final16(hashes + g * 32);
+    b.final16(hashes + g * 32 + 16);
+}
+
 // total present-cell value bytes (counter arena sizing)
 __global__ void k_sum_vallen(UnfCols in, uint64_t n_cells, unsigned long long* out) {
     __shared__ unsigned long long sh;

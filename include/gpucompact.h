@@ -186,6 +186,17 @@ typedef struct gpuc_gen_spec {
 
 int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* error, size_t error_len);
 
+/* VALIDATION compaction (CompactionManager.doValidationCompaction +
+ * repair/Validator.rowHash): merge + purge the inputs exactly as a
+ * compaction would, then write per-partition repair digests — records of
+ * (Murmur3 token, int64 BE) + 32-byte concat(murmur3_128(1000),
+ * murmur3_128(2000)) hash (db/Digest.java:53-59) — to out_path instead of
+ * an output sstable. The host feeds these to its MerkleTree
+ * (Validator.add). Uses the same gpuc_job fields (now/gcBefore/overlaps);
+ * sharding/tombstone-source/keep-range modes are rejected. */
+int gpuc_validate(const gpuc_job* job, const char* out_path, uint64_t* n_partitions,
+                  char* error, size_t error_len);
+
 /* Memtable flush (the write-path seed): n unsorted UNIQUE-key rows ->
  * token-sorted on device -> one complete `oa` sstable under output_base.
  * Schema: `pk blob PRIMARY KEY, val blob`. values[i] == NULL makes row i a

@@ -71,6 +71,90 @@ static const Cell& reconcile_cells(const Cell& left, const Cell& right) {
 }
 
 // ---------------------------------------------------------------------------
+// Repair-validation digest (repair/Validator.rowHash:207-216 +
+// UnfilteredRowIterators.digest + Rows/Cells/DeletionTime/ClusteringPrefix
+// .digest): concat(murmur3_128(seed 1000), murmur3_128(seed 2000)) over the
+// partition's field stream — big-endian ints/longs, INVERTED booleans
+// (Digest.updateWithBoolean), counter contexts digested body-only
+// (Digest.updateWithCounterContext skips the header), cell
+// localDeletionTime deliberately excluded (DeletionTime.digest).
+// ---------------------------------------------------------------------------
+void validator_digest(const Partition& p, const Header& h, uint8_t out[32]) {
+    M3Stream a, b;
+    a.init(1000);
+    b.init(2000);
+    auto put = [&](const uint8_t* d, size_t n) { a.put(d, n); b.put(d, n); };
+    auto u8 = [&](uint8_t v) { a.put_u8(v); b.put_u8(v); };
+    auto i32 = [&](int32_t v) { a.put_i32be(v); b.put_i32be(v); };
+    auto i64 = [&](int64_t v) { a.put_i64be(v); b.put_i64be(v); };
+    auto bl = [&](bool v) { a.put_bool(v); b.put_bool(v); };
+    auto put_cell = [&](const Cell& c, bool counter_col, const bytes* path) {
+        bool is_counter_cell = counter_col && c.ldt == LDT_NONE_U32;
+        if (is_counter_cell) {
+            if (!c.value.empty()) {
+                int16_t hn = (int16_t)((c.value[0] << 8) | c.value[1]);
+                size_t hl = 2 + (size_t)(hn < 0 ? -hn : hn) * 2;
+                if (c.value.size() > hl) put(c.value.data() + hl, c.value.size() - hl);
+            }
+        } else {
+            put(c.value.data(), c.value.size());
+        }
+        i64(c.ts);
+        i32(c.ttl == NO_TTL ? 0 : c.ttl);
+        bl(is_counter_cell);
+        if (path) put(path->data(), path->size());
+    };
+    auto digest_row = [&](const Row& r, bool is_static) {
+        u8(0);  // Unfiltered.Kind.ROW
+        if (!is_static)
+            for (auto& cv : r.clustering) put(cv.v.data(), cv.v.size());
+        u8(is_static ? 3 : 4);  // ClusteringPrefix.Kind ordinal
+        i64(r.del.mfda);
+        bl(false);  // Row.Deletion.isShadowable
+        i64(r.live.ts);
+        const auto& cols = is_static ? h.static_cols : h.regular_cols;
+        for (size_t ci = 0; ci < cols.size(); ci++) {
+            if (is_complex_type(cols[ci].second)) {
+                if (ci >= r.complex.size() || !r.complex[ci]) continue;
+                const ComplexData& cd = *r.complex[ci];
+                if (!cd.del.live()) i64(cd.del.mfda);
+                for (const Cell& c : cd.cells) put_cell(c, false, &c.path);
+            } else if (ci < r.cells.size() && r.cells[ci]) {
+                put_cell(*r.cells[ci], is_counter_type(cols[ci].second), nullptr);
+            }
+        }
+    };
+    put(p.key.data(), p.key.size());
+    i64(p.del.mfda);
+    for (auto& [nm, t2] : h.regular_cols) { (void)t2; put(nm.data(), nm.size()); }
+    bool static_present = h.has_static() && !row_is_empty(p.static_row);
+    if (static_present)
+        for (auto& [nm, t2] : h.static_cols) { (void)t2; put(nm.data(), nm.size()); }
+    bl(false);  // isReverseOrder
+    if (static_present) {
+        Row sr = p.static_row;
+        digest_row(sr, true);
+    } else {
+        Row empty;  // Rows.EMPTY_STATIC_ROW
+        digest_row(empty, true);
+    }
+    for (const Unfiltered& u : p.items) {
+        if (u.kind == Unfiltered::ROW) {
+            digest_row(u.row, false);
+        } else {
+            const Marker& m = u.marker;
+            u8(1);  // Unfiltered.Kind.RANGE_TOMBSTONE_MARKER
+            for (auto& cv : m.values) put(cv.v.data(), cv.v.size());
+            u8((uint8_t)m.kind);
+            i64(m.end_dt.mfda);
+            if (m.boundary()) i64(m.start_dt.mfda);
+        }
+    }
+    a.final16(out);
+    b.final16(out + 16);
+}
+
+// ---------------------------------------------------------------------------
 // CounterContext (db/context/CounterContext.java): header of global/local
 // element flags + (CounterId[16], clock i64, count i64) shards in unsigned
 // id order. merge() transcribed pairwise (compare(): global beats all with

@@ -52,6 +52,9 @@ CompactionResult compact(const CompactionJob& job);
 // merge ≤k versions of one partition (UnfilteredRowIterators.merge semantics)
 Partition merge_partition_versions(const std::vector<const Partition*>& versions, const Header& h);
 // purge one partition in place; returns false if empty post-purge (drop it)
+// repair-validation digest (Validator.rowHash): 32 bytes per partition
+void validator_digest(const Partition& p, const Header& h, uint8_t out[32]);
+
 bool purge_partition(Partition& p, int64_t now_sec, int64_t gc_before, bool never_purge,
                      const std::vector<PurgeRange>& overlaps, bool enforce_strict_liveness);
 int compare_clustering_prefix(const Header& h, BoundKind ka, const Clustering& a,

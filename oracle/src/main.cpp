@@ -372,6 +372,28 @@ int main(int argc, char** argv) {
         if (cmd == "dump") return cmd_dump(pos.at(0));
         if (cmd == "roundtrip") return cmd_roundtrip(pos.at(0));
         if (cmd == "rewrite") return cmd_rewrite(pos.at(0), pos.at(1));
+        if (cmd == "validate") {
+            // validate <outfile> <inputs...> [now= gcbefore= ...]: the
+            // VALIDATION compaction epilogue (CompactionManager.doValidation
+            // + Validator.rowHash): merge+purge, then per-partition repair
+            // digests written as (token i64 BE, 32-byte hash) records
+            CompactionJob job;
+            for (size_t i2 = 1; i2 < pos.size(); i2++)
+                job.inputs.push_back(read_sstable(pos[i2], true));
+            job.now_sec = kv.count("now") ? strtoll(kv["now"].c_str(), nullptr, 10) : 1800000000LL;
+            job.gc_before = kv.count("gcbefore") ? strtoll(kv["gcbefore"].c_str(), nullptr, 10) : INT64_MIN;
+            CompactionResult res = compact(job);
+            bytes outb;
+            for (const Partition& p2 : res.out.parts) {
+                for (int b = 7; b >= 0; b--) outb.push_back((uint8_t)((uint64_t)p2.token >> (8 * b)));
+                uint8_t hsh[32];
+                validator_digest(p2, res.out.header, hsh);
+                outb.insert(outb.end(), hsh, hsh + 32);
+            }
+            write_file(pos.at(0), outb);
+            printf("validated %zu partitions\n", res.out.parts.size());
+            return 0;
+        }
         if (cmd == "btidump") {
             // BTI (da) Partitions.db [+ Rows.db] dump — reader scaffolding
             bytes pf = read_file(pos.at(0) + "-Partitions.db");

@@ -766,6 +766,40 @@ def test_counter_columns_pipeline(ca, oracle_bin, tmp_path):
     _assert_dirs_equal(f"{dw}/oa-60-big", f"{dw}/oa-80-big")
 
 
+def test_validation_compaction(ca, oracle_bin, tmp_path):
+    """VALIDATION compaction (the repair digest epilogue — the last §8(f)3
+    piece): gpuc_validate merges+purges like a compaction and emits per-
+    partition (token, 32-byte concat-murmur3_128(1000|2000)) repair digests
+    (Validator.rowHash, Digest.forValidator), byte-identical to the oracle's
+    transcription across simple, wide+tombstone, complex, TTL and counter
+    tables."""
+    d = str(tmp_path)
+    cases = [
+        ("s", dict(seed=5, n=3, rows=900, vlen=100, overlap=25, tomb=15, pdel=2), {}),
+        ("w", dict(seed=6, n=2, rows=40, crows=60, vlen=120, rtomb=25, tomb=10,
+                   overlap=20, statics=40, cktext=1), {}),
+        ("x", dict(seed=7, n=2, rows=600, vlen=80, overlap=25, tomb=10, cpx=40,
+                   cpxdel=25), {}),
+        ("t", dict(seed=8, n=2, rows=600, vlen=80, overlap=25, ttl=35), dict(now=1800000000)),
+        ("c", dict(seed=9, n=2, rows=600, overlap=30, tomb=10, counter=1), {}),
+    ]
+    for tag, genkw, ckw in cases:
+        dd = os.path.join(d, tag)
+        os.makedirs(dd)
+        _oracle_gen(dd, **genkw)
+        ins = [f"{dd}/oa-{g}-big" for g in range(1, genkw["n"] + 1)]
+        subprocess.run([ORACLE, "validate", f"{dd}/v_cpu.bin", *ins,
+                        *[f"{k}={v}" for k, v in ckw.items()]],
+                       check=True, capture_output=True)
+        n = ca.validate(ins, f"{dd}/v_gpu.bin",
+                        now_sec=ckw.get("now", 1800000000))
+        a = open(f"{dd}/v_cpu.bin", "rb").read()
+        b = open(f"{dd}/v_gpu.bin", "rb").read()
+        assert n == len(a) // 40 and a == b, (
+            tag, n, len(a) // 40, len(b) // 40,
+            next((i for i in range(min(len(a), len(b))) if a[i] != b[i]), -1))
+
+
 def test_cancellation(ca, oracle_bin, tmp_path):
     """Cooperative cancel (CompactionIterator.isStopRequested): a set
     cancel_flag aborts the task with GPUC_ERR_CANCELLED; a zero flag is
