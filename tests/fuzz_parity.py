@@ -18,10 +18,12 @@ sys.path.insert(0, REPO)
 ORACLE = os.path.join(REPO, "oracle", "bin", "oracle_tool")
 COMPONENTS = ["Data.db", "Index.db", "CompressionInfo.db", "Filter.db",
               "Digest.crc32", "Statistics.db", "Summary.db", "TOC.txt"]
+BTI_COMPONENTS = ["Data.db", "CompressionInfo.db", "Filter.db", "Digest.crc32",
+                  "Statistics.db", "Partitions.db", "Rows.db", "TOC.txt"]
 
 
-def dirs_equal(a, b):
-    for c in COMPONENTS:
+def dirs_equal(a, b, components=COMPONENTS):
+    for c in components:
         with open(f"{a}-{c}", "rb") as f1, open(f"{b}-{c}", "rb") as f2:
             if f1.read() != f2.read():
                 return c
@@ -69,7 +71,28 @@ def main():
         if rng.random() < 0.4:
             gen["ncols"] = rng.choice([2, 4, 7])
             gen["colmiss"] = rng.choice([0, 25, 60])
+        # round-2 feature knobs: counters (exclusive of blob-column knobs),
+        # or TTL / complex columns; BTI (da) format orthogonal to all
+        if rng.random() < 0.15:
+            gen["counter"] = 1
+            for k in ("vrep", "ncols", "colmiss", "statics", "vlen"):
+                gen.pop(k, None)
+        else:
+            if rng.random() < 0.3:
+                gen["ttl"] = rng.choice([15, 40, 80])
+            if rng.random() < 0.3:
+                gen["cpx"] = rng.choice([20, 45])
+                gen["cpxdel"] = rng.choice([0, 25, 60])
+        use_bti = rng.random() < 0.25
+        if use_bti:
+            gen["bti"] = 1
+        comps = BTI_COMPONENTS if use_bti else COMPONENTS
+        stem = "da" if use_bti else "oa"
+        sfx = "bti" if use_bti else "big"
         job, okw = {}, {}
+        if "ttl" in gen:
+            job["now_sec"] = 1800000000
+            okw["now"] = 1800000000
         if rng.random() < 0.4:
             job["gc_before"] = 2000000000
             okw["gcbefore"] = 2000000000
@@ -78,46 +101,63 @@ def main():
             okw["nevergc"] = 1
             okw.setdefault("gcbefore", 2000000000)
             job.setdefault("gc_before", 2000000000)
-        use_gc_sources = not wide and rng.random() < 0.3 and "gc_before" not in job
-        args = [f"{k}={v}" for k, v in gen.items()]
-        subprocess.run([ORACLE, "gen", d, *args], check=True, capture_output=True)
-        ins = [f"{d}/oa-{g}-big" for g in range(1, gen["n"] + 1)]
-        if use_gc_sources:
-            os.makedirs(d + "/src")
-            sargs = dict(gen)
-            sargs["seed"] = gen["seed"]
-            sargs["tomb"] = 50
-            sargs["n"] = 2
-            sargs["ts0"] = 1700000500000000
-            subprocess.run([ORACLE, "gen", d + "/src",
-                            *[f"{k}={v}" for k, v in sargs.items()]],
-                           check=True, capture_output=True)
-            srcs = [f"{d}/src/oa-{g}-big" for g in (1, 2)]
-            okw["tombsrc"] = ",".join(srcs)
-            cell = rng.random() < 0.5
-            if cell:
-                okw["cellgc"] = 1
-            job["tombstone_sources"] = srcs
-            job["cell_level_gc"] = cell
-        use_shards = (not use_gc_sources) and rng.random() < 0.2
-        oargs = [f"{k}={v}" for k, v in okw.items()]
-        if use_shards:
-            from cassandra_amd.sharding import split_token_range
-            S = rng.choice([2, 3])
-            ca.compact(ins, f"{d}/oa-91-big", n_output_shards=S, **job)
-            bad = None
-            for i in range(S):
-                lo, hi = split_token_range(S, i)
-                subprocess.run([ORACLE, "compact", f"{d}/oa-{90 + 10 * i}-big", *ins,
-                                f"shard={lo}:{hi}", *oargs], check=True, capture_output=True)
-                bad = bad or dirs_equal(f"{d}/oa-{90 + 10 * i}-big", f"{d}/oa-{91 + i}-big")
-        else:
-            subprocess.run([ORACLE, "compact", f"{d}/oa-90-big", *ins, *oargs],
-                           check=True, capture_output=True)
-            ca.compact(ins, f"{d}/oa-91-big", **job)
-            bad = dirs_equal(f"{d}/oa-90-big", f"{d}/oa-91-big")
-            if not bad and rng.random() < 0.3:
-                ca.verify(f"{d}/oa-91-big")
+        use_gc_sources = (not wide and rng.random() < 0.3 and "gc_before" not in job
+                          and "counter" not in gen and "cpx" not in gen)
+        use_shards = False
+        bad = None
+        try:
+            args = [f"{k}={v}" for k, v in gen.items()]
+            subprocess.run([ORACLE, "gen", d, *args], check=True, capture_output=True)
+            ins = [f"{d}/{stem}-{g}-{sfx}" for g in range(1, gen["n"] + 1)]
+            if use_gc_sources:
+                os.makedirs(d + "/src")
+                sargs = dict(gen)
+                sargs["seed"] = gen["seed"]
+                sargs["tomb"] = 50
+                sargs["n"] = 2
+                sargs["ts0"] = 1700000500000000
+                subprocess.run([ORACLE, "gen", d + "/src",
+                                *[f"{k}={v}" for k, v in sargs.items()]],
+                               check=True, capture_output=True)
+                srcs = [f"{d}/src/{stem}-{g}-{sfx}" for g in (1, 2)]
+                okw["tombsrc"] = ",".join(srcs)
+                cell = rng.random() < 0.5
+                if cell:
+                    okw["cellgc"] = 1
+                job["tombstone_sources"] = srcs
+                job["cell_level_gc"] = cell
+            use_shards = (not use_gc_sources) and (not use_bti) and rng.random() < 0.2
+            oargs = [f"{k}={v}" for k, v in okw.items()]
+            if use_shards:
+                from cassandra_amd.sharding import split_token_range
+                S = rng.choice([2, 3])
+                ca.compact(ins, f"{d}/{stem}-91-{sfx}", n_output_shards=S, **job)
+                bad = None
+                for i in range(S):
+                    lo, hi = split_token_range(S, i)
+                    subprocess.run([ORACLE, "compact", f"{d}/{stem}-{90 + 10 * i}-{sfx}",
+                                    *ins, f"shard={lo}:{hi}", *oargs],
+                                   check=True, capture_output=True)
+                    bad = bad or dirs_equal(f"{d}/{stem}-{90 + 10 * i}-{sfx}",
+                                            f"{d}/{stem}-{91 + i}-{sfx}", comps)
+            else:
+                subprocess.run([ORACLE, "compact", f"{d}/{stem}-90-{sfx}", *ins, *oargs],
+                               check=True, capture_output=True)
+                ca.compact(ins, f"{d}/{stem}-91-{sfx}", **job)
+                bad = dirs_equal(f"{d}/{stem}-90-{sfx}", f"{d}/{stem}-91-{sfx}", comps)
+                if not bad and not use_bti and rng.random() < 0.3:
+                    ca.verify(f"{d}/{stem}-91-{sfx}")
+                if not bad and rng.random() < 0.3:
+                    # validation compaction on the same inputs
+                    vkw = [f"now={job['now_sec']}"] if "now_sec" in job else []
+                    subprocess.run([ORACLE, "validate", f"{d}/v_cpu.bin", *ins, *vkw],
+                                   check=True, capture_output=True)
+                    ca.validate(ins, f"{d}/v_gpu.bin",
+                                now_sec=job.get("now_sec", 1800000000))
+                    if open(f"{d}/v_cpu.bin", "rb").read() != open(f"{d}/v_gpu.bin", "rb").read():
+                        bad = "validate-digests"
+        except Exception as e:
+            bad = f"crash {type(e).__name__}: {e}"
         if bad:
             fails += 1
             print(f"FAIL cfg {t}: component {bad}; gen={gen} job={job}")
