@@ -769,10 +769,15 @@ inline bytes byte_comparable_clustering(const std::vector<std::pair<const uint8_
             bc_escape(out, tmp);
         }
     }
+    // Kind.asByteComparableValue at ByteComparable.Version.OSS50 (the version
+    // all tries use, io/tries/Walker.java:64): CLUSTERING -> TERMINATOR 0x38,
+    // GT side (incl-end / excl-start bounds + boundary) -> 0x60, LT side
+    // (excl-end / incl-start bounds + boundary) -> 0x20, STATIC -> 0x18.
     switch (kind) {
-        case 4: out.push_back(0x40); break;          // CLUSTERING
-        case 6: case 7: out.push_back(0x60); break;  // INCL_END / EXCL_START
-        case 0: case 1: out.push_back(0x20); break;  // EXCL_END / INCL_START
+        case 4: out.push_back(0x38); break;                  // CLUSTERING
+        case 5: case 6: case 7: out.push_back(0x60); break;  // INCL_END_EXCL_START/INCL_END/EXCL_START
+        case 0: case 1: case 2: out.push_back(0x20); break;  // EXCL_END/INCL_START/EXCL_END_INCL_START
+        case 3: out.push_back(0x18); break;                  // STATIC
         default: throw std::runtime_error("unsupported bound kind for byte-comparable");
     }
     return out;

@@ -611,10 +611,16 @@ bytes bti_byte_comparable_clustering(const Clustering& c,
                 bc_escape(out, c[i].v);
         }
     }
+    // Kind.asByteComparableValue at ByteComparable.Version.OSS50 — the version
+    // every trie writer/reader uses (io/tries/Walker.java:64;
+    // ClusteringPrefix.java:70-79): CLUSTERING -> TERMINATOR 0x38 (the LEGACY
+    // value is 0x40), end/start bounds and boundaries by their LT/GT side,
+    // STATIC_CLUSTERING -> EXCLUDED 0x18.
     switch (kind) {
-        case CLUSTERING_K: out.push_back(0x40); break;
-        case INCL_END: case EXCL_START: out.push_back(0x60); break;
-        case EXCL_END: case INCL_START: out.push_back(0x20); break;
+        case CLUSTERING_K: out.push_back(0x38); break;
+        case INCL_END: case EXCL_START: case INCL_END_EXCL_START: out.push_back(0x60); break;
+        case EXCL_END: case INCL_START: case EXCL_END_INCL_START: out.push_back(0x20); break;
+        case STATIC_K: out.push_back(0x18); break;
         default: throw std::runtime_error("unsupported bound kind for byte-comparable");
     }
     return out;

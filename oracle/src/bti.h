@@ -89,8 +89,9 @@ uint64_t append_bti_row_index(bytes& file, const BtiRowIndexBlockSpec& spec);
 
 // OSS50 byte-comparable encoding of a ClusteringPrefix
 // (ClusteringComparator.ByteComparableClustering: per component
-// NEXT_COMPONENT 0x40 + the type's OSS50 encoding; kind terminator —
-// CLUSTERING_K 0x40, INCL_END/EXCL_START 0x60, EXCL_END/INCL_START 0x20).
+// NEXT_COMPONENT 0x40 + the type's OSS50 encoding; kind terminator per
+// Kind.asByteComparableValue at Version.OSS50 — CLUSTERING_K 0x38, GT-side
+// bounds/boundary 0x60, LT-side 0x20, STATIC 0x18).
 // Types supported: UTF8/ASCII/BYTES (escaped) and LONG/INT
 // (ByteSource.variableLengthInteger).
 bytes bti_byte_comparable_clustering(const Clustering& c,

@@ -665,6 +665,19 @@ def test_bti_da_pipeline(ca, oracle_bin, tmp_path):
     ca.compact(sins, f"{ds}/da-60-bti")
     _oracle_compact(f"{ds}/da-80-bti", sins)
     _assert_dirs_equal(f"{ds}/da-60-bti", f"{ds}/da-80-bti", BTI_COMPONENTS)
+    # boundary-kind block keys (fuzz regression): dense range tombstones +
+    # large cells force row-index blocks that START at range-tombstone
+    # boundary markers, so separatorGt stores the marker's OSS50 terminator
+    # (LT 0x20 / GT 0x60; Kind.asByteComparableValue at Walker.java:64's
+    # Version.OSS50) — the kinds the first encoder version rejected
+    dbb = d + "/bb"
+    os.makedirs(dbb)
+    _oracle_gen(dbb, seed=138011248, n=4, rows=20, crows=5, vlen=20000,
+                overlap=40, tomb=10, rtomb=70, cktext=1, statics=30, bti=1)
+    bbins = [f"{dbb}/da-{g}-bti" for g in (1, 2, 3, 4)]
+    ca.compact(bbins, f"{dbb}/da-60-bti")
+    _oracle_compact(f"{dbb}/da-80-bti", bbins)
+    _assert_dirs_equal(f"{dbb}/da-60-bti", f"{dbb}/da-80-bti", BTI_COMPONENTS)
     # GPU writer parity: generate(bti=True) == oracle gen bti=1
     dg, do = d + "/g", d + "/o"
     os.makedirs(dg), os.makedirs(do)
