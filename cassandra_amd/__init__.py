@@ -161,6 +161,10 @@ def load_library(path=None):
     lib.gpuc_flush.argtypes = [ctypes.POINTER(GpucFlushRows), ctypes.c_char_p, ctypes.c_int32,
                                ctypes.c_char_p, ctypes.c_size_t]
     lib.gpuc_flush.restype = ctypes.c_int
+    lib.gpuc_flush_table.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_uint64,
+                                     ctypes.c_char_p, ctypes.c_int32,
+                                     ctypes.c_char_p, ctypes.c_size_t]
+    lib.gpuc_flush_table.restype = ctypes.c_int
     lib.gpuc_version.restype = ctypes.c_char_p
     lib.gpuc_device_count.restype = ctypes.c_int
     _lib = lib
@@ -202,6 +206,306 @@ def flush(rows, output_base, device=0):
     rc = lib.gpuc_flush(ctypes.byref(fr), output_base.encode(), device, err, 256)
     if rc != 0:
         raise GpuCompactError(f"gpuc_flush rc={rc}: {err.value.decode(errors='replace')}")
+
+
+# ---- full-schema memtable flush (gpuc_flush_table) ----
+LDT_NONE = 0xFFFFFFFF
+CELLF_PRESENT, CELLF_HAS_VALUE = 1, 2
+
+
+class GpucCell(ctypes.Structure):
+    _fields_ = [
+        ("flags", ctypes.c_uint8),
+        ("ts", ctypes.c_int64),
+        ("ldt", ctypes.c_uint32),
+        ("ttl", ctypes.c_int32),
+        ("value", ctypes.POINTER(ctypes.c_uint8)),
+        ("value_len", ctypes.c_uint32),
+    ]
+
+
+class GpucCpxCell(ctypes.Structure):
+    _fields_ = [
+        ("cell", GpucCell),
+        ("path", ctypes.POINTER(ctypes.c_uint8)),
+        ("path_len", ctypes.c_uint32),
+    ]
+
+
+class GpucUnfiltered(ctypes.Structure):
+    _fields_ = [
+        ("kind", ctypes.c_uint8),
+        ("ck_count", ctypes.c_uint8),
+        ("row_flags", ctypes.c_uint8),
+        ("ck", ctypes.POINTER(ctypes.POINTER(ctypes.c_uint8))),
+        ("ck_lens", ctypes.POINTER(ctypes.c_uint32)),
+        ("live_ts", ctypes.c_int64),
+        ("live_ttl", ctypes.c_int32),
+        ("live_let", ctypes.c_int64),
+        ("del_mfda", ctypes.c_int64),
+        ("del_ldt", ctypes.c_uint32),
+        ("open_mfda", ctypes.c_int64),
+        ("open_ldt", ctypes.c_uint32),
+        ("cells", ctypes.POINTER(GpucCell)),
+        ("has_cpx", ctypes.c_uint8),
+        ("cpx_del_mfda", ctypes.c_int64),
+        ("cpx_del_ldt", ctypes.c_uint32),
+        ("n_cpx_cells", ctypes.c_uint32),
+        ("cpx_cells", ctypes.POINTER(GpucCpxCell)),
+    ]
+
+
+class GpucFlushPart(ctypes.Structure):
+    _fields_ = [
+        ("key", ctypes.POINTER(ctypes.c_uint8)),
+        ("key_len", ctypes.c_uint16),
+        ("pdel_mfda", ctypes.c_int64),
+        ("pdel_ldt", ctypes.c_uint32),
+        ("static_flags", ctypes.c_uint8),
+        ("static_live_ts", ctypes.c_int64),
+        ("static_live_ttl", ctypes.c_int32),
+        ("static_live_let", ctypes.c_int64),
+        ("static_del_mfda", ctypes.c_int64),
+        ("static_del_ldt", ctypes.c_uint32),
+        ("static_cells", ctypes.POINTER(GpucCell)),
+        ("n_unf", ctypes.c_uint64),
+        ("unf", ctypes.POINTER(GpucUnfiltered)),
+    ]
+
+
+class GpucFlushSchema(ctypes.Structure):
+    _fields_ = [
+        ("key_type", ctypes.c_char_p),
+        ("n_ck", ctypes.c_uint32),
+        ("ck_types", ctypes.POINTER(ctypes.c_char_p)),
+        ("n_cols", ctypes.c_uint32),
+        ("col_names", ctypes.POINTER(ctypes.POINTER(ctypes.c_uint8))),
+        ("col_name_lens", ctypes.POINTER(ctypes.c_uint32)),
+        ("col_types", ctypes.POINTER(ctypes.c_char_p)),
+        ("n_static", ctypes.c_uint32),
+        ("static_names", ctypes.POINTER(ctypes.POINTER(ctypes.c_uint8))),
+        ("static_name_lens", ctypes.POINTER(ctypes.c_uint32)),
+        ("static_types", ctypes.POINTER(ctypes.c_char_p)),
+        ("n_cpx", ctypes.c_uint32),
+        ("column_index_size", ctypes.c_uint32),
+        ("snappy", ctypes.c_uint32),
+        ("bti", ctypes.c_uint32),
+    ]
+
+
+# oracle CqlType ordinal -> AbstractType class name (sstable.h:30,
+# cql_type_name) — the memdump stores the ordinal
+_CQL_TYPE_NAMES = [
+    b"org.apache.cassandra.db.marshal.BytesType",
+    b"org.apache.cassandra.db.marshal.UTF8Type",
+    b"org.apache.cassandra.db.marshal.AsciiType",
+    b"org.apache.cassandra.db.marshal.LongType",
+    b"org.apache.cassandra.db.marshal.Int32Type",
+    b"org.apache.cassandra.db.marshal.MapType"
+    b"(org.apache.cassandra.db.marshal.BytesType,"
+    b"org.apache.cassandra.db.marshal.BytesType)",
+    b"org.apache.cassandra.db.marshal.CounterColumnType",
+]
+
+
+class _MemdumpReader:
+    """Parses the oracle's .memdump interchange (write_memdump) into the
+    gpuc_flush_table ctypes graph. Keeps every referenced buffer alive via
+    self.bufs (ctypes pointers do not own memory)."""
+
+    def __init__(self, data):
+        self.d = data
+        self.o = 0
+        self.bufs = []
+
+    def u8(self):
+        v = self.d[self.o]; self.o += 1; return v
+
+    def u32(self):
+        v = int.from_bytes(self.d[self.o:self.o + 4], "little"); self.o += 4; return v
+
+    def i32(self):
+        v = int.from_bytes(self.d[self.o:self.o + 4], "little", signed=True); self.o += 4; return v
+
+    def u64(self):
+        v = int.from_bytes(self.d[self.o:self.o + 8], "little"); self.o += 8; return v
+
+    def i64(self):
+        v = int.from_bytes(self.d[self.o:self.o + 8], "little", signed=True); self.o += 8; return v
+
+    def blob(self):
+        n = self.u32()
+        v = self.d[self.o:self.o + n]; self.o += n; return v
+
+    def ptr(self, b):
+        buf = ctypes.create_string_buffer(bytes(b), max(len(b), 1))
+        self.bufs.append(buf)
+        return ctypes.cast(buf, ctypes.POINTER(ctypes.c_uint8)), len(b)
+
+    def cell_into(self, c):
+        cf = self.u8()
+        c.flags = cf
+        if not cf:
+            return
+        c.ts = self.i64()
+        c.ldt = self.u32()
+        c.ttl = self.i32()
+        v = self.blob()
+        c.value, c.value_len = self.ptr(v)
+
+    def row_head(self):
+        f = self.u8()
+        return f, self.i64(), self.i32(), self.i64(), self.i64(), self.u32()
+
+
+def _parse_memdump(data):
+    r = _MemdumpReader(data)
+    if data[:4] != b"GMD1":
+        raise GpuCompactError("bad memdump magic")
+    r.o = 4
+    key_type = _CQL_TYPE_NAMES[r.u8()]
+    n_ck = r.u32()
+    ck_types = [_CQL_TYPE_NAMES[r.u8()] for _ in range(n_ck)]
+    n_static = r.u32()
+    statics = [(r.blob(), _CQL_TYPE_NAMES[r.u8()]) for _ in range(n_static)]
+    n_cols = r.u32()
+    cols = [(r.blob(), r.u8()) for _ in range(n_cols)]
+    n_cpx = sum(1 for _, t in cols if t == 5)
+    cols = [(nm, _CQL_TYPE_NAMES[t]) for nm, t in cols]
+    snappy = r.u8()
+    bti = r.u8()
+    cis = r.u32()
+    n_parts = r.u64()
+
+    S = GpucFlushSchema()
+    S.key_type = key_type
+    S.n_ck = n_ck
+    S.n_cols = n_cols
+    S.n_static = n_static
+    cn = [r.ptr(nm) for nm, _ in cols]
+    sn = [r.ptr(nm) for nm, _ in statics]
+    # ctypes Structure POINTER fields do not keep their targets alive: build
+    # the arrays as locals, stash them in r.bufs, THEN point S at them
+    a_ckt = (ctypes.c_char_p * max(n_ck, 1))(*ck_types)
+    a_cnp = (ctypes.POINTER(ctypes.c_uint8) * n_cols)(*[p for p, _ in cn])
+    a_cnl = (ctypes.c_uint32 * n_cols)(*[l for _, l in cn])
+    a_ct = (ctypes.c_char_p * n_cols)(*[t for _, t in cols])
+    a_snp = (ctypes.POINTER(ctypes.c_uint8) * max(n_static, 1))(*[p for p, _ in sn])
+    a_snl = (ctypes.c_uint32 * max(n_static, 1))(*[l for _, l in sn])
+    a_st = (ctypes.c_char_p * max(n_static, 1))(*[t for _, t in statics])
+    r.bufs.extend((a_ckt, a_cnp, a_cnl, a_ct, a_snp, a_snl, a_st))
+    S.ck_types, S.col_names, S.col_name_lens, S.col_types = a_ckt, a_cnp, a_cnl, a_ct
+    S.static_names, S.static_name_lens, S.static_types = a_snp, a_snl, a_st
+    S.n_cpx = n_cpx
+    S.column_index_size = cis
+    S.snappy = snappy
+    S.bti = bti
+    n_simple = n_cols - n_cpx
+    cpx_idx = [i for i, (_, t) in enumerate(cols) if b"MapType" in t]
+
+    parts = (GpucFlushPart * n_parts)()
+    for pi in range(n_parts):
+        P = parts[pi]
+        key = r.blob()
+        P.key, P.key_len = r.ptr(key)
+        P.pdel_mfda = r.i64()
+        P.pdel_ldt = r.u32()
+        P.static_flags = 0
+        if n_static:
+            f = r.u8()
+            if f:
+                P.static_flags = f
+                P.static_live_ts = r.i64()
+                P.static_live_ttl = r.i32()
+                P.static_live_let = r.i64()
+                P.static_del_mfda = r.i64()
+                P.static_del_ldt = r.u32()
+                sc = (GpucCell * n_static)()
+                for c in range(n_static):
+                    r.cell_into(sc[c])
+                r.bufs.append(sc)
+                P.static_cells = sc
+        n_unf = r.u32()
+        P.n_unf = n_unf
+        unf = (GpucUnfiltered * max(n_unf, 1))()
+        r.bufs.append(unf)
+        P.unf = unf
+        for ui in range(n_unf):
+            U = unf[ui]
+            tag = r.u8()
+            if tag == 1:
+                U.kind = r.u8()
+            else:
+                U.kind = 4
+            ckc = r.u8()
+            U.ck_count = ckc
+            ckp = (ctypes.POINTER(ctypes.c_uint8) * max(ckc, 1))()
+            ckl = (ctypes.c_uint32 * max(ckc, 1))()
+            for c in range(ckc):
+                state = r.u8()
+                comp = r.blob()
+                if state != 0:
+                    raise GpuCompactError("EMPTY/NULL clustering components unsupported")
+                ckp[c], ckl[c] = r.ptr(comp)
+            r.bufs.extend((ckp, ckl))
+            U.ck = ckp
+            U.ck_lens = ckl
+            if tag == 1:
+                U.del_mfda = r.i64()
+                U.del_ldt = r.u32()
+                U.open_mfda = r.i64()
+                U.open_ldt = r.u32()
+                continue
+            f, lts, lttl, llet, dm, dl = r.row_head()
+            U.row_flags = f
+            U.live_ts, U.live_ttl, U.live_let = lts, lttl, llet
+            U.del_mfda, U.del_ldt = dm, dl
+            cells = (GpucCell * max(n_simple, 1))()
+            r.bufs.append(cells)
+            U.cells = cells
+            si = 0
+            for i in range(n_cols):
+                if i in cpx_idx:
+                    has = r.u8()
+                    U.has_cpx = has
+                    if has:
+                        U.cpx_del_mfda = r.i64()
+                        U.cpx_del_ldt = r.u32()
+                        nc = r.u32()
+                        U.n_cpx_cells = nc
+                        xc = (GpucCpxCell * max(nc, 1))()
+                        r.bufs.append(xc)
+                        U.cpx_cells = xc
+                        for e in range(nc):
+                            r.cell_into(xc[e].cell)
+                            path = r.blob()
+                            xc[e].path, xc[e].path_len = r.ptr(path)
+                else:
+                    r.cell_into(cells[si])
+                    si += 1
+    if r.o != len(data):
+        raise GpuCompactError(f"memdump trailing bytes: {r.o} != {len(data)}")
+    return S, parts, n_parts, r.bufs
+
+
+def _flush_table_parsed(S, parts, n_parts, output_base, device=0):
+    lib = load_library()
+    err = ctypes.create_string_buffer(256)
+    rc = lib.gpuc_flush_table(ctypes.byref(S), parts, n_parts, output_base.encode(),
+                              device, err, 256)
+    if rc != 0:
+        raise GpuCompactError(f"gpuc_flush_table rc={rc}: {err.value.decode(errors='replace')}")
+
+
+def flush_table(memdump_path, output_base, device=0):
+    """Full-schema memtable flush: marshal an oracle .memdump (the logical
+    memtable content) through gpuc_flush_table. A real host would build the
+    same structs straight from its memtable."""
+    with open(memdump_path, "rb") as f:
+        data = f.read()
+    S, parts, n_parts, bufs = _parse_memdump(data)
+    _flush_table_parsed(S, parts, n_parts, output_base, device)
+    del bufs
 
 
 def scrub(input_base, output_base, device=0):

@@ -3147,6 +3147,470 @@ extern "C" int gpuc_flush(const gpuc_flush_rows* rows, const char* output_base,
     }
 }
 
+
+// ---- full-schema memtable flush (gpuc_flush_table) -----------------------
+// Memtable.FlushablePartitionSet -> SortedTablePartitionWriter: the caller
+// hands partitions with within-partition clustering order already
+// established (memtables are sorted maps); the engine token-sorts the
+// partitions on device, validates order/uniqueness, marshals rows into the
+// same OutParts/UnfCols layout the generator and parse kernels produce, and
+// runs the shared writer kernels (serialize/compress/index/bloom/stats).
+namespace flushv2 {
+
+// host mirrors of the device ck encodings (kernels_rows.hip ck_sortable /
+// ck_prefix_var)
+static uint64_t h_ck_sortable(const uint8_t* p, int width) {
+    uint64_t v = 0;
+    for (int b = 0; b < width; b++) v = (v << 8) | p[b];
+    return v ^ (1ULL << (8 * width - 1));
+}
+static uint64_t h_ck_prefix_var(const uint8_t* p, uint32_t len) {
+    uint64_t v = 0;
+    uint32_t n = len < 8 ? len : 8;
+    for (uint32_t b = 0; b < n; b++) v |= (uint64_t)p[b] << (8 * (7 - b));
+    return v;
+}
+static int h_bk_cmp_to_clustering(uint8_t kk) {
+    static const int tbl[8] = {-1, -1, -1, -1, 0, 1, 1, 1};
+    return tbl[kk];
+}
+static int h_bk_comparison(uint8_t k) {
+    static const int tbl[8] = {0, 0, 0, 1, 2, 3, 3, 3};
+    return tbl[k];
+}
+
+// ClusteringComparator.compare over two ABI unfiltereds (mirrors pos_cmp)
+static int h_unf_cmp(const gpuc_unfiltered& a, const gpuc_unfiltered& b,
+                     const std::vector<int32_t>& ckw) {
+    uint32_t mn = a.ck_count < b.ck_count ? a.ck_count : b.ck_count;
+    for (uint32_t c = 0; c < mn; c++) {
+        int32_t w = ckw[c];
+        const uint8_t* pa = a.ck[c];
+        const uint8_t* pb = b.ck[c];
+        uint32_t la = a.ck_lens[c], lb = b.ck_lens[c];
+        uint64_t sa, sb;
+        if (w > 0) { sa = h_ck_sortable(pa, w); sb = h_ck_sortable(pb, w); }
+        else { sa = h_ck_prefix_var(pa, la); sb = h_ck_prefix_var(pb, lb); }
+        if (sa != sb) return sa < sb ? -1 : 1;
+        if (w < 0) {
+            uint32_t n = la < lb ? la : lb;
+            for (uint32_t i = 8; i < n; i++)
+                if (pa[i] != pb[i]) return pa[i] < pb[i] ? -1 : 1;
+            if (la != lb) return la < lb ? -1 : 1;
+        }
+    }
+    if (a.ck_count != b.ck_count)
+        return a.ck_count < b.ck_count ? h_bk_cmp_to_clustering(a.kind)
+                                       : -h_bk_cmp_to_clustering(b.kind);
+    return h_bk_comparison(a.kind) - h_bk_comparison(b.kind);
+}
+
+}  // namespace flushv2
+
+extern "C" int gpuc_flush_table(const gpuc_flush_schema* schema, const gpuc_flush_part* parts,
+                                uint64_t n_parts, const char* output_base, int32_t device,
+                                char* error, size_t error_len) {
+    using namespace flushv2;
+    try {
+        if (gpuc_device_count() <= 0) { set_err(error, error_len, "no HIP device"); return GPUC_ERR_NO_GPU; }
+        if (!schema || !parts || n_parts == 0) throw std::runtime_error("no partitions");
+        const gpuc_flush_schema& S = *schema;
+        if (S.n_cols == 0 || S.n_cols + S.n_static > 63)
+            throw std::runtime_error("1..63 columns supported");
+        if (S.n_cpx > 1) throw std::runtime_error("at most one complex column");
+        if (S.n_ck > 32) throw std::runtime_error("at most 32 clustering columns");
+        const std::string MAPT_PFX = "org.apache.cassandra.db.marshal.MapType";
+        const std::string CTRT = "org.apache.cassandra.db.marshal.CounterColumnType";
+        std::string key_type = S.key_type ? S.key_type
+                                          : "org.apache.cassandra.db.marshal.BytesType";
+        std::vector<std::string> ck_types;
+        std::vector<int32_t> ckw_h;
+        for (uint32_t i = 0; i < S.n_ck; i++) {
+            ck_types.emplace_back(S.ck_types[i]);
+            ckw_h.push_back(ck_type_width(ck_types.back()));
+        }
+        // n_cols in the ABI counts ALL regular columns; SchemaParams.n_cols
+        // counts the SIMPLE ones (complex last, excluded)
+        const uint32_t NSIMPLE = S.n_cols - S.n_cpx;
+        std::vector<std::pair<bytes, std::string>> regular_cols, static_cols;
+        std::vector<int32_t> colw_h, staticw_h;
+        bool counters = false;
+        for (uint32_t i = 0; i < S.n_cols; i++) {
+            std::string t(S.col_types[i]);
+            bool is_map = t.compare(0, MAPT_PFX.size(), MAPT_PFX) == 0;
+            if (is_map != (S.n_cpx == 1 && i == S.n_cols - 1))
+                throw std::runtime_error("a complex (MapType) column must be the last regular column");
+            if (t == CTRT) counters = true;
+            regular_cols.emplace_back(bytes(S.col_names[i], S.col_names[i] + S.col_name_lens[i]), t);
+            if (i < NSIMPLE)
+                colw_h.push_back(t == "org.apache.cassandra.db.marshal.LongType" ? 8
+                                 : t == "org.apache.cassandra.db.marshal.Int32Type" ? 4 : -1);
+        }
+        for (uint32_t i = 0; i < S.n_static; i++) {
+            std::string t(S.static_types[i]);
+            static_cols.emplace_back(bytes(S.static_names[i], S.static_names[i] + S.static_name_lens[i]), t);
+            staticw_h.push_back(t == "org.apache.cassandra.db.marshal.LongType" ? 8
+                                : t == "org.apache.cassandra.db.marshal.Int32Type" ? 4 : -1);
+        }
+        // ---- pass 0: totals + validation ----
+        uint64_t key_bytes = 0, n_unf = 0, val_bytes = 0, cpx_total = 0;
+        for (uint64_t p = 0; p < n_parts; p++) {
+            const gpuc_flush_part& P = parts[p];
+            if (P.key_len == 0) throw std::runtime_error("empty partition key");
+            key_bytes += P.key_len;
+            n_unf += P.n_unf;
+            if (P.static_flags && S.n_static)
+                for (uint32_t c = 0; c < S.n_static; c++)
+                    if (P.static_cells[c].flags & GPUC_CELLF_PRESENT)
+                        val_bytes += P.static_cells[c].value_len;
+            const gpuc_unfiltered* prev = nullptr;
+            for (uint64_t r = 0; r < P.n_unf; r++) {
+                const gpuc_unfiltered& U = P.unf[r];
+                if (U.kind > 7 || U.kind == 3) throw std::runtime_error("bad unfiltered kind");
+                bool is_row = U.kind == 4;
+                if (is_row && U.ck_count != S.n_ck)
+                    throw std::runtime_error("rows must have all clustering components");
+                if (U.ck_count > S.n_ck) throw std::runtime_error("ck_count exceeds schema");
+                for (uint32_t c = 0; c < U.ck_count; c++) {
+                    int32_t w = ckw_h[c];
+                    if (w > 0 && U.ck_lens[c] != (uint32_t)w)
+                        throw std::runtime_error("fixed clustering component width mismatch");
+                    if (w < 0) val_bytes += U.ck_lens[c];
+                }
+                if (is_row) {
+                    for (uint32_t c = 0; c < NSIMPLE; c++)
+                        if (U.cells[c].flags & GPUC_CELLF_PRESENT) val_bytes += U.cells[c].value_len;
+                    if (U.has_cpx && !S.n_cpx)
+                        throw std::runtime_error("complex data on a schema without a complex column");
+                    if (U.has_cpx) {
+                        cpx_total += U.n_cpx_cells;
+                        for (uint32_t e = 0; e < U.n_cpx_cells; e++)
+                            val_bytes += U.cpx_cells[e].path_len + U.cpx_cells[e].cell.value_len;
+                    }
+                }
+                if (prev && h_unf_cmp(*prev, U, ckw_h) >= 0)
+                    throw std::runtime_error("unfiltereds out of clustering order in partition " +
+                                             std::to_string(p));
+                prev = &U;
+            }
+        }
+        HIP_CHECK(hipSetDevice(device));
+        hipStream_t stream;
+        HIP_CHECK(hipStreamCreate(&stream));
+        ensure_crc_tables(stream);
+        // ---- key arena + device partition sort (token, key bytes) ----
+        std::vector<uint8_t> kbuf(key_bytes);
+        std::vector<uint64_t> koff(n_parts + 1, 0);
+        for (uint64_t p = 0; p < n_parts; p++) {
+            memcpy(kbuf.data() + koff[p], parts[p].key, parts[p].key_len);
+            koff[p + 1] = koff[p] + parts[p].key_len;
+        }
+        DevBuf d_keys, d_koff, d_a, d_b, d_koff_adj, d_err;
+        d_keys.alloc(kbuf.size());
+        d_koff.alloc((n_parts + 1) * 8);
+        HIP_CHECK(hipMemcpyAsync(d_keys.p, kbuf.data(), kbuf.size(), hipMemcpyHostToDevice, stream));
+        HIP_CHECK(hipMemcpyAsync(d_koff.p, koff.data(), (n_parts + 1) * 8, hipMemcpyHostToDevice, stream));
+        FlushParams fp{};
+        fp.keys = d_keys.as<uint8_t>();
+        fp.key_off = d_koff.as<uint64_t>();
+        fp.n = n_parts;
+        d_a.alloc(n_parts * sizeof(MRec));
+        d_b.alloc(n_parts * sizeof(MRec));
+        uint32_t pblocks = (uint32_t)((n_parts + 255) / 256);
+        hipLaunchKernelGGL(k_flush_recs, dim3(pblocks), dim3(256), 0, stream, fp, d_a.as<MRec>());
+        std::vector<uint64_t> koff_adj(n_parts);
+        for (uint64_t p = 0; p < n_parts; p++) koff_adj[p] = koff[p] - 2;
+        d_koff_adj.alloc(n_parts * 8);
+        HIP_CHECK(hipMemcpyAsync(d_koff_adj.p, koff_adj.data(), n_parts * 8, hipMemcpyHostToDevice, stream));
+        KeyLut lut{};
+        lut.base[0] = d_keys.as<uint8_t>();
+        lut.pos[0] = d_koff_adj.as<uint64_t>();
+        lut.enabled = 1;
+        MRec* d_sorted = merge_sort_recs(d_a.as<MRec>(), d_b.as<MRec>(), n_parts, stream, lut);
+        d_err.alloc(8);
+        HIP_CHECK(hipMemsetAsync(d_err.p, 0, 8, stream));
+        hipLaunchKernelGGL(k_flush_dupcheck, dim3(pblocks), dim3(256), 0, stream, fp, d_sorted,
+                           d_err.as<unsigned long long>());
+        std::vector<MRec> sorted(n_parts);
+        HIP_CHECK(hipStreamSynchronize(stream));
+        HIP_CHECK(hipMemcpy(sorted.data(), d_sorted, n_parts * sizeof(MRec), hipMemcpyDeviceToHost));
+        {
+            unsigned long long e = 0;
+            HIP_CHECK(hipMemcpy(&e, d_err.p, 8, hipMemcpyDeviceToHost));
+            if (e) throw std::runtime_error("duplicate partition keys in flush input");
+        }
+        // ---- device layout + host mirrors (filled in sorted order) ----
+        OutPartsBuf opb;
+        opb.alloc(n_parts, S.n_static);
+        UnfColsBuf urows;
+        urows.alloc(n_unf, NSIMPLE ? NSIMPLE : 1, S.n_ck, S.n_cpx);
+        if (S.n_cpx) urows.alloc_cpx_arena(cpx_total);
+        DevBuf d_vals;
+        d_vals.alloc(val_bytes ? val_bytes : 1);
+        uint8_t* const VBASE = d_vals.as<uint8_t>();
+        std::vector<uint8_t> h_vals(val_bytes ? val_bytes : 1);
+        uint64_t vcur = 0;
+        auto stash = [&](const uint8_t* src, uint32_t len) -> uint64_t {
+            uint64_t at = (uint64_t)(VBASE + vcur);
+            if (len) memcpy(h_vals.data() + vcur, src, len);
+            vcur += len;
+            return at;
+        };
+        const uint32_t NCK = S.n_ck ? S.n_ck : 1;
+        const uint32_t NCV = NSIMPLE ? NSIMPLE : 1;
+        const uint32_t NST = S.n_static ? S.n_static : 1;
+        // OutParts host
+        std::vector<uint64_t> h_keypfx(n_parts), h_keyaddr(n_parts), h_rowbase(n_parts);
+        std::vector<int64_t> h_token(n_parts), h_pdelm(n_parts);
+        std::vector<uint16_t> h_klen(n_parts);
+        std::vector<uint32_t> h_pdell(n_parts), h_rowcnt(n_parts);
+        std::vector<uint8_t> h_keep(n_parts, 1);
+        // statics host
+        std::vector<uint8_t> h_sflags(n_parts, 0), h_scf(n_parts * NST, 0);
+        std::vector<int64_t> h_sts(n_parts, NO_TIMESTAMP), h_slet(n_parts, NO_DELETION_TIME),
+            h_sdm(n_parts, INT64_MIN), h_scts(n_parts * NST, 0);
+        std::vector<int32_t> h_sttl(n_parts, 0), h_scttl(n_parts * NST, 0);
+        std::vector<uint32_t> h_sdl(n_parts, LDT_NONE_U32), h_scldt(n_parts * NST, 0),
+            h_scvl(n_parts * NST, 0);
+        std::vector<uint64_t> h_scva(n_parts * NST, 0);
+        // UnfCols host
+        std::vector<uint64_t> h_ck(n_unf * NCK, 0), h_ckaddr(n_unf * NCK, 0);
+        std::vector<uint32_t> h_cklen(n_unf * NCK, 0);
+        std::vector<uint8_t> h_ckcnt(n_unf, 0), h_rkind(n_unf), h_flags(n_unf, 0),
+            h_cf(n_unf * NCV, 0);
+        std::vector<int64_t> h_lts(n_unf, NO_TIMESTAMP), h_llet(n_unf, NO_DELETION_TIME),
+            h_rdm(n_unf, INT64_MIN), h_sm(n_unf, INT64_MIN), h_cts(n_unf * NCV, 0);
+        std::vector<int32_t> h_lttl(n_unf, 0), h_cttl(n_unf * NCV, 0);
+        std::vector<uint32_t> h_rdl(n_unf, LDT_NONE_U32), h_sl(n_unf, LDT_NONE_U32),
+            h_cldt(n_unf * NCV, 0), h_cvl(n_unf * NCV, 0);
+        std::vector<uint64_t> h_cva(n_unf * NCV, 0);
+        std::vector<int64_t> h_xdm(S.n_cpx ? n_unf : 0);
+        std::vector<uint32_t> h_xdl(S.n_cpx ? n_unf : 0), h_xcnt(S.n_cpx ? n_unf : 0);
+        std::vector<uint64_t> h_xstart(S.n_cpx ? n_unf : 0);
+        std::vector<int64_t> h_xts(cpx_total);
+        std::vector<uint32_t> h_xldt(cpx_total), h_xpl(cpx_total), h_xvl(cpx_total);
+        std::vector<int32_t> h_xttl(cpx_total);
+        std::vector<uint8_t> h_xf(cpx_total);
+        std::vector<uint64_t> h_xpa(cpx_total), h_xva(cpx_total);
+        auto conv_cell = [&](const gpuc_cell& C, int64_t* ts, uint32_t* ldt, int32_t* ttl,
+                             uint64_t* va, uint32_t* vl, uint8_t* cf) {
+            if (!(C.flags & GPUC_CELLF_PRESENT)) { *cf = 0; return; }
+            uint8_t f = CELLF_PRESENT;
+            if (C.flags & GPUC_CELLF_HAS_VALUE) f |= CELLF_HAS_VALUE;
+            if (C.ttl != 0) f |= CELLF_EXPIRING;
+            *cf = f;
+            *ts = C.ts;
+            *ldt = C.ldt;
+            *ttl = C.ttl;
+            *vl = C.value_len;
+            *va = (C.flags & GPUC_CELLF_HAS_VALUE) ? stash(C.value, C.value_len)
+                                                   : (uint64_t)VBASE;
+        };
+        uint64_t ob = 0, xcur = 0;
+        for (uint64_t si = 0; si < n_parts; si++) {
+            const uint64_t p = sorted[si].idx;
+            const gpuc_flush_part& P = parts[p];
+            h_keypfx[si] = sorted[si].pfx;
+            h_token[si] = (int64_t)(sorted[si].tok ^ 0x8000000000000000ULL);
+            h_keyaddr[si] = (uint64_t)(d_keys.as<uint8_t>() + koff[p]);
+            h_klen[si] = P.key_len;
+            h_pdelm[si] = P.pdel_mfda;
+            h_pdell[si] = P.pdel_ldt;
+            h_rowbase[si] = ob;
+            h_rowcnt[si] = (uint32_t)P.n_unf;
+            if (S.n_static && P.static_flags) {
+                h_sflags[si] = P.static_flags & 7;  // PF_HAS_ROW|PF_LIVE_TS|PF_ROW_DEL
+                h_sts[si] = P.static_live_ts;
+                h_sttl[si] = P.static_live_ttl;
+                h_slet[si] = P.static_live_let;
+                h_sdm[si] = P.static_del_mfda;
+                h_sdl[si] = P.static_del_ldt;
+                for (uint32_t c = 0; c < S.n_static; c++)
+                    conv_cell(P.static_cells[c], &h_scts[si * NST + c], &h_scldt[si * NST + c],
+                              &h_scttl[si * NST + c], &h_scva[si * NST + c], &h_scvl[si * NST + c],
+                              &h_scf[si * NST + c]);
+            }
+            for (uint64_t r = 0; r < P.n_unf; r++, ob++) {
+                const gpuc_unfiltered& U = P.unf[r];
+                h_rkind[ob] = U.kind;
+                h_ckcnt[ob] = U.ck_count;
+                for (uint32_t c = 0; c < U.ck_count; c++) {
+                    uint64_t oc = ob * NCK + c;
+                    int32_t w = ckw_h[c];
+                    if (w > 0) {
+                        h_ck[oc] = h_ck_sortable(U.ck[c], w);
+                        h_ckaddr[oc] = 0;
+                        h_cklen[oc] = (uint32_t)w;
+                    } else {
+                        h_ck[oc] = h_ck_prefix_var(U.ck[c], U.ck_lens[c]);
+                        h_ckaddr[oc] = stash(U.ck[c], U.ck_lens[c]);
+                        h_cklen[oc] = U.ck_lens[c];
+                    }
+                }
+                if (U.kind == 4) {
+                    uint8_t f = PF_HAS_ROW;
+                    if (U.row_flags & GPUC_ROWF_LIVE_TS) {
+                        f |= PF_LIVE_TS;
+                        h_lts[ob] = U.live_ts;
+                        h_lttl[ob] = U.live_ttl;
+                        h_llet[ob] = U.live_let;
+                    }
+                    if (U.row_flags & GPUC_ROWF_DELETED) {
+                        f |= PF_ROW_DEL;
+                        h_rdm[ob] = U.del_mfda;
+                        h_rdl[ob] = U.del_ldt;
+                    }
+                    for (uint32_t c = 0; c < NSIMPLE; c++)
+                        conv_cell(U.cells[c], &h_cts[ob * NCV + c], &h_cldt[ob * NCV + c],
+                                  &h_cttl[ob * NCV + c], &h_cva[ob * NCV + c],
+                                  &h_cvl[ob * NCV + c], &h_cf[ob * NCV + c]);
+                    if (S.n_cpx) {
+                        h_xdm[ob] = INT64_MIN;
+                        h_xdl[ob] = LDT_NONE_U32;
+                        h_xstart[ob] = xcur;
+                        h_xcnt[ob] = 0;
+                        if (U.has_cpx) {
+                            f |= PF_HAS_CPX;
+                            h_xdm[ob] = U.cpx_del_mfda;
+                            h_xdl[ob] = U.cpx_del_ldt;
+                            h_xcnt[ob] = U.n_cpx_cells;
+                            for (uint32_t e = 0; e < U.n_cpx_cells; e++, xcur++) {
+                                const gpuc_cpx_cell& X = U.cpx_cells[e];
+                                conv_cell(X.cell, &h_xts[xcur], &h_xldt[xcur], &h_xttl[xcur],
+                                          &h_xva[xcur], &h_xvl[xcur], &h_xf[xcur]);
+                                h_xpa[xcur] = stash(X.path, X.path_len);
+                                h_xpl[xcur] = X.path_len;
+                            }
+                        }
+                    }
+                    h_flags[ob] = f;
+                } else {
+                    h_flags[ob] = 0;
+                    h_rdm[ob] = U.del_mfda;
+                    h_rdl[ob] = U.del_ldt;
+                    if (U.kind == 2 || U.kind == 5) {  // boundary: open deletion
+                        h_sm[ob] = U.open_mfda;
+                        h_sl[ob] = U.open_ldt;
+                    }
+                    if (S.n_cpx) {
+                        h_xdm[ob] = INT64_MIN;
+                        h_xdl[ob] = LDT_NONE_U32;
+                        h_xstart[ob] = xcur;
+                        h_xcnt[ob] = 0;
+                    }
+                }
+            }
+        }
+        // ---- upload ----
+        auto up = [&](const DevBuf& d, const void* src, size_t len) {
+            if (len) HIP_CHECK(hipMemcpyAsync(d.p, src, len, hipMemcpyHostToDevice, stream));
+        };
+        up(d_vals, h_vals.data(), vcur);
+        up(opb.keypfx, h_keypfx.data(), n_parts * 8);
+        up(opb.key_addr, h_keyaddr.data(), n_parts * 8);
+        up(opb.token, h_token.data(), n_parts * 8);
+        up(opb.klen, h_klen.data(), n_parts * 2);
+        up(opb.pdel_mfda, h_pdelm.data(), n_parts * 8);
+        up(opb.pdel_ldt, h_pdell.data(), n_parts * 4);
+        up(opb.row_base, h_rowbase.data(), n_parts * 8);
+        up(opb.row_count, h_rowcnt.data(), n_parts * 4);
+        up(opb.keep, h_keep.data(), n_parts);
+        if (S.n_static) {
+            up(opb.stb.flags, h_sflags.data(), n_parts);
+            up(opb.stb.live_ts, h_sts.data(), n_parts * 8);
+            up(opb.stb.live_ttl, h_sttl.data(), n_parts * 4);
+            up(opb.stb.live_let, h_slet.data(), n_parts * 8);
+            up(opb.stb.rdel_mfda, h_sdm.data(), n_parts * 8);
+            up(opb.stb.rdel_ldt, h_sdl.data(), n_parts * 4);
+            up(opb.stb.cell_flags, h_scf.data(), n_parts * NST);
+            up(opb.stb.cell_ts, h_scts.data(), n_parts * NST * 8);
+            up(opb.stb.cell_ldt, h_scldt.data(), n_parts * NST * 4);
+            up(opb.stb.cell_ttl, h_scttl.data(), n_parts * NST * 4);
+            up(opb.stb.val_addr, h_scva.data(), n_parts * NST * 8);
+            up(opb.stb.val_len, h_scvl.data(), n_parts * NST * 4);
+        }
+        up(urows.ck, h_ck.data(), n_unf * NCK * 8);
+        up(urows.ck_addr, h_ckaddr.data(), n_unf * NCK * 8);
+        up(urows.ck_len, h_cklen.data(), n_unf * NCK * 4);
+        up(urows.ck_count, h_ckcnt.data(), n_unf);
+        up(urows.rkind, h_rkind.data(), n_unf);
+        up(urows.flags, h_flags.data(), n_unf);
+        up(urows.live_ts, h_lts.data(), n_unf * 8);
+        up(urows.live_ttl, h_lttl.data(), n_unf * 4);
+        up(urows.live_let, h_llet.data(), n_unf * 8);
+        up(urows.rdel_mfda, h_rdm.data(), n_unf * 8);
+        up(urows.rdel_ldt, h_rdl.data(), n_unf * 4);
+        up(urows.start_mfda, h_sm.data(), n_unf * 8);
+        up(urows.start_ldt, h_sl.data(), n_unf * 4);
+        up(urows.cell_ts, h_cts.data(), n_unf * NCV * 8);
+        up(urows.cell_ldt, h_cldt.data(), n_unf * NCV * 4);
+        up(urows.cell_ttl, h_cttl.data(), n_unf * NCV * 4);
+        up(urows.val_addr, h_cva.data(), n_unf * NCV * 8);
+        up(urows.val_len, h_cvl.data(), n_unf * NCV * 4);
+        up(urows.cell_flags, h_cf.data(), n_unf * NCV);
+        if (S.n_cpx) {
+            up(urows.cpx_del_mfda, h_xdm.data(), n_unf * 8);
+            up(urows.cpx_del_ldt, h_xdl.data(), n_unf * 4);
+            up(urows.cpx_start, h_xstart.data(), n_unf * 8);
+            up(urows.cpx_count, h_xcnt.data(), n_unf * 4);
+            if (cpx_total) {
+                up(urows.cx_ts, h_xts.data(), cpx_total * 8);
+                up(urows.cx_ldt, h_xldt.data(), cpx_total * 4);
+                up(urows.cx_ttl, h_xttl.data(), cpx_total * 4);
+                up(urows.cx_flags, h_xf.data(), cpx_total);
+                up(urows.cx_pa, h_xpa.data(), cpx_total * 8);
+                up(urows.cx_pl, h_xpl.data(), cpx_total * 4);
+                up(urows.cx_va, h_xva.data(), cpx_total * 8);
+                up(urows.cx_vl, h_xvl.data(), cpx_total * 4);
+            }
+        }
+        // ---- header mins from a pre-collect (gpuc_generate's recipe) ----
+        DevBuf d_stats, d_tomb;
+        d_stats.alloc(sizeof(OutStats));
+        init_outstats(d_stats, stream);
+        uint32_t tomb_cap = (uint32_t)std::min<uint64_t>(n_unf * 2 + cpx_total + 1024, 400000000ull);
+        d_tomb.alloc((uint64_t)tomb_cap * 4);
+        uint32_t blocks = (uint32_t)((n_parts + 255) / 256);
+        hipLaunchKernelGGL(k_collect_rows, dim3(blocks), dim3(256), 0, stream, opb.op, urows.uc,
+                           n_parts, NCV, d_stats.as<OutStats>(), d_tomb.as<uint32_t>(), tomb_cap);
+        OutStats hs0;
+        HIP_CHECK(hipStreamSynchronize(stream));
+        HIP_CHECK(hipMemcpy(&hs0, d_stats.p, sizeof(OutStats), hipMemcpyDeviceToHost));
+        SerParams2 sp{};
+        sp.hs.min_ts = hs0.min_ts_flip == 0xFFFFFFFFFFFFFFFFULL
+                           ? TIMESTAMP_EPOCH : (int64_t)(hs0.min_ts_flip ^ 0x8000000000000000ULL);
+        sp.hs.min_ldt = hs0.min_ldt_flip == 0xFFFFFFFFFFFFFFFFULL
+                            ? DELETION_TIME_EPOCH : (int64_t)(hs0.min_ldt_flip ^ 0x8000000000000000ULL);
+        if (sp.hs.min_ldt == NO_DELETION_TIME) sp.hs.min_ldt = DELETION_TIME_EPOCH;
+        sp.hs.min_ttl = hs0.min_ttl == 0xFFFFFFFFu ? 0 : (int32_t)hs0.min_ttl;
+        sp.sch.n_ck = S.n_ck;
+        DevBuf d_ckw, d_cf2, d_sf2;
+        d_ckw.alloc(ckw_h.size() * 4 + 8);
+        if (S.n_ck) up(d_ckw, ckw_h.data(), ckw_h.size() * 4);
+        sp.sch.ck_w = d_ckw.as<int32_t>();
+        sp.sch.n_cols = NSIMPLE;
+        d_cf2.alloc(colw_h.size() * 4 + 8);
+        if (NSIMPLE) up(d_cf2, colw_h.data(), colw_h.size() * 4);
+        sp.sch.col_fixed = d_cf2.as<int32_t>();
+        sp.sch.n_static = S.n_static;
+        d_sf2.alloc(staticw_h.size() * 4 + 8);
+        if (S.n_static) up(d_sf2, staticw_h.data(), staticw_h.size() * 4);
+        sp.sch.static_fixed = d_sf2.as<int32_t>();
+        sp.sch.n_cpx = S.n_cpx;
+        sp.sch.counters = counters ? 1u : 0u;
+        sp.sch.column_index_size = S.column_index_size ? S.column_index_size : 64 * 1024;
+        init_outstats(d_stats, stream);
+        write_sstable_device(opb, urows, n_parts, sp, d_stats, d_tomb, tomb_cap, output_base,
+                             key_type, ck_types, regular_cols, static_cols, stream, 0,
+                             S.snappy != 0, S.bti != 0);
+        HIP_CHECK(hipStreamDestroy(stream));
+        return GPUC_OK;
+    } catch (const std::exception& e) {
+        set_err(error, error_len, e.what());
+        return GPUC_ERR_INTERNAL;
+    }
+}
+
 extern "C" int gpuc_validate(const gpuc_job* job, const char* out_path, uint64_t* n_partitions,
                              char* error, size_t error_len) {
     if (job->n_output_shards > 1 || job->n_tomb_sources > 0 || job->n_keep_ranges > 0) {

@@ -216,6 +216,112 @@ typedef struct gpuc_flush_rows {
 int gpuc_flush(const gpuc_flush_rows* rows, const char* output_base, int32_t device,
                char* error, size_t error_len);
 
+/* ---- full-schema memtable flush ----------------------------------------
+ * The real Memtable -> BigTableWriter path (Memtable.FlushablePartitionSet
+ * -> SortedTablePartitionWriter): the caller hands over partitions in any
+ * order (the engine token-sorts them on device, like gpuc_flush), each with
+ * its partition deletion, optional static row, and its unfiltereds (rows +
+ * range-tombstone markers) ALREADY in clustering order — exactly what a
+ * memtable holds (reference memtables are sorted maps; flush iterates them
+ * in order, Memtable.java getFlushSet). The engine validates the
+ * within-partition order and key uniqueness and fails loudly on violations.
+ *
+ * Cell/row field conventions match the serialization they produce:
+ * ldt is the u32 localDeletionTime encoding (GPUC_LDT_NONE = none), ttl 0 =
+ * none, expiring cells carry ttl>0 + ldt = localExpirationTime. A cell with
+ * flags bit HAS_VALUE cleared and ldt set is a cell tombstone. */
+#define GPUC_LDT_NONE 0xFFFFFFFFu
+#define GPUC_CELLF_PRESENT 1u   /* column set on this row */
+#define GPUC_CELLF_HAS_VALUE 2u
+#define GPUC_CELLF_EXPIRING 4u
+#define GPUC_ROWF_HAS_ROW 1u    /* always set for rows */
+#define GPUC_ROWF_LIVE_TS 2u    /* primaryKeyLivenessInfo present */
+#define GPUC_ROWF_DELETED 4u    /* row deletion present */
+
+typedef struct gpuc_cell {
+    uint8_t flags;              /* GPUC_CELLF_*; 0 = column absent */
+    int64_t ts;
+    uint32_t ldt;               /* GPUC_LDT_NONE = live */
+    int32_t ttl;                /* 0 = none */
+    const uint8_t* value;
+    uint32_t value_len;
+} gpuc_cell;
+
+typedef struct gpuc_cpx_cell {  /* one complex (collection) cell */
+    gpuc_cell cell;
+    const uint8_t* path;        /* CellPath: the map key bytes */
+    uint32_t path_len;
+} gpuc_cpx_cell;
+
+typedef struct gpuc_unfiltered {
+    uint8_t kind;               /* ClusteringPrefix.Kind ordinal
+                                 * (ClusteringPrefix.java:65-85): 4 = row,
+                                 * 0/1/6/7 bounds, 2/5 boundaries */
+    uint8_t ck_count;           /* clustering components present (rows: n_ck;
+                                 * marker bounds may be shorter prefixes) */
+    uint8_t row_flags;          /* GPUC_ROWF_* (rows only) */
+    const uint8_t* const* ck;   /* per component: serialized value bytes */
+    const uint32_t* ck_lens;
+    int64_t live_ts;            /* with GPUC_ROWF_LIVE_TS */
+    int32_t live_ttl;           /* >0 = ExpiringLivenessInfo */
+    int64_t live_let;           /* localExpirationTime (long semantics) */
+    int64_t del_mfda;           /* row deletion / marker close(or single) */
+    uint32_t del_ldt;
+    int64_t open_mfda;          /* boundary markers: open deletion */
+    uint32_t open_ldt;
+    const gpuc_cell* cells;     /* rows: schema->n_cols entries (the complex
+                                 * column's entry is ignored; see cpx_*) */
+    uint8_t has_cpx;            /* complex column present on this row */
+    int64_t cpx_del_mfda;       /* complexDeletion */
+    uint32_t cpx_del_ldt;
+    uint32_t n_cpx_cells;
+    const gpuc_cpx_cell* cpx_cells;  /* in CellPath order */
+} gpuc_unfiltered;
+
+typedef struct gpuc_flush_part {
+    const uint8_t* key;
+    uint16_t key_len;
+    int64_t pdel_mfda;          /* partition deletion (INT64_MIN = live) */
+    uint32_t pdel_ldt;
+    uint8_t static_flags;       /* GPUC_ROWF_*; 0 = no/empty static row */
+    int64_t static_live_ts;
+    int32_t static_live_ttl;
+    int64_t static_live_let;
+    int64_t static_del_mfda;
+    uint32_t static_del_ldt;
+    const gpuc_cell* static_cells;  /* schema->n_static entries when
+                                     * static_flags != 0 */
+    uint64_t n_unf;
+    const gpuc_unfiltered* unf; /* clustering order */
+} gpuc_flush_part;
+
+typedef struct gpuc_flush_schema {
+    /* AbstractType class names as the reference serializes them into the
+     * Statistics.db HEADER (e.g. "org.apache.cassandra.db.marshal.LongType");
+     * fixed widths are derived from the type (LongType 8, Int32Type 4,
+     * otherwise variable). NULL key_type = BytesType. */
+    const char* key_type;
+    uint32_t n_ck;
+    const char* const* ck_types;
+    uint32_t n_cols;            /* regular columns, header order; a complex
+                                 * (MapType) column must be LAST */
+    const uint8_t* const* col_names;
+    const uint32_t* col_name_lens;
+    const char* const* col_types;
+    uint32_t n_static;
+    const uint8_t* const* static_names;
+    const uint32_t* static_name_lens;
+    const char* const* static_types;
+    uint32_t n_cpx;             /* 0/1: last regular column is complex */
+    uint32_t column_index_size; /* 0 = 64 KiB */
+    uint32_t snappy;            /* chunk codec for the output */
+    uint32_t bti;               /* write the `da` (trie-indexed) components */
+} gpuc_flush_schema;
+
+int gpuc_flush_table(const gpuc_flush_schema* schema, const gpuc_flush_part* parts,
+                     uint64_t n_partitions, const char* output_base, int32_t device,
+                     char* error, size_t error_len);
+
 /* Scrub one sstable (SortedTableScrubber): salvage every partition whose
  * byte range touches only CRC/decode-clean 16 KiB chunks and rewrite them as
  * a clean sstable under output_base (recovery granularity documented in

@@ -175,6 +175,10 @@ static int cmd_gen(const std::string& outdir, std::map<std::string, std::string>
         std::string base = g.bti ? outdir + "/da-" + std::to_string(t.generation) + "-bti"
                                  : outdir + "/oa-" + std::to_string(t.generation) + "-big";
         write_components(w, base);
+        if (kv.count("dump")) {
+            t.bti = g.bti != 0;  // write_sstable takes the flag separately
+            write_memdump(t, base + ".memdump");
+        }
         total_unc += w.uncompressed_data_len;
         printf("wrote %s: parts=%llu uncompressed=%llu compressed=%zu\n", base.c_str(),
                (unsigned long long)w.partition_count, (unsigned long long)w.uncompressed_data_len,

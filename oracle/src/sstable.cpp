@@ -1280,4 +1280,120 @@ SSTable read_sstable(const std::string& base, bool keep_raw) {
     return t;
 }
 
+
+// ---- memtable dump (flush-parity interchange) ----------------------------
+// Serializes the in-memory SSTable (the logical memtable content) into a
+// flat little-endian file that the GPU engine's gpuc_flush_table parity
+// tests re-marshal through the C ABI. Test infrastructure: the format is
+// ours (not a Cassandra component); the parity anchor is that
+// gpuc_flush_table(memdump of t) must reproduce write_sstable(t) byte-
+// for-byte.
+namespace {
+void md_u8(bytes& o, uint8_t v) { o.push_back(v); }
+void md_u32(bytes& o, uint32_t v) { for (int i = 0; i < 4; i++) o.push_back((uint8_t)(v >> (8 * i))); }
+void md_i32(bytes& o, int32_t v) { md_u32(o, (uint32_t)v); }
+void md_u64(bytes& o, uint64_t v) { for (int i = 0; i < 8; i++) o.push_back((uint8_t)(v >> (8 * i))); }
+void md_i64(bytes& o, int64_t v) { md_u64(o, (uint64_t)v); }
+void md_blob(bytes& o, const bytes& b) { md_u32(o, (uint32_t)b.size()); o.insert(o.end(), b.begin(), b.end()); }
+void md_cell(bytes& o, const Cell& c) {
+    uint8_t cf = 1;                              // CELLF_PRESENT
+    if (!c.tombstone()) cf |= 2;                 // CELLF_HAS_VALUE
+    if (c.expiring()) cf |= 4;                   // CELLF_EXPIRING
+    md_u8(o, cf);
+    md_i64(o, c.ts);
+    md_u32(o, c.ldt);
+    md_i32(o, c.ttl);
+    md_blob(o, c.value);
+}
+void md_clustering(bytes& o, const Clustering& c) {
+    md_u8(o, (uint8_t)c.size());
+    for (const auto& v : c) {
+        md_u8(o, (uint8_t)v.state);
+        md_blob(o, v.v);
+    }
+}
+void md_row_head(bytes& o, const Row& r) {
+    uint8_t f = 1;                               // PF_HAS_ROW analog
+    if (!r.live.empty()) f |= 2;                 // PF_LIVE_TS
+    if (!r.del.live()) f |= 4;                   // PF_ROW_DEL
+    md_u8(o, f);
+    md_i64(o, r.live.ts);
+    md_i32(o, r.live.ttl);
+    md_i64(o, r.live.let);
+    md_i64(o, r.del.mfda);
+    md_u32(o, r.del.ldt);
+}
+}  // namespace
+
+void write_memdump(const SSTable& t, const std::string& path) {
+    bytes o;
+    o.insert(o.end(), {'G', 'M', 'D', '1'});
+    md_u8(o, (uint8_t)t.header.key_type);
+    md_u32(o, (uint32_t)t.header.clustering_types.size());
+    for (CqlType ct : t.header.clustering_types) md_u8(o, (uint8_t)ct);
+    md_u32(o, (uint32_t)t.header.static_cols.size());
+    for (auto& [nm, ct] : t.header.static_cols) { md_blob(o, nm); md_u8(o, (uint8_t)ct); }
+    md_u32(o, (uint32_t)t.header.regular_cols.size());
+    for (auto& [nm, ct] : t.header.regular_cols) { md_blob(o, nm); md_u8(o, (uint8_t)ct); }
+    md_u8(o, t.comp.algo == Compressor::SNAPPY ? 1 : 0);
+    md_u8(o, t.bti ? 1 : 0);
+    md_u32(o, t.column_index_size);
+    md_u64(o, t.parts.size());
+    const auto& regs = t.header.regular_cols;
+    for (const Partition& p : t.parts) {
+        md_u32(o, (uint32_t)p.key.size());
+        o.insert(o.end(), p.key.begin(), p.key.end());
+        md_i64(o, p.del.mfda);
+        md_u32(o, p.del.ldt);
+        if (!t.header.static_cols.empty()) {
+            const Row& sr = p.static_row;
+            if (row_is_empty(sr)) {
+                md_u8(o, 0);
+            } else {
+                md_row_head(o, sr);
+                for (size_t i = 0; i < t.header.static_cols.size(); i++) {
+                    if (i < sr.cells.size() && sr.cells[i]) md_cell(o, *sr.cells[i]);
+                    else md_u8(o, 0);
+                }
+            }
+        }
+        md_u32(o, (uint32_t)p.items.size());
+        for (const Unfiltered& u : p.items) {
+            if (u.kind == Unfiltered::ROW) {
+                md_u8(o, 0);
+                md_clustering(o, u.row.clustering);
+                md_row_head(o, u.row);
+                for (size_t i = 0; i < regs.size(); i++) {
+                    if (regs[i].second == CqlType::MAP_BB) {
+                        const auto& cd = i < u.row.complex.size() ? u.row.complex[i]
+                                                                  : std::optional<ComplexData>{};
+                        if (!cd) { md_u8(o, 0); continue; }
+                        md_u8(o, 1);
+                        md_i64(o, cd->del.mfda);
+                        md_u32(o, cd->del.ldt);
+                        md_u32(o, (uint32_t)cd->cells.size());
+                        for (const Cell& c : cd->cells) {
+                            md_cell(o, c);
+                            md_blob(o, c.path);
+                        }
+                    } else if (i < u.row.cells.size() && u.row.cells[i]) {
+                        md_cell(o, *u.row.cells[i]);
+                    } else {
+                        md_u8(o, 0);
+                    }
+                }
+            } else {
+                md_u8(o, 1);
+                md_u8(o, (uint8_t)u.marker.kind);
+                md_clustering(o, u.marker.values);
+                md_i64(o, u.marker.end_dt.mfda);
+                md_u32(o, u.marker.end_dt.ldt);
+                md_i64(o, u.marker.start_dt.mfda);
+                md_u32(o, u.marker.start_dt.ldt);
+            }
+        }
+    }
+    write_file(path, o);
+}
+
 }  // namespace oracle

@@ -279,6 +279,8 @@ struct WriterOut {
 // and write_components emits the `da` component set instead of Index/Summary.
 WriterOut write_sstable(const SSTable& t, bool bti = false);
 void write_components(const WriterOut& w, const std::string& base);
+// memtable-dump interchange for gpuc_flush_table parity (format ours; see .cpp)
+void write_memdump(const SSTable& t, const std::string& path);
 
 // serialize one partition into `out` (Data.db stream) and append its Index.db
 // entry to `index_out`. Exposed for round-trip tests.

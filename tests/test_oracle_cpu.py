@@ -553,3 +553,34 @@ def test_counter_columns_oracle(oracle_bin, tmp_path):
     for c in ("Data.db", "Index.db", "Digest.crc32"):
         assert filecmp.cmp(f"{d}/oa-90-big-{c}", f"{d}/oa-91-big-{c}", shallow=False), c
     assert "FAIL" not in oracle_run("roundtrip", f"{d}/oa-90-big").stdout
+
+
+def test_memdump_roundtrip_parse(oracle_bin, tmp_path):
+    """gen dump=1 writes a .memdump of the logical memtable content; the
+    host-side parser must consume it fully and reconstruct the schema across
+    every feature combination (the GPU flush_table parity test reuses it)."""
+    import cassandra_amd as ca
+    cases = [
+        (dict(seed=7, rows=120, vlen=50, tomb=15, pdel=3), dict(n_ck=0, n_cols=1)),
+        (dict(seed=8, rows=12, crows=10, vlen=60, rtomb=40, tomb=10, statics=40,
+              cktext=1, ttl=30), dict(n_ck=1, n_static=1)),
+        (dict(seed=9, rows=80, cpx=45, cpxdel=25, vlen=40), dict(n_cpx=1, n_cols=2)),
+        (dict(seed=10, rows=80, counter=1), dict(n_cols=1)),
+        (dict(seed=11, rows=20, crows=8, ckcols=2, ncols=4, colmiss=30, snappy=1),
+         dict(n_ck=2, n_cols=4, snappy=1)),
+        (dict(seed=12, rows=60, vlen=30, bti=1), dict(bti=1)),
+    ]
+    for i, (genkw, want) in enumerate(cases):
+        d = os.path.join(str(tmp_path), str(i))
+        os.makedirs(d)
+        args = [f"{k}={v}" for k, v in dict(n=1, dump=1, **genkw).items()]
+        subprocess.run([oracle_bin, "gen", d, *args], check=True, capture_output=True)
+        stem = "da-1-bti" if genkw.get("bti") else "oa-1-big"
+        data = open(f"{d}/{stem}.memdump", "rb").read()
+        S, parts, n, bufs = ca._parse_memdump(data)  # raises on trailing bytes
+        assert n > 0
+        for k, v in want.items():
+            assert getattr(S, k) == v, (i, k, getattr(S, k), v)
+        # spot-check the first partition graph is materialized
+        assert parts[0].key_len > 0
+        del bufs
