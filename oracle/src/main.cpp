@@ -53,7 +53,7 @@ static int cmd_roundtrip(const std::string& base) {
         if (index != want_idx) {
             bytes d2, i2;
             for (auto& p : t.parts) serialize_partition(p, t.header, d2, i2, 4096);
-            if (i2 == want_idx) index = std::move(i2);
+            if (i2 == want_idx) { index = std::move(i2); t.column_index_size = 4096; }
         }
     } else {
         // column_index_size is config, not recorded: try the 64 KiB default,
@@ -118,6 +118,7 @@ static int cmd_roundtrip(const std::string& base) {
             return bytes(st.begin() + start, st.begin() + end2 - 4);  // strip component CRC
         };
         check("Stats[COMPACTION]", comp_slice(w.statistics), comp_slice(t.raw_statistics));
+        if (!t.bti) check("Summary.db", w.summary, read_file(base + "-Summary.db"));
     }
     return rc;
 }

@@ -862,10 +862,12 @@ WriterOut write_sstable(const SSTable& t, bool bti) {
     st.mins = StatsMins{};
     uint64_t total_cells = 0, total_rows = 0, total_cols_set = 0;
     bool has_partition_deletions = false;
+    std::vector<uint64_t> index_entry_pos;
     for (const Partition& p : t.parts) {
         size_t before = data_raw.size();
         part_pos.push_back(before);
         part_blocks.emplace_back();
+        index_entry_pos.push_back(w.index_db.size());
         serialize_partition_ex(p, t.header, data_raw, w.index_db, t.column_index_size,
                                bti ? &part_blocks.back() : nullptr);
         bloom.add(p.key);
@@ -949,13 +951,47 @@ WriterOut write_sstable(const SSTable& t, bool bti) {
     }
     st.has_partition_deletions = has_partition_deletions;
     w.statistics = serialize_statistics(t.header, st);
-    // Summary.db + TOC.txt: minimal valid placeholders (not in bit-exact set).
+    // Summary.db: IndexSummary at BASE_SAMPLING_LEVEL — entries for keys
+    // 0, 128, 256, ... (IndexSummaryBuilder.maybeAddEntry with empty
+    // Downsampling start points at full sampling), serialized per
+    // IndexSummary.IndexSummarySerializer.serialize: BE header
+    // (minIndexInterval, offsetCount, offHeapSize, samplingLevel,
+    // sizeAtFullSampling), then the off-heap image in NATIVE (LE) order —
+    // per-entry offsets rebased by 4*offsetCount, then key bytes + LE
+    // position of the entry in Index.db.
     {
         bytes& s = w.summary;
-        put_be32(s, 128);                       // minIndexInterval
-        put_be32(s, 0);                         // entries (unsampled placeholder)
-        put_be64(s, 8);                         // offheap size of entries section
-        put_be32(s, 128); put_be32(s, 0);       // samplingLevel? sizeAtFullSampling?
+        const uint32_t MIN_INTERVAL = 128;
+        std::vector<uint64_t> sampled;
+        for (uint64_t k = 0; k < t.parts.size(); k += MIN_INTERVAL) sampled.push_back(k);
+        uint32_t cnt = (uint32_t)sampled.size();
+        bytes entries;
+        std::vector<uint32_t> offs;
+        for (uint64_t k : sampled) {
+            offs.push_back((uint32_t)entries.size());
+            const bytes& kb = t.parts[k].key;
+            entries.insert(entries.end(), kb.begin(), kb.end());
+            uint64_t pos = index_entry_pos[k];
+            for (int b = 0; b < 8; b++) entries.push_back((uint8_t)(pos >> (8 * b)));
+        }
+        put_be32(s, MIN_INTERVAL);
+        put_be32(s, cnt);
+        put_be64(s, 4ull * cnt + entries.size());
+        put_be32(s, 128);                        // samplingLevel = BASE
+        put_be32(s, cnt);                        // sizeAtFullSampling
+        for (uint32_t o : offs) {
+            uint32_t v = o + 4 * cnt;
+            for (int b = 0; b < 4; b++) s.push_back((uint8_t)(v >> (8 * b)));  // LE
+        }
+        s.insert(s.end(), entries.begin(), entries.end());
+        if (!t.parts.empty()) {
+            // first/last key with 4-byte BE lengths (SSTableReader.saveSummary:
+            // ByteBufferUtil.writeWithLength)
+            put_be32(s, (uint32_t)t.parts.front().key.size());
+            s.insert(s.end(), t.parts.front().key.begin(), t.parts.front().key.end());
+            put_be32(s, (uint32_t)t.parts.back().key.size());
+            s.insert(s.end(), t.parts.back().key.begin(), t.parts.back().key.end());
+        }
         std::string toc = w.bti ? "Data.db\nStatistics.db\nDigest.crc32\nTOC.txt\nCompressionInfo.db\nFilter.db\nPartitions.db\nRows.db\n"
                                 : "Data.db\nStatistics.db\nDigest.crc32\nTOC.txt\nCompressionInfo.db\nFilter.db\nIndex.db\nSummary.db\n";
         w.toc.assign(toc.begin(), toc.end());
