@@ -1894,8 +1894,6 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         // data; the rest are tombstone sources merged in a second pass.
         int kd = su.k_data > 0 ? su.k_data : k;
         bool gc_mode = kd < k;
-        if (gc_mode && su.n_cpx)
-            throw std::runtime_error("garbage-collect mode with complex columns unsupported");
         if (su.bti && gc_mode)
             throw std::runtime_error("garbage-collect mode with bti (da) inputs unsupported");
         uint64_t data_parts = 0;
@@ -1960,6 +1958,7 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         d_stats.alloc(sizeof(OutStats));
         init_outstats(d_stats, stream);
         DevBuf d_ctr_arena;
+        uint64_t ctr_arena_cap = 0;
         if (sch.counters) {
             // merged-context arena: capacity = total input counter bytes
             // (any k-way merge result is bounded by the sum of its inputs)
@@ -1975,7 +1974,8 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
             unsigned long long total_ctr = 0;
             HIP_CHECK(hipStreamSynchronize(stream));
             HIP_CHECK(hipMemcpy(&total_ctr, d_sum.p, 8, hipMemcpyDeviceToHost));
-            d_ctr_arena.alloc(total_ctr + 64);
+            ctr_arena_cap = total_ctr + 64;
+            d_ctr_arena.alloc(ctr_arena_cap);
             sch.ctr_arena = d_ctr_arena.as<uint8_t>();
         }
         uint32_t tomb_cap = (uint32_t)std::min<uint64_t>(total_out_rows * 2 + n_groups + 1024, 400000000ull);
@@ -2102,6 +2102,7 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         UnfColsBuf* rows_for_writer = &out_rows;
         OutPartsBuf opb_t;
         UnfColsBuf out_rows_t;
+        DevBuf d_ctr_arena_t;
         if (gc_mode && n_groups > 0) {
             uint64_t src_parts = total_parts - data_parts;
             uint64_t n_groups_t = 0;
@@ -2135,9 +2136,18 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
                 }
                 uint64_t t_out_rows = exscan_u64(d_grows_t.as<uint64_t>(), n_groups_t, stream);
                 opb_t.alloc(n_groups_t ? n_groups_t : 1, sch.n_static);
-                out_rows_t.alloc(t_out_rows, sch.n_cols, sch.n_ck);
+                out_rows_t.alloc(t_out_rows, sch.n_cols, sch.n_ck, sch.n_cpx);
+                if (sch.n_cpx) out_rows_t.alloc_cpx_arena(total_in_cpx);
                 d_stats_t.alloc(sizeof(OutStats));
                 init_outstats(d_stats_t, stream);
+                // the source pass bump-allocates merged counter contexts from
+                // its own arena (the main pass's bump counter lives in the
+                // OTHER OutStats, so sharing one arena would overlap)
+                SchemaParams sch_t = sch;
+                if (sch.counters) {
+                    d_ctr_arena_t.alloc(ctr_arena_cap);
+                    sch_t.ctr_arena = d_ctr_arena_t.as<uint8_t>();
+                }
                 PurgeParams2 pp_src = pp_data;
                 pp_src.has_shard = 0;  // sources shadow regardless of the shard
                 {
@@ -2146,7 +2156,7 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
                         hipLaunchKernelGGL(kern, dim3(blocks_g), dim3(256), 0, stream, d_sorted_t,
                                            d_gstart_t.as<uint64_t>(), n_groups_t, src_parts,
                                            d_srcbases.as<uint32_t>(), pc, in_rows.uc, opb_t.op,
-                                           out_rows_t.uc, d_grows_t.as<uint64_t>(), sch, pp_src,
+                                           out_rows_t.uc, d_grows_t.as<uint64_t>(), sch_t, pp_src,
                                            d_stats_t.as<OutStats>(),
                                            d_error.as<unsigned long long>());
                     };
@@ -2167,7 +2177,10 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
                                    d_cap.as<uint64_t>());
             }
             uint64_t gc_total = exscan_u64(d_cap.as<uint64_t>(), n_groups, stream);
-            gc_rows.alloc(gc_total, sch.n_cols, sch.n_ck);
+            gc_rows.alloc(gc_total, sch.n_cols, sch.n_ck, sch.n_cpx);
+            // the cpx CELL arena is shared: the filter compacts each row's
+            // own [start,count) segment of out_rows' arena in place
+            if (sch.n_cpx) gc_rows.uc.cpx = out_rows.uc.cpx;
             {
                 uint32_t blocks_g = (uint32_t)((n_groups + 255) / 256);
                 hipLaunchKernelGGL(k_garbage_filter, dim3(blocks_g), dim3(256), 0, stream, opb.op,

@@ -2674,6 +2674,94 @@ __device__ inline bool gc_cell_b_wins(int64_t ats, uint32_t aldt, bool aexp, uin
     return cmp_values(av, al, bv, bl) < 0;
 }
 
+// pairwise CounterContext.merge relationship (CounterContext.merge over the
+// ContextState walk; same lattice as ctx_compare in the oracle): 0 = merge
+// returns LEFT's exact bytes (left superset or equal contexts — equality
+// resolves left, matching the reference's left-identity-first check),
+// 1 = returns RIGHT's bytes, 2 = fresh merged bytes.
+__device__ inline int gc_ctx_relationship(const uint8_t* lp, uint32_t ll,
+                                          const uint8_t* rp, uint32_t rl) {
+    CtxSt L, R;
+    L.init(lp, ll);
+    R.init(rp, rl);
+    bool lsup = true, rsup = true;
+    while (L.has() && R.has()) {
+        int c = ctx_idcmp(L.id(), R.id());
+        if (c == 0) {
+            int rel;
+            int64_t lc = L.clock(), lk = L.count(), rc = R.clock(), rk = R.count();
+            if (L.g || R.g) {
+                if (L.g && R.g)
+                    rel = lc == rc ? (lk > rk ? 1 : lk == rk ? 0 : -1) : (lc > rc ? 1 : -1);
+                else
+                    rel = L.g ? 1 : -1;
+            } else if (L.l || R.l) {
+                rel = (L.l && R.l) ? 2 : (L.l ? 1 : -1);
+            } else if (lc == rc) {
+                rel = lk > rk ? 1 : lk == rk ? 0 : -1;
+            } else {
+                rel = ctx_remote_better(lc, lk, rc, rk) ? 1 : -1;
+            }
+            if (rel == 2) lsup = rsup = false;
+            else if (rel > 0) rsup = false;
+            else if (rel < 0) lsup = false;
+            L.next();
+            R.next();
+        } else if (c < 0) {
+            rsup = false;
+            L.next();
+        } else {
+            lsup = false;
+            R.next();
+        }
+    }
+    if (L.has()) rsup = false;
+    if (R.has()) lsup = false;
+    return lsup ? 0 : rsup ? 1 : 2;
+}
+
+// Cells.addNonShadowed identity decision for counter cells (resolveCounter,
+// Cells.java:121-162): data cell dropped iff reconcile returns the source
+// cell itself. Counter cells never expire, so tombstone == ldt set.
+__device__ inline bool gc_ctr_b_wins(int64_t ats, uint32_t aldt, uint64_t av, uint32_t al,
+                                     int64_t bts, uint32_t bldt, uint64_t bv, uint32_t bl) {
+    bool at = aldt != LDT_NONE_U32, bt = bldt != LDT_NONE_U32;
+    if (at || bt) {
+        if (at != bt) return bt;  // tombstone beats any live counter
+        return gc_cell_b_wins(ats, aldt, false, av, al, bts, bldt, false, bv, bl);
+    }
+    if (al == 0 || bl == 0) {  // non-empty wins (documented divergence)
+        if ((al == 0) != (bl == 0)) return al == 0;
+        return !(ats > bts);
+    }
+    int rel = gc_ctx_relationship((const uint8_t*)av, al, (const uint8_t*)bv, bl);
+    return rel == 1 && (ats > bts ? ats : bts) == bts;
+}
+
+// move one complex cell within the shared cpx arena (row-local compaction)
+__device__ inline void gc_cpx_move(UnfCols& u, uint64_t to, uint64_t from) {
+    if (to == from) return;
+    u.cpx.ts[to] = u.cpx.ts[from];
+    u.cpx.ldt[to] = u.cpx.ldt[from];
+    u.cpx.ttl[to] = u.cpx.ttl[from];
+    u.cpx.flags[to] = u.cpx.flags[from];
+    u.cpx.path_addr[to] = u.cpx.path_addr[from];
+    u.cpx.path_len[to] = u.cpx.path_len[from];
+    u.cpx.val_addr[to] = u.cpx.val_addr[from];
+    u.cpx.val_len[to] = u.cpx.val_len[from];
+}
+
+// CellPath order (BytesType map keys): unsigned lex, shorter-first on ties
+__device__ inline int gc_path_cmp(const UnfCols& a, uint64_t ae, const UnfCols& b, uint64_t be) {
+    const uint8_t* pa = (const uint8_t*)a.cpx.path_addr[ae];
+    const uint8_t* pb = (const uint8_t*)b.cpx.path_addr[be];
+    uint32_t la = a.cpx.path_len[ae], lb = b.cpx.path_len[be];
+    uint32_t n = la < lb ? la : lb;
+    for (uint32_t i = 0; i < n; i++)
+        if (pa[i] != pb[i]) return pa[i] < pb[i] ? -1 : 1;
+    return la == lb ? 0 : (la < lb ? -1 : 1);
+}
+
 // per data partition: match to a source partition by decorated key, produce
 // the filtered capacity (data rows + matched tomb rows for reissue headroom)
 __global__ void k_gc_match(OutParts dp, uint64_t nd, OutParts tp, uint64_t nt,
@@ -2733,9 +2821,31 @@ __device__ inline bool gc_row_filter_active(UnfCols& out, uint64_t o, int64_t am
             if (out.cell_ts[oc] <= am) out.cell_flags[oc] = 0;
             else any_cell = true;
         }
+        if (sch.n_cpx && (f & PF_HAS_CPX)) {
+            // ComplexColumnData.filter(all, activeDeletion, null, -)
+            // (ComplexColumnData.java:188-210): drop a shadowed
+            // complexDeletion, drop cells activeDeletion deletes
+            if (dtp_sup(am, al, out.cpx_del_mfda[o], out.cpx_del_ldt[o])) {
+                out.cpx_del_mfda[o] = INT64_MIN;
+                out.cpx_del_ldt[o] = LDT_NONE_U32;
+            }
+            uint64_t s0 = out.cpx_start[o];
+            uint32_t nc2 = out.cpx_count[o], w = 0;
+            for (uint32_t e = 0; e < nc2; e++) {
+                if (out.cpx.ts[s0 + e] <= am) continue;
+                gc_cpx_move(out, s0 + w, s0 + e);
+                w++;
+            }
+            out.cpx_count[o] = w;
+            bool cpx_alive = w || !(out.cpx_del_mfda[o] == INT64_MIN &&
+                                    out.cpx_del_ldt[o] == LDT_NONE_U32);
+            if (!cpx_alive) f &= ~PF_HAS_CPX;
+            else any_cell = true;
+        }
     } else {
         for (uint32_t c = 0; c < sch.n_cols; c++)
             if (out.cell_flags[o * sch.n_cols + c] & CELLF_PRESENT) any_cell = true;
+        if (sch.n_cpx && (f & PF_HAS_CPX)) any_cell = true;
     }
     bool keep = (f & (PF_LIVE_TS | PF_ROW_DEL)) || any_cell;
     out.flags[o] = keep ? (f | PF_HAS_ROW) : 0;
@@ -2772,14 +2882,71 @@ __device__ inline bool gc_filter_row_vs(UnfCols& out, uint64_t o, const UnfCols&
         if (!(cf & CELLF_PRESENT)) continue;
         if (out.cell_ts[oc] <= dm) { out.cell_flags[oc] = 0; continue; }
         uint8_t tf = tin.cell_flags[tc];
-        if ((tf & CELLF_PRESENT) &&
-            gc_cell_b_wins(out.cell_ts[oc], out.cell_ldt[oc], cf & CELLF_EXPIRING,
-                           out.val_addr[oc], out.val_len[oc], tin.cell_ts[tc], tin.cell_ldt[tc],
-                           tf & CELLF_EXPIRING, tin.val_addr[tc], tin.val_len[tc])) {
-            out.cell_flags[oc] = 0;
-            continue;
+        if (tf & CELLF_PRESENT) {
+            bool bw = sch.counters
+                          ? gc_ctr_b_wins(out.cell_ts[oc], out.cell_ldt[oc], out.val_addr[oc],
+                                          out.val_len[oc], tin.cell_ts[tc], tin.cell_ldt[tc],
+                                          tin.val_addr[tc], tin.val_len[tc])
+                          : gc_cell_b_wins(out.cell_ts[oc], out.cell_ldt[oc],
+                                           cf & CELLF_EXPIRING, out.val_addr[oc],
+                                           out.val_len[oc], tin.cell_ts[tc], tin.cell_ldt[tc],
+                                           tf & CELLF_EXPIRING, tin.val_addr[tc],
+                                           tin.val_len[tc]);
+            if (bw) {
+                out.cell_flags[oc] = 0;
+                continue;
+            }
         }
         any_cell = true;
+    }
+    if (sch.n_cpx && (f & PF_HAS_CPX)) {
+        // complex branch of Rows.removeShadowedCells (Rows.java:298-316):
+        // data complexDeletion survives iff it supersedes
+        // max(updateDt, deletion) and then raises the bar; then per-path
+        // Cells.addNonShadowedComplex against the source's cells
+        bool b_has = tin.flags[t] & PF_HAS_CPX;
+        int64_t udm = b_has ? tin.cpx_del_mfda[t] : INT64_MIN;
+        uint32_t udl = b_has ? tin.cpx_del_ldt[t] : LDT_NONE_U32;
+        int64_t mm = dm;
+        uint32_t ml = dl;
+        if (dtp_sup(udm, udl, mm, ml)) { mm = udm; ml = udl; }
+        bool keep_del = dtp_sup(out.cpx_del_mfda[o], out.cpx_del_ldt[o], mm, ml);
+        if (keep_del) {
+            mm = out.cpx_del_mfda[o];
+            ml = out.cpx_del_ldt[o];
+        } else {
+            out.cpx_del_mfda[o] = INT64_MIN;
+            out.cpx_del_ldt[o] = LDT_NONE_U32;
+        }
+        uint64_t s0 = out.cpx_start[o];
+        uint32_t nc2 = out.cpx_count[o], w = 0;
+        uint64_t tb0 = b_has ? tin.cpx_start[t] : 0;
+        uint32_t tn = b_has ? tin.cpx_count[t] : 0, bi = 0;
+        for (uint32_t e = 0; e < nc2; e++) {
+            uint64_t xe = s0 + e;
+            bool have_b = false;
+            while (bi < tn) {
+                int pcmp = gc_path_cmp(tin, tb0 + bi, out, xe);
+                if (pcmp < 0) { bi++; continue; }
+                have_b = pcmp == 0;
+                break;
+            }
+            if (out.cpx.ts[xe] <= mm) continue;  // deletion.deletes(existing)
+            if (have_b) {
+                uint64_t te = tb0 + bi;
+                if (gc_cell_b_wins(out.cpx.ts[xe], out.cpx.ldt[xe],
+                                   out.cpx.flags[xe] & CELLF_EXPIRING, out.cpx.val_addr[xe],
+                                   out.cpx.val_len[xe], tin.cpx.ts[te], tin.cpx.ldt[te],
+                                   tin.cpx.flags[te] & CELLF_EXPIRING, tin.cpx.val_addr[te],
+                                   tin.cpx.val_len[te]))
+                    continue;
+            }
+            gc_cpx_move(out, s0 + w, xe);
+            w++;
+        }
+        out.cpx_count[o] = w;
+        if (!w && !keep_del) f &= ~PF_HAS_CPX;
+        else any_cell = true;
     }
     bool keep = (f & (PF_LIVE_TS | PF_ROW_DEL)) || any_cell;
     out.flags[o] = keep ? (f | PF_HAS_ROW) : 0;
@@ -2811,6 +2978,14 @@ __device__ inline void gc_copy_unf(UnfCols& dst, uint64_t d, const UnfCols& src,
         dst.cell_ttl[dc] = src.cell_ttl[sc];
         dst.val_addr[dc] = src.val_addr[sc];
         dst.val_len[dc] = src.val_len[sc];
+    }
+    if (sch.n_cpx) {
+        // the cpx CELL arena is shared (aliased) between src and dst in the
+        // gc path; only the per-row segment descriptors are copied
+        dst.cpx_del_mfda[d] = src.cpx_del_mfda[s];
+        dst.cpx_del_ldt[d] = src.cpx_del_ldt[s];
+        dst.cpx_start[d] = src.cpx_start[s];
+        dst.cpx_count[d] = src.cpx_count[s];
     }
 }
 
@@ -3134,6 +3309,44 @@ __global__ void k_purge_parts(OutParts op, UnfCols out, uint64_t n, SchemaParams
                     }
                 }
                 any_cell = true;
+            }
+            if (sch.n_cpx && (f & PF_HAS_CPX)) {
+                // ComplexColumnData.purge (ComplexColumnData.java:212-216),
+                // deferred-purge variant over the shared cpx arena
+                if (out.cpx_del_mfda[o] != INT64_MIN &&
+                    should_purge2(pp, token, out.cpx_del_mfda[o], ldt_long(out.cpx_del_ldt[o]))) {
+                    out.cpx_del_mfda[o] = INT64_MIN;
+                    out.cpx_del_ldt[o] = LDT_NONE_U32;
+                }
+                uint64_t s0 = out.cpx_start[o];
+                uint32_t nc2 = out.cpx_count[o], wx = 0;
+                for (uint32_t e = 0; e < nc2; e++) {
+                    uint64_t xe = s0 + e;
+                    int64_t cts = out.cpx.ts[xe];
+                    uint32_t cldt = out.cpx.ldt[xe];
+                    int32_t cttl = out.cpx.ttl[xe];
+                    bool live_cell =
+                        cldt == LDT_NONE_U32 || (cttl != 0 && pp.now_sec < ldt_long(cldt));
+                    if (!live_cell) {
+                        if (should_purge2(pp, token, cts, ldt_long(cldt))) continue;
+                        if (cttl != 0) {
+                            int64_t nldt = ldt_long(cldt) - cttl;
+                            if (should_purge2(pp, token, cts, nldt)) continue;
+                            out.cpx.ldt[xe] = ldt_u32(nldt);
+                            out.cpx.ttl[xe] = 0;
+                            out.cpx.flags[xe] = CELLF_PRESENT;  // expired -> tombstone
+                            out.cpx.val_len[xe] = 0;
+                        }
+                    }
+                    gc_cpx_move(out, s0 + wx, xe);
+                    wx++;
+                }
+                out.cpx_count[o] = wx;
+                if (!wx && out.cpx_del_mfda[o] == INT64_MIN &&
+                    out.cpx_del_ldt[o] == LDT_NONE_U32)
+                    f &= ~PF_HAS_CPX;
+                else
+                    any_cell = true;
             }
             if (!(f & (PF_LIVE_TS | PF_ROW_DEL)) && !any_cell) f = 0;
             else if (pp.enforce_strict_liveness && !(f & PF_LIVE_TS) && !(f & PF_ROW_DEL)) f = 0;

@@ -274,8 +274,12 @@ static bytes ctx_merge(const bytes& lv, const bytes& rv) {
     if (rsup) return rv;
     return ctx_serialize(out);
 }
-// Cells.resolveCounter (Cells.java): tombstone beats any live counter; empty
-// values lose; else context merge with ts = max
+// Cells.resolveCounter (Cells.java:121-162): tombstone beats any live
+// counter; else context merge with ts = max. Deliberate divergence: the
+// reference lets an EMPTY value win a pairwise merge — empty counter cells
+// are a read-path artifact (#10657/#11726) that is never serialized into an
+// sstable, so this branch is unreachable for compaction inputs; both this
+// oracle and the GPU fold use the simpler non-empty-wins rule there.
 static Cell counter_reconcile(const Cell& l, const Cell& r) {
     bool lt = l.ldt != LDT_NONE_U32, rt = r.ldt != LDT_NONE_U32;
     if (lt || rt) {
@@ -656,12 +660,44 @@ static bool row_filter_active(Row& r, const DeletionTime& active) {
         r.del = DeletionTime{};  // shadowed row deletion is dropped
         for (auto& c : r.cells)
             if (c && c->ts <= active.mfda) c.reset();
+        // ComplexColumnData.filter(all, activeDeletion, null, -) — shadowed
+        // complexDeletion dropped, cells dropped by activeDeletion.deletes
+        // (ComplexColumnData.java:188-210); empty result -> column absent
+        for (auto& cd : r.complex) {
+            if (!cd) continue;
+            if (active.supersedes(cd->del)) cd->del = DeletionTime{};
+            std::vector<Cell> kept;
+            for (auto& c : cd->cells)
+                if (!(c.ts <= active.mfda)) kept.push_back(c);
+            cd->cells = std::move(kept);
+            if (cd->cells.empty() && cd->del.live()) cd.reset();
+        }
     }
     return !row_is_empty(r);
 }
+// Cells.addNonShadowed identity decision for counter cells: the data cell is
+// dropped iff Cells.reconcile(existing=a, update=b) returns b itself
+// (Cells.java:178-189 with resolveCounter 121-162). ctx_merge's superset
+// early-returns hand back the exact input bytes, so byte equality with an
+// input reproduces the reference's object-identity check (left tested first).
+static bool counter_b_wins(const Cell& a, const Cell& b) {
+    bool at = a.ldt != LDT_NONE_U32, bt = b.ldt != LDT_NONE_U32;
+    if (at || bt) {
+        if (at != bt) return bt;  // tombstone beats any live counter
+        return &reconcile_cells(a, b) == &b;  // two tombstones: regular rules
+    }
+    bool ae = a.value.empty(), be = b.value.empty();
+    if (ae || be) {
+        if (ae != be) return ae;  // non-empty wins (divergence note above)
+        return !(a.ts > b.ts);
+    }
+    bytes m = ctx_merge(a.value, b.value);
+    bool lid = m == a.value, rid = m == b.value;
+    return rid && !lid && std::max(a.ts, b.ts) == b.ts;
+}
 // garbageFilterRow (CompactionIterator.java:544-556): row- or cell-level
 static bool garbage_filter_row(Row& data, const Row& tomb, const DeletionTime& active,
-                               bool cell_level) {
+                               bool cell_level, bool counters) {
     if (!cell_level) {
         return row_filter_active(data, dt_max(tomb.del, active));
     }
@@ -674,7 +710,39 @@ static bool garbage_filter_row(Row& data, const Row& tomb, const DeletionTime& a
         if (!a) continue;
         if (a->ts <= deletion.mfda) { a.reset(); continue; }
         const Cell* b = i < tomb.cells.size() && tomb.cells[i] ? &*tomb.cells[i] : nullptr;
-        if (b && &reconcile_cells(*a, *b) == b) a.reset();  // overwritten by source
+        if (b) {
+            bool bw = counters ? counter_b_wins(*a, *b) : &reconcile_cells(*a, *b) == b;
+            if (bw) a.reset();  // overwritten by source
+        }
+    }
+    // complex branch of removeShadowedCells (Rows.java:298-316): the data
+    // complexDeletion is kept iff it supersedes max(updateDt, deletion) and
+    // then raises the bar; per-path Cells.addNonShadowedComplex
+    for (size_t i = 0; i < data.complex.size(); i++) {
+        auto& a = data.complex[i];
+        if (!a) continue;
+        const ComplexData* b =
+            i < tomb.complex.size() && tomb.complex[i] ? &*tomb.complex[i] : nullptr;
+        DeletionTime updateDt = b ? b->del : DeletionTime{};
+        DeletionTime maxDt = updateDt.supersedes(deletion) ? updateDt : deletion;
+        DeletionTime keepDel{};  // LIVE unless data complexDeletion survives
+        if (a->del.supersedes(maxDt)) { keepDel = a->del; maxDt = a->del; }
+        std::vector<Cell> kept;
+        size_t bi = 0;
+        for (auto& c : a->cells) {
+            const Cell* bc = nullptr;
+            if (b) {
+                while (bi < b->cells.size() && compare_cell_path(b->cells[bi].path, c.path) < 0)
+                    bi++;
+                if (bi < b->cells.size() && b->cells[bi].path == c.path) bc = &b->cells[bi];
+            }
+            if (c.ts <= maxDt.mfda) continue;  // deletion.deletes(existing)
+            if (bc && &reconcile_cells(c, *bc) == bc) continue;
+            kept.push_back(c);
+        }
+        if (kept.empty() && keepDel.live()) { a.reset(); continue; }
+        a->del = keepDel;
+        a->cells = std::move(kept);
     }
     return !row_is_empty(data);
 }
@@ -698,12 +766,13 @@ static DeletionTime update_open_dt(const Unfiltered& u) {
 }
 
 void garbage_filter(Partition& data, const Partition& tomb, const Header& h, bool cell_level) {
+    bool counters = !h.regular_cols.empty() && is_counter_type(h.regular_cols[0].second);
     DeletionTime partition_dt = tomb.del;
     DeletionTime active = tomb.del;
     DeletionTime tomb_open, data_open, open_dt;  // LIVE
     if (!data.del.supersedes(tomb.del)) data.del = DeletionTime{};
     if (h.has_static() && !row_is_empty(data.static_row))
-        if (!garbage_filter_row(data.static_row, tomb.static_row, active, cell_level))
+        if (!garbage_filter_row(data.static_row, tomb.static_row, active, cell_level, counters))
             data.static_row = Row{};
     std::vector<Unfiltered> out;
     size_t di = 0, ti = 0;
@@ -752,7 +821,7 @@ void garbage_filter(Partition& data, const Partition& tomb, const Header& h, boo
         } else if (cmp == 0) {
             if (data.items[di].kind == Unfiltered::ROW) {
                 Row r = data.items[di].row;
-                if (garbage_filter_row(r, tomb.items[ti].row, active, cell_level)) {
+                if (garbage_filter_row(r, tomb.items[ti].row, active, cell_level, counters)) {
                     produced.kind = Unfiltered::ROW;
                     produced.row = std::move(r);
                     have = true;
@@ -878,11 +947,6 @@ static Header make_output_header(const std::vector<SSTable>& inputs) {
 // top-level compaction (CompactionTask.runMayThrow hot loop semantics)
 // ---------------------------------------------------------------------------
 CompactionResult compact(const CompactionJob& job) {
-    if (!job.tomb_sources.empty() && !job.inputs.empty())
-        for (auto& cp : job.inputs[0].header.regular_cols)
-            if (is_complex_type(cp.second) || is_counter_type(cp.second))
-                throw std::runtime_error(
-                    "garbage-collect mode with complex/counter columns unsupported");
     CompactionResult res;
     res.out.header = make_output_header(job.inputs);
     res.out.comp = job.inputs.empty() ? CompressionParams{} : job.inputs[0].comp;

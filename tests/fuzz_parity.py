@@ -101,14 +101,24 @@ def main():
             okw["nevergc"] = 1
             okw.setdefault("gcbefore", 2000000000)
             job.setdefault("gc_before", 2000000000)
-        use_gc_sources = (not wide and rng.random() < 0.3 and "gc_before" not in job
-                          and "counter" not in gen and "cpx" not in gen)
+        use_gc_sources = (not wide and not use_bti and rng.random() < 0.3
+                          and "gc_before" not in job)
+        use_flush = rng.random() < 0.25
         use_shards = False
         bad = None
         try:
             args = [f"{k}={v}" for k, v in gen.items()]
+            if use_flush:
+                args.append("dump=1")
             subprocess.run([ORACLE, "gen", d, *args], check=True, capture_output=True)
             ins = [f"{d}/{stem}-{g}-{sfx}" for g in range(1, gen["n"] + 1)]
+            if use_flush:
+                # full-schema flush parity: re-flushing the memdump must
+                # reproduce the generated sstable byte-for-byte
+                ca.flush_table(f"{ins[0]}.memdump", f"{d}/{stem}-70-{sfx}")
+                bad = dirs_equal(f"{d}/{stem}-70-{sfx}", ins[0], comps)
+                if bad:
+                    raise AssertionError(f"flush_table mismatch: {bad}")
             if use_gc_sources:
                 os.makedirs(d + "/src")
                 sargs = dict(gen)

@@ -742,6 +742,65 @@ def test_cross_feature_combinations(ca, oracle_bin, tmp_path):
     _assert_dirs_equal(f"{dd}/oa-60-big", f"{dd}/oa-80-big")
 
 
+def test_gc_complex_counter(ca, oracle_bin, tmp_path):
+    """nodetool garbagecollect over complex (collection) and counter tables
+    (CompactionIterator.GarbageSkipper with Rows.removeShadowedCells'
+    complex branch, Rows.java:298-316, and Cells.addNonShadowed's
+    resolveCounter identity rule, Cells.java:121-189) — both TombstoneOption
+    ROW and CELL modes, byte-identical to the oracle."""
+    d = str(tmp_path)
+    # complex columns
+    _oracle_gen(d, seed=41, n=2, rows=700, vlen=60, overlap=30, tomb=20,
+                cpx=45, cpxdel=25)
+    os.makedirs(d + "/src")
+    _oracle_gen(d + "/src", seed=41, n=2, rows=700, vlen=60, overlap=30,
+                tomb=50, cpx=45, cpxdel=25, ts0=1700000500000000)
+    ins = [f"{d}/oa-{g}-big" for g in (1, 2)]
+    srcs = [f"{d}/src/oa-{g}-big" for g in (1, 2)]
+    for cell in (False, True):
+        out = f"{d}/oa-{60 + cell}-big"
+        ref = f"{d}/oa-{80 + cell}-big"
+        ca.compact(ins, out, tombstone_sources=srcs, cell_level_gc=cell)
+        kw = dict(tombsrc=",".join(srcs))
+        if cell:
+            kw["cellgc"] = 1
+        _oracle_compact(ref, ins, **kw)
+        _assert_dirs_equal(out, ref)
+    # counter columns
+    dc = d + "/ctr"
+    os.makedirs(dc)
+    _oracle_gen(dc, seed=42, n=2, rows=700, overlap=30, tomb=15, counter=1)
+    os.makedirs(dc + "/src")
+    _oracle_gen(dc + "/src", seed=42, n=1, rows=700, tomb=60, counter=1,
+                ts0=1700000500000000)
+    cins = [f"{dc}/oa-{g}-big" for g in (1, 2)]
+    csrc = [f"{dc}/src/oa-1-big"]
+    for cell in (False, True):
+        out = f"{dc}/oa-{60 + cell}-big"
+        ref = f"{dc}/oa-{80 + cell}-big"
+        ca.compact(cins, out, tombstone_sources=csrc, cell_level_gc=cell)
+        kw = dict(tombsrc=",".join(csrc))
+        if cell:
+            kw["cellgc"] = 1
+        _oracle_compact(ref, cins, **kw)
+        _assert_dirs_equal(out, ref)
+    # complex + TTL through gc, with a purge pass after the filter
+    dt = d + "/ttl"
+    os.makedirs(dt)
+    _oracle_gen(dt, seed=43, n=2, rows=500, vlen=50, overlap=25, tomb=15,
+                cpx=40, cpxdel=25, ttl=30)
+    os.makedirs(dt + "/src")
+    _oracle_gen(dt + "/src", seed=43, n=1, rows=500, tomb=55, cpx=40,
+                cpxdel=30, ts0=1700000500000000)
+    tins = [f"{dt}/oa-{g}-big" for g in (1, 2)]
+    tsrc = [f"{dt}/src/oa-1-big"]
+    ca.compact(tins, f"{dt}/oa-60-big", tombstone_sources=tsrc, cell_level_gc=True,
+               now_sec=1800000000, gc_before=1700001000)
+    _oracle_compact(f"{dt}/oa-80-big", tins, tombsrc=tsrc[0], cellgc=1,
+                    now=1800000000, gcbefore=1700001000)
+    _assert_dirs_equal(f"{dt}/oa-60-big", f"{dt}/oa-80-big")
+
+
 def test_flush_table_full_schema(ca, oracle_bin, tmp_path):
     """gpuc_flush_table — the REAL memtable flush path (VERDICT round-1 weak
     item: the v1 flush schema was `pk blob, val blob` only): partitions with
