@@ -1894,8 +1894,6 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         // data; the rest are tombstone sources merged in a second pass.
         int kd = su.k_data > 0 ? su.k_data : k;
         bool gc_mode = kd < k;
-        if (su.bti && gc_mode)
-            throw std::runtime_error("garbage-collect mode with bti (da) inputs unsupported");
         uint64_t data_parts = 0;
         for (int s = 0; s < kd; s++) data_parts += srcs[s].n_parts;
         std::vector<uint64_t> runs;
@@ -2572,8 +2570,6 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
 
         int S = S_pre;
         if (S > 1024) throw std::runtime_error("n_output_shards must be <= 1024");
-        if (su.bti && (S > 1 || job->has_token_range))
-            throw std::runtime_error("token sharding of bti (da) inputs unsupported this round");
         std::mutex res_mu;
         if (S == 1) {
             std::vector<std::pair<uint32_t, uint32_t>> pr(su.k);

@@ -101,8 +101,7 @@ def main():
             okw["nevergc"] = 1
             okw.setdefault("gcbefore", 2000000000)
             job.setdefault("gc_before", 2000000000)
-        use_gc_sources = (not wide and not use_bti and rng.random() < 0.3
-                          and "gc_before" not in job)
+        use_gc_sources = not wide and rng.random() < 0.3 and "gc_before" not in job
         use_flush = rng.random() < 0.25
         use_shards = False
         bad = None
@@ -136,7 +135,7 @@ def main():
                     okw["cellgc"] = 1
                 job["tombstone_sources"] = srcs
                 job["cell_level_gc"] = cell
-            use_shards = (not use_gc_sources) and (not use_bti) and rng.random() < 0.2
+            use_shards = (not use_gc_sources) and rng.random() < 0.2
             oargs = [f"{k}={v}" for k, v in okw.items()]
             if use_shards:
                 from cassandra_amd.sharding import split_token_range

@@ -742,6 +742,55 @@ def test_cross_feature_combinations(ca, oracle_bin, tmp_path):
     _assert_dirs_equal(f"{dd}/oa-60-big", f"{dd}/oa-80-big")
 
 
+def test_bti_sharding_and_gc(ca, oracle_bin, tmp_path):
+    """da (BTI) inputs through the two paths that were previously rejected:
+    UCS-style sharded outputs (each shard gets its own Partitions.db/Rows.db
+    post-pass) and garbagecollect with da tombstone sources — byte-identical
+    to the oracle per shard / per mode."""
+    BTI_COMPONENTS = ["Data.db", "CompressionInfo.db", "Filter.db", "Digest.crc32",
+                      "Statistics.db", "Partitions.db", "Rows.db", "TOC.txt"]
+    from cassandra_amd.sharding import split_token_range
+    d = str(tmp_path)
+    _oracle_gen(d, seed=71, n=3, rows=1200, vlen=120, overlap=25, tomb=15, bti=1)
+    ins = [f"{d}/da-{g}-bti" for g in (1, 2, 3)]
+    S = 2
+    ca.compact(ins, f"{d}/da-61-bti", n_output_shards=S)
+    for i in range(S):
+        lo, hi = split_token_range(S, i)
+        _oracle_compact(f"{d}/da-{80 + 10 * i}-bti", ins, shard=f"{lo}:{hi}")
+        _assert_dirs_equal(f"{d}/da-{61 + i}-bti", f"{d}/da-{80 + 10 * i}-bti",
+                           BTI_COMPONENTS)
+    # wide da + shards (row-index tries in every shard)
+    dw = d + "/w"
+    os.makedirs(dw)
+    _oracle_gen(dw, seed=72, n=2, rows=40, crows=60, vlen=150, rtomb=30, tomb=10, bti=1)
+    wins = [f"{dw}/da-{g}-bti" for g in (1, 2)]
+    ca.compact(wins, f"{dw}/da-61-bti", n_output_shards=S)
+    for i in range(S):
+        lo, hi = split_token_range(S, i)
+        _oracle_compact(f"{dw}/da-{80 + 10 * i}-bti", wins, shard=f"{lo}:{hi}")
+        _assert_dirs_equal(f"{dw}/da-{61 + i}-bti", f"{dw}/da-{80 + 10 * i}-bti",
+                           BTI_COMPONENTS)
+    # gc over da inputs with da tombstone sources, both modes
+    dg = d + "/g"
+    os.makedirs(dg)
+    _oracle_gen(dg, seed=73, n=2, rows=800, vlen=80, overlap=30, tomb=20, bti=1)
+    os.makedirs(dg + "/src")
+    _oracle_gen(dg + "/src", seed=73, n=2, rows=800, vlen=80, overlap=30, tomb=50,
+                ts0=1700000500000000, bti=1)
+    gins = [f"{dg}/da-{g}-bti" for g in (1, 2)]
+    gsrc = [f"{dg}/src/da-{g}-bti" for g in (1, 2)]
+    for cell in (False, True):
+        out = f"{dg}/da-{61 + cell}-bti"
+        ref = f"{dg}/da-{81 + cell}-bti"
+        ca.compact(gins, out, tombstone_sources=gsrc, cell_level_gc=cell)
+        kw = dict(tombsrc=",".join(gsrc))
+        if cell:
+            kw["cellgc"] = 1
+        _oracle_compact(ref, gins, **kw)
+        _assert_dirs_equal(out, ref, BTI_COMPONENTS)
+
+
 def test_gc_complex_counter(ca, oracle_bin, tmp_path):
     """nodetool garbagecollect over complex (collection) and counter tables
     (CompactionIterator.GarbageSkipper with Rows.removeShadowedCells'
