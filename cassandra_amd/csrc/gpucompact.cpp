@@ -1551,6 +1551,7 @@ struct CompactSetup {
     std::vector<uint64_t> generations;
     std::vector<std::vector<uint64_t>> positions;   // n_parts+1 absolute offsets
     std::vector<std::vector<uint64_t>> entry_offs;  // Index.db entry byte offsets
+    std::vector<std::vector<bti::bytes>> bti_prefixes;  // da: trie separator per partition
     std::vector<size_t> comp_file_sz;
     std::vector<int32_t> col_fixed_h;    // SIMPLE regular columns only
     uint32_t n_cpx = 0;                  // one complex (map<blob,blob>) column, last
@@ -1597,7 +1598,8 @@ static inline void check_cancel(const gpuc_job* job) {
 static void compact_one(const gpuc_job* job, const CompactSetup& su,
                         const std::vector<std::pair<uint32_t, uint32_t>>& pr,
                         const std::string& out_base_str, int wslot,
-                        gpuc_result* res, std::mutex& res_mu) {
+                        gpuc_result* res, std::mutex& res_mu,
+                        bool shard_exact = false, int64_t sh_lo = 0, int64_t sh_hi = 0) {
     auto wall = []() {
         struct timespec ts;
         clock_gettime(CLOCK_MONOTONIC, &ts);
@@ -1988,6 +1990,19 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
         pp.has_shard = job->has_token_range;
         pp.shard_lo = job->token_lo;
         pp.shard_hi = job->token_hi;
+        if (shard_exact) {
+            // da internal shards: trie-separator windows are one partition
+            // wide of exact; this token filter (on GPU-computed tokens) is
+            // the exact boundary, intersected with any job-level range
+            if (pp.has_shard) {
+                pp.shard_lo = std::max(pp.shard_lo, sh_lo);
+                pp.shard_hi = std::min(pp.shard_hi, sh_hi);
+            } else {
+                pp.has_shard = 1;
+                pp.shard_lo = sh_lo;
+                pp.shard_hi = sh_hi;
+            }
+        }
         DevBuf d_kr_lo, d_kr_hi;
         if (job->n_keep_ranges > 0) {
             klo_h.resize(job->n_keep_ranges);
@@ -2358,6 +2373,7 @@ static void compact_setup(const gpuc_job* job, CompactSetup& su, bool preread_fu
     su.generations.resize(k);
     su.positions.resize(k);
     su.entry_offs.resize(k);
+    su.bti_prefixes.resize(k);
     su.comp_file_sz.resize(k);
     // Data.db reads start FIRST (they are the long pole); metadata parse and
     // index decode run while they stream into the pinned arenas
@@ -2424,9 +2440,13 @@ static void compact_setup(const gpuc_job* job, CompactSetup& su, bool preread_fu
                         auto bp = bti::read_bti_partitions(pfb);
                         auto& pos = su.positions[s];
                         pos.reserve(bp.entries.size() + 1);
-                        for (auto& e : bp.entries)
+                        auto& pfx = su.bti_prefixes[s];
+                        pfx.reserve(bp.entries.size());
+                        for (auto& e : bp.entries) {
                             pos.push_back(e.idxpos < 0 ? (uint64_t)~e.idxpos
                                                        : bti::row_index_data_pos(rfb, (uint64_t)e.idxpos));
+                            pfx.push_back(std::move(e.prefix));
+                        }
                     } else {
                         su.index_data[s] = read_file(base + "-Index.db");
                     }
@@ -2553,10 +2573,6 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
             set_err(res->error, sizeof(res->error), "n_inputs (+ tombstone sources) must be 1..64");
             return GPUC_ERR_UNSUPPORTED;
         }
-        if (job->n_tomb_sources > 0 && job->n_output_shards > 1) {
-            set_err(res->error, sizeof(res->error), "garbage collect + sharded outputs unsupported");
-            return GPUC_ERR_UNSUPPORTED;
-        }
         double t0 = wall();
         int S_pre = job->n_output_shards > 1 ? job->n_output_shards : 1;
         CompactSetup su;
@@ -2596,19 +2612,59 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
             }
             // per input: shard boundary partition = first with token >= lo_tok[i]
             std::vector<std::vector<uint32_t>> bounds(su.k, std::vector<uint32_t>(S + 1));
+            std::vector<std::vector<uint32_t>> boundsR(su.k, std::vector<uint32_t>(S + 1));
             for (int s = 0; s < su.k; s++) {
                 uint32_t n = (uint32_t)(su.positions[s].size() - 1);
-                bounds[s][0] = 0;
-                bounds[s][S] = n;
+                bounds[s][0] = boundsR[s][0] = 0;
+                bounds[s][S] = boundsR[s][S] = n;
                 for (int i = 1; i < S; i++) {
-                    uint32_t lo = 0, hi = n;  // lower_bound(token >= lo_tok[i])
-                    while (lo < hi) {
-                        uint32_t mid = (lo + hi) >> 1;
-                        if (index_entry_token(su.index_data[s], su.entry_offs[s][mid]) < lo_tok[i])
-                            lo = mid + 1;
-                        else hi = mid;
+                    if (!su.bti) {
+                        uint32_t lo = 0, hi = n;  // lower_bound(token >= lo_tok[i])
+                        while (lo < hi) {
+                            uint32_t mid = (lo + hi) >> 1;
+                            if (index_entry_token(su.index_data[s], su.entry_offs[s][mid]) < lo_tok[i])
+                                lo = mid + 1;
+                            else hi = mid;
+                        }
+                        bounds[s][i] = lo;
+                    } else {
+                        // da inputs: Partitions.db stores CUT byte-comparable
+                        // separators S_j with key_{j-1} < S_j <= key_j, whose
+                        // leading bytes are the sign-flipped BE token
+                        // (bti_byte_comparable_m3). A cut separator can be a
+                        // strict PREFIX of the boundary string B, which makes
+                        // its side ambiguous — so each boundary gets a
+                        // left-biased lower bound (prefix-of-B counts as > B;
+                        // first hit is provably <= j0+1, where j0 = first
+                        // partition with token >= the boundary, because
+                        // S_{j0+1} > key_{j0} >= B is never a prefix of B)
+                        // and a right-biased upper bound (prefix counts as
+                        // <= B; every separator before j0 is < B, so the
+                        // first hit is >= j0 and <= j0+1). Window i =
+                        // [L[i]-1, R[i+1]) then OVERLAPS its neighbours by at
+                        // most two partitions; compact_one's per-shard token
+                        // filter (exact, on GPU-computed tokens) assigns each
+                        // partition to exactly one shard.
+                        uint8_t B[8];
+                        uint64_t tb = (uint64_t)lo_tok[i] ^ (1ull << 63);
+                        for (int b3 = 0; b3 < 8; b3++) B[b3] = (uint8_t)(tb >> (8 * (7 - b3)));
+                        auto search = [&](bool prefix_is_gt) {
+                            uint32_t lo = 0, hi = n;
+                            while (lo < hi) {
+                                uint32_t mid = (lo + hi) >> 1;
+                                const bti::bytes& p2 = su.bti_prefixes[s][mid];
+                                size_t m2 = p2.size() < 8 ? p2.size() : 8;
+                                int c2 = m2 ? memcmp(p2.data(), B, m2) : 0;
+                                bool gt = c2 ? c2 > 0 : (p2.size() >= 8 ? false : prefix_is_gt);
+                                if (gt) hi = mid;
+                                else lo = mid + 1;
+                            }
+                            return lo;
+                        };
+                        uint32_t L = search(true);
+                        boundsR[s][i] = search(false);
+                        bounds[s][i] = L > 0 ? L - 1 : 0;
                     }
-                    bounds[s][i] = lo;
                 }
             }
             // two workers: front-phase kernels of one shard overlap the
@@ -2620,10 +2676,16 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
                     int i = next.fetch_add(1);
                     if (i >= S) break;
                     std::vector<std::pair<uint32_t, uint32_t>> pr(su.k);
-                    for (int s = 0; s < su.k; s++) pr[s] = {bounds[s][i], bounds[s][i + 1]};
+                    for (int s = 0; s < su.k; s++) {
+                        // bti: window = [left-biased start, right-biased end)
+                        uint32_t e2 = su.bti ? boundsR[s][i + 1] : bounds[s][i + 1];
+                        pr[s] = {std::min(bounds[s][i], e2), e2};
+                    }
                     std::string out_i = dir + "/" + pre + std::to_string(gen0 + i) + post;
+                    int64_t sh_lo = lo_tok[i];
+                    int64_t sh_hi = i + 1 < S ? lo_tok[i + 1] - 1 : INT64_MAX;
                     try {
-                        compact_one(job, su, pr, out_i, w, res, res_mu);
+                        compact_one(job, su, pr, out_i, w, res, res_mu, su.bti, sh_lo, sh_hi);
                     } catch (const std::exception& e) {
                         werr[w] = e.what();
                         break;

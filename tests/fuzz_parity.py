@@ -135,7 +135,7 @@ def main():
                     okw["cellgc"] = 1
                 job["tombstone_sources"] = srcs
                 job["cell_level_gc"] = cell
-            use_shards = (not use_gc_sources) and rng.random() < 0.2
+            use_shards = rng.random() < 0.2
             oargs = [f"{k}={v}" for k, v in okw.items()]
             if use_shards:
                 from cassandra_amd.sharding import split_token_range

@@ -791,6 +791,33 @@ def test_bti_sharding_and_gc(ca, oracle_bin, tmp_path):
         _assert_dirs_equal(out, ref, BTI_COMPONENTS)
 
 
+def test_gc_with_sharded_outputs(ca, oracle_bin, tmp_path):
+    """garbagecollect + UCS sharded outputs in one call: each shard applies
+    the GarbageSkipper before its writer (sources are NOT shard-filtered —
+    they shadow regardless of the shard), byte-identical to the oracle's
+    shard= + tombsrc= runs."""
+    from cassandra_amd.sharding import split_token_range
+    d = str(tmp_path)
+    _oracle_gen(d, seed=75, n=2, rows=900, vlen=70, overlap=30, tomb=20)
+    os.makedirs(d + "/src")
+    _oracle_gen(d + "/src", seed=75, n=2, rows=900, vlen=70, overlap=30, tomb=50,
+                ts0=1700000500000000)
+    ins = [f"{d}/oa-{g}-big" for g in (1, 2)]
+    srcs = [f"{d}/src/oa-{g}-big" for g in (1, 2)]
+    S = 2
+    for cell in (False, True):
+        ca.compact(ins, f"{d}/oa-{61 + 10 * cell}-big", n_output_shards=S,
+                   tombstone_sources=srcs, cell_level_gc=cell)
+        for i in range(S):
+            lo, hi = split_token_range(S, i)
+            kw = dict(tombsrc=",".join(srcs), shard=f"{lo}:{hi}")
+            if cell:
+                kw["cellgc"] = 1
+            ref = f"{d}/oa-{81 + 10 * cell + i * 2}-big"
+            _oracle_compact(ref, ins, **kw)
+            _assert_dirs_equal(f"{d}/oa-{61 + 10 * cell + i}-big", ref)
+
+
 def test_gc_complex_counter(ca, oracle_bin, tmp_path):
     """nodetool garbagecollect over complex (collection) and counter tables
     (CompactionIterator.GarbageSkipper with Rows.removeShadowedCells'
