@@ -2645,17 +2645,20 @@ extern "C" int gpuc_compact(const gpuc_job* job, gpuc_result* res) {
                         // most two partitions; compact_one's per-shard token
                         // filter (exact, on GPU-computed tokens) assigns each
                         // partition to exactly one shard.
-                        uint8_t B[8];
+                        // boundary string: NEXT_COMPONENT then the
+                        // sign-flipped BE token (bti_byte_comparable_m3)
+                        uint8_t B[9];
+                        B[0] = 0x40;
                         uint64_t tb = (uint64_t)lo_tok[i] ^ (1ull << 63);
-                        for (int b3 = 0; b3 < 8; b3++) B[b3] = (uint8_t)(tb >> (8 * (7 - b3)));
+                        for (int b3 = 0; b3 < 8; b3++) B[1 + b3] = (uint8_t)(tb >> (8 * (7 - b3)));
                         auto search = [&](bool prefix_is_gt) {
                             uint32_t lo = 0, hi = n;
                             while (lo < hi) {
                                 uint32_t mid = (lo + hi) >> 1;
                                 const bti::bytes& p2 = su.bti_prefixes[s][mid];
-                                size_t m2 = p2.size() < 8 ? p2.size() : 8;
+                                size_t m2 = p2.size() < 9 ? p2.size() : 9;
                                 int c2 = m2 ? memcmp(p2.data(), B, m2) : 0;
-                                bool gt = c2 ? c2 > 0 : (p2.size() >= 8 ? false : prefix_is_gt);
+                                bool gt = c2 ? c2 > 0 : (p2.size() >= 9 ? false : prefix_is_gt);
                                 if (gt) hi = mid;
                                 else lo = mid + 1;
                             }
