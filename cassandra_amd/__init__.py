@@ -161,10 +161,11 @@ def load_library(path=None):
     lib.gpuc_flush.argtypes = [ctypes.POINTER(GpucFlushRows), ctypes.c_char_p, ctypes.c_int32,
                                ctypes.c_char_p, ctypes.c_size_t]
     lib.gpuc_flush.restype = ctypes.c_int
-    lib.gpuc_flush_table.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_uint64,
-                                     ctypes.c_char_p, ctypes.c_int32,
-                                     ctypes.c_char_p, ctypes.c_size_t]
-    lib.gpuc_flush_table.restype = ctypes.c_int
+    if hasattr(lib, "gpuc_flush_table"):  # absent only in frozen A/B probe builds
+        lib.gpuc_flush_table.argtypes = [ctypes.c_void_p, ctypes.c_void_p, ctypes.c_uint64,
+                                         ctypes.c_char_p, ctypes.c_int32,
+                                         ctypes.c_char_p, ctypes.c_size_t]
+        lib.gpuc_flush_table.restype = ctypes.c_int
     lib.gpuc_version.restype = ctypes.c_char_p
     lib.gpuc_device_count.restype = ctypes.c_int
     _lib = lib

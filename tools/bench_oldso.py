@@ -5,4 +5,5 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import cassandra_amd as ca
 ca.load_library(sys.argv[1])
 sys.argv = ["bench.py"] + sys.argv[2:]
-exec(open(os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))), "bench.py")).read())
+_bench = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))), "bench.py")
+exec(compile(open(_bench).read(), _bench, "exec"), {"__name__": "__main__", "__file__": _bench})
