@@ -1095,24 +1095,32 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
         seg_end_byte[NSEG - 1] = total_unc;
     }
     std::vector<hipEvent_t> ev_seg(NSEG);
+    // NSEG == 1 keeps serialize on the MAIN stream with no cross-stream
+    // events: an A/B bisect (profiles/r02_drain_regression.md) measured the
+    // side-stream + per-slab hipStreamWaitEvent structure at -30% whole-step
+    // on this pool even with a single segment, so the segmented form only
+    // exists behind GPUC_NSEG > 1.
     {
         uint32_t waves_per_block = 4;
-        HIP_CHECK(hipStreamWaitEvent(sstream, ev1, 0));
+        hipStream_t ser_stream = NSEG > 1 ? sstream : stream;
+        if (NSEG > 1) HIP_CHECK(hipStreamWaitEvent(sstream, ev1, 0));
         for (int j = 0; j < NSEG; j++) {
             uint64_t gs = seg_g[j], ge = seg_g[j + 1];
             uint64_t blocks = (std::max<uint64_t>(ge - gs, 1) + waves_per_block - 1) / waves_per_block;
             if (ge > gs)
                 hipLaunchKernelGGL(k_serialize_rows, dim3((uint32_t)blocks),
-                                   dim3(WAVE * waves_per_block), 0, sstream, opb.op, rows.uc, ge,
+                                   dim3(WAVE * waves_per_block), 0, ser_stream, opb.op, rows.uc, ge,
                                    sp, d_psize.as<uint64_t>(), d_isize.as<uint64_t>(),
                                    d_nblocks.as<uint32_t>(), d_infsz.as<uint64_t>(),
                                    d_out_data.as<uint8_t>(), d_out_index.as<uint8_t>(),
                                    d_bloom.as<uint32_t>(), words * 64, bs.k, gs);
-            HIP_CHECK(hipEventCreate(&ev_seg[j]));
-            HIP_CHECK(hipEventRecord(ev_seg[j], sstream));
+            if (NSEG > 1) {
+                HIP_CHECK(hipEventCreate(&ev_seg[j]));
+                HIP_CHECK(hipEventRecord(ev_seg[j], sstream));
+            }
         }
+        HIP_CHECK(hipEventRecord(ev2, ser_stream));
     }
-    HIP_CHECK(hipEventRecord(ev2, sstream));
 
     // ---- compress + gather + D2H + write: slab-pipelined ----
     // Compress launches for all slabs are enqueued on `stream` back to back;
@@ -1141,10 +1149,12 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
     int next_seg = 0;
     for (uint32_t i = 0; i < n_slabs; i++) {
         uint32_t cb = i * SLAB, m = std::min(SLAB, n_chunks - cb);
-        // gate this slab on the serialize segment that completes its bytes
-        uint64_t slab_end = std::min<uint64_t>((uint64_t)(cb + m) * CHUNK_LEN, total_unc);
-        while (next_seg < NSEG && seg_end_byte[next_seg] < slab_end) next_seg++;
-        HIP_CHECK(hipStreamWaitEvent(stream, next_seg < NSEG ? ev_seg[next_seg] : ev2, 0));
+        if (NSEG > 1) {
+            // gate this slab on the serialize segment that completes its bytes
+            uint64_t slab_end = std::min<uint64_t>((uint64_t)(cb + m) * CHUNK_LEN, total_unc);
+            while (next_seg < NSEG && seg_end_byte[next_seg] < slab_end) next_seg++;
+            HIP_CHECK(hipStreamWaitEvent(stream, next_seg < NSEG ? ev_seg[next_seg] : ev2, 0));
+        }
         // all kernel arguments shift uniformly per chunk, so a slab launch is
         // just base-offset pointers with a local chunk count
         if (snappy_out)
@@ -1495,7 +1505,8 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
     }
     for (uint32_t i = 0; i < n_slabs; i++) (void)hipEventDestroy(ev_c[i]);
     HIP_CHECK(hipStreamSynchronize(sstream));
-    for (int j = 0; j < NSEG; j++) (void)hipEventDestroy(ev_seg[j]);
+    if (NSEG > 1)
+        for (int j = 0; j < NSEG; j++) (void)hipEventDestroy(ev_seg[j]);
     HIP_CHECK(hipStreamDestroy(sstream));
     HIP_CHECK(hipStreamDestroy(cstream));
     w.uncompressed_len = total_unc;
