@@ -1070,12 +1070,15 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
     // serialize in byte-balanced SEGMENTS on a side stream so the compress
     // slabs below start as soon as their byte range is fully written (the
     // output bytes are offset-addressed, so the split cannot change them)
-    hipStream_t sstream;
-    HIP_CHECK(hipStreamCreate(&sstream));
     int NSEG = 1;  // >1 overlaps serialize under compress; no win measured on
                    // write-wall-bound boxes, kept behind GPUC_NSEG for fast ones
     if (const char* e = getenv("GPUC_NSEG")) NSEG = std::max(1, std::min(64, atoi(e)));
     if (n_groups < 4096) NSEG = 1;
+    // the side stream exists ONLY in segmented mode: merely CREATING a third
+    // stream here shifts the runtime's stream->HW-queue mapping under the
+    // drain (cstream) and costs ~30% whole-step (r02_drain_regression.md)
+    hipStream_t sstream = nullptr;
+    if (NSEG > 1) HIP_CHECK(hipStreamCreate(&sstream));
     std::vector<uint64_t> seg_g(NSEG + 1), seg_end_byte(NSEG);
     if (NSEG > 1) {
         // d_psize holds the exscanned per-group data offsets
@@ -1504,10 +1507,11 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
         w.ms_d2h = t_drain1 - t_drain0;  // slab drain wall (overlaps compress)
     }
     for (uint32_t i = 0; i < n_slabs; i++) (void)hipEventDestroy(ev_c[i]);
-    HIP_CHECK(hipStreamSynchronize(sstream));
-    if (NSEG > 1)
+    if (NSEG > 1) {
+        HIP_CHECK(hipStreamSynchronize(sstream));
         for (int j = 0; j < NSEG; j++) (void)hipEventDestroy(ev_seg[j]);
-    HIP_CHECK(hipStreamDestroy(sstream));
+        HIP_CHECK(hipStreamDestroy(sstream));
+    }
     HIP_CHECK(hipStreamDestroy(cstream));
     w.uncompressed_len = total_unc;
     float t01, t12, t23;
