@@ -1048,7 +1048,6 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
                            d_infsz.as<uint64_t>(), d_isize.as<uint64_t>());
     }
     uint64_t total_idx = exscan_u64(d_isize.as<uint64_t>(), n_groups, stream);
-    HIP_CHECK(hipEventRecord(ev1, stream));
 
     OutStats hst;
     HIP_CHECK(hipStreamSynchronize(stream));
@@ -1067,62 +1066,53 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
     d_out_index.alloc(total_idx);
     d_bloom.alloc(words * 8);
     HIP_CHECK(hipMemsetAsync(d_bloom.p, 0, words * 8, stream));
-    // serialize in byte-balanced SEGMENTS on a side stream so the compress
-    // slabs below start as soon as their byte range is fully written (the
-    // output bytes are offset-addressed, so the split cannot change them)
-    int NSEG = 1;  // >1 overlaps serialize under compress; no win measured on
-                   // write-wall-bound boxes, kept behind GPUC_NSEG for fast ones
+    // ev1 AFTER the bloom memset: segmented serialize (cstream) gates on it
+    HIP_CHECK(hipEventRecord(ev1, stream));
+    // serialize in SEGMENTS INTERLEAVED with the compress slabs on the MAIN
+    // stream (in-order launches are the only synchronization): compress of a
+    // byte range starts once the segment covering it has been issued ahead
+    // of it, so the drain receives early slabs ~7/8 of a serialize earlier.
+    // NEVER a third stream, and nothing ahead of the drain ops on cstream:
+    // both alternatives measured -30% whole-step
+    // (profiles/r02_drain_regression.md).
+    hipStream_t cstream;
+    HIP_CHECK(hipStreamCreate(&cstream));
+    int NSEG = n_groups >= 4096 && total_unc > (256ull << 20) ? 8 : 1;
     if (const char* e = getenv("GPUC_NSEG")) NSEG = std::max(1, std::min(64, atoi(e)));
     if (n_groups < 4096) NSEG = 1;
-    // the side stream exists ONLY in segmented mode: merely CREATING a third
-    // stream here shifts the runtime's stream->HW-queue mapping under the
-    // drain (cstream) and costs ~30% whole-step (r02_drain_regression.md)
-    hipStream_t sstream = nullptr;
-    if (NSEG > 1) HIP_CHECK(hipStreamCreate(&sstream));
     std::vector<uint64_t> seg_g(NSEG + 1), seg_end_byte(NSEG);
     if (NSEG > 1) {
-        // d_psize holds the exscanned per-group data offsets
-        std::vector<uint64_t> h_off(n_groups);
-        HIP_CHECK(hipMemcpy(h_off.data(), d_psize.p, n_groups * 8, hipMemcpyDeviceToHost));
+        // equal-GROUP segments; only the NSEG-1 boundary byte offsets come
+        // back from d_psize (not the whole offset array)
         seg_g[0] = 0;
-        for (int j = 1; j < NSEG; j++) {
-            uint64_t target = total_unc / NSEG * j;
-            seg_g[j] = std::lower_bound(h_off.begin(), h_off.end(), target) - h_off.begin();
-        }
         seg_g[NSEG] = n_groups;
-        for (int j = 0; j < NSEG; j++)
-            seg_end_byte[j] = j + 1 < NSEG ? h_off[seg_g[j + 1]] : total_unc;
+        for (int j = 1; j < NSEG; j++) seg_g[j] = n_groups * (uint64_t)j / NSEG;
+        for (int j = 0; j < NSEG - 1; j++)
+            HIP_CHECK(hipMemcpy(&seg_end_byte[j], d_psize.as<uint64_t>() + seg_g[j + 1], 8,
+                                hipMemcpyDeviceToHost));
+        seg_end_byte[NSEG - 1] = total_unc;
     } else {
         seg_g[0] = 0;
         seg_g[NSEG] = n_groups;
         seg_end_byte[NSEG - 1] = total_unc;
     }
-    std::vector<hipEvent_t> ev_seg(NSEG);
-    // NSEG == 1 keeps serialize on the MAIN stream with no cross-stream
-    // events: an A/B bisect (profiles/r02_drain_regression.md) measured the
-    // side-stream + per-slab hipStreamWaitEvent structure at -30% whole-step
-    // on this pool even with a single segment, so the segmented form only
-    // exists behind GPUC_NSEG > 1.
-    {
+    auto launch_serialize_seg = [&](int j) {
         uint32_t waves_per_block = 4;
-        hipStream_t ser_stream = NSEG > 1 ? sstream : stream;
-        if (NSEG > 1) HIP_CHECK(hipStreamWaitEvent(sstream, ev1, 0));
-        for (int j = 0; j < NSEG; j++) {
-            uint64_t gs = seg_g[j], ge = seg_g[j + 1];
-            uint64_t blocks = (std::max<uint64_t>(ge - gs, 1) + waves_per_block - 1) / waves_per_block;
-            if (ge > gs)
-                hipLaunchKernelGGL(k_serialize_rows, dim3((uint32_t)blocks),
-                                   dim3(WAVE * waves_per_block), 0, ser_stream, opb.op, rows.uc, ge,
-                                   sp, d_psize.as<uint64_t>(), d_isize.as<uint64_t>(),
-                                   d_nblocks.as<uint32_t>(), d_infsz.as<uint64_t>(),
-                                   d_out_data.as<uint8_t>(), d_out_index.as<uint8_t>(),
-                                   d_bloom.as<uint32_t>(), words * 64, bs.k, gs);
-            if (NSEG > 1) {
-                HIP_CHECK(hipEventCreate(&ev_seg[j]));
-                HIP_CHECK(hipEventRecord(ev_seg[j], sstream));
-            }
-        }
-        HIP_CHECK(hipEventRecord(ev2, ser_stream));
+        uint64_t gs = seg_g[j], ge = seg_g[j + 1];
+        uint64_t blocks = (std::max<uint64_t>(ge - gs, 1) + waves_per_block - 1) / waves_per_block;
+        if (ge > gs)
+            hipLaunchKernelGGL(k_serialize_rows, dim3((uint32_t)blocks),
+                               dim3(WAVE * waves_per_block), 0, stream, opb.op, rows.uc, ge,
+                               sp, d_psize.as<uint64_t>(), d_isize.as<uint64_t>(),
+                               d_nblocks.as<uint32_t>(), d_infsz.as<uint64_t>(),
+                               d_out_data.as<uint8_t>(), d_out_index.as<uint8_t>(),
+                               d_bloom.as<uint32_t>(), words * 64, bs.k, gs);
+    };
+    int seg_issued = 0;
+    if (NSEG == 1) {
+        launch_serialize_seg(0);
+        seg_issued = 1;
+        HIP_CHECK(hipEventRecord(ev2, stream));
     }
 
     // ---- compress + gather + D2H + write: slab-pipelined ----
@@ -1139,24 +1129,28 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
     d_slots.alloc((uint64_t)n_chunks * SLOT_STRIDE + 16);
     d_csize.alloc((uint64_t)n_chunks * 4 + 16);
     d_ccrc.alloc((uint64_t)n_chunks * 4 + 16);
-    hipStream_t cstream;
-    HIP_CHECK(hipStreamCreate(&cstream));
-
-    // Index.db image is final right after serialize: drain it early on cstream
+    // Index.db image is final right after serialize: drain it early on
+    // cstream (NSEG == 1; segmented mode enqueues it after ev2 exists, below)
     uint8_t* h_index = (uint8_t*)g_pin_out[wslot][1].get(total_idx ? total_idx : 1);
     if (!h_index) throw std::runtime_error("pinned out alloc failed");
-    HIP_CHECK(hipStreamWaitEvent(cstream, ev2, 0));
-    HIP_CHECK(hipMemcpyAsync(h_index, d_out_index.p, total_idx, hipMemcpyDeviceToHost, cstream));
+    if (NSEG == 1) {
+        HIP_CHECK(hipStreamWaitEvent(cstream, ev2, 0));
+        HIP_CHECK(hipMemcpyAsync(h_index, d_out_index.p, total_idx, hipMemcpyDeviceToHost, cstream));
+    }
 
     std::vector<hipEvent_t> ev_c(n_slabs);
-    int next_seg = 0;
     for (uint32_t i = 0; i < n_slabs; i++) {
         uint32_t cb = i * SLAB, m = std::min(SLAB, n_chunks - cb);
         if (NSEG > 1) {
-            // gate this slab on the serialize segment that completes its bytes
+            // issue the serialize segments covering this slab's bytes AHEAD
+            // of it on the same in-order stream (ordering IS the dependency)
             uint64_t slab_end = std::min<uint64_t>((uint64_t)(cb + m) * CHUNK_LEN, total_unc);
-            while (next_seg < NSEG && seg_end_byte[next_seg] < slab_end) next_seg++;
-            HIP_CHECK(hipStreamWaitEvent(stream, next_seg < NSEG ? ev_seg[next_seg] : ev2, 0));
+            while (seg_issued < NSEG && (seg_issued ? seg_end_byte[seg_issued - 1] : 0) < slab_end)
+                launch_serialize_seg(seg_issued++);
+            if (seg_issued == NSEG) {
+                HIP_CHECK(hipEventRecord(ev2, stream));
+                seg_issued++;  // record ev2 once
+            }
         }
         // all kernel arguments shift uniformly per chunk, so a slab launch is
         // just base-offset pointers with a local chunk count
@@ -1177,7 +1171,11 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
         HIP_CHECK(hipEventCreate(&ev_c[i]));
         HIP_CHECK(hipEventRecord(ev_c[i], stream));
     }
-    HIP_CHECK(hipEventRecord(ev3, stream));  // ev2..ev3: pure compress GPU time
+    if (NSEG > 1 && seg_issued <= NSEG) {
+        while (seg_issued < NSEG) launch_serialize_seg(seg_issued++);
+        HIP_CHECK(hipEventRecord(ev2, stream));
+    }
+    HIP_CHECK(hipEventRecord(ev3, stream));  // ev2..ev3: compress (+interleaved serialize) GPU time
     TR("wsd: compress issued");
 
     std::string data_path = out_base + "-Data.db";
@@ -1278,6 +1276,14 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
     }
     w.compressed_len = file_off;
     TR("wsd: drain done");
+    if (NSEG > 1) {
+        // segmented mode: the index image drains AFTER the data drain — an
+        // earlier enqueue would gate the whole in-order cstream on ev2 (the
+        // LAST serialize segment) and collapse the drain to a tail
+        HIP_CHECK(hipStreamWaitEvent(cstream, ev2, 0));
+        HIP_CHECK(hipMemcpyAsync(h_index, d_out_index.p, total_idx, hipMemcpyDeviceToHost, cstream));
+        HIP_CHECK(hipStreamSynchronize(cstream));
+    }
     HIP_CHECK(hipEventRecord(ev4, stream));
 
     {
@@ -1507,11 +1513,7 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
         w.ms_d2h = t_drain1 - t_drain0;  // slab drain wall (overlaps compress)
     }
     for (uint32_t i = 0; i < n_slabs; i++) (void)hipEventDestroy(ev_c[i]);
-    if (NSEG > 1) {
-        HIP_CHECK(hipStreamSynchronize(sstream));
-        for (int j = 0; j < NSEG; j++) (void)hipEventDestroy(ev_seg[j]);
-        HIP_CHECK(hipStreamDestroy(sstream));
-    }
+
     HIP_CHECK(hipStreamDestroy(cstream));
     w.uncompressed_len = total_unc;
     float t01, t12, t23;
