@@ -791,6 +791,20 @@ def test_bti_sharding_and_gc(ca, oracle_bin, tmp_path):
         _assert_dirs_equal(out, ref, BTI_COMPONENTS)
 
 
+def test_segmented_serialize_parity(ca, oracle_bin, tmp_path):
+    """The serialize-interleaved-with-compress path (NSEG > 1) engages only
+    above 256 MiB of output — every other parity test is smaller and runs
+    the single-segment path. This case crosses the threshold (~320 MiB
+    uncompressed output) so segment boundaries, the in-order issue
+    interleave and the post-drain index copy are byte-compared too."""
+    d = str(tmp_path)
+    _oracle_gen(d, seed=61, n=2, rows=160000, vlen=1024, overlap=15, tomb=10)
+    ins = [f"{d}/oa-{g}-big" for g in (1, 2)]
+    ca.compact(ins, f"{d}/oa-60-big")
+    _oracle_compact(f"{d}/oa-80-big", ins)
+    _assert_dirs_equal(f"{d}/oa-60-big", f"{d}/oa-80-big")
+
+
 def test_gc_with_sharded_outputs(ca, oracle_bin, tmp_path):
     """garbagecollect + UCS sharded outputs in one call: each shard applies
     the GarbageSkipper before its writer (sources are NOT shard-filtered —
