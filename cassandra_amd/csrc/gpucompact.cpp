@@ -1297,6 +1297,100 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
         HIP_CHECK(hipStreamSynchronize(stream));
         HIP_CHECK(hipStreamSynchronize(cstream));
         TR("wsd: tail synced");
+        // digest fold and the Summary.db index walk are independent of the
+        // meta build: run all three concurrently (tail was serial: ~145 ms)
+        static Crc32Combiner comb;
+        std::future<uint32_t> fut_digest = std::async(std::launch::async, [&]() {
+            int nth = n_chunks > 4096 ? 16 : 1;
+            std::vector<uint32_t> pcrc(nth, 0);
+            std::vector<uint64_t> plen(nth, 0);
+            uint32_t per = (n_chunks + nth - 1) / nth;
+            std::vector<std::thread> th;
+            for (int t = 0; t < nth; t++) {
+                th.emplace_back([&, t]() {
+                    uint32_t c0 = t * per, c1 = std::min(n_chunks, (t + 1) * per);
+                    uint32_t d = 0;
+                    uint64_t l = 0;
+                    for (uint32_t c = c0; c < c1; c++) {
+                        d = comb.combine(d, h_crc[c], cs[c]);
+                        uint8_t cb2[4] = {(uint8_t)(h_crc[c] >> 24), (uint8_t)(h_crc[c] >> 16),
+                                          (uint8_t)(h_crc[c] >> 8), (uint8_t)h_crc[c]};
+                        d = comb.combine(d, crc32_update_bitwise(0, cb2, 4), 4);
+                        l += cs[c] + 4;
+                    }
+                    pcrc[t] = d;
+                    plen[t] = l;
+                });
+            }
+            for (auto& x : th) x.join();
+            uint32_t dg = 0;
+            for (int t = 0; t < nth; t++) dg = comb.combine(dg, pcrc[t], plen[t]);
+            return dg;
+        });
+
+        std::future<bytes> fut_summary;
+        if (!bti_out)
+            fut_summary = std::async(std::launch::async, [&]() -> bytes {
+            // IndexSummary at BASE_SAMPLING_LEVEL (IndexSummaryBuilder
+            // with empty Downsampling start points: entries for keys
+            // 0, 128, 256, ...), serialized per
+            // IndexSummary.IndexSummarySerializer.serialize — BE header,
+            // then the off-heap image in NATIVE (LE) order: rebased
+            // int offsets, then key bytes + LE u64 Index.db position;
+            // trailing first/last key with BE lengths
+            // (SSTableReader.saveSummary). Fixture-pinned by the
+            // oracle roundtrip of all four legacy_oa tables.
+            const uint32_t MIN_INTERVAL = 128;
+            bytes entries, s;
+            std::vector<uint32_t> offs;
+            const uint8_t *first_k = nullptr, *last_k = nullptr;
+            uint32_t first_kl = 0, last_kl = 0;
+            uint64_t q = 0, part_i = 0;
+            while (q < total_idx) {
+                uint64_t entry_off = q;
+                uint32_t klen = ((uint32_t)h_index[q] << 8) | h_index[q + 1];
+                const uint8_t* kp = h_index + q + 2;
+                q += 2 + klen;
+                if (!first_k) { first_k = kp; first_kl = klen; }
+                last_k = kp; last_kl = klen;
+                if (part_i % MIN_INTERVAL == 0) {
+                    offs.push_back((uint32_t)entries.size());
+                    entries.insert(entries.end(), kp, kp + klen);
+                    for (int b = 0; b < 8; b++) entries.push_back((uint8_t)(entry_off >> (8 * b)));
+                }
+                // skip position + promoted-index payload vints
+                auto skip_uvint = [&]() -> uint64_t {
+                    uint8_t f2 = h_index[q++];
+                    int extra = 0;
+                    uint8_t x = f2;
+                    while (x & 0x80) { extra++; x <<= 1; }
+                    uint64_t r2 = f2 & (uint8_t)(0xFFu >> extra);
+                    for (int i2 = 0; i2 < extra; i2++) r2 = (r2 << 8) | h_index[q++];
+                    return r2;
+                };
+                (void)skip_uvint();
+                q += skip_uvint();
+                part_i++;
+            }
+            uint32_t cnt = (uint32_t)offs.size();
+            put_be32(s, MIN_INTERVAL);
+            put_be32(s, cnt);
+            put_be64(s, 4ull * cnt + entries.size());
+            put_be32(s, 128);
+            put_be32(s, cnt);
+            for (uint32_t o : offs) {
+                uint32_t v = o + 4 * cnt;
+                for (int b = 0; b < 4; b++) s.push_back((uint8_t)(v >> (8 * b)));
+            }
+            s.insert(s.end(), entries.begin(), entries.end());
+            if (first_k) {
+                put_be32(s, first_kl);
+                s.insert(s.end(), first_k, first_k + first_kl);
+                put_be32(s, last_kl);
+                s.insert(s.end(), last_k, last_k + last_kl);
+            }
+            return s;
+            });
 
         std::vector<uint32_t> tombs;
         if (hst.tomb_count) {
@@ -1361,37 +1455,9 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
             }
         }
         TR("wsd: meta built");
-        static Crc32Combiner comb;
-        // Digest = CRC of the whole Data.db, folded from per-chunk CRCs.
-        // Thread-parallel: each worker folds a contiguous chunk range into
-        // (crc, len); the partials combine left-to-right (CRC concatenation
-        // is associative over (crc, len) pairs).
-        uint32_t digest = 0;
-        {
-            int nth = n_chunks > 4096 ? 16 : 1;
-            std::vector<uint32_t> pcrc(nth, 0);
-            std::vector<uint64_t> plen(nth, 0);
-            uint32_t per = (n_chunks + nth - 1) / nth;
-            std::vector<std::thread> th;
-            for (int t = 0; t < nth; t++) {
-                th.emplace_back([&, t]() {
-                    uint32_t c0 = t * per, c1 = std::min(n_chunks, (t + 1) * per);
-                    uint32_t d = 0;
-                    uint64_t l = 0;
-                    for (uint32_t c = c0; c < c1; c++) {
-                        d = comb.combine(d, h_crc[c], cs[c]);
-                        uint8_t cb2[4] = {(uint8_t)(h_crc[c] >> 24), (uint8_t)(h_crc[c] >> 16),
-                                          (uint8_t)(h_crc[c] >> 8), (uint8_t)h_crc[c]};
-                        d = comb.combine(d, crc32_update_bitwise(0, cb2, 4), 4);
-                        l += cs[c] + 4;
-                    }
-                    pcrc[t] = d;
-                    plen[t] = l;
-                });
-            }
-            for (auto& x : th) x.join();
-            for (int t = 0; t < nth; t++) digest = comb.combine(digest, pcrc[t], plen[t]);
-        }
+        // Digest = CRC of the whole Data.db, folded from per-chunk CRCs in
+        // the async above (CRC concatenation is associative over (crc, len))
+        uint32_t digest = fut_digest.get();
 
         TR("wsd: digest done");
         struct timespec ts0, ts1;
@@ -1441,64 +1507,7 @@ static WriteDeviceOut write_sstable_device(const OutPartsBuf& opb, const UnfCols
         }
         {
             if (!bti_out) {
-                // IndexSummary at BASE_SAMPLING_LEVEL (IndexSummaryBuilder
-                // with empty Downsampling start points: entries for keys
-                // 0, 128, 256, ...), serialized per
-                // IndexSummary.IndexSummarySerializer.serialize — BE header,
-                // then the off-heap image in NATIVE (LE) order: rebased
-                // int offsets, then key bytes + LE u64 Index.db position;
-                // trailing first/last key with BE lengths
-                // (SSTableReader.saveSummary). Fixture-pinned by the
-                // oracle roundtrip of all four legacy_oa tables.
-                const uint32_t MIN_INTERVAL = 128;
-                bytes entries, s;
-                std::vector<uint32_t> offs;
-                const uint8_t *first_k = nullptr, *last_k = nullptr;
-                uint32_t first_kl = 0, last_kl = 0;
-                uint64_t q = 0, part_i = 0;
-                while (q < total_idx) {
-                    uint64_t entry_off = q;
-                    uint32_t klen = ((uint32_t)h_index[q] << 8) | h_index[q + 1];
-                    const uint8_t* kp = h_index + q + 2;
-                    q += 2 + klen;
-                    if (!first_k) { first_k = kp; first_kl = klen; }
-                    last_k = kp; last_kl = klen;
-                    if (part_i % MIN_INTERVAL == 0) {
-                        offs.push_back((uint32_t)entries.size());
-                        entries.insert(entries.end(), kp, kp + klen);
-                        for (int b = 0; b < 8; b++) entries.push_back((uint8_t)(entry_off >> (8 * b)));
-                    }
-                    // skip position + promoted-index payload vints
-                    auto skip_uvint = [&]() -> uint64_t {
-                        uint8_t f2 = h_index[q++];
-                        int extra = 0;
-                        uint8_t x = f2;
-                        while (x & 0x80) { extra++; x <<= 1; }
-                        uint64_t r2 = f2 & (uint8_t)(0xFFu >> extra);
-                        for (int i2 = 0; i2 < extra; i2++) r2 = (r2 << 8) | h_index[q++];
-                        return r2;
-                    };
-                    (void)skip_uvint();
-                    q += skip_uvint();
-                    part_i++;
-                }
-                uint32_t cnt = (uint32_t)offs.size();
-                put_be32(s, MIN_INTERVAL);
-                put_be32(s, cnt);
-                put_be64(s, 4ull * cnt + entries.size());
-                put_be32(s, 128);
-                put_be32(s, cnt);
-                for (uint32_t o : offs) {
-                    uint32_t v = o + 4 * cnt;
-                    for (int b = 0; b < 4; b++) s.push_back((uint8_t)(v >> (8 * b)));
-                }
-                s.insert(s.end(), entries.begin(), entries.end());
-                if (first_k) {
-                    put_be32(s, first_kl);
-                    s.insert(s.end(), first_k, first_k + first_kl);
-                    put_be32(s, last_kl);
-                    s.insert(s.end(), last_k, last_k + last_kl);
-                }
+                bytes s = fut_summary.get();
                 write_file(out_base + "-Summary.db", s.data(), s.size());
             }
             std::string toc = bti_out
