@@ -377,6 +377,15 @@ int main(int argc, char** argv) {
         if (cmd == "dump") return cmd_dump(pos.at(0));
         if (cmd == "roundtrip") return cmd_roundtrip(pos.at(0));
         if (cmd == "rewrite") return cmd_rewrite(pos.at(0), pos.at(1));
+        if (cmd == "dumpsst") {
+            // memdump of ANY existing sstable (gen writes dumps only for its
+            // own tables; this lets compact outputs — pdel-only partitions,
+            // counter tables, purged shapes — round-trip through the
+            // gpuc_flush_table parity harness)
+            SSTable t = read_sstable(pos.at(0), false);
+            write_memdump(t, pos.at(1));
+            return 0;
+        }
         if (cmd == "validate") {
             // validate <outfile> <inputs...> [now= gcbefore= ...]: the
             // VALIDATION compaction epilogue (CompactionManager.doValidation

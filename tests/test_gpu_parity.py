@@ -921,6 +921,24 @@ def test_flush_table_full_schema(ca, oracle_bin, tmp_path):
         stem = "da-1-bti" if genkw.get("bti") else "oa-1-big"
         ca.flush_table(f"{d}/{stem}.memdump", f"{d}/out")
         _assert_dirs_equal(f"{d}/out", f"{d}/{stem}", comps)
+    # dumpsst round-trip: ANY written sstable (here a compact OUTPUT with
+    # pdel-only partitions — shapes the generator never emits — and a
+    # compacted counter table) re-flushes byte-identically
+    for tag, genkw, job in [
+        ("pdel", dict(seed=31, n=2, rows=300, vlen=60, overlap=30, tomb=40, pdel=8),
+         ["nevergc=1", "gcbefore=2000000000"]),
+        ("rectr", dict(seed=32, n=2, rows=400, overlap=30, tomb=20, counter=1), []),
+    ]:
+        dr = os.path.join(str(tmp_path), "rt_" + tag)
+        os.makedirs(dr)
+        _oracle_gen(dr, **genkw)
+        subprocess.run([ORACLE, "compact", f"{dr}/oa-90-big",
+                        *[f"{dr}/oa-{g}-big" for g in (1, 2)], *job],
+                       check=True, capture_output=True)
+        subprocess.run([ORACLE, "dumpsst", f"{dr}/oa-90-big", f"{dr}/rt.memdump"],
+                       check=True, capture_output=True)
+        ca.flush_table(f"{dr}/rt.memdump", f"{dr}/oa-60-big")
+        _assert_dirs_equal(f"{dr}/oa-60-big", f"{dr}/oa-90-big")
     # loud failures: out-of-order unfiltereds and duplicate keys
     d = os.path.join(str(tmp_path), "bad")
     os.makedirs(d)
