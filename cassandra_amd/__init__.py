@@ -291,6 +291,10 @@ class GpucFlushSchema(ctypes.Structure):
         ("column_index_size", ctypes.c_uint32),
         ("snappy", ctypes.c_uint32),
         ("bti", ctypes.c_uint32),
+        ("has_stats", ctypes.c_uint8),
+        ("stats_min_ts", ctypes.c_int64),
+        ("stats_min_ldt", ctypes.c_int64),
+        ("stats_min_ttl", ctypes.c_int32),
     ]
 
 
@@ -361,7 +365,7 @@ class _MemdumpReader:
 
 def _parse_memdump(data):
     r = _MemdumpReader(data)
-    if data[:4] != b"GMD1":
+    if data[:4] != b"GMD2":
         raise GpuCompactError("bad memdump magic")
     r.o = 4
     key_type = _CQL_TYPE_NAMES[r.u8()]
@@ -376,6 +380,7 @@ def _parse_memdump(data):
     snappy = r.u8()
     bti = r.u8()
     cis = r.u32()
+    stats_ts, stats_ldt, stats_ttl = r.i64(), r.i64(), r.i32()
     n_parts = r.u64()
 
     S = GpucFlushSchema()
@@ -401,6 +406,8 @@ def _parse_memdump(data):
     S.column_index_size = cis
     S.snappy = snappy
     S.bti = bti
+    S.has_stats = 1
+    S.stats_min_ts, S.stats_min_ldt, S.stats_min_ttl = stats_ts, stats_ldt, stats_ttl
     n_simple = n_cols - n_cpx
     cpx_idx = [i for i, (_, t) in enumerate(cols) if b"MapType" in t]
 

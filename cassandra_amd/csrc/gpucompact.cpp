@@ -3733,12 +3733,19 @@ extern "C" int gpuc_flush_table(const gpuc_flush_schema* schema, const gpuc_flus
         HIP_CHECK(hipStreamSynchronize(stream));
         HIP_CHECK(hipMemcpy(&hs0, d_stats.p, sizeof(OutStats), hipMemcpyDeviceToHost));
         SerParams2 sp{};
-        sp.hs.min_ts = hs0.min_ts_flip == 0xFFFFFFFFFFFFFFFFULL
-                           ? TIMESTAMP_EPOCH : (int64_t)(hs0.min_ts_flip ^ 0x8000000000000000ULL);
-        sp.hs.min_ldt = hs0.min_ldt_flip == 0xFFFFFFFFFFFFFFFFULL
-                            ? DELETION_TIME_EPOCH : (int64_t)(hs0.min_ldt_flip ^ 0x8000000000000000ULL);
-        if (sp.hs.min_ldt == NO_DELETION_TIME) sp.hs.min_ldt = DELETION_TIME_EPOCH;
-        sp.hs.min_ttl = hs0.min_ttl == 0xFFFFFFFFu ? 0 : (int32_t)hs0.min_ttl;
+        if (S.has_stats) {
+            // caller-provided EncodingStats (Memtable -> SerializationHeader.make)
+            sp.hs.min_ts = S.stats_min_ts;
+            sp.hs.min_ldt = S.stats_min_ldt;
+            sp.hs.min_ttl = S.stats_min_ttl;
+        } else {
+            sp.hs.min_ts = hs0.min_ts_flip == 0xFFFFFFFFFFFFFFFFULL
+                               ? TIMESTAMP_EPOCH : (int64_t)(hs0.min_ts_flip ^ 0x8000000000000000ULL);
+            sp.hs.min_ldt = hs0.min_ldt_flip == 0xFFFFFFFFFFFFFFFFULL
+                                ? DELETION_TIME_EPOCH : (int64_t)(hs0.min_ldt_flip ^ 0x8000000000000000ULL);
+            if (sp.hs.min_ldt == NO_DELETION_TIME) sp.hs.min_ldt = DELETION_TIME_EPOCH;
+            sp.hs.min_ttl = hs0.min_ttl == 0xFFFFFFFFu ? 0 : (int32_t)hs0.min_ttl;
+        }
         sp.sch.n_ck = S.n_ck;
         DevBuf d_ckw, d_cf2, d_sf2;
         d_ckw.alloc(ckw_h.size() * 4 + 8);

@@ -316,6 +316,13 @@ typedef struct gpuc_flush_schema {
     uint32_t column_index_size; /* 0 = 64 KiB */
     uint32_t snappy;            /* chunk codec for the output */
     uint32_t bti;               /* write the `da` (trie-indexed) components */
+    /* header EncodingStats, as the reference's memtable hands its collected
+     * stats to SerializationHeader.make; has_stats == 0 derives them from
+     * the rows instead */
+    uint8_t has_stats;
+    int64_t stats_min_ts;
+    int64_t stats_min_ldt;      /* long-seconds semantics */
+    int32_t stats_min_ttl;
 } gpuc_flush_schema;
 
 int gpuc_flush_table(const gpuc_flush_schema* schema, const gpuc_flush_part* parts,

@@ -1363,7 +1363,7 @@ void md_row_head(bytes& o, const Row& r) {
 
 void write_memdump(const SSTable& t, const std::string& path) {
     bytes o;
-    o.insert(o.end(), {'G', 'M', 'D', '1'});
+    o.insert(o.end(), {'G', 'M', 'D', '2'});
     md_u8(o, (uint8_t)t.header.key_type);
     md_u32(o, (uint32_t)t.header.clustering_types.size());
     for (CqlType ct : t.header.clustering_types) md_u8(o, (uint8_t)ct);
@@ -1374,6 +1374,13 @@ void write_memdump(const SSTable& t, const std::string& path) {
     md_u8(o, t.comp.algo == Compressor::SNAPPY ? 1 : 0);
     md_u8(o, t.bti ? 1 : 0);
     md_u32(o, t.column_index_size);
+    // header EncodingStats: a flush CALLER provides these (the reference's
+    // memtable passes its collected stats to SerializationHeader.make); a
+    // compaction output's header mins come from its INPUT headers and are
+    // not derivable from the data
+    md_i64(o, t.header.stats.min_ts);
+    md_i64(o, t.header.stats.min_ldt);
+    md_i32(o, t.header.stats.min_ttl);
     md_u64(o, t.parts.size());
     const auto& regs = t.header.regular_cols;
     for (const Partition& p : t.parts) {
