@@ -156,6 +156,15 @@ def main():
                 bad = dirs_equal(f"{d}/{stem}-90-{sfx}", f"{d}/{stem}-91-{sfx}", comps)
                 if not bad and not use_bti and rng.random() < 0.3:
                     ca.verify(f"{d}/{stem}-91-{sfx}")
+                if not bad and rng.random() < 0.25:
+                    # flush round-trip of the compact OUTPUT: dumpsst ->
+                    # gpuc_flush_table must reproduce it byte-for-byte
+                    subprocess.run([ORACLE, "dumpsst", f"{d}/{stem}-90-{sfx}",
+                                    f"{d}/rt.memdump"], check=True, capture_output=True)
+                    ca.flush_table(f"{d}/rt.memdump", f"{d}/{stem}-95-{sfx}")
+                    bad = dirs_equal(f"{d}/{stem}-95-{sfx}", f"{d}/{stem}-90-{sfx}", comps)
+                    if bad:
+                        bad = f"flush-roundtrip {bad}"
                 if not bad and rng.random() < 0.3:
                     # validation compaction on the same inputs
                     vkw = [f"now={job['now_sec']}"] if "now_sec" in job else []
