@@ -2004,7 +2004,10 @@ static void compact_one(const gpuc_job* job, const CompactSetup& su,
             d_ctr_arena.alloc(ctr_arena_cap);
             sch.ctr_arena = d_ctr_arena.as<uint8_t>();
         }
-        uint32_t tomb_cap = (uint32_t)std::min<uint64_t>(total_out_rows * 2 + n_groups + 1024, 400000000ull);
+        uint32_t tomb_cap = (uint32_t)std::min<uint64_t>(
+            total_out_rows * ((uint64_t)sch.n_cols + sch.n_cpx + 2) + total_in_cpx +
+                n_groups * ((uint64_t)sch.n_static + 2) + 1024,
+            400000000ull);
         d_tomb.alloc((uint64_t)tomb_cap * 4);
         DevBuf d_ov_lo, d_ov_hi, d_ov_ts;
         PurgeParams2 pp{};
@@ -3136,7 +3139,9 @@ extern "C" int gpuc_scrub(const char* input_base, const char* output_base, int32
         out_rows.alloc(out_total, sch.n_cols, sch.n_ck);
         d_stats.alloc(sizeof(OutStats));
         init_outstats(d_stats, stream);
-        uint32_t tomb_cap = (uint32_t)std::min<uint64_t>(out_total * 2 + n_groups + 1024, 400000000ull);
+        uint32_t tomb_cap = (uint32_t)std::min<uint64_t>(
+            out_total * ((uint64_t)sch.n_cols + 2) + n_groups * ((uint64_t)sch.n_static + 2) + 1024,
+            400000000ull);
         d_tomb.alloc((uint64_t)tomb_cap * 4);
         PurgeParams2 pp{};
         pp.gc_before = INT64_MIN;
@@ -3259,7 +3264,7 @@ extern "C" int gpuc_flush(const gpuc_flush_rows* rows, const char* output_base,
         DevBuf d_stats, d_tomb;
         d_stats.alloc(sizeof(OutStats));
         init_outstats(d_stats, stream);
-        uint32_t tomb_cap = (uint32_t)std::min<uint64_t>(n * 2 + 1024, 400000000ull);
+        uint32_t tomb_cap = (uint32_t)std::min<uint64_t>(n * 3 + 1024, 400000000ull);
         d_tomb.alloc((uint64_t)tomb_cap * 4);
         hipLaunchKernelGGL(k_collect_rows, dim3(blocks), dim3(256), 0, stream, opb.op,
                            urows.uc, n, 1u, d_stats.as<OutStats>(), d_tomb.as<uint32_t>(), tomb_cap);
@@ -3724,7 +3729,10 @@ extern "C" int gpuc_flush_table(const gpuc_flush_schema* schema, const gpuc_flus
         DevBuf d_stats, d_tomb;
         d_stats.alloc(sizeof(OutStats));
         init_outstats(d_stats, stream);
-        uint32_t tomb_cap = (uint32_t)std::min<uint64_t>(n_unf * 2 + cpx_total + 1024, 400000000ull);
+        uint32_t tomb_cap = (uint32_t)std::min<uint64_t>(
+            n_unf * ((uint64_t)NSIMPLE + S.n_cpx + 2) + cpx_total +
+                n_parts * ((uint64_t)S.n_static + 2) + 1024,
+            400000000ull);
         d_tomb.alloc((uint64_t)tomb_cap * 4);
         uint32_t blocks = (uint32_t)((n_parts + 255) / 256);
         hipLaunchKernelGGL(k_collect_rows, dim3(blocks), dim3(256), 0, stream, opb.op, urows.uc,
@@ -3896,7 +3904,10 @@ extern "C" int gpuc_generate(const gpuc_gen_spec* spec, const char* dir, char* e
                                d_prows.as<uint64_t>());
             d_stats.alloc(sizeof(OutStats));
             init_outstats(d_stats, stream);
-            uint32_t tomb_cap = (uint32_t)std::min<uint64_t>(total_rows * 2 + R + 1024, 400000000ull);
+            uint32_t tomb_cap = (uint32_t)std::min<uint64_t>(
+                total_rows * ((uint64_t)gp.n_value_cols + (gp.complex_pct ? 7 : 0) + 2) +
+                    R * 3 + 1024,
+                400000000ull);
             d_tomb.alloc((uint64_t)tomb_cap * 4);
             {
                 hipLaunchKernelGGL(k_collect_rows, dim3(blocks), dim3(256), 0, stream, opb.op,
