@@ -437,6 +437,30 @@ def test_scrub_parity(ca, oracle_bin, tmp_path):
     kept, dropped = ca.scrub(f"{d}/c/oa-1-big", f"{d}/c/oa-81-big")
     assert dropped == 0 and kept > 0
     ca.verify(f"{d}/c/oa-81-big")
+    # round-2 schemas through scrub: complex, counter and TTL sstables
+    for tag, genkw in [("x", dict(seed=174, n=1, rows=900, vlen=80, overlap=0,
+                                  tomb=10, cpx=45, cpxdel=25)),
+                       ("k", dict(seed=175, n=1, rows=900, overlap=0, tomb=10,
+                                  counter=1)),
+                       ("t", dict(seed=176, n=1, rows=900, vlen=80, overlap=0,
+                                  ttl=40))]:
+        dd = f"{d}/{tag}"
+        os.makedirs(dd)
+        _oracle_gen(dd, **genkw)
+        base = f"{dd}/oa-1-big"
+        with open(base + "-Data.db", "r+b") as f:
+            f.seek(os.path.getsize(base + "-Data.db") // 3)
+            b0 = f.read(1)
+            f.seek(-1, 1)
+            f.write(bytes([b0[0] ^ 0x10]))
+        out = subprocess.run([ORACLE, "scrub", f"{dd}/oa-80-big", base],
+                             capture_output=True, text=True, check=True)
+        ores = json.loads(out.stdout.splitlines()[-1])
+        kept, dropped = ca.scrub(base, f"{dd}/oa-81-big")
+        assert dropped > 0, tag
+        assert (kept, dropped) == (ores["partitions_kept"], ores["partitions_dropped"]), tag
+        _assert_dirs_equal(f"{dd}/oa-80-big", f"{dd}/oa-81-big")
+        ca.verify(f"{dd}/oa-81-big")
 
 
 def test_compaction_associativity_gpu(ca, oracle_bin, tmp_path):
