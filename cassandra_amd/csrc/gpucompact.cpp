@@ -3027,7 +3027,10 @@ extern "C" int gpuc_scrub(const char* input_base, const char* output_base, int32
         for (auto& [nm, ct] : st.regular_cols) {
             (void)nm;
             if (ct == "org.apache.cassandra.db.marshal.MapType(org.apache.cassandra.db.marshal.BytesType,org.apache.cassandra.db.marshal.BytesType)") sch.n_cpx = 1;  // last regular column
-            else cfh.push_back(ck_type_width(ct));
+            else if (ct == "org.apache.cassandra.db.marshal.CounterColumnType") {
+                cfh.push_back(-1);
+                sch.counters = 1;
+            } else cfh.push_back(ck_type_width(ct));
         }
         for (auto& ct : st.clustering_types) vckw.push_back(ck_type_width(ct));
         for (auto& [nm, ct] : st.static_cols) { (void)nm; vsf.push_back(ck_type_width(ct)); }
@@ -3090,10 +3093,28 @@ extern "C" int gpuc_scrub(const char* input_base, const char* output_base, int32
         uint64_t total_rows = n_kept ? exscan_u64(pc.row_base, n_kept, stream) : 0;
         UnfColsBuf in_rows;
         in_rows.alloc(total_rows, sch.n_cols, sch.n_ck, sch.n_cpx);
+        DevBuf p_cpxtot, p_cpxbase;
+        uint64_t total_in_cpx = 0;
+        if (n_kept && sch.n_cpx) {
+            // complex-cell counting pass, then arena alloc (compact's B1/B2)
+            p_cpxtot.alloc(n_kept * 4 + 8);
+            pc.cpx_total = p_cpxtot.as<uint32_t>();
+            hipLaunchKernelGGL(k_parse_rows, dim3(blocks), dim3(256), 0, stream,
+                               d_src.as<SrcDesc2>(), 1u, (uint32_t)n_kept, pc, in_rows.uc, sch,
+                               d_error.as<unsigned long long>(),
+                               d_rows_in.as<unsigned long long>(), nullptr);
+            HIP_CHECK(hipMemsetAsync(d_rows_in.p, 0, 8, stream));
+            p_cpxbase.alloc(n_kept * 8 + 8);
+            hipLaunchKernelGGL(k_widen_u32, dim3(blocks), dim3(256), 0, stream, pc.cpx_total,
+                               p_cpxbase.as<uint64_t>(), n_kept);
+            total_in_cpx = exscan_u64(p_cpxbase.as<uint64_t>(), n_kept, stream);
+            in_rows.alloc_cpx_arena(total_in_cpx);
+        }
         if (n_kept)
             hipLaunchKernelGGL(k_parse_rows, dim3(blocks), dim3(256), 0, stream,
                                d_src.as<SrcDesc2>(), 1u, (uint32_t)n_kept, pc, in_rows.uc, sch,
-                               d_error.as<unsigned long long>(), d_rows_in.as<unsigned long long>());
+                               d_error.as<unsigned long long>(), d_rows_in.as<unsigned long long>(),
+                               sch.n_cpx ? p_cpxbase.as<uint64_t>() : nullptr);
         {
             unsigned long long e = 0;
             HIP_CHECK(hipStreamSynchronize(stream));
@@ -3136,11 +3157,32 @@ extern "C" int gpuc_scrub(const char* input_base, const char* output_base, int32
         OutPartsBuf opb;
         opb.alloc(n_groups ? n_groups : 1, sch.n_static);
         UnfColsBuf out_rows;
-        out_rows.alloc(out_total, sch.n_cols, sch.n_ck);
+        out_rows.alloc(out_total, sch.n_cols, sch.n_ck, sch.n_cpx);
+        if (sch.n_cpx) out_rows.alloc_cpx_arena(total_in_cpx);
+        DevBuf d_ctr_arena_s;
+        if (sch.counters) {
+            // merged-context arena (arity-1 reconcile still routes counter
+            // cells through the context branch)
+            DevBuf d_sum;
+            d_sum.alloc(8);
+            HIP_CHECK(hipMemsetAsync(d_sum.p, 0, 8, stream));
+            uint64_t n_cells = total_rows * sch.n_cols;
+            if (n_cells) {
+                uint32_t blocks2 = (uint32_t)((n_cells + 255) / 256);
+                hipLaunchKernelGGL(k_sum_vallen, dim3(blocks2), dim3(256), 0, stream,
+                                   in_rows.uc, n_cells, d_sum.as<unsigned long long>());
+            }
+            unsigned long long total_ctr = 0;
+            HIP_CHECK(hipStreamSynchronize(stream));
+            HIP_CHECK(hipMemcpy(&total_ctr, d_sum.p, 8, hipMemcpyDeviceToHost));
+            d_ctr_arena_s.alloc(total_ctr + 64);
+            sch.ctr_arena = d_ctr_arena_s.as<uint8_t>();
+        }
         d_stats.alloc(sizeof(OutStats));
         init_outstats(d_stats, stream);
         uint32_t tomb_cap = (uint32_t)std::min<uint64_t>(
-            out_total * ((uint64_t)sch.n_cols + 2) + n_groups * ((uint64_t)sch.n_static + 2) + 1024,
+            out_total * ((uint64_t)sch.n_cols + sch.n_cpx + 2) + total_in_cpx +
+                n_groups * ((uint64_t)sch.n_static + 2) + 1024,
             400000000ull);
         d_tomb.alloc((uint64_t)tomb_cap * 4);
         PurgeParams2 pp{};
