@@ -584,3 +584,29 @@ def test_memdump_roundtrip_parse(oracle_bin, tmp_path):
         # spot-check the first partition graph is materialized
         assert parts[0].key_len > 0
         del bufs
+
+
+def test_dumpsst_reference_fixtures(oracle_bin, tmp_path):
+    """dumpsst over the REFERENCE's own committed sstables (including the
+    clustering fixture written with 4 KiB column_index_size and the counter
+    fixtures): the dump parses fully and reconstructs the schema."""
+    import cassandra_amd as ca
+    cases = [
+        ("legacy_oa_simple", dict(n_ck=0)),
+        ("legacy_oa_clust", dict(n_ck=1)),
+        ("legacy_oa_simple_counter", dict(n_ck=0)),
+        ("legacy_oa_clust_counter", dict(n_ck=1)),
+    ]
+    for name, want in cases:
+        base = os.path.join(GOLDEN, name, "oa-1-big")
+        if not os.path.exists(base + "-Data.db"):
+            continue
+        out = os.path.join(str(tmp_path), name + ".memdump")
+        subprocess.run([ORACLE, "dumpsst", base, out], check=True, capture_output=True)
+        S, parts, n, bufs = ca._parse_memdump(open(out, "rb").read())
+        assert n > 0
+        for k, v in want.items():
+            assert getattr(S, k) == v, (name, k)
+        if "counter" in name:
+            assert b"CounterColumnType" in bytes(S.col_types[0])
+        del bufs
