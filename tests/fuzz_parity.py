@@ -165,6 +165,28 @@ def main():
                     bad = dirs_equal(f"{d}/{stem}-95-{sfx}", f"{d}/{stem}-90-{sfx}", comps)
                     if bad:
                         bad = f"flush-roundtrip {bad}"
+                if not bad and not use_bti and rng.random() < 0.2:
+                    # scrub leg: corrupt one byte of the compact output's
+                    # Data.db, scrub on GPU and in the oracle, byte-compare
+                    import json as _json
+                    sb = f"{d}/{stem}-90-{sfx}"
+                    sz = os.path.getsize(sb + "-Data.db")
+                    if sz > 256:
+                        with open(sb + "-Data.db", "r+b") as fh:
+                            fh.seek(sz // 2)
+                            b0 = fh.read(1)
+                            fh.seek(-1, 1)
+                            fh.write(bytes([b0[0] ^ 0x10]))
+                        out2 = subprocess.run([ORACLE, "scrub", f"{d}/{stem}-96-{sfx}", sb],
+                                              check=True, capture_output=True, text=True)
+                        ores = _json.loads(out2.stdout.splitlines()[-1])
+                        kept, dropped = ca.scrub(sb, f"{d}/{stem}-97-{sfx}")
+                        if (kept, dropped) != (ores["partitions_kept"], ores["partitions_dropped"]):
+                            bad = f"scrub-counts {kept},{dropped} vs {ores}"
+                        else:
+                            bad = dirs_equal(f"{d}/{stem}-97-{sfx}", f"{d}/{stem}-96-{sfx}", comps)
+                            if bad:
+                                bad = f"scrub {bad}"
                 if not bad and rng.random() < 0.3:
                     # validation compaction on the same inputs
                     vkw = [f"now={job['now_sec']}"] if "now_sec" in job else []
