@@ -1241,10 +1241,7 @@ ScrubResult scrub_sstable(const std::string& inbase, const std::string& outbase)
     SSTable t;
     t.generation = 1;
     t.comp = ci.params;
-    t.header = sf.header;
-    t.header.stats.min_ts = sf.mins.min_timestamp == INT64_MIN ? TIMESTAMP_EPOCH : sf.mins.min_timestamp;
-    t.header.stats.min_ldt = sf.mins.min_ldt == INT64_MAX ? DELETION_TIME_EPOCH : sf.mins.min_ldt;
-    t.header.stats.min_ttl = sf.mins.min_ttl == INT32_MAX ? 0 : sf.mins.min_ttl;
+    t.header = sf.header;  // rows DECODE with the input's HEADER mins
     uint64_t dropped = 0;
     for (size_t i = 0; i + 1 < positions.size(); i++) {
         uint64_t c0 = positions[i] / ci.params.chunk_len;
@@ -1260,6 +1257,14 @@ ScrubResult scrub_sstable(const std::string& inbase, const std::string& outbase)
                       ? murmur3_token(p.key.data(), p.key.size()) : 0;
         t.parts.push_back(std::move(p));
     }
+    // the REWRITE declares StatsMetadata mins as its header mins (what a
+    // 1-input compaction would take) — set them only AFTER parsing: on a
+    // compaction OUTPUT they can differ from the input's header mins, and
+    // decoding with the writer's mins shifted every delta by a constant
+    // (fuzz-caught: scrub of a TTL compaction output)
+    t.header.stats.min_ts = sf.mins.min_timestamp == INT64_MIN ? TIMESTAMP_EPOCH : sf.mins.min_timestamp;
+    t.header.stats.min_ldt = sf.mins.min_ldt == INT64_MAX ? DELETION_TIME_EPOCH : sf.mins.min_ldt;
+    t.header.stats.min_ttl = sf.mins.min_ttl == INT32_MAX ? 0 : sf.mins.min_ttl;
     WriterOut w = write_sstable(t);
     write_components(w, outbase);
     ScrubResult sr;
